@@ -1,0 +1,139 @@
+"""Op dispatch for arks_amd.
+
+On GPU tensors every op runs the in-tree gfx950 HIP extension
+(``arks_amd._C``) — there is no eager fallback: if the extension is missing
+on a CUDA/ROCm device we raise immediately so a silently-slow PyTorch path
+can never masquerade as the native one. On CPU tensors ops run the plain
+PyTorch reference (arks_amd/ops/ref.py) so the engine/scheduler/model layers
+are unit-testable without a GPU.
+"""
+
+from __future__ import annotations
+
+import torch
+
+from . import ref
+
+_C = None
+_C_ERR: str | None = None
+
+
+def _load_extension():
+    global _C, _C_ERR
+    if _C is not None or _C_ERR is not None:
+        return _C
+    try:
+        from . import _load  # noqa: F401  (imports the built .so in-tree)
+
+        _C = _load.C
+    except Exception as e:  # pragma: no cover - exercised only when unbuilt
+        _C_ERR = f"{type(e).__name__}: {e}"
+    return _C
+
+
+def native_available() -> bool:
+    return _load_extension() is not None
+
+
+def _native():
+    c = _load_extension()
+    if c is None:
+        raise RuntimeError(
+            "arks_amd HIP extension (arks_amd._C) is not built but a GPU op was "
+            "requested. Build it with `python -m arks_amd.ops.build` "
+            f"(import error: {_C_ERR})"
+        )
+    return c
+
+
+def rmsnorm(x: torch.Tensor, weight: torch.Tensor, eps: float = 1e-6) -> torch.Tensor:
+    if x.is_cuda:
+        out = torch.empty_like(x)
+        _native().rmsnorm(out, x, weight, eps)
+        return out
+    return ref.rmsnorm(x, weight, eps)
+
+
+def fused_add_rmsnorm(x, residual, weight, eps: float = 1e-6):
+    """In-place on GPU: residual += x (written to residual), out = norm(residual)."""
+    if x.is_cuda:
+        out = torch.empty_like(x)
+        _native().fused_add_rmsnorm(out, x, residual, weight, eps)
+        return out, residual
+    return ref.fused_add_rmsnorm(x, residual, weight, eps)
+
+
+def silu_mul(gate_up: torch.Tensor) -> torch.Tensor:
+    if gate_up.is_cuda:
+        shape = list(gate_up.shape)
+        shape[-1] //= 2
+        out = gate_up.new_empty(shape)
+        _native().silu_mul(out, gate_up)
+        return out
+    return ref.silu_mul(gate_up)
+
+
+def rope_apply_inplace(positions, q, k, cos_sin, head_dim: int):
+    """Rotates q and k in place (GPU). CPU path returns new tensors."""
+    if q.is_cuda:
+        _native().rope_inplace(positions, q, k, cos_sin, head_dim)
+        return q, k
+    return ref.rope_apply(positions, q, k, cos_sin, head_dim)
+
+
+def reshape_and_cache(k, v, k_cache, v_cache, slot_mapping):
+    if k.is_cuda:
+        _native().reshape_and_cache(k, v, k_cache, v_cache, slot_mapping)
+        return
+    ref.reshape_and_cache(k, v, k_cache, v_cache, slot_mapping)
+
+
+PREFILL_QTILE = 64  # q rows per workgroup in the prefill kernel
+
+
+def build_prefill_tiles(seq_lens: list[int], device) -> torch.Tensor:
+    """int32 [ntiles, 2] of (seq_idx, q0) covering every sequence in 64-row
+    tiles — the prefill kernel's grid."""
+    tiles = []
+    for i, n in enumerate(seq_lens):
+        for q0 in range(0, n, PREFILL_QTILE):
+            tiles.append((i, q0))
+    return torch.tensor(tiles, dtype=torch.int32, device=device).reshape(-1, 2)
+
+
+def attention_prefill_varlen(q, k, v, cu_seqlens, seq_lens: list[int], scale: float):
+    """Causal varlen prefill attention. seq_lens is the host-side list of
+    per-sequence lengths (used to build the q-tile grid without a D2H sync)."""
+    if q.is_cuda:
+        out = torch.empty_like(q)
+        tile_info = build_prefill_tiles(seq_lens, q.device)
+        _native().attention_prefill_varlen(out, q, k, v, cu_seqlens, tile_info, scale)
+        return out
+    return ref.attention_prefill_varlen(q, k, v, cu_seqlens, scale)
+
+
+def attention_decode_paged(q, k_cache, v_cache, block_tables, seq_lens, scale: float):
+    if q.is_cuda:
+        out = torch.empty_like(q)
+        _native().attention_decode_paged(
+            out, q, k_cache, v_cache, block_tables, seq_lens, scale
+        )
+        return out
+    return ref.attention_decode_paged(q, k_cache, v_cache, block_tables, seq_lens, scale)
+
+
+def sample_tokens(logits, temperatures, uniform):
+    """Gumbel-max categorical sampling; rows with temperature 0 are greedy."""
+    if logits.is_cuda:
+        out = torch.empty(logits.shape[0], dtype=torch.int64, device=logits.device)
+        _native().gumbel_sample(out, logits, temperatures, uniform)
+        return out
+    return ref.gumbel_sample(logits, temperatures, uniform)
+
+
+def greedy_sample(logits):
+    if logits.is_cuda:
+        out = torch.empty(logits.shape[0], dtype=torch.int64, device=logits.device)
+        _native().greedy_sample(out, logits)
+        return out
+    return ref.greedy_sample(logits)

@@ -1,0 +1,292 @@
+// Flash-style varlen causal prefill attention on MFMA (gfx950).
+//
+// Grid: (num_q_heads, num_q_tiles). A workgroup = 256 threads = 4 waves
+// handles one 64-row Q tile of one sequence for one q-head; each wave owns 16
+// Q rows. K/V are iterated in 32-key tiles staged in LDS (K XOR-swizzled
+// against the D=128 row-major bank conflict — guide §6 G4; V stored
+// transposed so the PV B-fragment is a contiguous ds_read_b128).
+//
+// QK^T uses swapped operands (A = K-tile, B = Q-tile) so each lane's 16x16x32
+// C-fragment holds scores for ONE q-row (col = lane&15) — the online-softmax
+// row reduction is then 8 regs + 2 shuffles, never a serial-lane loop
+// (guide common-mistake #6). P is staged through a per-wave padded LDS buffer
+// to re-shape for the PV A-fragment.
+//
+// This kernel is the prefill half of the runtime-slot contract described in
+// SURVEY.md §2.4 (vLLM's flash prefill at reference
+// arksapplication_controller.go:941-955 delegated it to an external image).
+#include "common.h"
+
+#include <cfloat>
+
+namespace arks {
+
+// gfx950 MFMA fragment types.
+typedef __attribute__((ext_vector_type(8))) __bf16 bf16x8;
+typedef __attribute__((ext_vector_type(4))) float f32x4;
+
+__device__ __forceinline__ f32x4 mfma16x16x32(bf16x8 a, bf16x8 b, f32x4 c) {
+  return __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, c, 0, 0, 0);
+}
+
+// A/B fragment k-index for lane quarter `a` (= lane/16), element jj in 0..7.
+// Layout hypothesis H1: contiguous 8 elements per lane. Verified on hardware
+// by tests/test_ops_gpu.py::test_mfma_probe.
+__device__ __forceinline__ int frag_k(int a, int jj) { return 8 * a + jj; }
+
+constexpr int QTILE_WAVE = 16;  // q rows per wave
+constexpr int NUM_WAVES = 4;
+constexpr int QTILE = QTILE_WAVE * NUM_WAVES;  // 64 q rows per workgroup
+constexpr int KTILE = 32;                      // keys per LDS tile
+constexpr int VT_PAD = 40;                     // padded VT row length (keys)
+
+template <int HEAD_DIM>
+__global__ __launch_bounds__(256) void attn_prefill_kernel(
+    bf16* __restrict__ out,      // [T, Hq, D]
+    const bf16* __restrict__ q,  // [T, Hq, D]
+    const bf16* __restrict__ k,  // [T, Hkv, D]
+    const bf16* __restrict__ v,  // [T, Hkv, D]
+    const int* __restrict__ cu_seqlens,  // [num_seqs + 1]
+    const int* __restrict__ tile_info,   // [ntiles, 2] = (seq_idx, q0)
+    const float scale, const int num_q_heads, const int num_kv_heads) {
+  constexpr int CHUNKS = HEAD_DIM / 16;  // dim chunks for PV output
+  constexpr int STEPS = HEAD_DIM / 32;   // K-contraction steps for QK^T
+
+  const int h = blockIdx.x;
+  const int kvh = h / (num_q_heads / num_kv_heads);
+  const int seq_idx = tile_info[blockIdx.y * 2];
+  const int q0 = tile_info[blockIdx.y * 2 + 1];
+  const int seq_start = cu_seqlens[seq_idx];
+  const int seq_len = cu_seqlens[seq_idx + 1] - seq_start;
+
+  const int tid = threadIdx.x;
+  const int wave = tid / WAVE_SIZE;
+  const int lane = tid % WAVE_SIZE;
+  const int lq = lane % 16;   // q-row (QK^T) / dim-col (PV) within fragment
+  const int la = lane / 16;   // lane quarter
+
+  // LDS: K tile (XOR-swizzled rows), transposed V tile (padded), per-wave P.
+  // 16-B alignment required for the ushort8 (b128) accesses — guide §6 G17.
+  __shared__ __attribute__((aligned(16))) bf16 k_lds[KTILE][HEAD_DIM];
+  __shared__ __attribute__((aligned(16))) bf16 vt_lds[HEAD_DIM][VT_PAD];
+  __shared__ __attribute__((aligned(16))) bf16 p_lds[NUM_WAVES][QTILE_WAVE][VT_PAD];
+
+  // --- Q fragments: registers, loaded once. Wave w covers rows q0+16w..+15.
+  const int my_qrow = q0 + wave * QTILE_WAVE + lq;  // row in sequence
+  const bool qrow_valid = my_qrow < seq_len;
+  bf16x8 qfrag[STEPS];
+  {
+    const int64_t qbase =
+        ((int64_t)(seq_start + (qrow_valid ? my_qrow : 0)) * num_q_heads + h) *
+        HEAD_DIM;
+#pragma unroll
+    for (int st = 0; st < STEPS; ++st) {
+      ushort8 u = *reinterpret_cast<const ushort8*>(q + qbase + st * 32 +
+                                                    frag_k(la, 0));
+      qfrag[st] = *reinterpret_cast<bf16x8*>(&u);
+    }
+  }
+
+  // Online softmax state: per lane for q-row `my_qrow` (replicated over the 4
+  // lanes sharing lq).
+  float m_run = -FLT_MAX;
+  float l_run = 0.f;
+  f32x4 oacc[CHUNKS];
+#pragma unroll
+  for (int c = 0; c < CHUNKS; ++c) oacc[c] = {0.f, 0.f, 0.f, 0.f};
+
+  // Keys visible to this workgroup: strictly below kmax.
+  const int kmax = min(seq_len, q0 + QTILE);
+  const int ntiles = (kmax + KTILE - 1) / KTILE;
+
+  for (int j = 0; j < ntiles; ++j) {
+    const int key_base = j * KTILE;
+    // ---- Cooperative staging: K (swizzled) and V^T. 512 vec8 elements.
+    {
+      const int nvec = KTILE * HEAD_DIM / 8;
+      for (int i = tid; i < nvec; i += 256) {
+        const int key = i / (HEAD_DIM / 8);
+        const int col8 = (i % (HEAD_DIM / 8)) * 8;
+        const int kg = key_base + key;
+        ushort8 kv{}, vv{};
+        if (kg < kmax) {
+          const int64_t src =
+              ((int64_t)(seq_start + kg) * num_kv_heads + kvh) * HEAD_DIM + col8;
+          kv = *reinterpret_cast<const ushort8*>(k + src);
+          vv = *reinterpret_cast<const ushort8*>(v + src);
+        }
+        // K: swizzle byte offset within the 2*HEAD_DIM-byte row.
+        const int row_byte = col8 * 2;
+        const int swz = row_byte ^ ((key & 7) << 4);
+        *reinterpret_cast<ushort8*>(
+            reinterpret_cast<char*>(&k_lds[key][0]) + swz) = kv;
+        // V^T: scalar scatter (optimized via tr_b16 in a later revision).
+#pragma unroll
+        for (int e = 0; e < 8; ++e) vt_lds[col8 + e][key] = vv[e];
+      }
+    }
+    __syncthreads();
+
+    // ---- QK^T: two 16-key subtiles; A = K from LDS, B = Q registers.
+    f32x4 sc[2];
+    sc[0] = {0.f, 0.f, 0.f, 0.f};
+    sc[1] = {0.f, 0.f, 0.f, 0.f};
+#pragma unroll
+    for (int sub = 0; sub < 2; ++sub) {
+      const int key = sub * 16 + lq;
+#pragma unroll
+      for (int st = 0; st < STEPS; ++st) {
+        const int col_byte = (st * 32 + frag_k(la, 0)) * 2;
+        const int swz = col_byte ^ ((key & 7) << 4);
+        ushort8 u = *reinterpret_cast<const ushort8*>(
+            reinterpret_cast<const char*>(&k_lds[key][0]) + swz);
+        sc[sub] = mfma16x16x32(*reinterpret_cast<bf16x8*>(&u), qfrag[st], sc[sub]);
+      }
+    }
+
+    // ---- Masked online softmax. Lane holds 8 scores for q-row `my_qrow`,
+    // keys key_base + sub*16 + 4*la + r.
+    float p[8];
+    float tile_max = -FLT_MAX;
+    bool msk[8];
+#pragma unroll
+    for (int sub = 0; sub < 2; ++sub) {
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const int kg = key_base + sub * 16 + 4 * la + r;
+        const int i = sub * 4 + r;
+        msk[i] = qrow_valid && (kg <= my_qrow) && (kg < kmax);
+        p[i] = msk[i] ? sc[sub][r] * scale : -FLT_MAX;
+        tile_max = fmaxf(tile_max, p[i]);
+      }
+    }
+    // Row max across the 4 lanes sharing this q-row.
+    tile_max = fmaxf(tile_max, __shfl_xor(tile_max, 16, WAVE_SIZE));
+    tile_max = fmaxf(tile_max, __shfl_xor(tile_max, 32, WAVE_SIZE));
+
+    const float m_new = fmaxf(m_run, tile_max);
+    float alpha = 1.f;
+    float psum = 0.f;
+    if (m_new > -FLT_MAX) {
+      alpha = __expf(m_run - m_new);
+#pragma unroll
+      for (int i = 0; i < 8; ++i) {
+        p[i] = msk[i] ? __expf(p[i] - m_new) : 0.f;
+        psum += p[i];
+      }
+    } else {
+#pragma unroll
+      for (int i = 0; i < 8; ++i) p[i] = 0.f;
+    }
+    psum += __shfl_xor(psum, 16, WAVE_SIZE);
+    psum += __shfl_xor(psum, 32, WAVE_SIZE);
+    l_run = l_run * alpha + psum;
+    m_run = m_new;
+
+    // Rescale O accumulator. oacc rows are q-rows 4*la + r — fetch each row's
+    // alpha from the lane that owns it (lane 4*la + r has lq == that row).
+    float row_alpha[4];
+#pragma unroll
+    for (int r = 0; r < 4; ++r)
+      row_alpha[r] = __shfl(alpha, 4 * la + r, WAVE_SIZE);
+#pragma unroll
+    for (int c = 0; c < CHUNKS; ++c) {
+#pragma unroll
+      for (int r = 0; r < 4; ++r) oacc[c][r] *= row_alpha[r];
+    }
+
+    // ---- P to LDS (bf16), then PV.
+#pragma unroll
+    for (int sub = 0; sub < 2; ++sub) {
+      ushort4v pk;
+#pragma unroll
+      for (int r = 0; r < 4; ++r) pk[r] = float_to_bf16_bits(p[sub * 4 + r]);
+      *reinterpret_cast<ushort4v*>(&p_lds[wave][lq][sub * 16 + 4 * la]) = pk;
+    }
+    __syncthreads();  // P visible across lanes; also fences K/VT reuse below
+
+    // PV: one MFMA per 16-dim chunk contracts the full 32-key tile.
+    ushort8 pa =
+        *reinterpret_cast<const ushort8*>(&p_lds[wave][lq][frag_k(la, 0)]);
+#pragma unroll
+    for (int c = 0; c < CHUNKS; ++c) {
+      ushort8 vb =
+          *reinterpret_cast<const ushort8*>(&vt_lds[c * 16 + lq][frag_k(la, 0)]);
+      oacc[c] = mfma16x16x32(*reinterpret_cast<bf16x8*>(&pa),
+                             *reinterpret_cast<bf16x8*>(&vb), oacc[c]);
+    }
+    __syncthreads();  // all reads of K/VT/P done before next tile's staging
+  }
+
+  // ---- Epilogue: normalize and store. oacc[c][r] is q-row 4*la+r, dim
+  // c*16+lq. Each needed l_run lives in lane 4*la+r.
+  float row_inv[4];
+#pragma unroll
+  for (int r = 0; r < 4; ++r) {
+    const float lr = __shfl(l_run, 4 * la + r, WAVE_SIZE);
+    row_inv[r] = lr > 0.f ? 1.f / lr : 0.f;
+  }
+#pragma unroll
+  for (int r = 0; r < 4; ++r) {
+    const int row = q0 + wave * QTILE_WAVE + 4 * la + r;
+    if (row >= seq_len) continue;
+    const int64_t obase = ((int64_t)(seq_start + row) * num_q_heads + h) * HEAD_DIM;
+#pragma unroll
+    for (int c = 0; c < CHUNKS; ++c) {
+      out[obase + c * 16 + lq] = float_to_bf16_bits(oacc[c][r] * row_inv[r]);
+    }
+  }
+}
+
+// Single-wave probe: D[16,16] = A[16,32] @ B[32,16] using the fragment
+// layout conventions of the kernels above. Lets the GPU test suite verify the
+// lane->element mapping against torch.matmul before trusting the attention
+// kernel (guide §3: always check with asymmetric operands).
+__global__ void mfma_probe_kernel(float* __restrict__ d,
+                                  const bf16* __restrict__ a,
+                                  const bf16* __restrict__ b) {
+  const int lane = threadIdx.x % WAVE_SIZE;
+  const int lq = lane % 16;
+  const int la = lane / 16;
+  bf16x8 af, bf_;
+#pragma unroll
+  for (int jj = 0; jj < 8; ++jj) {
+    // A[row=lq][k], B[k][col=lq]
+    ushort au = a[lq * 32 + frag_k(la, jj)];
+    ushort bu = b[frag_k(la, jj) * 16 + lq];
+    af[jj] = *reinterpret_cast<__bf16*>(&au);
+    bf_[jj] = *reinterpret_cast<__bf16*>(&bu);
+  }
+  f32x4 c = {0.f, 0.f, 0.f, 0.f};
+  c = mfma16x16x32(af, bf_, c);
+#pragma unroll
+  for (int r = 0; r < 4; ++r) d[(4 * la + r) * 16 + lq] = c[r];
+}
+
+}  // namespace arks
+
+using namespace arks;
+
+extern "C" void arks_mfma_probe(void* d, const void* a, const void* b,
+                                hipStream_t stream) {
+  hipLaunchKernelGGL(mfma_probe_kernel, dim3(1), dim3(64), 0, stream, (float*)d,
+                     (const bf16*)a, (const bf16*)b);
+}
+
+extern "C" void arks_attn_prefill_varlen(
+    void* out, const void* q, const void* k, const void* v,
+    const void* cu_seqlens, const void* tile_info, int ntiles, float scale,
+    int num_q_heads, int num_kv_heads, int head_dim, hipStream_t stream) {
+  dim3 grid(num_q_heads, ntiles), block(256);
+  if (head_dim == 128) {
+    hipLaunchKernelGGL((attn_prefill_kernel<128>), grid, block, 0, stream,
+                       (bf16*)out, (const bf16*)q, (const bf16*)k,
+                       (const bf16*)v, (const int*)cu_seqlens,
+                       (const int*)tile_info, scale, num_q_heads, num_kv_heads);
+  } else if (head_dim == 64) {
+    hipLaunchKernelGGL((attn_prefill_kernel<64>), grid, block, 0, stream,
+                       (bf16*)out, (const bf16*)q, (const bf16*)k,
+                       (const bf16*)v, (const int*)cu_seqlens,
+                       (const int*)tile_info, scale, num_q_heads, num_kv_heads);
+  }
+}

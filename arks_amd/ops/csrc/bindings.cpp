@@ -1,0 +1,217 @@
+// Python bindings for the arks_amd gfx950 kernels (torch extension).
+// Host-side validation lives here; kernels are in the *.hip files.
+#include <torch/extension.h>
+
+#include <c10/hip/HIPStream.h>
+#include <hip/hip_runtime.h>
+
+#include <vector>
+
+// --- kernel launchers (see the .hip files) ---
+extern "C" {
+void arks_rmsnorm(void* out, const void* input, const void* weight, float eps,
+                  int rows, int hidden, hipStream_t stream);
+void arks_fused_add_rmsnorm(void* out, const void* input, void* residual,
+                            const void* weight, float eps, int rows, int hidden,
+                            hipStream_t stream);
+void arks_silu_mul(void* out, const void* gate_up, int64_t rows, int d,
+                   hipStream_t stream);
+void arks_rope_inplace(const void* positions, void* q, void* k,
+                       const void* cos_sin, int num_tokens, int head_dim,
+                       int num_q_heads, int num_kv_heads, hipStream_t stream);
+void arks_reshape_and_cache(const void* k, const void* v, void* k_cache,
+                            void* v_cache, const void* slot_mapping,
+                            int num_tokens, int num_kv_heads, int head_dim,
+                            int block_size, hipStream_t stream);
+void arks_attn_decode_paged(void* out, const void* q, const void* k_cache,
+                            const void* v_cache, const void* block_tables,
+                            const void* seq_lens, float scale, int num_seqs,
+                            int num_q_heads, int num_kv_heads, int head_dim,
+                            int max_blocks, hipStream_t stream);
+void arks_attn_prefill_varlen(void* out, const void* q, const void* k,
+                              const void* v, const void* cu_seqlens,
+                              const void* tile_info, int ntiles, float scale,
+                              int num_q_heads, int num_kv_heads, int head_dim,
+                              hipStream_t stream);
+void arks_greedy_sample(void* out, const void* logits, int rows, int vocab,
+                        hipStream_t stream);
+void arks_gumbel_sample(void* out, const void* logits, const void* temperatures,
+                        const void* uniform, int rows, int vocab,
+                        hipStream_t stream);
+void arks_mfma_probe(void* d, const void* a, const void* b, hipStream_t stream);
+}
+
+namespace {
+
+hipStream_t current_stream() {
+  return c10::hip::getCurrentHIPStream().stream();
+}
+
+void check_bf16_contig(const torch::Tensor& t, const char* name) {
+  TORCH_CHECK(t.is_cuda(), name, " must be on GPU");
+  TORCH_CHECK(t.scalar_type() == torch::kBFloat16, name, " must be bf16");
+  TORCH_CHECK(t.is_contiguous(), name, " must be contiguous");
+}
+
+void rmsnorm(torch::Tensor out, torch::Tensor input, torch::Tensor weight,
+             double eps) {
+  check_bf16_contig(out, "out");
+  check_bf16_contig(input, "input");
+  check_bf16_contig(weight, "weight");
+  const int hidden = input.size(-1);
+  TORCH_CHECK(hidden % 8 == 0, "hidden must be a multiple of 8");
+  const int rows = input.numel() / hidden;
+  arks_rmsnorm(out.data_ptr(), input.data_ptr(), weight.data_ptr(), (float)eps,
+               rows, hidden, current_stream());
+}
+
+void fused_add_rmsnorm(torch::Tensor out, torch::Tensor input,
+                       torch::Tensor residual, torch::Tensor weight,
+                       double eps) {
+  check_bf16_contig(out, "out");
+  check_bf16_contig(input, "input");
+  check_bf16_contig(residual, "residual");
+  const int hidden = input.size(-1);
+  TORCH_CHECK(hidden % 8 == 0, "hidden must be a multiple of 8");
+  const int rows = input.numel() / hidden;
+  arks_fused_add_rmsnorm(out.data_ptr(), input.data_ptr(), residual.data_ptr(),
+                         weight.data_ptr(), (float)eps, rows, hidden,
+                         current_stream());
+}
+
+void silu_mul(torch::Tensor out, torch::Tensor gate_up) {
+  check_bf16_contig(out, "out");
+  check_bf16_contig(gate_up, "gate_up");
+  const int d = out.size(-1);
+  TORCH_CHECK(gate_up.size(-1) == 2 * d, "gate_up last dim must be 2*d");
+  TORCH_CHECK(d % 8 == 0, "d must be a multiple of 8");
+  const int64_t rows = out.numel() / d;
+  arks_silu_mul(out.data_ptr(), gate_up.data_ptr(), rows, d, current_stream());
+}
+
+void rope_inplace(torch::Tensor positions, torch::Tensor q, torch::Tensor k,
+                  torch::Tensor cos_sin, int64_t head_dim) {
+  check_bf16_contig(q, "q");
+  check_bf16_contig(k, "k");
+  TORCH_CHECK(positions.scalar_type() == torch::kInt64, "positions must be i64");
+  TORCH_CHECK(cos_sin.scalar_type() == torch::kFloat32, "cos_sin must be f32");
+  TORCH_CHECK(head_dim % 4 == 0, "head_dim must be a multiple of 4");
+  const int num_tokens = q.size(0);
+  const int num_q_heads = q.size(-1) / head_dim;
+  const int num_kv_heads = k.size(-1) / head_dim;
+  arks_rope_inplace(positions.data_ptr(), q.data_ptr(), k.data_ptr(),
+                    cos_sin.data_ptr(), num_tokens, (int)head_dim, num_q_heads,
+                    num_kv_heads, current_stream());
+}
+
+void reshape_and_cache(torch::Tensor k, torch::Tensor v, torch::Tensor k_cache,
+                       torch::Tensor v_cache, torch::Tensor slot_mapping) {
+  check_bf16_contig(k, "k");
+  check_bf16_contig(v, "v");
+  check_bf16_contig(k_cache, "k_cache");
+  check_bf16_contig(v_cache, "v_cache");
+  TORCH_CHECK(slot_mapping.scalar_type() == torch::kInt64);
+  const int num_tokens = k.size(0);
+  const int num_kv_heads = k_cache.size(1);
+  const int block_size = k_cache.size(2);
+  const int head_dim = k_cache.size(3);
+  TORCH_CHECK(head_dim % 8 == 0);
+  arks_reshape_and_cache(k.data_ptr(), v.data_ptr(), k_cache.data_ptr(),
+                         v_cache.data_ptr(), slot_mapping.data_ptr(),
+                         num_tokens, num_kv_heads, head_dim, block_size,
+                         current_stream());
+}
+
+void attention_decode_paged(torch::Tensor out, torch::Tensor q,
+                            torch::Tensor k_cache, torch::Tensor v_cache,
+                            torch::Tensor block_tables, torch::Tensor seq_lens,
+                            double scale) {
+  check_bf16_contig(out, "out");
+  check_bf16_contig(q, "q");
+  check_bf16_contig(k_cache, "k_cache");
+  check_bf16_contig(v_cache, "v_cache");
+  TORCH_CHECK(block_tables.scalar_type() == torch::kInt32);
+  TORCH_CHECK(seq_lens.scalar_type() == torch::kInt32);
+  const int num_seqs = q.size(0);
+  const int num_q_heads = q.size(1);
+  const int head_dim = q.size(2);
+  const int num_kv_heads = k_cache.size(1);
+  TORCH_CHECK(k_cache.size(2) == 16, "KV block size must be 16");
+  TORCH_CHECK(head_dim == 64 || head_dim == 128, "head_dim must be 64 or 128");
+  const int gq = num_q_heads / num_kv_heads;
+  TORCH_CHECK(gq >= 1 && gq <= 8 && gq * num_kv_heads == num_q_heads,
+              "q/kv head ratio must be integral and <= 8");
+  const int max_blocks = block_tables.size(1);
+  arks_attn_decode_paged(out.data_ptr(), q.data_ptr(), k_cache.data_ptr(),
+                         v_cache.data_ptr(), block_tables.data_ptr(),
+                         seq_lens.data_ptr(), (float)scale, num_seqs,
+                         num_q_heads, num_kv_heads, head_dim, max_blocks,
+                         current_stream());
+}
+
+void attention_prefill_varlen(torch::Tensor out, torch::Tensor q,
+                              torch::Tensor k, torch::Tensor v,
+                              torch::Tensor cu_seqlens, torch::Tensor tile_info,
+                              double scale) {
+  check_bf16_contig(out, "out");
+  check_bf16_contig(q, "q");
+  check_bf16_contig(k, "k");
+  check_bf16_contig(v, "v");
+  TORCH_CHECK(cu_seqlens.scalar_type() == torch::kInt32);
+  TORCH_CHECK(tile_info.scalar_type() == torch::kInt32);
+  TORCH_CHECK(tile_info.dim() == 2 && tile_info.size(1) == 2);
+  const int num_q_heads = q.size(1);
+  const int head_dim = q.size(2);
+  const int num_kv_heads = k.size(1);
+  TORCH_CHECK(head_dim == 64 || head_dim == 128, "head_dim must be 64 or 128");
+  const int ntiles = tile_info.size(0);
+  arks_attn_prefill_varlen(out.data_ptr(), q.data_ptr(), k.data_ptr(),
+                           v.data_ptr(), cu_seqlens.data_ptr(),
+                           tile_info.data_ptr(), ntiles, (float)scale,
+                           num_q_heads, num_kv_heads, head_dim,
+                           current_stream());
+}
+
+void greedy_sample(torch::Tensor out, torch::Tensor logits) {
+  check_bf16_contig(logits, "logits");
+  TORCH_CHECK(out.scalar_type() == torch::kInt64);
+  arks_greedy_sample(out.data_ptr(), logits.data_ptr(), logits.size(0),
+                     logits.size(1), current_stream());
+}
+
+void gumbel_sample(torch::Tensor out, torch::Tensor logits,
+                   torch::Tensor temperatures, torch::Tensor uniform) {
+  check_bf16_contig(logits, "logits");
+  TORCH_CHECK(out.scalar_type() == torch::kInt64);
+  TORCH_CHECK(temperatures.scalar_type() == torch::kFloat32);
+  TORCH_CHECK(uniform.scalar_type() == torch::kFloat32);
+  arks_gumbel_sample(out.data_ptr(), logits.data_ptr(), temperatures.data_ptr(),
+                     uniform.data_ptr(), logits.size(0), logits.size(1),
+                     current_stream());
+}
+
+// D[16,16] = A[16,32] @ B[32,16], all bf16 in / f32 out. Verifies the MFMA
+// fragment layout assumption on hardware.
+void mfma_probe(torch::Tensor d, torch::Tensor a, torch::Tensor b) {
+  check_bf16_contig(a, "a");
+  check_bf16_contig(b, "b");
+  TORCH_CHECK(d.scalar_type() == torch::kFloat32 && d.is_contiguous());
+  TORCH_CHECK(a.size(0) == 16 && a.size(1) == 32);
+  TORCH_CHECK(b.size(0) == 32 && b.size(1) == 16);
+  arks_mfma_probe(d.data_ptr(), a.data_ptr(), b.data_ptr(), current_stream());
+}
+
+}  // namespace
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("rmsnorm", &rmsnorm);
+  m.def("fused_add_rmsnorm", &fused_add_rmsnorm);
+  m.def("silu_mul", &silu_mul);
+  m.def("rope_inplace", &rope_inplace);
+  m.def("reshape_and_cache", &reshape_and_cache);
+  m.def("attention_decode_paged", &attention_decode_paged);
+  m.def("attention_prefill_varlen", &attention_prefill_varlen);
+  m.def("greedy_sample", &greedy_sample);
+  m.def("gumbel_sample", &gumbel_sample);
+  m.def("mfma_probe", &mfma_probe);
+}

@@ -1,0 +1,200 @@
+"""Numerics tests: every gfx950 HIP kernel vs the plain-PyTorch fp32 reference
+(arks_amd/ops/ref.py). All tests here need the GPU."""
+
+import math
+
+import pytest
+import torch
+
+import arks_amd.ops as ops
+from arks_amd.ops import ref
+
+pytestmark = pytest.mark.gpu
+
+DEV = "cuda"
+
+
+def assert_native():
+    assert ops.native_available(), "HIP extension must be built on a GPU box"
+
+
+@pytest.mark.parametrize("rows,hidden", [(7, 128), (64, 3584), (3, 8192)])
+def test_rmsnorm(rows, hidden):
+    assert_native()
+    torch.manual_seed(0)
+    x = torch.randn(rows, hidden, dtype=torch.bfloat16, device=DEV)
+    w = torch.randn(hidden, dtype=torch.bfloat16, device=DEV)
+    out = ops.rmsnorm(x, w, 1e-6)
+    expect = ref.rmsnorm(x.float().cpu(), w.float().cpu(), 1e-6)
+    torch.testing.assert_close(out.float().cpu(), expect, atol=2e-2, rtol=2e-2)
+
+
+@pytest.mark.parametrize("rows,hidden", [(64, 3584), (5, 256)])
+def test_fused_add_rmsnorm(rows, hidden):
+    assert_native()
+    torch.manual_seed(1)
+    x = torch.randn(rows, hidden, dtype=torch.bfloat16, device=DEV)
+    res = torch.randn(rows, hidden, dtype=torch.bfloat16, device=DEV)
+    w = torch.randn(hidden, dtype=torch.bfloat16, device=DEV)
+    res_cpu = res.float().cpu().clone()
+    out, new_res = ops.fused_add_rmsnorm(x, res, w, 1e-6)
+    expect_out, expect_res = ref.fused_add_rmsnorm(
+        x.float().cpu(), res_cpu, w.float().cpu(), 1e-6
+    )
+    torch.testing.assert_close(new_res.float().cpu(), expect_res, atol=2e-2, rtol=2e-2)
+    torch.testing.assert_close(out.float().cpu(), expect_out, atol=3e-2, rtol=3e-2)
+
+
+def test_silu_mul():
+    assert_native()
+    torch.manual_seed(2)
+    x = torch.randn(33, 2 * 1024, dtype=torch.bfloat16, device=DEV)
+    out = ops.silu_mul(x)
+    expect = ref.silu_mul(x.float().cpu())
+    torch.testing.assert_close(out.float().cpu(), expect, atol=2e-2, rtol=2e-2)
+
+
+@pytest.mark.parametrize("hq,hkv,hd", [(8, 2, 128), (4, 4, 64)])
+def test_rope(hq, hkv, hd):
+    assert_native()
+    torch.manual_seed(3)
+    T = 50
+    pos = torch.randint(0, 2000, (T,), dtype=torch.int64, device=DEV)
+    q = torch.randn(T, hq * hd, dtype=torch.bfloat16, device=DEV)
+    k = torch.randn(T, hkv * hd, dtype=torch.bfloat16, device=DEV)
+    cs = ref.rope_cos_sin_cache(hd, 2048, 10000.0, device=DEV)
+    q_ref, k_ref = ref.rope_apply(
+        pos.cpu(), q.float().cpu(), k.float().cpu(), cs.cpu(), hd
+    )
+    q2, k2 = ops.rope_apply_inplace(pos, q, k, cs, hd)
+    torch.testing.assert_close(q2.float().cpu(), q_ref, atol=2e-2, rtol=2e-2)
+    torch.testing.assert_close(k2.float().cpu(), k_ref, atol=2e-2, rtol=2e-2)
+
+
+def test_reshape_and_cache():
+    assert_native()
+    torch.manual_seed(4)
+    T, hkv, hd, nb, bs = 37, 4, 128, 16, 16
+    k = torch.randn(T, hkv, hd, dtype=torch.bfloat16, device=DEV)
+    v = torch.randn(T, hkv, hd, dtype=torch.bfloat16, device=DEV)
+    kc = torch.zeros(nb, hkv, bs, hd, dtype=torch.bfloat16, device=DEV)
+    vc = torch.zeros_like(kc)
+    slots = torch.randperm(nb * bs, device=DEV)[:T].to(torch.int64)
+    kc_ref, vc_ref = kc.cpu().clone(), vc.cpu().clone()
+    ref.reshape_and_cache(k.cpu(), v.cpu(), kc_ref, vc_ref, slots.cpu())
+    ops.reshape_and_cache(k, v, kc, vc, slots)
+    torch.testing.assert_close(kc.cpu(), kc_ref)
+    torch.testing.assert_close(vc.cpu(), vc_ref)
+
+
+def test_mfma_probe():
+    """Verify the MFMA fragment-layout hypothesis vs torch.matmul.
+    Asymmetric operands so a transpose cannot pass (guide §3)."""
+    assert_native()
+    torch.manual_seed(5)
+    a = torch.randn(16, 32, dtype=torch.bfloat16, device=DEV)
+    b = torch.randn(32, 16, dtype=torch.bfloat16, device=DEV)
+    d = torch.zeros(16, 16, dtype=torch.float32, device=DEV)
+    from arks_amd.ops import _load
+
+    _load.C.mfma_probe(d, a, b)
+    expect = a.float() @ b.float()
+    torch.testing.assert_close(d.cpu(), expect.cpu(), atol=2e-2, rtol=2e-2)
+
+
+@pytest.mark.parametrize(
+    "hq,hkv,hd,seq_lens",
+    [
+        (28, 4, 128, [1, 16, 33, 256]),
+        (8, 8, 128, [64, 127]),
+        (8, 1, 128, [300]),
+        (4, 2, 64, [17, 90]),
+    ],
+)
+def test_attention_decode_paged(hq, hkv, hd, seq_lens):
+    assert_native()
+    torch.manual_seed(6)
+    bs = 16
+    S = len(seq_lens)
+    max_blocks = (max(seq_lens) + bs - 1) // bs
+    total_blocks = sum((n + bs - 1) // bs for n in seq_lens) + 2
+    q = torch.randn(S, hq, hd, dtype=torch.bfloat16, device=DEV)
+    kc = torch.randn(total_blocks, hkv, bs, hd, dtype=torch.bfloat16, device=DEV)
+    vc = torch.randn_like(kc)
+    # assign blocks sequentially
+    bt = torch.zeros(S, max_blocks, dtype=torch.int32, device=DEV)
+    nxt = 0
+    for i, n in enumerate(seq_lens):
+        nb = (n + bs - 1) // bs
+        bt[i, :nb] = torch.arange(nxt, nxt + nb, dtype=torch.int32)
+        nxt += nb
+    sl = torch.tensor(seq_lens, dtype=torch.int32, device=DEV)
+    scale = 1.0 / math.sqrt(hd)
+    out = ops.attention_decode_paged(q, kc, vc, bt, sl, scale)
+    expect = ref.attention_decode_paged(
+        q.float().cpu(), kc.float().cpu(), vc.float().cpu(), bt.cpu(), sl.cpu(), scale
+    )
+    torch.testing.assert_close(out.float().cpu(), expect, atol=2e-2, rtol=2e-2)
+
+
+@pytest.mark.parametrize(
+    "hq,hkv,hd,seq_lens",
+    [
+        (28, 4, 128, [1, 5, 64, 200]),
+        (8, 8, 128, [129]),
+        (4, 1, 128, [64, 64]),
+        (4, 2, 64, [100, 33]),
+    ],
+)
+def test_attention_prefill_varlen(hq, hkv, hd, seq_lens):
+    assert_native()
+    torch.manual_seed(7)
+    T = sum(seq_lens)
+    q = torch.randn(T, hq, hd, dtype=torch.bfloat16, device=DEV)
+    k = torch.randn(T, hkv, hd, dtype=torch.bfloat16, device=DEV)
+    v = torch.randn(T, hkv, hd, dtype=torch.bfloat16, device=DEV)
+    cu = torch.tensor(
+        [0] + list(torch.tensor(seq_lens).cumsum(0)), dtype=torch.int32, device=DEV
+    )
+    scale = 1.0 / math.sqrt(hd)
+    out = ops.attention_prefill_varlen(q, k, v, cu, seq_lens, scale)
+    expect = ref.attention_prefill_varlen(
+        q.float().cpu(), k.float().cpu(), v.float().cpu(), cu.cpu(), scale
+    )
+    torch.testing.assert_close(out.float().cpu(), expect, atol=3e-2, rtol=3e-2)
+
+
+def test_greedy_sample():
+    assert_native()
+    torch.manual_seed(8)
+    logits = torch.randn(9, 32000, dtype=torch.bfloat16, device=DEV)
+    out = ops.greedy_sample(logits)
+    expect = logits.float().argmax(dim=-1).cpu()
+    torch.testing.assert_close(out.cpu(), expect)
+
+
+def test_gumbel_sample_greedy_rows():
+    """Rows with temperature 0 must match argmax exactly."""
+    assert_native()
+    torch.manual_seed(9)
+    logits = torch.randn(8, 5000, dtype=torch.bfloat16, device=DEV)
+    temps = torch.zeros(8, dtype=torch.float32, device=DEV)
+    u = torch.rand(8, 5000, dtype=torch.float32, device=DEV)
+    out = ops.sample_tokens(logits, temps, u)
+    expect = logits.float().argmax(dim=-1).cpu()
+    torch.testing.assert_close(out.cpu(), expect)
+
+
+def test_gumbel_sample_distribution():
+    """Sampling a 4-token vocab many times approximates the softmax dist."""
+    assert_native()
+    torch.manual_seed(10)
+    N = 20000
+    logits = torch.tensor([[2.0, 1.0, 0.0, -1.0]], dtype=torch.bfloat16, device=DEV)
+    logits = logits.expand(N, 4).contiguous()
+    temps = torch.ones(N, dtype=torch.float32, device=DEV)
+    u = torch.rand(N, 4, dtype=torch.float32, device=DEV)
+    out = ops.sample_tokens(logits, temps, u)
+    counts = torch.bincount(out, minlength=4).float() / N
+    expect = torch.softmax(torch.tensor([2.0, 1.0, 0.0, -1.0]), dim=0)
+    assert (counts.cpu() - expect).abs().max() < 0.02
