@@ -1,0 +1,166 @@
+"""Model and engine configuration.
+
+ModelConfig is HF-config-driven: it reads the same config.json fields the
+reference's delegated runtimes (vLLM/SGLang at reference
+arksapplication_controller.go:941-1014) consume, so any Qwen2/Llama-family
+checkpoint directory works unchanged.
+"""
+
+from __future__ import annotations
+
+import json
+import os
+from dataclasses import dataclass, field
+
+
+@dataclass
+class ModelConfig:
+    architecture: str = "Qwen2ForCausalLM"
+    vocab_size: int = 151936
+    hidden_size: int = 3584
+    intermediate_size: int = 18944
+    num_hidden_layers: int = 28
+    num_attention_heads: int = 28
+    num_key_value_heads: int = 4
+    head_dim: int = 128
+    rms_norm_eps: float = 1e-6
+    rope_theta: float = 1000000.0
+    max_position_embeddings: int = 32768
+    tie_word_embeddings: bool = False
+    attention_bias: bool = True  # qwen2 has qkv bias; llama does not
+    eos_token_id: int = 151645
+    bos_token_id: int = 151643
+    torch_dtype: str = "bfloat16"
+
+    @classmethod
+    def from_hf_config(cls, cfg: dict) -> "ModelConfig":
+        arch = (cfg.get("architectures") or ["Qwen2ForCausalLM"])[0]
+        num_heads = cfg.get("num_attention_heads", 28)
+        hidden = cfg.get("hidden_size", 3584)
+        eos = cfg.get("eos_token_id", 151645)
+        if isinstance(eos, list):
+            eos = eos[0]
+        return cls(
+            architecture=arch,
+            vocab_size=cfg.get("vocab_size", 151936),
+            hidden_size=hidden,
+            intermediate_size=cfg.get("intermediate_size", 18944),
+            num_hidden_layers=cfg.get("num_hidden_layers", 28),
+            num_attention_heads=num_heads,
+            num_key_value_heads=cfg.get("num_key_value_heads", num_heads),
+            head_dim=cfg.get("head_dim", hidden // num_heads),
+            rms_norm_eps=cfg.get("rms_norm_eps", 1e-6),
+            rope_theta=cfg.get("rope_theta", 10000.0),
+            max_position_embeddings=cfg.get("max_position_embeddings", 32768),
+            tie_word_embeddings=cfg.get("tie_word_embeddings", False),
+            attention_bias=cfg.get("attention_bias", arch.startswith("Qwen2")),
+            eos_token_id=eos,
+            bos_token_id=cfg.get("bos_token_id", 1),
+            torch_dtype=cfg.get("torch_dtype", "bfloat16"),
+        )
+
+    @classmethod
+    def from_pretrained(cls, model_path: str) -> "ModelConfig":
+        with open(os.path.join(model_path, "config.json")) as f:
+            return cls.from_hf_config(json.load(f))
+
+    @property
+    def num_qo_heads(self) -> int:
+        return self.num_attention_heads
+
+
+# Named synthetic configs for benchmarks / tests (random-init weights; no
+# network access for real checkpoints — BASELINE.json's models by shape).
+PRESET_CONFIGS: dict[str, ModelConfig] = {
+    "qwen2.5-0.5b": ModelConfig(
+        architecture="Qwen2ForCausalLM",
+        vocab_size=151936,
+        hidden_size=896,
+        intermediate_size=4864,
+        num_hidden_layers=24,
+        num_attention_heads=14,
+        num_key_value_heads=2,
+        head_dim=64,
+        rope_theta=1000000.0,
+        tie_word_embeddings=True,
+    ),
+    "qwen2.5-7b": ModelConfig(
+        architecture="Qwen2ForCausalLM",
+        vocab_size=152064,
+        hidden_size=3584,
+        intermediate_size=18944,
+        num_hidden_layers=28,
+        num_attention_heads=28,
+        num_key_value_heads=4,
+        head_dim=128,
+        rope_theta=1000000.0,
+    ),
+    "llama-3-70b": ModelConfig(
+        architecture="LlamaForCausalLM",
+        vocab_size=128256,
+        hidden_size=8192,
+        intermediate_size=28672,
+        num_hidden_layers=80,
+        num_attention_heads=64,
+        num_key_value_heads=8,
+        head_dim=128,
+        rms_norm_eps=1e-5,
+        rope_theta=500000.0,
+        attention_bias=False,
+        eos_token_id=128001,
+        bos_token_id=128000,
+    ),
+    "tiny": ModelConfig(  # CPU-test-sized
+        architecture="Qwen2ForCausalLM",
+        vocab_size=512,
+        hidden_size=128,
+        intermediate_size=256,
+        num_hidden_layers=2,
+        num_attention_heads=4,
+        num_key_value_heads=2,
+        head_dim=32,
+        rope_theta=10000.0,
+        max_position_embeddings=2048,
+        eos_token_id=2,
+        bos_token_id=1,
+    ),
+    "tiny-gpu": ModelConfig(  # GPU-test-sized (head_dim 128 kernels)
+        architecture="Qwen2ForCausalLM",
+        vocab_size=2048,
+        hidden_size=512,
+        intermediate_size=1024,
+        num_hidden_layers=2,
+        num_attention_heads=4,
+        num_key_value_heads=2,
+        head_dim=128,
+        rope_theta=10000.0,
+        max_position_embeddings=4096,
+        eos_token_id=2,
+        bos_token_id=1,
+    ),
+}
+
+
+@dataclass
+class EngineConfig:
+    model_path: str | None = None  # dir with config.json + *.safetensors
+    preset: str | None = None  # or a PRESET_CONFIGS key (random-init)
+    served_model_name: str = "model"
+    tensor_parallel_size: int = 1
+    block_size: int = 16  # KV page size (tokens) — fixed by the decode kernel
+    gpu_memory_utilization: float = 0.90
+    max_num_seqs: int = 256
+    max_num_batched_tokens: int = 8192
+    max_model_len: int = 8192
+    kv_cache_blocks: int | None = None  # override (else sized from free HBM)
+    enforce_eager: bool = False  # disable hipGraph decode capture
+    device: str = "cuda"
+    dtype: str = "bfloat16"
+    seed: int = 0
+
+    def model_config(self) -> ModelConfig:
+        if self.model_path:
+            return ModelConfig.from_pretrained(self.model_path)
+        if self.preset:
+            return PRESET_CONFIGS[self.preset]
+        raise ValueError("EngineConfig needs model_path or preset")

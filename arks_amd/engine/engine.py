@@ -1,0 +1,90 @@
+"""LLMEngine: the synchronous continuous-batching core.
+
+add_request() -> step() loop -> StepOutputs. The OpenAI server wraps this in
+an asyncio loop (arks_amd.server); bench.py drives it directly.
+"""
+
+from __future__ import annotations
+
+import time
+
+from ..config import EngineConfig
+from .model_runner import ModelRunner
+from .scheduler import Scheduler
+from .sequence import SamplingParams, Sequence, SeqStatus, StepOutput
+
+
+class LLMEngine:
+    def __init__(self, cfg: EngineConfig):
+        self.cfg = cfg
+        self.model_cfg = cfg.model_config()
+        self.runner = ModelRunner(cfg, self.model_cfg)
+        t0 = time.time()
+        self.runner.load_weights()
+        self.load_seconds = time.time() - t0
+        allocator = self.runner.init_kv_cache()
+        self.scheduler = Scheduler(
+            allocator,
+            max_num_seqs=cfg.max_num_seqs,
+            max_num_batched_tokens=cfg.max_num_batched_tokens,
+            max_model_len=cfg.max_model_len,
+        )
+        # serving metrics
+        self.total_prompt_tokens = 0
+        self.total_output_tokens = 0
+
+    # ---- request API ----
+    def add_request(
+        self,
+        prompt_token_ids: list[int],
+        sampling: SamplingParams | None = None,
+        request_id: str | None = None,
+        arrival_time: float | None = None,
+    ) -> Sequence:
+        seq = Sequence(prompt_token_ids, sampling, request_id, arrival_time)
+        self.scheduler.add(seq)
+        self.total_prompt_tokens += seq.num_prompt_tokens
+        return seq
+
+    def abort_request(self, request_id: str) -> bool:
+        return self.scheduler.abort(request_id)
+
+    def has_work(self) -> bool:
+        return self.scheduler.has_work()
+
+    # ---- the step loop ----
+    def step(self) -> list[StepOutput]:
+        sb = self.scheduler.schedule()
+        if sb is None:
+            return []
+        new_tokens = self.runner.execute(sb)
+        outputs: list[StepOutput] = []
+        eos = self.model_cfg.eos_token_id
+        for seq, tok in zip(sb.seqs, new_tokens):
+            if seq.status is not SeqStatus.RUNNING:  # aborted mid-step
+                continue
+            seq.append_token(tok)
+            self.total_output_tokens += 1
+            finished = seq.check_finished(eos)
+            outputs.append(
+                StepOutput(
+                    request_id=seq.request_id,
+                    seq_id=seq.seq_id,
+                    new_token_id=tok,
+                    finished=finished,
+                    finish_reason=seq.finish_reason,
+                    num_prompt_tokens=seq.num_prompt_tokens,
+                    num_output_tokens=seq.num_output_tokens,
+                )
+            )
+        self.scheduler.free_finished()
+        return outputs
+
+    # ---- convenience: run to completion (tests / offline use) ----
+    def generate(
+        self, prompts: list[list[int]], sampling: SamplingParams | None = None
+    ) -> list[list[int]]:
+        seqs = [self.add_request(p, sampling) for p in prompts]
+        while self.has_work():
+            self.step()
+        return [s.output_token_ids for s in seqs]

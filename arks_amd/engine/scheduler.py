@@ -1,0 +1,184 @@
+"""Continuous-batching scheduler (the subset of vLLM's engineering the
+runtime slot needs — SURVEY.md §7 'hard parts' #2).
+
+Policy (v0, vLLM-v0-like):
+- Prefill-priority: if waiting requests fit the token budget and KV blocks,
+  schedule them as one varlen prefill batch.
+- Otherwise run one decode step over all RUNNING sequences, allocating one
+  new block per sequence when it crosses a page boundary.
+- On KV exhaustion, preempt the most-recently-admitted sequence
+  (free its pages, recompute later) until the rest fits.
+"""
+
+from __future__ import annotations
+
+from collections import deque
+from dataclasses import dataclass, field
+
+from .kv_cache import BlockAllocator
+from .sequence import Sequence, SeqStatus
+
+
+@dataclass
+class ScheduledBatch:
+    seqs: list[Sequence]
+    is_prefill: bool
+    # number of new tokens to run for each seq (prefill: whole prompt;
+    # decode: 1)
+    num_new_tokens: list[int] = field(default_factory=list)
+
+    @property
+    def total_tokens(self) -> int:
+        return sum(self.num_new_tokens)
+
+
+class Scheduler:
+    def __init__(
+        self,
+        allocator: BlockAllocator,
+        max_num_seqs: int = 256,
+        max_num_batched_tokens: int = 8192,
+        max_model_len: int = 8192,
+    ):
+        self.allocator = allocator
+        self.max_num_seqs = max_num_seqs
+        self.max_num_batched_tokens = max_num_batched_tokens
+        self.max_model_len = max_model_len
+        self.waiting: deque[Sequence] = deque()
+        self.running: list[Sequence] = []
+        self.num_preemptions = 0
+
+    # --- API ---
+    def add(self, seq: Sequence) -> None:
+        if seq.num_prompt_tokens > self.max_model_len:
+            raise ValueError(
+                f"prompt length {seq.num_prompt_tokens} exceeds max_model_len "
+                f"{self.max_model_len}"
+            )
+        if (
+            BlockAllocator.blocks_needed(seq.num_prompt_tokens, self.allocator.block_size)
+            > self.allocator.num_blocks
+        ):
+            raise ValueError("prompt does not fit in the KV cache at all")
+        seq.status = SeqStatus.WAITING
+        self.waiting.append(seq)
+
+    def abort(self, request_id: str) -> bool:
+        for i, s in enumerate(self.waiting):
+            if s.request_id == request_id:
+                s.status = SeqStatus.ABORTED
+                del self.waiting[i]
+                return True
+        for s in self.running:
+            if s.request_id == request_id:
+                s.status = SeqStatus.ABORTED
+                self._release(s)
+                self.running.remove(s)
+                return True
+        return False
+
+    def has_work(self) -> bool:
+        return bool(self.waiting) or bool(self.running)
+
+    @property
+    def num_waiting(self) -> int:
+        return len(self.waiting)
+
+    @property
+    def num_running(self) -> int:
+        return len(self.running)
+
+    # --- scheduling ---
+    def schedule(self) -> ScheduledBatch | None:
+        batch = self._schedule_prefill()
+        if batch is not None:
+            return batch
+        return self._schedule_decode()
+
+    def _schedule_prefill(self) -> ScheduledBatch | None:
+        seqs: list[Sequence] = []
+        ntoks: list[int] = []
+        budget = self.max_num_batched_tokens
+        while self.waiting:
+            seq = self.waiting[0]
+            # num_tokens (not num_prompt_tokens): a preempted sequence is
+            # recomputed over prompt + already-generated tokens.
+            n = seq.num_tokens
+            if seqs and n > budget:
+                break
+            if len(self.running) + len(seqs) >= self.max_num_seqs:
+                break
+            need = BlockAllocator.blocks_needed(n, self.allocator.block_size)
+            if not self.allocator.can_allocate(need):
+                break
+            self.waiting.popleft()
+            seq.block_table = self.allocator.allocate(need)
+            seq.status = SeqStatus.RUNNING
+            seqs.append(seq)
+            ntoks.append(n)
+            budget -= n
+            if budget <= 0:
+                break
+        if not seqs:
+            return None
+        self.running.extend(seqs)
+        return ScheduledBatch(seqs=seqs, is_prefill=True, num_new_tokens=ntoks)
+
+    def _schedule_decode(self) -> ScheduledBatch | None:
+        if not self.running:
+            return None
+        # Ensure each running sequence has a slot for its next token;
+        # preempt from the back (most recent) on exhaustion.
+        scheduled: list[Sequence] = []
+        i = 0
+        while i < len(self.running):
+            seq = self.running[i]
+            need = BlockAllocator.blocks_needed(
+                seq.num_tokens, self.allocator.block_size
+            ) - len(seq.block_table)
+            if need > 0 and not self.allocator.can_allocate(need):
+                # Victims must be unvisited (not already in this batch).
+                if not self._preempt_last(exclude=seq, protected=scheduled):
+                    # nothing left to preempt but this seq itself
+                    self._preempt(seq)
+                    self.running.remove(seq)
+                    continue
+                continue  # retry same seq
+            if need > 0:
+                seq.block_table.extend(self.allocator.allocate(need))
+            scheduled.append(seq)
+            i += 1
+        if not scheduled:
+            return None
+        return ScheduledBatch(
+            seqs=scheduled, is_prefill=False, num_new_tokens=[1] * len(scheduled)
+        )
+
+    def _preempt_last(self, exclude: Sequence, protected: list[Sequence]) -> bool:
+        for seq in reversed(self.running):
+            if seq is exclude or seq in protected:
+                continue
+            self._preempt(seq)
+            self.running.remove(seq)
+            return True
+        return False
+
+    def _preempt(self, seq: Sequence) -> None:
+        self.num_preemptions += 1
+        self._release(seq)
+        seq.status = SeqStatus.WAITING
+        # Recompute-style preemption: prompt grows to include generated tokens
+        # so the whole context is prefil­led again on readmission.
+        seq.num_cached_tokens = 0
+        self.waiting.appendleft(seq)
+
+    def _release(self, seq: Sequence) -> None:
+        if seq.block_table:
+            self.allocator.free(seq.block_table)
+            seq.block_table = []
+
+    def free_finished(self) -> None:
+        for seq in self.running:
+            if seq.is_finished:
+                self._release(seq)
+        self.running = [s for s in self.running if not s.is_finished]

@@ -1,0 +1,22 @@
+"""Model registry: HF `architectures[0]` -> implementation class."""
+
+from __future__ import annotations
+
+from ..config import ModelConfig
+from .llama_family import LlamaFamilyForCausalLM
+
+_REGISTRY = {
+    "Qwen2ForCausalLM": LlamaFamilyForCausalLM,
+    "LlamaForCausalLM": LlamaFamilyForCausalLM,
+}
+
+
+def create_model(cfg: ModelConfig, dtype=None):
+    import torch
+
+    cls = _REGISTRY.get(cfg.architecture)
+    if cls is None:
+        raise ValueError(
+            f"unsupported architecture {cfg.architecture!r}; known: {sorted(_REGISTRY)}"
+        )
+    return cls(cfg, dtype=dtype or torch.bfloat16)

@@ -1,0 +1,250 @@
+"""Qwen2 / Llama decoder family for the arks_amd engine.
+
+One implementation covers both architectures (Qwen2 = Llama + QKV bias +
+different rope_theta/vocab). HF-config-driven (arks_amd.config.ModelConfig);
+weight names follow the HF checkpoint layout so safetensors load directly.
+
+Compute path: F.linear GEMMs (hipBLASLt) + arks_amd.ops HIP kernels for
+RMSNorm / RoPE / paged attention / SwiGLU; TP via column/row-parallel shards
+with one RCCL all-reduce per attention and per MLP block.
+"""
+
+from __future__ import annotations
+
+import math
+
+import torch
+import torch.nn as nn
+
+from .. import ops
+from ..config import ModelConfig
+from ..engine.forward_batch import ForwardBatch
+from ..parallel.comm import get_tp_world_size
+from ..parallel.layers import (
+    MergedColumnParallelLinear,
+    ParallelLMHead,
+    QKVParallelLinear,
+    RowParallelLinear,
+)
+
+
+class RMSNorm(nn.Module):
+    def __init__(self, hidden: int, eps: float, dtype=torch.bfloat16):
+        super().__init__()
+        self.weight = nn.Parameter(torch.ones(hidden, dtype=dtype), requires_grad=False)
+        self.eps = eps
+
+    def forward(self, x, residual=None):
+        if residual is None:
+            return ops.rmsnorm(x, self.weight, self.eps)
+        return ops.fused_add_rmsnorm(x, residual, self.weight, self.eps)
+
+
+class Attention(nn.Module):
+    def __init__(self, cfg: ModelConfig, dtype=torch.bfloat16):
+        super().__init__()
+        tp = get_tp_world_size()
+        self.head_dim = cfg.head_dim
+        self.nq_local = cfg.num_attention_heads // tp
+        self.nkv_local = cfg.num_key_value_heads // tp
+        self.scale = 1.0 / math.sqrt(cfg.head_dim)
+        self.qkv_proj = QKVParallelLinear(
+            cfg.hidden_size, cfg.head_dim, cfg.num_attention_heads,
+            cfg.num_key_value_heads, bias=cfg.attention_bias, dtype=dtype,
+        )
+        self.o_proj = RowParallelLinear(
+            cfg.num_attention_heads * cfg.head_dim, cfg.hidden_size, bias=False,
+            dtype=dtype,
+        )
+
+    def forward(self, x: torch.Tensor, batch: ForwardBatch, kv_cache) -> torch.Tensor:
+        T = x.shape[0]
+        qkv = self.qkv_proj(x)
+        q, k, v = self.qkv_proj.split_qkv(qkv)
+        q = q.contiguous()
+        k = k.contiguous()
+        v = v.contiguous()
+        q, k = ops.rope_apply_inplace(
+            batch.positions, q, k, self._cos_sin, self.head_dim
+        )
+        k = k.view(T, self.nkv_local, self.head_dim)
+        v = v.view(T, self.nkv_local, self.head_dim)
+        k_cache, v_cache = kv_cache
+        ops.reshape_and_cache(k, v, k_cache, v_cache, batch.slot_mapping)
+        q = q.view(T, self.nq_local, self.head_dim)
+        if batch.is_prefill:
+            out = ops.attention_prefill_varlen(
+                q, k, v, batch.cu_seqlens, batch.seq_lens_list, self.scale
+            )
+        else:
+            out = ops.attention_decode_paged(
+                q, k_cache, v_cache, batch.block_tables, batch.seq_lens, self.scale
+            )
+        return self.o_proj(out.view(T, -1))
+
+
+class MLP(nn.Module):
+    def __init__(self, cfg: ModelConfig, dtype=torch.bfloat16):
+        super().__init__()
+        self.gate_up_proj = MergedColumnParallelLinear(
+            cfg.hidden_size, cfg.intermediate_size, bias=False, dtype=dtype
+        )
+        self.down_proj = RowParallelLinear(
+            cfg.intermediate_size, cfg.hidden_size, bias=False, dtype=dtype
+        )
+
+    def forward(self, x):
+        return self.down_proj(ops.silu_mul(self.gate_up_proj(x)))
+
+
+class DecoderLayer(nn.Module):
+    def __init__(self, cfg: ModelConfig, dtype=torch.bfloat16):
+        super().__init__()
+        self.input_layernorm = RMSNorm(cfg.hidden_size, cfg.rms_norm_eps, dtype)
+        self.self_attn = Attention(cfg, dtype)
+        self.post_attention_layernorm = RMSNorm(cfg.hidden_size, cfg.rms_norm_eps, dtype)
+        self.mlp = MLP(cfg, dtype)
+
+    def forward(self, x, residual, batch: ForwardBatch, kv_cache):
+        if residual is None:
+            residual = x
+            x = self.input_layernorm(x)
+        else:
+            x, residual = self.input_layernorm(x, residual)
+        x = self.self_attn(x, batch, kv_cache)
+        x, residual = self.post_attention_layernorm(x, residual)
+        x = self.mlp(x)
+        return x, residual
+
+
+class LlamaFamilyForCausalLM(nn.Module):
+    """Covers Qwen2ForCausalLM and LlamaForCausalLM."""
+
+    def __init__(self, cfg: ModelConfig, dtype=torch.bfloat16):
+        super().__init__()
+        self.cfg = cfg
+        self.dtype = dtype
+        self.embed_tokens = nn.Embedding(
+            cfg.vocab_size, cfg.hidden_size, dtype=dtype
+        )
+        self.embed_tokens.weight.requires_grad_(False)
+        self.layers = nn.ModuleList(
+            DecoderLayer(cfg, dtype) for _ in range(cfg.num_hidden_layers)
+        )
+        self.norm = RMSNorm(cfg.hidden_size, cfg.rms_norm_eps, dtype)
+        self.lm_head = ParallelLMHead(cfg.hidden_size, cfg.vocab_size, dtype)
+        cos_sin = self._build_cos_sin()
+        self.register_buffer("rope_cos_sin", cos_sin, persistent=False)
+        for layer in self.layers:
+            layer.self_attn._cos_sin = cos_sin
+
+    def _build_cos_sin(self):
+        from ..ops import ref
+
+        return ref.rope_cos_sin_cache(
+            self.cfg.head_dim, self.cfg.max_position_embeddings, self.cfg.rope_theta
+        )
+
+    def _apply(self, fn, recurse=True):  # keep the shared cos_sin in sync
+        out = super()._apply(fn, recurse)
+        for layer in self.layers:
+            layer.self_attn._cos_sin = self.rope_cos_sin
+        return out
+
+    def forward(self, batch: ForwardBatch, kv_caches: list) -> torch.Tensor:
+        """Returns logits for batch.logits_indices rows."""
+        x = self.embed_tokens(batch.input_ids)
+        residual = None
+        for i, layer in enumerate(self.layers):
+            x, residual = layer(x, residual, batch, kv_caches[i])
+        x, _ = self.norm(x, residual)
+        if batch.logits_indices is not None:
+            x = x[batch.logits_indices]
+        return self.lm_head(x)
+
+    # ------------------------- weight loading -------------------------
+    def load_hf_state_dict(self, tensors: dict[str, torch.Tensor]) -> None:
+        """Load an HF-layout state dict (full tensors; sharded here)."""
+        pending_qkv: dict[int, dict] = {}
+        pending_mlp: dict[int, dict] = {}
+        own = dict(self.named_parameters())
+
+        def put(name, value):
+            p = own[name]
+            assert p.shape == value.shape, (name, p.shape, value.shape)
+            p.data.copy_(value.to(p.dtype))
+
+        for name, w in tensors.items():
+            name = name.removeprefix("model.")
+            if name == "embed_tokens.weight":
+                put("embed_tokens.weight", w)
+                if self.cfg.tie_word_embeddings:
+                    put("lm_head.weight", self.lm_head.shard(w))
+            elif name in ("lm_head.weight",):
+                put("lm_head.weight", self.lm_head.shard(w))
+            elif name == "norm.weight":
+                put("norm.weight", w)
+            elif name.startswith("layers."):
+                parts = name.split(".")
+                li = int(parts[1])
+                sub = ".".join(parts[2:])
+                layer = self.layers[li]
+                if sub in ("input_layernorm.weight", "post_attention_layernorm.weight"):
+                    put(f"layers.{li}.{sub}", w)
+                elif sub.startswith("self_attn.") and sub.split(".")[1] in (
+                    "q_proj", "k_proj", "v_proj",
+                ):
+                    proj, param = sub.split(".")[1], sub.split(".")[2]
+                    pending_qkv.setdefault(li, {})[f"{proj}.{param}"] = w
+                    d = pending_qkv[li]
+                    keys_w = {"q_proj.weight", "k_proj.weight", "v_proj.weight"}
+                    keys_b = {"q_proj.bias", "k_proj.bias", "v_proj.bias"}
+                    if keys_w <= d.keys():
+                        fused = layer.self_attn.qkv_proj.shard_qkv(
+                            d["q_proj.weight"], d["k_proj.weight"], d["v_proj.weight"]
+                        )
+                        put(f"layers.{li}.self_attn.qkv_proj.weight", fused)
+                        for kk in keys_w:
+                            del d[kk]
+                    if layer.self_attn.qkv_proj.bias is not None and keys_b <= d.keys():
+                        fused = layer.self_attn.qkv_proj.shard_qkv(
+                            d["q_proj.bias"], d["k_proj.bias"], d["v_proj.bias"]
+                        )
+                        put(f"layers.{li}.self_attn.qkv_proj.bias", fused)
+                        for kk in keys_b:
+                            del d[kk]
+                elif sub == "self_attn.o_proj.weight":
+                    put(
+                        f"layers.{li}.self_attn.o_proj.weight",
+                        layer.self_attn.o_proj.shard(w),
+                    )
+                elif sub in ("mlp.gate_proj.weight", "mlp.up_proj.weight"):
+                    pending_mlp.setdefault(li, {})[sub] = w
+                    d = pending_mlp[li]
+                    if {"mlp.gate_proj.weight", "mlp.up_proj.weight"} <= d.keys():
+                        fused = layer.mlp.gate_up_proj.shard_merged(
+                            d["mlp.gate_proj.weight"], d["mlp.up_proj.weight"]
+                        )
+                        put(f"layers.{li}.mlp.gate_up_proj.weight", fused)
+                        d.clear()
+                elif sub == "mlp.down_proj.weight":
+                    put(f"layers.{li}.mlp.down_proj.weight", layer.mlp.down_proj.shard(w))
+                # rotary_emb.inv_freq etc. are ignored (recomputed)
+
+    @torch.no_grad()
+    def random_init(self, seed: int = 0) -> None:
+        g = torch.Generator().manual_seed(seed)
+        for p in self.parameters():
+            if p.dim() >= 2:
+                std = 0.02
+                p.data.copy_(
+                    torch.randn(p.shape, generator=g, dtype=torch.float32).mul_(std).to(p.dtype)
+                )
+            else:
+                p.data.zero_()
+        # norms to 1
+        for name, p in self.named_parameters():
+            if "layernorm" in name or name == "norm.weight":
+                p.data.fill_(1.0)
+            elif name.endswith("qkv_proj.bias"):
+                p.data.zero_()

@@ -1,0 +1,99 @@
+"""Tensor-parallel process-group management over RCCL/xGMI.
+
+One process per GPU; torch.distributed backend "nccl" IS RCCL on ROCm. The
+TP group is the whole world for a single serving instance (the LWS
+leader/worker topology of the reference maps each inference group to one
+node's xGMI-connected GPUs — SURVEY.md §5 'Distributed communication
+backend'). Falls back to gloo on CPU so the distributed path is testable
+without hardware.
+"""
+
+from __future__ import annotations
+
+import os
+from datetime import timedelta
+
+import torch
+import torch.distributed as dist
+
+_TP_GROUP = None
+_TP_RANK = 0
+_TP_WORLD = 1
+
+
+def init_tp(
+    tp_size: int | None = None,
+    rank: int | None = None,
+    backend: str | None = None,
+    master_addr: str = "127.0.0.1",
+    master_port: int = 29500,
+    timeout_s: int = 600,
+) -> None:
+    """Initialize the TP process group. Reads torchrun env (RANK/WORLD_SIZE/
+    MASTER_ADDR/MASTER_PORT) when present."""
+    global _TP_GROUP, _TP_RANK, _TP_WORLD
+    world = int(os.environ.get("WORLD_SIZE", tp_size or 1))
+    if world <= 1:
+        _TP_GROUP, _TP_RANK, _TP_WORLD = None, 0, 1
+        return
+    r = int(os.environ.get("RANK", rank if rank is not None else 0))
+    if backend is None:
+        backend = "nccl" if torch.cuda.is_available() else "gloo"
+    if not dist.is_initialized():
+        os.environ.setdefault("MASTER_ADDR", master_addr)
+        os.environ.setdefault("MASTER_PORT", str(master_port))
+        dist.init_process_group(
+            backend=backend,
+            rank=r,
+            world_size=world,
+            timeout=timedelta(seconds=timeout_s),
+        )
+    if torch.cuda.is_available():
+        torch.cuda.set_device(int(os.environ.get("LOCAL_RANK", r)))
+    _TP_GROUP = dist.group.WORLD
+    _TP_RANK = dist.get_rank()
+    _TP_WORLD = dist.get_world_size()
+
+
+def destroy_tp() -> None:
+    global _TP_GROUP, _TP_RANK, _TP_WORLD
+    if dist.is_initialized():
+        dist.destroy_process_group()
+    _TP_GROUP, _TP_RANK, _TP_WORLD = None, 0, 1
+
+
+def get_tp_group():
+    return _TP_GROUP
+
+
+def get_tp_rank() -> int:
+    return _TP_RANK
+
+
+def get_tp_world_size() -> int:
+    return _TP_WORLD
+
+
+def tp_all_reduce(x: torch.Tensor) -> torch.Tensor:
+    """In-place sum all-reduce across the TP group (no-op at TP=1)."""
+    if _TP_WORLD > 1:
+        dist.all_reduce(x, op=dist.ReduceOp.SUM, group=_TP_GROUP)
+    return x
+
+
+def tp_all_gather(x: torch.Tensor, dim: int = -1) -> torch.Tensor:
+    """All-gather shards along `dim` (no-op at TP=1)."""
+    if _TP_WORLD <= 1:
+        return x
+    parts = [torch.empty_like(x) for _ in range(_TP_WORLD)]
+    dist.all_gather(parts, x.contiguous(), group=_TP_GROUP)
+    return torch.cat(parts, dim=dim)
+
+
+def tp_broadcast_object(obj, src: int = 0):
+    """Broadcast a picklable object from the driver rank (no-op at TP=1)."""
+    if _TP_WORLD <= 1:
+        return obj
+    buf = [obj]
+    dist.broadcast_object_list(buf, src=src, group=_TP_GROUP)
+    return buf[0]

@@ -1,0 +1,160 @@
+"""Tensor-parallel linear layers (Megatron-style column/row split).
+
+GEMMs go through F.linear (hipBLASLt on ROCm — the guide's 'plain library
+GEMM' path); the fused hot ops around them are the hand-written kernels in
+arks_amd.ops. Weight loading maps full HF tensors to local shards via each
+layer's `shard()`.
+"""
+
+from __future__ import annotations
+
+import torch
+import torch.nn as nn
+import torch.nn.functional as F
+
+from .comm import get_tp_rank, get_tp_world_size, tp_all_gather, tp_all_reduce
+
+
+class ColumnParallelLinear(nn.Module):
+    """Y = X W^T with W row-sharded over TP ranks (output features split).
+    Output stays sharded (gather_output=False semantics)."""
+
+    def __init__(self, in_features: int, out_features: int, bias: bool,
+                 dtype: torch.dtype = torch.bfloat16):
+        super().__init__()
+        tp = get_tp_world_size()
+        assert out_features % tp == 0, (out_features, tp)
+        self.in_features = in_features
+        self.out_features = out_features
+        self.out_per_rank = out_features // tp
+        self.weight = nn.Parameter(
+            torch.empty(self.out_per_rank, in_features, dtype=dtype),
+            requires_grad=False,
+        )
+        self.bias = (
+            nn.Parameter(torch.empty(self.out_per_rank, dtype=dtype), requires_grad=False)
+            if bias
+            else None
+        )
+
+    def shard(self, full: torch.Tensor, param: str = "weight") -> torch.Tensor:
+        r = get_tp_rank()
+        return full[r * self.out_per_rank : (r + 1) * self.out_per_rank]
+
+    def forward(self, x: torch.Tensor) -> torch.Tensor:
+        return F.linear(x, self.weight, self.bias)
+
+
+class QKVParallelLinear(ColumnParallelLinear):
+    """Fused QKV projection; q/k/v head groups are sharded per rank.
+    Weight layout per rank: [q_shard | k_shard | v_shard]."""
+
+    def __init__(self, hidden: int, head_dim: int, num_q_heads: int,
+                 num_kv_heads: int, bias: bool, dtype=torch.bfloat16):
+        tp = get_tp_world_size()
+        assert num_q_heads % tp == 0 and num_kv_heads % tp == 0
+        self.head_dim = head_dim
+        self.nq, self.nkv = num_q_heads, num_kv_heads
+        self.nq_local = num_q_heads // tp
+        self.nkv_local = num_kv_heads // tp
+        out = (self.nq_local + 2 * self.nkv_local) * head_dim * tp  # per-rank x tp
+        super().__init__(hidden, out, bias, dtype)
+
+    def shard_qkv(self, q_full, k_full, v_full) -> torch.Tensor:
+        """Build this rank's fused weight (or bias) from full q/k/v tensors."""
+        r = get_tp_rank()
+        hd = self.head_dim
+        qs = q_full[r * self.nq_local * hd : (r + 1) * self.nq_local * hd]
+        ks = k_full[r * self.nkv_local * hd : (r + 1) * self.nkv_local * hd]
+        vs = v_full[r * self.nkv_local * hd : (r + 1) * self.nkv_local * hd]
+        return torch.cat([qs, ks, vs], dim=0)
+
+    def split_qkv(self, qkv: torch.Tensor):
+        hd = self.head_dim
+        q, k, v = qkv.split(
+            [self.nq_local * hd, self.nkv_local * hd, self.nkv_local * hd], dim=-1
+        )
+        return q, k, v
+
+
+class MergedColumnParallelLinear(ColumnParallelLinear):
+    """Two column-parallel projections fused (gate_proj|up_proj)."""
+
+    def __init__(self, in_features: int, each_out: int, bias: bool,
+                 dtype=torch.bfloat16):
+        self.each_out = each_out
+        super().__init__(in_features, 2 * each_out, bias, dtype)
+
+    def shard_merged(self, gate_full: torch.Tensor, up_full: torch.Tensor) -> torch.Tensor:
+        r = get_tp_rank()
+        tp = get_tp_world_size()
+        per = self.each_out // tp
+        return torch.cat(
+            [gate_full[r * per : (r + 1) * per], up_full[r * per : (r + 1) * per]], dim=0
+        )
+
+
+class RowParallelLinear(nn.Module):
+    """Y = sum_ranks X_shard W_shard^T, W column-sharded (input features
+    split); all-reduce over the TP group after the local GEMM."""
+
+    def __init__(self, in_features: int, out_features: int, bias: bool,
+                 dtype: torch.dtype = torch.bfloat16):
+        super().__init__()
+        tp = get_tp_world_size()
+        assert in_features % tp == 0, (in_features, tp)
+        self.in_per_rank = in_features // tp
+        self.weight = nn.Parameter(
+            torch.empty(out_features, self.in_per_rank, dtype=dtype),
+            requires_grad=False,
+        )
+        # bias applied once (post-reduce) on every rank identically
+        self.bias = (
+            nn.Parameter(torch.empty(out_features, dtype=dtype), requires_grad=False)
+            if bias
+            else None
+        )
+
+    def shard(self, full: torch.Tensor, param: str = "weight") -> torch.Tensor:
+        if param == "bias":
+            return full
+        r = get_tp_rank()
+        return full[:, r * self.in_per_rank : (r + 1) * self.in_per_rank]
+
+    def forward(self, x: torch.Tensor) -> torch.Tensor:
+        y = F.linear(x, self.weight)
+        y = tp_all_reduce(y)
+        if self.bias is not None:
+            y = y + self.bias
+        return y
+
+
+class ParallelLMHead(nn.Module):
+    """Vocab-sharded output projection; logits are all-gathered so every rank
+    samples identically."""
+
+    def __init__(self, hidden: int, vocab: int, dtype=torch.bfloat16):
+        super().__init__()
+        tp = get_tp_world_size()
+        # pad vocab shard up so it divides evenly
+        self.vocab = vocab
+        self.vocab_padded = ((vocab + tp - 1) // tp) * tp
+        self.per_rank = self.vocab_padded // tp
+        self.weight = nn.Parameter(
+            torch.empty(self.per_rank, hidden, dtype=dtype), requires_grad=False
+        )
+
+    def shard(self, full: torch.Tensor, param: str = "weight") -> torch.Tensor:
+        r = get_tp_rank()
+        lo, hi = r * self.per_rank, (r + 1) * self.per_rank
+        if hi <= full.shape[0]:
+            return full[lo:hi]
+        pad = torch.full(
+            (hi - min(hi, full.shape[0]), full.shape[1]), 0, dtype=full.dtype
+        )
+        return torch.cat([full[lo:], pad], dim=0)
+
+    def forward(self, x: torch.Tensor) -> torch.Tensor:
+        logits = F.linear(x, self.weight)
+        logits = tp_all_gather(logits, dim=-1)
+        return logits[..., : self.vocab]
