@@ -1,0 +1,171 @@
+#!/usr/bin/env python3
+"""Serving benchmark for the arks_amd engine (driver contract).
+
+Measures the BASELINE.json headline: output tok/s (whole node) for
+Qwen2.5-7B, bf16, synthetic data, random-init weights, plus p50 TTFT
+(reported inside config). A "step" is one continuous-batching engine step
+(decode over the full in-flight batch during the timed region).
+
+Modes:
+  --parallel dp  (default): each rank is an independent engine replica
+                 (TP=1) — weak scaling, the reference's replica-scaling shape
+                 (ArksApplication.spec.replicas).
+  --parallel tp : all ranks form one tensor-parallel engine over RCCL/xGMI.
+
+Launch (driver): python -m torch.distributed.run --nnodes=1 --nproc-per-node N
+  --master-addr 127.0.0.1 bench.py --gpus N --steps K --warmup W
+"""
+
+from __future__ import annotations
+
+import argparse
+import json
+import os
+import time
+
+import torch
+
+
+def parse_args():
+    p = argparse.ArgumentParser()
+    p.add_argument("--gpus", type=int, default=1)
+    p.add_argument("--steps", type=int, default=32)
+    p.add_argument("--warmup", type=int, default=8)
+    p.add_argument("--model", type=str, default="qwen2.5-7b")
+    p.add_argument("--parallel", choices=["dp", "tp"], default="dp")
+    p.add_argument("--batch", type=int, default=64, help="in-flight requests per engine")
+    p.add_argument("--input-len", type=int, default=1024)
+    p.add_argument("--seed", type=int, default=0)
+    return p.parse_args()
+
+
+def main():
+    args = parse_args()
+    rank = int(os.environ.get("RANK", 0))
+    world = int(os.environ.get("WORLD_SIZE", 1))
+    local_rank = int(os.environ.get("LOCAL_RANK", rank))
+    use_gpu = torch.cuda.is_available()
+    if use_gpu:
+        torch.cuda.set_device(local_rank)
+
+    import torch.distributed as dist
+
+    from arks_amd.config import EngineConfig
+    from arks_amd.engine import LLMEngine, SamplingParams
+    from arks_amd.parallel import comm as tp_comm
+
+    if args.parallel == "tp" and world > 1:
+        tp_comm.init_tp()  # process group == TP group
+    elif world > 1:
+        dist.init_process_group(
+            backend="nccl" if use_gpu else "gloo", rank=rank, world_size=world
+        )
+
+    def barrier():
+        if world > 1:
+            dist.barrier()
+        if use_gpu:
+            torch.cuda.synchronize()
+
+    # KV pool sized for this run (keeps init fast; cache could hold far more
+    # of the 288 GB).
+    horizon = args.input_len + args.warmup + args.steps + 128
+    blocks = (args.batch * horizon + 15) // 16 + 64
+    cfg = EngineConfig(
+        preset=args.model,
+        device="cuda" if use_gpu else "cpu",
+        max_num_seqs=max(args.batch, 256),
+        max_num_batched_tokens=max(8192, args.input_len * 2),
+        max_model_len=horizon,
+        kv_cache_blocks=blocks,
+        seed=args.seed,
+    )
+    t_load0 = time.time()
+    engine = LLMEngine(cfg)
+    load_s = time.time() - t_load0
+
+    # Synthetic requests: random token ids, long enough to stay in decode
+    # through the whole timed region.
+    g = torch.Generator().manual_seed(args.seed + (0 if args.parallel == "tp" else rank))
+    prompts = [
+        torch.randint(0, engine.model_cfg.vocab_size - 1, (args.input_len,), generator=g).tolist()
+        for _ in range(args.batch)
+    ]
+    sp = SamplingParams(max_tokens=args.warmup + args.steps + 64, ignore_eos=True)
+    for i, pr in enumerate(prompts):
+        engine.add_request(pr, sp, request_id=f"bench-{i}")
+
+    # Warmup: run all prefills (measuring TTFT of the first wave), then W
+    # decode steps.
+    ttft_ms = None
+    t0 = time.time()
+    while engine.scheduler.num_waiting > 0:
+        engine.step()
+        if ttft_ms is None:
+            if use_gpu:
+                torch.cuda.synchronize()
+            ttft_ms = (time.time() - t0) * 1000.0
+    for _ in range(args.warmup):
+        out = engine.step()
+    barrier()
+
+    # Timed region: exactly K decode steps over the full batch.
+    t_start = time.time()
+    tokens = 0
+    for _ in range(args.steps):
+        out = engine.step()
+        tokens += len(out)
+    if use_gpu:
+        torch.cuda.synchronize()
+    elapsed = time.time() - t_start
+    barrier()
+
+    # MAX elapsed across ranks; tokens aggregate over the job.
+    if world > 1:
+        te = torch.tensor([elapsed], dtype=torch.float64)
+        tk = torch.tensor([tokens], dtype=torch.float64)
+        if use_gpu:
+            te, tk = te.cuda(), tk.cuda()
+        dist.all_reduce(te, op=dist.ReduceOp.MAX)
+        if args.parallel == "dp":
+            dist.all_reduce(tk, op=dist.ReduceOp.SUM)  # replicas add up
+        elapsed = float(te.item())
+        tokens = int(tk.item())
+
+    assert tokens == (args.batch * args.steps * (world if args.parallel == "dp" else 1)), (
+        "a sequence finished or was preempted inside the timed region"
+    )
+
+    if rank == 0:
+        n_gpus = world if use_gpu else args.gpus
+        value = tokens / elapsed
+        result = {
+            "metric": "output_tok_s",
+            "value": round(value, 2),
+            "unit": "tok/s",
+            "n_gpus": n_gpus,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": round(elapsed / args.steps * 1000.0, 3),
+            "higher_is_better": True,
+            "scaling": "weak" if args.parallel == "dp" else "strong",
+            "vs_baseline": None,
+            "dtype": "bf16",
+            "data": "synthetic",
+            "config": {
+                "model": args.model,
+                "global_batch": args.batch * (world if args.parallel == "dp" else 1),
+                "seq_len": args.input_len,
+                "parallelism": f"{args.parallel}{world}",
+                "ttft_ms_p50": round(ttft_ms, 2) if ttft_ms is not None else None,
+                "weight_load_s": round(load_s, 2),
+            },
+        }
+        print(json.dumps(result))
+
+    if world > 1:
+        dist.destroy_process_group()
+
+
+if __name__ == "__main__":
+    main()
