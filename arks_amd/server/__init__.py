@@ -1,0 +1,1 @@
+from .async_engine import AsyncEngine, worker_loop  # noqa: F401

@@ -1,0 +1,171 @@
+"""Asyncio wrapper around LLMEngine for the HTTP server, including the TP
+driver/worker protocol.
+
+TP>1 runs SPMD: every rank owns an identical scheduler+runner; rank 0
+broadcasts {adds, aborts} before each step so all schedulers take identical
+decisions, and the model's collectives (all-reduce / all-gather) keep ranks
+in lockstep. Sampling noise comes from each rank's identically-seeded
+generator, so sampled tokens agree without an extra broadcast.
+"""
+
+from __future__ import annotations
+
+import asyncio
+import concurrent.futures
+import time
+from dataclasses import dataclass, field
+
+from ..config import EngineConfig
+from ..engine import LLMEngine, SamplingParams, StepOutput
+from ..parallel.comm import get_tp_world_size, tp_broadcast_object
+from .metrics import EngineMetrics
+
+
+@dataclass
+class RequestAdd:
+    request_id: str
+    token_ids: list[int]
+    sampling: dict  # SamplingParams fields (picklable for broadcast)
+
+
+def apply_msg(engine: LLMEngine, msg: dict) -> None:
+    for add in msg.get("adds", ()):
+        engine.add_request(
+            add.token_ids, SamplingParams(**add.sampling), add.request_id
+        )
+    for rid in msg.get("aborts", ()):
+        engine.abort_request(rid)
+
+
+def worker_loop(cfg: EngineConfig) -> None:
+    """Ranks > 0: follow the driver's broadcasts forever."""
+    engine = LLMEngine(cfg)
+    while True:
+        msg = tp_broadcast_object(None, src=0)
+        if msg is None or msg.get("stop"):
+            break
+        apply_msg(engine, msg)
+        engine.step()
+
+
+@dataclass
+class _Stream:
+    queue: asyncio.Queue = field(default_factory=asyncio.Queue)
+    prev_token_time: float | None = None
+
+
+class AsyncEngine:
+    """Drives LLMEngine from an asyncio loop (rank 0)."""
+
+    def __init__(self, cfg: EngineConfig, model_name: str = "model"):
+        self.cfg = cfg
+        self.engine = LLMEngine(cfg)
+        self.metrics = EngineMetrics(model_name)
+        self.streams: dict[str, _Stream] = {}
+        self.pending_adds: list[RequestAdd] = []
+        self.pending_aborts: list[str] = []
+        self._wakeup: asyncio.Event | None = None
+        self._task: asyncio.Task | None = None
+        # Collectives + forward run on ONE dedicated thread.
+        self._executor = concurrent.futures.ThreadPoolExecutor(max_workers=1)
+        self._stopped = False
+
+    @property
+    def model_cfg(self):
+        return self.engine.model_cfg
+
+    async def start(self):
+        self._wakeup = asyncio.Event()
+        self._task = asyncio.get_running_loop().create_task(self._run())
+
+    async def stop(self):
+        self._stopped = True
+        if self._wakeup:
+            self._wakeup.set()
+        if self._task:
+            await self._task
+        if get_tp_world_size() > 1:
+            await asyncio.get_running_loop().run_in_executor(
+                self._executor, tp_broadcast_object, {"stop": True}, 0
+            )
+        self._executor.shutdown(wait=False)
+
+    # ---- request API ----
+    def submit(self, request_id: str, token_ids: list[int], sampling: SamplingParams) -> _Stream:
+        st = _Stream()
+        self.streams[request_id] = st
+        self.pending_adds.append(
+            RequestAdd(request_id, token_ids, sampling.__dict__.copy())
+        )
+        self.metrics.prompt_tokens.inc(len(token_ids))
+        if self._wakeup:
+            self._wakeup.set()
+        return st
+
+    def abort(self, request_id: str) -> None:
+        self.pending_aborts.append(request_id)
+        st = self.streams.pop(request_id, None)
+        if st:
+            st.queue.put_nowait(None)
+        if self._wakeup:
+            self._wakeup.set()
+
+    # ---- engine loop ----
+    def _sync_iteration(self, msg: dict) -> list[StepOutput]:
+        if get_tp_world_size() > 1:
+            tp_broadcast_object(msg, src=0)
+        apply_msg(self.engine, msg)
+        return self.engine.step()
+
+    async def _run(self):
+        loop = asyncio.get_running_loop()
+        while not self._stopped:
+            if not (self.pending_adds or self.pending_aborts or self.engine.has_work()):
+                self._wakeup.clear()
+                await self._wakeup.wait()
+                continue
+            msg = {"adds": self.pending_adds, "aborts": self.pending_aborts}
+            self.pending_adds, self.pending_aborts = [], []
+            outputs = await loop.run_in_executor(
+                self._executor, self._sync_iteration, msg
+            )
+            now = time.time()
+            for out in outputs:
+                st = self.streams.get(out.request_id)
+                if st is None:
+                    continue
+                if st.prev_token_time is None:
+                    # TTFT measured against request arrival
+                    seq = next(
+                        (s for s in self.engine.scheduler.running
+                         if s.request_id == out.request_id),
+                        None,
+                    )
+                    if seq is not None:
+                        self.metrics.ttft.observe(now - seq.arrival_time)
+                else:
+                    self.metrics.tpot.observe(now - st.prev_token_time)
+                st.prev_token_time = now
+                self.metrics.generation_tokens.inc()
+                st.queue.put_nowait(out)
+                if out.finished:
+                    self.metrics.request_success.inc()
+                    st.queue.put_nowait(None)  # sentinel
+                    self.streams.pop(out.request_id, None)
+            sched = self.engine.scheduler
+            self.metrics.num_requests_running.set(sched.num_running)
+            self.metrics.num_requests_waiting.set(sched.num_waiting)
+            alloc = sched.allocator
+            self.metrics.gpu_cache_usage_perc.set(
+                1.0 - alloc.num_free / max(alloc.num_blocks, 1)
+            )
+
+    async def generate_stream(self, request_id: str, token_ids: list[int],
+                              sampling: SamplingParams):
+        """Async iterator of StepOutputs for one request."""
+        st = self.submit(request_id, token_ids, sampling)
+        while True:
+            out = await st.queue.get()
+            if out is None:
+                return
+            yield out

@@ -1,0 +1,63 @@
+"""Prometheus metrics with vLLM-compatible names.
+
+The reference's runtime ServiceMonitor relabels metrics matching
+`(sglang|vllm|...)[:_].+` into a common `runtime` label space
+(reference config/prometheus/monitor-runtime.yaml:14-37), so the engine
+exports the vllm:* family it expects.
+"""
+
+from __future__ import annotations
+
+from prometheus_client import (
+    CollectorRegistry,
+    Counter,
+    Gauge,
+    Histogram,
+    generate_latest,
+)
+
+_TTFT_BUCKETS = (0.001, 0.005, 0.01, 0.02, 0.04, 0.06, 0.08, 0.1, 0.25, 0.5,
+                 0.75, 1.0, 2.5, 5.0, 7.5, 10.0)
+_TPOT_BUCKETS = (0.01, 0.025, 0.05, 0.075, 0.1, 0.15, 0.2, 0.3, 0.4, 0.5,
+                 0.75, 1.0, 2.5)
+_E2E_BUCKETS = (0.3, 0.5, 0.8, 1.0, 1.5, 2.0, 2.5, 5.0, 10.0, 15.0, 20.0,
+                30.0, 40.0, 50.0, 60.0)
+
+
+class EngineMetrics:
+    def __init__(self, model_name: str, registry: CollectorRegistry | None = None):
+        self.registry = registry or CollectorRegistry()
+        label = {"model_name": model_name}
+        mk = lambda cls, name, doc, **kw: cls(
+            name, doc, labelnames=["model_name"], registry=self.registry, **kw
+        ).labels(**label)
+        self.num_requests_running = mk(
+            Gauge, "vllm:num_requests_running", "running requests"
+        )
+        self.num_requests_waiting = mk(
+            Gauge, "vllm:num_requests_waiting", "waiting requests"
+        )
+        self.gpu_cache_usage_perc = mk(
+            Gauge, "vllm:gpu_cache_usage_perc", "KV cache usage fraction"
+        )
+        self.prompt_tokens = mk(Counter, "vllm:prompt_tokens", "prompt tokens")
+        self.generation_tokens = mk(
+            Counter, "vllm:generation_tokens", "generated tokens"
+        )
+        self.request_success = mk(Counter, "vllm:request_success", "finished requests")
+        self.ttft = mk(
+            Histogram, "vllm:time_to_first_token_seconds", "TTFT",
+            buckets=_TTFT_BUCKETS,
+        )
+        self.tpot = mk(
+            Histogram, "vllm:time_per_output_token_seconds", "TPOT",
+            buckets=_TPOT_BUCKETS,
+        )
+        self.e2e = mk(
+            Histogram, "vllm:e2e_request_latency_seconds", "E2E latency",
+            buckets=_E2E_BUCKETS,
+        )
+        self.preemptions = mk(Counter, "vllm:num_preemptions", "preemptions")
+
+    def render(self) -> bytes:
+        return generate_latest(self.registry)

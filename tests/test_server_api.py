@@ -1,0 +1,164 @@
+"""HTTP API tests (CPU, tiny model): OpenAI contract incl. the SSE
+final-usage-chunk shape the gateway depends on."""
+
+import asyncio
+import json
+
+import httpx
+import pytest
+
+from arks_amd.config import EngineConfig
+from arks_amd.server.api import create_app
+from arks_amd.server.async_engine import AsyncEngine
+from arks_amd.server.tokenizer import ByteTokenizer
+
+
+@pytest.fixture()
+def app():
+    cfg = EngineConfig(
+        preset="tiny", device="cpu", kv_cache_blocks=256, max_model_len=512
+    )
+    engine = AsyncEngine(cfg, model_name="tiny")
+    tok = ByteTokenizer(cfg.model_config().vocab_size, cfg.model_config().eos_token_id)
+    return create_app(engine, "tiny", tok)
+
+
+def run_with_client(app, fn):
+    async def go():
+        async with app.router.lifespan_context(app):
+            transport = httpx.ASGITransport(app=app)
+            async with httpx.AsyncClient(
+                transport=transport, base_url="http://t", timeout=60
+            ) as client:
+                await fn(client)
+
+    asyncio.new_event_loop().run_until_complete(go())
+
+
+def test_models_and_health(app):
+    async def fn(client):
+        r = await client.get("/health")
+        assert r.status_code == 200
+        r = await client.get("/v1/models")
+        assert r.json()["data"][0]["id"] == "tiny"
+
+    run_with_client(app, fn)
+
+
+def test_chat_completion_non_stream(app):
+    async def fn(client):
+        r = await client.post(
+            "/v1/chat/completions",
+            json={
+                "model": "tiny",
+                "messages": [{"role": "user", "content": "hi"}],
+                "max_tokens": 4,
+                "temperature": 0,
+                "ignore_eos": True,
+            },
+        )
+        assert r.status_code == 200, r.text
+        body = r.json()
+        assert body["object"] == "chat.completion"
+        assert body["usage"]["completion_tokens"] == 4
+        assert body["usage"]["total_tokens"] == body["usage"]["prompt_tokens"] + 4
+        assert body["choices"][0]["message"]["role"] == "assistant"
+
+    run_with_client(app, fn)
+
+
+def test_chat_completion_stream_final_usage_chunk(app):
+    """Streaming with include_usage: final chunk has EMPTY choices + usage,
+    then [DONE] — exactly what reference handle_response.go:113-133 parses."""
+
+    async def fn(client):
+        chunks = []
+        async with client.stream(
+            "POST",
+            "/v1/chat/completions",
+            json={
+                "model": "tiny",
+                "messages": [{"role": "user", "content": "hello"}],
+                "max_tokens": 3,
+                "temperature": 0,
+                "stream": True,
+                "stream_options": {"include_usage": True},
+                "ignore_eos": True,
+            },
+        ) as r:
+            assert r.status_code == 200
+            async for line in r.aiter_lines():
+                if line.startswith("data: "):
+                    chunks.append(line[len("data: "):])
+        assert chunks[-1] == "[DONE]"
+        final = json.loads(chunks[-2])
+        assert final["choices"] == []
+        assert final["usage"]["completion_tokens"] == 3
+        content_chunks = [json.loads(c) for c in chunks[:-2]]
+        assert content_chunks[0]["choices"][0]["delta"]["role"] == "assistant"
+        deltas = [
+            c["choices"][0]["delta"].get("content", "") for c in content_chunks[1:]
+        ]
+        assert len(deltas) == 3
+
+    run_with_client(app, fn)
+
+
+def test_wrong_model_404(app):
+    async def fn(client):
+        r = await client.post(
+            "/v1/chat/completions",
+            json={"model": "nope", "messages": [{"role": "user", "content": "x"}]},
+        )
+        assert r.status_code == 404
+        assert "error" in r.json()
+
+    run_with_client(app, fn)
+
+
+def test_completions_and_metrics(app):
+    async def fn(client):
+        r = await client.post(
+            "/v1/completions",
+            json={
+                "model": "tiny",
+                "prompt": "abc",
+                "max_tokens": 2,
+                "temperature": 0,
+                "ignore_eos": True,
+            },
+        )
+        assert r.status_code == 200, r.text
+        assert r.json()["usage"]["completion_tokens"] == 2
+        m = await client.get("/metrics")
+        text = m.text
+        assert "vllm:generation_tokens" in text
+        assert "vllm:num_requests_running" in text
+        assert "vllm:time_to_first_token_seconds" in text
+
+    run_with_client(app, fn)
+
+
+def test_concurrent_requests_batched(app):
+    """Multiple concurrent HTTP requests share engine steps (continuous
+    batching) and each gets its own complete stream."""
+
+    async def fn(client):
+        async def one(i):
+            r = await client.post(
+                "/v1/completions",
+                json={
+                    "model": "tiny",
+                    "prompt": f"request {i}",
+                    "max_tokens": 5,
+                    "temperature": 0,
+                    "ignore_eos": True,
+                },
+            )
+            assert r.status_code == 200
+            return r.json()["usage"]["completion_tokens"]
+
+        results = await asyncio.gather(*[one(i) for i in range(6)])
+        assert results == [5] * 6
+
+    run_with_client(app, fn)
