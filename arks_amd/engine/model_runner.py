@@ -39,6 +39,8 @@ class ModelRunner:
 
             load_model_weights(self.model, self.cfg.model_path, self.device)
         else:
+            # move first so random-init generates directly on the GPU
+            self.model.to(self.device)
             self.model.random_init(self.cfg.seed)
         self.model.to(self.device)
         self._weights_loaded = True

@@ -233,18 +233,51 @@ class LlamaFamilyForCausalLM(nn.Module):
 
     @torch.no_grad()
     def random_init(self, seed: int = 0) -> None:
-        g = torch.Generator().manual_seed(seed)
-        for p in self.parameters():
-            if p.dim() >= 2:
-                std = 0.02
-                p.data.copy_(
-                    torch.randn(p.shape, generator=g, dtype=torch.float32).mul_(std).to(p.dtype)
-                )
+        """Random-init with TP-consistent semantics: full HF-layout tensors
+        are generated deterministically per name (so every TP rank sees
+        slices of the SAME logical weights) and routed through the normal
+        shard-aware loader. Generates on the model's device when possible."""
+        import zlib
+
+        cfg = self.cfg
+        dev = self.embed_tokens.weight.device
+
+        def gen(name: str, *shape, std: float = 0.02) -> torch.Tensor:
+            s = (seed * 1000003 + zlib.crc32(name.encode())) % (2**63 - 1)
+            if dev.type == "cuda":
+                g = torch.Generator(device=dev).manual_seed(s)
+                t = torch.randn(shape, generator=g, dtype=torch.float32, device=dev)
             else:
-                p.data.zero_()
-        # norms to 1
-        for name, p in self.named_parameters():
-            if "layernorm" in name or name == "norm.weight":
-                p.data.fill_(1.0)
-            elif name.endswith("qkv_proj.bias"):
-                p.data.zero_()
+                g = torch.Generator().manual_seed(s)
+                t = torch.randn(shape, generator=g, dtype=torch.float32)
+            return t.mul_(std)
+
+        H, I, hd = cfg.hidden_size, cfg.intermediate_size, cfg.head_dim
+        nq, nkv = cfg.num_attention_heads, cfg.num_key_value_heads
+        glob: dict[str, torch.Tensor] = {
+            "model.embed_tokens.weight": gen("embed", cfg.vocab_size, H),
+            "model.norm.weight": torch.ones(H),
+        }
+        if not cfg.tie_word_embeddings:
+            glob["lm_head.weight"] = gen("lm_head", cfg.vocab_size, H)
+        self.load_hf_state_dict(glob)
+        del glob
+        for i in range(cfg.num_hidden_layers):
+            pre = f"model.layers.{i}"
+            tensors = {
+                f"{pre}.self_attn.q_proj.weight": gen(f"{pre}.q", nq * hd, H),
+                f"{pre}.self_attn.k_proj.weight": gen(f"{pre}.k", nkv * hd, H),
+                f"{pre}.self_attn.v_proj.weight": gen(f"{pre}.v", nkv * hd, H),
+                f"{pre}.self_attn.o_proj.weight": gen(f"{pre}.o", H, nq * hd),
+                f"{pre}.mlp.gate_proj.weight": gen(f"{pre}.gate", I, H),
+                f"{pre}.mlp.up_proj.weight": gen(f"{pre}.up", I, H),
+                f"{pre}.mlp.down_proj.weight": gen(f"{pre}.down", H, I),
+                f"{pre}.input_layernorm.weight": torch.ones(H),
+                f"{pre}.post_attention_layernorm.weight": torch.ones(H),
+            }
+            if cfg.attention_bias:
+                tensors[f"{pre}.self_attn.q_proj.bias"] = torch.zeros(nq * hd)
+                tensors[f"{pre}.self_attn.k_proj.bias"] = torch.zeros(nkv * hd)
+                tensors[f"{pre}.self_attn.v_proj.bias"] = torch.zeros(nkv * hd)
+            # load layer-by-layer to bound peak memory
+            self.load_hf_state_dict(tensors)

@@ -1,0 +1,93 @@
+"""Tensor-parallel correctness over gloo (world_size=2, CPU):
+the TP=2 SPMD engine must produce exactly the TP=1 greedy tokens."""
+
+import multiprocessing as mp
+import os
+import socket
+
+import pytest
+import torch
+
+PROMPTS = [[1, 5, 9, 20, 31, 7], [3, 3, 7, 90]]
+MAX_TOKENS = 6
+
+
+def _free_port():
+    with socket.socket() as s:
+        s.bind(("127.0.0.1", 0))
+        return s.getsockname()[1]
+
+
+def _tp1_reference():
+    from arks_amd.config import EngineConfig
+    from arks_amd.engine import LLMEngine, SamplingParams
+
+    torch.manual_seed(0)
+    e = LLMEngine(
+        EngineConfig(preset="tiny", device="cpu", kv_cache_blocks=128, max_model_len=512)
+    )
+    return e.generate(PROMPTS, SamplingParams(max_tokens=MAX_TOKENS, ignore_eos=True))
+
+
+def _tp_worker(rank: int, world: int, port: int, q):
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    os.environ["RANK"] = str(rank)
+    os.environ["WORLD_SIZE"] = str(world)
+    try:
+        from arks_amd.config import EngineConfig
+        from arks_amd.engine import LLMEngine, SamplingParams
+        from arks_amd.parallel import comm
+
+        comm.init_tp(backend="gloo")
+        torch.manual_seed(0)
+        e = LLMEngine(
+            EngineConfig(
+                preset="tiny", device="cpu", kv_cache_blocks=128, max_model_len=512
+            )
+        )
+        out = e.generate(PROMPTS, SamplingParams(max_tokens=MAX_TOKENS, ignore_eos=True))
+        if rank == 0:
+            q.put(("ok", out))
+        comm.destroy_tp()
+    except Exception as e:  # pragma: no cover
+        import traceback
+
+        q.put(("err", f"{e}\n{traceback.format_exc()}"))
+
+
+@pytest.mark.timeout(180)
+def test_tp2_matches_tp1_gloo():
+    ref = _tp1_reference()
+
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    port = _free_port()
+    procs = [ctx.Process(target=_tp_worker, args=(r, 2, port, q)) for r in range(2)]
+    for p in procs:
+        p.start()
+    status, payload = q.get(timeout=150)
+    for p in procs:
+        p.join(timeout=60)
+    assert status == "ok", payload
+    assert payload == ref, f"TP=2 output {payload} != TP=1 {ref}"
+
+
+@pytest.mark.timeout(180)
+def test_tp2_sharded_weights_random_init_consistent():
+    """random_init must give identical full tensors regardless of TP degree:
+    verified indirectly by the generation equality above; here check the
+    sharding math: column+row shard shapes on the tiny config."""
+    from arks_amd.config import PRESET_CONFIGS
+    from arks_amd.models import create_model
+
+    cfg = PRESET_CONFIGS["tiny"]
+    m = create_model(cfg)
+    assert m.layers[0].self_attn.qkv_proj.weight.shape == (
+        (cfg.num_attention_heads + 2 * cfg.num_key_value_heads) * cfg.head_dim,
+        cfg.hidden_size,
+    )
+    assert m.layers[0].mlp.gate_up_proj.weight.shape == (
+        2 * cfg.intermediate_size,
+        cfg.hidden_size,
+    )
