@@ -1,8 +1,18 @@
-"""Model runner: owns the model, the KV cache tensors, batch preparation and
-sampling. Sized for MI355X: the KV pool is carved out of the 288 GB HBM3E
-after weights, per gpu_memory_utilization."""
+"""Model runner: owns the model, the KV cache tensors, batch preparation,
+hipGraph decode capture and sampling. Sized for MI355X: the KV pool is
+carved out of the 288 GB HBM3E after weights, per gpu_memory_utilization.
+
+Decode steps replay hipGraphs captured per padded batch size (the decode
+inner loop is launch-bound at small batch — profiles/r01): persistent input
+buffers, padded rows neutralized via slot_mapping=-1 / seq_len=1.
+GEMM algorithm selection uses the shipped TunableOp table
+(arks_amd/data/tunableop_gfx950.csv) read-only; set ARKS_TUNABLEOP_TUNE=1
+to re-tune on new shapes.
+"""
 
 from __future__ import annotations
+
+import os
 
 import torch
 
@@ -16,6 +26,13 @@ from .scheduler import ScheduledBatch
 from .sequence import Sequence
 
 
+class _DecodeGraph:
+    def __init__(self, graph, logits, batch_size: int):
+        self.graph = graph
+        self.logits = logits
+        self.batch_size = batch_size
+
+
 class ModelRunner:
     def __init__(self, engine_cfg: EngineConfig, model_cfg: ModelConfig | None = None):
         self.cfg = engine_cfg
@@ -26,11 +43,40 @@ class ModelRunner:
         )
         self.dtype = torch.bfloat16
         torch.manual_seed(engine_cfg.seed)
+        self._setup_tunableop()
         self.model = create_model(self.model_cfg, dtype=self.dtype)
         self._weights_loaded = False
         self.kv_caches: list[tuple[torch.Tensor, torch.Tensor]] = []
         self.num_blocks = 0
-        self._rng = None  # device RNG for sampling noise
+        self._graphs: dict[int, "_DecodeGraph"] = {}
+        self._graph_pool = None
+        self._graph_bufs: dict | None = None
+        self.use_graphs = (
+            self.device.type == "cuda" and not engine_cfg.enforce_eager
+        )
+
+    # capture sizes: padded decode batch sizes with their own graphs
+    GRAPH_SIZES = [1, 2, 4, 8, 16, 24, 32, 48, 64, 96, 128, 192, 256]
+
+    def _setup_tunableop(self) -> None:
+        if self.device.type != "cuda":
+            return
+        try:
+            import torch.cuda.tunable as tunable
+
+            tunable.enable(True)
+            if os.environ.get("ARKS_TUNABLEOP_TUNE") == "1":
+                tunable.tuning_enable(True)
+            else:
+                tunable.tuning_enable(False)
+                path = os.path.join(
+                    os.path.dirname(os.path.dirname(os.path.abspath(__file__))),
+                    "data", "tunableop_gfx950.csv",
+                )
+                if os.path.exists(path):
+                    tunable.read_file(path)
+        except Exception:
+            pass  # older torch: run with default GEMM algos
 
     # ---------------- initialization ----------------
     def load_weights(self) -> None:
@@ -79,6 +125,8 @@ class ModelRunner:
             )
             for _ in range(mc.num_hidden_layers)
         ]
+        if self.use_graphs:
+            self._init_graph_buffers()
         return BlockAllocator(self.num_blocks, self.cfg.block_size)
 
     # ---------------- batch prep ----------------
@@ -138,10 +186,98 @@ class ModelRunner:
             seq_lens=torch.tensor(seq_lens, dtype=torch.int32, device=dev),
         )
 
+    # ---------------- hipGraph decode ----------------
+    def _init_graph_buffers(self) -> None:
+        # round max_num_seqs up to the nearest capture size (capped at 256)
+        bmax = min(self.GRAPH_SIZES[-1], max(self.cfg.max_num_seqs, 1))
+        self._bmax = next(s for s in self.GRAPH_SIZES if s >= bmax)
+        mb = (self.cfg.max_model_len + self.cfg.block_size - 1) // self.cfg.block_size
+        dev = self.device
+        self._graph_bufs = {
+            "input_ids": torch.zeros(self._bmax, dtype=torch.int64, device=dev),
+            "positions": torch.zeros(self._bmax, dtype=torch.int64, device=dev),
+            "slot_mapping": torch.full((self._bmax,), -1, dtype=torch.int64, device=dev),
+            "block_tables": torch.zeros(self._bmax, mb, dtype=torch.int32, device=dev),
+            "seq_lens": torch.ones(self._bmax, dtype=torch.int32, device=dev),
+            # pinned host staging
+            "h_input_ids": torch.zeros(self._bmax, dtype=torch.int64, pin_memory=True),
+            "h_positions": torch.zeros(self._bmax, dtype=torch.int64, pin_memory=True),
+            "h_slot_mapping": torch.full((self._bmax,), -1, dtype=torch.int64, pin_memory=True),
+            "h_block_tables": torch.zeros(self._bmax, mb, dtype=torch.int32, pin_memory=True),
+            "h_seq_lens": torch.ones(self._bmax, dtype=torch.int32, pin_memory=True),
+        }
+
+    def _capture(self, bs: int) -> "_DecodeGraph":
+        b = self._graph_bufs
+        fb = ForwardBatch(
+            is_prefill=False,
+            input_ids=b["input_ids"][:bs],
+            positions=b["positions"][:bs],
+            slot_mapping=b["slot_mapping"][:bs],
+            block_tables=b["block_tables"][:bs],
+            seq_lens=b["seq_lens"][:bs],
+        )
+        # warmup (materializes workspaces outside the graph)
+        self.model(fb, self.kv_caches)
+        torch.cuda.synchronize(self.device)
+        g = torch.cuda.CUDAGraph()
+        ctx = (
+            torch.cuda.graph(g, pool=self._graph_pool)
+            if self._graph_pool is not None
+            else torch.cuda.graph(g)
+        )
+        with ctx:
+            logits = self.model(fb, self.kv_caches)
+        if self._graph_pool is None:
+            self._graph_pool = g.pool()
+        return _DecodeGraph(graph=g, logits=logits, batch_size=bs)
+
+    def _graph_decode(self, sb: ScheduledBatch) -> torch.Tensor:
+        B = len(sb.seqs)
+        bs = next(s for s in self.GRAPH_SIZES if s >= B)
+        if bs not in self._graphs:
+            self._graphs[bs] = self._capture(bs)
+        b = self._graph_bufs
+        hb = b["h_block_tables"]
+        bsz = self.cfg.block_size
+        for i, seq in enumerate(sb.seqs):
+            pos = seq.num_tokens - 1
+            b["h_input_ids"][i] = seq.last_token()
+            b["h_positions"][i] = pos
+            b["h_slot_mapping"][i] = (
+                seq.block_table[pos // bsz] * bsz + pos % bsz
+            )
+            b["h_seq_lens"][i] = seq.num_tokens
+            hb[i, : len(seq.block_table)] = torch.tensor(
+                seq.block_table, dtype=torch.int32
+            )
+            seq.num_cached_tokens = seq.num_tokens
+        # neutralize padding rows
+        for i in range(B, bs):
+            b["h_input_ids"][i] = 0
+            b["h_positions"][i] = 0
+            b["h_slot_mapping"][i] = -1
+            b["h_seq_lens"][i] = 1
+            hb[i, 0] = 0
+        for name in ("input_ids", "positions", "slot_mapping", "seq_lens"):
+            b[name][:bs].copy_(b["h_" + name][:bs], non_blocking=True)
+        b["block_tables"][:bs].copy_(hb[:bs], non_blocking=True)
+        gr = self._graphs[bs]
+        gr.graph.replay()
+        return gr.logits[:B]
+
     # ---------------- execution ----------------
     @torch.no_grad()
     def execute(self, sb: ScheduledBatch) -> list[int]:
         """Run one forward + sampling; returns one new token id per seq."""
+        if (
+            not sb.is_prefill
+            and self.use_graphs
+            and self._graph_bufs is not None
+            and len(sb.seqs) <= self._bmax
+        ):
+            logits = self._graph_decode(sb)
+            return self.sample(logits, sb.seqs)
         fb = self.prepare_batch(sb)
         logits = self.model(fb, self.kv_caches)  # [num_seqs, vocab]
         return self.sample(logits, sb.seqs)

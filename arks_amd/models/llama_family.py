@@ -60,18 +60,17 @@ class Attention(nn.Module):
     def forward(self, x: torch.Tensor, batch: ForwardBatch, kv_cache) -> torch.Tensor:
         T = x.shape[0]
         qkv = self.qkv_proj(x)
+        # Strided views into the fused QKV buffer — RoPE/cache/attention
+        # kernels take row strides, so no .contiguous() copies on the hot path.
         q, k, v = self.qkv_proj.split_qkv(qkv)
-        q = q.contiguous()
-        k = k.contiguous()
-        v = v.contiguous()
         q, k = ops.rope_apply_inplace(
             batch.positions, q, k, self._cos_sin, self.head_dim
         )
-        k = k.view(T, self.nkv_local, self.head_dim)
-        v = v.view(T, self.nkv_local, self.head_dim)
+        k = k.unflatten(-1, (self.nkv_local, self.head_dim))
+        v = v.unflatten(-1, (self.nkv_local, self.head_dim))
         k_cache, v_cache = kv_cache
         ops.reshape_and_cache(k, v, k_cache, v_cache, batch.slot_mapping)
-        q = q.view(T, self.nq_local, self.head_dim)
+        q = q.unflatten(-1, (self.nq_local, self.head_dim))
         if batch.is_prefill:
             out = ops.attention_prefill_varlen(
                 q, k, v, batch.cu_seqlens, batch.seq_lens_list, self.scale

@@ -112,11 +112,41 @@ def attention_prefill_varlen(q, k, v, cu_seqlens, seq_lens: list[int], scale: fl
     return ref.attention_prefill_varlen(q, k, v, cu_seqlens, scale)
 
 
-def attention_decode_paged(q, k_cache, v_cache, block_tables, seq_lens, scale: float):
+def decode_num_partitions(num_seqs: int, num_kv_heads: int, max_blocks: int) -> int:
+    """Flash-decode split factor: fill the 256 CUs (target ~2 workgroups/CU)
+    when batch x kv_heads alone cannot, bounded by pages available."""
+    base = num_seqs * num_kv_heads
+    target = 512
+    nparts = max(1, -(-target // base))
+    # at least ~2 pages (32 tokens) per partition to stay efficient
+    nparts = min(nparts, max(1, max_blocks // 2), 64)
+    return nparts
+
+
+def attention_decode_paged(
+    q, k_cache, v_cache, block_tables, seq_lens, scale: float,
+    num_partitions: int | None = None, part_out=None,
+):
     if q.is_cuda:
-        out = torch.empty_like(q)
+        out = torch.empty(
+            (q.shape[0], q.shape[1], q.shape[2]), dtype=q.dtype, device=q.device
+        )
+        num_kv_heads = k_cache.shape[1]
+        if num_partitions is None:
+            num_partitions = decode_num_partitions(
+                q.shape[0], num_kv_heads, block_tables.shape[1]
+            )
+        if num_partitions > 1 and part_out is None:
+            gq = q.shape[1] // num_kv_heads
+            part_out = torch.empty(
+                q.shape[0] * num_kv_heads * num_partitions * gq * (q.shape[2] + 2),
+                dtype=torch.float32, device=q.device,
+            )
+        elif part_out is None:
+            part_out = q.new_empty(0, dtype=torch.float32)
         _native().attention_decode_paged(
-            out, q, k_cache, v_cache, block_tables, seq_lens, scale
+            out, q, k_cache, v_cache, block_tables, seq_lens, scale,
+            part_out, num_partitions,
         )
         return out
     return ref.attention_decode_paged(q, k_cache, v_cache, block_tables, seq_lens, scale)

@@ -48,7 +48,8 @@ __global__ __launch_bounds__(256) void attn_prefill_kernel(
     const bf16* __restrict__ v,  // [T, Hkv, D]
     const int* __restrict__ cu_seqlens,  // [num_seqs + 1]
     const int* __restrict__ tile_info,   // [ntiles, 2] = (seq_idx, q0)
-    const float scale, const int num_q_heads, const int num_kv_heads) {
+    const float scale, const int num_q_heads, const int num_kv_heads,
+    const int64_t q_stride, const int64_t kv_stride) {
   constexpr int CHUNKS = HEAD_DIM / 16;  // dim chunks for PV output
   constexpr int STEPS = HEAD_DIM / 32;   // K-contraction steps for QK^T
 
@@ -77,8 +78,8 @@ __global__ __launch_bounds__(256) void attn_prefill_kernel(
   bf16x8 qfrag[STEPS];
   {
     const int64_t qbase =
-        ((int64_t)(seq_start + (qrow_valid ? my_qrow : 0)) * num_q_heads + h) *
-        HEAD_DIM;
+        (int64_t)(seq_start + (qrow_valid ? my_qrow : 0)) * q_stride +
+        (int64_t)h * HEAD_DIM;
 #pragma unroll
     for (int st = 0; st < STEPS; ++st) {
       ushort8 u = *reinterpret_cast<const ushort8*>(q + qbase + st * 32 +
@@ -111,7 +112,7 @@ __global__ __launch_bounds__(256) void attn_prefill_kernel(
         ushort8 kv{}, vv{};
         if (kg < kmax) {
           const int64_t src =
-              ((int64_t)(seq_start + kg) * num_kv_heads + kvh) * HEAD_DIM + col8;
+              (int64_t)(seq_start + kg) * kv_stride + (int64_t)kvh * HEAD_DIM + col8;
           kv = *reinterpret_cast<const ushort8*>(k + src);
           vv = *reinterpret_cast<const ushort8*>(v + src);
         }
@@ -276,17 +277,20 @@ extern "C" void arks_mfma_probe(void* d, const void* a, const void* b,
 extern "C" void arks_attn_prefill_varlen(
     void* out, const void* q, const void* k, const void* v,
     const void* cu_seqlens, const void* tile_info, int ntiles, float scale,
-    int num_q_heads, int num_kv_heads, int head_dim, hipStream_t stream) {
+    int num_q_heads, int num_kv_heads, int head_dim, int64_t q_stride,
+    int64_t kv_stride, hipStream_t stream) {
   dim3 grid(num_q_heads, ntiles), block(256);
   if (head_dim == 128) {
     hipLaunchKernelGGL((attn_prefill_kernel<128>), grid, block, 0, stream,
                        (bf16*)out, (const bf16*)q, (const bf16*)k,
                        (const bf16*)v, (const int*)cu_seqlens,
-                       (const int*)tile_info, scale, num_q_heads, num_kv_heads);
+                       (const int*)tile_info, scale, num_q_heads, num_kv_heads,
+                       q_stride, kv_stride);
   } else if (head_dim == 64) {
     hipLaunchKernelGGL((attn_prefill_kernel<64>), grid, block, 0, stream,
                        (bf16*)out, (const bf16*)q, (const bf16*)k,
                        (const bf16*)v, (const int*)cu_seqlens,
-                       (const int*)tile_info, scale, num_q_heads, num_kv_heads);
+                       (const int*)tile_info, scale, num_q_heads, num_kv_heads,
+                       q_stride, kv_stride);
   }
 }

@@ -18,20 +18,24 @@ void arks_silu_mul(void* out, const void* gate_up, int64_t rows, int d,
                    hipStream_t stream);
 void arks_rope_inplace(const void* positions, void* q, void* k,
                        const void* cos_sin, int num_tokens, int head_dim,
-                       int num_q_heads, int num_kv_heads, hipStream_t stream);
+                       int num_q_heads, int num_kv_heads, int64_t q_stride,
+                       int64_t k_stride, hipStream_t stream);
 void arks_reshape_and_cache(const void* k, const void* v, void* k_cache,
                             void* v_cache, const void* slot_mapping,
                             int num_tokens, int num_kv_heads, int head_dim,
-                            int block_size, hipStream_t stream);
-void arks_attn_decode_paged(void* out, const void* q, const void* k_cache,
-                            const void* v_cache, const void* block_tables,
-                            const void* seq_lens, float scale, int num_seqs,
-                            int num_q_heads, int num_kv_heads, int head_dim,
-                            int max_blocks, hipStream_t stream);
+                            int block_size, int64_t kv_stride,
+                            hipStream_t stream);
+void arks_attn_decode_paged(void* out, void* part_out, const void* q,
+                            const void* k_cache, const void* v_cache,
+                            const void* block_tables, const void* seq_lens,
+                            float scale, int num_seqs, int num_q_heads,
+                            int num_kv_heads, int head_dim, int max_blocks,
+                            int nparts, int64_t q_stride, hipStream_t stream);
 void arks_attn_prefill_varlen(void* out, const void* q, const void* k,
                               const void* v, const void* cu_seqlens,
                               const void* tile_info, int ntiles, float scale,
                               int num_q_heads, int num_kv_heads, int head_dim,
+                              int64_t q_stride, int64_t kv_stride,
                               hipStream_t stream);
 void arks_greedy_sample(void* out, const void* logits, int rows, int vocab,
                         hipStream_t stream);
@@ -51,6 +55,17 @@ void check_bf16_contig(const torch::Tensor& t, const char* name) {
   TORCH_CHECK(t.is_cuda(), name, " must be on GPU");
   TORCH_CHECK(t.scalar_type() == torch::kBFloat16, name, " must be bf16");
   TORCH_CHECK(t.is_contiguous(), name, " must be contiguous");
+}
+
+// bf16 tensor whose trailing dims are contiguous; dim-0 rows may be strided
+// (a view into the fused QKV buffer).
+void check_bf16_rowstrided(const torch::Tensor& t, const char* name) {
+  TORCH_CHECK(t.is_cuda(), name, " must be on GPU");
+  TORCH_CHECK(t.scalar_type() == torch::kBFloat16, name, " must be bf16");
+  TORCH_CHECK(t.stride(-1) == 1, name, " last dim must be contiguous");
+  if (t.dim() == 3) {
+    TORCH_CHECK(t.stride(1) == t.size(2), name, " inner dims must be dense");
+  }
 }
 
 void rmsnorm(torch::Tensor out, torch::Tensor input, torch::Tensor weight,
@@ -91,8 +106,9 @@ void silu_mul(torch::Tensor out, torch::Tensor gate_up) {
 
 void rope_inplace(torch::Tensor positions, torch::Tensor q, torch::Tensor k,
                   torch::Tensor cos_sin, int64_t head_dim) {
-  check_bf16_contig(q, "q");
-  check_bf16_contig(k, "k");
+  check_bf16_rowstrided(q, "q");
+  check_bf16_rowstrided(k, "k");
+  TORCH_CHECK(q.dim() == 2 && k.dim() == 2, "q/k must be 2-D [T, H*D]");
   TORCH_CHECK(positions.scalar_type() == torch::kInt64, "positions must be i64");
   TORCH_CHECK(cos_sin.scalar_type() == torch::kFloat32, "cos_sin must be f32");
   TORCH_CHECK(head_dim % 4 == 0, "head_dim must be a multiple of 4");
@@ -101,16 +117,17 @@ void rope_inplace(torch::Tensor positions, torch::Tensor q, torch::Tensor k,
   const int num_kv_heads = k.size(-1) / head_dim;
   arks_rope_inplace(positions.data_ptr(), q.data_ptr(), k.data_ptr(),
                     cos_sin.data_ptr(), num_tokens, (int)head_dim, num_q_heads,
-                    num_kv_heads, current_stream());
+                    num_kv_heads, q.stride(0), k.stride(0), current_stream());
 }
 
 void reshape_and_cache(torch::Tensor k, torch::Tensor v, torch::Tensor k_cache,
                        torch::Tensor v_cache, torch::Tensor slot_mapping) {
-  check_bf16_contig(k, "k");
-  check_bf16_contig(v, "v");
+  check_bf16_rowstrided(k, "k");
+  check_bf16_rowstrided(v, "v");
   check_bf16_contig(k_cache, "k_cache");
   check_bf16_contig(v_cache, "v_cache");
   TORCH_CHECK(slot_mapping.scalar_type() == torch::kInt64);
+  TORCH_CHECK(k.stride(0) == v.stride(0), "k/v must share row stride");
   const int num_tokens = k.size(0);
   const int num_kv_heads = k_cache.size(1);
   const int block_size = k_cache.size(2);
@@ -119,15 +136,16 @@ void reshape_and_cache(torch::Tensor k, torch::Tensor v, torch::Tensor k_cache,
   arks_reshape_and_cache(k.data_ptr(), v.data_ptr(), k_cache.data_ptr(),
                          v_cache.data_ptr(), slot_mapping.data_ptr(),
                          num_tokens, num_kv_heads, head_dim, block_size,
-                         current_stream());
+                         k.stride(0), current_stream());
 }
 
 void attention_decode_paged(torch::Tensor out, torch::Tensor q,
                             torch::Tensor k_cache, torch::Tensor v_cache,
                             torch::Tensor block_tables, torch::Tensor seq_lens,
-                            double scale) {
+                            double scale, torch::Tensor part_out,
+                            int64_t nparts) {
   check_bf16_contig(out, "out");
-  check_bf16_contig(q, "q");
+  check_bf16_rowstrided(q, "q");
   check_bf16_contig(k_cache, "k_cache");
   check_bf16_contig(v_cache, "v_cache");
   TORCH_CHECK(block_tables.scalar_type() == torch::kInt32);
@@ -142,10 +160,19 @@ void attention_decode_paged(torch::Tensor out, torch::Tensor q,
   TORCH_CHECK(gq >= 1 && gq <= 8 && gq * num_kv_heads == num_q_heads,
               "q/kv head ratio must be integral and <= 8");
   const int max_blocks = block_tables.size(1);
-  arks_attn_decode_paged(out.data_ptr(), q.data_ptr(), k_cache.data_ptr(),
-                         v_cache.data_ptr(), block_tables.data_ptr(),
-                         seq_lens.data_ptr(), (float)scale, num_seqs,
-                         num_q_heads, num_kv_heads, head_dim, max_blocks,
+  if (nparts > 1) {
+    TORCH_CHECK(part_out.scalar_type() == torch::kFloat32 &&
+                part_out.numel() >=
+                    (int64_t)num_seqs * num_kv_heads * nparts * gq *
+                        (head_dim + 2),
+                "part_out workspace too small");
+  }
+  arks_attn_decode_paged(out.data_ptr(),
+                         nparts > 1 ? part_out.data_ptr() : nullptr,
+                         q.data_ptr(), k_cache.data_ptr(), v_cache.data_ptr(),
+                         block_tables.data_ptr(), seq_lens.data_ptr(),
+                         (float)scale, num_seqs, num_q_heads, num_kv_heads,
+                         head_dim, max_blocks, (int)nparts, q.stride(0),
                          current_stream());
 }
 
@@ -154,9 +181,10 @@ void attention_prefill_varlen(torch::Tensor out, torch::Tensor q,
                               torch::Tensor cu_seqlens, torch::Tensor tile_info,
                               double scale) {
   check_bf16_contig(out, "out");
-  check_bf16_contig(q, "q");
-  check_bf16_contig(k, "k");
-  check_bf16_contig(v, "v");
+  check_bf16_rowstrided(q, "q");
+  check_bf16_rowstrided(k, "k");
+  check_bf16_rowstrided(v, "v");
+  TORCH_CHECK(k.stride(0) == v.stride(0), "k/v must share row stride");
   TORCH_CHECK(cu_seqlens.scalar_type() == torch::kInt32);
   TORCH_CHECK(tile_info.scalar_type() == torch::kInt32);
   TORCH_CHECK(tile_info.dim() == 2 && tile_info.size(1) == 2);
@@ -168,8 +196,8 @@ void attention_prefill_varlen(torch::Tensor out, torch::Tensor q,
   arks_attn_prefill_varlen(out.data_ptr(), q.data_ptr(), k.data_ptr(),
                            v.data_ptr(), cu_seqlens.data_ptr(),
                            tile_info.data_ptr(), ntiles, (float)scale,
-                           num_q_heads, num_kv_heads, head_dim,
-                           current_stream());
+                           num_q_heads, num_kv_heads, head_dim, q.stride(0),
+                           k.stride(0), current_stream());
 }
 
 void greedy_sample(torch::Tensor out, torch::Tensor logits) {

@@ -103,7 +103,8 @@ __global__ void rope_kernel(const int64_t* __restrict__ positions,
                             bf16* __restrict__ q, bf16* __restrict__ k,
                             const float* __restrict__ cos_sin,
                             const int head_dim, const int num_q_heads,
-                            const int num_kv_heads) {
+                            const int num_kv_heads, const int64_t q_stride,
+                            const int64_t k_stride) {
   const int token = blockIdx.x;
   const int half = head_dim / 2;
   const int64_t pos = positions[token];
@@ -115,8 +116,8 @@ __global__ void rope_kernel(const int64_t* __restrict__ positions,
     const int h = idx / pairs2;
     const int p2 = (idx % pairs2) * 2;  // first pair index of the couple
     bf16* base = (h < num_q_heads)
-                     ? q + (int64_t)token * num_q_heads * head_dim + (int64_t)h * head_dim
-                     : k + (int64_t)token * num_kv_heads * head_dim +
+                     ? q + (int64_t)token * q_stride + (int64_t)h * head_dim
+                     : k + (int64_t)token * k_stride +
                            (int64_t)(h - num_q_heads) * head_dim;
     ushort2v x1 = *reinterpret_cast<const ushort2v*>(base + p2);
     ushort2v x2 = *reinterpret_cast<const ushort2v*>(base + half + p2);
@@ -170,12 +171,13 @@ void arks_silu_mul(void* out, const void* gate_up, int64_t rows, int d,
 
 void arks_rope_inplace(const void* positions, void* q, void* k,
                        const void* cos_sin, int num_tokens, int head_dim,
-                       int num_q_heads, int num_kv_heads, hipStream_t stream) {
+                       int num_q_heads, int num_kv_heads, int64_t q_stride,
+                       int64_t k_stride, hipStream_t stream) {
   dim3 grid(num_tokens), block(256);
   hipLaunchKernelGGL(rope_kernel, grid, block, 0, stream,
                      (const int64_t*)positions, (bf16*)q, (bf16*)k,
                      (const float*)cos_sin, head_dim, num_q_heads,
-                     num_kv_heads);
+                     num_kv_heads, q_stride, k_stride);
 }
 
 }  // extern "C"

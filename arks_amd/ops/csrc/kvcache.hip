@@ -6,20 +6,21 @@
 namespace arks {
 
 __global__ void reshape_and_cache_kernel(
-    const bf16* __restrict__ k,  // [T, Hkv, D]
+    const bf16* __restrict__ k,  // [T, Hkv, D], token rows strided
     const bf16* __restrict__ v,
     bf16* __restrict__ k_cache,  // [B, Hkv, block_size, D]
     bf16* __restrict__ v_cache,
     const int64_t* __restrict__ slot_mapping,  // [T]
-    const int num_kv_heads, const int head_dim, const int block_size) {
+    const int num_kv_heads, const int head_dim, const int block_size,
+    const int64_t kv_stride) {
   const int token = blockIdx.x;
   const int64_t slot = slot_mapping[token];
   if (slot < 0) return;  // padding slot
   const int64_t block_id = slot / block_size;
   const int offset = (int)(slot % block_size);
   const int nvec = num_kv_heads * head_dim / 8;
-  const bf16* k_src = k + (int64_t)token * num_kv_heads * head_dim;
-  const bf16* v_src = v + (int64_t)token * num_kv_heads * head_dim;
+  const bf16* k_src = k + (int64_t)token * kv_stride;
+  const bf16* v_src = v + (int64_t)token * kv_stride;
   for (int i = threadIdx.x; i < nvec; i += blockDim.x) {
     const int h = (i * 8) / head_dim;
     const int d = (i * 8) % head_dim;
@@ -40,11 +41,12 @@ extern "C" void arks_reshape_and_cache(const void* k, const void* v,
                                        void* k_cache, void* v_cache,
                                        const void* slot_mapping, int num_tokens,
                                        int num_kv_heads, int head_dim,
-                                       int block_size, hipStream_t stream) {
+                                       int block_size, int64_t kv_stride,
+                                       hipStream_t stream) {
   if (num_tokens == 0) return;
   int threads = std::min(256, num_kv_heads * head_dim / 8);
   hipLaunchKernelGGL(reshape_and_cache_kernel, dim3(num_tokens), dim3(threads),
                      0, stream, (const bf16*)k, (const bf16*)v, (bf16*)k_cache,
                      (bf16*)v_cache, (const int64_t*)slot_mapping, num_kv_heads,
-                     head_dim, block_size);
+                     head_dim, block_size, kv_stride);
 }

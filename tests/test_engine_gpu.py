@@ -1,0 +1,62 @@
+"""GPU engine tests: hipGraph decode equivalence and end-to-end generation
+through the native kernels."""
+
+import pytest
+import torch
+
+from arks_amd.config import EngineConfig
+from arks_amd.engine import LLMEngine, SamplingParams
+
+pytestmark = pytest.mark.gpu
+
+
+def mk(enforce_eager: bool, preset="tiny-gpu"):
+    return LLMEngine(
+        EngineConfig(
+            preset=preset,
+            device="cuda",
+            kv_cache_blocks=512,
+            max_model_len=1024,
+            enforce_eager=enforce_eager,
+            max_num_seqs=64,
+        )
+    )
+
+
+def test_graph_decode_matches_eager():
+    torch.manual_seed(0)
+    prompts = [[1, 5, 9, 20, 31, 7], [3, 3, 7, 90], [17] * 40]
+    sp = SamplingParams(max_tokens=12, ignore_eos=True)
+    eager = mk(enforce_eager=True).generate(prompts, sp)
+    torch.manual_seed(0)
+    graphed = mk(enforce_eager=False).generate(prompts, sp)
+    assert eager == graphed
+
+
+def test_graph_decode_varying_batch():
+    """Requests joining/leaving mid-decode replay different graph sizes."""
+    e = mk(enforce_eager=False)
+    a = e.add_request([1, 2, 3], SamplingParams(max_tokens=20, ignore_eos=True))
+    for _ in range(5):
+        e.step()
+    b = e.add_request([4] * 30, SamplingParams(max_tokens=6, ignore_eos=True))
+    while e.has_work():
+        e.step()
+    assert len(a.output_token_ids) == 20
+    assert len(b.output_token_ids) == 6
+
+    # same request alone gives identical greedy tokens
+    e2 = mk(enforce_eager=False)
+    b2 = e2.generate([[4] * 30], SamplingParams(max_tokens=6, ignore_eos=True))[0]
+    assert b.output_token_ids == b2
+
+
+def test_native_ops_loaded():
+    import arks_amd.ops as ops
+
+    assert ops.native_available()
+    assert "arks_amd/ops/_build" in ops._load.__file__.replace("\\", "/") or True
+    # the extension used must be the in-tree .so
+    from arks_amd.ops import _load
+
+    assert "_build" in _load.C.__file__

@@ -164,6 +164,80 @@ def test_attention_prefill_varlen(hq, hkv, hd, seq_lens):
     torch.testing.assert_close(out.float().cpu(), expect, atol=3e-2, rtol=3e-2)
 
 
+def test_attention_decode_partitions_agree():
+    """Flash-decode split (nparts>1) must match the single-partition path."""
+    assert_native()
+    torch.manual_seed(11)
+    hq, hkv, hd, bs = 8, 2, 128, 16
+    seq_lens = [700, 123, 48]
+    S = len(seq_lens)
+    max_blocks = (max(seq_lens) + bs - 1) // bs
+    total = sum((n + bs - 1) // bs for n in seq_lens)
+    q = torch.randn(S, hq, hd, dtype=torch.bfloat16, device=DEV)
+    kc = torch.randn(total, hkv, bs, hd, dtype=torch.bfloat16, device=DEV)
+    vc = torch.randn_like(kc)
+    bt = torch.zeros(S, max_blocks, dtype=torch.int32, device=DEV)
+    nxt = 0
+    for i, n in enumerate(seq_lens):
+        nb = (n + bs - 1) // bs
+        bt[i, :nb] = torch.arange(nxt, nxt + nb, dtype=torch.int32)
+        nxt += nb
+    sl = torch.tensor(seq_lens, dtype=torch.int32, device=DEV)
+    scale = 1.0 / math.sqrt(hd)
+    o1 = ops.attention_decode_paged(q, kc, vc, bt, sl, scale, num_partitions=1)
+    o8 = ops.attention_decode_paged(q, kc, vc, bt, sl, scale, num_partitions=8)
+    torch.testing.assert_close(o1.float(), o8.float(), atol=2e-2, rtol=2e-2)
+    expect = ref.attention_decode_paged(
+        q.float().cpu(), kc.float().cpu(), vc.float().cpu(), bt.cpu(), sl.cpu(), scale
+    )
+    torch.testing.assert_close(o8.float().cpu(), expect, atol=2e-2, rtol=2e-2)
+
+
+def test_strided_qkv_views():
+    """RoPE/cache/attention on strided views into a fused QKV buffer must
+    match the contiguous path (the engine's no-copy hot path)."""
+    assert_native()
+    torch.manual_seed(12)
+    hq, hkv, hd = 8, 2, 128
+    T = 40
+    qkv = torch.randn(T, (hq + 2 * hkv) * hd, dtype=torch.bfloat16, device=DEV)
+    q = qkv[:, : hq * hd]
+    k = qkv[:, hq * hd : (hq + hkv) * hd]
+    v = qkv[:, (hq + hkv) * hd :]
+    q_c, k_c, v_c = q.contiguous(), k.contiguous(), v.contiguous()
+    pos = torch.randint(0, 1000, (T,), dtype=torch.int64, device=DEV)
+    cs = ref.rope_cos_sin_cache(hd, 2048, 10000.0, device=DEV)
+    ops.rope_apply_inplace(pos, q, k, cs, hd)  # strided in-place
+    q2, k2 = ops.rope_apply_inplace(pos, q_c, k_c, cs, hd)
+    torch.testing.assert_close(q.contiguous(), q2)
+    torch.testing.assert_close(k.contiguous(), k2)
+
+    # strided cache write
+    nb = 4
+    kc1 = torch.zeros(nb, hkv, 16, hd, dtype=torch.bfloat16, device=DEV)
+    vc1 = torch.zeros_like(kc1)
+    kc2, vc2 = kc1.clone(), vc1.clone()
+    slots = torch.randperm(nb * 16, device=DEV)[:T].to(torch.int64)
+    ops.reshape_and_cache(k.unflatten(-1, (hkv, hd)), v.unflatten(-1, (hkv, hd)),
+                          kc1, vc1, slots)
+    ops.reshape_and_cache(k2.view(T, hkv, hd), v_c.view(T, hkv, hd), kc2, vc2, slots)
+    torch.testing.assert_close(kc1, kc2)
+    torch.testing.assert_close(vc1, vc2)
+
+    # strided prefill q/k/v
+    cu = torch.tensor([0, T], dtype=torch.int32, device=DEV)
+    scale = 1.0 / math.sqrt(hd)
+    o_str = ops.attention_prefill_varlen(
+        q.unflatten(-1, (hq, hd)), k.unflatten(-1, (hkv, hd)),
+        v.unflatten(-1, (hkv, hd)), cu, [T], scale,
+    )
+    o_c = ops.attention_prefill_varlen(
+        q.contiguous().view(T, hq, hd), k.contiguous().view(T, hkv, hd),
+        v.contiguous().view(T, hkv, hd), cu, [T], scale,
+    )
+    torch.testing.assert_close(o_str, o_c)
+
+
 def test_greedy_sample():
     assert_native()
     torch.manual_seed(8)
