@@ -54,6 +54,7 @@ def main():
 
     tunable.enable(True)
     tunable.tuning_enable(True)
+    tunable.set_filename(args.out)  # flushed via atexit / set_filename
     if os.path.exists(args.out):
         tunable.read_file(args.out)  # extend the existing table
 
@@ -75,8 +76,8 @@ def main():
         del w, b
         torch.cuda.empty_cache()
 
-    tunable.write_file(args.out)
-    print(f"wrote {args.out}")
+    # results are flushed to the filename at interpreter exit
+    print(f"tuning done; results flush to {args.out} at exit")
 
 
 if __name__ == "__main__":
