@@ -154,6 +154,7 @@ class EngineConfig:
     max_model_len: int = 8192
     kv_cache_blocks: int | None = None  # override (else sized from free HBM)
     enforce_eager: bool = False  # disable hipGraph decode capture
+    enable_prefix_caching: bool = True  # content-addressed KV block reuse
     device: str = "cuda"
     dtype: str = "bfloat16"
     seed: int = 0

@@ -33,6 +33,12 @@ class LLMEngine:
         self.total_prompt_tokens = 0
         self.total_output_tokens = 0
 
+    @property
+    def prefix_cache_stats(self) -> tuple[int, int]:
+        """(hit_tokens, query_tokens) of the prefix cache (0, 0 if disabled)."""
+        alloc = self.scheduler.allocator
+        return getattr(alloc, "hit_tokens", 0), getattr(alloc, "query_tokens", 0)
+
     # ---- request API ----
     def add_request(
         self,

@@ -21,7 +21,7 @@ from ..config import EngineConfig, ModelConfig
 from ..models import create_model
 from ..parallel.comm import get_tp_world_size
 from .forward_batch import ForwardBatch
-from .kv_cache import BlockAllocator, kv_cache_block_bytes
+from .kv_cache import BlockAllocator, PrefixCachingAllocator, kv_cache_block_bytes
 from .scheduler import ScheduledBatch
 from .sequence import Sequence
 
@@ -127,38 +127,60 @@ class ModelRunner:
         ]
         if self.use_graphs:
             self._init_graph_buffers()
+        if self.cfg.enable_prefix_caching:
+            return PrefixCachingAllocator(self.num_blocks, self.cfg.block_size)
         return BlockAllocator(self.num_blocks, self.cfg.block_size)
 
     # ---------------- batch prep ----------------
     def prepare_batch(self, sb: ScheduledBatch) -> ForwardBatch:
         bs = self.cfg.block_size
         if sb.is_prefill:
+            # Cached prefixes (prefix cache hits) are skipped: only the new
+            # suffix runs as q rows; attention then reads the full paged KV
+            # through the extend kernel.
+            use_extend = any(s.num_cached_tokens > 0 for s in sb.seqs)
             input_ids: list[int] = []
             positions: list[int] = []
             slot_mapping: list[int] = []
             cu = [0]
-            seq_lens_list: list[int] = []
+            q_lens: list[int] = []
+            kv_lens: list[int] = []
             logits_idx: list[int] = []
             for seq in sb.seqs:
                 toks = seq.all_token_ids
                 n = len(toks)
-                input_ids.extend(toks)
-                positions.extend(range(n))
-                for pos in range(n):
+                c = seq.num_cached_tokens
+                input_ids.extend(toks[c:])
+                positions.extend(range(c, n))
+                for pos in range(c, n):
                     b = seq.block_table[pos // bs]
                     slot_mapping.append(b * bs + pos % bs)
-                cu.append(cu[-1] + n)
-                seq_lens_list.append(n)
+                cu.append(cu[-1] + (n - c))
+                q_lens.append(n - c)
+                kv_lens.append(n)
                 logits_idx.append(cu[-1] - 1)
                 seq.num_cached_tokens = n
             dev = self.device
+            block_tables = None
+            seq_lens = None
+            if use_extend:
+                max_blocks = max(len(s.block_table) for s in sb.seqs)
+                bt = torch.zeros((len(sb.seqs), max_blocks), dtype=torch.int32)
+                for i, seq in enumerate(sb.seqs):
+                    bt[i, : len(seq.block_table)] = torch.tensor(
+                        seq.block_table, dtype=torch.int32
+                    )
+                block_tables = bt.to(dev)
+                seq_lens = torch.tensor(kv_lens, dtype=torch.int32, device=dev)
             return ForwardBatch(
                 is_prefill=True,
                 input_ids=torch.tensor(input_ids, dtype=torch.int64, device=dev),
                 positions=torch.tensor(positions, dtype=torch.int64, device=dev),
                 slot_mapping=torch.tensor(slot_mapping, dtype=torch.int64, device=dev),
                 cu_seqlens=torch.tensor(cu, dtype=torch.int32, device=dev),
-                seq_lens_list=seq_lens_list,
+                seq_lens_list=q_lens,
+                block_tables=block_tables,
+                seq_lens=seq_lens,
                 logits_indices=torch.tensor(logits_idx, dtype=torch.int64, device=dev),
             )
         # decode
@@ -265,7 +287,7 @@ class ModelRunner:
             np_bt[B:bs, 0] = 0
         for name in ("input_ids", "positions", "slot_mapping", "seq_lens"):
             b[name][:bs].copy_(b["h_" + name][:bs], non_blocking=True)
-        b["block_tables"][:bs].copy_(hb[:bs], non_blocking=True)
+        b["block_tables"][:bs].copy_(b["h_block_tables"][:bs], non_blocking=True)
         gr = self._graphs[bs]
         gr.graph.replay()
         return gr.logits[:B]

@@ -104,19 +104,30 @@ class Scheduler:
             # num_tokens (not num_prompt_tokens): a preempted sequence is
             # recomputed over prompt + already-generated tokens.
             n = seq.num_tokens
-            if seqs and n > budget:
+            toks = seq.all_token_ids
+            # Prefix cache: reuse cached full blocks of the prompt (refs are
+            # taken here and must be released on every non-admit path). Cap
+            # at n-1 so at least one token runs to produce logits.
+            reused, ncached = self.allocator.match_prefix(toks, n - 1)
+            n_new = n - ncached
+            if seqs and n_new > budget:
+                self.allocator.free(reused)
                 break
             if len(self.running) + len(seqs) >= self.max_num_seqs:
+                self.allocator.free(reused)
                 break
-            need = BlockAllocator.blocks_needed(n, self.allocator.block_size)
+            need = BlockAllocator.blocks_needed(n, self.allocator.block_size) - len(reused)
             if not self.allocator.can_allocate(need):
+                self.allocator.free(reused)
                 break
             self.waiting.popleft()
-            seq.block_table = self.allocator.allocate(need)
+            seq.block_table = reused + self.allocator.allocate(need)
+            self.allocator.register_prefix(toks, seq.block_table, len(reused))
+            seq.num_cached_tokens = ncached
             seq.status = SeqStatus.RUNNING
             seqs.append(seq)
-            ntoks.append(n)
-            budget -= n
+            ntoks.append(n_new)
+            budget -= n_new
             if budget <= 0:
                 break
         if not seqs:

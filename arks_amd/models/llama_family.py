@@ -72,9 +72,17 @@ class Attention(nn.Module):
         ops.reshape_and_cache(k, v, k_cache, v_cache, batch.slot_mapping)
         q = q.unflatten(-1, (self.nq_local, self.head_dim))
         if batch.is_prefill:
-            out = ops.attention_prefill_varlen(
-                q, k, v, batch.cu_seqlens, batch.seq_lens_list, self.scale
-            )
+            if batch.block_tables is not None:
+                # Prefix-cache hit in the batch: new tokens attend over the
+                # full paged KV (cached prefix + the rows just written above).
+                out = ops.attention_extend_paged(
+                    q, k_cache, v_cache, batch.block_tables, batch.seq_lens,
+                    batch.cu_seqlens, batch.seq_lens_list, self.scale,
+                )
+            else:
+                out = ops.attention_prefill_varlen(
+                    q, k, v, batch.cu_seqlens, batch.seq_lens_list, self.scale
+                )
         else:
             out = ops.attention_decode_paged(
                 q, k_cache, v_cache, batch.block_tables, batch.seq_lens, self.scale

@@ -112,6 +112,30 @@ def attention_prefill_varlen(q, k, v, cu_seqlens, seq_lens: list[int], scale: fl
     return ref.attention_prefill_varlen(q, k, v, cu_seqlens, scale)
 
 
+def attention_extend_paged(
+    q, k_cache, v_cache, block_tables, kv_lens, cu_seqlens_q,
+    q_lens: list[int], scale: float,
+):
+    """Causal attention of packed NEW tokens over each sequence's full paged
+    KV history (cached prefix + new tokens already written by
+    reshape_and_cache). The prefix-caching / chunked-prefill attention path.
+    q_lens is the host-side per-seq new-token count (tile grid without a
+    D2H sync); kv_lens is the device int32 total kv length per seq."""
+    if q.is_cuda:
+        out = torch.empty(
+            (q.shape[0], q.shape[1], q.shape[2]), dtype=q.dtype, device=q.device
+        )
+        tile_info = build_prefill_tiles(q_lens, q.device)
+        _native().attention_extend_paged(
+            out, q, k_cache, v_cache, block_tables, kv_lens, cu_seqlens_q,
+            tile_info, scale,
+        )
+        return out
+    return ref.attention_extend_paged(
+        q, k_cache, v_cache, block_tables, kv_lens, cu_seqlens_q, scale
+    )
+
+
 def decode_num_partitions(num_seqs: int, num_kv_heads: int, max_blocks: int) -> int:
     """Flash-decode split factor: fill the 256 CUs (target ~2 workgroups/CU)
     when batch x kv_heads alone cannot, bounded by pages available."""

@@ -24,6 +24,7 @@ SOURCES = [
     os.path.join(CSRC, "kvcache.hip"),
     os.path.join(CSRC, "attn_decode.hip"),
     os.path.join(CSRC, "attn_prefill.hip"),
+    os.path.join(CSRC, "attn_extend.hip"),
     os.path.join(CSRC, "sampling.hip"),
 ]
 

@@ -37,6 +37,13 @@ void arks_attn_prefill_varlen(void* out, const void* q, const void* k,
                               int num_q_heads, int num_kv_heads, int head_dim,
                               int64_t q_stride, int64_t kv_stride,
                               hipStream_t stream);
+void arks_attn_extend_paged(void* out, const void* q, const void* k_cache,
+                            const void* v_cache, const void* block_tables,
+                            const void* kv_lens, const void* cu_seqlens_q,
+                            const void* tile_info, int ntiles, float scale,
+                            int num_q_heads, int num_kv_heads, int head_dim,
+                            int max_blocks, int64_t q_stride,
+                            hipStream_t stream);
 void arks_greedy_sample(void* out, const void* logits, int rows, int vocab,
                         hipStream_t stream);
 void arks_gumbel_sample(void* out, const void* logits, const void* temperatures,
@@ -200,6 +207,35 @@ void attention_prefill_varlen(torch::Tensor out, torch::Tensor q,
                            k.stride(0), current_stream());
 }
 
+void attention_extend_paged(torch::Tensor out, torch::Tensor q,
+                            torch::Tensor k_cache, torch::Tensor v_cache,
+                            torch::Tensor block_tables, torch::Tensor kv_lens,
+                            torch::Tensor cu_seqlens_q, torch::Tensor tile_info,
+                            double scale) {
+  check_bf16_contig(out, "out");
+  check_bf16_rowstrided(q, "q");
+  check_bf16_contig(k_cache, "k_cache");
+  check_bf16_contig(v_cache, "v_cache");
+  TORCH_CHECK(block_tables.scalar_type() == torch::kInt32);
+  TORCH_CHECK(kv_lens.scalar_type() == torch::kInt32);
+  TORCH_CHECK(cu_seqlens_q.scalar_type() == torch::kInt32);
+  TORCH_CHECK(tile_info.scalar_type() == torch::kInt32);
+  TORCH_CHECK(tile_info.dim() == 2 && tile_info.size(1) == 2);
+  const int num_q_heads = q.size(1);
+  const int head_dim = q.size(2);
+  const int num_kv_heads = k_cache.size(1);
+  TORCH_CHECK(k_cache.size(2) == 16, "KV block size must be 16");
+  TORCH_CHECK(head_dim == 64 || head_dim == 128, "head_dim must be 64 or 128");
+  const int ntiles = tile_info.size(0);
+  const int max_blocks = block_tables.size(1);
+  arks_attn_extend_paged(out.data_ptr(), q.data_ptr(), k_cache.data_ptr(),
+                         v_cache.data_ptr(), block_tables.data_ptr(),
+                         kv_lens.data_ptr(), cu_seqlens_q.data_ptr(),
+                         tile_info.data_ptr(), ntiles, (float)scale,
+                         num_q_heads, num_kv_heads, head_dim, max_blocks,
+                         q.stride(0), current_stream());
+}
+
 void greedy_sample(torch::Tensor out, torch::Tensor logits) {
   check_bf16_contig(logits, "logits");
   TORCH_CHECK(out.scalar_type() == torch::kInt64);
@@ -239,6 +275,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("reshape_and_cache", &reshape_and_cache);
   m.def("attention_decode_paged", &attention_decode_paged);
   m.def("attention_prefill_varlen", &attention_prefill_varlen);
+  m.def("attention_extend_paged", &attention_extend_paged);
   m.def("greedy_sample", &greedy_sample);
   m.def("gumbel_sample", &gumbel_sample);
   m.def("mfma_probe", &mfma_probe);

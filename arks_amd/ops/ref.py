@@ -162,6 +162,46 @@ def attention_decode_paged(
     return out
 
 
+def attention_extend_paged(
+    q: torch.Tensor,  # [total_new_tokens, num_q_heads, head_dim]
+    k_cache: torch.Tensor,  # [num_blocks, num_kv_heads, block_size, head_dim]
+    v_cache: torch.Tensor,
+    block_tables: torch.Tensor,  # [num_seqs, max_blocks] int32
+    kv_lens: torch.Tensor,  # [num_seqs] int32 total kv length per seq
+    cu_seqlens_q: torch.Tensor,  # [num_seqs + 1] int32 (packed new tokens)
+    scale: float,
+) -> torch.Tensor:
+    """Causal attention of each sequence's new tokens over its full paged KV
+    (prefix caching / chunked prefill). New token j of seq i sits at global
+    position kv_len - q_len + j."""
+    num_q_heads, head_dim = q.shape[1], q.shape[2]
+    num_kv_heads = k_cache.shape[1]
+    block_size = k_cache.shape[2]
+    rep = num_q_heads // num_kv_heads
+    out = torch.empty_like(q)
+    for i in range(cu_seqlens_q.numel() - 1):
+        s, e = int(cu_seqlens_q[i]), int(cu_seqlens_q[i + 1])
+        qn = e - s
+        L = int(kv_lens[i])
+        off = L - qn
+        nblocks = (L + block_size - 1) // block_size
+        bt = block_tables[i, :nblocks].long()
+        ks = k_cache[bt].permute(0, 2, 1, 3).reshape(nblocks * block_size, num_kv_heads, head_dim)[:L]
+        vs = v_cache[bt].permute(0, 2, 1, 3).reshape(nblocks * block_size, num_kv_heads, head_dim)[:L]
+        kf = ks.float().repeat_interleave(rep, dim=1).transpose(0, 1)  # [H, L, D]
+        vf = vs.float().repeat_interleave(rep, dim=1).transpose(0, 1)
+        qi = q[s:e].float().transpose(0, 1)  # [H, qn, D]
+        scores = torch.matmul(qi, kf.transpose(-1, -2)) * scale  # [H, qn, L]
+        kpos = torch.arange(L, device=q.device)
+        qpos = torch.arange(off, L, device=q.device)
+        mask = kpos[None, :] > qpos[:, None]  # [qn, L] True = hidden
+        scores = scores.masked_fill(mask, float("-inf"))
+        p = torch.softmax(scores, dim=-1)
+        o = torch.matmul(p, vf)  # [H, qn, D]
+        out[s:e] = o.transpose(0, 1).to(q.dtype)
+    return out
+
+
 def greedy_sample(logits: torch.Tensor) -> torch.Tensor:
     """[num_seqs, vocab] -> [num_seqs] int64 argmax."""
     return logits.float().argmax(dim=-1)

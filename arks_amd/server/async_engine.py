@@ -159,6 +159,14 @@ class AsyncEngine:
             self.metrics.gpu_cache_usage_perc.set(
                 1.0 - alloc.num_free / max(alloc.num_blocks, 1)
             )
+            hits, queries = self.engine.prefix_cache_stats
+            if queries:  # counters are monotone: set via inc of the delta
+                self.metrics.prefix_cache_queries.inc(
+                    queries - self.metrics.prefix_cache_queries._value.get()
+                )
+                self.metrics.prefix_cache_hits.inc(
+                    hits - self.metrics.prefix_cache_hits._value.get()
+                )
 
     async def generate_stream(self, request_id: str, token_ids: list[int],
                               sampling: SamplingParams):

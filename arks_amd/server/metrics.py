@@ -58,6 +58,14 @@ class EngineMetrics:
             buckets=_E2E_BUCKETS,
         )
         self.preemptions = mk(Counter, "vllm:num_preemptions", "preemptions")
+        self.prefix_cache_queries = mk(
+            Counter, "vllm:prefix_cache_queries",
+            "prompt tokens queried against the prefix cache",
+        )
+        self.prefix_cache_hits = mk(
+            Counter, "vllm:prefix_cache_hits",
+            "prompt tokens served from the prefix cache",
+        )
 
     def render(self) -> bytes:
         return generate_latest(self.registry)

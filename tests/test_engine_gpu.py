@@ -60,3 +60,25 @@ def test_native_ops_loaded():
     from arks_amd.ops import _load
 
     assert "_build" in _load.C.__file__
+
+
+def test_prefix_cache_outputs_match_gpu():
+    """Cache-hit decode (extend kernel path) must reproduce the uncached
+    outputs bit-for-bit at temperature 0."""
+    torch.manual_seed(0)
+    prompts = [[5, 9, 2, 8] * 12, [5, 9, 2, 8] * 12, [3, 1, 4] * 11]
+    sp = SamplingParams(max_tokens=8, ignore_eos=True)
+    e = LLMEngine(EngineConfig(
+        preset="tiny-gpu", device="cuda", kv_cache_blocks=512,
+        max_model_len=1024, enable_prefix_caching=True, seed=11,
+    ))
+    first = e.generate(prompts, sp)
+    second = e.generate(prompts, sp)  # full-prefix hits
+    hits, queries = e.prefix_cache_stats
+    assert hits > 0
+    e0 = LLMEngine(EngineConfig(
+        preset="tiny-gpu", device="cuda", kv_cache_blocks=512,
+        max_model_len=1024, enable_prefix_caching=False, seed=11,
+    ))
+    base = e0.generate(prompts, sp)
+    assert first == base and second == base

@@ -272,3 +272,46 @@ def test_gumbel_sample_distribution():
     counts = torch.bincount(out, minlength=4).float() / N
     expect = torch.softmax(torch.tensor([2.0, 1.0, 0.0, -1.0]), dim=0)
     assert (counts.cpu() - expect).abs().max() < 0.02
+
+
+@pytest.mark.parametrize(
+    "hq,hkv,hd,spec",
+    [
+        # (kv_len, q_len) per seq; q_len < kv_len = cached prefix
+        (8, 2, 128, [(80, 80), (45, 13), (200, 1)]),
+        (28, 4, 128, [(512, 128), (100, 100)]),
+        (4, 4, 64, [(33, 17), (16, 1), (64, 64)]),
+    ],
+)
+def test_attention_extend_paged(hq, hkv, hd, spec):
+    """Extend kernel (paged KV, cached prefixes) vs torch fp32 reference."""
+    assert_native()
+    torch.manual_seed(13)
+    bs = 16
+    kv_lens = [s[0] for s in spec]
+    q_lens = [s[1] for s in spec]
+    S = len(spec)
+    Tq = sum(q_lens)
+    nb = [(n + bs - 1) // bs for n in kv_lens]
+    total_blocks = sum(nb) + 3
+    q = torch.randn(Tq, hq, hd, dtype=torch.bfloat16, device=DEV)
+    kc = torch.randn(total_blocks, hkv, bs, hd, dtype=torch.bfloat16, device=DEV)
+    vc = torch.randn_like(kc)
+    bt = torch.zeros(S, max(nb), dtype=torch.int32, device=DEV)
+    # scrambled non-contiguous block ids
+    perm = torch.randperm(total_blocks - 1) + 1
+    idx = 0
+    for i in range(S):
+        for j in range(nb[i]):
+            bt[i, j] = perm[idx]
+            idx += 1
+    cu_q = torch.tensor([0] + list(torch.tensor(q_lens).cumsum(0)),
+                        dtype=torch.int32, device=DEV)
+    kvl = torch.tensor(kv_lens, dtype=torch.int32, device=DEV)
+    scale = 1.0 / math.sqrt(hd)
+    out = ops.attention_extend_paged(q, kc, vc, bt, kvl, cu_q, q_lens, scale)
+    expect = ref.attention_extend_paged(
+        q.float().cpu(), kc.float().cpu(), vc.float().cpu(), bt.cpu(),
+        kvl.cpu(), cu_q.cpu(), scale,
+    )
+    torch.testing.assert_close(out.float().cpu(), expect, atol=3e-2, rtol=3e-2)
