@@ -46,14 +46,64 @@ class LLMEngine:
         sampling: SamplingParams | None = None,
         request_id: str | None = None,
         arrival_time: float | None = None,
+        hold_pages: bool = False,
     ) -> Sequence:
         seq = Sequence(prompt_token_ids, sampling, request_id, arrival_time)
+        seq.hold_pages = hold_pages
         self.scheduler.add(seq)
         self.total_prompt_tokens += seq.num_prompt_tokens
         return seq
 
     def abort_request(self, request_id: str) -> bool:
         return self.scheduler.abort(request_id)
+
+    # ---- prefill/decode disaggregation ----
+    def extract_prefilled(self, request_id: str):
+        """After a hold_pages request finished (1 token), pull its KV pages
+        for transfer and release them. Returns (prompt_len, kv tensor)."""
+        seq = self.scheduler.take_held(request_id)
+        if seq is None:
+            raise KeyError(f"no held sequence {request_id!r}")
+        # KV covers the prompt positions only (the generated token's KV is
+        # written by the decode instance's first step).
+        nb = (seq.num_prompt_tokens + self.cfg.block_size - 1) // self.cfg.block_size
+        kv = self.runner.extract_kv(seq.block_table[:nb])
+        self.scheduler.allocator.free(seq.block_table)
+        seq.block_table = []
+        return seq.num_prompt_tokens, kv
+
+    def add_prefilled(
+        self,
+        prompt_token_ids: list[int],
+        first_token: int,
+        kv,
+        sampling: SamplingParams | None = None,
+        request_id: str | None = None,
+        arrival_time: float | None = None,
+    ) -> Sequence:
+        """Decode-instance entry: admit a sequence whose prompt KV was
+        computed remotely. Injects the pages and enters RUNNING directly."""
+        from .kv_cache import BlockAllocator
+
+        seq = Sequence(prompt_token_ids, sampling, request_id, arrival_time)
+        need = BlockAllocator.blocks_needed(seq.num_prompt_tokens, self.cfg.block_size)
+        if not self.scheduler.allocator.can_allocate(need):
+            raise RuntimeError("KV cache exhausted on decode instance")
+        seq.block_table = self.scheduler.allocator.allocate(need)
+        self.runner.inject_kv(seq.block_table, kv)
+        self.scheduler.allocator.register_prefix(
+            prompt_token_ids, seq.block_table, 0
+        )
+        seq.num_cached_tokens = seq.num_prompt_tokens
+        seq.append_token(first_token)
+        self.total_prompt_tokens += seq.num_prompt_tokens
+        self.total_output_tokens += 1
+        if seq.check_finished(self.model_cfg.eos_token_id):
+            self.scheduler._release(seq)
+        else:
+            seq.status = SeqStatus.RUNNING
+            self.scheduler.running.append(seq)
+        return seq
 
     def has_work(self) -> bool:
         return self.scheduler.has_work()

@@ -131,6 +131,27 @@ class ModelRunner:
             return PrefixCachingAllocator(self.num_blocks, self.cfg.block_size)
         return BlockAllocator(self.num_blocks, self.cfg.block_size)
 
+    # ---------------- disaggregated prefill (KV page transfer) ----------------
+    def extract_kv(self, block_table: list[int]) -> torch.Tensor:
+        """Gather a sequence's KV pages to one host tensor
+        [layers, 2(k/v), nblocks, num_kv_heads, block_size, head_dim] for
+        transfer to a decode instance (the reference's PD
+        split delegates this to SGLang's disaggregation-mode, reference
+        arksdisaggregatedapplication_controller.go:1672-1724)."""
+        bt = torch.tensor(block_table, dtype=torch.long, device=self.device)
+        layers = [torch.stack((kc[bt], vc[bt]), 0) for kc, vc in self.kv_caches]
+        return torch.stack(layers, 0).cpu()
+
+    def inject_kv(self, block_table: list[int], kv: torch.Tensor) -> None:
+        """Scatter a transferred KV tensor (extract_kv layout) into this
+        runner's pages."""
+        assert kv.shape[0] == len(self.kv_caches) and kv.shape[2] == len(block_table)
+        bt = torch.tensor(block_table, dtype=torch.long, device=self.device)
+        kv = kv.to(device=self.device, dtype=self.dtype, non_blocking=True)
+        for li, (kc, vc) in enumerate(self.kv_caches):
+            kc[bt] = kv[li, 0]
+            vc[bt] = kv[li, 1]
+
     # ---------------- batch prep ----------------
     def prepare_batch(self, sb: ScheduledBatch) -> ForwardBatch:
         bs = self.cfg.block_size

@@ -46,6 +46,9 @@ class Scheduler:
         self.max_model_len = max_model_len
         self.waiting: deque[Sequence] = deque()
         self.running: list[Sequence] = []
+        # finished seqs with hold_pages: pages stay allocated until the
+        # disaggregation layer extracts their KV (release via take_held)
+        self.held: dict[str, Sequence] = {}
         self.num_preemptions = 0
 
     # --- API ---
@@ -191,5 +194,13 @@ class Scheduler:
     def free_finished(self) -> None:
         for seq in self.running:
             if seq.is_finished:
-                self._release(seq)
+                if seq.hold_pages:
+                    self.held[seq.request_id] = seq
+                else:
+                    self._release(seq)
         self.running = [s for s in self.running if not s.is_finished]
+
+    def take_held(self, request_id: str) -> Sequence | None:
+        """Remove and return a held finished sequence (pages still ref'd;
+        caller must allocator.free(seq.block_table) when done)."""
+        return self.held.pop(request_id, None)

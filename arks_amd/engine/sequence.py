@@ -45,6 +45,9 @@ class Sequence:
         self.sampling = sampling or SamplingParams()
         self.status = SeqStatus.WAITING
         self.block_table: list[int] = []
+        # Disaggregated prefill: keep KV pages allocated after finish so the
+        # decode instance can pull them (scheduler.held).
+        self.hold_pages = False
         self.num_cached_tokens = 0  # tokens whose KV is already in cache
         self.arrival_time = arrival_time if arrival_time is not None else time.time()
         self.first_token_time: float | None = None

@@ -32,6 +32,12 @@ def parse_args(argv=None):
     p.add_argument("--gpu-memory-utilization", type=float, default=0.90)
     p.add_argument("--kv-cache-blocks", type=int, default=None)
     p.add_argument("--enforce-eager", action="store_true")
+    p.add_argument("--disaggregation-mode", choices=["prefill", "decode"],
+                   default=None)
+    # multi-node group flags (LWS leader/worker topology): workers join the
+    # leader's torch.distributed rendezvous
+    p.add_argument("--leader-address", default=None)
+    p.add_argument("--node-rank", type=int, default=0)
     return p.parse_args(argv)
 
 
@@ -91,11 +97,12 @@ def main(argv=None):
     from arks_amd.server.tokenizer import load_tokenizer
 
     served = args.served_model_name or os.path.basename(args.model.rstrip("/"))
-    engine = AsyncEngine(cfg, model_name=served)
+    engine = AsyncEngine(cfg, model_name=served,
+                         disagg_mode=args.disaggregation_mode)
     tok = load_tokenizer(
         cfg.model_path, engine.model_cfg.vocab_size, engine.model_cfg.eos_token_id
     )
-    app = create_app(engine, served, tok)
+    app = create_app(engine, served, tok, disagg_mode=args.disaggregation_mode)
     uvicorn.run(app, host=args.host, port=args.port, log_level="info")
 
 

@@ -44,7 +44,8 @@ def _error(status: int, message: str) -> JSONResponse:
     )
 
 
-def create_app(engine: AsyncEngine, served_model_name: str, tokenizer) -> FastAPI:
+def create_app(engine: AsyncEngine, served_model_name: str, tokenizer,
+               disagg_mode: str | None = None) -> FastAPI:
     from contextlib import asynccontextmanager
 
     @asynccontextmanager
@@ -59,6 +60,11 @@ def create_app(engine: AsyncEngine, served_model_name: str, tokenizer) -> FastAP
     app.state.engine = engine
     app.state.tokenizer = tokenizer
     app.state.model_name = served_model_name
+
+    if disagg_mode == "prefill":
+        from .disagg import add_prefill_routes
+
+        add_prefill_routes(app, engine)
 
     @app.get("/health")
     async def health():
@@ -98,7 +104,9 @@ def create_app(engine: AsyncEngine, served_model_name: str, tokenizer) -> FastAP
             )
         text_ids: list[int] = []
         finish = None
-        async for out in engine.generate_stream(rid, token_ids, sp):
+        async for out in engine.generate_stream(
+            rid, token_ids, sp, prefill_addr=raw.headers.get("x-arks-prefill-addr")
+        ):
             text_ids.append(out.new_token_id)
             if out.finished:
                 finish = out.finish_reason
@@ -130,7 +138,9 @@ def create_app(engine: AsyncEngine, served_model_name: str, tokenizer) -> FastAP
         n_out = 0
         finish = None
         try:
-            async for out in engine.generate_stream(rid, token_ids, sp):
+            async for out in engine.generate_stream(
+            rid, token_ids, sp, prefill_addr=raw.headers.get("x-arks-prefill-addr")
+        ):
                 if await raw.is_disconnected():
                     engine.abort(rid)
                     return
@@ -186,7 +196,9 @@ def create_app(engine: AsyncEngine, served_model_name: str, tokenizer) -> FastAP
             )
         out_ids: list[int] = []
         finish = None
-        async for out in engine.generate_stream(rid, token_ids, sp):
+        async for out in engine.generate_stream(
+            rid, token_ids, sp, prefill_addr=raw.headers.get("x-arks-prefill-addr")
+        ):
             out_ids.append(out.new_token_id)
             if out.finished:
                 finish = out.finish_reason
@@ -206,7 +218,9 @@ def create_app(engine: AsyncEngine, served_model_name: str, tokenizer) -> FastAP
         include_usage = bool(req.stream_options and req.stream_options.include_usage)
         n_out = 0
         try:
-            async for out in engine.generate_stream(rid, token_ids, sp):
+            async for out in engine.generate_stream(
+            rid, token_ids, sp, prefill_addr=raw.headers.get("x-arks-prefill-addr")
+        ):
                 if await raw.is_disconnected():
                     engine.abort(rid)
                     return

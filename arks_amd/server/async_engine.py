@@ -26,12 +26,14 @@ class RequestAdd:
     request_id: str
     token_ids: list[int]
     sampling: dict  # SamplingParams fields (picklable for broadcast)
+    hold_pages: bool = False  # disaggregated prefill: keep pages for extract
 
 
 def apply_msg(engine: LLMEngine, msg: dict) -> None:
     for add in msg.get("adds", ()):
         engine.add_request(
-            add.token_ids, SamplingParams(**add.sampling), add.request_id
+            add.token_ids, SamplingParams(**add.sampling), add.request_id,
+            hold_pages=add.hold_pages,
         )
     for rid in msg.get("aborts", ()):
         engine.abort_request(rid)
@@ -57,10 +59,13 @@ class _Stream:
 class AsyncEngine:
     """Drives LLMEngine from an asyncio loop (rank 0)."""
 
-    def __init__(self, cfg: EngineConfig, model_name: str = "model"):
+    def __init__(self, cfg: EngineConfig, model_name: str = "model",
+                 disagg_mode: str | None = None):
         self.cfg = cfg
         self.engine = LLMEngine(cfg)
         self.metrics = EngineMetrics(model_name)
+        self.disagg_mode = disagg_mode  # None | "prefill" | "decode"
+        self.http_transport = None  # httpx transport override (tests)
         self.streams: dict[str, _Stream] = {}
         self.pending_adds: list[RequestAdd] = []
         self.pending_aborts: list[str] = []
@@ -91,13 +96,74 @@ class AsyncEngine:
         self._executor.shutdown(wait=False)
 
     # ---- request API ----
-    def submit(self, request_id: str, token_ids: list[int], sampling: SamplingParams) -> _Stream:
+    def submit(self, request_id: str, token_ids: list[int],
+               sampling: SamplingParams, hold_pages: bool = False) -> _Stream:
         st = _Stream()
         self.streams[request_id] = st
         self.pending_adds.append(
-            RequestAdd(request_id, token_ids, sampling.__dict__.copy())
+            RequestAdd(request_id, token_ids, sampling.__dict__.copy(), hold_pages)
         )
         self.metrics.prompt_tokens.inc(len(token_ids))
+        if self._wakeup:
+            self._wakeup.set()
+        return st
+
+    # ---- prefill/decode disaggregation (reference delegates this to
+    # SGLang's --disaggregation-mode; here it is first-party over a TCP KV
+    # page transfer — SURVEY.md §2.2) ----
+    async def disagg_prefill(self, request_id: str, token_ids: list[int],
+                             sampling: SamplingParams):
+        """Prefill-instance side: run the prompt, sample the first token,
+        return (first_token, finish_reason, kv pages tensor)."""
+        if get_tp_world_size() > 1:
+            raise NotImplementedError("disaggregation requires TP=1 instances")
+        import dataclasses
+
+        sp = dataclasses.replace(sampling, max_tokens=1)
+        st = self.submit(request_id, token_ids, sp, hold_pages=True)
+        out = await st.queue.get()
+        assert out is not None and out.finished
+        await st.queue.get()  # sentinel
+        loop = asyncio.get_running_loop()
+
+        def _extract():
+            _, kv = self.engine.extract_prefilled(request_id)
+            return kv
+
+        kv = await loop.run_in_executor(self._executor, _extract)
+        return out.new_token_id, out.finish_reason, kv
+
+    async def disagg_inject(self, request_id: str, token_ids: list[int],
+                            first_token: int, kv,
+                            sampling: SamplingParams) -> _Stream:
+        """Decode-instance side: admit a remotely prefilled sequence and
+        return its output stream (first token already queued)."""
+        if get_tp_world_size() > 1:
+            raise NotImplementedError("disaggregation requires TP=1 instances")
+        st = _Stream()
+        self.streams[request_id] = st
+        loop = asyncio.get_running_loop()
+
+        def _inject():
+            return self.engine.add_prefilled(
+                token_ids, first_token, kv, sampling, request_id
+            )
+
+        seq = await loop.run_in_executor(self._executor, _inject)
+        self.metrics.prompt_tokens.inc(len(token_ids))
+        self.metrics.generation_tokens.inc()
+        now = time.time()
+        self.metrics.ttft.observe(now - seq.arrival_time)
+        st.prev_token_time = now
+        st.queue.put_nowait(StepOutput(
+            request_id=request_id, seq_id=seq.seq_id, new_token_id=first_token,
+            finished=seq.is_finished, finish_reason=seq.finish_reason,
+            num_prompt_tokens=seq.num_prompt_tokens, num_output_tokens=1,
+        ))
+        if seq.is_finished:
+            self.metrics.request_success.inc()
+            st.queue.put_nowait(None)
+            self.streams.pop(request_id, None)
         if self._wakeup:
             self._wakeup.set()
         return st
@@ -169,9 +235,23 @@ class AsyncEngine:
                 )
 
     async def generate_stream(self, request_id: str, token_ids: list[int],
-                              sampling: SamplingParams):
-        """Async iterator of StepOutputs for one request."""
-        st = self.submit(request_id, token_ids, sampling)
+                              sampling: SamplingParams,
+                              prefill_addr: str | None = None):
+        """Async iterator of StepOutputs for one request. With a
+        prefill_addr (disaggregated decode instance), the prompt is prefilled
+        remotely and its KV pages pulled before decoding locally."""
+        if prefill_addr and self.disagg_mode == "decode":
+            from .disagg import remote_prefill
+
+            first, reason, kv = await remote_prefill(
+                prefill_addr, request_id, token_ids, sampling,
+                transport=self.http_transport,
+            )
+            st = await self.disagg_inject(
+                request_id, token_ids, first, kv, sampling
+            )
+        else:
+            st = self.submit(request_id, token_ids, sampling)
         while True:
             out = await st.queue.get()
             if out is None:
