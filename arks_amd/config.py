@@ -155,6 +155,9 @@ class EngineConfig:
     kv_cache_blocks: int | None = None  # override (else sized from free HBM)
     enforce_eager: bool = False  # disable hipGraph decode capture
     enable_prefix_caching: bool = True  # content-addressed KV block reuse
+    # None | "fp8": W8A8 OCP-e4m3 for qkv/gate_up/down/lm_head GEMMs
+    # (dynamic per-token activation scales; o_proj and KV stay bf16)
+    quantization: str | None = None
     device: str = "cuda"
     dtype: str = "bfloat16"
     seed: int = 0

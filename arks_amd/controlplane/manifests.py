@@ -287,7 +287,7 @@ def generate_download_pod(model: ArksModel) -> dict[str, Any]:
                 {
                     "name": "download",
                     "image": os.environ.get(SCRIPTS_IMAGE_ENV, DEFAULT_SCRIPTS_IMAGE),
-                    "command": ["python3", "/scripts/download.py"],
+                    "command": ["python3", "-m", "arks_amd.loader.download"],
                     "env": env,
                     "volumeMounts": [mount],
                     "terminationMessagePolicy": "FallbackToLogsOnError",

@@ -89,6 +89,10 @@ class ModelRunner:
             self.model.to(self.device)
             self.model.random_init(self.cfg.seed)
         self.model.to(self.device)
+        if self.cfg.quantization == "fp8":
+            self.model.quantize_fp8()
+        elif self.cfg.quantization is not None:
+            raise ValueError(f"unknown quantization {self.cfg.quantization!r}")
         self._weights_loaded = True
 
     def profile_num_blocks(self) -> int:

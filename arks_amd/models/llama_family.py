@@ -145,6 +145,19 @@ class LlamaFamilyForCausalLM(nn.Module):
         for layer in self.layers:
             layer.self_attn._cos_sin = cos_sin
 
+    def quantize_fp8(self) -> None:
+        """W8A8 fp8 for the weight-bandwidth-bound GEMMs (profiles/r01_p2:
+        gate_up/down/lm_head dominate the decode step; o_proj is left bf16 —
+        no gain at its size and its input is the attention output, which
+        has no producing kernel to fuse a quant into)."""
+        from ..parallel.layers import quantize_module_fp8
+
+        for layer in self.layers:
+            quantize_module_fp8(layer.self_attn.qkv_proj)
+            quantize_module_fp8(layer.mlp.gate_up_proj)
+            quantize_module_fp8(layer.mlp.down_proj)
+        quantize_module_fp8(self.lm_head)
+
     def _build_cos_sin(self):
         from ..ops import ref
 

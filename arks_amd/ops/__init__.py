@@ -176,6 +176,18 @@ def attention_decode_paged(
     return ref.attention_decode_paged(q, k_cache, v_cache, block_tables, seq_lens, scale)
 
 
+def quant_fp8_rows(x):
+    """Dynamic per-token fp8 e4m3 quantization: [M, K] bf16 ->
+    ([M, K] float8_e4m3fn, [M] fp32 dequant scales) for W8A8 GEMMs
+    (torch._scaled_mm rowwise)."""
+    if x.is_cuda:
+        out = torch.empty(x.shape, dtype=torch.float8_e4m3fn, device=x.device)
+        inv_scale = torch.empty(x.shape[0], dtype=torch.float32, device=x.device)
+        _native().quant_fp8_rows(out, inv_scale, x.contiguous())
+        return out, inv_scale
+    return ref.quant_fp8_rows(x)
+
+
 def sample_tokens(logits, temperatures, uniform):
     """Gumbel-max categorical sampling; rows with temperature 0 are greedy."""
     if logits.is_cuda:

@@ -44,6 +44,8 @@ void arks_attn_extend_paged(void* out, const void* q, const void* k_cache,
                             int num_q_heads, int num_kv_heads, int head_dim,
                             int max_blocks, int64_t q_stride,
                             hipStream_t stream);
+void arks_quant_fp8_rows(void* out, void* inv_scale, const void* x, int rows,
+                         int cols, hipStream_t stream);
 void arks_greedy_sample(void* out, const void* logits, int rows, int vocab,
                         hipStream_t stream);
 void arks_gumbel_sample(void* out, const void* logits, const void* temperatures,
@@ -236,6 +238,16 @@ void attention_extend_paged(torch::Tensor out, torch::Tensor q,
                          q.stride(0), current_stream());
 }
 
+void quant_fp8_rows(torch::Tensor out, torch::Tensor inv_scale,
+                    torch::Tensor x) {
+  check_bf16_contig(x, "x");
+  TORCH_CHECK(out.scalar_type() == torch::kFloat8_e4m3fn && out.is_contiguous());
+  TORCH_CHECK(inv_scale.scalar_type() == torch::kFloat32);
+  TORCH_CHECK(x.size(1) % 8 == 0, "cols must be a multiple of 8");
+  arks_quant_fp8_rows(out.data_ptr(), inv_scale.data_ptr(), x.data_ptr(),
+                      x.size(0), x.size(1), current_stream());
+}
+
 void greedy_sample(torch::Tensor out, torch::Tensor logits) {
   check_bf16_contig(logits, "logits");
   TORCH_CHECK(out.scalar_type() == torch::kInt64);
@@ -276,6 +288,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("attention_decode_paged", &attention_decode_paged);
   m.def("attention_prefill_varlen", &attention_prefill_varlen);
   m.def("attention_extend_paged", &attention_extend_paged);
+  m.def("quant_fp8_rows", &quant_fp8_rows);
   m.def("greedy_sample", &greedy_sample);
   m.def("gumbel_sample", &gumbel_sample);
   m.def("mfma_probe", &mfma_probe);

@@ -202,6 +202,14 @@ def attention_extend_paged(
     return out
 
 
+def quant_fp8_rows(x: torch.Tensor) -> tuple[torch.Tensor, torch.Tensor]:
+    """Per-row symmetric quantization to OCP e4m3 (range +-448)."""
+    amax = x.float().abs().amax(dim=1).clamp(min=1e-6)
+    scale = 448.0 / amax
+    q = (x.float() * scale[:, None]).clamp(-448.0, 448.0).to(torch.float8_e4m3fn)
+    return q, (1.0 / scale)
+
+
 def greedy_sample(logits: torch.Tensor) -> torch.Tensor:
     """[num_seqs, vocab] -> [num_seqs] int64 argmax."""
     return logits.float().argmax(dim=-1)

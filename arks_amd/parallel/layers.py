@@ -15,6 +15,46 @@ import torch.nn.functional as F
 from .comm import get_tp_rank, get_tp_world_size, tp_all_gather, tp_all_reduce
 
 
+def quantize_module_fp8(mod: nn.Module) -> None:
+    """Switch a linear layer to W8A8 fp8 (OCP e4m3, per-output-channel weight
+    scales, dynamic per-token activation scales — ops.quant_fp8_rows).
+    The bf16 weight parameter is dropped on CUDA to free HBM; on CPU a
+    quant-dequant copy emulates the GPU numerics for tests."""
+    w = mod.weight.data
+    amax = w.float().abs().amax(dim=1).clamp(min=1e-6)
+    scale = 448.0 / amax
+    w8 = (w.float() * scale[:, None]).clamp(-448.0, 448.0).to(torch.float8_e4m3fn)
+    mod.register_buffer("weight_fp8", w8, persistent=False)
+    mod.register_buffer(
+        "w_inv_scale", (1.0 / scale).float()[None, :], persistent=False
+    )
+    if w.is_cuda:
+        del mod._parameters["weight"]
+        mod.weight = None
+    else:
+        mod.register_buffer(
+            "weight_qdq",
+            (w8.float() * (1.0 / scale)[:, None]).to(w.dtype),
+            persistent=False,
+        )
+    mod.fp8 = True
+
+
+def _fp8_linear(mod: nn.Module, x: torch.Tensor) -> torch.Tensor:
+    from .. import ops
+
+    if x.is_cuda:
+        x8, sa = ops.quant_fp8_rows(x)
+        return torch._scaled_mm(
+            x8, mod.weight_fp8.t(), scale_a=sa[:, None], scale_b=mod.w_inv_scale,
+            bias=mod.bias, out_dtype=x.dtype,
+        )
+    # CPU emulation: quant-dequant both operands, accumulate in f32
+    x8, sa = ops.quant_fp8_rows(x)
+    xd = (x8.float() * sa[:, None]).to(x.dtype)
+    return F.linear(xd, mod.weight_qdq, mod.bias)
+
+
 class ColumnParallelLinear(nn.Module):
     """Y = X W^T with W row-sharded over TP ranks (output features split).
     Output stays sharded (gather_output=False semantics)."""
@@ -41,7 +81,11 @@ class ColumnParallelLinear(nn.Module):
         r = get_tp_rank()
         return full[r * self.out_per_rank : (r + 1) * self.out_per_rank]
 
+    fp8 = False
+
     def forward(self, x: torch.Tensor) -> torch.Tensor:
+        if self.fp8:
+            return _fp8_linear(self, x)
         return F.linear(x, self.weight, self.bias)
 
 
@@ -121,8 +165,18 @@ class RowParallelLinear(nn.Module):
         r = get_tp_rank()
         return full[:, r * self.in_per_rank : (r + 1) * self.in_per_rank]
 
+    fp8 = False
+
     def forward(self, x: torch.Tensor) -> torch.Tensor:
-        y = F.linear(x, self.weight)
+        if self.fp8:
+            bias = self.bias
+            self.bias = None  # bias must be applied post-reduce, once
+            try:
+                y = _fp8_linear(self, x)
+            finally:
+                self.bias = bias
+        else:
+            y = F.linear(x, self.weight)
         y = tp_all_reduce(y)
         if self.bias is not None:
             y = y + self.bias
@@ -154,7 +208,13 @@ class ParallelLMHead(nn.Module):
         )
         return torch.cat([full[lo:], pad], dim=0)
 
+    fp8 = False
+
     def forward(self, x: torch.Tensor) -> torch.Tensor:
-        logits = F.linear(x, self.weight)
+        if self.fp8:
+            self.bias = None
+            logits = _fp8_linear(self, x)
+        else:
+            logits = F.linear(x, self.weight)
         logits = tp_all_gather(logits, dim=-1)
         return logits[..., : self.vocab]

@@ -34,6 +34,7 @@ def parse_args(argv=None):
     p.add_argument("--enforce-eager", action="store_true")
     p.add_argument("--disaggregation-mode", choices=["prefill", "decode"],
                    default=None)
+    p.add_argument("--quantization", choices=["fp8"], default=None)
     # multi-node group flags (LWS leader/worker topology): workers join the
     # leader's torch.distributed rendezvous
     p.add_argument("--leader-address", default=None)
@@ -60,6 +61,7 @@ def build_engine_config(args):
         gpu_memory_utilization=args.gpu_memory_utilization,
         kv_cache_blocks=args.kv_cache_blocks,
         enforce_eager=args.enforce_eager,
+        quantization=args.quantization,
     )
 
 
