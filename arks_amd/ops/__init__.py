@@ -176,6 +176,61 @@ def attention_decode_paged(
     return ref.attention_decode_paged(q, k_cache, v_cache, block_tables, seq_lens, scale)
 
 
+_SKINNY_WS: dict = {}
+
+
+def skinny_gemm(x, weight, bias=None):
+    """C = x @ weight^T for decode-sized M (<=64) via the streaming MFMA
+    kernel (profiles/r01_p2: hipBLASLt leaves 2-4x on the table at these
+    shapes). Deterministic split-K for small N."""
+    M, K = x.shape
+    N = weight.shape[0]
+    out = torch.empty((M, N), dtype=x.dtype, device=x.device)
+    ntiles = N // 64
+    if ntiles >= 384:
+        nsplits = 1
+    else:
+        nsplits = min(-(-512 // ntiles), -(-K // 256))
+    k_ceil = -(-K // nsplits)
+    k_per_split = -(-k_ceil // 32) * 32
+    nsplits = -(-K // k_per_split)
+    part = _skinny_ws(nsplits, N, M, x.device) if nsplits > 1 else x.new_empty(
+        0, dtype=torch.float32
+    )
+    _native().skinny_gemm(out, part, x, weight, bias, k_per_split, nsplits)
+    return out
+
+
+def _skinny_ws(nsplits: int, N: int, M: int, device):
+    mrows = ((M + 15) // 16) * 16
+    key = (device.index,)
+    need = nsplits * N * mrows
+    ws = _SKINNY_WS.get(key)
+    if ws is None or ws.numel() < need:
+        ws = torch.empty(need, dtype=torch.float32, device=device)
+        _SKINNY_WS[key] = ws
+    return ws
+
+
+def linear_bf16(x, weight, bias=None):
+    """Linear dispatch: the skinny streaming kernel for decode-shaped
+    (M<=64) GEMMs on GPU, hipBLASLt (F.linear) otherwise."""
+    if (
+        x.is_cuda
+        and x.dim() == 2
+        and 1 <= x.shape[0] <= 64
+        and x.dtype == torch.bfloat16
+        and weight.shape[0] % 64 == 0
+        and weight.shape[1] % 32 == 0
+        and weight.is_contiguous()
+        and native_available()
+    ):
+        return skinny_gemm(x, weight, bias)
+    import torch.nn.functional as F
+
+    return F.linear(x, weight, bias)
+
+
 def quant_fp8_rows(x):
     """Dynamic per-token fp8 e4m3 quantization: [M, K] bf16 ->
     ([M, K] float8_e4m3fn, [M] fp32 dequant scales) for W8A8 GEMMs

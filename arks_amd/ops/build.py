@@ -26,6 +26,7 @@ SOURCES = [
     os.path.join(CSRC, "attn_prefill.hip"),
     os.path.join(CSRC, "attn_extend.hip"),
     os.path.join(CSRC, "quant_fp8.hip"),
+    os.path.join(CSRC, "skinny_gemm.hip"),
     os.path.join(CSRC, "sampling.hip"),
 ]
 

@@ -46,6 +46,10 @@ void arks_attn_extend_paged(void* out, const void* q, const void* k_cache,
                             hipStream_t stream);
 void arks_quant_fp8_rows(void* out, void* inv_scale, const void* x, int rows,
                          int cols, hipStream_t stream);
+void arks_skinny_gemm(void* part, void* out, const void* a, const void* w,
+                      const void* bias, int m_rows, int n_total, int k_total,
+                      int k_per_split, int nsplits, int64_t a_stride,
+                      hipStream_t stream);
 void arks_greedy_sample(void* out, const void* logits, int rows, int vocab,
                         hipStream_t stream);
 void arks_gumbel_sample(void* out, const void* logits, const void* temperatures,
@@ -238,6 +242,34 @@ void attention_extend_paged(torch::Tensor out, torch::Tensor q,
                          q.stride(0), current_stream());
 }
 
+void skinny_gemm(torch::Tensor out, torch::Tensor part, torch::Tensor a,
+                 torch::Tensor w, c10::optional<torch::Tensor> bias,
+                 int64_t k_per_split, int64_t nsplits) {
+  check_bf16_contig(out, "out");
+  check_bf16_rowstrided(a, "a");
+  check_bf16_contig(w, "w");
+  const int m = a.size(0), k = a.size(1), n = w.size(0);
+  TORCH_CHECK(m >= 1 && m <= 64, "skinny_gemm needs 1 <= M <= 64");
+  TORCH_CHECK(n % 64 == 0 && k % 32 == 0, "N%64==0 and K%32==0 required");
+  TORCH_CHECK(w.size(1) == k && out.size(0) == m && out.size(1) == n);
+  const void* bias_ptr = nullptr;
+  if (bias.has_value()) {
+    check_bf16_contig(*bias, "bias");
+    TORCH_CHECK(bias->numel() == n);
+    bias_ptr = bias->data_ptr();
+  }
+  const int mrows = ((m + 15) / 16) * 16;
+  if (nsplits > 1) {
+    TORCH_CHECK(part.scalar_type() == torch::kFloat32 &&
+                part.numel() >= (int64_t)nsplits * n * mrows,
+                "skinny_gemm workspace too small");
+  }
+  arks_skinny_gemm(nsplits > 1 ? part.data_ptr() : nullptr, out.data_ptr(),
+                   a.data_ptr(), w.data_ptr(), bias_ptr, m, n, k,
+                   (int)k_per_split, (int)nsplits, a.stride(0),
+                   current_stream());
+}
+
 void quant_fp8_rows(torch::Tensor out, torch::Tensor inv_scale,
                     torch::Tensor x) {
   check_bf16_contig(x, "x");
@@ -289,6 +321,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("attention_prefill_varlen", &attention_prefill_varlen);
   m.def("attention_extend_paged", &attention_extend_paged);
   m.def("quant_fp8_rows", &quant_fp8_rows);
+  m.def("skinny_gemm", &skinny_gemm);
   m.def("greedy_sample", &greedy_sample);
   m.def("gumbel_sample", &gumbel_sample);
   m.def("mfma_probe", &mfma_probe);

@@ -86,7 +86,9 @@ class ColumnParallelLinear(nn.Module):
     def forward(self, x: torch.Tensor) -> torch.Tensor:
         if self.fp8:
             return _fp8_linear(self, x)
-        return F.linear(x, self.weight, self.bias)
+        from .. import ops
+
+        return ops.linear_bf16(x, self.weight, self.bias)
 
 
 class QKVParallelLinear(ColumnParallelLinear):
@@ -176,7 +178,9 @@ class RowParallelLinear(nn.Module):
             finally:
                 self.bias = bias
         else:
-            y = F.linear(x, self.weight)
+            from .. import ops
+
+            y = ops.linear_bf16(x, self.weight)
         y = tp_all_reduce(y)
         if self.bias is not None:
             y = y + self.bias
@@ -215,6 +219,8 @@ class ParallelLMHead(nn.Module):
             self.bias = None
             logits = _fp8_linear(self, x)
         else:
-            logits = F.linear(x, self.weight)
+            from .. import ops
+
+            logits = ops.linear_bf16(x, self.weight)
         logits = tp_all_gather(logits, dim=-1)
         return logits[..., : self.vocab]

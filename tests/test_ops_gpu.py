@@ -343,3 +343,37 @@ def test_engine_fp8_gpu():
     out1 = e1.generate(prompts, sp)
     out2 = e1.generate(prompts, sp)  # prefix-cache hit path + fp8
     assert out1 == out2 and all(len(o) == 6 for o in out1)
+
+
+@pytest.mark.parametrize("m,k,n", [
+    (1, 3584, 4608), (5, 96, 64), (17, 3584, 3584), (64, 18944, 3584),
+    (64, 3584, 37888), (33, 3584, 152064), (64, 128, 64),
+])
+def test_skinny_gemm(m, k, n):
+    """Streaming decode GEMM vs hipBLASLt, both split and direct paths."""
+    assert_native()
+    torch.manual_seed(19)
+    a = torch.randn(m, k, dtype=torch.bfloat16, device=DEV) * 0.5
+    w = torch.randn(n, k, dtype=torch.bfloat16, device=DEV) * 0.05
+    bias = torch.randn(n, dtype=torch.bfloat16, device=DEV)
+    ref_out = torch.nn.functional.linear(a, w, bias).float()
+    out = ops.skinny_gemm(a, w, bias).float()
+    torch.testing.assert_close(out, ref_out, atol=5e-2, rtol=5e-2)
+    # no-bias path
+    out2 = ops.skinny_gemm(a, w).float()
+    torch.testing.assert_close(
+        out2, torch.nn.functional.linear(a, w).float(), atol=5e-2, rtol=5e-2
+    )
+
+
+def test_skinny_gemm_strided_input():
+    """Row-strided activations (views into a fused buffer)."""
+    assert_native()
+    torch.manual_seed(23)
+    buf = torch.randn(16, 512, dtype=torch.bfloat16, device=DEV)
+    a = buf[:, 128:128 + 256]  # stride(0)=512
+    w = torch.randn(128, 256, dtype=torch.bfloat16, device=DEV) * 0.1
+    out = ops.skinny_gemm(a, w).float()
+    torch.testing.assert_close(
+        out, torch.nn.functional.linear(a, w).float(), atol=5e-2, rtol=5e-2
+    )
