@@ -1,0 +1,61 @@
+"""Operator entrypoint (the reference's /manager binary, cmd/main.go).
+
+    python -m arks_amd.controlplane [--kube-api https://...] [--resync 5]
+
+Runs the four reconcilers against the cluster through KubeStore, with a
+poll-resync loop feeding the work queue and /healthz + /readyz probes on
+--health-port (reference cmd/main.go:157-196).
+"""
+
+from __future__ import annotations
+
+import argparse
+import threading
+from http.server import BaseHTTPRequestHandler, ThreadingHTTPServer
+
+
+def main(argv=None):
+    ap = argparse.ArgumentParser(prog="arks_amd.controlplane")
+    ap.add_argument("--kube-api", default=None,
+                    help="API server base URL (default: in-cluster)")
+    ap.add_argument("--resync", type=float, default=5.0)
+    ap.add_argument("--health-port", type=int, default=8081)
+    args = ap.parse_args(argv)
+
+    from .kubestore import KubeStore
+    from .operator import Operator
+
+    store = KubeStore(api_base=args.kube_api)
+    op = Operator(store)
+
+    class Probe(BaseHTTPRequestHandler):
+        def do_GET(self):
+            if self.path in ("/healthz", "/readyz"):
+                self.send_response(200)
+                self.end_headers()
+                self.wfile.write(b"ok")
+            else:
+                self.send_response(404)
+                self.end_headers()
+
+        def log_message(self, *a):
+            pass
+
+    health = ThreadingHTTPServer(("0.0.0.0", args.health_port), Probe)
+    threading.Thread(target=health.serve_forever, daemon=True).start()
+    threading.Thread(target=store.run_resync, args=(args.resync,),
+                     daemon=True).start()
+    print(f"arks operator running (resync {args.resync}s, "
+          f"health :{args.health_port})", flush=True)
+    try:
+        op.run()
+    except KeyboardInterrupt:
+        pass
+    finally:
+        store.stop()
+        op.stop()
+        health.shutdown()
+
+
+if __name__ == "__main__":
+    main()

@@ -1,0 +1,261 @@
+"""Real-cluster Store backend: the same CRUD surface as
+arks_amd.controlplane.store.Store, implemented against the Kubernetes API
+server over REST (httpx, serviceaccount auth — no client library needed).
+
+This is what makes the operator deployable (reference cmd/main.go runs a
+controller-runtime manager; here Operator + KubeStore fill that role):
+  * arks.ai kinds round-trip through the pydantic types (arks_amd.crd.types)
+    so reconcilers see the same objects as with the in-memory store;
+  * workload kinds (Pod/Service/PVC/Deployment/LWS/RBGS/HTTPRoute) stay raw
+    dicts;
+  * status is written through the /status subresource when present;
+  * watch is a poll-based resync loop (list + diff) that feeds the
+    Operator's work queue — simple, restart-safe and sufficient at
+    control-plane rates.
+"""
+
+from __future__ import annotations
+
+import os
+import threading
+import time
+from typing import Any, Callable
+
+import httpx
+
+from ..crd import types as T
+from .store import Conflict, NotFound
+
+SA_DIR = "/var/run/secrets/kubernetes.io/serviceaccount"
+
+# kind -> (api prefix, plural, namespaced)
+API_MAP: dict[str, tuple[str, str]] = {
+    "Pod": ("/api/v1", "pods"),
+    "Service": ("/api/v1", "services"),
+    "PersistentVolumeClaim": ("/api/v1", "persistentvolumeclaims"),
+    "Secret": ("/api/v1", "secrets"),
+    "Deployment": ("/apis/apps/v1", "deployments"),
+    "ServiceAccount": ("/api/v1", "serviceaccounts"),
+    "Role": ("/apis/rbac.authorization.k8s.io/v1", "roles"),
+    "RoleBinding": ("/apis/rbac.authorization.k8s.io/v1", "rolebindings"),
+    "LeaderWorkerSet": ("/apis/leaderworkerset.x-k8s.io/v1", "leaderworkersets"),
+    "RoleBasedGroupSet": ("/apis/workloads.x-k8s.io/v1alpha1", "rolebasedgroupsets"),
+    "HTTPRoute": ("/apis/gateway.networking.k8s.io/v1", "httproutes"),
+    "ArksModel": ("/apis/arks.ai/v1", "arksmodels"),
+    "ArksApplication": ("/apis/arks.ai/v1", "arksapplications"),
+    "ArksDisaggregatedApplication": ("/apis/arks.ai/v1",
+                                     "arksdisaggregatedapplications"),
+    "ArksEndpoint": ("/apis/arks.ai/v1", "arksendpoints"),
+    "ArksToken": ("/apis/arks.ai/v1", "arkstokens"),
+    "ArksQuota": ("/apis/arks.ai/v1", "arksquotas"),
+}
+
+ARKS_TYPES: dict[str, type] = {
+    "ArksModel": T.ArksModel,
+    "ArksApplication": T.ArksApplication,
+    "ArksDisaggregatedApplication": T.ArksDisaggregatedApplication,
+    "ArksEndpoint": T.ArksEndpoint,
+    "ArksToken": T.ArksToken,
+    "ArksQuota": T.ArksQuota,
+}
+
+WATCHED_KINDS = [
+    "ArksModel", "ArksApplication", "ArksDisaggregatedApplication",
+    "ArksEndpoint", "Pod", "LeaderWorkerSet", "RoleBasedGroupSet",
+    "Deployment",
+]
+
+
+def _to_dict(obj: Any) -> dict:
+    if isinstance(obj, dict):
+        return obj
+    d = obj.model_dump(by_alias=True, exclude_none=True)
+    kind = type(obj).__name__
+    d["apiVersion"] = "arks.ai/v1"
+    d["kind"] = kind
+    meta = d.get("metadata", {})
+    # internal metadata back to K8s shape
+    rv = meta.pop("resourceVersion", 0)
+    if rv:
+        meta["resourceVersion"] = str(rv)
+    meta.pop("deletionTimestamp", None)
+    d["metadata"] = meta
+    return d
+
+
+def _from_dict(kind: str, d: dict) -> Any:
+    cls = ARKS_TYPES.get(kind)
+    if cls is None:
+        return d
+    meta = dict(d.get("metadata", {}))
+    rv = meta.get("resourceVersion")
+    if isinstance(rv, str) and rv.isdigit():
+        meta["resourceVersion"] = int(rv)
+    elif not isinstance(rv, int):
+        meta.pop("resourceVersion", None)
+    dt = meta.get("deletionTimestamp")
+    if isinstance(dt, str):
+        meta["deletionTimestamp"] = time.time()  # presence is what matters
+    obj = cls.model_validate({**d, "metadata": meta})
+    return obj
+
+
+class KubeStore:
+    """Store-compatible CRUD against a real API server."""
+
+    def __init__(self, api_base: str | None = None, token: str | None = None,
+                 verify: Any = None, transport=None, timeout: float = 15.0):
+        host = os.environ.get("KUBERNETES_SERVICE_HOST", "kubernetes.default.svc")
+        port = os.environ.get("KUBERNETES_SERVICE_PORT", "443")
+        base = api_base or f"https://{host}:{port}"
+        headers = {"Content-Type": "application/json"}
+        token_path = os.path.join(SA_DIR, "token")
+        if token is None and os.path.exists(token_path):
+            with open(token_path) as f:
+                token = f.read().strip()
+        if token:
+            headers["Authorization"] = f"Bearer {token}"
+        if verify is None:
+            ca = os.path.join(SA_DIR, "ca.crt")
+            verify = ca if os.path.exists(ca) else False
+        self._client = httpx.Client(base_url=base, headers=headers,
+                                    verify=verify, transport=transport,
+                                    timeout=timeout)
+        self._watchers: list[Callable[[str, Any], None]] = []
+        self._stop = threading.Event()
+        self._known: dict[tuple, str] = {}  # (kind, ns, name) -> resourceVersion
+
+    # -- paths --
+    @staticmethod
+    def _path(kind: str, namespace: str, name: str | None = None) -> str:
+        prefix, plural = API_MAP[kind]
+        p = f"{prefix}/namespaces/{namespace}/{plural}"
+        return f"{p}/{name}" if name else p
+
+    # -- events --
+    def subscribe(self, fn: Callable[[str, Any], None]) -> None:
+        self._watchers.append(fn)
+
+    def _notify(self, event: str, obj: Any) -> None:
+        for fn in list(self._watchers):
+            fn(event, obj)
+
+    # -- CRUD (Store interface) --
+    def create(self, obj: Any) -> Any:
+        d = _to_dict(obj)
+        kind = d["kind"]
+        ns = d["metadata"].get("namespace", "default")
+        r = self._client.post(self._path(kind, ns), json=d)
+        if r.status_code == 409:
+            raise Conflict(r.text)
+        r.raise_for_status()
+        return _from_dict(kind, r.json())
+
+    def get(self, kind: str, namespace: str, name: str) -> Any:
+        r = self._client.get(self._path(kind, namespace, name))
+        if r.status_code == 404:
+            raise NotFound(f"{kind}/{namespace}/{name}")
+        r.raise_for_status()
+        return _from_dict(kind, r.json())
+
+    def get_opt(self, kind: str, namespace: str, name: str) -> Any | None:
+        try:
+            return self.get(kind, namespace, name)
+        except NotFound:
+            return None
+
+    def update(self, obj: Any) -> Any:
+        d = _to_dict(obj)
+        kind = d["kind"]
+        meta = d["metadata"]
+        ns = meta.get("namespace", "default")
+        name = meta["name"]
+        path = self._path(kind, ns, name)
+        # fetch live resourceVersion (reconcilers are level-triggered and
+        # idempotent; last-write-wins is the reference's RetryOnConflict
+        # pattern collapsed into one step)
+        live = self._client.get(path)
+        if live.status_code == 404:
+            raise NotFound(f"{kind}/{ns}/{name}")
+        live.raise_for_status()
+        live_obj = live.json()
+        meta["resourceVersion"] = live_obj["metadata"]["resourceVersion"]
+        status = d.pop("status", None)
+        r = self._client.put(path, json=d)
+        if r.status_code == 409:
+            raise Conflict(r.text)
+        r.raise_for_status()
+        out = r.json()
+        if status is not None:
+            sub = dict(out)
+            sub["status"] = status
+            rs = self._client.put(f"{path}/status", json=sub)
+            if rs.status_code not in (404, 405):  # kinds without status sub
+                rs.raise_for_status()
+                out = rs.json()
+        return _from_dict(kind, out)
+
+    def apply(self, obj: Any) -> Any:
+        try:
+            return self.update(obj)
+        except NotFound:
+            return self.create(obj)
+
+    def delete(self, kind: str, namespace: str, name: str) -> None:
+        r = self._client.delete(self._path(kind, namespace, name))
+        if r.status_code not in (200, 202, 404):
+            r.raise_for_status()
+
+    def finalize(self, obj: Any) -> None:
+        """Called by reconcilers after removing the last finalizer: the
+        update() above already persisted the empty finalizer list, so the
+        API server completes the deletion itself."""
+
+    def mark_deleted(self, kind: str, namespace: str, name: str) -> Any | None:
+        self.delete(kind, namespace, name)
+        return self.get_opt(kind, namespace, name)
+
+    def list(self, kind: str, namespace: str | None = None) -> list:
+        if namespace is not None:
+            r = self._client.get(self._path(kind, namespace))
+            r.raise_for_status()
+            return [_from_dict(kind, i) for i in r.json().get("items", [])]
+        prefix, plural = API_MAP[kind]
+        r = self._client.get(f"{prefix}/{plural}")
+        r.raise_for_status()
+        return [_from_dict(kind, i) for i in r.json().get("items", [])]
+
+    # -- resync loop (drives Operator's queue like informers would) --
+    def resync_once(self) -> None:
+        seen: set[tuple] = set()
+        for kind in WATCHED_KINDS:
+            try:
+                items = self.list(kind)
+            except Exception:
+                continue
+            for obj in items:
+                m = obj["metadata"] if isinstance(obj, dict) else obj.metadata
+                ns = (m.get("namespace", "default") if isinstance(m, dict)
+                      else m.namespace)
+                name = m.get("name", "") if isinstance(m, dict) else m.name
+                rv = str(m.get("resourceVersion", "") if isinstance(m, dict)
+                         else m.resource_version)
+                key = (kind, ns, name)
+                seen.add(key)
+                if self._known.get(key) != rv:
+                    event = "MODIFIED" if key in self._known else "ADDED"
+                    self._known[key] = rv
+                    self._notify(event, obj)
+        for key in [k for k in self._known if k not in seen]:
+            kind, ns, name = key
+            del self._known[key]
+            self._notify("DELETED", {"kind": kind,
+                                     "metadata": {"namespace": ns, "name": name}})
+
+    def run_resync(self, interval_s: float = 5.0) -> None:
+        while not self._stop.is_set():
+            self.resync_once()
+            self._stop.wait(interval_s)
+
+    def stop(self) -> None:
+        self._stop.set()

@@ -1,0 +1,47 @@
+"""Gateway entrypoint (the reference's gateway-plugins binary,
+cmd/gateway/main.go — here the gateway IS the HTTP hop instead of an Envoy
+ext_proc side-call).
+
+    python -m arks_amd.gateway --port 8080 [--kube-api https://...]
+
+Token/Quota/Endpoint CRs are read live from the cluster (KubeStore resync);
+rate-limit and quota state is in-process (fixed-window counters with the
+reference's redis key semantics — arks_amd/gateway/limiter.py).
+"""
+
+from __future__ import annotations
+
+import argparse
+import threading
+
+
+def main(argv=None):
+    ap = argparse.ArgumentParser(prog="arks_amd.gateway")
+    ap.add_argument("--host", default="0.0.0.0")
+    ap.add_argument("--port", type=int, default=8080)
+    ap.add_argument("--kube-api", default=None)
+    ap.add_argument("--resync", type=float, default=5.0)
+    ap.add_argument("--standalone", action="store_true",
+                    help="in-memory store (no cluster; for local testing)")
+    args = ap.parse_args(argv)
+
+    import uvicorn
+
+    from .app import create_gateway_app
+
+    if args.standalone:
+        from ..controlplane.store import Store
+
+        store = Store()
+    else:
+        from ..controlplane.kubestore import KubeStore
+
+        store = KubeStore(api_base=args.kube_api)
+        threading.Thread(target=store.run_resync, args=(args.resync,),
+                         daemon=True).start()
+    app = create_gateway_app(store)
+    uvicorn.run(app, host=args.host, port=args.port, log_level="info")
+
+
+if __name__ == "__main__":
+    main()
