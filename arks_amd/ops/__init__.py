@@ -10,6 +10,8 @@ are unit-testable without a GPU.
 
 from __future__ import annotations
 
+import os
+
 import torch
 
 from . import ref
@@ -212,11 +214,16 @@ def _skinny_ws(nsplits: int, N: int, M: int, device):
     return ws
 
 
+_USE_SKINNY = os.environ.get("ARKS_SKINNY_GEMM", "0") == "1"
+
+
 def linear_bf16(x, weight, bias=None):
     """Linear dispatch: the skinny streaming kernel for decode-shaped
-    (M<=64) GEMMs on GPU, hipBLASLt (F.linear) otherwise."""
+    (M<=64) GEMMs on GPU, hipBLASLt (F.linear) otherwise. Off by default
+    until it beats the tuned hipBLASLt table in-engine (ARKS_SKINNY_GEMM=1)."""
     if (
-        x.is_cuda
+        _USE_SKINNY
+        and x.is_cuda
         and x.dim() == 2
         and 1 <= x.shape[0] <= 64
         and x.dtype == torch.bfloat16

@@ -60,6 +60,21 @@ __global__ __launch_bounds__(256) void skinny_gemm_kernel(
   for (int t = 0; t < MT; ++t) acc[t] = {0.f, 0.f, 0.f, 0.f};
 
   const bf16* wrow = w + (int64_t)(n0 + wave * 16 + lq) * k_total;
+  constexpr int KSTEPS = SK_KC / 32;
+
+  // W fragments are double-buffered in registers: chunk i+1's 8 global
+  // loads issue while chunk i's MFMAs run, so the weight stream never
+  // stalls the MFMA pipe (the whole kernel is W-bandwidth-bound).
+  ushort8 wreg[KSTEPS];
+  {
+    const int kw0 = min(SK_KC, ke - kb);
+#pragma unroll
+    for (int s = 0; s < KSTEPS; ++s)
+      wreg[s] = (s * 32 < kw0)
+                    ? *reinterpret_cast<const ushort8*>(wrow + kb + s * 32 +
+                                                        8 * la)
+                    : ushort8{};
+  }
 
   for (int kc = kb; kc < ke; kc += SK_KC) {
     const int kw = min(SK_KC, ke - kc);  // multiple of 32
@@ -75,18 +90,31 @@ __global__ __launch_bounds__(256) void skinny_gemm_kernel(
     }
     __syncthreads();
 
-    for (int kk = 0; kk < kw; kk += 32) {
-      // W fragment: A-operand row = W row (n0 + wave*16 + lq), k = 8*la..+7.
-      ushort8 wf =
-          *reinterpret_cast<const ushort8*>(wrow + kc + kk + 8 * la);
-      bf16x8 wfrag = *reinterpret_cast<bf16x8*>(&wf);
+    ushort8 wnext[KSTEPS];
+    const int kn = kc + SK_KC;
+    if (kn < ke) {
+      const int kwn = min(SK_KC, ke - kn);
+#pragma unroll
+      for (int s = 0; s < KSTEPS; ++s)
+        wnext[s] = (s * 32 < kwn)
+                       ? *reinterpret_cast<const ushort8*>(wrow + kn + s * 32 +
+                                                           8 * la)
+                       : ushort8{};
+    }
+
+#pragma unroll
+    for (int s = 0; s < KSTEPS; ++s) {
+      if (s * 32 >= kw) break;
+      bf16x8 wfrag = *reinterpret_cast<bf16x8*>(&wreg[s]);
 #pragma unroll
       for (int t = 0; t < MT; ++t) {
-        ushort8 af =
-            *reinterpret_cast<const ushort8*>(&a_lds[t * 16 + lq][kk + 8 * la]);
+        ushort8 af = *reinterpret_cast<const ushort8*>(
+            &a_lds[t * 16 + lq][s * 32 + 8 * la]);
         acc[t] = sk_mfma(wfrag, *reinterpret_cast<bf16x8*>(&af), acc[t]);
       }
     }
+#pragma unroll
+    for (int s = 0; s < KSTEPS; ++s) wreg[s] = wnext[s];
     __syncthreads();  // a_lds reuse
   }
 

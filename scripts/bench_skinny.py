@@ -2,9 +2,13 @@
 """Per-shape microbench: skinny_gemm vs hipBLASLt (tuned table loaded) on the
 decode GEMM shapes. Prints achieved weight-streaming TB/s."""
 
+import os
+import sys
 import time
 
 import torch
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 
 import arks_amd.ops as ops
 
