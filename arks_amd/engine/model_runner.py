@@ -338,9 +338,43 @@ class ModelRunner:
         if all(t == 0.0 for t in temps):
             ids = ops.greedy_sample(logits.contiguous())
         else:
+            logits = self._apply_top_p_top_k(logits, seqs)
             t = torch.tensor(temps, dtype=torch.float32, device=logits.device)
             u = torch.rand(
                 logits.shape, dtype=torch.float32, device=logits.device
             )
             ids = ops.sample_tokens(logits.contiguous(), t, u)
         return ids.tolist()
+
+    @staticmethod
+    def _apply_top_p_top_k(logits: torch.Tensor, seqs: list[Sequence]) -> torch.Tensor:
+        """Mask logits outside each row's top-p nucleus / top-k set (rows
+        with temperature 0 or no constraint pass through)."""
+        rows = [
+            i for i, s in enumerate(seqs)
+            if s.sampling.temperature > 0.0
+            and (s.sampling.top_p < 1.0 or s.sampling.top_k > 0)
+        ]
+        if not rows:
+            return logits
+        logits = logits.clone()
+        idx = torch.tensor(rows, dtype=torch.long, device=logits.device)
+        sub = logits[idx].float()
+        sorted_logits, sorted_idx = sub.sort(dim=-1, descending=True)
+        keep = torch.ones_like(sorted_logits, dtype=torch.bool)
+        probs = torch.softmax(sorted_logits, dim=-1)
+        cum = probs.cumsum(dim=-1)
+        for j, i in enumerate(rows):
+            sp = seqs[i].sampling
+            if sp.top_p < 1.0:
+                # keep tokens while cumulative prob (exclusive) < top_p
+                keep[j] &= (cum[j] - probs[j]) < sp.top_p
+            if sp.top_k > 0:
+                keep[j, sp.top_k:] = False
+            keep[j, 0] = True  # always keep the best token
+        sub = sub.masked_fill(~keep, float("-inf"))
+        # scatter back to original column order
+        restored = torch.full_like(sub, float("-inf"))
+        restored.scatter_(1, sorted_idx, sub)
+        logits[idx] = restored.to(logits.dtype)
+        return logits

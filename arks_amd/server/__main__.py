@@ -69,12 +69,24 @@ def main(argv=None):
     args = parse_args(argv)
     tp = args.tensor_parallel_size
     if tp > 1 and "RANK" not in os.environ:
+        # LWS leader/worker topology: workers join the leader's rendezvous
+        # (reference arksapplication_controller.go:982-1014 wires
+        # LWS_LEADER_ADDRESS / LWS_GROUP_SIZE / LWS_WORKER_INDEX; our
+        # command builders pass them as --leader-address / --node-rank).
+        nnodes = int(os.environ.get("LWS_GROUP_SIZE", "1"))
+        master = args.leader_address or "127.0.0.1"
+        if tp % nnodes != 0:
+            raise SystemExit(
+                f"tensor-parallel-size {tp} must divide across "
+                f"LWS_GROUP_SIZE {nnodes} nodes"
+            )
         os.execv(
             sys.executable,
             [
                 sys.executable, "-m", "torch.distributed.run",
-                "--nnodes=1", f"--nproc-per-node={tp}",
-                "--master-addr=127.0.0.1", "--master-port=29517",
+                f"--nnodes={nnodes}", f"--nproc-per-node={tp // nnodes}",
+                f"--node-rank={args.node_rank}",
+                f"--master-addr={master}", "--master-port=29517",
                 "-m", "arks_amd.server",
             ] + sys.argv[1:],
         )

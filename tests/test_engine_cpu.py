@@ -98,3 +98,24 @@ def test_random_sampling_runs():
     )[0]
     assert len(out) == 5
     assert all(0 <= t < e.model_cfg.vocab_size for t in out)
+
+
+def test_top_k_sampling_restricts_support():
+    """top_k=1 must equal greedy; top_p tiny must track the argmax too."""
+    from arks_amd.config import EngineConfig
+    from arks_amd.engine import LLMEngine, SamplingParams
+
+    def run(sp):
+        eng = LLMEngine(EngineConfig(
+            preset="tiny", device="cpu", kv_cache_blocks=128,
+            max_model_len=256, seed=9,
+        ))
+        return eng.generate([[4, 8, 15, 16, 23, 42]], sp)[0]
+
+    greedy = run(SamplingParams(max_tokens=8, temperature=0.0, ignore_eos=True))
+    topk1 = run(SamplingParams(max_tokens=8, temperature=1.5, top_k=1,
+                               ignore_eos=True))
+    assert topk1 == greedy
+    topp = run(SamplingParams(max_tokens=8, temperature=1.5, top_p=1e-9,
+                              ignore_eos=True))
+    assert topp == greedy
