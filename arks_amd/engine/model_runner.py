@@ -334,6 +334,8 @@ class ModelRunner:
         return self.sample(logits, sb.seqs)
 
     def sample(self, logits: torch.Tensor, seqs: list[Sequence]) -> list[int]:
+        if any(s.sampling.has_penalties for s in seqs):
+            logits = self._apply_penalties(logits, seqs)
         temps = [s.sampling.temperature for s in seqs]
         if all(t == 0.0 for t in temps):
             ids = ops.greedy_sample(logits.contiguous())
@@ -345,6 +347,35 @@ class ModelRunner:
             )
             ids = ops.sample_tokens(logits.contiguous(), t, u)
         return ids.tolist()
+
+    @staticmethod
+    def _apply_penalties(logits: torch.Tensor, seqs: list[Sequence]) -> torch.Tensor:
+        """OpenAI presence/frequency penalties (on generated tokens) and
+        vLLM-style repetition penalty (prompt + generated)."""
+        logits = logits.clone()
+        dev = logits.device
+        for i, seq in enumerate(seqs):
+            sp = seq.sampling
+            if not sp.has_penalties:
+                continue
+            row = logits[i].float()
+            if sp.presence_penalty or sp.frequency_penalty:
+                if seq.output_token_ids:
+                    ids, counts = torch.unique(
+                        torch.tensor(seq.output_token_ids, device=dev),
+                        return_counts=True,
+                    )
+                    row[ids] -= sp.presence_penalty
+                    row[ids] -= sp.frequency_penalty * counts.float()
+            if sp.repetition_penalty != 1.0:
+                seen = torch.unique(torch.tensor(seq.all_token_ids, device=dev))
+                vals = row[seen]
+                row[seen] = torch.where(
+                    vals > 0, vals / sp.repetition_penalty,
+                    vals * sp.repetition_penalty,
+                )
+            logits[i] = row.to(logits.dtype)
+        return logits
 
     @staticmethod
     def _apply_top_p_top_k(logits: torch.Tensor, seqs: list[Sequence]) -> torch.Tensor:

@@ -13,8 +13,16 @@ class SamplingParams:
     temperature: float = 0.0  # 0 => greedy
     top_p: float = 1.0  # applied via logit filtering when < 1.0
     top_k: int = 0  # 0 => disabled
+    presence_penalty: float = 0.0  # flat penalty on seen tokens
+    frequency_penalty: float = 0.0  # per-occurrence penalty
+    repetition_penalty: float = 1.0  # >1 divides positive seen-logits
     stop_token_ids: tuple[int, ...] = ()
     ignore_eos: bool = False
+
+    @property
+    def has_penalties(self) -> bool:
+        return (self.presence_penalty != 0.0 or self.frequency_penalty != 0.0
+                or self.repetition_penalty != 1.0)
 
 
 class SeqStatus(enum.Enum):

@@ -214,13 +214,17 @@ def _skinny_ws(nsplits: int, N: int, M: int, device):
     return ws
 
 
-_USE_SKINNY = os.environ.get("ARKS_SKINNY_GEMM", "0") == "1"
+_USE_SKINNY = os.environ.get("ARKS_SKINNY_GEMM", "1") == "1"
+# Weight-elements threshold: the split-K streaming kernel beats hipBLASLt's
+# ~19 us latency floor on the small decode shapes (qkv/o: 1.0-1.6x,
+# bench_skinny v3); the big streaming shapes (gate_up/down/lm_head) stay on
+# the tuned hipBLASLt algos which reach 5.4-6.4 TB/s there.
+_SKINNY_MAX_ELEMS = 34_000_000
 
 
 def linear_bf16(x, weight, bias=None):
-    """Linear dispatch: the skinny streaming kernel for decode-shaped
-    (M<=64) GEMMs on GPU, hipBLASLt (F.linear) otherwise. Off by default
-    until it beats the tuned hipBLASLt table in-engine (ARKS_SKINNY_GEMM=1)."""
+    """Linear dispatch: the skinny split-K kernel for small decode-shaped
+    GEMMs on GPU, hipBLASLt (F.linear) otherwise."""
     if (
         _USE_SKINNY
         and x.is_cuda
@@ -229,6 +233,7 @@ def linear_bf16(x, weight, bias=None):
         and x.dtype == torch.bfloat16
         and weight.shape[0] % 64 == 0
         and weight.shape[1] % 32 == 0
+        and weight.numel() <= _SKINNY_MAX_ELEMS
         and weight.is_contiguous()
         and native_available()
     ):

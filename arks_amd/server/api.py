@@ -38,6 +38,42 @@ from .openai_types import (
 )
 
 
+
+
+def _stop_list(stop) -> list[str]:
+    if stop is None:
+        return []
+    return [stop] if isinstance(stop, str) else [x for x in stop if x]
+
+
+class StopStringTracker:
+    """Incremental stop-string matching over streamed text: holds back the
+    longest possible partial match so a stop split across two pieces is
+    still caught before it is emitted."""
+
+    def __init__(self, stops: list[str]):
+        self.stops = stops
+        self.buf = ""
+        self.hold = max((len(x) for x in stops), default=1) - 1
+
+    def feed(self, piece: str) -> tuple[str, bool]:
+        if not self.stops:
+            return piece, False
+        self.buf += piece
+        cut = min((i for i in (self.buf.find(x) for x in self.stops) if i != -1),
+                  default=-1)
+        if cut != -1:
+            emit, self.buf = self.buf[:cut], ""
+            return emit, True
+        emit = self.buf[: len(self.buf) - self.hold] if self.hold else self.buf
+        self.buf = self.buf[len(emit):]
+        return emit, False
+
+    def flush(self) -> str:
+        out, self.buf = self.buf, ""
+        return out
+
+
 def _error(status: int, message: str) -> JSONResponse:
     return JSONResponse(
         status_code=status, content=ErrorResponse.make(message, status).model_dump()
@@ -78,12 +114,16 @@ def create_app(engine: AsyncEngine, served_model_name: str, tokenizer,
     async def metrics():
         return PlainTextResponse(engine.metrics.render().decode())
 
-    def _sampling(max_tokens, temperature, top_p, ignore_eos) -> SamplingParams:
+    def _sampling(req, max_tokens) -> SamplingParams:
         return SamplingParams(
             max_tokens=max_tokens if max_tokens is not None else 1024,
-            temperature=max(0.0, temperature),
-            top_p=top_p,
-            ignore_eos=ignore_eos,
+            temperature=max(0.0, req.temperature),
+            top_p=req.top_p,
+            top_k=getattr(req, "top_k", 0),
+            presence_penalty=getattr(req, "presence_penalty", 0.0),
+            frequency_penalty=getattr(req, "frequency_penalty", 0.0),
+            repetition_penalty=getattr(req, "repetition_penalty", 1.0),
+            ignore_eos=req.ignore_eos,
         )
 
     @app.post("/v1/chat/completions")
@@ -92,24 +132,33 @@ def create_app(engine: AsyncEngine, served_model_name: str, tokenizer,
             return _error(404, f"model {req.model!r} not found")
         prompt = tokenizer.apply_chat_template([m.model_dump() for m in req.messages])
         token_ids = tokenizer.encode(prompt)
-        sp = _sampling(
-            req.max_completion_tokens or req.max_tokens,
-            req.temperature, req.top_p, req.ignore_eos,
-        )
+        sp = _sampling(req, req.max_completion_tokens or req.max_tokens)
         rid = f"chatcmpl-{uuid.uuid4().hex}"
         if req.stream:
             return StreamingResponse(
                 _chat_stream(engine, tokenizer, req, rid, token_ids, sp, raw),
                 media_type="text/event-stream",
             )
+        stops = _stop_list(req.stop)
+        tracker = StopStringTracker(stops)
         text_ids: list[int] = []
+        text_acc = ""
         finish = None
         async for out in engine.generate_stream(
             rid, token_ids, sp, prefill_addr=raw.headers.get("x-arks-prefill-addr")
         ):
             text_ids.append(out.new_token_id)
+            if stops:
+                emit, stopped = tracker.feed(tokenizer.decode([out.new_token_id]))
+                text_acc += emit
+                if stopped:
+                    finish = "stop"
+                    engine.abort(rid)
+                    break
             if out.finished:
                 finish = out.finish_reason
+        if stops and finish != "stop":
+            text_acc += tracker.flush()
         usage = Usage(
             prompt_tokens=len(token_ids),
             completion_tokens=len(text_ids),
@@ -120,7 +169,10 @@ def create_app(engine: AsyncEngine, served_model_name: str, tokenizer,
             model=req.model,
             choices=[
                 ChatChoice(
-                    message=ChatMessage(role="assistant", content=tokenizer.decode(text_ids)),
+                    message=ChatMessage(
+                        role="assistant",
+                        content=text_acc if stops else tokenizer.decode(text_ids),
+                    ),
                     finish_reason=finish or "stop",
                 )
             ],
@@ -137,6 +189,7 @@ def create_app(engine: AsyncEngine, served_model_name: str, tokenizer,
         yield f"data: {first.model_dump_json(exclude_none=True)}\n\n"
         n_out = 0
         finish = None
+        tracker = StopStringTracker(_stop_list(req.stop))
         try:
             async for out in engine.generate_stream(
             rid, token_ids, sp, prefill_addr=raw.headers.get("x-arks-prefill-addr")
@@ -145,15 +198,24 @@ def create_app(engine: AsyncEngine, served_model_name: str, tokenizer,
                     engine.abort(rid)
                     return
                 n_out += 1
-                piece = tokenizer.decode([out.new_token_id])
-                chunk = ChatCompletionChunk(
-                    id=rid, model=req.model, created=created,
-                    choices=[ChatDeltaChoice(delta={"content": piece},
-                                             finish_reason=out.finish_reason if out.finished else None)],
-                )
-                if out.finished:
+                piece, stopped = tracker.feed(tokenizer.decode([out.new_token_id]))
+                if stopped:
+                    finish = "stop"
+                    engine.abort(rid)
+                elif out.finished:
                     finish = out.finish_reason
-                yield f"data: {chunk.model_dump_json(exclude_none=True)}\n\n"
+                    piece += tracker.flush()
+                done = stopped or out.finished
+                if piece or done:
+                    chunk = ChatCompletionChunk(
+                        id=rid, model=req.model, created=created,
+                        choices=[ChatDeltaChoice(
+                            delta={"content": piece},
+                            finish_reason=finish if done else None)],
+                    )
+                    yield f"data: {chunk.model_dump_json(exclude_none=True)}\n\n"
+                if stopped:
+                    break
         except asyncio.CancelledError:
             engine.abort(rid)
             raise
@@ -187,25 +249,38 @@ def create_app(engine: AsyncEngine, served_model_name: str, tokenizer,
             token_ids = list(p[0])
         else:
             return _error(400, "invalid prompt")
-        sp = _sampling(req.max_tokens, req.temperature, req.top_p, req.ignore_eos)
+        sp = _sampling(req, req.max_tokens)
         rid = f"cmpl-{uuid.uuid4().hex}"
         if req.stream:
             return StreamingResponse(
                 _completion_stream(engine, tokenizer, req, rid, token_ids, sp, raw),
                 media_type="text/event-stream",
             )
+        stops = _stop_list(req.stop)
+        tracker = StopStringTracker(stops)
         out_ids: list[int] = []
+        text_acc = ""
         finish = None
         async for out in engine.generate_stream(
             rid, token_ids, sp, prefill_addr=raw.headers.get("x-arks-prefill-addr")
         ):
             out_ids.append(out.new_token_id)
+            if stops:
+                emit, stopped = tracker.feed(tokenizer.decode([out.new_token_id]))
+                text_acc += emit
+                if stopped:
+                    finish = "stop"
+                    engine.abort(rid)
+                    break
             if out.finished:
                 finish = out.finish_reason
+        if stops and finish != "stop":
+            text_acc += tracker.flush()
         return CompletionResponse(
             id=rid, model=req.model,
-            choices=[CompletionChoice(text=tokenizer.decode(out_ids),
-                                      finish_reason=finish or "stop")],
+            choices=[CompletionChoice(
+                text=text_acc if stops else tokenizer.decode(out_ids),
+                finish_reason=finish or "stop")],
             usage=Usage(
                 prompt_tokens=len(token_ids),
                 completion_tokens=len(out_ids),
@@ -217,6 +292,7 @@ def create_app(engine: AsyncEngine, served_model_name: str, tokenizer,
         created = int(time.time())
         include_usage = bool(req.stream_options and req.stream_options.include_usage)
         n_out = 0
+        tracker = StopStringTracker(_stop_list(req.stop))
         try:
             async for out in engine.generate_stream(
             rid, token_ids, sp, prefill_addr=raw.headers.get("x-arks-prefill-addr")
@@ -225,16 +301,29 @@ def create_app(engine: AsyncEngine, served_model_name: str, tokenizer,
                     engine.abort(rid)
                     return
                 n_out += 1
-                chunk = {
-                    "id": rid, "object": "text_completion", "created": created,
-                    "model": req.model,
-                    "choices": [{
-                        "index": 0,
-                        "text": tokenizer.decode([out.new_token_id]),
-                        "finish_reason": out.finish_reason if out.finished else None,
-                    }],
-                }
-                yield f"data: {json.dumps(chunk)}\n\n"
+                piece, stopped = tracker.feed(tokenizer.decode([out.new_token_id]))
+                if stopped:
+                    finish = "stop"
+                    engine.abort(rid)
+                elif out.finished:
+                    finish = out.finish_reason
+                    piece += tracker.flush()
+                else:
+                    finish = None
+                done = stopped or out.finished
+                if piece or done:
+                    chunk = {
+                        "id": rid, "object": "text_completion", "created": created,
+                        "model": req.model,
+                        "choices": [{
+                            "index": 0,
+                            "text": piece,
+                            "finish_reason": finish if done else None,
+                        }],
+                    }
+                    yield f"data: {json.dumps(chunk)}\n\n"
+                if stopped:
+                    break
         except asyncio.CancelledError:
             engine.abort(rid)
             raise

@@ -31,6 +31,10 @@ class ChatCompletionRequest(BaseModel):
     stream: bool = False
     stream_options: StreamOptions | None = None
     stop: list[str] | str | None = None
+    top_k: int = 0  # extension (vLLM-compatible)
+    presence_penalty: float = 0.0
+    frequency_penalty: float = 0.0
+    repetition_penalty: float = 1.0  # extension (vLLM-compatible)
     seed: int | None = None
     ignore_eos: bool = False  # extension (load testing)
 
@@ -44,6 +48,11 @@ class CompletionRequest(BaseModel):
     n: int = 1
     stream: bool = False
     stream_options: StreamOptions | None = None
+    stop: list[str] | str | None = None
+    top_k: int = 0
+    presence_penalty: float = 0.0
+    frequency_penalty: float = 0.0
+    repetition_penalty: float = 1.0
     seed: int | None = None
     ignore_eos: bool = False
 

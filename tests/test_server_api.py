@@ -162,3 +162,43 @@ def test_concurrent_requests_batched(app):
         assert results == [5] * 6
 
     run_with_client(app, fn)
+
+
+def test_completion_stop_string(app):
+    async def fn(client):
+        # ByteTokenizer round-trips ASCII; with temperature 0 on random
+        # weights the output is arbitrary bytes, so use a stop that is
+        # guaranteed to appear: a single char from the decoded output of a
+        # no-stop run.
+        r0 = await client.post("/v1/completions", json={
+            "model": "tiny", "prompt": "abc", "max_tokens": 8,
+            "temperature": 0, "ignore_eos": True,
+        })
+        full = r0.json()["choices"][0]["text"]
+        assert full
+        stop_ch = full[len(full) // 2]
+        r = await client.post("/v1/completions", json={
+            "model": "tiny", "prompt": "abc", "max_tokens": 8,
+            "temperature": 0, "ignore_eos": True, "stop": [stop_ch],
+        })
+        body = r.json()
+        text = body["choices"][0]["text"]
+        assert stop_ch not in text
+        assert text == full.split(stop_ch)[0]
+        assert body["choices"][0]["finish_reason"] == "stop"
+
+    run_with_client(app, fn)
+
+
+def test_stop_string_tracker_split_across_pieces():
+    from arks_amd.server.api import StopStringTracker
+
+    t = StopStringTracker(["END"])
+    out = ""
+    stopped = False
+    for piece in ["hello E", "N", "D tail"]:
+        emit, stopped = t.feed(piece)
+        out += emit
+        if stopped:
+            break
+    assert stopped and out == "hello "
