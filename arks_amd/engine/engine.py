@@ -114,14 +114,16 @@ class LLMEngine:
         if sb is None:
             return []
         new_tokens = self.runner.execute(sb)
+        logprobs = self.runner.take_logprobs()
         outputs: list[StepOutput] = []
         eos = self.model_cfg.eos_token_id
-        for seq, tok in zip(sb.seqs, new_tokens):
+        for i, (seq, tok) in enumerate(zip(sb.seqs, new_tokens)):
             if seq.status is not SeqStatus.RUNNING:  # aborted mid-step
                 continue
             seq.append_token(tok)
             self.total_output_tokens += 1
             finished = seq.check_finished(eos)
+            lp, top = logprobs.get(i, (None, None))
             outputs.append(
                 StepOutput(
                     request_id=seq.request_id,
@@ -131,6 +133,8 @@ class LLMEngine:
                     finish_reason=seq.finish_reason,
                     num_prompt_tokens=seq.num_prompt_tokens,
                     num_output_tokens=seq.num_output_tokens,
+                    logprob=lp,
+                    top_logprobs=top,
                 )
             )
         self.scheduler.free_finished()

@@ -346,7 +346,34 @@ class ModelRunner:
                 logits.shape, dtype=torch.float32, device=logits.device
             )
             ids = ops.sample_tokens(logits.contiguous(), t, u)
-        return ids.tolist()
+        out = ids.tolist()
+        self._last_logprobs = self._gather_logprobs(logits, seqs, out)
+        return out
+
+    _last_logprobs: dict | None = None
+
+    def take_logprobs(self) -> dict:
+        """Per-seq-index logprob info from the last sample() call."""
+        lp, self._last_logprobs = self._last_logprobs, None
+        return lp or {}
+
+    @staticmethod
+    def _gather_logprobs(logits, seqs, chosen: list[int]) -> dict | None:
+        rows = [i for i, s in enumerate(seqs) if s.sampling.logprobs is not None]
+        if not rows:
+            return None
+        out: dict[int, tuple[float, dict[int, float] | None]] = {}
+        idx = torch.tensor(rows, dtype=torch.long, device=logits.device)
+        lps = torch.log_softmax(logits[idx].float(), dim=-1)
+        for j, i in enumerate(rows):
+            n = seqs[i].sampling.logprobs
+            chosen_lp = float(lps[j, chosen[i]])
+            top = None
+            if n:
+                v, t = lps[j].topk(n)
+                top = {int(ti): float(vi) for vi, ti in zip(v, t)}
+            out[i] = (chosen_lp, top)
+        return out
 
     @staticmethod
     def _apply_penalties(logits: torch.Tensor, seqs: list[Sequence]) -> torch.Tensor:

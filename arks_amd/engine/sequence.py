@@ -18,6 +18,7 @@ class SamplingParams:
     repetition_penalty: float = 1.0  # >1 divides positive seen-logits
     stop_token_ids: tuple[int, ...] = ()
     ignore_eos: bool = False
+    logprobs: int | None = None  # None = off; 0 = chosen only; N = top-N too
 
     @property
     def has_penalties(self) -> bool:
@@ -118,3 +119,5 @@ class StepOutput:
     finish_reason: str | None
     num_prompt_tokens: int
     num_output_tokens: int
+    logprob: float | None = None  # chosen token's logprob (when requested)
+    top_logprobs: dict[int, float] | None = None

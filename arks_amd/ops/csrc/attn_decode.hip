@@ -26,6 +26,33 @@ namespace arks {
 
 constexpr int KV_BLOCK_SIZE = 16;  // tokens per KV page (matches engine)
 
+// MFMA fragment types (shared convention with attn_prefill.hip; layout
+// verified on hardware by tests/test_ops_gpu.py::test_mfma_probe).
+typedef __attribute__((ext_vector_type(8))) __bf16 bf16x8;
+typedef __attribute__((ext_vector_type(4))) float f32x4;
+
+__device__ __forceinline__ f32x4 dec_mfma(bf16x8 a, bf16x8 b, f32x4 c) {
+  return __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, c, 0, 0, 0);
+}
+
+__device__ __forceinline__ int dec_frag_k(int a, int jj) { return 8 * a + jj; }
+
+constexpr int DEC_CHUNK_TOK = 32;  // tokens per wave iteration (2 KV pages)
+constexpr int DEC_VT_PAD = 40;     // padded token axis in V^T / P LDS tiles
+
+// v3 (MFMA): one workgroup per (seq, kv_head, partition); the GQA head
+// group (GQ <= 16 q-heads padded to a 16-row tile) is the MFMA M axis, so
+// QK^T for all heads of the group over a 16-token page is ONE mfma chain
+// instead of GQ serial 16-lane shuffle reductions (v2 spent ~112 cyc/token
+// on shuffles at GQ=7 and reached only 3.0 TB/s of KV stream). Each of the
+// 4 waves owns a 32-token chunk (wave-strided over the partition): K feeds
+// the A-operand straight from cache pages (16 B/lane), V is transposed
+// through a per-wave LDS tile for the PV B-operand, P goes through a
+// per-wave LDS tile exactly as in attn_prefill.hip. Column-wise online
+// softmax: lane lq owns head lq, so the per-tile reduction is two
+// shfl_xor's for ALL heads at once. Wave partials merge flash-style in LDS
+// (the V/P tiles are re-used as the merge buffer), and the partition
+// output/partial format is unchanged from v2 (combine kernel below).
 template <int HEAD_DIM, int GQ>
 __global__ __launch_bounds__(256) void attn_decode_kernel(
     bf16* __restrict__ out,        // [S, Hq, D] (used when gridDim.z == 1)
@@ -37,8 +64,10 @@ __global__ __launch_bounds__(256) void attn_decode_kernel(
     const int* __restrict__ seq_lens,      // [S]
     const float scale, const int num_kv_heads, const int max_blocks,
     const int64_t q_stride) {
-  constexpr int E = HEAD_DIM / 16;  // elems per lane (8 for D=128)
   constexpr int NUM_WAVES = 4;
+  constexpr int STEPS = HEAD_DIM / 32;   // QK^T contraction steps
+  constexpr int CHUNKS = HEAD_DIM / 16;  // PV output dim chunks
+  constexpr int E = HEAD_DIM / 16;       // dim elems per lane16 slice
 
   const int seq = blockIdx.y;
   const int kvh = blockIdx.x;
@@ -48,153 +77,208 @@ __global__ __launch_bounds__(256) void attn_decode_kernel(
   const int nblocks = (L + KV_BLOCK_SIZE - 1) / KV_BLOCK_SIZE;
   const int num_q_heads = num_kv_heads * GQ;
 
-  // Partition p owns pages [p*chunk, min((p+1)*chunk, nblocks)).
-  const int chunk = (nblocks + nparts - 1) / nparts;
+  const int chunk = (nblocks + nparts - 1) / nparts;  // pages per partition
   const int pb_lo = part * chunk;
   const int pb_hi = min(pb_lo + chunk, nblocks);
-
   const int tid = threadIdx.x;
   const int wave = tid / WAVE_SIZE;
   const int lane = tid % WAVE_SIZE;
-  const int grp = lane / 16;     // 16-lane group within wave
-  const int lane16 = lane % 16;  // position within group
+  const int lq = lane % 16;
+  const int la = lane / 16;
 
-  // Q fragments (scale folded in).
-  float qreg[GQ][E];
+  // Per-wave LDS region: V^T tile + P tile during the main loop, merge
+  // buffer afterwards (union; a wave touches only its own region between
+  // the loop and the final barrier).
+  constexpr int VT_BYTES = HEAD_DIM * DEC_VT_PAD * 2;
+  constexpr int P_BYTES = 16 * DEC_VT_PAD * 2;
+  constexpr int MERGE_BYTES = 16 * (HEAD_DIM + 2) * 4;
+  constexpr int WAVE_BYTES0 =
+      VT_BYTES + P_BYTES > MERGE_BYTES ? VT_BYTES + P_BYTES : MERGE_BYTES;
+  constexpr int WAVE_BYTES = (WAVE_BYTES0 + 15) & ~15;
+  __shared__ __attribute__((aligned(16))) char smem[NUM_WAVES][WAVE_BYTES];
+  bf16* vt_lds = reinterpret_cast<bf16*>(smem[wave]);              // [D][40]
+  bf16* p_lds = reinterpret_cast<bf16*>(smem[wave] + VT_BYTES);    // [16][40]
+  float* merge_lds = reinterpret_cast<float*>(smem[wave]);         // [16][D+2]
+
+  // Q fragments: B-operand [32k x 16 heads] per step; rows g >= GQ zero.
+  bf16x8 qfrag[STEPS];
+  {
+    const bool valid = lq < GQ;
+    const bf16* qp = q + (int64_t)seq * q_stride +
+                     (kvh * GQ + (valid ? lq : 0)) * HEAD_DIM;
 #pragma unroll
-  for (int g = 0; g < GQ; ++g) {
-    const bf16* qp =
-        q + (int64_t)seq * q_stride + (kvh * GQ + g) * HEAD_DIM + lane16 * E;
-    ushort8 qv8{};
-    if constexpr (E == 8) {
-      qv8 = *reinterpret_cast<const ushort8*>(qp);
-    } else {
-      ushort4v q4 = *reinterpret_cast<const ushort4v*>(qp);
-#pragma unroll
-      for (int e = 0; e < 4; ++e) qv8[e] = q4[e];
+    for (int st = 0; st < STEPS; ++st) {
+      ushort8 u{};
+      if (valid)
+        u = *reinterpret_cast<const ushort8*>(qp + st * 32 + dec_frag_k(la, 0));
+      qfrag[st] = *reinterpret_cast<bf16x8*>(&u);
     }
-#pragma unroll
-    for (int e = 0; e < E; ++e) qreg[g][e] = bf16_bits_to_float(qv8[e]) * scale;
   }
 
-  float m[GQ], lsum[GQ], acc[GQ][E];
+  float m_run = -FLT_MAX;  // per head lq (replicated over la)
+  float l_run = 0.f;
+  f32x4 oacc[CHUNKS];
 #pragma unroll
-  for (int g = 0; g < GQ; ++g) {
-    m[g] = -FLT_MAX;
-    lsum[g] = 0.f;
-#pragma unroll
-    for (int e = 0; e < E; ++e) acc[g][e] = 0.f;
-  }
+  for (int c = 0; c < CHUNKS; ++c) oacc[c] = {0.f, 0.f, 0.f, 0.f};
 
   const int* bt = block_tables + (int64_t)seq * max_blocks;
+  const int64_t page_elems = (int64_t)num_kv_heads * KV_BLOCK_SIZE * HEAD_DIM;
+  const int64_t head_off = (int64_t)kvh * KV_BLOCK_SIZE * HEAD_DIM;
 
-  for (int pb = pb_lo + wave; pb < pb_hi; pb += NUM_WAVES) {
-    const int64_t phys = bt[pb];
-    const bf16* kb = k_cache + ((phys * num_kv_heads + kvh) * KV_BLOCK_SIZE) *
-                                   (int64_t)HEAD_DIM;
-    const bf16* vb = v_cache + ((phys * num_kv_heads + kvh) * KV_BLOCK_SIZE) *
-                                   (int64_t)HEAD_DIM;
-#pragma unroll
-    for (int it = 0; it < KV_BLOCK_SIZE / 4; ++it) {
-      const int t = grp + it * 4;  // token within page
-      const int tok = pb * KV_BLOCK_SIZE + t;
-      const bool active = tok < L;
-      float kf[E], vf[E];
-      if (active) {
-        const bf16* kp = kb + t * HEAD_DIM + lane16 * E;
-        const bf16* vp = vb + t * HEAD_DIM + lane16 * E;
-        if constexpr (E == 8) {
-          ushort8 kv8 = *reinterpret_cast<const ushort8*>(kp);
-          ushort8 vv8 = *reinterpret_cast<const ushort8*>(vp);
-#pragma unroll
-          for (int e = 0; e < 8; ++e) {
-            kf[e] = bf16_bits_to_float(kv8[e]);
-            vf[e] = bf16_bits_to_float(vv8[e]);
-          }
-        } else {
-          ushort4v kv4 = *reinterpret_cast<const ushort4v*>(kp);
-          ushort4v vv4 = *reinterpret_cast<const ushort4v*>(vp);
-#pragma unroll
-          for (int e = 0; e < E; ++e) {
-            kf[e] = bf16_bits_to_float(kv4[e]);
-            vf[e] = bf16_bits_to_float(vv4[e]);
-          }
+  // 32-token chunks, wave-strided over the partition's pages.
+  for (int pb = pb_lo + wave * 2; pb < pb_hi; pb += NUM_WAVES * 2) {
+    const int64_t pbase0 = (int64_t)bt[pb] * page_elems + head_off;
+    const bool have_p1 = pb + 1 < pb_hi;
+    const int64_t pbase1 =
+        have_p1 ? (int64_t)bt[pb + 1] * page_elems + head_off : pbase0;
+    const int tok0 = pb * KV_BLOCK_SIZE;  // first token of the chunk
+    // valid tokens here: bounded by L AND by the partition's page range
+    const int kmax =
+        min(min(L - tok0, (pb_hi - pb) * KV_BLOCK_SIZE), DEC_CHUNK_TOK);
+
+    // ---- stage V^T (transposed, zero-padded) for this chunk.
+    {
+      const int nvec = DEC_CHUNK_TOK * HEAD_DIM / 8;
+      for (int i = lane; i < nvec; i += WAVE_SIZE) {
+        const int t = i / (HEAD_DIM / 8);
+        const int col8 = (i % (HEAD_DIM / 8)) * 8;
+        ushort8 vv{};
+        if (t < kmax) {
+          const int64_t src = (t < KV_BLOCK_SIZE ? pbase0 : pbase1) +
+                              (int64_t)(t & (KV_BLOCK_SIZE - 1)) * HEAD_DIM +
+                              col8;
+          vv = *reinterpret_cast<const ushort8*>(v_cache + src);
         }
+#pragma unroll
+        for (int e = 0; e < 8; ++e) vt_lds[(col8 + e) * DEC_VT_PAD + t] = vv[e];
       }
+    }
+
+    // ---- QK^T: A-frag = K rows from the pages, B-frag = Q registers.
+    f32x4 sc[2];
+    sc[0] = {0.f, 0.f, 0.f, 0.f};
+    sc[1] = {0.f, 0.f, 0.f, 0.f};
 #pragma unroll
-      for (int g = 0; g < GQ; ++g) {
-        float s = 0.f;
-        if (active) {
+    for (int sub = 0; sub < 2; ++sub) {
+      const int64_t pbase = sub == 0 ? pbase0 : pbase1;
+      const bool sub_any = sub == 0 || have_p1;
 #pragma unroll
-          for (int e = 0; e < E; ++e) s += qreg[g][e] * kf[e];
-        }
-        s = group16_reduce_sum(s);  // all 16 lanes get the dot product
-        if (active) {
-          const float mn = fmaxf(m[g], s);
-          const float alpha = __expf(m[g] - mn);
-          const float p = __expf(s - mn);
-          lsum[g] = lsum[g] * alpha + p;
-#pragma unroll
-          for (int e = 0; e < E; ++e) acc[g][e] = acc[g][e] * alpha + p * vf[e];
-          m[g] = mn;
-        }
+      for (int st = 0; st < STEPS; ++st) {
+        ushort8 u{};
+        if (sub_any)
+          u = *reinterpret_cast<const ushort8*>(
+              k_cache + pbase + (int64_t)lq * HEAD_DIM + st * 32 +
+              dec_frag_k(la, 0));
+        sc[sub] = dec_mfma(*reinterpret_cast<bf16x8*>(&u), qfrag[st], sc[sub]);
       }
+    }
+
+    // ---- masked column-wise online softmax (head = lq; tokens in lanes).
+    float p[8];
+    float tile_max = -FLT_MAX;
+#pragma unroll
+    for (int sub = 0; sub < 2; ++sub) {
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const int t = sub * 16 + 4 * la + r;
+        const int i = sub * 4 + r;
+        p[i] = (t < kmax) ? sc[sub][r] * scale : -FLT_MAX;
+        tile_max = fmaxf(tile_max, p[i]);
+      }
+    }
+    tile_max = fmaxf(tile_max, __shfl_xor(tile_max, 16, WAVE_SIZE));
+    tile_max = fmaxf(tile_max, __shfl_xor(tile_max, 32, WAVE_SIZE));
+
+    const float m_new = fmaxf(m_run, tile_max);
+    float alpha = 1.f;
+    float psum = 0.f;
+    if (m_new > -FLT_MAX) {
+      alpha = __expf(m_run - m_new);
+#pragma unroll
+      for (int i = 0; i < 8; ++i) {
+        p[i] = (p[i] > -FLT_MAX) ? __expf(p[i] - m_new) : 0.f;
+        psum += p[i];
+      }
+    } else {
+#pragma unroll
+      for (int i = 0; i < 8; ++i) p[i] = 0.f;
+    }
+    psum += __shfl_xor(psum, 16, WAVE_SIZE);
+    psum += __shfl_xor(psum, 32, WAVE_SIZE);
+    l_run = l_run * alpha + psum;
+    m_run = m_new;
+
+    // rescale O rows (head = 4*la + r in the PV fragment).
+    float row_alpha[4];
+#pragma unroll
+    for (int r = 0; r < 4; ++r)
+      row_alpha[r] = __shfl(alpha, 4 * la + r, WAVE_SIZE);
+#pragma unroll
+    for (int c = 0; c < CHUNKS; ++c) {
+#pragma unroll
+      for (int r = 0; r < 4; ++r) oacc[c][r] *= row_alpha[r];
+    }
+
+    // ---- P to LDS (bf16) then PV (contract the full 32-token chunk).
+#pragma unroll
+    for (int sub = 0; sub < 2; ++sub) {
+      ushort4v pk;
+#pragma unroll
+      for (int r = 0; r < 4; ++r) pk[r] = float_to_bf16_bits(p[sub * 4 + r]);
+      *reinterpret_cast<ushort4v*>(&p_lds[lq * DEC_VT_PAD + sub * 16 + 4 * la]) =
+          pk;
+    }
+    // Intra-wave LDS producer->consumer: lanes of ONE wave — no barrier,
+    // but the compiler must not reorder the reads above the writes; the
+    // lgkmcnt wait is enough within a wave.
+    __builtin_amdgcn_s_waitcnt(0);  // drain LDS writes (lgkmcnt)
+    ushort8 pa = *reinterpret_cast<const ushort8*>(
+        &p_lds[lq * DEC_VT_PAD + dec_frag_k(la, 0)]);
+#pragma unroll
+    for (int c = 0; c < CHUNKS; ++c) {
+      ushort8 vb = *reinterpret_cast<const ushort8*>(
+          &vt_lds[(c * 16 + lq) * DEC_VT_PAD + dec_frag_k(la, 0)]);
+      oacc[c] = dec_mfma(*reinterpret_cast<bf16x8*>(&pa),
+                         *reinterpret_cast<bf16x8*>(&vb), oacc[c]);
     }
   }
 
-  // Merge the four 16-lane groups of each wave (lanes l, l^16, l^32, l^48
-  // share the same dim slice lane16*E).
+  // ---- merge the 4 wave partials (flash-style) and emit.
+  __syncthreads();  // everyone done with vt/p before the union flips
+  {
+    // wave writes its [16 head][D] partial + per-head (m, l)
 #pragma unroll
-  for (int off = 16; off <= 32; off <<= 1) {
+    for (int c = 0; c < CHUNKS; ++c) {
 #pragma unroll
-    for (int g = 0; g < GQ; ++g) {
-      const float om = __shfl_xor(m[g], off, WAVE_SIZE);
-      const float ol = __shfl_xor(lsum[g], off, WAVE_SIZE);
-      const float mn = fmaxf(m[g], om);
-      const float a1 = __expf(m[g] - mn);
-      const float a2 = __expf(om - mn);
-      lsum[g] = lsum[g] * a1 + ol * a2;
-#pragma unroll
-      for (int e = 0; e < E; ++e) {
-        const float oa = __shfl_xor(acc[g][e], off, WAVE_SIZE);
-        acc[g][e] = acc[g][e] * a1 + oa * a2;
-      }
-      m[g] = mn;
+      for (int r = 0; r < 4; ++r)
+        merge_lds[(4 * la + r) * (HEAD_DIM + 2) + c * 16 + lq] = oacc[c][r];
     }
-  }
-
-  // Cross-wave merge via LDS: wave partials [w][g][D + 2] f32.
-  __shared__ float lds[NUM_WAVES][GQ][HEAD_DIM + 2];
-  if (grp == 0) {
-#pragma unroll
-    for (int g = 0; g < GQ; ++g) {
-#pragma unroll
-      for (int e = 0; e < E; ++e) lds[wave][g][lane16 * E + e] = acc[g][e];
-      if (lane16 == 0) {
-        lds[wave][g][HEAD_DIM] = m[g];
-        lds[wave][g][HEAD_DIM + 1] = lsum[g];
-      }
+    if (la == 0) {
+      merge_lds[lq * (HEAD_DIM + 2) + HEAD_DIM] = m_run;
+      merge_lds[lq * (HEAD_DIM + 2) + HEAD_DIM + 1] = l_run;
     }
   }
   __syncthreads();
 
-  // Head g is finalized by wave g % NUM_WAVES, lanes 0..15.
-  if (grp == 0) {
-    for (int g = wave; g < GQ; g += NUM_WAVES) {
+  // 256 threads = 16 heads x 16 dim slices; head g merged over the 4 waves.
+  {
+    const int g = tid / 16;
+    const int lane16 = tid % 16;
+    if (g < GQ) {
       float mf = -FLT_MAX, lf = 0.f, af[E];
 #pragma unroll
       for (int e = 0; e < E; ++e) af[e] = 0.f;
-#pragma unroll
       for (int w = 0; w < NUM_WAVES; ++w) {
-        const float mw = lds[w][g][HEAD_DIM];
-        const float lw = lds[w][g][HEAD_DIM + 1];
+        const float* wl = reinterpret_cast<const float*>(smem[w]);
+        const float mw = wl[g * (HEAD_DIM + 2) + HEAD_DIM];
+        const float lw = wl[g * (HEAD_DIM + 2) + HEAD_DIM + 1];
         const float mn = fmaxf(mf, mw);
         const float a1 = __expf(mf - mn);
         const float a2 = __expf(mw - mn);
         lf = lf * a1 + lw * a2;
 #pragma unroll
         for (int e = 0; e < E; ++e)
-          af[e] = af[e] * a1 + lds[w][g][lane16 * E + e] * a2;
+          af[e] = af[e] * a1 + wl[g * (HEAD_DIM + 2) + lane16 * E + e] * a2;
         mf = mn;
       }
       if (nparts == 1) {
@@ -215,8 +299,8 @@ __global__ __launch_bounds__(256) void attn_decode_kernel(
         }
       } else {
         float* pp = part_out +
-                    ((((int64_t)seq * num_kv_heads + kvh) * nparts + part) * GQ + g) *
-                        (HEAD_DIM + 2);
+                    ((((int64_t)seq * num_kv_heads + kvh) * nparts + part) * GQ +
+                     g) * (HEAD_DIM + 2);
 #pragma unroll
         for (int e = 0; e < E; ++e) pp[lane16 * E + e] = af[e];
         if (lane16 == 0) {
@@ -227,6 +311,7 @@ __global__ __launch_bounds__(256) void attn_decode_kernel(
     }
   }
 }
+
 
 // Combine partials: one wave per (seq, q_head); lanes 0..15 hold dim slices.
 template <int HEAD_DIM>

@@ -40,6 +40,13 @@ from .openai_types import (
 
 
 
+def _logprobs_of(req) -> int | None:
+    lp = getattr(req, "logprobs", None)
+    if isinstance(lp, bool):  # chat API shape: logprobs + top_logprobs
+        return getattr(req, "top_logprobs", 0) if lp else None
+    return lp  # completions API shape: int | None
+
+
 def _stop_list(stop) -> list[str]:
     if stop is None:
         return []
@@ -123,6 +130,7 @@ def create_app(engine: AsyncEngine, served_model_name: str, tokenizer,
             presence_penalty=getattr(req, "presence_penalty", 0.0),
             frequency_penalty=getattr(req, "frequency_penalty", 0.0),
             repetition_penalty=getattr(req, "repetition_penalty", 1.0),
+            logprobs=_logprobs_of(req),
             ignore_eos=req.ignore_eos,
         )
 
@@ -143,11 +151,21 @@ def create_app(engine: AsyncEngine, served_model_name: str, tokenizer,
         tracker = StopStringTracker(stops)
         text_ids: list[int] = []
         text_acc = ""
+        lp_content: list[dict] = []
         finish = None
         async for out in engine.generate_stream(
             rid, token_ids, sp, prefill_addr=raw.headers.get("x-arks-prefill-addr")
         ):
             text_ids.append(out.new_token_id)
+            if out.logprob is not None:
+                lp_content.append({
+                    "token": tokenizer.decode([out.new_token_id]),
+                    "logprob": out.logprob,
+                    "top_logprobs": [
+                        {"token": tokenizer.decode([t]), "logprob": v}
+                        for t, v in (out.top_logprobs or {}).items()
+                    ],
+                })
             if stops:
                 emit, stopped = tracker.feed(tokenizer.decode([out.new_token_id]))
                 text_acc += emit
@@ -174,6 +192,7 @@ def create_app(engine: AsyncEngine, served_model_name: str, tokenizer,
                         content=text_acc if stops else tokenizer.decode(text_ids),
                     ),
                     finish_reason=finish or "stop",
+                    logprobs={"content": lp_content} if lp_content else None,
                 )
             ],
             usage=usage,
@@ -261,10 +280,18 @@ def create_app(engine: AsyncEngine, served_model_name: str, tokenizer,
         out_ids: list[int] = []
         text_acc = ""
         finish = None
+        token_lps: list[float] = []
+        top_lps: list[dict] = []
         async for out in engine.generate_stream(
             rid, token_ids, sp, prefill_addr=raw.headers.get("x-arks-prefill-addr")
         ):
             out_ids.append(out.new_token_id)
+            if out.logprob is not None:
+                token_lps.append(out.logprob)
+                top_lps.append({
+                    tokenizer.decode([t]): v
+                    for t, v in (out.top_logprobs or {}).items()
+                })
             if stops:
                 emit, stopped = tracker.feed(tokenizer.decode([out.new_token_id]))
                 text_acc += emit
@@ -280,7 +307,15 @@ def create_app(engine: AsyncEngine, served_model_name: str, tokenizer,
             id=rid, model=req.model,
             choices=[CompletionChoice(
                 text=text_acc if stops else tokenizer.decode(out_ids),
-                finish_reason=finish or "stop")],
+                finish_reason=finish or "stop",
+                logprobs=(
+                    {
+                        "tokens": [tokenizer.decode([t]) for t in out_ids],
+                        "token_logprobs": token_lps,
+                        "top_logprobs": top_lps,
+                    }
+                    if token_lps else None
+                ))],
             usage=Usage(
                 prompt_tokens=len(token_ids),
                 completion_tokens=len(out_ids),

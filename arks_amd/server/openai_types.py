@@ -31,6 +31,8 @@ class ChatCompletionRequest(BaseModel):
     stream: bool = False
     stream_options: StreamOptions | None = None
     stop: list[str] | str | None = None
+    logprobs: bool = False
+    top_logprobs: int = 0
     top_k: int = 0  # extension (vLLM-compatible)
     presence_penalty: float = 0.0
     frequency_penalty: float = 0.0
@@ -49,6 +51,7 @@ class CompletionRequest(BaseModel):
     stream: bool = False
     stream_options: StreamOptions | None = None
     stop: list[str] | str | None = None
+    logprobs: int | None = None
     top_k: int = 0
     presence_penalty: float = 0.0
     frequency_penalty: float = 0.0
@@ -67,6 +70,7 @@ class ChatChoice(BaseModel):
     index: int = 0
     message: ChatMessage | None = None
     finish_reason: str | None = None
+    logprobs: dict | None = None
 
 
 class ChatDeltaChoice(BaseModel):
@@ -97,6 +101,7 @@ class CompletionChoice(BaseModel):
     index: int = 0
     text: str = ""
     finish_reason: str | None = None
+    logprobs: dict | None = None
 
 
 class CompletionResponse(BaseModel):
