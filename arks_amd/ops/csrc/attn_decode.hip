@@ -125,7 +125,37 @@ __global__ __launch_bounds__(256) void attn_decode_kernel(
   const int64_t page_elems = (int64_t)num_kv_heads * KV_BLOCK_SIZE * HEAD_DIM;
   const int64_t head_off = (int64_t)kvh * KV_BLOCK_SIZE * HEAD_DIM;
 
-  // 32-token chunks, wave-strided over the partition's pages.
+  // 32-token chunks, wave-strided over the partition's pages. Software
+  // pipeline: chunk i+1's K rides in registers and its loads issue during
+  // chunk i's MFMAs; V loads issue at the top of the chunk and their
+  // LDS-transpose happens after QK+softmax have covered the latency. The
+  // LDS producer->consumer wait is lgkmcnt-only so outstanding K/V global
+  // prefetches keep flowing (a full s_waitcnt(0) here serialized every
+  // chunk: v3 capped at ~3 TB/s of KV stream).
+  auto load_k = [&](int pb, ushort8 (*dst)[STEPS]) {
+    const int64_t pbase0 = (int64_t)bt[pb] * page_elems + head_off;
+    const bool have_p1 = pb + 1 < pb_hi;
+    const int64_t pbase1 =
+        have_p1 ? (int64_t)bt[pb + 1] * page_elems + head_off : pbase0;
+#pragma unroll
+    for (int sub = 0; sub < 2; ++sub) {
+      const int64_t pbase = sub == 0 ? pbase0 : pbase1;
+#pragma unroll
+      for (int st = 0; st < STEPS; ++st) {
+        ushort8 u{};
+        if (sub == 0 || have_p1)
+          u = *reinterpret_cast<const ushort8*>(
+              k_cache + pbase + (int64_t)lq * HEAD_DIM + st * 32 +
+              dec_frag_k(la, 0));
+        dst[sub][st] = u;
+      }
+    }
+  };
+
+  constexpr int NV = DEC_CHUNK_TOK * HEAD_DIM / 8 / WAVE_SIZE;
+  ushort8 kreg[2][STEPS];
+  if (pb_lo + wave * 2 < pb_hi) load_k(pb_lo + wave * 2, kreg);
+
   for (int pb = pb_lo + wave * 2; pb < pb_hi; pb += NUM_WAVES * 2) {
     const int64_t pbase0 = (int64_t)bt[pb] * page_elems + head_off;
     const bool have_p1 = pb + 1 < pb_hi;
@@ -136,40 +166,40 @@ __global__ __launch_bounds__(256) void attn_decode_kernel(
     const int kmax =
         min(min(L - tok0, (pb_hi - pb) * KV_BLOCK_SIZE), DEC_CHUNK_TOK);
 
-    // ---- stage V^T (transposed, zero-padded) for this chunk.
-    {
-      const int nvec = DEC_CHUNK_TOK * HEAD_DIM / 8;
-      for (int i = lane; i < nvec; i += WAVE_SIZE) {
-        const int t = i / (HEAD_DIM / 8);
-        const int col8 = (i % (HEAD_DIM / 8)) * 8;
-        ushort8 vv{};
-        if (t < kmax) {
-          const int64_t src = (t < KV_BLOCK_SIZE ? pbase0 : pbase1) +
-                              (int64_t)(t & (KV_BLOCK_SIZE - 1)) * HEAD_DIM +
-                              col8;
-          vv = *reinterpret_cast<const ushort8*>(v_cache + src);
-        }
+    // ---- V for this chunk into registers (zero-padded); transposed to
+    // LDS only after QK+softmax (in-order LDS keeps it behind the previous
+    // chunk's PV reads).
+    ushort8 vv[NV];
 #pragma unroll
-        for (int e = 0; e < 8; ++e) vt_lds[(col8 + e) * DEC_VT_PAD + t] = vv[e];
+    for (int j = 0; j < NV; ++j) {
+      const int i = lane + WAVE_SIZE * j;
+      const int t = i / (HEAD_DIM / 8);
+      const int col8 = (i % (HEAD_DIM / 8)) * 8;
+      ushort8 v{};
+      if (t < kmax) {
+        const int64_t src = (t < KV_BLOCK_SIZE ? pbase0 : pbase1) +
+                            (int64_t)(t & (KV_BLOCK_SIZE - 1)) * HEAD_DIM +
+                            col8;
+        v = *reinterpret_cast<const ushort8*>(v_cache + src);
       }
+      vv[j] = v;
     }
 
-    // ---- QK^T: A-frag = K rows from the pages, B-frag = Q registers.
+    // ---- prefetch next chunk's K while this chunk computes.
+    ushort8 knext[2][STEPS];
+    const int pnext = pb + NUM_WAVES * 2;
+    if (pnext < pb_hi) load_k(pnext, knext);
+
+    // ---- QK^T: A-frag = K registers, B-frag = Q registers.
     f32x4 sc[2];
     sc[0] = {0.f, 0.f, 0.f, 0.f};
     sc[1] = {0.f, 0.f, 0.f, 0.f};
 #pragma unroll
     for (int sub = 0; sub < 2; ++sub) {
-      const int64_t pbase = sub == 0 ? pbase0 : pbase1;
-      const bool sub_any = sub == 0 || have_p1;
 #pragma unroll
       for (int st = 0; st < STEPS; ++st) {
-        ushort8 u{};
-        if (sub_any)
-          u = *reinterpret_cast<const ushort8*>(
-              k_cache + pbase + (int64_t)lq * HEAD_DIM + st * 32 +
-              dec_frag_k(la, 0));
-        sc[sub] = dec_mfma(*reinterpret_cast<bf16x8*>(&u), qfrag[st], sc[sub]);
+        sc[sub] = dec_mfma(*reinterpret_cast<bf16x8*>(&kreg[sub][st]),
+                           qfrag[st], sc[sub]);
       }
     }
 
@@ -219,6 +249,17 @@ __global__ __launch_bounds__(256) void attn_decode_kernel(
       for (int r = 0; r < 4; ++r) oacc[c][r] *= row_alpha[r];
     }
 
+    // ---- scatter V^T to LDS (after the previous PV reads, in order).
+#pragma unroll
+    for (int j = 0; j < NV; ++j) {
+      const int i = lane + WAVE_SIZE * j;
+      const int t = i / (HEAD_DIM / 8);
+      const int col8 = (i % (HEAD_DIM / 8)) * 8;
+#pragma unroll
+      for (int e = 0; e < 8; ++e)
+        vt_lds[(col8 + e) * DEC_VT_PAD + t] = vv[j][e];
+    }
+
     // ---- P to LDS (bf16) then PV (contract the full 32-token chunk).
 #pragma unroll
     for (int sub = 0; sub < 2; ++sub) {
@@ -228,10 +269,9 @@ __global__ __launch_bounds__(256) void attn_decode_kernel(
       *reinterpret_cast<ushort4v*>(&p_lds[lq * DEC_VT_PAD + sub * 16 + 4 * la]) =
           pk;
     }
-    // Intra-wave LDS producer->consumer: lanes of ONE wave — no barrier,
-    // but the compiler must not reorder the reads above the writes; the
-    // lgkmcnt wait is enough within a wave.
-    __builtin_amdgcn_s_waitcnt(0);  // drain LDS writes (lgkmcnt)
+    // Intra-wave LDS producer->consumer (lanes of one wave): wait on the
+    // LDS counter ONLY — global prefetches must stay outstanding.
+    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
     ushort8 pa = *reinterpret_cast<const ushort8*>(
         &p_lds[lq * DEC_VT_PAD + dec_frag_k(la, 0)]);
 #pragma unroll
@@ -241,6 +281,10 @@ __global__ __launch_bounds__(256) void attn_decode_kernel(
       oacc[c] = dec_mfma(*reinterpret_cast<bf16x8*>(&pa),
                          *reinterpret_cast<bf16x8*>(&vb), oacc[c]);
     }
+#pragma unroll
+    for (int sub = 0; sub < 2; ++sub)
+#pragma unroll
+      for (int st = 0; st < STEPS; ++st) kreg[sub][st] = knext[sub][st];
   }
 
   // ---- merge the 4 wave partials (flash-style) and emit.
