@@ -66,10 +66,15 @@ __global__ __launch_bounds__(256) void attn_prefill_kernel(
   const int lq = lane % 16;   // q-row (QK^T) / dim-col (PV) within fragment
   const int la = lane / 16;   // lane quarter
 
-  // LDS: K tile (XOR-swizzled rows), transposed V tile (padded), per-wave P.
-  // 16-B alignment required for the ushort8 (b128) accesses — guide §6 G17.
-  __shared__ __attribute__((aligned(16))) bf16 k_lds[KTILE][HEAD_DIM];
-  __shared__ __attribute__((aligned(16))) bf16 vt_lds[HEAD_DIM][VT_PAD];
+  // LDS: ping-pong K tiles (XOR-swizzled rows) + transposed V tiles, and a
+  // per-wave P buffer. 16-B alignment required for the ushort8 (b128)
+  // accesses — guide §6 G17. Double buffering + register staging makes the
+  // tile loop a single-barrier software pipeline: tile j+1's global loads
+  // land in registers during tile j's MFMAs, are scattered to the idle
+  // buffer, and one __syncthreads flips the buffers (v1 took two barriers
+  // per tile and stalled on the staging round-trip, ~118 TF effective).
+  __shared__ __attribute__((aligned(16))) bf16 k_lds[2][KTILE][HEAD_DIM];
+  __shared__ __attribute__((aligned(16))) bf16 vt_lds[2][HEAD_DIM][VT_PAD];
   __shared__ __attribute__((aligned(16))) bf16 p_lds[NUM_WAVES][QTILE_WAVE][VT_PAD];
 
   // --- Q fragments: registers, loaded once. Wave w covers rows q0+16w..+15.
@@ -100,33 +105,60 @@ __global__ __launch_bounds__(256) void attn_prefill_kernel(
   const int kmax = min(seq_len, q0 + QTILE);
   const int ntiles = (kmax + KTILE - 1) / KTILE;
 
-  for (int j = 0; j < ntiles; ++j) {
+  // Per-thread cooperative staging share: tile elements i = tid + 256*v.
+  constexpr int NVEC = KTILE * HEAD_DIM / 8;      // vec8 per tile
+  constexpr int VPT = NVEC / 256;                 // vec8 per thread (2 at D=128)
+
+  auto load_tile = [&](int j, ushort8* kr, ushort8* vr) {
     const int key_base = j * KTILE;
-    // ---- Cooperative staging: K (swizzled) and V^T. 512 vec8 elements.
-    {
-      const int nvec = KTILE * HEAD_DIM / 8;
-      for (int i = tid; i < nvec; i += 256) {
-        const int key = i / (HEAD_DIM / 8);
-        const int col8 = (i % (HEAD_DIM / 8)) * 8;
-        const int kg = key_base + key;
-        ushort8 kv{}, vv{};
-        if (kg < kmax) {
-          const int64_t src =
-              (int64_t)(seq_start + kg) * kv_stride + (int64_t)kvh * HEAD_DIM + col8;
-          kv = *reinterpret_cast<const ushort8*>(k + src);
-          vv = *reinterpret_cast<const ushort8*>(v + src);
-        }
-        // K: swizzle byte offset within the 2*HEAD_DIM-byte row.
-        const int row_byte = col8 * 2;
-        const int swz = row_byte ^ ((key & 7) << 4);
-        *reinterpret_cast<ushort8*>(
-            reinterpret_cast<char*>(&k_lds[key][0]) + swz) = kv;
-        // V^T: scalar scatter (optimized via tr_b16 in a later revision).
 #pragma unroll
-        for (int e = 0; e < 8; ++e) vt_lds[col8 + e][key] = vv[e];
+    for (int vv8 = 0; vv8 < VPT; ++vv8) {
+      const int i = tid + 256 * vv8;
+      const int key = i / (HEAD_DIM / 8);
+      const int col8 = (i % (HEAD_DIM / 8)) * 8;
+      const int kg = key_base + key;
+      ushort8 kv{}, vv{};
+      if (kg < kmax) {
+        const int64_t src =
+            (int64_t)(seq_start + kg) * kv_stride + (int64_t)kvh * HEAD_DIM + col8;
+        kv = *reinterpret_cast<const ushort8*>(k + src);
+        vv = *reinterpret_cast<const ushort8*>(v + src);
       }
+      kr[vv8] = kv;
+      vr[vv8] = vv;
     }
-    __syncthreads();
+  };
+
+  auto store_tile = [&](int buf, const ushort8* kr, const ushort8* vr) {
+#pragma unroll
+    for (int vv8 = 0; vv8 < VPT; ++vv8) {
+      const int i = tid + 256 * vv8;
+      const int key = i / (HEAD_DIM / 8);
+      const int col8 = (i % (HEAD_DIM / 8)) * 8;
+      // K: swizzle byte offset within the 2*HEAD_DIM-byte row.
+      const int row_byte = col8 * 2;
+      const int swz = row_byte ^ ((key & 7) << 4);
+      *reinterpret_cast<ushort8*>(
+          reinterpret_cast<char*>(&k_lds[buf][key][0]) + swz) = kr[vv8];
+      // V^T: scalar scatter (optimized via tr_b16 in a later revision).
+#pragma unroll
+      for (int e = 0; e < 8; ++e) vt_lds[buf][col8 + e][key] = vr[vv8][e];
+    }
+  };
+
+  {
+    ushort8 kr0[VPT], vr0[VPT];
+    load_tile(0, kr0, vr0);
+    store_tile(0, kr0, vr0);
+  }
+  __syncthreads();
+
+  int buf = 0;
+  for (int j = 0; j < ntiles; ++j, buf ^= 1) {
+    const int key_base = j * KTILE;
+    // ---- prefetch tile j+1 into registers (global loads overlap MFMAs).
+    ushort8 krn[VPT], vrn[VPT];
+    if (j + 1 < ntiles) load_tile(j + 1, krn, vrn);
 
     // ---- QK^T: two 16-key subtiles; A = K from LDS, B = Q registers.
     f32x4 sc[2];
@@ -140,7 +172,7 @@ __global__ __launch_bounds__(256) void attn_prefill_kernel(
         const int col_byte = (st * 32 + frag_k(la, 0)) * 2;
         const int swz = col_byte ^ ((key & 7) << 4);
         ushort8 u = *reinterpret_cast<const ushort8*>(
-            reinterpret_cast<const char*>(&k_lds[key][0]) + swz);
+            reinterpret_cast<const char*>(&k_lds[buf][key][0]) + swz);
         sc[sub] = mfma16x16x32(*reinterpret_cast<bf16x8*>(&u), qfrag[st], sc[sub]);
       }
     }
@@ -204,19 +236,23 @@ __global__ __launch_bounds__(256) void attn_prefill_kernel(
       for (int r = 0; r < 4; ++r) pk[r] = float_to_bf16_bits(p[sub * 4 + r]);
       *reinterpret_cast<ushort4v*>(&p_lds[wave][lq][sub * 16 + 4 * la]) = pk;
     }
-    __syncthreads();  // P visible across lanes; also fences K/VT reuse below
+    // P is a per-wave buffer: an LDS-counter wait orders the read below
+    // behind the writes without a workgroup barrier.
+    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
 
     // PV: one MFMA per 16-dim chunk contracts the full 32-key tile.
     ushort8 pa =
         *reinterpret_cast<const ushort8*>(&p_lds[wave][lq][frag_k(la, 0)]);
 #pragma unroll
     for (int c = 0; c < CHUNKS; ++c) {
-      ushort8 vb =
-          *reinterpret_cast<const ushort8*>(&vt_lds[c * 16 + lq][frag_k(la, 0)]);
+      ushort8 vb = *reinterpret_cast<const ushort8*>(
+          &vt_lds[buf][c * 16 + lq][frag_k(la, 0)]);
       oacc[c] = mfma16x16x32(*reinterpret_cast<bf16x8*>(&pa),
                              *reinterpret_cast<bf16x8*>(&vb), oacc[c]);
     }
-    __syncthreads();  // all reads of K/VT/P done before next tile's staging
+    // ---- scatter the prefetched tile into the idle buffer and flip.
+    if (j + 1 < ntiles) store_tile(buf ^ 1, krn, vrn);
+    __syncthreads();
   }
 
   // ---- Epilogue: normalize and store. oacc[c][r] is q-row 4*la+r, dim
