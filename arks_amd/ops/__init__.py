@@ -140,12 +140,14 @@ def attention_extend_paged(
 
 def decode_num_partitions(num_seqs: int, num_kv_heads: int, max_blocks: int) -> int:
     """Flash-decode split factor: fill the 256 CUs (target ~2 workgroups/CU)
-    when batch x kv_heads alone cannot, bounded by pages available."""
+    when batch x kv_heads alone cannot. Each partition keeps >= 8 pages so
+    all 4 waves of a workgroup get a 2-page chunk and the K-prefetch
+    software pipeline has depth (bench_decode2: starved waves at <8
+    pages/partition cap short-seq shapes)."""
     base = num_seqs * num_kv_heads
     target = 512
     nparts = max(1, -(-target // base))
-    # at least ~2 pages (32 tokens) per partition to stay efficient
-    nparts = min(nparts, max(1, max_blocks // 2), 64)
+    nparts = min(nparts, max(1, max_blocks // 8), 64)
     return nparts
 
 
