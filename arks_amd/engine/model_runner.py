@@ -171,11 +171,13 @@ class ModelRunner:
             q_lens: list[int] = []
             kv_lens: list[int] = []
             logits_idx: list[int] = []
-            for seq in sb.seqs:
+            for i, seq in enumerate(sb.seqs):
                 toks = seq.all_token_ids
-                n = len(toks)
                 c = seq.num_cached_tokens
-                input_ids.extend(toks[c:])
+                # chunked prefill: run exactly the scheduled token count
+                # (== everything remaining unless the budget split it)
+                n = c + sb.num_new_tokens[i]
+                input_ids.extend(toks[c:n])
                 positions.extend(range(c, n))
                 for pos in range(c, n):
                     b = seq.block_table[pos // bs]

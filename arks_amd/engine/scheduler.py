@@ -70,6 +70,7 @@ class Scheduler:
         for i, s in enumerate(self.waiting):
             if s.request_id == request_id:
                 s.status = SeqStatus.ABORTED
+                self._release(s)  # chunked prefills hold pages while WAITING
                 del self.waiting[i]
                 return True
         for s in self.running:
@@ -102,40 +103,58 @@ class Scheduler:
         seqs: list[Sequence] = []
         ntoks: list[int] = []
         budget = self.max_num_batched_tokens
-        while self.waiting:
+        while self.waiting and budget > 0:
             seq = self.waiting[0]
             # num_tokens (not num_prompt_tokens): a preempted sequence is
             # recomputed over prompt + already-generated tokens.
             n = seq.num_tokens
             toks = seq.all_token_ids
-            # Prefix cache: reuse cached full blocks of the prompt (refs are
-            # taken here and must be released on every non-admit path). Cap
-            # at n-1 so at least one token runs to produce logits.
-            reused, ncached = self.allocator.match_prefix(toks, n - 1)
-            n_new = n - ncached
-            if seqs and n_new > budget:
-                self.allocator.free(reused)
-                break
             if len(self.running) + len(seqs) >= self.max_num_seqs:
+                break
+            if not seq.block_table:
+                # Fresh admission: reuse cached full blocks of the prompt
+                # (refs taken here must be released on every non-admit
+                # path). Cap at n-1 so at least one token produces logits.
+                reused, ncached = self.allocator.match_prefix(toks, n - 1)
+                seq.num_cached_tokens = ncached
+            else:
+                # Continuation of a chunked prefill: pages for the covered
+                # range are already held.
+                reused, ncached = [], seq.num_cached_tokens
+            n_new = n - ncached
+            # Chunked prefill: run at most `budget` new tokens this step;
+            # the remainder stays WAITING (the extend kernel attends the
+            # chunk over the full paged prefix).
+            n_run = min(n_new, budget)
+            partial = n_run < n_new
+            covered = ncached + n_run
+            need = (
+                BlockAllocator.blocks_needed(covered, self.allocator.block_size)
+                - len(reused) - len(seq.block_table)
+            )
+            if not self.allocator.can_allocate(max(need, 0)):
                 self.allocator.free(reused)
                 break
-            need = BlockAllocator.blocks_needed(n, self.allocator.block_size) - len(reused)
-            if not self.allocator.can_allocate(need):
-                self.allocator.free(reused)
-                break
-            self.waiting.popleft()
-            seq.block_table = reused + self.allocator.allocate(need)
-            self.allocator.register_prefix(toks, seq.block_table, len(reused))
-            seq.num_cached_tokens = ncached
-            seq.status = SeqStatus.RUNNING
+            if reused:
+                seq.block_table = reused + seq.block_table
+            if need > 0:
+                seq.block_table.extend(self.allocator.allocate(need))
+            self.allocator.register_prefix(
+                toks[:covered], seq.block_table, len(reused)
+            )
+            if not partial:
+                self.waiting.popleft()
+                seq.status = SeqStatus.RUNNING
             seqs.append(seq)
-            ntoks.append(n_new)
-            budget -= n_new
-            if budget <= 0:
-                break
+            ntoks.append(n_run)
+            budget -= n_run
+            if partial:
+                break  # budget exhausted on this seq
         if not seqs:
             return None
-        self.running.extend(seqs)
+        # partial (chunked) seqs stay in `waiting`; only completed prefills
+        # enter the decode set
+        self.running.extend(s for s in seqs if s.status is SeqStatus.RUNNING)
         return ScheduledBatch(seqs=seqs, is_prefill=True, num_new_tokens=ntoks)
 
     def _schedule_decode(self) -> ScheduledBatch | None:

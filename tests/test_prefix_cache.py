@@ -170,3 +170,45 @@ def test_extend_ref_cached_prefix_matches_full_prefill_suffix():
         D ** -0.5,
     )
     assert torch.allclose(full[cached:].float(), ext.float(), atol=2e-2, rtol=2e-2)
+
+
+def test_chunked_prefill_matches_unchunked():
+    """A prompt longer than max_num_batched_tokens prefills over several
+    steps (extend path) and must produce identical greedy output."""
+    prompt = [(7 * i + 3) % 250 for i in range(100)]
+    sp = SamplingParams(max_tokens=6, ignore_eos=True)
+
+    def run(budget, cache=True):
+        eng = LLMEngine(EngineConfig(
+            preset="tiny", device="cpu", kv_cache_blocks=128,
+            max_model_len=256, max_num_batched_tokens=budget,
+            enable_prefix_caching=cache, seed=13,
+        ))
+        out = eng.generate([prompt], sp)
+        return out[0], eng
+
+    full, _ = run(8192)
+    chunked, eng = run(32)
+    assert chunked == full
+    # chunked + prefix-cache hit on a repeat
+    again = eng.generate([prompt], sp)[0]
+    assert again == full
+    hits, _ = eng.prefix_cache_stats
+    assert hits > 0
+    # chunked with caching off
+    nocache, _ = run(32, cache=False)
+    assert nocache == full
+
+
+def test_chunked_prefill_interleaves_multiple_seqs():
+    prompts = [[1, 2, 3] * 20, [9, 8] * 25, [4] * 7]
+    sp = SamplingParams(max_tokens=5, ignore_eos=True)
+
+    def run(budget):
+        eng = LLMEngine(EngineConfig(
+            preset="tiny", device="cpu", kv_cache_blocks=128,
+            max_model_len=256, max_num_batched_tokens=budget, seed=5,
+        ))
+        return eng.generate(prompts, sp)
+
+    assert run(24) == run(8192)

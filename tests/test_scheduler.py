@@ -27,11 +27,17 @@ def test_prefill_batches_waiting():
     s.add(Sequence([1] * 40))
     s.add(Sequence([1] * 40))
     b = s.schedule()
-    assert b.is_prefill and len(b.seqs) == 2  # 3rd would blow the budget
-    assert all(seq.status is SeqStatus.RUNNING for seq in b.seqs)
-    assert all(len(seq.block_table) == 3 for seq in b.seqs)  # ceil(40/16)
+    # chunked prefill: the 3rd seq rides along with the remaining 20-token
+    # budget and stays WAITING
+    assert b.is_prefill and len(b.seqs) == 3
+    assert b.num_new_tokens == [40, 40, 20]
+    assert [seq.status for seq in b.seqs] == [
+        SeqStatus.RUNNING, SeqStatus.RUNNING, SeqStatus.WAITING]
+    assert s.num_running == 2 and s.num_waiting == 1
+    b.seqs[2].num_cached_tokens = 20  # (the runner does this after the step)
     b2 = s.schedule()
-    assert b2.is_prefill and len(b2.seqs) == 1
+    assert b2.is_prefill and len(b2.seqs) == 1 and b2.num_new_tokens == [20]
+    assert s.num_running == 3 and s.num_waiting == 0
 
 
 def test_decode_after_prefill_grows_blocks():
