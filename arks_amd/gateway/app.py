@@ -92,11 +92,13 @@ class BackendResolver:
         return self.url_for_service(namespace, refs[-1]["name"])
 
 
-def _err(status: int, message: str) -> JSONResponse:
-    # same JSON error shape as reference util.go:40-77
+def _err(status: int, message: str, headers: dict | None = None) -> JSONResponse:
+    # same JSON error shape as reference util.go:40-77; limit/quota errors
+    # carry x-error-* headers naming the tripped rule (reference types.go)
     return JSONResponse(
         status_code=status,
         content={"error": {"message": message, "code": status}},
+        headers=headers,
     )
 
 
@@ -156,12 +158,16 @@ def create_gateway_app(
         if not ok:
             metrics.rate_limit_hits.labels(**labels, rule=rule).inc()
             metrics.requests_total.labels(**labels, status="429").inc()
-            return _err(429, f"rate limit exceeded: {rule}")
+            return _err(429, f"rate limit exceeded: {rule}",
+                        headers={"x-error-type": "rate-limit",
+                                 "x-error-rule": str(rule)})
         qdesc = provider.get_quota_descriptors(qos)
         ok, qtype = quota_service.check(qdesc)
         if not ok:
             metrics.requests_total.labels(**labels, status="429").inc()
-            return _err(429, f"quota exceeded: {qtype}")
+            return _err(429, f"quota exceeded: {qtype}",
+                        headers={"x-error-type": "quota",
+                                 "x-error-rule": str(qtype)})
         # incr request-type counters (rpm/rpd)
         limiter.do_limit(
             [d for d in descriptors if RULES[d.rule].type == TYPE_REQUEST], 1
