@@ -46,6 +46,8 @@ def main():
     ap = argparse.ArgumentParser()
     ap.add_argument("--model", default="qwen2.5-7b")
     ap.add_argument("--tp", type=int, default=1)
+    ap.add_argument("--fp8", action="store_true",
+                    help="also tune rowwise-scaled fp8 _scaled_mm shapes")
     ap.add_argument("--out", default="arks_amd/data/tunableop_gfx950.csv")
     args = ap.parse_args()
 
@@ -73,6 +75,17 @@ def main():
             F.linear(x, w, b)  # tuning happens on first call per shape
             torch.cuda.synchronize()
             print(f"tuned {name} M={m}", flush=True)
+        if args.fp8:
+            w8 = (w.float() * 10).clamp(-448, 448).to(torch.float8_e4m3fn)
+            sb = torch.full((1, out_f), 0.1, device=dev)
+            for m in ms:
+                x8 = torch.randn(m, in_f, device=dev).clamp(-4, 4).to(
+                    torch.float8_e4m3fn)
+                sa = torch.full((m, 1), 0.1, device=dev)
+                torch._scaled_mm(x8, w8.t(), scale_a=sa, scale_b=sb,
+                                 bias=b, out_dtype=torch.bfloat16)
+                torch.cuda.synchronize()
+                print(f"tuned fp8 {name} M={m}", flush=True)
         del w, b
         torch.cuda.empty_cache()
 
