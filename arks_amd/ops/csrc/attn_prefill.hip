@@ -304,6 +304,44 @@ __global__ void mfma_probe_kernel(float* __restrict__ d,
 
 using namespace arks;
 
+// 32x32x16 probe: D[32,32] = A[32,16] @ B[16,32] using the layout
+// hypothesis A[row=l%32][k=8*(l/32)+j], B[k=8*(l/32)+j][col=l%32],
+// C[col=l%32][rows 4*(l/32) + 8*g + r] per f32x4 acc g. Verified on
+// hardware by tests/test_ops_gpu.py::test_mfma_probe_32x32 before any
+// kernel trusts it (guide §3).
+__global__ void mfma_probe32_kernel(float* __restrict__ d,
+                                    const bf16* __restrict__ a,
+                                    const bf16* __restrict__ b) {
+  typedef __attribute__((ext_vector_type(16))) float f32x16;
+  const int lane = threadIdx.x % WAVE_SIZE;
+  const int col = lane % 32;
+  const int half = lane / 32;
+  bf16x8 af, bf_;
+#pragma unroll
+  for (int j = 0; j < 8; ++j) {
+    ushort au = a[col * 16 + 8 * half + j];   // A[row=col][k]
+    ushort bu = b[(8 * half + j) * 32 + col]; // B[k][col]
+    af[j] = *reinterpret_cast<__bf16*>(&au);
+    bf_[j] = *reinterpret_cast<__bf16*>(&bu);
+  }
+  f32x16 c = {};
+  c = __builtin_amdgcn_mfma_f32_32x32x16_bf16(af, bf_, c, 0, 0, 0);
+#pragma unroll
+  for (int g = 0; g < 4; ++g) {
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      const int row = 8 * g + 4 * half + r;
+      d[row * 32 + col] = c[g * 4 + r];
+    }
+  }
+}
+
+extern "C" void arks_mfma_probe32(void* d, const void* a, const void* b,
+                                  hipStream_t stream) {
+  hipLaunchKernelGGL(mfma_probe32_kernel, dim3(1), dim3(64), 0, stream,
+                     (float*)d, (const bf16*)a, (const bf16*)b);
+}
+
 extern "C" void arks_mfma_probe(void* d, const void* a, const void* b,
                                 hipStream_t stream) {
   hipLaunchKernelGGL(mfma_probe_kernel, dim3(1), dim3(64), 0, stream, (float*)d,

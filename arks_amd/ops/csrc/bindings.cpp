@@ -56,6 +56,7 @@ void arks_gumbel_sample(void* out, const void* logits, const void* temperatures,
                         const void* uniform, int rows, int vocab,
                         hipStream_t stream);
 void arks_mfma_probe(void* d, const void* a, const void* b, hipStream_t stream);
+void arks_mfma_probe32(void* d, const void* a, const void* b, hipStream_t stream);
 }
 
 namespace {
@@ -300,6 +301,15 @@ void gumbel_sample(torch::Tensor out, torch::Tensor logits,
 
 // D[16,16] = A[16,32] @ B[32,16], all bf16 in / f32 out. Verifies the MFMA
 // fragment layout assumption on hardware.
+void mfma_probe32(torch::Tensor d, torch::Tensor a, torch::Tensor b) {
+  check_bf16_contig(a, "a");
+  check_bf16_contig(b, "b");
+  TORCH_CHECK(d.scalar_type() == torch::kFloat32 && d.is_contiguous());
+  TORCH_CHECK(a.size(0) == 32 && a.size(1) == 16);
+  TORCH_CHECK(b.size(0) == 16 && b.size(1) == 32);
+  arks_mfma_probe32(d.data_ptr(), a.data_ptr(), b.data_ptr(), current_stream());
+}
+
 void mfma_probe(torch::Tensor d, torch::Tensor a, torch::Tensor b) {
   check_bf16_contig(a, "a");
   check_bf16_contig(b, "b");
@@ -325,4 +335,5 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("greedy_sample", &greedy_sample);
   m.def("gumbel_sample", &gumbel_sample);
   m.def("mfma_probe", &mfma_probe);
+  m.def("mfma_probe32", &mfma_probe32);
 }

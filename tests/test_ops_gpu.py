@@ -383,3 +383,16 @@ def test_skinny_gemm_strided_input():
     torch.testing.assert_close(
         out, torch.nn.functional.linear(a, w).float(), atol=5e-2, rtol=5e-2
     )
+
+
+def test_mfma_probe_32x32():
+    """Verify the 32x32x16 bf16 MFMA fragment layout hypothesis on HW."""
+    assert_native()
+    torch.manual_seed(3)
+    a = torch.randn(32, 16, dtype=torch.bfloat16, device=DEV)
+    b = torch.randn(16, 32, dtype=torch.bfloat16, device=DEV)
+    d = torch.zeros(32, 32, dtype=torch.float32, device=DEV)
+    from arks_amd.ops import _native
+    _native().mfma_probe32(d, a, b)
+    ref_out = a.float() @ b.float()
+    torch.testing.assert_close(d.cpu(), ref_out.cpu(), atol=2e-2, rtol=2e-2)
