@@ -336,6 +336,35 @@ __global__ void mfma_probe32_kernel(float* __restrict__ d,
   }
 }
 
+
+// tr16 probe: LDS[i] = i (u16), each lane does ds_read_b64_tr_b16 at
+// address lane*8 + row_sel*stride; dumps its 4 values so the host can
+// derive the exact lane->element mapping empirically before any kernel
+// uses the transpose read (T10 pitfall: misaligned tr reads return
+// wrong data silently).
+__global__ void tr16_probe_kernel(uint16_t* __restrict__ out,
+                                  const int stride_bytes) {
+  typedef __attribute__((ext_vector_type(4))) __bf16 bf16x4v;
+  __shared__ uint16_t lds[4096];
+  const int lane = threadIdx.x;
+  for (int i = lane; i < 4096; i += 64) lds[i] = (uint16_t)i;
+  __syncthreads();
+  char* base = reinterpret_cast<char*>(&lds[0]);
+  auto addr = (__attribute__((address_space(3))) bf16x4v*)(
+      base + (int64_t)lane * stride_bytes);
+  bf16x4v v = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(addr);
+  union { bf16x4v bv; uint16_t u[4]; } cv;
+  cv.bv = v;
+#pragma unroll
+  for (int j = 0; j < 4; ++j) out[lane * 4 + j] = cv.u[j];
+}
+
+extern "C" void arks_tr16_probe(void* out, int stride_bytes,
+                                hipStream_t stream) {
+  hipLaunchKernelGGL(tr16_probe_kernel, dim3(1), dim3(64), 0, stream,
+                     (uint16_t*)out, stride_bytes);
+}
+
 extern "C" void arks_mfma_probe32(void* d, const void* a, const void* b,
                                   hipStream_t stream) {
   hipLaunchKernelGGL(mfma_probe32_kernel, dim3(1), dim3(64), 0, stream,

@@ -57,6 +57,7 @@ void arks_gumbel_sample(void* out, const void* logits, const void* temperatures,
                         hipStream_t stream);
 void arks_mfma_probe(void* d, const void* a, const void* b, hipStream_t stream);
 void arks_mfma_probe32(void* d, const void* a, const void* b, hipStream_t stream);
+void arks_tr16_probe(void* out, int stride_bytes, hipStream_t stream);
 }
 
 namespace {
@@ -301,6 +302,11 @@ void gumbel_sample(torch::Tensor out, torch::Tensor logits,
 
 // D[16,16] = A[16,32] @ B[32,16], all bf16 in / f32 out. Verifies the MFMA
 // fragment layout assumption on hardware.
+void tr16_probe(torch::Tensor out, int64_t stride_bytes) {
+  TORCH_CHECK(out.scalar_type() == torch::kInt16 && out.numel() >= 256);
+  arks_tr16_probe(out.data_ptr(), (int)stride_bytes, current_stream());
+}
+
 void mfma_probe32(torch::Tensor d, torch::Tensor a, torch::Tensor b) {
   check_bf16_contig(a, "a");
   check_bf16_contig(b, "b");
@@ -336,4 +342,5 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("gumbel_sample", &gumbel_sample);
   m.def("mfma_probe", &mfma_probe);
   m.def("mfma_probe32", &mfma_probe32);
+  m.def("tr16_probe", &tr16_probe);
 }
