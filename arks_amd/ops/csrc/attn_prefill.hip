@@ -23,6 +23,7 @@ namespace arks {
 
 // gfx950 MFMA fragment types.
 typedef __attribute__((ext_vector_type(8))) __bf16 bf16x8;
+typedef __attribute__((ext_vector_type(4))) __bf16 bf16x4v;
 typedef __attribute__((ext_vector_type(4))) float f32x4;
 
 __device__ __forceinline__ f32x4 mfma16x16x32(bf16x8 a, bf16x8 b, f32x4 c) {
@@ -74,7 +75,14 @@ __global__ __launch_bounds__(256) void attn_prefill_kernel(
   // buffer, and one __syncthreads flips the buffers (v1 took two barriers
   // per tile and stalled on the staging round-trip, ~118 TF effective).
   __shared__ __attribute__((aligned(16))) bf16 k_lds[2][KTILE][HEAD_DIM];
-  __shared__ __attribute__((aligned(16))) bf16 vt_lds[2][HEAD_DIM][VT_PAD];
+  // V tile as a tr16 image: [d-chunk cb][key-block kb][4 key x 16 col]
+  // blocks of 64 bf16 padded to 96 so the two 16-lane regions of one
+  // 32-lane ds_read_b64_tr_b16 group land on disjoint bank halves
+  // (stride 192 B = 48 dwords; probe: gpurun_out/probe_layouts.log).
+  // Each lane then receives the 4 keys of its column — B-fragment shape —
+  // and V staging becomes two b128 stores instead of 16 scalar b16 writes
+  // (the hot block of the v2 disassembly).
+  __shared__ __attribute__((aligned(16))) bf16 vt_img[2][HEAD_DIM / 16][8][96];
   __shared__ __attribute__((aligned(16))) bf16 p_lds[NUM_WAVES][QTILE_WAVE][VT_PAD];
 
   // --- Q fragments: registers, loaded once. Wave w covers rows q0+16w..+15.
@@ -140,9 +148,10 @@ __global__ __launch_bounds__(256) void attn_prefill_kernel(
       const int swz = row_byte ^ ((key & 7) << 4);
       *reinterpret_cast<ushort8*>(
           reinterpret_cast<char*>(&k_lds[buf][key][0]) + swz) = kr[vv8];
-      // V^T: scalar scatter (optimized via tr_b16 in a later revision).
-#pragma unroll
-      for (int e = 0; e < 8; ++e) vt_lds[buf][col8 + e][key] = vr[vv8][e];
+      // V into the tr16 image: one b128 store per vec8.
+      const int kb = key >> 2, r = key & 3;
+      const int cb = col8 >> 4, i0 = col8 & 15;
+      *reinterpret_cast<ushort8*>(&vt_img[buf][cb][kb][r * 16 + i0]) = vr[vv8];
     }
   };
 
@@ -245,10 +254,21 @@ __global__ __launch_bounds__(256) void attn_prefill_kernel(
         *reinterpret_cast<const ushort8*>(&p_lds[wave][lq][frag_k(la, 0)]);
 #pragma unroll
     for (int c = 0; c < CHUNKS; ++c) {
-      ushort8 vb = *reinterpret_cast<const ushort8*>(
-          &vt_lds[buf][c * 16 + lq][frag_k(la, 0)]);
-      oacc[c] = mfma16x16x32(*reinterpret_cast<bf16x8*>(&pa),
-                             *reinterpret_cast<bf16x8*>(&vb), oacc[c]);
+      // B-fragment via two hardware transpose reads: keys 8*la..+3, +4..7
+      // at column lq of chunk c.
+      bf16x4v v1 = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(
+          (__attribute__((address_space(3))) bf16x4v*)(
+              reinterpret_cast<char*>(&vt_img[buf][c][2 * la][0]) + lq * 8));
+      bf16x4v v2 = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(
+          (__attribute__((address_space(3))) bf16x4v*)(
+              reinterpret_cast<char*>(&vt_img[buf][c][2 * la + 1][0]) + lq * 8));
+      bf16x8 vb;
+#pragma unroll
+      for (int e = 0; e < 4; ++e) {
+        vb[e] = v1[e];
+        vb[e + 4] = v2[e];
+      }
+      oacc[c] = mfma16x16x32(*reinterpret_cast<bf16x8*>(&pa), vb, oacc[c]);
     }
     // ---- scatter the prefetched tile into the idle buffer and flip.
     if (j + 1 < ntiles) store_tile(buf ^ 1, krn, vrn);
