@@ -134,6 +134,30 @@ def create_app(engine: AsyncEngine, served_model_name: str, tokenizer,
             ignore_eos=req.ignore_eos,
         )
 
+
+    async def _collect_choice(rid, token_ids, sp, stops, prefill_addr):
+        """One non-streamed choice: (token_ids, text, finish, n_generated)."""
+        tracker = StopStringTracker(stops)
+        out_ids: list[int] = []
+        text_acc = ""
+        finish = None
+        async for out in engine.generate_stream(rid, token_ids, sp,
+                                                prefill_addr=prefill_addr):
+            out_ids.append(out.new_token_id)
+            if stops:
+                emit, stopped = tracker.feed(tokenizer.decode([out.new_token_id]))
+                text_acc += emit
+                if stopped:
+                    finish = "stop"
+                    engine.abort(rid)
+                    break
+            if out.finished:
+                finish = out.finish_reason
+        if stops and finish != "stop":
+            text_acc += tracker.flush()
+        text = text_acc if stops else tokenizer.decode(out_ids)
+        return out_ids, text, finish or "stop", len(out_ids)
+
     @app.post("/v1/chat/completions")
     async def chat_completions(req: ChatCompletionRequest, raw: Request):
         if req.model != served_model_name:
@@ -143,10 +167,32 @@ def create_app(engine: AsyncEngine, served_model_name: str, tokenizer,
         sp = _sampling(req, req.max_completion_tokens or req.max_tokens)
         rid = f"chatcmpl-{uuid.uuid4().hex}"
         if req.stream:
+            if req.n > 1:
+                return _error(400, "n > 1 is not supported with stream=true")
             return StreamingResponse(
                 _chat_stream(engine, tokenizer, req, rid, token_ids, sp, raw),
                 media_type="text/event-stream",
             )
+        if req.n > 1:
+            paddr = raw.headers.get("x-arks-prefill-addr")
+            stops_n = _stop_list(req.stop)
+            results = await asyncio.gather(*[
+                _collect_choice(f"{rid}-{i}", token_ids, sp, stops_n, paddr)
+                for i in range(req.n)
+            ])
+            total_out = sum(r[3] for r in results)
+            return ChatCompletionResponse(
+                id=rid, model=req.model,
+                choices=[
+                    ChatChoice(index=i,
+                               message=ChatMessage(role="assistant", content=r[1]),
+                               finish_reason=r[2])
+                    for i, r in enumerate(results)
+                ],
+                usage=Usage(prompt_tokens=len(token_ids),
+                            completion_tokens=total_out,
+                            total_tokens=len(token_ids) + total_out),
+            ).model_dump()
         stops = _stop_list(req.stop)
         tracker = StopStringTracker(stops)
         text_ids: list[int] = []
@@ -271,10 +317,30 @@ def create_app(engine: AsyncEngine, served_model_name: str, tokenizer,
         sp = _sampling(req, req.max_tokens)
         rid = f"cmpl-{uuid.uuid4().hex}"
         if req.stream:
+            if req.n > 1:
+                return _error(400, "n > 1 is not supported with stream=true")
             return StreamingResponse(
                 _completion_stream(engine, tokenizer, req, rid, token_ids, sp, raw),
                 media_type="text/event-stream",
             )
+        if req.n > 1:
+            paddr = raw.headers.get("x-arks-prefill-addr")
+            stops_n = _stop_list(req.stop)
+            results = await asyncio.gather(*[
+                _collect_choice(f"{rid}-{i}", token_ids, sp, stops_n, paddr)
+                for i in range(req.n)
+            ])
+            total_out = sum(r[3] for r in results)
+            return CompletionResponse(
+                id=rid, model=req.model,
+                choices=[
+                    CompletionChoice(index=i, text=r[1], finish_reason=r[2])
+                    for i, r in enumerate(results)
+                ],
+                usage=Usage(prompt_tokens=len(token_ids),
+                            completion_tokens=total_out,
+                            total_tokens=len(token_ids) + total_out),
+            ).model_dump()
         stops = _stop_list(req.stop)
         tracker = StopStringTracker(stops)
         out_ids: list[int] = []

@@ -202,3 +202,23 @@ def test_stop_string_tracker_split_across_pieces():
         if stopped:
             break
     assert stopped and out == "hello "
+
+
+def test_completion_n_choices(app):
+    async def fn(client):
+        r = await client.post("/v1/completions", json={
+            "model": "tiny", "prompt": "abc", "max_tokens": 4,
+            "temperature": 0.8, "n": 3, "ignore_eos": True,
+        })
+        body = r.json()
+        assert len(body["choices"]) == 3
+        assert [c["index"] for c in body["choices"]] == [0, 1, 2]
+        assert body["usage"]["completion_tokens"] == 12
+        # n>1 + stream is rejected
+        r2 = await client.post("/v1/completions", json={
+            "model": "tiny", "prompt": "abc", "max_tokens": 2, "n": 2,
+            "stream": True,
+        })
+        assert r2.status_code == 400
+
+    run_with_client(app, fn)
