@@ -29,6 +29,7 @@ constexpr int KV_BLOCK_SIZE = 16;  // tokens per KV page (matches engine)
 // MFMA fragment types (shared convention with attn_prefill.hip; layout
 // verified on hardware by tests/test_ops_gpu.py::test_mfma_probe).
 typedef __attribute__((ext_vector_type(8))) __bf16 bf16x8;
+typedef __attribute__((ext_vector_type(4))) __bf16 dec_bf16x4v;
 typedef __attribute__((ext_vector_type(4))) float f32x4;
 
 __device__ __forceinline__ f32x4 dec_mfma(bf16x8 a, bf16x8 b, f32x4 c) {
@@ -89,7 +90,8 @@ __global__ __launch_bounds__(256) void attn_decode_kernel(
   // Per-wave LDS region: V^T tile + P tile during the main loop, merge
   // buffer afterwards (union; a wave touches only its own region between
   // the loop and the final barrier).
-  constexpr int VT_BYTES = HEAD_DIM * DEC_VT_PAD * 2;
+  // V tr16 image: [cb = D/16][kb = 8][96] (64 data + 32 pad) bf16
+  constexpr int VT_BYTES = (HEAD_DIM / 16) * 8 * 96 * 2;
   constexpr int P_BYTES = 16 * DEC_VT_PAD * 2;
   constexpr int MERGE_BYTES = 16 * (HEAD_DIM + 2) * 4;
   constexpr int WAVE_BYTES0 =
@@ -249,15 +251,16 @@ __global__ __launch_bounds__(256) void attn_decode_kernel(
       for (int r = 0; r < 4; ++r) oacc[c][r] *= row_alpha[r];
     }
 
-    // ---- scatter V^T to LDS (after the previous PV reads, in order).
+    // ---- store V into the tr16 image (b128; after the previous PV reads,
+    // in order within the wave).
 #pragma unroll
     for (int j = 0; j < NV; ++j) {
       const int i = lane + WAVE_SIZE * j;
       const int t = i / (HEAD_DIM / 8);
       const int col8 = (i % (HEAD_DIM / 8)) * 8;
-#pragma unroll
-      for (int e = 0; e < 8; ++e)
-        vt_lds[(col8 + e) * DEC_VT_PAD + t] = vv[j][e];
+      *reinterpret_cast<ushort8*>(
+          &vt_lds[((col8 >> 4) * 8 + (t >> 2)) * 96 + (t & 3) * 16 +
+                  (col8 & 15)]) = vv[j];
     }
 
     // ---- P to LDS (bf16) then PV (contract the full 32-token chunk).
@@ -276,10 +279,21 @@ __global__ __launch_bounds__(256) void attn_decode_kernel(
         &p_lds[lq * DEC_VT_PAD + dec_frag_k(la, 0)]);
 #pragma unroll
     for (int c = 0; c < CHUNKS; ++c) {
-      ushort8 vb = *reinterpret_cast<const ushort8*>(
-          &vt_lds[(c * 16 + lq) * DEC_VT_PAD + dec_frag_k(la, 0)]);
-      oacc[c] = dec_mfma(*reinterpret_cast<bf16x8*>(&pa),
-                         *reinterpret_cast<bf16x8*>(&vb), oacc[c]);
+      dec_bf16x4v v1 = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(
+          (__attribute__((address_space(3))) dec_bf16x4v*)(
+              reinterpret_cast<char*>(&vt_lds[(c * 8 + 2 * la) * 96]) +
+              lq * 8));
+      dec_bf16x4v v2 = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(
+          (__attribute__((address_space(3))) dec_bf16x4v*)(
+              reinterpret_cast<char*>(&vt_lds[(c * 8 + 2 * la + 1) * 96]) +
+              lq * 8));
+      bf16x8 vb;
+#pragma unroll
+      for (int e = 0; e < 4; ++e) {
+        vb[e] = v1[e];
+        vb[e + 4] = v2[e];
+      }
+      oacc[c] = dec_mfma(*reinterpret_cast<bf16x8*>(&pa), vb, oacc[c]);
     }
 #pragma unroll
     for (int sub = 0; sub < 2; ++sub)

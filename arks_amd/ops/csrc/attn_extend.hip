@@ -23,6 +23,7 @@
 namespace arks {
 
 typedef __attribute__((ext_vector_type(8))) __bf16 bf16x8;
+typedef __attribute__((ext_vector_type(4))) __bf16 ext_bf16x4v;
 typedef __attribute__((ext_vector_type(4))) float f32x4;
 
 __device__ __forceinline__ f32x4 ext_mfma16x16x32(bf16x8 a, bf16x8 b, f32x4 c) {
@@ -70,7 +71,10 @@ __global__ __launch_bounds__(256) void attn_extend_kernel(
   const int la = lane / 16;
 
   __shared__ __attribute__((aligned(16))) bf16 k_lds[EXT_KTILE][HEAD_DIM];
-  __shared__ __attribute__((aligned(16))) bf16 vt_lds[HEAD_DIM][EXT_VT_PAD];
+  // V as a tr16 image (see attn_prefill.hip): [cb][kb][4 key x 16 col]
+  // 96-elem blocks; staged with b128 stores, consumed with
+  // ds_read_b64_tr_b16 B-fragments.
+  __shared__ __attribute__((aligned(16))) bf16 vt_img[HEAD_DIM / 16][8][96];
   __shared__ __attribute__((aligned(16)))
       bf16 p_lds[EXT_NUM_WAVES][EXT_QTILE_WAVE][EXT_VT_PAD];
 
@@ -128,8 +132,8 @@ __global__ __launch_bounds__(256) void attn_extend_kernel(
         const int swz = row_byte ^ ((key & 7) << 4);
         *reinterpret_cast<ushort8*>(reinterpret_cast<char*>(&k_lds[key][0]) +
                                     swz) = kv;
-#pragma unroll
-        for (int e = 0; e < 8; ++e) vt_lds[col8 + e][key] = vv[e];
+        *reinterpret_cast<ushort8*>(
+            &vt_img[col8 >> 4][key >> 2][(key & 3) * 16 + (col8 & 15)]) = vv;
       }
     }
     __syncthreads();
@@ -212,10 +216,19 @@ __global__ __launch_bounds__(256) void attn_extend_kernel(
         *reinterpret_cast<const ushort8*>(&p_lds[wave][lq][ext_frag_k(la, 0)]);
 #pragma unroll
     for (int c = 0; c < CHUNKS; ++c) {
-      ushort8 vb = *reinterpret_cast<const ushort8*>(
-          &vt_lds[c * 16 + lq][ext_frag_k(la, 0)]);
-      oacc[c] = ext_mfma16x16x32(*reinterpret_cast<bf16x8*>(&pa),
-                                 *reinterpret_cast<bf16x8*>(&vb), oacc[c]);
+      ext_bf16x4v v1 = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(
+          (__attribute__((address_space(3))) ext_bf16x4v*)(
+              reinterpret_cast<char*>(&vt_img[c][2 * la][0]) + lq * 8));
+      ext_bf16x4v v2 = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(
+          (__attribute__((address_space(3))) ext_bf16x4v*)(
+              reinterpret_cast<char*>(&vt_img[c][2 * la + 1][0]) + lq * 8));
+      bf16x8 vb;
+#pragma unroll
+      for (int e = 0; e < 4; ++e) {
+        vb[e] = v1[e];
+        vb[e + 4] = v2[e];
+      }
+      oacc[c] = ext_mfma16x16x32(*reinterpret_cast<bf16x8*>(&pa), vb, oacc[c]);
     }
     __syncthreads();
   }
