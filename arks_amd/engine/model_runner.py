@@ -347,6 +347,18 @@ class ModelRunner:
             u = torch.rand(
                 logits.shape, dtype=torch.float32, device=logits.device
             )
+            # per-request seeded rows draw from their own generator
+            for i, s in enumerate(seqs):
+                if s.sampling.seed is not None:
+                    gen = getattr(s, "_rng", None)
+                    if gen is None:
+                        gen = torch.Generator(device=logits.device)
+                        gen.manual_seed(s.sampling.seed)
+                        s._rng = gen
+                    u[i] = torch.rand(
+                        logits.shape[1], dtype=torch.float32,
+                        device=logits.device, generator=gen,
+                    )
             ids = ops.sample_tokens(logits.contiguous(), t, u)
         out = ids.tolist()
         self._last_logprobs = self._gather_logprobs(logits, seqs, out)

@@ -19,6 +19,7 @@ class SamplingParams:
     stop_token_ids: tuple[int, ...] = ()
     ignore_eos: bool = False
     logprobs: int | None = None  # None = off; 0 = chosen only; N = top-N too
+    seed: int | None = None  # per-request RNG seed (reproducible sampling)
 
     @property
     def has_penalties(self) -> bool:

@@ -131,6 +131,7 @@ def create_app(engine: AsyncEngine, served_model_name: str, tokenizer,
             frequency_penalty=getattr(req, "frequency_penalty", 0.0),
             repetition_penalty=getattr(req, "repetition_penalty", 1.0),
             logprobs=_logprobs_of(req),
+            seed=req.seed,
             ignore_eos=req.ignore_eos,
         )
 
