@@ -191,10 +191,8 @@ def skinny_gemm(x, weight, bias=None):
     N = weight.shape[0]
     out = torch.empty((M, N), dtype=x.dtype, device=x.device)
     ntiles = N // 64
-    if ntiles >= 1024:
+    if ntiles >= 384:
         nsplits = 1
-    elif ntiles >= 384:
-        nsplits = 2  # ~4 workgroups/CU for the prefetch ring to hide HBM
     else:
         nsplits = min(-(-512 // ntiles), -(-K // 256))
     k_ceil = -(-K // nsplits)

@@ -84,10 +84,8 @@ __global__ __launch_bounds__(256) void skinny_gemm_kernel(
     }
   };
 
-  // W fragments: depth-3 register ring so ~2 chunks (≈550 MFMA cycles) of
-  // compute cover each chunk's ~900-cycle HBM load latency (the depth-1
-  // version left the kernel ~65% wave-parked — gpurun_out/pmc).
-  ushort8 wring[3][SK_KSTEPS];
+  // W fragments, register-double-buffered.
+  ushort8 wreg[SK_KSTEPS];
   auto load_w = [&](int kc, ushort8* dst) {
     const int kw = min(SK_KC, ke - kc);
 #pragma unroll
@@ -98,23 +96,23 @@ __global__ __launch_bounds__(256) void skinny_gemm_kernel(
                    : ushort8{};
   };
 
-  load_w(kb, wring[0]);
-  if (kb + SK_KC < ke) load_w(kb + SK_KC, wring[1]);
+  load_w(kb, wreg);
   stage_a(kb, 0);
   __syncthreads();
 
   int buf = 0;
-  int ring = 0;
-  for (int kc = kb; kc < ke; kc += SK_KC, buf ^= 1, ring = (ring + 1) % 3) {
+  for (int kc = kb; kc < ke; kc += SK_KC, buf ^= 1) {
     const int kw = min(SK_KC, ke - kc);  // multiple of 32
     const int kn = kc + SK_KC;
-    const int k2 = kc + 2 * SK_KC;
-    if (k2 < ke) load_w(k2, wring[(ring + 2) % 3]);
-    if (kn < ke) stage_a(kn, buf ^ 1);
+    ushort8 wnext[SK_KSTEPS];
+    if (kn < ke) {
+      load_w(kn, wnext);   // issues during this chunk's MFMAs
+      stage_a(kn, buf ^ 1);
+    }
 #pragma unroll
     for (int s = 0; s < SK_KSTEPS; ++s) {
       if (s * 32 >= kw) break;
-      bf16x8 wfrag = *reinterpret_cast<bf16x8*>(&wring[ring][s]);
+      bf16x8 wfrag = *reinterpret_cast<bf16x8*>(&wreg[s]);
 #pragma unroll
       for (int t = 0; t < MT; ++t) {
         ushort8 af = *reinterpret_cast<const ushort8*>(
@@ -122,6 +120,8 @@ __global__ __launch_bounds__(256) void skinny_gemm_kernel(
         acc[t] = sk_mfma(wfrag, *reinterpret_cast<bf16x8*>(&af), acc[t]);
       }
     }
+#pragma unroll
+    for (int s = 0; s < SK_KSTEPS; ++s) wreg[s] = wnext[s];
     __syncthreads();  // next buffer's staging writes have had the whole
                       // compute phase to land; also fences buf reuse
   }
