@@ -88,3 +88,36 @@ def test_prefix_cache_outputs_match_gpu():
     assert all(len(o) == 8 for o in second)
     agree = sum(a == b for o1, o2 in zip(second, base) for a, b in zip(o1, o2))
     assert agree >= 16, (second, base)
+
+
+def test_disagg_kv_transfer_gpu():
+    """Extract KV pages on engine A, inject into engine B, decode matches a
+    monolithic engine (greedy) — the PD disaggregation hot path on HW."""
+    from arks_amd.engine.sequence import SamplingParams as SP
+
+    def cfg():
+        return EngineConfig(
+            preset="tiny-gpu", device="cuda", kv_cache_blocks=256,
+            max_model_len=512, seed=21,
+        )
+
+    prompts = [[7, 3, 9, 1] * 9, [2, 8] * 5]
+    sp = SP(max_tokens=6, ignore_eos=True)
+    mono = LLMEngine(cfg()).generate(prompts, sp)
+
+    a, b = LLMEngine(cfg()), LLMEngine(cfg())
+    outs = []
+    for i, prompt in enumerate(prompts):
+        rid = f"r{i}"
+        seq = a.add_request(prompt, SP(max_tokens=1, ignore_eos=True),
+                            request_id=rid, hold_pages=True)
+        while not seq.is_finished:
+            a.step()
+        first = seq.output_token_ids[0]
+        _, kv = a.extract_prefilled(rid)
+        b.add_prefilled(prompt, first, kv, sp, request_id=rid)
+        outs.append([first])
+    while b.has_work():
+        for o in b.step():
+            outs[int(o.request_id[1:])].append(o.new_token_id)
+    assert outs == mono
