@@ -123,6 +123,13 @@ class LLMEngine:
             seq.append_token(tok)
             self.total_output_tokens += 1
             finished = seq.check_finished(eos)
+            if not finished and seq.num_tokens >= self.cfg.max_model_len:
+                # context-window cap: never grow past max_model_len (the
+                # graph block-table buffers are sized to it)
+                seq.finish_reason = "length"
+                seq.status = SeqStatus.FINISHED
+                seq.finish_time = time.time()
+                finished = True
             lp, top = logprobs.get(i, (None, None))
             outputs.append(
                 StepOutput(

@@ -138,3 +138,16 @@ def test_seeded_sampling_reproducible():
 
     assert run(42) == run(42)
     assert run(42) != run(43) or run(7) != run(8)  # different seeds diverge
+
+
+def test_decode_capped_at_max_model_len():
+    from arks_amd.config import EngineConfig
+    from arks_amd.engine import LLMEngine, SamplingParams
+
+    eng = LLMEngine(EngineConfig(
+        preset="tiny", device="cpu", kv_cache_blocks=64, max_model_len=24,
+        seed=1,
+    ))
+    out = eng.generate([[1] * 20], SamplingParams(max_tokens=50, ignore_eos=True))
+    assert len(out[0]) == 4  # 20 prompt + 4 = 24 = max_model_len
+    assert not eng.has_work()
