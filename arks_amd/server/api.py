@@ -165,6 +165,9 @@ def create_app(engine: AsyncEngine, served_model_name: str, tokenizer,
             return _error(404, f"model {req.model!r} not found")
         prompt = tokenizer.apply_chat_template([m.model_dump() for m in req.messages])
         token_ids = tokenizer.encode(prompt)
+        if len(token_ids) > engine.cfg.max_model_len:
+            return _error(400, f"prompt is {len(token_ids)} tokens; "
+                               f"max_model_len is {engine.cfg.max_model_len}")
         sp = _sampling(req, req.max_completion_tokens or req.max_tokens)
         rid = f"chatcmpl-{uuid.uuid4().hex}"
         if req.stream:
@@ -315,6 +318,9 @@ def create_app(engine: AsyncEngine, served_model_name: str, tokenizer,
             token_ids = list(p[0])
         else:
             return _error(400, "invalid prompt")
+        if len(token_ids) > engine.cfg.max_model_len:
+            return _error(400, f"prompt is {len(token_ids)} tokens; "
+                               f"max_model_len is {engine.cfg.max_model_len}")
         sp = _sampling(req, req.max_tokens)
         rid = f"cmpl-{uuid.uuid4().hex}"
         if req.stream:

@@ -29,14 +29,22 @@ class RequestAdd:
     hold_pages: bool = False  # disaggregated prefill: keep pages for extract
 
 
-def apply_msg(engine: LLMEngine, msg: dict) -> None:
+def apply_msg(engine: LLMEngine, msg: dict) -> list[str]:
+    """Apply adds/aborts; returns request ids rejected at admission (e.g.
+    prompt too long) so the driver can fail their streams instead of
+    letting an exception kill the engine loop."""
+    rejected: list[str] = []
     for add in msg.get("adds", ()):
-        engine.add_request(
-            add.token_ids, SamplingParams(**add.sampling), add.request_id,
-            hold_pages=add.hold_pages,
-        )
+        try:
+            engine.add_request(
+                add.token_ids, SamplingParams(**add.sampling), add.request_id,
+                hold_pages=add.hold_pages,
+            )
+        except ValueError:
+            rejected.append(add.request_id)
     for rid in msg.get("aborts", ()):
         engine.abort_request(rid)
+    return rejected
 
 
 def worker_loop(cfg: EngineConfig) -> None:
@@ -177,11 +185,11 @@ class AsyncEngine:
             self._wakeup.set()
 
     # ---- engine loop ----
-    def _sync_iteration(self, msg: dict) -> list[StepOutput]:
+    def _sync_iteration(self, msg: dict):
         if get_tp_world_size() > 1:
             tp_broadcast_object(msg, src=0)
-        apply_msg(self.engine, msg)
-        return self.engine.step()
+        rejected = apply_msg(self.engine, msg)
+        return self.engine.step(), rejected
 
     async def _run(self):
         loop = asyncio.get_running_loop()
@@ -192,9 +200,13 @@ class AsyncEngine:
                 continue
             msg = {"adds": self.pending_adds, "aborts": self.pending_aborts}
             self.pending_adds, self.pending_aborts = [], []
-            outputs = await loop.run_in_executor(
+            outputs, rejected = await loop.run_in_executor(
                 self._executor, self._sync_iteration, msg
             )
+            for rid in rejected:
+                st = self.streams.pop(rid, None)
+                if st:
+                    st.queue.put_nowait(None)  # stream ends with no tokens
             now = time.time()
             for out in outputs:
                 st = self.streams.get(out.request_id)

@@ -222,3 +222,20 @@ def test_completion_n_choices(app):
         assert r2.status_code == 400
 
     run_with_client(app, fn)
+
+
+def test_prompt_too_long_rejected(app):
+    async def fn(client):
+        r = await client.post("/v1/completions", json={
+            "model": "tiny", "prompt": "x" * 2000, "max_tokens": 2,
+        })
+        assert r.status_code == 400
+        assert "max_model_len" in r.json()["error"]["message"]
+        # server still healthy afterwards (engine loop not killed)
+        r2 = await client.post("/v1/completions", json={
+            "model": "tiny", "prompt": "ok", "max_tokens": 2,
+            "ignore_eos": True,
+        })
+        assert r2.status_code == 200
+
+    run_with_client(app, fn)
