@@ -29,12 +29,19 @@ from ..parallel.layers import (
 
 
 class RMSNorm(nn.Module):
+    fp8_out = False  # emit (fp8, scales) for a W8A8 consumer GEMM
+
     def __init__(self, hidden: int, eps: float, dtype=torch.bfloat16):
         super().__init__()
         self.weight = nn.Parameter(torch.ones(hidden, dtype=dtype), requires_grad=False)
         self.eps = eps
 
     def forward(self, x, residual=None):
+        if self.fp8_out:
+            if residual is None:
+                return ops.rmsnorm_fp8(x, self.weight, self.eps)
+            q, s, res = ops.fused_add_rmsnorm_fp8(x, residual, self.weight, self.eps)
+            return (q, s), res
         if residual is None:
             return ops.rmsnorm(x, self.weight, self.eps)
         return ops.fused_add_rmsnorm(x, residual, self.weight, self.eps)
@@ -57,8 +64,8 @@ class Attention(nn.Module):
             dtype=dtype,
         )
 
-    def forward(self, x: torch.Tensor, batch: ForwardBatch, kv_cache) -> torch.Tensor:
-        T = x.shape[0]
+    def forward(self, x, batch: ForwardBatch, kv_cache) -> torch.Tensor:
+        T = (x[0] if isinstance(x, tuple) else x).shape[0]
         qkv = self.qkv_proj(x)
         # Strided views into the fused QKV buffer — RoPE/cache/attention
         # kernels take row strides, so no .contiguous() copies on the hot path.
@@ -100,8 +107,13 @@ class MLP(nn.Module):
             cfg.intermediate_size, cfg.hidden_size, bias=False, dtype=dtype
         )
 
+    fp8 = False
+
     def forward(self, x):
-        return self.down_proj(ops.silu_mul(self.gate_up_proj(x)))
+        gate_up = self.gate_up_proj(x)
+        if self.fp8:
+            return self.down_proj(ops.silu_mul_fp8(gate_up))
+        return self.down_proj(ops.silu_mul(gate_up))
 
 
 class DecoderLayer(nn.Module):
@@ -156,7 +168,12 @@ class LlamaFamilyForCausalLM(nn.Module):
             quantize_module_fp8(layer.self_attn.qkv_proj)
             quantize_module_fp8(layer.mlp.gate_up_proj)
             quantize_module_fp8(layer.mlp.down_proj)
+            # producing kernels emit fp8 directly (fused activation quant)
+            layer.input_layernorm.fp8_out = True
+            layer.post_attention_layernorm.fp8_out = True
+            layer.mlp.fp8 = True
         quantize_module_fp8(self.lm_head)
+        self.norm.fp8_out = True
 
     def _build_cos_sin(self):
         from ..ops import ref
@@ -178,6 +195,12 @@ class LlamaFamilyForCausalLM(nn.Module):
         for i, layer in enumerate(self.layers):
             x, residual = layer(x, residual, batch, kv_caches[i])
         x, _ = self.norm(x, residual)
+        if isinstance(x, tuple):  # fp8: (values, per-token scales)
+            q8, sc = x
+            if batch.logits_indices is not None:
+                q8 = q8[batch.logits_indices]
+                sc = sc[batch.logits_indices]
+            return self.lm_head((q8, sc))
         if batch.logits_indices is not None:
             x = x[batch.logits_indices]
         return self.lm_head(x)

@@ -247,6 +247,43 @@ def linear_bf16(x, weight, bias=None):
     return F.linear(x, weight, bias)
 
 
+def rmsnorm_fp8(x, weight, eps: float = 1e-6):
+    """Fused RMSNorm -> per-token fp8 e4m3: (out8, inv_scale). GPU-only
+    fast path (hidden <= 8192); otherwise compose the unfused ops."""
+    if x.is_cuda and x.shape[-1] <= 8192:
+        out = torch.empty(x.shape, dtype=torch.float8_e4m3fn, device=x.device)
+        inv_scale = torch.empty(x.shape[0], dtype=torch.float32, device=x.device)
+        _native().rmsnorm_fp8(out, inv_scale, x, None, weight, eps)
+        return out, inv_scale
+    return quant_fp8_rows(rmsnorm(x, weight, eps))
+
+
+def fused_add_rmsnorm_fp8(x, residual, weight, eps: float = 1e-6):
+    """Fused residual-add + RMSNorm -> fp8: (out8, inv_scale, residual)."""
+    if x.is_cuda and x.shape[-1] <= 8192:
+        out = torch.empty(x.shape, dtype=torch.float8_e4m3fn, device=x.device)
+        inv_scale = torch.empty(x.shape[0], dtype=torch.float32, device=x.device)
+        _native().rmsnorm_fp8(out, inv_scale, x, residual, weight, eps)
+        return out, inv_scale, residual
+    y, res = fused_add_rmsnorm(x, residual, weight, eps)
+    q, s = quant_fp8_rows(y)
+    return q, s, res
+
+
+def silu_mul_fp8(gate_up):
+    """Fused SwiGLU -> per-token fp8: (out8, inv_scale)."""
+    d = gate_up.shape[-1] // 2
+    # dynamic-LDS default cap is 64 KiB; stay under it (d*2 bytes staged)
+    if gate_up.is_cuda and d % 8 == 0 and d * 2 <= 60 * 1024:
+        out = torch.empty((gate_up.shape[0], d), dtype=torch.float8_e4m3fn,
+                          device=gate_up.device)
+        inv_scale = torch.empty(gate_up.shape[0], dtype=torch.float32,
+                                device=gate_up.device)
+        _native().silu_mul_fp8(out, inv_scale, gate_up.contiguous())
+        return out, inv_scale
+    return quant_fp8_rows(silu_mul(gate_up))
+
+
 def quant_fp8_rows(x):
     """Dynamic per-token fp8 e4m3 quantization: [M, K] bf16 ->
     ([M, K] float8_e4m3fn, [M] fp32 dequant scales) for W8A8 GEMMs

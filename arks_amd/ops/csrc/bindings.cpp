@@ -46,6 +46,11 @@ void arks_attn_extend_paged(void* out, const void* q, const void* k_cache,
                             hipStream_t stream);
 void arks_quant_fp8_rows(void* out, void* inv_scale, const void* x, int rows,
                          int cols, hipStream_t stream);
+void arks_rmsnorm_fp8(void* out, void* inv_scale, const void* input,
+                      void* residual, const void* weight, float eps, int rows,
+                      int hidden, int fused_add, hipStream_t stream);
+void arks_silu_mul_fp8(void* out, void* inv_scale, const void* gate_up,
+                       int rows, int d, hipStream_t stream);
 void arks_skinny_gemm(void* part, void* out, const void* a, const void* w,
                       const void* bias, int m_rows, int n_total, int k_total,
                       int k_per_split, int nsplits, int64_t a_stride,
@@ -272,6 +277,40 @@ void skinny_gemm(torch::Tensor out, torch::Tensor part, torch::Tensor a,
                    current_stream());
 }
 
+void rmsnorm_fp8(torch::Tensor out, torch::Tensor inv_scale,
+                 torch::Tensor input, c10::optional<torch::Tensor> residual,
+                 torch::Tensor weight, double eps) {
+  check_bf16_contig(input, "input");
+  check_bf16_contig(weight, "weight");
+  TORCH_CHECK(out.scalar_type() == torch::kFloat8_e4m3fn && out.is_contiguous());
+  TORCH_CHECK(inv_scale.scalar_type() == torch::kFloat32);
+  const int hidden = input.size(-1);
+  TORCH_CHECK(hidden % 8 == 0 && hidden <= 8192,
+              "rmsnorm_fp8 requires hidden % 8 == 0 and hidden <= 8192");
+  const int rows = input.numel() / hidden;
+  void* res_ptr = nullptr;
+  if (residual.has_value()) {
+    check_bf16_contig(*residual, "residual");
+    res_ptr = residual->data_ptr();
+  }
+  arks_rmsnorm_fp8(out.data_ptr(), inv_scale.data_ptr(), input.data_ptr(),
+                   res_ptr, weight.data_ptr(), (float)eps, rows, hidden,
+                   residual.has_value() ? 1 : 0, current_stream());
+}
+
+void silu_mul_fp8(torch::Tensor out, torch::Tensor inv_scale,
+                  torch::Tensor gate_up) {
+  check_bf16_contig(gate_up, "gate_up");
+  TORCH_CHECK(out.scalar_type() == torch::kFloat8_e4m3fn && out.is_contiguous());
+  TORCH_CHECK(inv_scale.scalar_type() == torch::kFloat32);
+  const int d = out.size(-1);
+  TORCH_CHECK(gate_up.size(-1) == 2 * d && d % 8 == 0);
+  TORCH_CHECK((int64_t)d * 2 <= 160 * 1024, "row too large for LDS staging");
+  const int rows = out.numel() / d;
+  arks_silu_mul_fp8(out.data_ptr(), inv_scale.data_ptr(), gate_up.data_ptr(),
+                    rows, d, current_stream());
+}
+
 void quant_fp8_rows(torch::Tensor out, torch::Tensor inv_scale,
                     torch::Tensor x) {
   check_bf16_contig(x, "x");
@@ -337,6 +376,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("attention_prefill_varlen", &attention_prefill_varlen);
   m.def("attention_extend_paged", &attention_extend_paged);
   m.def("quant_fp8_rows", &quant_fp8_rows);
+  m.def("rmsnorm_fp8", &rmsnorm_fp8);
+  m.def("silu_mul_fp8", &silu_mul_fp8);
   m.def("skinny_gemm", &skinny_gemm);
   m.def("greedy_sample", &greedy_sample);
   m.def("gumbel_sample", &gumbel_sample);

@@ -40,6 +40,18 @@ def quantize_module_fp8(mod: nn.Module) -> None:
     mod.fp8 = True
 
 
+def _fp8_linear_prequant(mod: nn.Module, x8, sa) -> torch.Tensor:
+    """fp8 GEMM on an already-quantized activation (the producing kernel —
+    rmsnorm_fp8 / silu_mul_fp8 — emitted x8 + per-token scales)."""
+    if x8.is_cuda:
+        return torch._scaled_mm(
+            x8, mod.weight_fp8.t(), scale_a=sa[:, None], scale_b=mod.w_inv_scale,
+            bias=mod.bias, out_dtype=torch.bfloat16,
+        )
+    xd = (x8.float() * sa[:, None]).to(torch.bfloat16)
+    return F.linear(xd, mod.weight_qdq, mod.bias)
+
+
 def _fp8_linear(mod: nn.Module, x: torch.Tensor) -> torch.Tensor:
     from .. import ops
 
@@ -83,7 +95,9 @@ class ColumnParallelLinear(nn.Module):
 
     fp8 = False
 
-    def forward(self, x: torch.Tensor) -> torch.Tensor:
+    def forward(self, x) -> torch.Tensor:
+        if isinstance(x, tuple):  # pre-quantized by the producing kernel
+            return _fp8_linear_prequant(self, *x)
         if self.fp8:
             return _fp8_linear(self, x)
         from .. import ops
@@ -169,12 +183,13 @@ class RowParallelLinear(nn.Module):
 
     fp8 = False
 
-    def forward(self, x: torch.Tensor) -> torch.Tensor:
-        if self.fp8:
+    def forward(self, x) -> torch.Tensor:
+        if isinstance(x, tuple) or self.fp8:
             bias = self.bias
             self.bias = None  # bias must be applied post-reduce, once
             try:
-                y = _fp8_linear(self, x)
+                y = (_fp8_linear_prequant(self, *x) if isinstance(x, tuple)
+                     else _fp8_linear(self, x))
             finally:
                 self.bias = bias
         else:
@@ -214,10 +229,11 @@ class ParallelLMHead(nn.Module):
 
     fp8 = False
 
-    def forward(self, x: torch.Tensor) -> torch.Tensor:
-        if self.fp8:
+    def forward(self, x) -> torch.Tensor:
+        if isinstance(x, tuple) or self.fp8:
             self.bias = None
-            logits = _fp8_linear(self, x)
+            logits = (_fp8_linear_prequant(self, *x) if isinstance(x, tuple)
+                      else _fp8_linear(self, x))
         else:
             from .. import ops
 

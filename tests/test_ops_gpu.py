@@ -396,3 +396,36 @@ def test_mfma_probe_32x32():
     _native().mfma_probe32(d, a, b)
     ref_out = a.float() @ b.float()
     torch.testing.assert_close(d.cpu(), ref_out.cpu(), atol=2e-2, rtol=2e-2)
+
+
+def test_rmsnorm_fp8_fused_matches_composed():
+    assert_native()
+    torch.manual_seed(31)
+    x = torch.randn(17, 3584, dtype=torch.bfloat16, device=DEV)
+    res = torch.randn_like(x)
+    w = torch.randn(3584, dtype=torch.bfloat16, device=DEV)
+    # no-residual variant
+    q, s = ops.rmsnorm_fp8(x, w, 1e-6)
+    y = ops.rmsnorm(x, w, 1e-6)
+    qr, sr = ops.quant_fp8_rows(y)
+    torch.testing.assert_close(s, sr, atol=1e-5, rtol=1e-4)
+    dq = q.float() * s[:, None]
+    dqr = qr.float() * sr[:, None]
+    torch.testing.assert_close(dq, dqr, atol=0.05, rtol=0.1)
+    # fused-add variant mutates residual identically
+    res2 = res.clone()
+    q2, s2, r2 = ops.fused_add_rmsnorm_fp8(x, res, w, 1e-6)
+    y2, r2b = ops.fused_add_rmsnorm(x, res2, w, 1e-6)
+    torch.testing.assert_close(r2, r2b)
+    dq2 = q2.float() * s2[:, None]
+    torch.testing.assert_close(dq2, y2.float(), atol=0.08, rtol=0.12)
+
+
+def test_silu_mul_fp8_fused_matches_composed():
+    assert_native()
+    torch.manual_seed(37)
+    gu = torch.randn(9, 2 * 18944, dtype=torch.bfloat16, device=DEV)
+    q, s = ops.silu_mul_fp8(gu)
+    y = ops.silu_mul(gu)
+    dq = q.float() * s[:, None]
+    torch.testing.assert_close(dq, y.float(), atol=0.08, rtol=0.12)
