@@ -95,6 +95,32 @@ PRESET_CONFIGS: dict[str, ModelConfig] = {
         head_dim=128,
         rope_theta=1000000.0,
     ),
+    "qwen2.5-32b": ModelConfig(
+        architecture="Qwen2ForCausalLM",
+        vocab_size=152064,
+        hidden_size=5120,
+        intermediate_size=27648,
+        num_hidden_layers=64,
+        num_attention_heads=40,
+        num_key_value_heads=8,
+        head_dim=128,
+        rope_theta=1000000.0,
+    ),
+    "llama-3-8b": ModelConfig(
+        architecture="LlamaForCausalLM",
+        vocab_size=128256,
+        hidden_size=4096,
+        intermediate_size=14336,
+        num_hidden_layers=32,
+        num_attention_heads=32,
+        num_key_value_heads=8,
+        head_dim=128,
+        rms_norm_eps=1e-5,
+        rope_theta=500000.0,
+        attention_bias=False,
+        eos_token_id=128001,
+        bos_token_id=128000,
+    ),
     "llama-3-70b": ModelConfig(
         architecture="LlamaForCausalLM",
         vocab_size=128256,
