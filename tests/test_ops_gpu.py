@@ -408,17 +408,19 @@ def test_rmsnorm_fp8_fused_matches_composed():
     q, s = ops.rmsnorm_fp8(x, w, 1e-6)
     y = ops.rmsnorm(x, w, 1e-6)
     qr, sr = ops.quant_fp8_rows(y)
-    torch.testing.assert_close(s, sr, atol=1e-5, rtol=1e-4)
+    # fused quantizes the f32 normalized values; composed goes through a
+    # bf16 round first -> scales/values differ by ~1 bf16 ulp
+    torch.testing.assert_close(s, sr, atol=1e-4, rtol=1e-2)
     dq = q.float() * s[:, None]
     dqr = qr.float() * sr[:, None]
-    torch.testing.assert_close(dq, dqr, atol=0.05, rtol=0.1)
+    torch.testing.assert_close(dq, dqr, atol=0.1, rtol=0.15)
     # fused-add variant mutates residual identically
     res2 = res.clone()
     q2, s2, r2 = ops.fused_add_rmsnorm_fp8(x, res, w, 1e-6)
     y2, r2b = ops.fused_add_rmsnorm(x, res2, w, 1e-6)
     torch.testing.assert_close(r2, r2b)
     dq2 = q2.float() * s2[:, None]
-    torch.testing.assert_close(dq2, y2.float(), atol=0.08, rtol=0.12)
+    torch.testing.assert_close(dq2, y2.float(), atol=0.1, rtol=0.15)
 
 
 def test_silu_mul_fp8_fused_matches_composed():
