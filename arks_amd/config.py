@@ -26,6 +26,8 @@ class ModelConfig:
     rms_norm_eps: float = 1e-6
     rope_theta: float = 1000000.0
     max_position_embeddings: int = 32768
+    # HF rope_scaling dict: {"rope_type": "llama3"|"yarn"|"linear", ...}
+    rope_scaling: dict | None = None
     tie_word_embeddings: bool = False
     attention_bias: bool = True  # qwen2 has qkv bias; llama does not
     eos_token_id: int = 151645
@@ -52,6 +54,7 @@ class ModelConfig:
             rms_norm_eps=cfg.get("rms_norm_eps", 1e-6),
             rope_theta=cfg.get("rope_theta", 10000.0),
             max_position_embeddings=cfg.get("max_position_embeddings", 32768),
+            rope_scaling=cfg.get("rope_scaling"),
             tie_word_embeddings=cfg.get("tie_word_embeddings", False),
             attention_bias=cfg.get("attention_bias", arch.startswith("Qwen2")),
             eos_token_id=eos,

@@ -179,7 +179,8 @@ class LlamaFamilyForCausalLM(nn.Module):
         from ..ops import ref
 
         return ref.rope_cos_sin_cache(
-            self.cfg.head_dim, self.cfg.max_position_embeddings, self.cfg.rope_theta
+            self.cfg.head_dim, self.cfg.max_position_embeddings,
+            self.cfg.rope_theta, rope_scaling=self.cfg.rope_scaling,
         )
 
     def _apply(self, fn, recurse=True):  # keep the shared cos_sin in sync

@@ -44,21 +44,70 @@ def silu_mul(gate_up: torch.Tensor) -> torch.Tensor:
     return (torch.nn.functional.silu(gate) * up).to(gate_up.dtype)
 
 
+def _scaled_inv_freq(inv_freq: torch.Tensor, rope_scaling: dict | None,
+                     base: float, head_dim: int) -> tuple[torch.Tensor, float]:
+    """Apply an HF rope_scaling config; returns (inv_freq, mscale).
+    Supports the llama3 ramp, YaRN (NTK-by-parts + attention scaling) and
+    plain linear scaling — matching transformers' rope_utils semantics."""
+    if not rope_scaling:
+        return inv_freq, 1.0
+    rtype = rope_scaling.get("rope_type") or rope_scaling.get("type")
+    factor = float(rope_scaling.get("factor", 1.0))
+    if rtype == "linear":
+        return inv_freq / factor, 1.0
+    if rtype == "llama3":
+        lo = float(rope_scaling.get("low_freq_factor", 1.0))
+        hi = float(rope_scaling.get("high_freq_factor", 4.0))
+        orig = float(rope_scaling.get("original_max_position_embeddings", 8192))
+        wavelen = 2 * math.pi / inv_freq
+        scaled = torch.where(wavelen > orig / lo, inv_freq / factor, inv_freq)
+        smooth = (orig / wavelen - lo) / (hi - lo)
+        mid = (1 - smooth) * inv_freq / factor + smooth * inv_freq
+        is_mid = (wavelen <= orig / lo) & (wavelen >= orig / hi)
+        return torch.where(is_mid, mid, scaled), 1.0
+    if rtype == "yarn":
+        orig = float(rope_scaling.get("original_max_position_embeddings", 4096))
+        beta_fast = float(rope_scaling.get("beta_fast", 32.0))
+        beta_slow = float(rope_scaling.get("beta_slow", 1.0))
+
+        def corr_dim(num_rot: float) -> float:
+            return (head_dim * math.log(orig / (num_rot * 2 * math.pi))) / (
+                2 * math.log(base))
+
+        low = max(math.floor(corr_dim(beta_fast)), 0)
+        high = min(math.ceil(corr_dim(beta_slow)), head_dim // 2 - 1)
+        ramp = torch.clamp(
+            (torch.arange(head_dim // 2, dtype=torch.float64,
+                          device=inv_freq.device) - low) / max(high - low, 1),
+            0, 1,
+        )
+        # ramp==0 (high-freq dims) -> extrapolate (keep original);
+        # ramp==1 (long wavelengths) -> interpolate (divide by factor)
+        out = (inv_freq / factor) * ramp + inv_freq * (1 - ramp)
+        mscale = float(rope_scaling.get("attention_factor") or
+                       (0.1 * math.log(factor) + 1.0))
+        return out, mscale
+    return inv_freq, 1.0
+
+
 def rope_cos_sin_cache(
     head_dim: int,
     max_positions: int,
     base: float = 10000.0,
     dtype: torch.dtype = torch.float32,
     device=None,
+    rope_scaling: dict | None = None,
 ) -> torch.Tensor:
     """Precomputed [max_positions, head_dim] cache: first half cos, second half sin
-    (per rotary pair), matching HF neox-style RoPE."""
+    (per rotary pair), matching HF neox-style RoPE (incl. llama3 / yarn /
+    linear rope_scaling)."""
     inv_freq = 1.0 / (
         base ** (torch.arange(0, head_dim, 2, dtype=torch.float64, device=device) / head_dim)
     )
+    inv_freq, mscale = _scaled_inv_freq(inv_freq, rope_scaling, base, head_dim)
     t = torch.arange(max_positions, dtype=torch.float64, device=device)
     freqs = torch.outer(t, inv_freq)  # [P, head_dim/2]
-    cache = torch.cat([freqs.cos(), freqs.sin()], dim=-1)  # [P, head_dim]
+    cache = torch.cat([freqs.cos() * mscale, freqs.sin() * mscale], dim=-1)
     return cache.to(dtype)
 
 

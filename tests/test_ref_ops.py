@@ -63,3 +63,34 @@ def test_gumbel_sample_greedy_when_t0():
     u = torch.rand(5, 100)
     out = ref.gumbel_sample(logits, temps, u)
     torch.testing.assert_close(out, logits.argmax(-1))
+
+
+def test_rope_scaling_matches_transformers():
+    """llama3 + yarn rope_scaling pinned against transformers' rope_utils."""
+    import torch
+    from transformers import LlamaConfig
+    from transformers.modeling_rope_utils import ROPE_INIT_FUNCTIONS
+
+    from arks_amd.ops.ref import _scaled_inv_freq
+
+    base_inv = 1.0 / (500000.0 ** (torch.arange(0, 128, 2, dtype=torch.float64) / 128))
+    l3 = {"rope_type": "llama3", "factor": 8.0, "low_freq_factor": 1.0,
+          "high_freq_factor": 4.0, "original_max_position_embeddings": 8192}
+    cfg = LlamaConfig(rope_theta=500000.0, hidden_size=4096,
+                      num_attention_heads=32,
+                      max_position_embeddings=131072, rope_scaling=dict(l3))
+    inv_hf, _ = ROPE_INIT_FUNCTIONS["llama3"](cfg, "cpu")
+    mine, ms = _scaled_inv_freq(base_inv, l3, 500000.0, 128)
+    assert ms == 1.0
+    torch.testing.assert_close(mine, inv_hf.double(), rtol=1e-5, atol=0)
+
+    yarn = {"rope_type": "yarn", "factor": 4.0,
+            "original_max_position_embeddings": 4096}
+    cfg2 = LlamaConfig(rope_theta=10000.0, hidden_size=4096,
+                       num_attention_heads=32,
+                       max_position_embeddings=16384, rope_scaling=dict(yarn))
+    inv_hf2, att2 = ROPE_INIT_FUNCTIONS["yarn"](cfg2, "cpu")
+    base2 = 1.0 / (10000.0 ** (torch.arange(0, 128, 2, dtype=torch.float64) / 128))
+    mine2, ms2 = _scaled_inv_freq(base2, yarn, 10000.0, 128)
+    torch.testing.assert_close(mine2, inv_hf2.double(), rtol=1e-5, atol=0)
+    assert abs(ms2 - att2) < 1e-9
