@@ -82,3 +82,55 @@ def test_logits_match_transformers(tmp_path):
     assert diff < 2e-3, f"max logits diff {diff}"
     # and the next-token argmax ranking agrees everywhere
     assert torch.equal(logits.argmax(-1), hf_logits.argmax(-1))
+
+
+def test_llama_rope_scaling_logits_match_transformers(tmp_path):
+    """Llama arch (no qkv bias) + llama3 rope_scaling vs transformers."""
+    import dataclasses
+
+    from arks_amd.loader.safetensors_loader import (
+        load_model_weights,
+        save_random_checkpoint,
+    )
+
+    base = PRESET_CONFIGS["tiny"]
+    cfg = dataclasses.replace(
+        base,
+        architecture="LlamaForCausalLM",
+        attention_bias=False,
+        rope_theta=500000.0,
+        rope_scaling={
+            "rope_type": "llama3", "factor": 8.0, "low_freq_factor": 1.0,
+            "high_freq_factor": 4.0, "original_max_position_embeddings": 256,
+        },
+    )
+    save_random_checkpoint(cfg, str(tmp_path), seed=11)
+    ours = create_model(cfg, dtype=torch.float32)
+    load_model_weights(ours, str(tmp_path), torch.device("cpu"))
+
+    hf_cfg = transformers.LlamaConfig(
+        vocab_size=cfg.vocab_size,
+        hidden_size=cfg.hidden_size,
+        intermediate_size=cfg.intermediate_size,
+        num_hidden_layers=cfg.num_hidden_layers,
+        num_attention_heads=cfg.num_attention_heads,
+        num_key_value_heads=cfg.num_key_value_heads,
+        rms_norm_eps=cfg.rms_norm_eps,
+        rope_theta=cfg.rope_theta,
+        max_position_embeddings=cfg.max_position_embeddings,
+        rope_scaling=dict(cfg.rope_scaling),
+        tie_word_embeddings=cfg.tie_word_embeddings,
+        attention_bias=False,
+        attention_dropout=0.0,
+    )
+    hf = transformers.LlamaForCausalLM.from_pretrained(
+        str(tmp_path), config=hf_cfg, torch_dtype=torch.float32
+    )
+    hf.eval()
+    ids = [3, 1, 4, 1, 5, 9, 2, 6, 5, 3]
+    logits = forward_ours(ours, cfg, ids)
+    with torch.no_grad():
+        hf_logits = hf(torch.tensor([ids])).logits[0]
+    diff = (logits - hf_logits).abs().max().item()
+    assert diff < 2e-3, f"max logits diff {diff}"
+    assert torch.equal(logits.argmax(-1), hf_logits.argmax(-1))
