@@ -28,6 +28,9 @@ class ModelConfig:
     max_position_embeddings: int = 32768
     # HF rope_scaling dict: {"rope_type": "llama3"|"yarn"|"linear", ...}
     rope_scaling: dict | None = None
+    # Mixtral-style sparse MoE (0 experts = dense MLP)
+    num_local_experts: int = 0
+    num_experts_per_tok: int = 2
     tie_word_embeddings: bool = False
     attention_bias: bool = True  # qwen2 has qkv bias; llama does not
     eos_token_id: int = 151645
@@ -55,6 +58,8 @@ class ModelConfig:
             rope_theta=cfg.get("rope_theta", 10000.0),
             max_position_embeddings=cfg.get("max_position_embeddings", 32768),
             rope_scaling=cfg.get("rope_scaling"),
+            num_local_experts=cfg.get("num_local_experts", 0),
+            num_experts_per_tok=cfg.get("num_experts_per_tok", 2),
             tie_word_embeddings=cfg.get("tie_word_embeddings", False),
             attention_bias=cfg.get("attention_bias", arch.startswith("Qwen2")),
             eos_token_id=eos,
@@ -138,6 +143,40 @@ PRESET_CONFIGS: dict[str, ModelConfig] = {
         attention_bias=False,
         eos_token_id=128001,
         bos_token_id=128000,
+    ),
+    "mixtral-8x7b": ModelConfig(
+        architecture="MixtralForCausalLM",
+        vocab_size=32000,
+        hidden_size=4096,
+        intermediate_size=14336,  # per expert
+        num_hidden_layers=32,
+        num_attention_heads=32,
+        num_key_value_heads=8,
+        head_dim=128,
+        rms_norm_eps=1e-5,
+        rope_theta=1000000.0,
+        attention_bias=False,
+        num_local_experts=8,
+        num_experts_per_tok=2,
+        eos_token_id=2,
+        bos_token_id=1,
+    ),
+    "tiny-moe": ModelConfig(  # CPU-test-sized sparse MoE
+        architecture="MixtralForCausalLM",
+        vocab_size=512,
+        hidden_size=128,
+        intermediate_size=192,
+        num_hidden_layers=2,
+        num_attention_heads=4,
+        num_key_value_heads=2,
+        head_dim=32,
+        rope_theta=10000.0,
+        max_position_embeddings=2048,
+        attention_bias=False,
+        num_local_experts=4,
+        num_experts_per_tok=2,
+        eos_token_id=2,
+        bos_token_id=1,
     ),
     "tiny": ModelConfig(  # CPU-test-sized
         architecture="Qwen2ForCausalLM",

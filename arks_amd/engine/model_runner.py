@@ -18,7 +18,6 @@ import torch
 
 from .. import ops
 from ..config import EngineConfig, ModelConfig
-from ..models import create_model
 from ..parallel.comm import get_tp_world_size
 from .forward_batch import ForwardBatch
 from .kv_cache import BlockAllocator, PrefixCachingAllocator, kv_cache_block_bytes
@@ -44,6 +43,8 @@ class ModelRunner:
         self.dtype = torch.bfloat16
         torch.manual_seed(engine_cfg.seed)
         self._setup_tunableop()
+        from ..models import create_model  # lazy: breaks the import cycle
+
         self.model = create_model(self.model_cfg, dtype=self.dtype)
         self._weights_loaded = False
         self.kv_caches: list[tuple[torch.Tensor, torch.Tensor]] = []

@@ -128,11 +128,20 @@ def save_random_checkpoint(cfg, out_dir: str, seed: int = 0) -> None:
             out[f"{pre}.self_attn.k_proj.bias"] = kb
             out[f"{pre}.self_attn.v_proj.bias"] = vb
         out[f"{pre}.self_attn.o_proj.weight"] = layer.self_attn.o_proj.weight.data
-        gu = layer.mlp.gate_up_proj.weight.data
-        g, u = gu.split(gu.shape[0] // 2, dim=0)
-        out[f"{pre}.mlp.gate_proj.weight"] = g
-        out[f"{pre}.mlp.up_proj.weight"] = u
-        out[f"{pre}.mlp.down_proj.weight"] = layer.mlp.down_proj.weight.data
+        if hasattr(layer.mlp, "w13"):  # sparse MoE block
+            out[f"{pre}.block_sparse_moe.gate.weight"] = layer.mlp.gate.data
+            I = layer.mlp.inter
+            for e in range(layer.mlp.local_experts):
+                ep = f"{pre}.block_sparse_moe.experts.{e}"
+                out[f"{ep}.w1.weight"] = layer.mlp.w13.data[e, :I]
+                out[f"{ep}.w3.weight"] = layer.mlp.w13.data[e, I:]
+                out[f"{ep}.w2.weight"] = layer.mlp.w2.data[e]
+        else:
+            gu = layer.mlp.gate_up_proj.weight.data
+            g, u = gu.split(gu.shape[0] // 2, dim=0)
+            out[f"{pre}.mlp.gate_proj.weight"] = g
+            out[f"{pre}.mlp.up_proj.weight"] = u
+            out[f"{pre}.mlp.down_proj.weight"] = layer.mlp.down_proj.weight.data
         out[f"{pre}.input_layernorm.weight"] = layer.input_layernorm.weight.data
         out[f"{pre}.post_attention_layernorm.weight"] = (
             layer.post_attention_layernorm.weight.data

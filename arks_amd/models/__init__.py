@@ -8,6 +8,7 @@ from .llama_family import LlamaFamilyForCausalLM
 _REGISTRY = {
     "Qwen2ForCausalLM": LlamaFamilyForCausalLM,
     "LlamaForCausalLM": LlamaFamilyForCausalLM,
+    "MixtralForCausalLM": LlamaFamilyForCausalLM,  # sparse MoE MLP branch
 }
 
 

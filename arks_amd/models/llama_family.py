@@ -19,7 +19,7 @@ import torch.nn as nn
 from .. import ops
 from ..config import ModelConfig
 from ..engine.forward_batch import ForwardBatch
-from ..parallel.comm import get_tp_world_size
+from ..parallel.comm import get_tp_rank, get_tp_world_size
 from ..parallel.layers import (
     MergedColumnParallelLinear,
     ParallelLMHead,
@@ -116,13 +116,73 @@ class MLP(nn.Module):
         return self.down_proj(ops.silu_mul(gate_up))
 
 
+class MoEMLP(nn.Module):
+    """Mixtral-style sparse MoE block, expert-parallel over the TP group:
+    each rank owns num_local_experts/tp whole experts (an expert's FFN is
+    never split), computes its experts' contributions for the tokens routed
+    to them, and one all-reduce sums the partial outputs — the same single
+    collective per block as the dense RowParallel MLP. The router (gate) is
+    replicated so every rank routes identically."""
+
+    fp8 = False  # MoE experts stay bf16 (v1)
+
+    def __init__(self, cfg: ModelConfig, dtype=torch.bfloat16):
+        super().__init__()
+        tp = get_tp_world_size()
+        E = cfg.num_local_experts
+        assert E % tp == 0, (E, tp)
+        self.num_experts = E
+        self.top_k = cfg.num_experts_per_tok
+        self.local_experts = E // tp
+        self.expert_base = get_tp_rank() * self.local_experts
+        self.hidden = cfg.hidden_size
+        self.inter = cfg.intermediate_size
+        self.gate = nn.Parameter(
+            torch.empty(E, cfg.hidden_size, dtype=dtype), requires_grad=False
+        )
+        # fused [gate|up] and down per local expert
+        self.w13 = nn.Parameter(
+            torch.empty(self.local_experts, 2 * self.inter, self.hidden,
+                        dtype=dtype), requires_grad=False,
+        )
+        self.w2 = nn.Parameter(
+            torch.empty(self.local_experts, self.hidden, self.inter,
+                        dtype=dtype), requires_grad=False,
+        )
+
+    def forward(self, x):
+        from ..parallel.comm import tp_all_reduce
+
+        T = x.shape[0]
+        router_logits = torch.nn.functional.linear(x, self.gate).float()
+        probs = torch.softmax(router_logits, dim=-1)
+        weights, selected = probs.topk(self.top_k, dim=-1)  # [T, k]
+        weights = weights / weights.sum(dim=-1, keepdim=True)
+        out = torch.zeros_like(x)
+        for le in range(self.local_experts):
+            ge = self.expert_base + le
+            tok, slot = (selected == ge).nonzero(as_tuple=True)
+            if tok.numel() == 0:
+                continue
+            h = ops.silu_mul(
+                torch.nn.functional.linear(x[tok], self.w13[le])
+            )
+            y = torch.nn.functional.linear(h, self.w2[le])
+            out.index_add_(
+                0, tok, y * weights[tok, slot, None].to(y.dtype)
+            )
+        return tp_all_reduce(out)
+
+
 class DecoderLayer(nn.Module):
     def __init__(self, cfg: ModelConfig, dtype=torch.bfloat16):
         super().__init__()
         self.input_layernorm = RMSNorm(cfg.hidden_size, cfg.rms_norm_eps, dtype)
         self.self_attn = Attention(cfg, dtype)
         self.post_attention_layernorm = RMSNorm(cfg.hidden_size, cfg.rms_norm_eps, dtype)
-        self.mlp = MLP(cfg, dtype)
+        self.mlp = (
+            MoEMLP(cfg, dtype) if cfg.num_local_experts > 0 else MLP(cfg, dtype)
+        )
 
     def forward(self, x, residual, batch: ForwardBatch, kv_cache):
         if residual is None:
@@ -166,10 +226,12 @@ class LlamaFamilyForCausalLM(nn.Module):
 
         for layer in self.layers:
             quantize_module_fp8(layer.self_attn.qkv_proj)
+            layer.input_layernorm.fp8_out = True
+            if isinstance(layer.mlp, MoEMLP):
+                continue  # MoE experts stay bf16 (v1)
             quantize_module_fp8(layer.mlp.gate_up_proj)
             quantize_module_fp8(layer.mlp.down_proj)
             # producing kernels emit fp8 directly (fused activation quant)
-            layer.input_layernorm.fp8_out = True
             layer.post_attention_layernorm.fp8_out = True
             layer.mlp.fp8 = True
         quantize_module_fp8(self.lm_head)
@@ -273,6 +335,23 @@ class LlamaFamilyForCausalLM(nn.Module):
                         d.clear()
                 elif sub == "mlp.down_proj.weight":
                     put(f"layers.{li}.mlp.down_proj.weight", layer.mlp.down_proj.shard(w))
+                elif sub == "block_sparse_moe.gate.weight":
+                    layer.mlp.gate.data.copy_(w.to(layer.mlp.gate.dtype))
+                elif sub.startswith("block_sparse_moe.experts."):
+                    # Mixtral expert naming: w1=gate, w3=up, w2=down.
+                    # EP-sharded: only this rank's experts are kept.
+                    moe = layer.mlp
+                    e = int(sub.split(".")[2])
+                    which = sub.split(".")[3]
+                    le = e - moe.expert_base
+                    if 0 <= le < moe.local_experts:
+                        I = moe.inter
+                        if which == "w1":
+                            moe.w13.data[le, :I].copy_(w.to(moe.w13.dtype))
+                        elif which == "w3":
+                            moe.w13.data[le, I:].copy_(w.to(moe.w13.dtype))
+                        elif which == "w2":
+                            moe.w2.data[le].copy_(w.to(moe.w2.dtype))
                 # rotary_emb.inv_freq etc. are ignored (recomputed)
 
     @torch.no_grad()
@@ -313,12 +392,22 @@ class LlamaFamilyForCausalLM(nn.Module):
                 f"{pre}.self_attn.k_proj.weight": gen(f"{pre}.k", nkv * hd, H),
                 f"{pre}.self_attn.v_proj.weight": gen(f"{pre}.v", nkv * hd, H),
                 f"{pre}.self_attn.o_proj.weight": gen(f"{pre}.o", H, nq * hd),
-                f"{pre}.mlp.gate_proj.weight": gen(f"{pre}.gate", I, H),
-                f"{pre}.mlp.up_proj.weight": gen(f"{pre}.up", I, H),
-                f"{pre}.mlp.down_proj.weight": gen(f"{pre}.down", H, I),
                 f"{pre}.input_layernorm.weight": torch.ones(H),
                 f"{pre}.post_attention_layernorm.weight": torch.ones(H),
             }
+            if cfg.num_local_experts > 0:
+                tensors[f"{pre}.block_sparse_moe.gate.weight"] = gen(
+                    f"{pre}.moe.gate", cfg.num_local_experts, H
+                )
+                for e in range(cfg.num_local_experts):
+                    ep = f"{pre}.block_sparse_moe.experts.{e}"
+                    tensors[f"{ep}.w1.weight"] = gen(f"{ep}.w1", I, H)
+                    tensors[f"{ep}.w3.weight"] = gen(f"{ep}.w3", I, H)
+                    tensors[f"{ep}.w2.weight"] = gen(f"{ep}.w2", H, I)
+            else:
+                tensors[f"{pre}.mlp.gate_proj.weight"] = gen(f"{pre}.gate", I, H)
+                tensors[f"{pre}.mlp.up_proj.weight"] = gen(f"{pre}.up", I, H)
+                tensors[f"{pre}.mlp.down_proj.weight"] = gen(f"{pre}.down", H, I)
             if cfg.attention_bias:
                 tensors[f"{pre}.self_attn.q_proj.bias"] = torch.zeros(nq * hd)
                 tensors[f"{pre}.self_attn.k_proj.bias"] = torch.zeros(nkv * hd)

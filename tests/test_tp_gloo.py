@@ -18,18 +18,18 @@ def _free_port():
         return s.getsockname()[1]
 
 
-def _tp1_reference():
+def _tp1_reference(preset="tiny"):
     from arks_amd.config import EngineConfig
     from arks_amd.engine import LLMEngine, SamplingParams
 
     torch.manual_seed(0)
     e = LLMEngine(
-        EngineConfig(preset="tiny", device="cpu", kv_cache_blocks=128, max_model_len=512)
+        EngineConfig(preset=preset, device="cpu", kv_cache_blocks=128, max_model_len=512)
     )
     return e.generate(PROMPTS, SamplingParams(max_tokens=MAX_TOKENS, ignore_eos=True))
 
 
-def _tp_worker(rank: int, world: int, port: int, q):
+def _tp_worker(rank: int, world: int, port: int, q, preset: str = "tiny"):
     os.environ["MASTER_ADDR"] = "127.0.0.1"
     os.environ["MASTER_PORT"] = str(port)
     os.environ["RANK"] = str(rank)
@@ -43,7 +43,7 @@ def _tp_worker(rank: int, world: int, port: int, q):
         torch.manual_seed(0)
         e = LLMEngine(
             EngineConfig(
-                preset="tiny", device="cpu", kv_cache_blocks=128, max_model_len=512
+                preset=preset, device="cpu", kv_cache_blocks=128, max_model_len=512
             )
         )
         out = e.generate(PROMPTS, SamplingParams(max_tokens=MAX_TOKENS, ignore_eos=True))
@@ -91,3 +91,25 @@ def test_tp2_sharded_weights_random_init_consistent():
         2 * cfg.intermediate_size,
         cfg.hidden_size,
     )
+
+
+@pytest.mark.timeout(240)
+def test_tp2_moe_expert_parallel_matches_tp1():
+    """Sparse-MoE expert parallelism: TP=2 shards whole experts per rank and
+    all-reduces partial outputs; greedy tokens must equal TP=1."""
+    ref = _tp1_reference("tiny-moe")
+
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    port = _free_port()
+    procs = [
+        ctx.Process(target=_tp_worker, args=(r, 2, port, q, "tiny-moe"))
+        for r in range(2)
+    ]
+    for p in procs:
+        p.start()
+    status, payload = q.get(timeout=200)
+    for p in procs:
+        p.join(timeout=60)
+    assert status == "ok", payload
+    assert payload == ref, f"TP=2 MoE output {payload} != TP=1 {ref}"
