@@ -53,7 +53,11 @@ class ModelRunner:
         self._graph_pool = None
         self._graph_bufs: dict | None = None
         self.use_graphs = (
-            self.device.type == "cuda" and not engine_cfg.enforce_eager
+            self.device.type == "cuda"
+            and not engine_cfg.enforce_eager
+            # sparse-MoE routing (nonzero/index_add) is data-dependent and
+            # not graph-capturable; MoE decodes run eager for now
+            and self.model_cfg.num_local_experts == 0
         )
 
     # capture sizes: padded decode batch sizes with their own graphs
