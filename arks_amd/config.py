@@ -31,6 +31,14 @@ class ModelConfig:
     # Mixtral-style sparse MoE (0 experts = dense MLP)
     num_local_experts: int = 0
     num_experts_per_tok: int = 2
+    sliding_window: int | None = None  # SWA not implemented: must be None
+
+    def __post_init__(self):
+        if self.sliding_window is not None:
+            raise ValueError(
+                "sliding-window attention is not supported yet; this config "
+                f"sets sliding_window={self.sliding_window}"
+            )
     tie_word_embeddings: bool = False
     attention_bias: bool = True  # qwen2 has qkv bias; llama does not
     eos_token_id: int = 151645
@@ -59,6 +67,7 @@ class ModelConfig:
             max_position_embeddings=cfg.get("max_position_embeddings", 32768),
             rope_scaling=cfg.get("rope_scaling"),
             num_local_experts=cfg.get("num_local_experts", 0),
+            sliding_window=cfg.get("sliding_window"),
             num_experts_per_tok=cfg.get("num_experts_per_tok", 2),
             tie_word_embeddings=cfg.get("tie_word_embeddings", False),
             attention_bias=cfg.get("attention_bias", arch.startswith("Qwen2")),

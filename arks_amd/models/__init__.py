@@ -9,6 +9,9 @@ _REGISTRY = {
     "Qwen2ForCausalLM": LlamaFamilyForCausalLM,
     "LlamaForCausalLM": LlamaFamilyForCausalLM,
     "MixtralForCausalLM": LlamaFamilyForCausalLM,  # sparse MoE MLP branch
+    # Mistral v0.3+ ships sliding_window=null -> plain llama arch; models
+    # that DO set a window are rejected at config load (no SWA kernels yet)
+    "MistralForCausalLM": LlamaFamilyForCausalLM,
 }
 
 
