@@ -187,6 +187,23 @@ PRESET_CONFIGS: dict[str, ModelConfig] = {
         eos_token_id=2,
         bos_token_id=1,
     ),
+    "tiny-moe-gpu": ModelConfig(  # GPU-test-sized sparse MoE (head_dim 128)
+        architecture="MixtralForCausalLM",
+        vocab_size=2048,
+        hidden_size=512,
+        intermediate_size=256,
+        num_hidden_layers=2,
+        num_attention_heads=4,
+        num_key_value_heads=2,
+        head_dim=128,
+        rope_theta=10000.0,
+        max_position_embeddings=4096,
+        attention_bias=False,
+        num_local_experts=4,
+        num_experts_per_tok=2,
+        eos_token_id=2,
+        bos_token_id=1,
+    ),
     "tiny": ModelConfig(  # CPU-test-sized
         architecture="Qwen2ForCausalLM",
         vocab_size=512,

@@ -126,7 +126,7 @@ def test_disagg_kv_transfer_gpu():
 def test_moe_engine_gpu():
     """Sparse-MoE model end-to-end on HW (router + experts + our kernels)."""
     e = LLMEngine(EngineConfig(
-        preset="tiny-moe", device="cuda", kv_cache_blocks=256,
+        preset="tiny-moe-gpu", device="cuda", kv_cache_blocks=256,
         max_model_len=512, seed=2,
     ))
     out = e.generate([[5, 2, 8, 1], [9] * 7],
