@@ -53,11 +53,7 @@ class ModelRunner:
         self._graph_pool = None
         self._graph_bufs: dict | None = None
         self.use_graphs = (
-            self.device.type == "cuda"
-            and not engine_cfg.enforce_eager
-            # sparse-MoE routing (nonzero/index_add) is data-dependent and
-            # not graph-capturable; MoE decodes run eager for now
-            and self.model_cfg.num_local_experts == 0
+            self.device.type == "cuda" and not engine_cfg.enforce_eager
         )
 
     # capture sizes: padded decode batch sizes with their own graphs
@@ -242,8 +238,15 @@ class ModelRunner:
 
     # ---------------- hipGraph decode ----------------
     def _init_graph_buffers(self) -> None:
-        # round max_num_seqs up to the nearest capture size (capped at 256)
-        bmax = min(self.GRAPH_SIZES[-1], max(self.cfg.max_num_seqs, 1))
+        # round max_num_seqs up to the nearest capture size (capped at 256;
+        # MoE decode is only capture-safe on its dense path, <= 64 tokens —
+        # larger decode batches fall back to the eager sparse route)
+        cap = self.GRAPH_SIZES[-1]
+        if self.model_cfg.num_local_experts > 0:
+            from ..models.llama_family import MoEMLP
+
+            cap = MoEMLP.DENSE_TOKENS
+        bmax = min(cap, max(self.cfg.max_num_seqs, 1))
         self._bmax = next(s for s in self.GRAPH_SIZES if s >= bmax)
         mb = (self.cfg.max_model_len + self.cfg.block_size - 1) // self.cfg.block_size
         dev = self.device

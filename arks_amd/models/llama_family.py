@@ -150,6 +150,13 @@ class MoEMLP(nn.Module):
                         dtype=dtype), requires_grad=False,
         )
 
+    # Below this many tokens the dense path wins: every expert's weights are
+    # streamed from HBM regardless (tokens scatter over all experts), so
+    # computing all tokens per expert costs nothing extra in bandwidth and
+    # removes the data-dependent nonzero/index_add — making decode steps
+    # hipGraph-capturable.
+    DENSE_TOKENS = 64
+
     def forward(self, x):
         from ..parallel.comm import tp_all_reduce
 
@@ -159,6 +166,18 @@ class MoEMLP(nn.Module):
         weights, selected = probs.topk(self.top_k, dim=-1)  # [T, k]
         weights = weights / weights.sum(dim=-1, keepdim=True)
         out = torch.zeros_like(x)
+        if T <= self.DENSE_TOKENS:
+            wdense = torch.zeros(
+                T, self.num_experts, dtype=torch.float32, device=x.device
+            )
+            wdense.scatter_(1, selected, weights)
+            wdense = wdense.to(x.dtype)
+            for le in range(self.local_experts):
+                ge = self.expert_base + le
+                h = ops.silu_mul(torch.nn.functional.linear(x, self.w13[le]))
+                y = torch.nn.functional.linear(h, self.w2[le])
+                out += y * wdense[:, ge:ge + 1]
+            return tp_all_reduce(out)
         for le in range(self.local_experts):
             ge = self.expert_base + le
             tok, slot = (selected == ge).nonzero(as_tuple=True)
