@@ -18,18 +18,20 @@ def _free_port():
         return s.getsockname()[1]
 
 
-def _tp1_reference(preset="tiny"):
+def _tp1_reference(preset="tiny", quantization=None):
     from arks_amd.config import EngineConfig
     from arks_amd.engine import LLMEngine, SamplingParams
 
     torch.manual_seed(0)
     e = LLMEngine(
-        EngineConfig(preset=preset, device="cpu", kv_cache_blocks=128, max_model_len=512)
+        EngineConfig(preset=preset, device="cpu", kv_cache_blocks=128,
+                     max_model_len=512, quantization=quantization)
     )
     return e.generate(PROMPTS, SamplingParams(max_tokens=MAX_TOKENS, ignore_eos=True))
 
 
-def _tp_worker(rank: int, world: int, port: int, q, preset: str = "tiny"):
+def _tp_worker(rank: int, world: int, port: int, q, preset: str = "tiny",
+               quantization=None):
     os.environ["MASTER_ADDR"] = "127.0.0.1"
     os.environ["MASTER_PORT"] = str(port)
     os.environ["RANK"] = str(rank)
@@ -43,7 +45,8 @@ def _tp_worker(rank: int, world: int, port: int, q, preset: str = "tiny"):
         torch.manual_seed(0)
         e = LLMEngine(
             EngineConfig(
-                preset=preset, device="cpu", kv_cache_blocks=128, max_model_len=512
+                preset=preset, device="cpu", kv_cache_blocks=128,
+                max_model_len=512, quantization=quantization,
             )
         )
         out = e.generate(PROMPTS, SamplingParams(max_tokens=MAX_TOKENS, ignore_eos=True))
@@ -113,3 +116,25 @@ def test_tp2_moe_expert_parallel_matches_tp1():
         p.join(timeout=60)
     assert status == "ok", payload
     assert payload == ref, f"TP=2 MoE output {payload} != TP=1 {ref}"
+
+
+@pytest.mark.timeout(240)
+def test_tp2_fp8_matches_tp1_fp8():
+    """fp8 W8A8 with TP sharding: per-shard weight scales + bf16 all-reduce
+    must reproduce the TP=1 fp8 greedy tokens."""
+    ref = _tp1_reference("tiny", quantization="fp8")
+
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    port = _free_port()
+    procs = [
+        ctx.Process(target=_tp_worker, args=(r, 2, port, q, "tiny", "fp8"))
+        for r in range(2)
+    ]
+    for p in procs:
+        p.start()
+    status, payload = q.get(timeout=200)
+    for p in procs:
+        p.join(timeout=60)
+    assert status == "ok", payload
+    assert payload == ref, f"TP=2 fp8 output {payload} != TP=1 {ref}"
