@@ -32,6 +32,7 @@ class ModelConfig:
     num_local_experts: int = 0
     num_experts_per_tok: int = 2
     sliding_window: int | None = None  # SWA not implemented: must be None
+    qk_norm: bool = False  # Qwen3: per-head RMSNorm on q/k before RoPE
 
     def __post_init__(self):
         if self.sliding_window is not None:
@@ -66,11 +67,13 @@ class ModelConfig:
             rope_theta=cfg.get("rope_theta", 10000.0),
             max_position_embeddings=cfg.get("max_position_embeddings", 32768),
             rope_scaling=cfg.get("rope_scaling"),
+            qk_norm=arch.startswith("Qwen3"),
             num_local_experts=cfg.get("num_local_experts", 0),
             sliding_window=cfg.get("sliding_window"),
             num_experts_per_tok=cfg.get("num_experts_per_tok", 2),
             tie_word_embeddings=cfg.get("tie_word_embeddings", False),
-            attention_bias=cfg.get("attention_bias", arch.startswith("Qwen2")),
+            attention_bias=cfg.get("attention_bias", arch == "Qwen2ForCausalLM"
+                                   or arch == "Qwen2MoeForCausalLM"),
             eos_token_id=eos,
             bos_token_id=cfg.get("bos_token_id", 1),
             torch_dtype=cfg.get("torch_dtype", "bfloat16"),
@@ -184,6 +187,35 @@ PRESET_CONFIGS: dict[str, ModelConfig] = {
         attention_bias=False,
         num_local_experts=4,
         num_experts_per_tok=2,
+        eos_token_id=2,
+        bos_token_id=1,
+    ),
+    "qwen3-8b": ModelConfig(
+        architecture="Qwen3ForCausalLM",
+        vocab_size=151936,
+        hidden_size=4096,
+        intermediate_size=12288,
+        num_hidden_layers=36,
+        num_attention_heads=32,
+        num_key_value_heads=8,
+        head_dim=128,
+        rope_theta=1000000.0,
+        attention_bias=False,
+        qk_norm=True,
+    ),
+    "tiny-qwen3": ModelConfig(  # CPU-test-sized qk-norm arch
+        architecture="Qwen3ForCausalLM",
+        vocab_size=512,
+        hidden_size=128,
+        intermediate_size=256,
+        num_hidden_layers=2,
+        num_attention_heads=4,
+        num_key_value_heads=2,
+        head_dim=32,
+        rope_theta=10000.0,
+        max_position_embeddings=2048,
+        attention_bias=False,
+        qk_norm=True,
         eos_token_id=2,
         bos_token_id=1,
     ),

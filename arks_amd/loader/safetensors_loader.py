@@ -128,6 +128,9 @@ def save_random_checkpoint(cfg, out_dir: str, seed: int = 0) -> None:
             out[f"{pre}.self_attn.k_proj.bias"] = kb
             out[f"{pre}.self_attn.v_proj.bias"] = vb
         out[f"{pre}.self_attn.o_proj.weight"] = layer.self_attn.o_proj.weight.data
+        if layer.self_attn.q_norm is not None:
+            out[f"{pre}.self_attn.q_norm.weight"] = layer.self_attn.q_norm.weight.data
+            out[f"{pre}.self_attn.k_norm.weight"] = layer.self_attn.k_norm.weight.data
         if hasattr(layer.mlp, "w13"):  # sparse MoE block
             out[f"{pre}.block_sparse_moe.gate.weight"] = layer.mlp.gate.data
             I = layer.mlp.inter

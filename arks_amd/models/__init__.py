@@ -12,6 +12,7 @@ _REGISTRY = {
     # Mistral v0.3+ ships sliding_window=null -> plain llama arch; models
     # that DO set a window are rejected at config load (no SWA kernels yet)
     "MistralForCausalLM": LlamaFamilyForCausalLM,
+    "Qwen3ForCausalLM": LlamaFamilyForCausalLM,  # + per-head q/k RMSNorm
 }
 
 

@@ -63,6 +63,13 @@ class Attention(nn.Module):
             cfg.num_attention_heads * cfg.head_dim, cfg.hidden_size, bias=False,
             dtype=dtype,
         )
+        # Qwen3: per-head RMSNorm on q and k before RoPE
+        self.q_norm = (
+            RMSNorm(cfg.head_dim, cfg.rms_norm_eps, dtype) if cfg.qk_norm else None
+        )
+        self.k_norm = (
+            RMSNorm(cfg.head_dim, cfg.rms_norm_eps, dtype) if cfg.qk_norm else None
+        )
 
     def forward(self, x, batch: ForwardBatch, kv_cache) -> torch.Tensor:
         T = (x[0] if isinstance(x, tuple) else x).shape[0]
@@ -70,6 +77,12 @@ class Attention(nn.Module):
         # Strided views into the fused QKV buffer — RoPE/cache/attention
         # kernels take row strides, so no .contiguous() copies on the hot path.
         q, k, v = self.qkv_proj.split_qkv(qkv)
+        if self.q_norm is not None:
+            hd = self.head_dim
+            q = self.q_norm(
+                q.contiguous().view(-1, hd)).view(T, self.nq_local * hd)
+            k = self.k_norm(
+                k.contiguous().view(-1, hd)).view(T, self.nkv_local * hd)
         q, k = ops.rope_apply_inplace(
             batch.positions, q, k, self._cos_sin, self.head_dim
         )
@@ -316,6 +329,8 @@ class LlamaFamilyForCausalLM(nn.Module):
                 layer = self.layers[li]
                 if sub in ("input_layernorm.weight", "post_attention_layernorm.weight"):
                     put(f"layers.{li}.{sub}", w)
+                elif sub in ("self_attn.q_norm.weight", "self_attn.k_norm.weight"):
+                    put(f"layers.{li}.{sub}", w)
                 elif sub.startswith("self_attn.") and sub.split(".")[1] in (
                     "q_proj", "k_proj", "v_proj",
                 ):
@@ -427,6 +442,9 @@ class LlamaFamilyForCausalLM(nn.Module):
                 tensors[f"{pre}.mlp.gate_proj.weight"] = gen(f"{pre}.gate", I, H)
                 tensors[f"{pre}.mlp.up_proj.weight"] = gen(f"{pre}.up", I, H)
                 tensors[f"{pre}.mlp.down_proj.weight"] = gen(f"{pre}.down", H, I)
+            if cfg.qk_norm:
+                tensors[f"{pre}.self_attn.q_norm.weight"] = torch.ones(hd)
+                tensors[f"{pre}.self_attn.k_norm.weight"] = torch.ones(hd)
             if cfg.attention_bias:
                 tensors[f"{pre}.self_attn.q_proj.bias"] = torch.zeros(nq * hd)
                 tensors[f"{pre}.self_attn.k_proj.bias"] = torch.zeros(nkv * hd)
