@@ -31,6 +31,8 @@ class ModelConfig:
     # Mixtral-style sparse MoE (0 experts = dense MLP)
     num_local_experts: int = 0
     num_experts_per_tok: int = 2
+    moe_intermediate_size: int | None = None  # per-expert FFN width (MoE)
+    norm_topk_prob: bool = True  # renormalize top-k routing weights
     sliding_window: int | None = None  # SWA not implemented: must be None
     qk_norm: bool = False  # Qwen3: per-head RMSNorm on q/k before RoPE
 
@@ -68,9 +70,12 @@ class ModelConfig:
             max_position_embeddings=cfg.get("max_position_embeddings", 32768),
             rope_scaling=cfg.get("rope_scaling"),
             qk_norm=arch.startswith("Qwen3"),
-            num_local_experts=cfg.get("num_local_experts", 0),
+            num_local_experts=cfg.get("num_local_experts",
+                                      cfg.get("num_experts", 0)),
             sliding_window=cfg.get("sliding_window"),
             num_experts_per_tok=cfg.get("num_experts_per_tok", 2),
+            moe_intermediate_size=cfg.get("moe_intermediate_size"),
+            norm_topk_prob=cfg.get("norm_topk_prob", "num_local_experts" in cfg),
             tie_word_embeddings=cfg.get("tie_word_embeddings", False),
             attention_bias=cfg.get("attention_bias", arch == "Qwen2ForCausalLM"
                                    or arch == "Qwen2MoeForCausalLM"),
@@ -218,6 +223,43 @@ PRESET_CONFIGS: dict[str, ModelConfig] = {
         qk_norm=True,
         eos_token_id=2,
         bos_token_id=1,
+    ),
+    "tiny-qwen3moe": ModelConfig(  # CPU-test-sized Qwen3-MoE
+        architecture="Qwen3MoeForCausalLM",
+        vocab_size=512,
+        hidden_size=128,
+        intermediate_size=256,
+        moe_intermediate_size=96,
+        num_hidden_layers=2,
+        num_attention_heads=4,
+        num_key_value_heads=2,
+        head_dim=32,
+        rope_theta=10000.0,
+        max_position_embeddings=2048,
+        attention_bias=False,
+        qk_norm=True,
+        num_local_experts=4,
+        num_experts_per_tok=2,
+        norm_topk_prob=False,
+        eos_token_id=2,
+        bos_token_id=1,
+    ),
+    "qwen3-30b-a3b": ModelConfig(  # Qwen3-30B-A3B MoE shape
+        architecture="Qwen3MoeForCausalLM",
+        vocab_size=151936,
+        hidden_size=2048,
+        intermediate_size=6144,
+        moe_intermediate_size=768,
+        num_hidden_layers=48,
+        num_attention_heads=32,
+        num_key_value_heads=4,
+        head_dim=128,
+        rope_theta=1000000.0,
+        attention_bias=False,
+        qk_norm=True,
+        num_local_experts=128,
+        num_experts_per_tok=8,
+        norm_topk_prob=True,
     ),
     "tiny-moe-gpu": ModelConfig(  # GPU-test-sized sparse MoE (head_dim 128)
         architecture="MixtralForCausalLM",

@@ -132,13 +132,21 @@ def save_random_checkpoint(cfg, out_dir: str, seed: int = 0) -> None:
             out[f"{pre}.self_attn.q_norm.weight"] = layer.self_attn.q_norm.weight.data
             out[f"{pre}.self_attn.k_norm.weight"] = layer.self_attn.k_norm.weight.data
         if hasattr(layer.mlp, "w13"):  # sparse MoE block
-            out[f"{pre}.block_sparse_moe.gate.weight"] = layer.mlp.gate.data
             I = layer.mlp.inter
+            qstyle = cfg.architecture.startswith("Qwen3Moe")
+            gname = "mlp.gate" if qstyle else "block_sparse_moe.gate"
+            out[f"{pre}.{gname}.weight"] = layer.mlp.gate.data
             for e in range(layer.mlp.local_experts):
-                ep = f"{pre}.block_sparse_moe.experts.{e}"
-                out[f"{ep}.w1.weight"] = layer.mlp.w13.data[e, :I]
-                out[f"{ep}.w3.weight"] = layer.mlp.w13.data[e, I:]
-                out[f"{ep}.w2.weight"] = layer.mlp.w2.data[e]
+                if qstyle:
+                    ep = f"{pre}.mlp.experts.{e}"
+                    out[f"{ep}.gate_proj.weight"] = layer.mlp.w13.data[e, :I]
+                    out[f"{ep}.up_proj.weight"] = layer.mlp.w13.data[e, I:]
+                    out[f"{ep}.down_proj.weight"] = layer.mlp.w2.data[e]
+                else:
+                    ep = f"{pre}.block_sparse_moe.experts.{e}"
+                    out[f"{ep}.w1.weight"] = layer.mlp.w13.data[e, :I]
+                    out[f"{ep}.w3.weight"] = layer.mlp.w13.data[e, I:]
+                    out[f"{ep}.w2.weight"] = layer.mlp.w2.data[e]
         else:
             gu = layer.mlp.gate_up_proj.weight.data
             g, u = gu.split(gu.shape[0] // 2, dim=0)

@@ -13,6 +13,7 @@ _REGISTRY = {
     # that DO set a window are rejected at config load (no SWA kernels yet)
     "MistralForCausalLM": LlamaFamilyForCausalLM,
     "Qwen3ForCausalLM": LlamaFamilyForCausalLM,  # + per-head q/k RMSNorm
+    "Qwen3MoeForCausalLM": LlamaFamilyForCausalLM,  # qk-norm + sparse MoE
 }
 
 

@@ -149,7 +149,8 @@ class MoEMLP(nn.Module):
         self.local_experts = E // tp
         self.expert_base = get_tp_rank() * self.local_experts
         self.hidden = cfg.hidden_size
-        self.inter = cfg.intermediate_size
+        self.inter = cfg.moe_intermediate_size or cfg.intermediate_size
+        self.norm_topk = cfg.norm_topk_prob
         self.gate = nn.Parameter(
             torch.empty(E, cfg.hidden_size, dtype=dtype), requires_grad=False
         )
@@ -177,7 +178,8 @@ class MoEMLP(nn.Module):
         router_logits = torch.nn.functional.linear(x, self.gate).float()
         probs = torch.softmax(router_logits, dim=-1)
         weights, selected = probs.topk(self.top_k, dim=-1)  # [T, k]
-        weights = weights / weights.sum(dim=-1, keepdim=True)
+        if self.norm_topk:
+            weights = weights / weights.sum(dim=-1, keepdim=True)
         out = torch.zeros_like(x)
         if T <= self.DENSE_TOKENS:
             wdense = torch.zeros(
@@ -369,8 +371,22 @@ class LlamaFamilyForCausalLM(nn.Module):
                         d.clear()
                 elif sub == "mlp.down_proj.weight":
                     put(f"layers.{li}.mlp.down_proj.weight", layer.mlp.down_proj.shard(w))
-                elif sub == "block_sparse_moe.gate.weight":
+                elif sub in ("block_sparse_moe.gate.weight", "mlp.gate.weight"):
                     layer.mlp.gate.data.copy_(w.to(layer.mlp.gate.dtype))
+                elif sub.startswith("mlp.experts."):
+                    # Qwen3-MoE naming: experts.N.{gate,up,down}_proj.weight
+                    moe = layer.mlp
+                    e = int(sub.split(".")[2])
+                    which = sub.split(".")[3]
+                    le = e - moe.expert_base
+                    if 0 <= le < moe.local_experts:
+                        I = moe.inter
+                        if which == "gate_proj":
+                            moe.w13.data[le, :I].copy_(w.to(moe.w13.dtype))
+                        elif which == "up_proj":
+                            moe.w13.data[le, I:].copy_(w.to(moe.w13.dtype))
+                        elif which == "down_proj":
+                            moe.w2.data[le].copy_(w.to(moe.w2.dtype))
                 elif sub.startswith("block_sparse_moe.experts."):
                     # Mixtral expert naming: w1=gate, w3=up, w2=down.
                     # EP-sharded: only this rank's experts are kept.
@@ -430,14 +446,23 @@ class LlamaFamilyForCausalLM(nn.Module):
                 f"{pre}.post_attention_layernorm.weight": torch.ones(H),
             }
             if cfg.num_local_experts > 0:
-                tensors[f"{pre}.block_sparse_moe.gate.weight"] = gen(
+                Ie = cfg.moe_intermediate_size or I
+                qstyle = cfg.architecture.startswith("Qwen3Moe")
+                gname = "mlp.gate" if qstyle else "block_sparse_moe.gate"
+                tensors[f"{pre}.{gname}.weight"] = gen(
                     f"{pre}.moe.gate", cfg.num_local_experts, H
                 )
                 for e in range(cfg.num_local_experts):
-                    ep = f"{pre}.block_sparse_moe.experts.{e}"
-                    tensors[f"{ep}.w1.weight"] = gen(f"{ep}.w1", I, H)
-                    tensors[f"{ep}.w3.weight"] = gen(f"{ep}.w3", I, H)
-                    tensors[f"{ep}.w2.weight"] = gen(f"{ep}.w2", H, I)
+                    if qstyle:
+                        ep = f"{pre}.mlp.experts.{e}"
+                        tensors[f"{ep}.gate_proj.weight"] = gen(f"{ep}.w1", Ie, H)
+                        tensors[f"{ep}.up_proj.weight"] = gen(f"{ep}.w3", Ie, H)
+                        tensors[f"{ep}.down_proj.weight"] = gen(f"{ep}.w2", H, Ie)
+                    else:
+                        ep = f"{pre}.block_sparse_moe.experts.{e}"
+                        tensors[f"{ep}.w1.weight"] = gen(f"{ep}.w1", Ie, H)
+                        tensors[f"{ep}.w3.weight"] = gen(f"{ep}.w3", Ie, H)
+                        tensors[f"{ep}.w2.weight"] = gen(f"{ep}.w2", H, Ie)
             else:
                 tensors[f"{pre}.mlp.gate_proj.weight"] = gen(f"{pre}.gate", I, H)
                 tensors[f"{pre}.mlp.up_proj.weight"] = gen(f"{pre}.up", I, H)

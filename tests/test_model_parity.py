@@ -214,3 +214,49 @@ def test_qwen3_qk_norm_logits_match_transformers(tmp_path):
     diff = (logits - hf_logits).abs().max().item()
     assert diff < 2e-3, f"max logits diff {diff}"
     assert torch.equal(logits.argmax(-1), hf_logits.argmax(-1))
+
+
+def test_qwen3_moe_logits_match_transformers(tmp_path):
+    """Qwen3-MoE (qk-norm + sparse MoE, norm_topk_prob=False) vs transformers."""
+    from arks_amd.loader.safetensors_loader import (
+        load_model_weights,
+        save_random_checkpoint,
+    )
+
+    cfg = PRESET_CONFIGS["tiny-qwen3moe"]
+    save_random_checkpoint(cfg, str(tmp_path), seed=19)
+    ours = create_model(cfg, dtype=torch.float32)
+    load_model_weights(ours, str(tmp_path), torch.device("cpu"))
+
+    hf_cfg = transformers.Qwen3MoeConfig(
+        vocab_size=cfg.vocab_size,
+        hidden_size=cfg.hidden_size,
+        intermediate_size=cfg.intermediate_size,
+        moe_intermediate_size=cfg.moe_intermediate_size,
+        num_hidden_layers=cfg.num_hidden_layers,
+        num_attention_heads=cfg.num_attention_heads,
+        num_key_value_heads=cfg.num_key_value_heads,
+        head_dim=cfg.head_dim,
+        rms_norm_eps=cfg.rms_norm_eps,
+        rope_theta=cfg.rope_theta,
+        max_position_embeddings=cfg.max_position_embeddings,
+        num_experts=cfg.num_local_experts,
+        num_experts_per_tok=cfg.num_experts_per_tok,
+        norm_topk_prob=cfg.norm_topk_prob,
+        decoder_sparse_step=1,
+        mlp_only_layers=[],
+        tie_word_embeddings=cfg.tie_word_embeddings,
+        attention_bias=False,
+        attention_dropout=0.0,
+    )
+    hf = transformers.Qwen3MoeForCausalLM.from_pretrained(
+        str(tmp_path), config=hf_cfg, torch_dtype=torch.float32
+    )
+    hf.eval()
+    ids = [11, 6, 42, 9, 127, 8]
+    logits = forward_ours(ours, cfg, ids)
+    with torch.no_grad():
+        hf_logits = hf(torch.tensor([ids])).logits[0]
+    diff = (logits - hf_logits).abs().max().item()
+    assert diff < 2e-3, f"max logits diff {diff}"
+    assert torch.equal(logits.argmax(-1), hf_logits.argmax(-1))
