@@ -182,16 +182,23 @@ class MoEMLP(nn.Module):
             weights = weights / weights.sum(dim=-1, keepdim=True)
         out = torch.zeros_like(x)
         if T <= self.DENSE_TOKENS:
+            # Batched over experts: two strided-batch GEMMs (bmm) instead of
+            # a per-expert loop — one launch pair regardless of E (Qwen3-MoE
+            # has 128 experts), and every expert's weights stream once.
             wdense = torch.zeros(
                 T, self.num_experts, dtype=torch.float32, device=x.device
             )
             wdense.scatter_(1, selected, weights)
             wdense = wdense.to(x.dtype)
-            for le in range(self.local_experts):
-                ge = self.expert_base + le
-                h = ops.silu_mul(torch.nn.functional.linear(x, self.w13[le]))
-                y = torch.nn.functional.linear(h, self.w2[le])
-                out += y * wdense[:, ge:ge + 1]
+            E = self.local_experts
+            xb = x.unsqueeze(0).expand(E, T, self.hidden)
+            gu = torch.bmm(xb, self.w13.transpose(1, 2))  # [E, T, 2I]
+            h = ops.silu_mul(gu.reshape(E * T, 2 * self.inter))
+            y = torch.bmm(h.view(E, T, self.inter),
+                          self.w2.transpose(1, 2))  # [E, T, H]
+            wl = wdense[:, self.expert_base:self.expert_base + E]  # [T, E]
+            out = torch.einsum("eth,te->th", y.float(),
+                               wl.float()).to(x.dtype)
             return tp_all_reduce(out)
         for le in range(self.local_experts):
             ge = self.expert_base + le
