@@ -83,6 +83,7 @@ class Attention(nn.Module):
                 q.contiguous().view(-1, hd)).view(T, self.nq_local * hd)
             k = self.k_norm(
                 k.contiguous().view(-1, hd)).view(T, self.nkv_local * hd)
+            v = v.contiguous()  # reshape_and_cache wants one shared kv stride
         q, k = ops.rope_apply_inplace(
             batch.positions, q, k, self._cos_sin, self.head_dim
         )

@@ -21,6 +21,7 @@ from fastapi.responses import JSONResponse, PlainTextResponse, Response, Streami
 
 from ..engine import SamplingParams
 from .async_engine import AsyncEngine
+from .tokenizer import IncrementalDetokenizer
 from .openai_types import (
     ChatChoice,
     ChatCompletionChunk,
@@ -259,6 +260,7 @@ def create_app(engine: AsyncEngine, served_model_name: str, tokenizer,
         n_out = 0
         finish = None
         tracker = StopStringTracker(_stop_list(req.stop))
+        detok = IncrementalDetokenizer(tokenizer)
         try:
             async for out in engine.generate_stream(
             rid, token_ids, sp, prefill_addr=raw.headers.get("x-arks-prefill-addr")
@@ -267,7 +269,10 @@ def create_app(engine: AsyncEngine, served_model_name: str, tokenizer,
                     engine.abort(rid)
                     return
                 n_out += 1
-                piece, stopped = tracker.feed(tokenizer.decode([out.new_token_id]))
+                text_in = detok.feed(out.new_token_id)
+                if out.finished:
+                    text_in += detok.flush()
+                piece, stopped = tracker.feed(text_in)
                 if stopped:
                     finish = "stop"
                     engine.abort(rid)
@@ -401,6 +406,7 @@ def create_app(engine: AsyncEngine, served_model_name: str, tokenizer,
         include_usage = bool(req.stream_options and req.stream_options.include_usage)
         n_out = 0
         tracker = StopStringTracker(_stop_list(req.stop))
+        detok = IncrementalDetokenizer(tokenizer)
         try:
             async for out in engine.generate_stream(
             rid, token_ids, sp, prefill_addr=raw.headers.get("x-arks-prefill-addr")
@@ -409,7 +415,10 @@ def create_app(engine: AsyncEngine, served_model_name: str, tokenizer,
                     engine.abort(rid)
                     return
                 n_out += 1
-                piece, stopped = tracker.feed(tokenizer.decode([out.new_token_id]))
+                text_in = detok.feed(out.new_token_id)
+                if out.finished:
+                    text_in += detok.flush()
+                piece, stopped = tracker.feed(text_in)
                 if stopped:
                     finish = "stop"
                     engine.abort(rid)

@@ -72,3 +72,34 @@ def load_tokenizer(model_path: str | None, vocab_size: int, eos_token_id: int):
         except Exception:
             pass
     return ByteTokenizer(vocab_size, eos_token_id)
+
+
+class IncrementalDetokenizer:
+    """Streaming-safe detokenization: decoding token-by-token glues
+    sentencepiece/BPE pieces wrongly (missing spaces, broken multibyte
+    runes). vLLM-style offsets: decode a short suffix window and emit only
+    complete text, holding back incomplete runes (U+FFFD at the cut)."""
+
+    def __init__(self, tokenizer):
+        self.tok = tokenizer
+        self.ids: list[int] = []
+        self.prefix_offset = 0
+        self.read_offset = 0
+
+    def feed(self, token_id: int) -> str:
+        self.ids.append(token_id)
+        prefix_text = self.tok.decode(self.ids[self.prefix_offset:self.read_offset])
+        new_text = self.tok.decode(self.ids[self.prefix_offset:])
+        if len(new_text) > len(prefix_text) and not new_text.endswith("\ufffd"):
+            piece = new_text[len(prefix_text):]
+            self.prefix_offset = self.read_offset
+            self.read_offset = len(self.ids)
+            return piece
+        return ""
+
+    def flush(self) -> str:
+        """Emit whatever is still held (final chunk: incomplete runes too)."""
+        prefix_text = self.tok.decode(self.ids[self.prefix_offset:self.read_offset])
+        new_text = self.tok.decode(self.ids[self.prefix_offset:])
+        self.prefix_offset = self.read_offset = len(self.ids)
+        return new_text[len(prefix_text):]
