@@ -323,6 +323,9 @@ class EngineConfig:
     kv_cache_blocks: int | None = None  # override (else sized from free HBM)
     enforce_eager: bool = False  # disable hipGraph decode capture
     enable_prefix_caching: bool = True  # content-addressed KV block reuse
+    # mixed batching: prefill chunks share a step with in-flight decodes so
+    # long prompts never stall token streams (vLLM-v1-style scheduling)
+    enable_mixed_batching: bool = True
     # None | "fp8": W8A8 OCP-e4m3 for qkv/gate_up/down/lm_head GEMMs
     # (dynamic per-token activation scales; o_proj and KV stay bf16)
     quantization: str | None = None

@@ -28,6 +28,7 @@ class LLMEngine:
             max_num_seqs=cfg.max_num_seqs,
             max_num_batched_tokens=cfg.max_num_batched_tokens,
             max_model_len=cfg.max_model_len,
+            mixed_batching=cfg.enable_mixed_batching,
         )
         # serving metrics
         self.total_prompt_tokens = 0

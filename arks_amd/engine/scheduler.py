@@ -39,11 +39,13 @@ class Scheduler:
         max_num_seqs: int = 256,
         max_num_batched_tokens: int = 8192,
         max_model_len: int = 8192,
+        mixed_batching: bool = True,
     ):
         self.allocator = allocator
         self.max_num_seqs = max_num_seqs
         self.max_num_batched_tokens = max_num_batched_tokens
         self.max_model_len = max_model_len
+        self.mixed_batching = mixed_batching
         self.waiting: deque[Sequence] = deque()
         self.running: list[Sequence] = []
         # finished seqs with hold_pages: pages stay allocated until the
@@ -94,15 +96,39 @@ class Scheduler:
 
     # --- scheduling ---
     def schedule(self) -> ScheduledBatch | None:
+        """One step's batch. With work waiting AND sequences decoding, build
+        a MIXED batch: every running sequence contributes its next token and
+        the remaining token budget admits prefill chunks — all through the
+        paged extend path, so a long prefill never stalls in-flight decodes
+        (decode-only steps still take the hipGraph fast path)."""
+        if self.waiting and self.running and self.mixed_batching:
+            decode_part = self._schedule_decode()
+            if decode_part is not None:
+                budget_left = self.max_num_batched_tokens - len(decode_part.seqs)
+                chunk_part = (
+                    self._schedule_prefill(budget=max(budget_left, 0))
+                    if budget_left > 0 else None
+                )
+                if chunk_part is None:
+                    return decode_part
+                for seq in decode_part.seqs:
+                    seq.num_cached_tokens = seq.num_tokens - 1
+                return ScheduledBatch(
+                    seqs=decode_part.seqs + chunk_part.seqs,
+                    is_prefill=True,  # runs the paged extend path
+                    num_new_tokens=[1] * len(decode_part.seqs)
+                    + chunk_part.num_new_tokens,
+                )
         batch = self._schedule_prefill()
         if batch is not None:
             return batch
         return self._schedule_decode()
 
-    def _schedule_prefill(self) -> ScheduledBatch | None:
+    def _schedule_prefill(self, budget: int | None = None) -> ScheduledBatch | None:
         seqs: list[Sequence] = []
         ntoks: list[int] = []
-        budget = self.max_num_batched_tokens
+        if budget is None:
+            budget = self.max_num_batched_tokens
         while self.waiting and budget > 0:
             seq = self.waiting[0]
             # num_tokens (not num_prompt_tokens): a preempted sequence is

@@ -151,3 +151,31 @@ def test_decode_capped_at_max_model_len():
     out = eng.generate([[1] * 20], SamplingParams(max_tokens=50, ignore_eos=True))
     assert len(out[0]) == 4  # 20 prompt + 4 = 24 = max_model_len
     assert not eng.has_work()
+
+
+def test_mixed_batching_greedy_equivalence():
+    """Requests arriving mid-decode: mixed batching must produce the same
+    greedy tokens as strictly separated scheduling."""
+    from arks_amd.config import EngineConfig
+    from arks_amd.engine import LLMEngine, SamplingParams
+
+    def run(mixed):
+        eng = LLMEngine(EngineConfig(
+            preset="tiny", device="cpu", kv_cache_blocks=128,
+            max_model_len=256, max_num_batched_tokens=64,
+            enable_mixed_batching=mixed, seed=8,
+        ))
+        sp = SamplingParams(max_tokens=10, ignore_eos=True)
+        s1 = eng.add_request([3, 1, 4] * 6, sp, request_id="a")
+        # let the first request get into decode
+        for _ in range(3):
+            eng.step()
+        s2 = eng.add_request([2, 7] * 20, sp, request_id="b")
+        while eng.has_work():
+            eng.step()
+        return s1.output_token_ids, s2.output_token_ids
+
+    a_mixed, b_mixed = run(True)
+    a_sep, b_sep = run(False)
+    assert a_mixed == a_sep
+    assert b_mixed == b_sep

@@ -75,11 +75,13 @@ def test_scheduler_prefix_reuse_counts_new_tokens_only():
     s.add(s1)
     b1 = s.schedule()
     assert b1.is_prefill and b1.num_new_tokens == [20]
-    # same prompt again: 4 full blocks (16 tokens) cached
+    # same prompt again: 4 full blocks (16 tokens) cached. s1 is decoding,
+    # so this schedules as a MIXED batch: s1's next token + s2's new suffix.
     s2 = Sequence(prompt)
     s.add(s2)
     b2 = s.schedule()
-    assert b2.is_prefill and b2.num_new_tokens == [4]
+    assert b2.is_prefill and b2.num_new_tokens == [1, 4]
+    assert b2.seqs == [s1, s2]
     assert s2.num_cached_tokens == 16
     assert s2.block_table[:4] == s1.block_table[:4]  # shared pages
     assert s2.block_table[4] != s1.block_table[4]

@@ -36,7 +36,9 @@ def test_prefill_batches_waiting():
     assert s.num_running == 2 and s.num_waiting == 1
     b.seqs[2].num_cached_tokens = 20  # (the runner does this after the step)
     b2 = s.schedule()
-    assert b2.is_prefill and len(b2.seqs) == 1 and b2.num_new_tokens == [20]
+    # mixed batch: the two decoding seqs ride along with the chunk remainder
+    assert b2.is_prefill and len(b2.seqs) == 3
+    assert b2.num_new_tokens == [1, 1, 20]
     assert s.num_running == 3 and s.num_waiting == 0
 
 
