@@ -201,17 +201,33 @@ class MoEMLP(nn.Module):
             out = torch.einsum("eth,te->th", y.float(),
                                wl.float()).to(x.dtype)
             return tp_all_reduce(out)
+        # Sparse path (prefill-sized T): sort token-expert pairs once so
+        # each expert sees a contiguous segment — one host sync for the
+        # segment table instead of a .nonzero() sync per expert.
+        k = self.top_k
+        flat_sel = selected.reshape(-1)
+        flat_tok = torch.arange(T, device=x.device).repeat_interleave(k)
+        flat_w = weights.reshape(-1)
+        order = torch.argsort(flat_sel, stable=True)
+        tok_sorted = flat_tok[order]
+        w_sorted = flat_w[order]
+        counts = torch.bincount(flat_sel, minlength=self.num_experts)
+        offs = torch.cumsum(counts, 0)
+        counts_h = counts.cpu().tolist()  # the single sync
+        offs_h = offs.cpu().tolist()
+        x_g = x[tok_sorted]  # [T*k, H] gathered once
         for le in range(self.local_experts):
             ge = self.expert_base + le
-            tok, slot = (selected == ge).nonzero(as_tuple=True)
-            if tok.numel() == 0:
+            c = counts_h[ge]
+            if c == 0:
                 continue
+            seg = slice(offs_h[ge] - c, offs_h[ge])
             h = ops.silu_mul(
-                torch.nn.functional.linear(x[tok], self.w13[le])
+                torch.nn.functional.linear(x_g[seg], self.w13[le])
             )
             y = torch.nn.functional.linear(h, self.w2[le])
             out.index_add_(
-                0, tok, y * weights[tok, slot, None].to(y.dtype)
+                0, tok_sorted[seg], y * w_sorted[seg, None].to(y.dtype)
             )
         return tp_all_reduce(out)
 
