@@ -167,7 +167,8 @@ def test_mixtral_moe_logits_match_transformers(tmp_path):
         str(tmp_path), config=hf_cfg, torch_dtype=torch.float32
     )
     hf.eval()
-    ids = [7, 99, 3, 42, 17, 5, 200, 6]
+    # > MoEMLP.DENSE_TOKENS so this exercises the sorted sparse dispatch
+    ids = [(13 * i + 7) % 500 for i in range(80)]
     logits = forward_ours(ours, cfg, ids)
     with torch.no_grad():
         hf_logits = hf(torch.tensor([ids])).logits[0]
