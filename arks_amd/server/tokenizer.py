@@ -86,11 +86,15 @@ class IncrementalDetokenizer:
         self.prefix_offset = 0
         self.read_offset = 0
 
+    MAX_HOLD = 4  # never withhold more than this many tokens
+
     def feed(self, token_id: int) -> str:
         self.ids.append(token_id)
         prefix_text = self.tok.decode(self.ids[self.prefix_offset:self.read_offset])
         new_text = self.tok.decode(self.ids[self.prefix_offset:])
-        if len(new_text) > len(prefix_text) and not new_text.endswith("\ufffd"):
+        complete = not new_text.endswith("\ufffd")
+        overdue = len(self.ids) - self.read_offset >= self.MAX_HOLD
+        if len(new_text) > len(prefix_text) and (complete or overdue):
             piece = new_text[len(prefix_text):]
             self.prefix_offset = self.read_offset
             self.read_offset = len(self.ids)

@@ -55,10 +55,11 @@ async def one_request(client, args, rid, results):
             if not line.startswith("data: ") or line == "data: [DONE]":
                 continue
             chunk = json.loads(line[6:])
-            if chunk.get("choices"):
+            if chunk.get("choices") and chunk["choices"][0].get("text"):
                 if ttft is None:
                     ttft = time.time() - t0
-                n_tok += 1
+            if chunk.get("usage"):
+                n_tok = chunk["usage"]["completion_tokens"]
     results.append({"ttft": ttft, "e2e": time.time() - t0, "tokens": n_tok})
 
 
