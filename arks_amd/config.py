@@ -32,6 +32,7 @@ class ModelConfig:
     num_local_experts: int = 0
     num_experts_per_tok: int = 2
     moe_intermediate_size: int | None = None  # per-expert FFN width (MoE)
+    shared_expert_intermediate_size: int | None = None  # Qwen2-MoE
     norm_topk_prob: bool = True  # renormalize top-k routing weights
     sliding_window: int | None = None  # SWA not implemented: must be None
     qk_norm: bool = False  # Qwen3: per-head RMSNorm on q/k before RoPE
@@ -75,6 +76,8 @@ class ModelConfig:
             sliding_window=cfg.get("sliding_window"),
             num_experts_per_tok=cfg.get("num_experts_per_tok", 2),
             moe_intermediate_size=cfg.get("moe_intermediate_size"),
+            shared_expert_intermediate_size=cfg.get(
+                "shared_expert_intermediate_size"),
             norm_topk_prob=cfg.get("norm_topk_prob", "num_local_experts" in cfg),
             tie_word_embeddings=cfg.get("tie_word_embeddings", False),
             attention_bias=cfg.get("attention_bias", arch == "Qwen2ForCausalLM"
@@ -221,6 +224,26 @@ PRESET_CONFIGS: dict[str, ModelConfig] = {
         max_position_embeddings=2048,
         attention_bias=False,
         qk_norm=True,
+        eos_token_id=2,
+        bos_token_id=1,
+    ),
+    "tiny-qwen2moe": ModelConfig(  # CPU-test-sized Qwen2-MoE (shared expert)
+        architecture="Qwen2MoeForCausalLM",
+        vocab_size=512,
+        hidden_size=128,
+        intermediate_size=256,
+        moe_intermediate_size=96,
+        shared_expert_intermediate_size=160,
+        num_hidden_layers=2,
+        num_attention_heads=4,
+        num_key_value_heads=2,
+        head_dim=32,
+        rope_theta=10000.0,
+        max_position_embeddings=2048,
+        attention_bias=True,
+        num_local_experts=4,
+        num_experts_per_tok=2,
+        norm_topk_prob=False,
         eos_token_id=2,
         bos_token_id=1,
     ),

@@ -133,7 +133,15 @@ def save_random_checkpoint(cfg, out_dir: str, seed: int = 0) -> None:
             out[f"{pre}.self_attn.k_norm.weight"] = layer.self_attn.k_norm.weight.data
         if hasattr(layer.mlp, "w13"):  # sparse MoE block
             I = layer.mlp.inter
-            qstyle = cfg.architecture.startswith("Qwen3Moe")
+            qstyle = not cfg.architecture.startswith("MixtralFor")
+            if layer.mlp.shared is not None:
+                sp = f"{pre}.mlp.shared_expert"
+                sgu = layer.mlp.shared.gate_up_proj.weight.data
+                sg, su = sgu.split(sgu.shape[0] // 2, dim=0)
+                out[f"{sp}.gate_proj.weight"] = sg
+                out[f"{sp}.up_proj.weight"] = su
+                out[f"{sp}.down_proj.weight"] = layer.mlp.shared.down_proj.weight.data
+                out[f"{pre}.mlp.shared_expert_gate.weight"] = layer.mlp.shared_gate.data
             gname = "mlp.gate" if qstyle else "block_sparse_moe.gate"
             out[f"{pre}.{gname}.weight"] = layer.mlp.gate.data
             for e in range(layer.mlp.local_experts):

@@ -14,6 +14,7 @@ _REGISTRY = {
     "MistralForCausalLM": LlamaFamilyForCausalLM,
     "Qwen3ForCausalLM": LlamaFamilyForCausalLM,  # + per-head q/k RMSNorm
     "Qwen3MoeForCausalLM": LlamaFamilyForCausalLM,  # qk-norm + sparse MoE
+    "Qwen2MoeForCausalLM": LlamaFamilyForCausalLM,  # MoE + shared expert
 }
 
 

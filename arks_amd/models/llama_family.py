@@ -164,6 +164,21 @@ class MoEMLP(nn.Module):
             torch.empty(self.local_experts, self.hidden, self.inter,
                         dtype=dtype), requires_grad=False,
         )
+        # Qwen2-MoE shared expert (dense, TP-sharded like a normal MLP) with
+        # a sigmoid scalar gate
+        self.shared = None
+        self.shared_gate = None
+        if cfg.shared_expert_intermediate_size:
+            import dataclasses
+
+            dense_cfg = dataclasses.replace(
+                cfg, intermediate_size=cfg.shared_expert_intermediate_size
+            )
+            self.shared = MLP(dense_cfg, dtype)
+            self.shared_gate = nn.Parameter(
+                torch.zeros(1, cfg.hidden_size, dtype=dtype),
+                requires_grad=False,
+            )
 
     # Below this many tokens the dense path wins: every expert's weights are
     # streamed from HBM regardless (tokens scatter over all experts), so
@@ -200,7 +215,13 @@ class MoEMLP(nn.Module):
             wl = wdense[:, self.expert_base:self.expert_base + E]  # [T, E]
             out = torch.einsum("eth,te->th", y.float(),
                                wl.float()).to(x.dtype)
-            return tp_all_reduce(out)
+            out = tp_all_reduce(out)
+            if self.shared is not None:
+                gate = torch.sigmoid(
+                    torch.nn.functional.linear(x, self.shared_gate).float()
+                ).to(x.dtype)
+                out = out + gate * self.shared(x)
+            return out
         # Sparse path (prefill-sized T): sort token-expert pairs once so
         # each expert sees a contiguous segment — one host sync for the
         # segment table instead of a .nonzero() sync per expert.
@@ -229,7 +250,13 @@ class MoEMLP(nn.Module):
             out.index_add_(
                 0, tok_sorted[seg], y * w_sorted[seg, None].to(y.dtype)
             )
-        return tp_all_reduce(out)
+        out = tp_all_reduce(out)
+        if self.shared is not None:
+            gate = torch.sigmoid(
+                torch.nn.functional.linear(x, self.shared_gate).float()
+            ).to(x.dtype)
+            out = out + gate * self.shared(x)
+        return out
 
 
 class DecoderLayer(nn.Module):
@@ -395,6 +422,25 @@ class LlamaFamilyForCausalLM(nn.Module):
                         d.clear()
                 elif sub == "mlp.down_proj.weight":
                     put(f"layers.{li}.mlp.down_proj.weight", layer.mlp.down_proj.shard(w))
+                elif sub == "mlp.shared_expert_gate.weight":
+                    layer.mlp.shared_gate.data.copy_(
+                        w.to(layer.mlp.shared_gate.dtype))
+                elif sub in ("mlp.shared_expert.gate_proj.weight",
+                             "mlp.shared_expert.up_proj.weight"):
+                    pending_mlp.setdefault((li, "se"), {})[sub] = w
+                    d = pending_mlp[(li, "se")]
+                    if len(d) == 2:
+                        fused = layer.mlp.shared.gate_up_proj.shard_merged(
+                            d["mlp.shared_expert.gate_proj.weight"],
+                            d["mlp.shared_expert.up_proj.weight"],
+                        )
+                        layer.mlp.shared.gate_up_proj.weight.data.copy_(
+                            fused.to(layer.mlp.shared.gate_up_proj.weight.dtype))
+                        d.clear()
+                elif sub == "mlp.shared_expert.down_proj.weight":
+                    layer.mlp.shared.down_proj.weight.data.copy_(
+                        layer.mlp.shared.down_proj.shard(w).to(
+                            layer.mlp.shared.down_proj.weight.dtype))
                 elif sub in ("block_sparse_moe.gate.weight", "mlp.gate.weight"):
                     layer.mlp.gate.data.copy_(w.to(layer.mlp.gate.dtype))
                 elif sub.startswith("mlp.experts."):
@@ -471,7 +517,16 @@ class LlamaFamilyForCausalLM(nn.Module):
             }
             if cfg.num_local_experts > 0:
                 Ie = cfg.moe_intermediate_size or I
-                qstyle = cfg.architecture.startswith("Qwen3Moe")
+                if cfg.shared_expert_intermediate_size:
+                    Is = cfg.shared_expert_intermediate_size
+                    sp = f"{pre}.mlp.shared_expert"
+                    tensors[f"{sp}.gate_proj.weight"] = gen(f"{sp}.g", Is, H)
+                    tensors[f"{sp}.up_proj.weight"] = gen(f"{sp}.u", Is, H)
+                    tensors[f"{sp}.down_proj.weight"] = gen(f"{sp}.d", H, Is)
+                    tensors[f"{pre}.mlp.shared_expert_gate.weight"] = gen(
+                        f"{sp}.sg", 1, H
+                    )
+                qstyle = not cfg.architecture.startswith("MixtralFor")
                 gname = "mlp.gate" if qstyle else "block_sparse_moe.gate"
                 tensors[f"{pre}.{gname}.weight"] = gen(
                     f"{pre}.moe.gate", cfg.num_local_experts, H

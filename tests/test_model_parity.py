@@ -261,3 +261,49 @@ def test_qwen3_moe_logits_match_transformers(tmp_path):
     diff = (logits - hf_logits).abs().max().item()
     assert diff < 2e-3, f"max logits diff {diff}"
     assert torch.equal(logits.argmax(-1), hf_logits.argmax(-1))
+
+
+def test_qwen2_moe_shared_expert_logits_match_transformers(tmp_path):
+    """Qwen2-MoE: sparse experts + sigmoid-gated shared expert."""
+    from arks_amd.loader.safetensors_loader import (
+        load_model_weights,
+        save_random_checkpoint,
+    )
+
+    cfg = PRESET_CONFIGS["tiny-qwen2moe"]
+    save_random_checkpoint(cfg, str(tmp_path), seed=23)
+    ours = create_model(cfg, dtype=torch.float32)
+    load_model_weights(ours, str(tmp_path), torch.device("cpu"))
+
+    hf_cfg = transformers.Qwen2MoeConfig(
+        vocab_size=cfg.vocab_size,
+        hidden_size=cfg.hidden_size,
+        intermediate_size=cfg.intermediate_size,
+        moe_intermediate_size=cfg.moe_intermediate_size,
+        shared_expert_intermediate_size=cfg.shared_expert_intermediate_size,
+        num_hidden_layers=cfg.num_hidden_layers,
+        num_attention_heads=cfg.num_attention_heads,
+        num_key_value_heads=cfg.num_key_value_heads,
+        rms_norm_eps=cfg.rms_norm_eps,
+        rope_theta=cfg.rope_theta,
+        max_position_embeddings=cfg.max_position_embeddings,
+        num_experts=cfg.num_local_experts,
+        num_experts_per_tok=cfg.num_experts_per_tok,
+        norm_topk_prob=cfg.norm_topk_prob,
+        decoder_sparse_step=1,
+        mlp_only_layers=[],
+        tie_word_embeddings=cfg.tie_word_embeddings,
+        attention_dropout=0.0,
+    )
+    hf = transformers.Qwen2MoeForCausalLM.from_pretrained(
+        str(tmp_path), config=hf_cfg, torch_dtype=torch.float32
+    )
+    hf.eval()
+    # both dense (<=64) and sparse (>64) dispatch paths
+    for ids in ([5, 77, 2, 30, 9], [(11 * i + 3) % 500 for i in range(80)]):
+        logits = forward_ours(ours, cfg, ids)
+        with torch.no_grad():
+            hf_logits = hf(torch.tensor([ids])).logits[0]
+        diff = (logits - hf_logits).abs().max().item()
+        assert diff < 2e-3, f"max logits diff {diff}"
+        assert torch.equal(logits.argmax(-1), hf_logits.argmax(-1))
