@@ -237,19 +237,49 @@ class MoEMLP(nn.Module):
         counts_h = counts.cpu().tolist()  # the single sync
         offs_h = offs.cpu().tolist()
         x_g = x[tok_sorted]  # [T*k, H] gathered once
-        for le in range(self.local_experts):
-            ge = self.expert_base + le
-            c = counts_h[ge]
-            if c == 0:
-                continue
-            seg = slice(offs_h[ge] - c, offs_h[ge])
-            h = ops.silu_mul(
-                torch.nn.functional.linear(x_g[seg], self.w13[le])
-            )
-            y = torch.nn.functional.linear(h, self.w2[le])
-            out.index_add_(
-                0, tok_sorted[seg], y * w_sorted[seg, None].to(y.dtype)
-            )
+        El = self.local_experts
+        base = self.expert_base
+        cap = max(counts_h[base:base + El] or [0])
+        if cap > 0 and El * cap <= 4 * flat_sel.numel():
+            # Grouped: pad each local expert's segment to `cap` rows and run
+            # TWO strided-batch GEMMs for the whole block (384 segment GEMMs
+            # per Qwen3-MoE layer otherwise). Padding rows index row 0 with
+            # weight 0, so they are compute-only noise.
+            local_counts = counts[base:base + El, None]        # [El, 1]
+            local_starts = (offs[base:base + El] -
+                            counts[base:base + El])[:, None]   # [El, 1]
+            ar = torch.arange(cap, device=x.device)[None, :]   # [1, cap]
+            valid = ar < local_counts
+            idx = torch.where(valid, local_starts + ar,
+                              torch.zeros_like(ar))
+            flat = idx.reshape(-1)
+            wpad = torch.where(valid, w_sorted[flat].view(El, cap),
+                               torch.zeros(1, dtype=w_sorted.dtype,
+                                           device=x.device))
+            tpad = torch.where(valid, tok_sorted[flat].view(El, cap),
+                               torch.zeros(1, dtype=torch.long,
+                                           device=x.device))
+            xp = x_g[flat].view(El, cap, self.hidden)
+            gu = torch.bmm(xp, self.w13.transpose(1, 2))
+            h = ops.silu_mul(gu.reshape(El * cap, 2 * self.inter))
+            y = torch.bmm(h.view(El, cap, self.inter),
+                          self.w2.transpose(1, 2))  # [El, cap, H]
+            y = y * wpad[..., None].to(y.dtype)
+            out.index_add_(0, tpad.reshape(-1), y.reshape(El * cap, -1))
+        else:
+            for le in range(El):
+                ge = base + le
+                c = counts_h[ge]
+                if c == 0:
+                    continue
+                seg = slice(offs_h[ge] - c, offs_h[ge])
+                h = ops.silu_mul(
+                    torch.nn.functional.linear(x_g[seg], self.w13[le])
+                )
+                y = torch.nn.functional.linear(h, self.w2[le])
+                out.index_add_(
+                    0, tok_sorted[seg], y * w_sorted[seg, None].to(y.dtype)
+                )
         out = tp_all_reduce(out)
         if self.shared is not None:
             gate = torch.sigmoid(
