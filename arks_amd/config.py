@@ -352,6 +352,9 @@ class EngineConfig:
     # None | "fp8": W8A8 OCP-e4m3 for qkv/gate_up/down/lm_head GEMMs
     # (dynamic per-token activation scales; o_proj and KV stay bf16)
     quantization: str | None = None
+    # "auto" (bf16) | "fp8": e4m3 KV pages (halves the decode KV stream;
+    # scale 1.0, opt-in like vLLM's --kv-cache-dtype fp8)
+    kv_cache_dtype: str = "auto"
     device: str = "cuda"
     dtype: str = "bfloat16"
     seed: int = 0

@@ -107,6 +107,7 @@ class ModelRunner:
             max(mc.num_key_value_heads // tp, 1),
             mc.head_dim,
             self.cfg.block_size,
+            dtype_bytes=1 if self.cfg.kv_cache_dtype == "fp8" else 2,
         )
         if self.device.type == "cuda":
             free, total = torch.cuda.mem_get_info(self.device)
@@ -123,10 +124,14 @@ class ModelRunner:
         self.num_blocks = self.profile_num_blocks()
         nkv = max(mc.num_key_value_heads // tp, 1)
         shape = (self.num_blocks, nkv, self.cfg.block_size, mc.head_dim)
+        kv_dtype = (
+            torch.float8_e4m3fn if self.cfg.kv_cache_dtype == "fp8"
+            else self.dtype
+        )
         self.kv_caches = [
             (
-                torch.zeros(shape, dtype=self.dtype, device=self.device),
-                torch.zeros(shape, dtype=self.dtype, device=self.device),
+                torch.zeros(shape, dtype=kv_dtype, device=self.device),
+                torch.zeros(shape, dtype=kv_dtype, device=self.device),
             )
             for _ in range(mc.num_hidden_layers)
         ]
@@ -152,7 +157,8 @@ class ModelRunner:
         runner's pages."""
         assert kv.shape[0] == len(self.kv_caches) and kv.shape[2] == len(block_table)
         bt = torch.tensor(block_table, dtype=torch.long, device=self.device)
-        kv = kv.to(device=self.device, dtype=self.dtype, non_blocking=True)
+        kv = kv.to(device=self.device, dtype=self.kv_caches[0][0].dtype,
+                   non_blocking=True)
         for li, (kc, vc) in enumerate(self.kv_caches):
             kc[bt] = kv[li, 0]
             vc[bt] = kv[li, 1]

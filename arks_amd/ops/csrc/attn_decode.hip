@@ -21,6 +21,7 @@
 #include "common.h"
 
 #include <cfloat>
+#include <hip/hip_fp8.h>
 
 namespace arks {
 
@@ -54,7 +55,13 @@ constexpr int DEC_VT_PAD = 40;     // padded token axis in V^T / P LDS tiles
 // shfl_xor's for ALL heads at once. Wave partials merge flash-style in LDS
 // (the V/P tiles are re-used as the merge buffer), and the partition
 // output/partial format is unchanged from v2 (combine kernel below).
-template <int HEAD_DIM, int GQ>
+__device__ __forceinline__ uint16_t dec_fp8_to_bf16(uint8_t b) {
+  __hip_fp8_e4m3 q;
+  q.__x = b;
+  return float_to_bf16_bits((float)q);
+}
+
+template <int HEAD_DIM, int GQ, bool KV_FP8>
 __global__ __launch_bounds__(256) void attn_decode_kernel(
     bf16* __restrict__ out,        // [S, Hq, D] (used when gridDim.z == 1)
     float* __restrict__ part_out,  // [S, Hkv, P, GQ, D+2] (when gridDim.z > 1)
@@ -145,10 +152,18 @@ __global__ __launch_bounds__(256) void attn_decode_kernel(
 #pragma unroll
       for (int st = 0; st < STEPS; ++st) {
         ushort8 u{};
-        if (sub == 0 || have_p1)
-          u = *reinterpret_cast<const ushort8*>(
-              k_cache + pbase + (int64_t)lq * HEAD_DIM + st * 32 +
-              dec_frag_k(la, 0));
+        if (sub == 0 || have_p1) {
+          const int64_t src =
+              pbase + (int64_t)lq * HEAD_DIM + st * 32 + dec_frag_k(la, 0);
+          if constexpr (KV_FP8) {
+            uchar8 u8 = *reinterpret_cast<const uchar8*>(
+                reinterpret_cast<const uint8_t*>(k_cache) + src);
+#pragma unroll
+            for (int e = 0; e < 8; ++e) u[e] = dec_fp8_to_bf16(u8[e]);
+          } else {
+            u = *reinterpret_cast<const ushort8*>(k_cache + src);
+          }
+        }
         dst[sub][st] = u;
       }
     }
@@ -182,7 +197,14 @@ __global__ __launch_bounds__(256) void attn_decode_kernel(
         const int64_t src = (t < KV_BLOCK_SIZE ? pbase0 : pbase1) +
                             (int64_t)(t & (KV_BLOCK_SIZE - 1)) * HEAD_DIM +
                             col8;
-        v = *reinterpret_cast<const ushort8*>(v_cache + src);
+        if constexpr (KV_FP8) {
+          uchar8 v8 = *reinterpret_cast<const uchar8*>(
+              reinterpret_cast<const uint8_t*>(v_cache) + src);
+#pragma unroll
+          for (int e = 0; e < 8; ++e) v[e] = dec_fp8_to_bf16(v8[e]);
+        } else {
+          v = *reinterpret_cast<const ushort8*>(v_cache + src);
+        }
       }
       vv[j] = v;
     }
@@ -431,13 +453,19 @@ void launch_decode_gq(bf16* out, float* part_out, const bf16* q,
                       const bf16* kc, const bf16* vc, const int* bt,
                       const int* sl, float scale, int num_seqs,
                       int num_kv_heads, int gq, int max_blocks, int nparts,
-                      int64_t q_stride, hipStream_t stream) {
+                      int64_t q_stride, bool kv_fp8, hipStream_t stream) {
   dim3 grid(num_kv_heads, num_seqs, nparts), block(256);
 #define ARKS_CASE(G)                                                          \
   case G:                                                                     \
-    hipLaunchKernelGGL((attn_decode_kernel<HEAD_DIM, G>), grid, block, 0,     \
-                       stream, out, part_out, q, kc, vc, bt, sl, scale,       \
-                       num_kv_heads, max_blocks, q_stride);                   \
+    if (kv_fp8) {                                                             \
+      hipLaunchKernelGGL((attn_decode_kernel<HEAD_DIM, G, true>), grid,       \
+                         block, 0, stream, out, part_out, q, kc, vc, bt, sl,  \
+                         scale, num_kv_heads, max_blocks, q_stride);          \
+    } else {                                                                  \
+      hipLaunchKernelGGL((attn_decode_kernel<HEAD_DIM, G, false>), grid,      \
+                         block, 0, stream, out, part_out, q, kc, vc, bt, sl,  \
+                         scale, num_kv_heads, max_blocks, q_stride);          \
+    }                                                                         \
     break;
   switch (gq) {
     ARKS_CASE(1)
@@ -471,19 +499,20 @@ extern "C" void arks_attn_decode_paged(void* out, void* part_out, const void* q,
                                        int num_seqs, int num_q_heads,
                                        int num_kv_heads, int head_dim,
                                        int max_blocks, int nparts,
-                                       int64_t q_stride, hipStream_t stream) {
+                                       int64_t q_stride, int kv_fp8,
+                                       hipStream_t stream) {
   const int gq = num_q_heads / num_kv_heads;
   if (head_dim == 128) {
     launch_decode_gq<128>((bf16*)out, (float*)part_out, (const bf16*)q,
                           (const bf16*)k_cache, (const bf16*)v_cache,
                           (const int*)block_tables, (const int*)seq_lens,
                           scale, num_seqs, num_kv_heads, gq, max_blocks,
-                          nparts, q_stride, stream);
+                          nparts, q_stride, kv_fp8 != 0, stream);
   } else if (head_dim == 64) {
     launch_decode_gq<64>((bf16*)out, (float*)part_out, (const bf16*)q,
                          (const bf16*)k_cache, (const bf16*)v_cache,
                          (const int*)block_tables, (const int*)seq_lens, scale,
                          num_seqs, num_kv_heads, gq, max_blocks, nparts,
-                         q_stride, stream);
+                         q_stride, kv_fp8 != 0, stream);
   }
 }

@@ -19,6 +19,7 @@
 #include "common.h"
 
 #include <cfloat>
+#include <hip/hip_fp8.h>
 
 namespace arks {
 
@@ -39,7 +40,13 @@ constexpr int EXT_KTILE = 32;   // keys per LDS tile (= 2 KV pages)
 constexpr int EXT_VT_PAD = 40;  // padded VT row length
 constexpr int PAGE = 16;        // KV page (block) size in tokens
 
-template <int HEAD_DIM>
+__device__ __forceinline__ uint16_t ext_fp8_to_bf16(uint8_t b) {
+  __hip_fp8_e4m3 q;
+  q.__x = b;
+  return float_to_bf16_bits((float)q);
+}
+
+template <int HEAD_DIM, bool KV_FP8>
 __global__ __launch_bounds__(256) void attn_extend_kernel(
     bf16* __restrict__ out,            // [Tq, Hq, D]
     const bf16* __restrict__ q,        // [Tq, Hq, D] (packed new tokens)
@@ -125,8 +132,20 @@ __global__ __launch_bounds__(256) void attn_extend_kernel(
         if (kg < kmax) {
           const int64_t src = (key < PAGE ? pbase0 : pbase1) +
                               (int64_t)(kg % PAGE) * HEAD_DIM + col8;
-          kv = *reinterpret_cast<const ushort8*>(k_cache + src);
-          vv = *reinterpret_cast<const ushort8*>(v_cache + src);
+          if constexpr (KV_FP8) {
+            uchar8 k8 = *reinterpret_cast<const uchar8*>(
+                reinterpret_cast<const uint8_t*>(k_cache) + src);
+            uchar8 v8 = *reinterpret_cast<const uchar8*>(
+                reinterpret_cast<const uint8_t*>(v_cache) + src);
+#pragma unroll
+            for (int e = 0; e < 8; ++e) {
+              kv[e] = ext_fp8_to_bf16(k8[e]);
+              vv[e] = ext_fp8_to_bf16(v8[e]);
+            }
+          } else {
+            kv = *reinterpret_cast<const ushort8*>(k_cache + src);
+            vv = *reinterpret_cast<const ushort8*>(v_cache + src);
+          }
         }
         const int row_byte = col8 * 2;
         const int swz = row_byte ^ ((key & 7) << 4);
@@ -260,17 +279,31 @@ extern "C" void arks_attn_extend_paged(
     const void* block_tables, const void* kv_lens, const void* cu_seqlens_q,
     const void* tile_info, int ntiles, float scale, int num_q_heads,
     int num_kv_heads, int head_dim, int max_blocks, int64_t q_stride,
-    hipStream_t stream) {
+    int kv_fp8, hipStream_t stream) {
   dim3 grid(num_q_heads, ntiles), block(256);
-  if (head_dim == 128) {
-    hipLaunchKernelGGL((attn_extend_kernel<128>), grid, block, 0, stream,
+  if (head_dim == 128 && kv_fp8) {
+    hipLaunchKernelGGL((attn_extend_kernel<128, true>), grid, block, 0, stream,
+                       (bf16*)out, (const bf16*)q, (const bf16*)k_cache,
+                       (const bf16*)v_cache, (const int*)block_tables,
+                       (const int*)kv_lens, (const int*)cu_seqlens_q,
+                       (const int*)tile_info, scale, num_q_heads, num_kv_heads,
+                       max_blocks, q_stride);
+  } else if (head_dim == 64 && kv_fp8) {
+    hipLaunchKernelGGL((attn_extend_kernel<64, true>), grid, block, 0, stream,
+                       (bf16*)out, (const bf16*)q, (const bf16*)k_cache,
+                       (const bf16*)v_cache, (const int*)block_tables,
+                       (const int*)kv_lens, (const int*)cu_seqlens_q,
+                       (const int*)tile_info, scale, num_q_heads, num_kv_heads,
+                       max_blocks, q_stride);
+  } else if (head_dim == 128) {
+    hipLaunchKernelGGL((attn_extend_kernel<128, false>), grid, block, 0, stream,
                        (bf16*)out, (const bf16*)q, (const bf16*)k_cache,
                        (const bf16*)v_cache, (const int*)block_tables,
                        (const int*)kv_lens, (const int*)cu_seqlens_q,
                        (const int*)tile_info, scale, num_q_heads, num_kv_heads,
                        max_blocks, q_stride);
   } else if (head_dim == 64) {
-    hipLaunchKernelGGL((attn_extend_kernel<64>), grid, block, 0, stream,
+    hipLaunchKernelGGL((attn_extend_kernel<64, false>), grid, block, 0, stream,
                        (bf16*)out, (const bf16*)q, (const bf16*)k_cache,
                        (const bf16*)v_cache, (const int*)block_tables,
                        (const int*)kv_lens, (const int*)cu_seqlens_q,

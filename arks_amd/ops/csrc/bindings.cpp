@@ -25,12 +25,18 @@ void arks_reshape_and_cache(const void* k, const void* v, void* k_cache,
                             int num_tokens, int num_kv_heads, int head_dim,
                             int block_size, int64_t kv_stride,
                             hipStream_t stream);
+void arks_reshape_and_cache_fp8(const void* k, const void* v, void* k_cache,
+                                void* v_cache, const void* slot_mapping,
+                                int num_tokens, int num_kv_heads, int head_dim,
+                                int block_size, int64_t kv_stride,
+                                hipStream_t stream);
 void arks_attn_decode_paged(void* out, void* part_out, const void* q,
                             const void* k_cache, const void* v_cache,
                             const void* block_tables, const void* seq_lens,
                             float scale, int num_seqs, int num_q_heads,
                             int num_kv_heads, int head_dim, int max_blocks,
-                            int nparts, int64_t q_stride, hipStream_t stream);
+                            int nparts, int64_t q_stride, int kv_fp8,
+                            hipStream_t stream);
 void arks_attn_prefill_varlen(void* out, const void* q, const void* k,
                               const void* v, const void* cu_seqlens,
                               const void* tile_info, int ntiles, float scale,
@@ -42,7 +48,7 @@ void arks_attn_extend_paged(void* out, const void* q, const void* k_cache,
                             const void* kv_lens, const void* cu_seqlens_q,
                             const void* tile_info, int ntiles, float scale,
                             int num_q_heads, int num_kv_heads, int head_dim,
-                            int max_blocks, int64_t q_stride,
+                            int max_blocks, int64_t q_stride, int kv_fp8,
                             hipStream_t stream);
 void arks_quant_fp8_rows(void* out, void* inv_scale, const void* x, int rows,
                          int cols, hipStream_t stream);
@@ -69,6 +75,15 @@ namespace {
 
 hipStream_t current_stream() {
   return c10::hip::getCurrentHIPStream().stream();
+}
+
+// KV caches may be bf16 or fp8 e4m3 (kv_cache_dtype="fp8"); returns fp8?
+bool check_kv_cache(const torch::Tensor& t, const char* name) {
+  TORCH_CHECK(t.is_cuda(), name, " must be on GPU");
+  TORCH_CHECK(t.is_contiguous(), name, " must be contiguous");
+  if (t.scalar_type() == torch::kFloat8_e4m3fn) return true;
+  TORCH_CHECK(t.scalar_type() == torch::kBFloat16, name, " must be bf16/fp8");
+  return false;
 }
 
 void check_bf16_contig(const torch::Tensor& t, const char* name) {
@@ -144,8 +159,8 @@ void reshape_and_cache(torch::Tensor k, torch::Tensor v, torch::Tensor k_cache,
                        torch::Tensor v_cache, torch::Tensor slot_mapping) {
   check_bf16_rowstrided(k, "k");
   check_bf16_rowstrided(v, "v");
-  check_bf16_contig(k_cache, "k_cache");
-  check_bf16_contig(v_cache, "v_cache");
+  const bool fp8 = check_kv_cache(k_cache, "k_cache");
+  check_kv_cache(v_cache, "v_cache");
   TORCH_CHECK(slot_mapping.scalar_type() == torch::kInt64);
   TORCH_CHECK(k.stride(0) == v.stride(0), "k/v must share row stride");
   const int num_tokens = k.size(0);
@@ -153,10 +168,11 @@ void reshape_and_cache(torch::Tensor k, torch::Tensor v, torch::Tensor k_cache,
   const int block_size = k_cache.size(2);
   const int head_dim = k_cache.size(3);
   TORCH_CHECK(head_dim % 8 == 0);
-  arks_reshape_and_cache(k.data_ptr(), v.data_ptr(), k_cache.data_ptr(),
-                         v_cache.data_ptr(), slot_mapping.data_ptr(),
-                         num_tokens, num_kv_heads, head_dim, block_size,
-                         k.stride(0), current_stream());
+  auto fn = fp8 ? arks_reshape_and_cache_fp8 : arks_reshape_and_cache;
+  fn(k.data_ptr(), v.data_ptr(), k_cache.data_ptr(),
+     v_cache.data_ptr(), slot_mapping.data_ptr(),
+     num_tokens, num_kv_heads, head_dim, block_size,
+     k.stride(0), current_stream());
 }
 
 void attention_decode_paged(torch::Tensor out, torch::Tensor q,
@@ -166,8 +182,8 @@ void attention_decode_paged(torch::Tensor out, torch::Tensor q,
                             int64_t nparts) {
   check_bf16_contig(out, "out");
   check_bf16_rowstrided(q, "q");
-  check_bf16_contig(k_cache, "k_cache");
-  check_bf16_contig(v_cache, "v_cache");
+  const bool kv_fp8 = check_kv_cache(k_cache, "k_cache");
+  check_kv_cache(v_cache, "v_cache");
   TORCH_CHECK(block_tables.scalar_type() == torch::kInt32);
   TORCH_CHECK(seq_lens.scalar_type() == torch::kInt32);
   const int num_seqs = q.size(0);
@@ -193,7 +209,7 @@ void attention_decode_paged(torch::Tensor out, torch::Tensor q,
                          block_tables.data_ptr(), seq_lens.data_ptr(),
                          (float)scale, num_seqs, num_q_heads, num_kv_heads,
                          head_dim, max_blocks, (int)nparts, q.stride(0),
-                         current_stream());
+                         kv_fp8 ? 1 : 0, current_stream());
 }
 
 void attention_prefill_varlen(torch::Tensor out, torch::Tensor q,
@@ -227,8 +243,8 @@ void attention_extend_paged(torch::Tensor out, torch::Tensor q,
                             double scale) {
   check_bf16_contig(out, "out");
   check_bf16_rowstrided(q, "q");
-  check_bf16_contig(k_cache, "k_cache");
-  check_bf16_contig(v_cache, "v_cache");
+  const bool kv_fp8 = check_kv_cache(k_cache, "k_cache");
+  check_kv_cache(v_cache, "v_cache");
   TORCH_CHECK(block_tables.scalar_type() == torch::kInt32);
   TORCH_CHECK(kv_lens.scalar_type() == torch::kInt32);
   TORCH_CHECK(cu_seqlens_q.scalar_type() == torch::kInt32);
@@ -246,7 +262,7 @@ void attention_extend_paged(torch::Tensor out, torch::Tensor q,
                          kv_lens.data_ptr(), cu_seqlens_q.data_ptr(),
                          tile_info.data_ptr(), ntiles, (float)scale,
                          num_q_heads, num_kv_heads, head_dim, max_blocks,
-                         q.stride(0), current_stream());
+                         q.stride(0), kv_fp8 ? 1 : 0, current_stream());
 }
 
 void skinny_gemm(torch::Tensor out, torch::Tensor part, torch::Tensor a,

@@ -86,6 +86,11 @@ __device__ __forceinline__ float block_reduce_sum(float v, float* scratch) {
 
 __host__ __forceinline__ int ceil_div(int a, int b) { return (a + b - 1) / b; }
 
+// OCP e4m3 <-> bf16 bit conversions for the fp8 KV cache (scale 1.0; the
+// e4m3 range of +-448 covers K/V magnitudes). Uses the gfx950 hardware cvt
+// via __hip_fp8_e4m3 on device.
+typedef __attribute__((ext_vector_type(8))) uint8_t uchar8;
+
 }  // namespace arks
 
 #define HIP_CHECK_KERNEL()                                    \

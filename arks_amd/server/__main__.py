@@ -35,6 +35,7 @@ def parse_args(argv=None):
     p.add_argument("--disaggregation-mode", choices=["prefill", "decode"],
                    default=None)
     p.add_argument("--quantization", choices=["fp8"], default=None)
+    p.add_argument("--kv-cache-dtype", choices=["auto", "fp8"], default="auto")
     # multi-node group flags (LWS leader/worker topology): workers join the
     # leader's torch.distributed rendezvous
     p.add_argument("--leader-address", default=None)
@@ -62,6 +63,7 @@ def build_engine_config(args):
         kv_cache_blocks=args.kv_cache_blocks,
         enforce_eager=args.enforce_eager,
         quantization=args.quantization,
+        kv_cache_dtype=args.kv_cache_dtype,
     )
 
 

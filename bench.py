@@ -39,6 +39,7 @@ def parse_args():
     # NOT the headline config (BASELINE dtype is bf16): opt-in fp8 W8A8 run,
     # reported with dtype="fp8" so it is never mistaken for the bf16 number.
     p.add_argument("--quantization", choices=["fp8"], default=None)
+    p.add_argument("--kv-cache-dtype", choices=["auto", "fp8"], default="auto")
     return p.parse_args()
 
 
@@ -83,6 +84,7 @@ def main():
         kv_cache_blocks=blocks,
         seed=args.seed,
         quantization=args.quantization,
+        kv_cache_dtype=args.kv_cache_dtype,
     )
     t_load0 = time.time()
     engine = LLMEngine(cfg)
@@ -162,7 +164,8 @@ def main():
             "higher_is_better": True,
             "scaling": "weak" if args.parallel == "dp" else "strong",
             "vs_baseline": None,
-            "dtype": args.quantization or "bf16",
+            "dtype": (args.quantization or "bf16")
+            + ("+kv8" if args.kv_cache_dtype == "fp8" else ""),
             "data": "synthetic",
             "config": {
                 "model": args.model,

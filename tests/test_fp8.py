@@ -47,3 +47,18 @@ def test_engine_fp8_cpu_runs_deterministic():
 
     a, b = run(), run()
     assert a == b and all(len(o) == 6 for o in a)
+
+
+def test_engine_fp8_kv_cache_cpu():
+    """kv_cache_dtype=fp8: e4m3 pages end to end on the CPU ref path."""
+    eng = LLMEngine(EngineConfig(
+        preset="tiny", device="cpu", kv_cache_blocks=128, max_model_len=256,
+        kv_cache_dtype="fp8", seed=12,
+    ))
+    assert eng.runner.kv_caches[0][0].dtype == torch.float8_e4m3fn
+    out = eng.generate([[5, 2, 8, 1] * 5, [7] * 30],
+                       SamplingParams(max_tokens=6, ignore_eos=True))
+    assert all(len(o) == 6 for o in out)
+    out2 = eng.generate([[5, 2, 8, 1] * 5, [7] * 30],
+                        SamplingParams(max_tokens=6, ignore_eos=True))
+    assert out == out2
