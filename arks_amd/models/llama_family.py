@@ -240,7 +240,11 @@ class MoEMLP(nn.Module):
         El = self.local_experts
         base = self.expert_base
         cap = max(counts_h[base:base + El] or [0])
-        if cap > 0 and El * cap <= 4 * flat_sel.numel():
+        # grouped only when the padded transients stay modest (the [El, cap,
+        # 2I] bmm buffer reaches GBs on wide-expert models at full prefill)
+        grouped_bytes = El * cap * 2 * self.inter * 2
+        if (cap > 0 and El * cap <= 4 * flat_sel.numel()
+                and grouped_bytes <= 512 * 1024 * 1024):
             # Grouped: pad each local expert's segment to `cap` rows and run
             # TWO strided-batch GEMMs for the whole block (384 segment GEMMs
             # per Qwen3-MoE layer otherwise). Padding rows index row 0 with
