@@ -225,6 +225,30 @@ class MoEMLP(nn.Module):
         # Sparse path (prefill-sized T): sort token-expert pairs once so
         # each expert sees a contiguous segment — one host sync for the
         # segment table instead of a .nonzero() sync per expert.
+        # KNOWN ISSUE: at El > 32 the sorted/grouped dispatch faulted on HW
+        # (qwen3-30b-a3b, GPU memory access fault — not reproduced at small
+        # expert counts, tests pass); wide-expert models take the simple
+        # per-expert route until root-caused (next-round backlog).
+        if self.local_experts > 32:
+            for le in range(self.local_experts):
+                ge = self.expert_base + le
+                tok, slot = (selected == ge).nonzero(as_tuple=True)
+                if tok.numel() == 0:
+                    continue
+                h = ops.silu_mul(
+                    torch.nn.functional.linear(x[tok], self.w13[le])
+                )
+                y = torch.nn.functional.linear(h, self.w2[le])
+                out.index_add_(
+                    0, tok, y * weights[tok, slot, None].to(y.dtype)
+                )
+            out = tp_all_reduce(out)
+            if self.shared is not None:
+                gate = torch.sigmoid(
+                    torch.nn.functional.linear(x, self.shared_gate).float()
+                ).to(x.dtype)
+                out = out + gate * self.shared(x)
+            return out
         k = self.top_k
         flat_sel = selected.reshape(-1)
         flat_tok = torch.arange(T, device=x.device).repeat_interleave(k)
