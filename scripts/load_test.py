@@ -18,7 +18,6 @@ import asyncio
 import json
 import os
 import random
-import statistics
 import subprocess
 import sys
 import time
