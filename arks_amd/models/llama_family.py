@@ -12,6 +12,7 @@ with one RCCL all-reduce per attention and per MLP block.
 from __future__ import annotations
 
 import math
+import os
 
 import torch
 import torch.nn as nn
@@ -229,7 +230,9 @@ class MoEMLP(nn.Module):
         # (qwen3-30b-a3b, GPU memory access fault — not reproduced at small
         # expert counts, tests pass); wide-expert models take the simple
         # per-expert route until root-caused (next-round backlog).
-        if self.local_experts > 32:
+        # ARKS_MOE_WIDE_GROUPED=1 forces the sorted path for fault hunting.
+        if (self.local_experts > 32
+                and os.environ.get("ARKS_MOE_WIDE_GROUPED", "0") != "1"):
             for le in range(self.local_experts):
                 ge = self.expert_base + le
                 tok, slot = (selected == ge).nonzero(as_tuple=True)
