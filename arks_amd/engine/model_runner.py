@@ -18,7 +18,7 @@ import torch
 
 from .. import ops
 from ..config import EngineConfig, ModelConfig
-from ..parallel.comm import get_tp_world_size
+from ..parallel.comm import get_tp_rank, get_tp_world_size, tp_all_gather
 from .forward_batch import ForwardBatch
 from .kv_cache import BlockAllocator, PrefixCachingAllocator, kv_cache_block_bytes
 from .scheduler import ScheduledBatch
@@ -150,12 +150,24 @@ class ModelRunner:
         arksdisaggregatedapplication_controller.go:1672-1724)."""
         bt = torch.tensor(block_table, dtype=torch.long, device=self.device)
         layers = [torch.stack((kc[bt], vc[bt]), 0) for kc, vc in self.kv_caches]
-        return torch.stack(layers, 0).cpu()
+        kv = torch.stack(layers, 0)
+        # TP>1: each rank holds a contiguous nkv/tp head slice — all-gather
+        # to the full head set so the wire format is TP-degree independent
+        # (prefill TP degree need not match the decode instance's).
+        kv = tp_all_gather(kv, dim=3)
+        return kv.cpu()
 
     def inject_kv(self, block_table: list[int], kv: torch.Tensor) -> None:
-        """Scatter a transferred KV tensor (extract_kv layout) into this
-        runner's pages."""
+        """Scatter a transferred KV tensor (extract_kv layout, full head
+        set) into this runner's pages — each TP rank takes its head slice."""
+        nkv_local = self.kv_caches[0][0].shape[1]
+        world, rank = get_tp_world_size(), get_tp_rank()
         assert kv.shape[0] == len(self.kv_caches) and kv.shape[2] == len(block_table)
+        assert kv.shape[3] == nkv_local * world, (
+            f"KV head count {kv.shape[3]} != {nkv_local}*tp{world}"
+        )
+        if world > 1:
+            kv = kv[:, :, :, rank * nkv_local:(rank + 1) * nkv_local]
         bt = torch.tensor(block_table, dtype=torch.long, device=self.device)
         kv = kv.to(device=self.device, dtype=self.kv_caches[0][0].dtype,
                    non_blocking=True)

@@ -97,3 +97,22 @@ def tp_broadcast_object(obj, src: int = 0):
     buf = [obj]
     dist.broadcast_object_list(buf, src=src, group=_TP_GROUP)
     return buf[0]
+
+
+def tp_broadcast_tensor(x: torch.Tensor | None, src: int = 0) -> torch.Tensor:
+    """Broadcast a tensor from `src` (shape/dtype via a small object
+    broadcast, payload via dist.broadcast on the backend's device — avoids
+    pickling multi-GB KV pages). Non-src ranks pass None. No-op at TP=1."""
+    if _TP_WORLD <= 1:
+        return x
+    meta = None
+    if _TP_RANK == src:
+        meta = (tuple(x.shape), str(x.dtype).removeprefix("torch."))
+    meta = tp_broadcast_object(meta, src)
+    dev = "cuda" if dist.get_backend(_TP_GROUP) == "nccl" else "cpu"
+    if _TP_RANK == src:
+        t = x.to(dev).contiguous()
+    else:
+        t = torch.empty(meta[0], dtype=getattr(torch, meta[1]), device=dev)
+    dist.broadcast(t, src=src, group=_TP_GROUP)
+    return t

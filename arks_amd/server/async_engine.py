@@ -17,7 +17,11 @@ from dataclasses import dataclass, field
 
 from ..config import EngineConfig
 from ..engine import LLMEngine, SamplingParams, StepOutput
-from ..parallel.comm import get_tp_world_size, tp_broadcast_object
+from ..parallel.comm import (
+    get_tp_world_size,
+    tp_broadcast_object,
+    tp_broadcast_tensor,
+)
 from .metrics import EngineMetrics
 
 
@@ -54,6 +58,19 @@ def worker_loop(cfg: EngineConfig) -> None:
         msg = tp_broadcast_object(None, src=0)
         if msg is None or msg.get("stop"):
             break
+        if "extract" in msg:
+            # Disaggregated prefill: every rank joins the head all-gather
+            # inside extract_kv; only rank 0 ships the result over the wire.
+            engine.extract_prefilled(msg["extract"])
+            continue
+        if "inject" in msg:
+            m = msg["inject"]
+            kv = tp_broadcast_tensor(None, src=0)
+            engine.add_prefilled(
+                m["token_ids"], m["first_token"], kv,
+                SamplingParams(**m["sampling"]), m["request_id"],
+            )
+            continue
         apply_msg(engine, msg)
         engine.step()
 
@@ -122,9 +139,8 @@ class AsyncEngine:
     async def disagg_prefill(self, request_id: str, token_ids: list[int],
                              sampling: SamplingParams):
         """Prefill-instance side: run the prompt, sample the first token,
-        return (first_token, finish_reason, kv pages tensor)."""
-        if get_tp_world_size() > 1:
-            raise NotImplementedError("disaggregation requires TP=1 instances")
+        return (first_token, finish_reason, kv pages tensor). TP>1: workers
+        are told to join the extract (head all-gather) in lockstep."""
         import dataclasses
 
         sp = dataclasses.replace(sampling, max_tokens=1)
@@ -135,6 +151,8 @@ class AsyncEngine:
         loop = asyncio.get_running_loop()
 
         def _extract():
+            if get_tp_world_size() > 1:
+                tp_broadcast_object({"extract": request_id}, src=0)
             _, kv = self.engine.extract_prefilled(request_id)
             return kv
 
@@ -145,14 +163,21 @@ class AsyncEngine:
                             first_token: int, kv,
                             sampling: SamplingParams) -> _Stream:
         """Decode-instance side: admit a remotely prefilled sequence and
-        return its output stream (first token already queued)."""
-        if get_tp_world_size() > 1:
-            raise NotImplementedError("disaggregation requires TP=1 instances")
+        return its output stream (first token already queued). TP>1: the
+        full-head KV is broadcast; each rank injects its head slice."""
         st = _Stream()
         self.streams[request_id] = st
         loop = asyncio.get_running_loop()
 
         def _inject():
+            if get_tp_world_size() > 1:
+                tp_broadcast_object({"inject": {
+                    "request_id": request_id,
+                    "token_ids": token_ids,
+                    "first_token": first_token,
+                    "sampling": sampling.__dict__.copy(),
+                }}, src=0)
+                tp_broadcast_tensor(kv, src=0)
             return self.engine.add_prefilled(
                 token_ids, first_token, kv, sampling, request_id
             )

@@ -16,6 +16,10 @@ paged cache, and continues the decode loop (engine.add_prefilled). On one
 MI355X node the transfer rides loopback; cross-node it is bounded by the
 fabric, and the layout is a single contiguous buffer so an RDMA transport
 can replace httpx without touching the engine.
+
+TP>1 on either side: extract_kv all-gathers the KV head shards so the wire
+tensor always carries the FULL head set, and inject_kv takes each rank's
+slice — prefill and decode instances may run different TP degrees.
 """
 
 from __future__ import annotations
