@@ -110,9 +110,12 @@ def tp_broadcast_tensor(x: torch.Tensor | None, src: int = 0) -> torch.Tensor:
         meta = (tuple(x.shape), str(x.dtype).removeprefix("torch."))
     meta = tp_broadcast_object(meta, src)
     dev = "cuda" if dist.get_backend(_TP_GROUP) == "nccl" else "cpu"
+    dtype = getattr(torch, meta[1])
     if _TP_RANK == src:
         t = x.to(dev).contiguous()
     else:
-        t = torch.empty(meta[0], dtype=getattr(torch, meta[1]), device=dev)
-    dist.broadcast(t, src=src, group=_TP_GROUP)
+        t = torch.empty(meta[0], dtype=dtype, device=dev)
+    # fp8 has no collective support on gloo — ship the bytes
+    wire = t.view(torch.uint8) if dtype == torch.float8_e4m3fn else t
+    dist.broadcast(wire, src=src, group=_TP_GROUP)
     return t

@@ -29,20 +29,32 @@ from fastapi import Request, Response
 
 from ..engine import SamplingParams
 
-KV_DTYPE = torch.bfloat16
+# wire dtypes: bf16 pages (default) or fp8 e4m3 pages (kv_cache_dtype=fp8)
+_WIRE = {
+    "bfloat16": (torch.bfloat16, torch.uint16),
+    "float8_e4m3fn": (torch.float8_e4m3fn, torch.uint8),
+}
 
 
 def encode_kv(kv: torch.Tensor) -> tuple[str, bytes]:
-    """-> (shape header, raw bytes). kv must be the extract_kv host tensor."""
+    """-> (shape header incl. dtype, raw bytes) for the extract_kv host
+    tensor. Header: "<dtype>:<d0>,<d1>,..."."""
     kv = kv.contiguous()
+    name = str(kv.dtype).removeprefix("torch.")
+    _, raw = _WIRE[name]
     shape = ",".join(str(s) for s in kv.shape)
-    return shape, kv.view(torch.uint16).numpy().tobytes()
+    return f"{name}:{shape}", kv.view(raw).numpy().tobytes()
 
 
 def decode_kv(shape_header: str, body: bytes) -> torch.Tensor:
-    shape = tuple(int(s) for s in shape_header.split(","))
-    t = torch.frombuffer(bytearray(body), dtype=torch.uint16).view(shape)
-    return t.view(KV_DTYPE)
+    if ":" in shape_header:
+        name, shape_s = shape_header.split(":", 1)
+    else:  # older peers sent bare shapes (bf16 implied)
+        name, shape_s = "bfloat16", shape_header
+    dtype, raw = _WIRE[name]
+    shape = tuple(int(s) for s in shape_s.split(","))
+    t = torch.frombuffer(bytearray(body), dtype=raw).view(shape)
+    return t.view(dtype)
 
 
 def add_prefill_routes(app, engine) -> None:

@@ -49,6 +49,51 @@ def test_kv_wire_roundtrip():
     shape, body = encode_kv(kv)
     back = decode_kv(shape, body)
     assert torch.equal(kv, back)
+    # bare-shape headers from older peers imply bf16
+    legacy = decode_kv(shape.split(":", 1)[1], body)
+    assert torch.equal(kv, legacy)
+
+
+def test_kv_wire_roundtrip_fp8():
+    import torch
+
+    kv = (torch.randn(2, 2, 3, 2, 16, 8) * 4).to(torch.float8_e4m3fn)
+    shape, body = encode_kv(kv)
+    assert shape.startswith("float8_e4m3fn:")
+    back = decode_kv(shape, body)
+    assert back.dtype == torch.float8_e4m3fn
+    assert torch.equal(kv.view(torch.uint8), back.view(torch.uint8))
+
+
+def test_disagg_engine_fp8_kv_pages():
+    """fp8 KV pages end to end through extract -> wire -> inject (both
+    instances on kv_cache_dtype=fp8)."""
+    def cfg():
+        return EngineConfig(
+            preset="tiny", device="cpu", kv_cache_blocks=256,
+            max_model_len=512, seed=5, kv_cache_dtype="fp8",
+        )
+
+    prompt = [7, 3, 9, 1] * 9
+    sp = SamplingParams(max_tokens=6, ignore_eos=True)
+    mono = LLMEngine(cfg()).generate([prompt], sp)[0]
+    a, b = LLMEngine(cfg()), LLMEngine(cfg())
+    seq = a.add_request(prompt, SamplingParams(max_tokens=1, ignore_eos=True),
+                        request_id="r0", hold_pages=True)
+    while not seq.is_finished:
+        a.step()
+    _, kv = a.extract_prefilled("r0")
+    import torch
+
+    assert kv.dtype == torch.float8_e4m3fn
+    shape, body = encode_kv(kv)
+    kv2 = decode_kv(shape, body)
+    b.add_prefilled(prompt, seq.output_token_ids[0], kv2, sp, request_id="r0")
+    out = [seq.output_token_ids[0]]
+    while b.has_work():
+        for o in b.step():
+            out.append(o.new_token_id)
+    assert out == mono
 
 
 def test_disagg_engine_matches_monolithic():
