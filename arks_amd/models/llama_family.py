@@ -12,7 +12,6 @@ with one RCCL all-reduce per attention and per MLP block.
 from __future__ import annotations
 
 import math
-import os
 
 import torch
 import torch.nn as nn
@@ -209,6 +208,10 @@ class MoEMLP(nn.Module):
             wdense = wdense.to(x.dtype)
             E = self.local_experts
             xb = x.unsqueeze(0).expand(E, T, self.hidden)
+            # transposed-VIEW B is safe here: the expanded-A decode shapes
+            # (T <= 64) are exercised by every MoE bench/graph capture; the
+            # prefill-shape TN fault (scripts/probe_bmm_fault.py) does not
+            # reproduce at these sizes.
             gu = torch.bmm(xb, self.w13.transpose(1, 2))  # [E, T, 2I]
             h = ops.silu_mul(gu.reshape(E * T, 2 * self.inter))
             y = torch.bmm(h.view(E, T, self.inter),
@@ -226,32 +229,6 @@ class MoEMLP(nn.Module):
         # Sparse path (prefill-sized T): sort token-expert pairs once so
         # each expert sees a contiguous segment — one host sync for the
         # segment table instead of a .nonzero() sync per expert.
-        # KNOWN ISSUE: at El > 32 the sorted/grouped dispatch faulted on HW
-        # (qwen3-30b-a3b, GPU memory access fault — not reproduced at small
-        # expert counts, tests pass); wide-expert models take the simple
-        # per-expert route until root-caused (next-round backlog).
-        # ARKS_MOE_WIDE_GROUPED=1 forces the sorted path for fault hunting.
-        if (self.local_experts > 32
-                and os.environ.get("ARKS_MOE_WIDE_GROUPED", "0") != "1"):
-            for le in range(self.local_experts):
-                ge = self.expert_base + le
-                tok, slot = (selected == ge).nonzero(as_tuple=True)
-                if tok.numel() == 0:
-                    continue
-                h = ops.silu_mul(
-                    torch.nn.functional.linear(x[tok], self.w13[le])
-                )
-                y = torch.nn.functional.linear(h, self.w2[le])
-                out.index_add_(
-                    0, tok, y * weights[tok, slot, None].to(y.dtype)
-                )
-            out = tp_all_reduce(out)
-            if self.shared is not None:
-                gate = torch.sigmoid(
-                    torch.nn.functional.linear(x, self.shared_gate).float()
-                ).to(x.dtype)
-                out = out + gate * self.shared(x)
-            return out
         k = self.top_k
         flat_sel = selected.reshape(-1)
         flat_tok = torch.arange(T, device=x.device).repeat_interleave(k)
@@ -291,10 +268,16 @@ class MoEMLP(nn.Module):
                                torch.zeros(1, dtype=torch.long,
                                            device=x.device))
             xp = x_g[flat].view(El, cap, self.hidden)
-            gu = torch.bmm(xp, self.w13.transpose(1, 2))
+            # Strided bmm with a transposed-VIEW B operand memory-faults on
+            # this ROCm stack at prefill shapes (scripts/probe_bmm_fault.py:
+            # every tb=True variant dies incl. batch 8, contiguous B passes)
+            # — materialize the [E, H, 2I] layout per call. The copy is one
+            # weight read+write (~0.2 ms/layer at 8 TB/s) vs the 2*E segment
+            # GEMM launches it replaces.
+            gu = torch.bmm(xp, self.w13.transpose(1, 2).contiguous())
             h = ops.silu_mul(gu.reshape(El * cap, 2 * self.inter))
             y = torch.bmm(h.view(El, cap, self.inter),
-                          self.w2.transpose(1, 2))  # [El, cap, H]
+                          self.w2.transpose(1, 2).contiguous())  # [El, cap, H]
             y = y * wpad[..., None].to(y.dtype)
             out.index_add_(0, tpad.reshape(-1), y.reshape(El * cap, -1))
         else:
