@@ -307,3 +307,41 @@ def test_qwen2_moe_shared_expert_logits_match_transformers(tmp_path):
         diff = (logits - hf_logits).abs().max().item()
         assert diff < 2e-3, f"max logits diff {diff}"
         assert torch.equal(logits.argmax(-1), hf_logits.argmax(-1))
+
+
+def test_wide_expert_grouped_dispatch_matches_naive():
+    """E=40 (>32) sparse grouped dispatch — the path that memory-faulted on
+    HW with transposed-view bmm operands (scripts/probe_bmm_fault.py) — must
+    match a naive per-token fp32 loop."""
+    import dataclasses
+
+    from arks_amd.config import PRESET_CONFIGS
+    from arks_amd.models.llama_family import MoEMLP
+
+    cfg = dataclasses.replace(
+        PRESET_CONFIGS["tiny-moe"], num_local_experts=40, num_experts_per_tok=4
+    )
+    torch.manual_seed(3)
+    m = MoEMLP(cfg)
+    for p in m.parameters():
+        p.data.normal_(0, 0.05)
+    T = 100  # > DENSE_TOKENS -> sparse path; 40 experts -> grouped bmm
+    x = torch.randn(T, cfg.hidden_size, dtype=torch.bfloat16)
+    y = m(x)
+
+    # naive reference: per-token, per-selected-expert in fp32
+    logits = (x.float() @ m.gate.float().t())
+    probs = torch.softmax(logits, dim=-1)
+    w, sel = probs.topk(cfg.num_experts_per_tok, dim=-1)
+    if cfg.norm_topk_prob:
+        w = w / w.sum(-1, keepdim=True)
+    ref = torch.zeros(T, cfg.hidden_size)
+    for t in range(T):
+        for j in range(cfg.num_experts_per_tok):
+            e = int(sel[t, j])
+            gu = m.w13[e].float() @ x[t].float()
+            g, u = gu.chunk(2)
+            h = torch.nn.functional.silu(g) * u
+            ref[t] += w[t, j].float() * (m.w2[e].float() @ h)
+    rel = (y.float() - ref).abs().mean() / ref.abs().mean()
+    assert rel < 0.05, rel
