@@ -144,17 +144,25 @@ def save_random_checkpoint(cfg, out_dir: str, seed: int = 0) -> None:
                 out[f"{pre}.mlp.shared_expert_gate.weight"] = layer.mlp.shared_gate.data
             gname = "mlp.gate" if qstyle else "block_sparse_moe.gate"
             out[f"{pre}.{gname}.weight"] = layer.mlp.gate.data
+            # w13/w2 are stored pre-transposed [in, out]; export back to the
+            # HF [out, in] convention
             for e in range(layer.mlp.local_experts):
                 if qstyle:
                     ep = f"{pre}.mlp.experts.{e}"
-                    out[f"{ep}.gate_proj.weight"] = layer.mlp.w13.data[e, :I]
-                    out[f"{ep}.up_proj.weight"] = layer.mlp.w13.data[e, I:]
-                    out[f"{ep}.down_proj.weight"] = layer.mlp.w2.data[e]
+                    out[f"{ep}.gate_proj.weight"] = (
+                        layer.mlp.w13.data[e, :, :I].t().contiguous())
+                    out[f"{ep}.up_proj.weight"] = (
+                        layer.mlp.w13.data[e, :, I:].t().contiguous())
+                    out[f"{ep}.down_proj.weight"] = (
+                        layer.mlp.w2.data[e].t().contiguous())
                 else:
                     ep = f"{pre}.block_sparse_moe.experts.{e}"
-                    out[f"{ep}.w1.weight"] = layer.mlp.w13.data[e, :I]
-                    out[f"{ep}.w3.weight"] = layer.mlp.w13.data[e, I:]
-                    out[f"{ep}.w2.weight"] = layer.mlp.w2.data[e]
+                    out[f"{ep}.w1.weight"] = (
+                        layer.mlp.w13.data[e, :, :I].t().contiguous())
+                    out[f"{ep}.w3.weight"] = (
+                        layer.mlp.w13.data[e, :, I:].t().contiguous())
+                    out[f"{ep}.w2.weight"] = (
+                        layer.mlp.w2.data[e].t().contiguous())
         else:
             gu = layer.mlp.gate_up_proj.weight.data
             g, u = gu.split(gu.shape[0] // 2, dim=0)

@@ -155,13 +155,18 @@ class MoEMLP(nn.Module):
         self.gate = nn.Parameter(
             torch.empty(E, cfg.hidden_size, dtype=dtype), requires_grad=False
         )
-        # fused [gate|up] and down per local expert
+        # fused [gate|up] and down per local expert, stored PRE-TRANSPOSED
+        # ([in, out] per expert) so every MoE GEMM is an NN strided bmm /
+        # mm with contiguous operands: strided bmm with a transposed-VIEW B
+        # memory-faults on this ROCm stack at prefill shapes
+        # (scripts/probe_bmm_fault.py — all tb=True variants die, contiguous
+        # B passes), and NN needs no per-call transpose copies.
         self.w13 = nn.Parameter(
-            torch.empty(self.local_experts, 2 * self.inter, self.hidden,
+            torch.empty(self.local_experts, self.hidden, 2 * self.inter,
                         dtype=dtype), requires_grad=False,
         )
         self.w2 = nn.Parameter(
-            torch.empty(self.local_experts, self.hidden, self.inter,
+            torch.empty(self.local_experts, self.inter, self.hidden,
                         dtype=dtype), requires_grad=False,
         )
         # Qwen2-MoE shared expert (dense, TP-sharded like a normal MLP) with
@@ -208,14 +213,9 @@ class MoEMLP(nn.Module):
             wdense = wdense.to(x.dtype)
             E = self.local_experts
             xb = x.unsqueeze(0).expand(E, T, self.hidden)
-            # transposed-VIEW B is safe here: the expanded-A decode shapes
-            # (T <= 64) are exercised by every MoE bench/graph capture; the
-            # prefill-shape TN fault (scripts/probe_bmm_fault.py) does not
-            # reproduce at these sizes.
-            gu = torch.bmm(xb, self.w13.transpose(1, 2))  # [E, T, 2I]
+            gu = torch.bmm(xb, self.w13)  # [E, T, 2I]
             h = ops.silu_mul(gu.reshape(E * T, 2 * self.inter))
-            y = torch.bmm(h.view(E, T, self.inter),
-                          self.w2.transpose(1, 2))  # [E, T, H]
+            y = torch.bmm(h.view(E, T, self.inter), self.w2)  # [E, T, H]
             wl = wdense[:, self.expert_base:self.expert_base + E]  # [T, E]
             out = torch.einsum("eth,te->th", y.float(),
                                wl.float()).to(x.dtype)
@@ -268,16 +268,9 @@ class MoEMLP(nn.Module):
                                torch.zeros(1, dtype=torch.long,
                                            device=x.device))
             xp = x_g[flat].view(El, cap, self.hidden)
-            # Strided bmm with a transposed-VIEW B operand memory-faults on
-            # this ROCm stack at prefill shapes (scripts/probe_bmm_fault.py:
-            # every tb=True variant dies incl. batch 8, contiguous B passes)
-            # — materialize the [E, H, 2I] layout per call. The copy is one
-            # weight read+write (~0.2 ms/layer at 8 TB/s) vs the 2*E segment
-            # GEMM launches it replaces.
-            gu = torch.bmm(xp, self.w13.transpose(1, 2).contiguous())
+            gu = torch.bmm(xp, self.w13)
             h = ops.silu_mul(gu.reshape(El * cap, 2 * self.inter))
-            y = torch.bmm(h.view(El, cap, self.inter),
-                          self.w2.transpose(1, 2).contiguous())  # [El, cap, H]
+            y = torch.bmm(h.view(El, cap, self.inter), self.w2)  # [El, cap, H]
             y = y * wpad[..., None].to(y.dtype)
             out.index_add_(0, tpad.reshape(-1), y.reshape(El * cap, -1))
         else:
@@ -287,10 +280,8 @@ class MoEMLP(nn.Module):
                 if c == 0:
                     continue
                 seg = slice(offs_h[ge] - c, offs_h[ge])
-                h = ops.silu_mul(
-                    torch.nn.functional.linear(x_g[seg], self.w13[le])
-                )
-                y = torch.nn.functional.linear(h, self.w2[le])
+                h = ops.silu_mul(x_g[seg] @ self.w13[le])
+                y = h @ self.w2[le]
                 out.index_add_(
                     0, tok_sorted[seg], y * w_sorted[seg, None].to(y.dtype)
                 )
@@ -494,13 +485,14 @@ class LlamaFamilyForCausalLM(nn.Module):
                     which = sub.split(".")[3]
                     le = e - moe.expert_base
                     if 0 <= le < moe.local_experts:
+                        # HF [out, in] -> our pre-transposed [in, out]
                         I = moe.inter
                         if which == "gate_proj":
-                            moe.w13.data[le, :I].copy_(w.to(moe.w13.dtype))
+                            moe.w13.data[le, :, :I].copy_(w.t().to(moe.w13.dtype))
                         elif which == "up_proj":
-                            moe.w13.data[le, I:].copy_(w.to(moe.w13.dtype))
+                            moe.w13.data[le, :, I:].copy_(w.t().to(moe.w13.dtype))
                         elif which == "down_proj":
-                            moe.w2.data[le].copy_(w.to(moe.w2.dtype))
+                            moe.w2.data[le].copy_(w.t().to(moe.w2.dtype))
                 elif sub.startswith("block_sparse_moe.experts."):
                     # Mixtral expert naming: w1=gate, w3=up, w2=down.
                     # EP-sharded: only this rank's experts are kept.
@@ -509,13 +501,14 @@ class LlamaFamilyForCausalLM(nn.Module):
                     which = sub.split(".")[3]
                     le = e - moe.expert_base
                     if 0 <= le < moe.local_experts:
+                        # HF [out, in] -> our pre-transposed [in, out]
                         I = moe.inter
                         if which == "w1":
-                            moe.w13.data[le, :I].copy_(w.to(moe.w13.dtype))
+                            moe.w13.data[le, :, :I].copy_(w.t().to(moe.w13.dtype))
                         elif which == "w3":
-                            moe.w13.data[le, I:].copy_(w.to(moe.w13.dtype))
+                            moe.w13.data[le, :, I:].copy_(w.t().to(moe.w13.dtype))
                         elif which == "w2":
-                            moe.w2.data[le].copy_(w.to(moe.w2.dtype))
+                            moe.w2.data[le].copy_(w.t().to(moe.w2.dtype))
                 # rotary_emb.inv_freq etc. are ignored (recomputed)
 
     @torch.no_grad()

@@ -339,9 +339,10 @@ def test_wide_expert_grouped_dispatch_matches_naive():
     for t in range(T):
         for j in range(cfg.num_experts_per_tok):
             e = int(sel[t, j])
-            gu = m.w13[e].float() @ x[t].float()
+            # w13/w2 are stored pre-transposed [in, out]
+            gu = x[t].float() @ m.w13[e].float()
             g, u = gu.chunk(2)
             h = torch.nn.functional.silu(g) * u
-            ref[t] += w[t, j].float() * (m.w2[e].float() @ h)
+            ref[t] += w[t, j].float() * (h @ m.w2[e].float())
     rel = (y.float() - ref).abs().mean() / ref.abs().mean()
     assert rel < 0.05, rel
