@@ -20,13 +20,37 @@ def main(argv=None):
                     help="API server base URL (default: in-cluster)")
     ap.add_argument("--resync", type=float, default=5.0)
     ap.add_argument("--health-port", type=int, default=8081)
+    ap.add_argument("--leader-elect", action="store_true",
+                    help="gate reconciling on a coordination.k8s.io Lease "
+                         "(reference cmd/main.go:198-216)")
+    ap.add_argument("--leader-elect-namespace", default="arks-system")
     args = ap.parse_args(argv)
+
+    import os as _os
 
     from .kubestore import KubeStore
     from .operator import Operator
 
     store = KubeStore(api_base=args.kube_api)
     op = Operator(store)
+
+    elector = None
+    if args.leader_elect:
+        from .leaderelect import LeaderElector
+
+        identity = _os.environ.get("HOSTNAME") or f"pid-{_os.getpid()}"
+        elector = LeaderElector(store, identity,
+                                namespace=args.leader_elect_namespace)
+        print(f"waiting for leader lease as {identity!r} ...", flush=True)
+        elector.acquire()
+        print("acquired leader lease", flush=True)
+
+        def _lost():
+            print("leader lease lost — exiting for restart", flush=True)
+            _os._exit(1)  # pod restarts as a follower
+
+        threading.Thread(target=elector.run_renew, args=(_lost,),
+                         daemon=True).start()
 
     class Probe(BaseHTTPRequestHandler):
         def do_GET(self):
@@ -52,6 +76,8 @@ def main(argv=None):
     except KeyboardInterrupt:
         pass
     finally:
+        if elector is not None:
+            elector.release()
         store.stop()
         op.stop()
         health.shutdown()

@@ -41,6 +41,7 @@ API_MAP: dict[str, tuple[str, str]] = {
     "LeaderWorkerSet": ("/apis/leaderworkerset.x-k8s.io/v1", "leaderworkersets"),
     "RoleBasedGroupSet": ("/apis/workloads.x-k8s.io/v1alpha1", "rolebasedgroupsets"),
     "HTTPRoute": ("/apis/gateway.networking.k8s.io/v1", "httproutes"),
+    "Lease": ("/apis/coordination.k8s.io/v1", "leases"),
     "ArksModel": ("/apis/arks.ai/v1", "arksmodels"),
     "ArksApplication": ("/apis/arks.ai/v1", "arksapplications"),
     "ArksDisaggregatedApplication": ("/apis/arks.ai/v1",

@@ -62,3 +62,42 @@ def test_model_to_ready_and_app_to_running():
     assert store.get_opt("LeaderWorkerSet", "d", "app1") is not None or \
         store.get_opt("RoleBasedGroupSet", "d", "app1") is not None
     assert store.get_opt("Service", "d", "arks-application-app1") is not None
+
+
+def test_leader_election_lease():
+    """Two candidates on one Lease: only one leads; the other takes over
+    after expiry; renewal keeps leadership; release vacates."""
+    from arks_amd.controlplane.leaderelect import LeaderElector
+
+    fake, store = mk()
+    t = [1000.0]
+    clock = lambda: t[0]  # noqa: E731
+    a = LeaderElector(store, "op-a", lease_s=15, renew_s=5, clock=clock)
+    b = LeaderElector(store, "op-b", lease_s=15, renew_s=5, clock=clock)
+
+    assert a.try_acquire() and a.is_leader
+    assert not b.try_acquire() and not b.is_leader
+    lease = store.get_opt("Lease", "arks-system", a.name)
+    assert lease["spec"]["holderIdentity"] == "op-a"
+
+    # renewal within the window keeps op-a leading and blocks op-b
+    t[0] += 10
+    assert a.try_acquire()
+    t[0] += 10  # 10s since op-a's renewal < 15s lease
+    assert not b.try_acquire()
+
+    # op-a goes silent -> op-b steals after expiry with a transition bump
+    t[0] += 20
+    assert b.try_acquire() and b.is_leader
+    lease = store.get_opt("Lease", "arks-system", a.name)
+    assert lease["spec"]["holderIdentity"] == "op-b"
+    assert lease["spec"]["leaseTransitions"] == 1
+    # op-a notices it lost
+    assert not a.try_acquire() and not a.is_leader
+
+    # release vacates immediately; op-a re-acquires without waiting
+    b.release()
+    assert a.try_acquire()
+    lease = store.get_opt("Lease", "arks-system", a.name)
+    assert lease["spec"]["holderIdentity"] == "op-a"
+    assert lease["spec"]["leaseTransitions"] == 2
