@@ -208,3 +208,79 @@ def test_tp2_fp8_matches_tp1_fp8():
         p.join(timeout=60)
     assert status == "ok", payload
     assert payload == ref, f"TP=2 fp8 output {payload} != TP=1 {ref}"
+
+
+def _async_disagg_worker(rank: int, world: int, port: int, q):
+    """TP=2 disaggregation through the REAL serving stack: rank 0 drives an
+    AsyncEngine (broadcast protocol), rank 1 runs worker_loop — covering the
+    extract/inject worker messages end to end."""
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    os.environ["RANK"] = str(rank)
+    os.environ["WORLD_SIZE"] = str(world)
+    try:
+        import asyncio
+
+        from arks_amd.config import EngineConfig
+        from arks_amd.engine import SamplingParams
+        from arks_amd.parallel import comm
+        from arks_amd.server.async_engine import AsyncEngine, worker_loop
+
+        comm.init_tp(backend="gloo")
+        torch.manual_seed(0)
+        cfg = EngineConfig(preset="tiny", device="cpu", kv_cache_blocks=128,
+                           max_model_len=512)
+        if rank != 0:
+            worker_loop(cfg)  # returns on the driver's stop broadcast
+            comm.destroy_tp()
+            return
+
+        async def go():
+            eng = AsyncEngine(cfg, model_name="tiny")
+            await eng.start()
+            prompt = [4, 9, 2, 7, 7, 1]
+            sp = SamplingParams(max_tokens=5, ignore_eos=True)
+            # plain generate through the broadcast protocol
+            plain = []
+            async for out in eng.generate_stream("g0", list(prompt), sp):
+                plain.append(out.new_token_id)
+            # disaggregated: prefill+extract, then inject+decode
+            first, reason, kv = await eng.disagg_prefill("p0", list(prompt), sp)
+            # the inject stream queues the first token itself
+            st = await eng.disagg_inject("d0", list(prompt), first, kv, sp)
+            toks = []
+            while True:
+                out = await st.queue.get()
+                if out is None:
+                    break
+                toks.append(out.new_token_id)
+            await eng.stop()
+            return plain, toks
+
+        plain, toks = asyncio.new_event_loop().run_until_complete(go())
+        q.put(("ok", (plain, toks)))
+        comm.destroy_tp()
+    except Exception as exc:  # pragma: no cover
+        import traceback
+
+        q.put(("err", f"{exc}\n{traceback.format_exc()}"))
+
+
+@pytest.mark.timeout(240)
+def test_tp2_async_engine_disagg_worker_protocol():
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    port = _free_port()
+    procs = [
+        ctx.Process(target=_async_disagg_worker, args=(r, 2, port, q))
+        for r in range(2)
+    ]
+    for p in procs:
+        p.start()
+    status, payload = q.get(timeout=200)
+    for p in procs:
+        p.join(timeout=60)
+    assert status == "ok", payload
+    plain, toks = payload
+    assert len(plain) == 5
+    assert toks == plain, f"disagg {toks} != plain {plain}"
