@@ -355,6 +355,12 @@ class EngineConfig:
     # "auto" (bf16) | "fp8": e4m3 KV pages (halves the decode KV stream;
     # scale 1.0, opt-in like vLLM's --kv-cache-dtype fp8)
     kv_cache_dtype: str = "auto"
+    # None | "ngram": prompt-lookup speculative decoding (draft tokens from
+    # n-gram matches in the sequence's own history, verified in one extend
+    # forward; greedy-exact, applied only to temperature-0 requests without
+    # penalties/logprobs — same surface as vLLM's ngram speculator)
+    speculative: str | None = None
+    num_speculative_tokens: int = 4  # max draft length per step
     device: str = "cuda"
     dtype: str = "bfloat16"
     seed: int = 0

@@ -33,6 +33,10 @@ class LLMEngine:
         # serving metrics
         self.total_prompt_tokens = 0
         self.total_output_tokens = 0
+        # speculative decoding (prompt-lookup) stats
+        self.spec_enabled = cfg.speculative == "ngram"
+        self.spec_drafted_tokens = 0
+        self.spec_accepted_tokens = 0
 
     @property
     def prefix_cache_stats(self) -> tuple[int, int]:
@@ -114,6 +118,8 @@ class LLMEngine:
         sb = self.scheduler.schedule()
         if sb is None:
             return []
+        if self.spec_enabled and not sb.is_prefill:
+            return self._spec_step(sb)
         new_tokens = self.runner.execute(sb)
         logprobs = self.runner.take_logprobs()
         outputs: list[StepOutput] = []
@@ -145,6 +151,75 @@ class LLMEngine:
                     top_logprobs=top,
                 )
             )
+        self.scheduler.free_finished()
+        return outputs
+
+    def _spec_step(self, sb) -> list[StepOutput]:
+        """Decode step with prompt-lookup speculation: propose drafts from
+        each sequence's history, extend its KV pages to cover them, verify
+        in one forward, emit 1..k+1 tokens per seq (greedy-exact)."""
+        from .kv_cache import BlockAllocator
+        from .spec import eligible, propose_ngram
+
+        alloc = self.scheduler.allocator
+        bs = self.cfg.block_size
+        k = self.cfg.num_speculative_tokens
+        drafts: list[list[int]] = []
+        for seq in sb.seqs:
+            d: list[int] = []
+            room = self.cfg.max_model_len - seq.num_tokens
+            if room > 0 and eligible(seq):
+                d = propose_ngram(seq.all_token_ids, min(k, room))
+            if d:
+                need = (
+                    BlockAllocator.blocks_needed(seq.num_tokens + len(d), bs)
+                    - len(seq.block_table)
+                )
+                if need > 0:
+                    if alloc.can_allocate(need):
+                        seq.block_table.extend(alloc.allocate(need))
+                    else:
+                        d = []  # no KV room — plain decode this step
+            drafts.append(d)
+        if not any(drafts):
+            # nothing to verify: take the normal (hipGraph) decode path
+            emitted = [[t] for t in self.runner.execute(sb)]
+        else:
+            emitted = self.runner.execute_spec(sb, drafts)
+        logprobs = self.runner.take_logprobs()
+        outputs: list[StepOutput] = []
+        eos = self.model_cfg.eos_token_id
+        for i, (seq, toks) in enumerate(zip(sb.seqs, emitted)):
+            if seq.status is not SeqStatus.RUNNING:  # aborted mid-step
+                continue
+            self.spec_drafted_tokens += len(drafts[i])
+            if drafts[i]:
+                self.spec_accepted_tokens += len(toks) - 1
+            for tok in toks:
+                seq.append_token(tok)
+                self.total_output_tokens += 1
+                finished = seq.check_finished(eos)
+                if not finished and seq.num_tokens >= self.cfg.max_model_len:
+                    seq.finish_reason = "length"
+                    seq.status = SeqStatus.FINISHED
+                    seq.finish_time = time.time()
+                    finished = True
+                lp, top = logprobs.get(i, (None, None))
+                outputs.append(
+                    StepOutput(
+                        request_id=seq.request_id,
+                        seq_id=seq.seq_id,
+                        new_token_id=tok,
+                        finished=finished,
+                        finish_reason=seq.finish_reason,
+                        num_prompt_tokens=seq.num_prompt_tokens,
+                        num_output_tokens=seq.num_output_tokens,
+                        logprob=lp,
+                        top_logprobs=top,
+                    )
+                )
+                if finished:
+                    break
         self.scheduler.free_finished()
         return outputs
 

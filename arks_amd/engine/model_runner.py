@@ -361,6 +361,81 @@ class ModelRunner:
         logits = self.model(fb, self.kv_caches)  # [num_seqs, vocab]
         return self.sample(logits, sb.seqs)
 
+    def execute_spec(self, sb: ScheduledBatch,
+                     drafts: list[list[int]]) -> list[list[int]]:
+        """Speculative decode step: verify each sequence's draft tokens in
+        ONE extend-attention forward (row j scores position n-1+j given the
+        draft prefix) and return the emitted tokens per seq — the longest
+        agreeing draft prefix plus the model's own token at the first
+        disagreement (or the bonus token after a full match). Greedy-exact;
+        non-eligible seqs ride along with an empty draft and go through the
+        full sampler on their single row."""
+        from .spec import eligible
+
+        bs = self.cfg.block_size
+        dev = self.device
+        input_ids: list[int] = []
+        positions: list[int] = []
+        slot_mapping: list[int] = []
+        cu = [0]
+        q_lens: list[int] = []
+        kv_lens: list[int] = []
+        for seq, d in zip(sb.seqs, drafts):
+            toks = [seq.last_token()] + d
+            start = seq.num_tokens - 1
+            input_ids += toks
+            positions += range(start, start + len(toks))
+            for pos in range(start, start + len(toks)):
+                slot_mapping.append(seq.block_table[pos // bs] * bs + pos % bs)
+            cu.append(cu[-1] + len(toks))
+            q_lens.append(len(toks))
+            kv_lens.append(seq.num_tokens + len(d))
+            seq.num_cached_tokens = seq.num_tokens
+        max_blocks = max(len(s.block_table) for s in sb.seqs)
+        bt = torch.zeros((len(sb.seqs), max_blocks), dtype=torch.int32)
+        for i, seq in enumerate(sb.seqs):
+            bt[i, : len(seq.block_table)] = torch.tensor(
+                seq.block_table, dtype=torch.int32
+            )
+        fb = ForwardBatch(
+            is_prefill=True,
+            input_ids=torch.tensor(input_ids, dtype=torch.int64, device=dev),
+            positions=torch.tensor(positions, dtype=torch.int64, device=dev),
+            slot_mapping=torch.tensor(slot_mapping, dtype=torch.int64, device=dev),
+            cu_seqlens=torch.tensor(cu, dtype=torch.int32, device=dev),
+            seq_lens_list=q_lens,
+            block_tables=bt.to(dev),
+            seq_lens=torch.tensor(kv_lens, dtype=torch.int32, device=dev),
+            # logits at EVERY row, not just the last per seq
+            logits_indices=torch.arange(cu[-1], dtype=torch.int64, device=dev),
+        )
+        logits = self.model(fb, self.kv_caches)  # [rows, vocab]
+        greedy = ops.greedy_sample(logits.contiguous()).tolist()
+        emitted: list[list[int] | None] = [None] * len(sb.seqs)
+        other = [i for i, s in enumerate(sb.seqs) if not eligible(s)]
+        if other:
+            rows = torch.tensor([cu[i + 1] - 1 for i in other],
+                                dtype=torch.long, device=logits.device)
+            sub = self.sample(logits[rows], [sb.seqs[i] for i in other])
+            lp = self._last_logprobs
+            if lp:  # remap sampler sub-batch indices to batch positions
+                self._last_logprobs = {other[j]: v for j, v in lp.items()}
+            for j, i in enumerate(other):
+                emitted[i] = [sub[j]]
+        for i, (seq, d) in enumerate(zip(sb.seqs, drafts)):
+            if emitted[i] is not None:
+                continue
+            g = greedy[cu[i]:cu[i + 1]]
+            out: list[int] = []
+            for j, dt in enumerate(d):
+                out.append(g[j])
+                if g[j] != dt:
+                    break
+            else:
+                out.append(g[len(d)])  # bonus token after a full match
+            emitted[i] = out
+        return emitted  # type: ignore[return-value]
+
     def sample(self, logits: torch.Tensor, seqs: list[Sequence]) -> list[int]:
         if any(s.sampling.has_penalties for s in seqs):
             logits = self._apply_penalties(logits, seqs)

@@ -36,6 +36,9 @@ def parse_args(argv=None):
                    default=None)
     p.add_argument("--quantization", choices=["fp8"], default=None)
     p.add_argument("--kv-cache-dtype", choices=["auto", "fp8"], default="auto")
+    p.add_argument("--speculative", choices=["ngram"], default=None,
+                   help="prompt-lookup speculative decoding (greedy requests)")
+    p.add_argument("--num-speculative-tokens", type=int, default=4)
     # multi-node group flags (LWS leader/worker topology): workers join the
     # leader's torch.distributed rendezvous
     p.add_argument("--leader-address", default=None)
@@ -64,6 +67,8 @@ def build_engine_config(args):
         enforce_eager=args.enforce_eager,
         quantization=args.quantization,
         kv_cache_dtype=args.kv_cache_dtype,
+        speculative=args.speculative,
+        num_speculative_tokens=args.num_speculative_tokens,
     )
 
 

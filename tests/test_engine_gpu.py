@@ -169,3 +169,23 @@ def test_engine_fp8_kv_cache_gpu():
     # token-level agreement with bf16 KV is high on a tiny random model
     agree = sum(a == b for o1, o2 in zip(out, base) for a, b in zip(o1, o2))
     assert agree >= 8, (out, base)
+
+
+def test_spec_decode_matches_plain_gpu():
+    """n-gram speculative decoding through the native extend kernel must
+    emit exactly the plain engine's greedy tokens."""
+    prompts = [[1, 2, 3, 4] * 8, [9, 31, 7, 2, 55, 14, 3], [5, 6] * 12]
+    sp = SamplingParams(max_tokens=20, ignore_eos=True)
+
+    def run(spec):
+        torch.manual_seed(0)
+        e = LLMEngine(EngineConfig(
+            preset="tiny-gpu", device="cuda", kv_cache_blocks=512,
+            max_model_len=1024, max_num_seqs=64, speculative=spec,
+        ))
+        return e.generate(prompts, sp), e
+
+    ref, _ = run(None)
+    out, eng = run("ngram")
+    assert out == ref
+    assert eng.spec_drafted_tokens > 0
