@@ -12,6 +12,14 @@ The reference's runtime images (vLLM/SGLang) ship the same speculator as
 Opt-in via EngineConfig.speculative="ngram"; only greedy requests without
 penalties/logprobs are speculated (others decode normally in-batch), which
 keeps acceptance exact rather than distribution-approximate.
+
+Exactness caveat (same property as vLLM's greedy verify): "exact" means
+exact w.r.t. the verify forward's own logits. On GPU the verify runs the
+extend kernel while plain decode runs the graphed decode kernel — two bf16
+reduction orders — so a near-tie argmax can resolve differently between
+the speculative and plain engines; both are valid greedy outputs of the
+model. The CPU fp32 path is bitwise single-path and pins exact equality in
+tests/test_spec.py.
 """
 
 from __future__ import annotations
