@@ -270,6 +270,15 @@ class AsyncEngine:
                 self.metrics.prefix_cache_hits.inc(
                     hits - self.metrics.prefix_cache_hits._value.get()
                 )
+            if self.engine.spec_enabled:
+                self.metrics.spec_draft_tokens.inc(
+                    self.engine.spec_drafted_tokens
+                    - self.metrics.spec_draft_tokens._value.get()
+                )
+                self.metrics.spec_accepted_tokens.inc(
+                    self.engine.spec_accepted_tokens
+                    - self.metrics.spec_accepted_tokens._value.get()
+                )
 
     async def generate_stream(self, request_id: str, token_ids: list[int],
                               sampling: SamplingParams,

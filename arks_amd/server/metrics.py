@@ -66,6 +66,14 @@ class EngineMetrics:
             Counter, "vllm:prefix_cache_hits",
             "prompt tokens served from the prefix cache",
         )
+        self.spec_draft_tokens = mk(
+            Counter, "vllm:spec_decode_num_draft_tokens",
+            "speculative draft tokens proposed",
+        )
+        self.spec_accepted_tokens = mk(
+            Counter, "vllm:spec_decode_num_accepted_tokens",
+            "speculative draft tokens accepted",
+        )
 
     def render(self) -> bytes:
         return generate_latest(self.registry)
