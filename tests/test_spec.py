@@ -102,3 +102,63 @@ def test_spec_with_prefix_cache_and_mixed_batching_off():
     out = LLMEngine(_cfg(spec="ngram", enable_prefix_caching=False,
                          enable_mixed_batching=False)).generate(prompts, sp)
     assert out == ref
+
+
+def test_spec_fuzz_equivalence():
+    """Random prompt soups from a tiny alphabet (heavy repetition): spec
+    engine output must equal the plain engine's on the CPU fp32 path."""
+    import random
+
+    rng = random.Random(11)
+    for trial in range(4):
+        prompts = [
+            [rng.randrange(1, 6) for _ in range(rng.randrange(3, 40))]
+            for _ in range(rng.randrange(1, 6))
+        ]
+        sp = SamplingParams(max_tokens=rng.randrange(3, 20), ignore_eos=True)
+        torch.manual_seed(0)
+        ref = LLMEngine(_cfg()).generate(prompts, sp)
+        torch.manual_seed(0)
+        eng = LLMEngine(_cfg(spec="ngram"))
+        assert eng.generate(prompts, sp) == ref, f"trial {trial}"
+
+
+def test_spec_server_stream_e2e():
+    """HTTP SSE stream with speculation enabled: chunk tokens must add up
+    and usage must be exact even when a step emits several tokens."""
+    import asyncio
+    import json
+
+    import httpx
+
+    from arks_amd.server.api import create_app
+    from arks_amd.server.async_engine import AsyncEngine
+    from arks_amd.server.tokenizer import ByteTokenizer
+
+    cfg = _cfg(spec="ngram")
+    engine = AsyncEngine(cfg, model_name="tiny")
+    mc = cfg.model_config()
+    tok = ByteTokenizer(mc.vocab_size, mc.eos_token_id)
+    app = create_app(engine, "tiny", tok)
+
+    async def go():
+        async with app.router.lifespan_context(app):
+            rt = httpx.ASGITransport(app=app)
+            async with httpx.AsyncClient(transport=rt, base_url="http://t",
+                                         timeout=60) as client:
+                r = await client.post("/v1/completions", json={
+                    "model": "tiny",
+                    "prompt": "abab abab abab",
+                    "max_tokens": 12,
+                    "temperature": 0,
+                    "ignore_eos": True,
+                    "stream": True,
+                    "stream_options": {"include_usage": True},
+                })
+                assert r.status_code == 200, r.text
+                chunks = [json.loads(line[6:]) for line in r.text.splitlines()
+                          if line.startswith("data: ") and line != "data: [DONE]"]
+                usage = chunks[-1]["usage"]
+                assert usage["completion_tokens"] == 12
+
+    asyncio.new_event_loop().run_until_complete(go())
