@@ -9,7 +9,6 @@ from __future__ import annotations
 
 import queue
 import threading
-import time
 
 from ..crd.types import served_model_name
 from .reconcilers import (

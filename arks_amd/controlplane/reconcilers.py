@@ -31,7 +31,7 @@ from ..crd.types import (
     set_condition,
 )
 from . import manifests
-from .store import NotFound, Store
+from .store import Store
 
 
 def _has_finalizer(meta, fin: str) -> bool:

@@ -29,7 +29,7 @@ from starlette.routing import Route
 
 from ..controlplane.store import Store
 from .limiter import RULES, RateLimiter, TYPE_REQUEST, TYPE_TOKEN
-from .provider import ConfigProvider, UserQos
+from .provider import ConfigProvider
 from .quota import QuotaService
 
 
