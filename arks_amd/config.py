@@ -36,6 +36,11 @@ class ModelConfig:
     norm_topk_prob: bool = True  # renormalize top-k routing weights
     sliding_window: int | None = None  # SWA not implemented: must be None
     qk_norm: bool = False  # Qwen3: per-head RMSNorm on q/k before RoPE
+    tie_word_embeddings: bool = False
+    attention_bias: bool = True  # qwen2 has qkv bias; llama does not
+    eos_token_id: int = 151645
+    bos_token_id: int = 151643
+    torch_dtype: str = "bfloat16"
 
     def __post_init__(self):
         if self.sliding_window is not None:
@@ -43,11 +48,6 @@ class ModelConfig:
                 "sliding-window attention is not supported yet; this config "
                 f"sets sliding_window={self.sliding_window}"
             )
-    tie_word_embeddings: bool = False
-    attention_bias: bool = True  # qwen2 has qkv bias; llama does not
-    eos_token_id: int = 151645
-    bos_token_id: int = 151643
-    torch_dtype: str = "bfloat16"
 
     @classmethod
     def from_hf_config(cls, cfg: dict) -> "ModelConfig":
