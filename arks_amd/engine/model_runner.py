@@ -437,6 +437,8 @@ class ModelRunner:
         return emitted  # type: ignore[return-value]
 
     def sample(self, logits: torch.Tensor, seqs: list[Sequence]) -> list[int]:
+        if any(s.sampling.logit_bias or s.sampling.min_tokens for s in seqs):
+            logits = self._apply_bias_min_tokens(logits, seqs)
         if any(s.sampling.has_penalties for s in seqs):
             logits = self._apply_penalties(logits, seqs)
         temps = [s.sampling.temperature for s in seqs]
@@ -489,6 +491,23 @@ class ModelRunner:
                 top = {int(ti): float(vi) for vi, ti in zip(v, t)}
             out[i] = (chosen_lp, top)
         return out
+
+    def _apply_bias_min_tokens(self, logits: torch.Tensor,
+                               seqs: list[Sequence]) -> torch.Tensor:
+        """OpenAI logit_bias, and min_tokens (mask eos/stop token logits to
+        -inf until the sequence has emitted min_tokens — vLLM semantics)."""
+        logits = logits.clone()
+        eos = self.model_cfg.eos_token_id
+        for i, seq in enumerate(seqs):
+            sp = seq.sampling
+            if sp.logit_bias:
+                for t, b in sp.logit_bias.items():
+                    logits[i, t] += b
+            if sp.min_tokens and seq.num_output_tokens < sp.min_tokens:
+                logits[i, eos] = float("-inf")
+                for t in sp.stop_token_ids:
+                    logits[i, t] = float("-inf")
+        return logits
 
     @staticmethod
     def _apply_penalties(logits: torch.Tensor, seqs: list[Sequence]) -> torch.Tensor:

@@ -20,6 +20,8 @@ class SamplingParams:
     ignore_eos: bool = False
     logprobs: int | None = None  # None = off; 0 = chosen only; N = top-N too
     seed: int | None = None  # per-request RNG seed (reproducible sampling)
+    logit_bias: dict[int, float] | None = None  # OpenAI logit_bias
+    min_tokens: int = 0  # suppress eos/stop tokens until this many emitted
 
     @property
     def has_penalties(self) -> bool:

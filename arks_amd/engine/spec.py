@@ -55,4 +55,6 @@ def eligible(seq: Sequence) -> bool:
         sp.temperature == 0.0
         and not sp.has_penalties
         and sp.logprobs is None
+        and not sp.logit_bias
+        and sp.min_tokens == 0
     )

@@ -134,6 +134,11 @@ def create_app(engine: AsyncEngine, served_model_name: str, tokenizer,
             logprobs=_logprobs_of(req),
             seed=req.seed,
             ignore_eos=req.ignore_eos,
+            logit_bias=(
+                {int(k): float(v) for k, v in req.logit_bias.items()}
+                if getattr(req, "logit_bias", None) else None
+            ),
+            min_tokens=getattr(req, "min_tokens", 0),
         )
 
 

@@ -39,6 +39,8 @@ class ChatCompletionRequest(BaseModel):
     repetition_penalty: float = 1.0  # extension (vLLM-compatible)
     seed: int | None = None
     ignore_eos: bool = False  # extension (load testing)
+    logit_bias: dict[str, float] | None = None
+    min_tokens: int = 0  # extension (vLLM-compatible)
 
 
 class CompletionRequest(BaseModel):
@@ -58,6 +60,8 @@ class CompletionRequest(BaseModel):
     repetition_penalty: float = 1.0
     seed: int | None = None
     ignore_eos: bool = False
+    logit_bias: dict[str, float] | None = None
+    min_tokens: int = 0  # extension (vLLM-compatible)
 
 
 class Usage(BaseModel):

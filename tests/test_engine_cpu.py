@@ -179,3 +179,41 @@ def test_mixed_batching_greedy_equivalence():
     a_sep, b_sep = run(False)
     assert a_mixed == a_sep
     assert b_mixed == b_sep
+
+
+def test_logit_bias_and_min_tokens():
+    """OpenAI logit_bias steers greedy choice; min_tokens masks eos/stop
+    until satisfied (vLLM semantics)."""
+    from arks_amd.config import EngineConfig, PRESET_CONFIGS
+    from arks_amd.engine import LLMEngine, SamplingParams
+
+    eos = PRESET_CONFIGS["tiny"].eos_token_id
+
+    def mk():
+        return LLMEngine(EngineConfig(
+            preset="tiny", device="cpu", kv_cache_blocks=128,
+            max_model_len=256, seed=3,
+        ))
+
+    # +inf-ish bias on eos -> immediate stop
+    out = mk().generate(
+        [[5, 9, 2]],
+        SamplingParams(max_tokens=10, logit_bias={eos: 1000.0}),
+    )
+    assert out == [[eos]]
+    # same bias but min_tokens=4: eos masked until 4 tokens emitted, then
+    # the bias wins immediately -> exactly 5 tokens ending in eos
+    out = mk().generate(
+        [[5, 9, 2]],
+        SamplingParams(max_tokens=10, logit_bias={eos: 1000.0}, min_tokens=4),
+    )
+    assert len(out[0]) == 5 and out[0][-1] == eos and eos not in out[0][:4]
+    # negative bias bans a token the model would otherwise pick
+    base = mk().generate([[5, 9, 2]], SamplingParams(max_tokens=1, ignore_eos=True))
+    banned = base[0][0]
+    out = mk().generate(
+        [[5, 9, 2]],
+        SamplingParams(max_tokens=1, ignore_eos=True,
+                       logit_bias={banned: -1000.0}),
+    )
+    assert out[0][0] != banned
