@@ -254,3 +254,45 @@ def test_incremental_detokenizer_multibyte():
     # multibyte runes must never be emitted as U+FFFD fragments
     assert "�" not in out
     assert text.startswith(out) and len(text) - len(out) <= 3
+
+
+def test_hf_tokenizer_incremental_stream(tmp_path):
+    """A real BPE tokenizer (built in-process — no network) through
+    load_tokenizer + IncrementalDetokenizer: streamed pieces must
+    concatenate to the full decode."""
+    import json as _json
+
+    from tokenizers import Tokenizer, decoders, models, pre_tokenizers, trainers
+
+    from arks_amd.server.tokenizer import (
+        HFTokenizer,
+        IncrementalDetokenizer,
+        load_tokenizer,
+    )
+
+    tok = Tokenizer(models.BPE(unk_token="[UNK]"))
+    tok.pre_tokenizer = pre_tokenizers.ByteLevel()
+    tok.decoder = decoders.ByteLevel()
+    trainer = trainers.BpeTrainer(
+        vocab_size=300, special_tokens=["[UNK]", "<eos>"]
+    )
+    corpus = [
+        "the quick brown fox jumps over the lazy dog",
+        "hello world, hello tokenizer streams",
+        "paged attention over xgmi links",
+    ]
+    tok.train_from_iterator(corpus, trainer)
+    tok.save(str(tmp_path / "tokenizer.json"))
+    (tmp_path / "tokenizer_config.json").write_text(_json.dumps({
+        "tokenizer_class": "PreTrainedTokenizerFast", "eos_token": "<eos>",
+    }))
+
+    t = load_tokenizer(str(tmp_path), 512, 1)
+    assert isinstance(t, HFTokenizer)
+    text = "hello world, the quick brown fox streams"
+    ids = t.encode(text)
+    assert len(ids) > 3
+    full = t.decode(ids)
+    detok = IncrementalDetokenizer(t)
+    streamed = "".join(detok.feed(i) for i in ids) + detok.flush()
+    assert streamed == full
