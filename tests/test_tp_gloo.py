@@ -31,7 +31,7 @@ def _tp1_reference(preset="tiny", quantization=None):
 
 
 def _tp_worker(rank: int, world: int, port: int, q, preset: str = "tiny",
-               quantization=None):
+               quantization=None, speculative=None):
     os.environ["MASTER_ADDR"] = "127.0.0.1"
     os.environ["MASTER_PORT"] = str(port)
     os.environ["RANK"] = str(rank)
@@ -47,6 +47,7 @@ def _tp_worker(rank: int, world: int, port: int, q, preset: str = "tiny",
             EngineConfig(
                 preset=preset, device="cpu", kv_cache_blocks=128,
                 max_model_len=512, quantization=quantization,
+                speculative=speculative,
             )
         )
         out = e.generate(PROMPTS, SamplingParams(max_tokens=MAX_TOKENS, ignore_eos=True))
@@ -284,3 +285,26 @@ def test_tp2_async_engine_disagg_worker_protocol():
     plain, toks = payload
     assert len(plain) == 5
     assert toks == plain, f"disagg {toks} != plain {plain}"
+
+
+@pytest.mark.timeout(240)
+def test_tp2_speculative_matches_tp1_plain():
+    """Speculative decoding under TP=2 SPMD: every rank proposes/accepts
+    identically (drafts are a pure function of the shared token history),
+    so the output equals the plain TP=1 engine on the CPU fp32 path."""
+    ref = _tp1_reference()
+
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    port = _free_port()
+    procs = [
+        ctx.Process(target=_tp_worker, args=(r, 2, port, q, "tiny", None, "ngram"))
+        for r in range(2)
+    ]
+    for p in procs:
+        p.start()
+    status, payload = q.get(timeout=200)
+    for p in procs:
+        p.join(timeout=60)
+    assert status == "ok", payload
+    assert payload == ref, f"TP=2 spec output {payload} != TP=1 {ref}"
