@@ -49,10 +49,12 @@ def audit(engine: LLMEngine) -> None:
         min_size=4, max_size=30,
     ),
 )
-def test_engine_invariants_under_random_ops(actions):
+@pytest.mark.parametrize("speculative", [None, "ngram"])
+def test_engine_invariants_under_random_ops(actions, speculative):
     eng = LLMEngine(EngineConfig(
         preset="tiny", device="cpu", kv_cache_blocks=24, max_model_len=96,
         max_num_batched_tokens=48, max_num_seqs=8, seed=5,
+        speculative=speculative,
     ))
     rid = 0
     for kind, a, b, c in actions:
