@@ -130,6 +130,22 @@ class MLP(nn.Module):
         return self.down_proj(ops.silu_mul(gate_up))
 
 
+_ARANGE_CACHE: dict[tuple[int, int, str], torch.Tensor] = {}
+
+
+def _arange_interleave(T: int, k: int, device) -> torch.Tensor:
+    """Cached [0,0,..,1,1,..] token-index vector — identical for every MoE
+    layer of a forward, so build it once per (T, k) instead of 48 times."""
+    key = (T, k, str(device))
+    t = _ARANGE_CACHE.get(key)
+    if t is None:
+        if len(_ARANGE_CACHE) > 256:
+            _ARANGE_CACHE.clear()
+        t = torch.arange(T, device=device).repeat_interleave(k)
+        _ARANGE_CACHE[key] = t
+    return t
+
+
 class MoEMLP(nn.Module):
     """Mixtral-style sparse MoE block, expert-parallel over the TP group:
     each rank owns num_local_experts/tp whole experts (an expert's FFN is
@@ -231,15 +247,16 @@ class MoEMLP(nn.Module):
         # segment table instead of a .nonzero() sync per expert.
         k = self.top_k
         flat_sel = selected.reshape(-1)
-        flat_tok = torch.arange(T, device=x.device).repeat_interleave(k)
+        flat_tok = _arange_interleave(T, k, x.device)
         flat_w = weights.reshape(-1)
         order = torch.argsort(flat_sel, stable=True)
         tok_sorted = flat_tok[order]
         w_sorted = flat_w[order]
         counts = torch.bincount(flat_sel, minlength=self.num_experts)
         offs = torch.cumsum(counts, 0)
-        counts_h = counts.cpu().tolist()  # the single sync
-        offs_h = offs.cpu().tolist()
+        seg = torch.stack((counts, offs)).cpu()  # ONE host sync per layer
+        counts_h = seg[0].tolist()
+        offs_h = seg[1].tolist()
         x_g = x[tok_sorted]  # [T*k, H] gathered once
         El = self.local_experts
         base = self.expert_base
