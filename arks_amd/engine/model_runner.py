@@ -545,7 +545,8 @@ class ModelRunner:
         rows = [
             i for i, s in enumerate(seqs)
             if s.sampling.temperature > 0.0
-            and (s.sampling.top_p < 1.0 or s.sampling.top_k > 0)
+            and (s.sampling.top_p < 1.0 or s.sampling.top_k > 0
+                 or s.sampling.min_p > 0.0)
         ]
         if not rows:
             return logits
@@ -563,6 +564,9 @@ class ModelRunner:
                 keep[j] &= (cum[j] - probs[j]) < sp.top_p
             if sp.top_k > 0:
                 keep[j, sp.top_k:] = False
+            if sp.min_p > 0.0:
+                # vLLM min_p: drop tokens with prob < min_p * max prob
+                keep[j] &= probs[j] >= sp.min_p * probs[j, 0]
             keep[j, 0] = True  # always keep the best token
         sub = sub.masked_fill(~keep, float("-inf"))
         # scatter back to original column order

@@ -13,6 +13,7 @@ class SamplingParams:
     temperature: float = 0.0  # 0 => greedy
     top_p: float = 1.0  # applied via logit filtering when < 1.0
     top_k: int = 0  # 0 => disabled
+    min_p: float = 0.0  # drop tokens with prob < min_p * max-prob (vLLM)
     presence_penalty: float = 0.0  # flat penalty on seen tokens
     frequency_penalty: float = 0.0  # per-occurrence penalty
     repetition_penalty: float = 1.0  # >1 divides positive seen-logits

@@ -128,6 +128,7 @@ def create_app(engine: AsyncEngine, served_model_name: str, tokenizer,
             temperature=max(0.0, req.temperature),
             top_p=req.top_p,
             top_k=getattr(req, "top_k", 0),
+            min_p=getattr(req, "min_p", 0.0),
             presence_penalty=getattr(req, "presence_penalty", 0.0),
             frequency_penalty=getattr(req, "frequency_penalty", 0.0),
             repetition_penalty=getattr(req, "repetition_penalty", 1.0),

@@ -34,6 +34,7 @@ class ChatCompletionRequest(BaseModel):
     logprobs: bool = False
     top_logprobs: int = 0
     top_k: int = 0  # extension (vLLM-compatible)
+    min_p: float = 0.0  # extension (vLLM-compatible)
     presence_penalty: float = 0.0
     frequency_penalty: float = 0.0
     repetition_penalty: float = 1.0  # extension (vLLM-compatible)
@@ -55,6 +56,7 @@ class CompletionRequest(BaseModel):
     stop: list[str] | str | None = None
     logprobs: int | None = None
     top_k: int = 0
+    min_p: float = 0.0
     presence_penalty: float = 0.0
     frequency_penalty: float = 0.0
     repetition_penalty: float = 1.0

@@ -217,3 +217,26 @@ def test_logit_bias_and_min_tokens():
                        logit_bias={banned: -1000.0}),
     )
     assert out[0][0] != banned
+
+
+def test_min_p_filter():
+    """min_p keeps tokens with prob >= min_p * max-prob and masks the rest
+    (vLLM semantics), composing with top_k."""
+    import torch
+
+    from arks_amd.engine.model_runner import ModelRunner
+    from arks_amd.engine.sequence import Sequence, SamplingParams
+
+    logits = torch.tensor([[10.0, 9.9, 5.0, 1.0],
+                           [10.0, 9.9, 9.8, 9.7]])
+    seqs = [
+        Sequence([1], SamplingParams(temperature=1.0, min_p=0.5)),
+        Sequence([1], SamplingParams(temperature=1.0, min_p=0.5, top_k=2)),
+    ]
+    out = ModelRunner._apply_top_p_top_k(logits, seqs)
+    # row 0: 10 and 9.9 survive (ratio e^-0.1 ~ 0.90 > 0.5); 5 and 1 masked
+    assert torch.isfinite(out[0, 0]) and torch.isfinite(out[0, 1])
+    assert out[0, 2] == float("-inf") and out[0, 3] == float("-inf")
+    # row 1: all four pass min_p but top_k=2 cuts the tail
+    assert torch.isfinite(out[1, 0]) and torch.isfinite(out[1, 1])
+    assert out[1, 2] == float("-inf") and out[1, 3] == float("-inf")
