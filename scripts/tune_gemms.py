@@ -48,6 +48,9 @@ def main():
     ap.add_argument("--tp", type=int, default=1)
     ap.add_argument("--fp8", action="store_true",
                     help="also tune rowwise-scaled fp8 _scaled_mm shapes")
+    ap.add_argument("--moe", action="store_true",
+                    help="also tune the MoE strided-batch bmm shapes "
+                         "(NN layout, pre-transposed expert weights)")
     ap.add_argument("--out", default="arks_amd/data/tunableop_gfx950.csv")
     args = ap.parse_args()
 
@@ -87,6 +90,24 @@ def main():
                 torch.cuda.synchronize()
                 print(f"tuned fp8 {name} M={m}", flush=True)
         del w, b
+        torch.cuda.empty_cache()
+
+    if args.moe and cfg.num_local_experts:
+        # grouped/dense expert bmms: [E, rows, H] @ [E, H, 2I] and
+        # [E, rows, I] @ [E, I, H] (NN, contiguous — the engine layout).
+        E = cfg.num_local_experts // args.tp
+        I = cfg.moe_intermediate_size or cfg.intermediate_size
+        H = cfg.hidden_size
+        w13 = torch.randn(E, H, 2 * I, dtype=torch.bfloat16, device=dev) * 0.01
+        w2 = torch.randn(E, I, H, dtype=torch.bfloat16, device=dev) * 0.01
+        # decode dense path rows (= batch <= 64) and typical prefill caps
+        for rows in [1, 2, 4, 8, 16, 32, 64, 128, 256, 512, 768, 1024]:
+            x = torch.randn(E, rows, H, dtype=torch.bfloat16, device=dev) * 0.01
+            h = torch.bmm(x, w13)[..., :I].contiguous()
+            torch.bmm(h, w2)
+            torch.cuda.synchronize()
+            print(f"tuned moe bmm rows={rows}", flush=True)
+        del w13, w2
         torch.cuda.empty_cache()
 
     # results are flushed to the filename at interpreter exit
