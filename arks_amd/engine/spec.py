@@ -20,6 +20,11 @@ reduction orders — so a near-tie argmax can resolve differently between
 the speculative and plain engines; both are valid greedy outputs of the
 model. The CPU fp32 path is bitwise single-path and pins exact equality in
 tests/test_spec.py.
+
+Scheduling note: speculation applies to decode-only steps. When mixed
+batching folds prefill chunks into a step, that step decodes normally —
+under sustained prefill arrival the engine alternates between the two,
+which is the right trade (prefill throughput dominates those phases).
 """
 
 from __future__ import annotations
