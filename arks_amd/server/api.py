@@ -149,18 +149,23 @@ def create_app(engine: AsyncEngine, served_model_name: str, tokenizer,
         out_ids: list[int] = []
         text_acc = ""
         finish = None
-        async for out in engine.generate_stream(rid, token_ids, sp,
-                                                prefill_addr=prefill_addr):
-            out_ids.append(out.new_token_id)
-            if stops:
-                emit, stopped = tracker.feed(tokenizer.decode([out.new_token_id]))
-                text_acc += emit
-                if stopped:
-                    finish = "stop"
-                    engine.abort(rid)
-                    break
-            if out.finished:
-                finish = out.finish_reason
+        try:
+            async for out in engine.generate_stream(rid, token_ids, sp,
+                                                    prefill_addr=prefill_addr):
+                out_ids.append(out.new_token_id)
+                if stops:
+                    emit, stopped = tracker.feed(
+                        tokenizer.decode([out.new_token_id]))
+                    text_acc += emit
+                    if stopped:
+                        finish = "stop"
+                        engine.abort(rid)
+                        break
+                if out.finished:
+                    finish = out.finish_reason
+        except asyncio.CancelledError:
+            engine.abort(rid)  # client went away mid-request
+            raise
         if stops and finish != "stop":
             text_acc += tracker.flush()
         text = text_acc if stops else tokenizer.decode(out_ids)
@@ -210,28 +215,34 @@ def create_app(engine: AsyncEngine, served_model_name: str, tokenizer,
         text_acc = ""
         lp_content: list[dict] = []
         finish = None
-        async for out in engine.generate_stream(
-            rid, token_ids, sp, prefill_addr=raw.headers.get("x-arks-prefill-addr")
-        ):
-            text_ids.append(out.new_token_id)
-            if out.logprob is not None:
-                lp_content.append({
-                    "token": tokenizer.decode([out.new_token_id]),
-                    "logprob": out.logprob,
-                    "top_logprobs": [
-                        {"token": tokenizer.decode([t]), "logprob": v}
-                        for t, v in (out.top_logprobs or {}).items()
-                    ],
-                })
-            if stops:
-                emit, stopped = tracker.feed(tokenizer.decode([out.new_token_id]))
-                text_acc += emit
-                if stopped:
-                    finish = "stop"
-                    engine.abort(rid)
-                    break
-            if out.finished:
-                finish = out.finish_reason
+        try:
+            async for out in engine.generate_stream(
+                rid, token_ids, sp,
+                prefill_addr=raw.headers.get("x-arks-prefill-addr"),
+            ):
+                text_ids.append(out.new_token_id)
+                if out.logprob is not None:
+                    lp_content.append({
+                        "token": tokenizer.decode([out.new_token_id]),
+                        "logprob": out.logprob,
+                        "top_logprobs": [
+                            {"token": tokenizer.decode([t]), "logprob": v}
+                            for t, v in (out.top_logprobs or {}).items()
+                        ],
+                    })
+                if stops:
+                    emit, stopped = tracker.feed(
+                        tokenizer.decode([out.new_token_id]))
+                    text_acc += emit
+                    if stopped:
+                        finish = "stop"
+                        engine.abort(rid)
+                        break
+                if out.finished:
+                    finish = out.finish_reason
+        except asyncio.CancelledError:
+            engine.abort(rid)  # client went away mid-request
+            raise
         if stops and finish != "stop":
             text_acc += tracker.flush()
         usage = Usage(

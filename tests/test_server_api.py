@@ -296,3 +296,40 @@ def test_hf_tokenizer_incremental_stream(tmp_path):
     detok = IncrementalDetokenizer(t)
     streamed = "".join(detok.feed(i) for i in ids) + detok.flush()
     assert streamed == full
+
+
+def test_non_stream_cancellation_aborts_engine(app):
+    """A client disconnect (task cancellation) mid non-stream request must
+    abort the engine sequence instead of decoding to completion."""
+
+    async def fn(client):
+        engine = app.state.engine
+        task = asyncio.get_running_loop().create_task(client.post(
+            "/v1/chat/completions",
+            json={
+                "model": "tiny",
+                "messages": [{"role": "user", "content": "hello there"}],
+                "max_tokens": 4000,
+                "temperature": 0,
+                "ignore_eos": True,
+            },
+        ))
+        # let the request get admitted and produce a few tokens
+        for _ in range(200):
+            await asyncio.sleep(0.01)
+            if engine.engine.scheduler.num_running > 0:
+                break
+        assert engine.engine.scheduler.num_running > 0
+        task.cancel()
+        try:
+            await task
+        except (asyncio.CancelledError, Exception):
+            pass
+        # the abort drains on the next engine iterations
+        for _ in range(300):
+            await asyncio.sleep(0.01)
+            if not engine.engine.has_work():
+                break
+        assert not engine.engine.has_work(), "sequence kept decoding after disconnect"
+
+    run_with_client(app, fn)
