@@ -149,6 +149,11 @@ class LLMEngine:
                     num_output_tokens=seq.num_output_tokens,
                     logprob=lp,
                     top_logprobs=top,
+                    prompt_logprobs=(
+                        list(seq.prompt_logprob_values)
+                        if seq.sampling.prompt_logprobs
+                        and seq.num_output_tokens == 1 else None
+                    ),
                 )
             )
         self.scheduler.free_finished()

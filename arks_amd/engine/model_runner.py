@@ -190,6 +190,8 @@ class ModelRunner:
             q_lens: list[int] = []
             kv_lens: list[int] = []
             logits_idx: list[int] = []
+            lp_meta: list[tuple[int, int, int, int]] = []
+            self._lp_prefill = None
             for i, seq in enumerate(sb.seqs):
                 toks = seq.all_token_ids
                 c = seq.num_cached_tokens
@@ -204,8 +206,30 @@ class ModelRunner:
                 cu.append(cu[-1] + (n - c))
                 q_lens.append(n - c)
                 kv_lens.append(n)
-                logits_idx.append(cu[-1] - 1)
+                if seq.sampling.prompt_logprobs and c < seq.num_prompt_tokens:
+                    # echo/prompt-scoring: lm_head on EVERY row of this seq
+                    # (sampling still reads the last row — see execute())
+                    off = len(logits_idx)
+                    logits_idx.extend(range(cu[-2], cu[-1]))
+                    lp_meta.append((i, off, n - c, c))
+                else:
+                    logits_idx.append(cu[-1] - 1)
                 seq.num_cached_tokens = n
+            if lp_meta:
+                # per-seq index of its SAMPLING row within the gathered
+                # logits (lp seqs contributed a whole span, others one row)
+                sample_pos = []
+                lp_by_seq = {m[0]: m for m in lp_meta}
+                g = 0
+                for i in range(len(sb.seqs)):
+                    m = lp_by_seq.get(i)
+                    if m is not None:
+                        sample_pos.append(g + m[2] - 1)
+                        g += m[2]
+                    else:
+                        sample_pos.append(g)
+                        g += 1
+                self._lp_prefill = (lp_meta, sample_pos)
             dev = self.device
             block_tables = None
             seq_lens = None
@@ -359,7 +383,34 @@ class ModelRunner:
             return self.sample(logits, sb.seqs)
         fb = self.prepare_batch(sb)
         logits = self.model(fb, self.kv_caches)  # [num_seqs, vocab]
+        if sb.is_prefill and self._lp_prefill is not None:
+            logits = self._finish_prompt_logprobs(logits, sb)
         return self.sample(logits, sb.seqs)
+
+    _lp_prefill: tuple | None = None
+
+    def _finish_prompt_logprobs(self, logits: torch.Tensor,
+                                sb: ScheduledBatch) -> torch.Tensor:
+        """Score prompt tokens for echo/prompt_logprobs requests: the row at
+        position p predicts token p+1, so targets 1..P-1 accumulate across
+        prefill chunks. Returns the [num_seqs, vocab] sampling logits."""
+        lp_meta, sample_pos = self._lp_prefill
+        self._lp_prefill = None
+        for i, off, qn, c in lp_meta:
+            seq = sb.seqs[i]
+            P = seq.num_prompt_tokens
+            toks = seq.all_token_ids
+            lsm = torch.log_softmax(logits[off:off + qn].float(), dim=-1)
+            vals = seq.prompt_logprob_values
+            for j in range(qn):
+                q = c + j + 1  # target position scored by this row
+                if q > P - 1:
+                    break
+                if q - 1 < len(vals):
+                    continue  # already scored (preemption recompute)
+                vals.append(float(lsm[j, toks[q]]))
+        idx = torch.tensor(sample_pos, dtype=torch.long, device=logits.device)
+        return logits[idx]
 
     def execute_spec(self, sb: ScheduledBatch,
                      drafts: list[list[int]]) -> list[list[int]]:

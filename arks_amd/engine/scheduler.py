@@ -141,7 +141,12 @@ class Scheduler:
                 # Fresh admission: reuse cached full blocks of the prompt
                 # (refs taken here must be released on every non-admit
                 # path). Cap at n-1 so at least one token produces logits.
-                reused, ncached = self.allocator.match_prefix(toks, n - 1)
+                # prompt_logprobs needs logits at EVERY prompt position, so
+                # those requests skip prefix reuse and recompute in full.
+                if seq.sampling.prompt_logprobs:
+                    reused, ncached = [], 0
+                else:
+                    reused, ncached = self.allocator.match_prefix(toks, n - 1)
                 seq.num_cached_tokens = ncached
             else:
                 # Continuation of a chunked prefill: pages for the covered

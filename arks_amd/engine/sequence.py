@@ -23,6 +23,7 @@ class SamplingParams:
     seed: int | None = None  # per-request RNG seed (reproducible sampling)
     logit_bias: dict[int, float] | None = None  # OpenAI logit_bias
     min_tokens: int = 0  # suppress eos/stop tokens until this many emitted
+    prompt_logprobs: bool = False  # score prompt tokens at prefill (echo)
 
     @property
     def has_penalties(self) -> bool:
@@ -62,6 +63,9 @@ class Sequence:
         # decode instance can pull them (scheduler.held).
         self.hold_pages = False
         self.num_cached_tokens = 0  # tokens whose KV is already in cache
+        # prompt-token logprobs (sampling.prompt_logprobs): value at index
+        # q-1 scores prompt token q given tokens < q (first token unscored)
+        self.prompt_logprob_values: list[float] = []
         self.arrival_time = arrival_time if arrival_time is not None else time.time()
         self.first_token_time: float | None = None
         self.finish_time: float | None = None
@@ -125,3 +129,5 @@ class StepOutput:
     num_output_tokens: int
     logprob: float | None = None  # chosen token's logprob (when requested)
     top_logprobs: dict[int, float] | None = None
+    # prompt-token logprobs (echo): set on the sequence's FIRST output
+    prompt_logprobs: list[float] | None = None

@@ -344,6 +344,11 @@ def create_app(engine: AsyncEngine, served_model_name: str, tokenizer,
             return _error(400, f"prompt is {len(token_ids)} tokens; "
                                f"max_model_len is {engine.cfg.max_model_len}")
         sp = _sampling(req, req.max_tokens)
+        if req.echo:
+            if req.n > 1 or req.stream:
+                return _error(400, "echo is not supported with n > 1 or stream")
+            # lm-eval-style prompt scoring: logits at every prompt position
+            sp.prompt_logprobs = req.logprobs is not None
         rid = f"cmpl-{uuid.uuid4().hex}"
         if req.stream:
             if req.n > 1:
@@ -377,40 +382,63 @@ def create_app(engine: AsyncEngine, served_model_name: str, tokenizer,
         finish = None
         token_lps: list[float] = []
         top_lps: list[dict] = []
-        async for out in engine.generate_stream(
-            rid, token_ids, sp, prefill_addr=raw.headers.get("x-arks-prefill-addr")
-        ):
-            out_ids.append(out.new_token_id)
-            if out.logprob is not None:
-                token_lps.append(out.logprob)
-                top_lps.append({
-                    tokenizer.decode([t]): v
-                    for t, v in (out.top_logprobs or {}).items()
-                })
-            if stops:
-                emit, stopped = tracker.feed(tokenizer.decode([out.new_token_id]))
-                text_acc += emit
-                if stopped:
-                    finish = "stop"
-                    engine.abort(rid)
-                    break
-            if out.finished:
-                finish = out.finish_reason
+        prompt_lps: list[float] | None = None
+        try:
+            async for out in engine.generate_stream(
+                rid, token_ids, sp,
+                prefill_addr=raw.headers.get("x-arks-prefill-addr"),
+            ):
+                out_ids.append(out.new_token_id)
+                if out.prompt_logprobs is not None:
+                    prompt_lps = out.prompt_logprobs
+                if out.logprob is not None:
+                    token_lps.append(out.logprob)
+                    top_lps.append({
+                        tokenizer.decode([t]): v
+                        for t, v in (out.top_logprobs or {}).items()
+                    })
+                if stops:
+                    emit, stopped = tracker.feed(
+                        tokenizer.decode([out.new_token_id]))
+                    text_acc += emit
+                    if stopped:
+                        finish = "stop"
+                        engine.abort(rid)
+                        break
+                if out.finished:
+                    finish = out.finish_reason
+        except asyncio.CancelledError:
+            engine.abort(rid)  # client went away mid-request
+            raise
         if stops and finish != "stop":
             text_acc += tracker.flush()
+        text = text_acc if stops else tokenizer.decode(out_ids)
+        lp_obj = None
+        if token_lps:
+            lp_obj = {
+                "tokens": [tokenizer.decode([t]) for t in out_ids],
+                "token_logprobs": list(token_lps),
+                "top_logprobs": list(top_lps),
+            }
+        if req.echo:
+            # OpenAI echo contract: prompt text precedes the completion and,
+            # with logprobs, prompt tokens are scored (first one null)
+            text = tokenizer.decode(token_ids) + text
+            if req.logprobs is not None:
+                lp_obj = {
+                    "tokens": [tokenizer.decode([t]) for t in token_ids]
+                    + (lp_obj["tokens"] if lp_obj else []),
+                    "token_logprobs": [None] + (prompt_lps or [])
+                    + (lp_obj["token_logprobs"] if lp_obj else []),
+                    "top_logprobs": [None] * len(token_ids)
+                    + (lp_obj["top_logprobs"] if lp_obj else []),
+                }
         return CompletionResponse(
             id=rid, model=req.model,
             choices=[CompletionChoice(
-                text=text_acc if stops else tokenizer.decode(out_ids),
+                text=text,
                 finish_reason=finish or "stop",
-                logprobs=(
-                    {
-                        "tokens": [tokenizer.decode([t]) for t in out_ids],
-                        "token_logprobs": token_lps,
-                        "top_logprobs": top_lps,
-                    }
-                    if token_lps else None
-                ))],
+                logprobs=lp_obj)],
             usage=Usage(
                 prompt_tokens=len(token_ids),
                 completion_tokens=len(out_ids),

@@ -64,6 +64,7 @@ class CompletionRequest(BaseModel):
     ignore_eos: bool = False
     logit_bias: dict[str, float] | None = None
     min_tokens: int = 0  # extension (vLLM-compatible)
+    echo: bool = False  # return the prompt (+ its logprobs) too
 
 
 class Usage(BaseModel):

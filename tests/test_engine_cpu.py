@@ -240,3 +240,40 @@ def test_min_p_filter():
     # row 1: all four pass min_p but top_k=2 cuts the tail
     assert torch.isfinite(out[1, 0]) and torch.isfinite(out[1, 1])
     assert out[1, 2] == float("-inf") and out[1, 3] == float("-inf")
+
+
+def test_prompt_logprobs_chunked_matches_unchunked():
+    """echo/prompt scoring: values accumulate across prefill chunks and
+    must equal the single-chunk run; prefix-cache hits are bypassed so a
+    repeated prompt still gets fully scored."""
+    import math
+
+    from arks_amd.config import EngineConfig
+    from arks_amd.engine import LLMEngine, SamplingParams
+
+    prompt = [(7 * i + 3) % 200 for i in range(50)]
+    sp = SamplingParams(max_tokens=2, ignore_eos=True, prompt_logprobs=True)
+
+    def run(chunk):
+        e = LLMEngine(EngineConfig(
+            preset="tiny", device="cpu", kv_cache_blocks=256,
+            max_model_len=256, max_num_batched_tokens=chunk, seed=6,
+        ))
+        s1 = e.add_request(list(prompt), sp)
+        while e.has_work():
+            e.step()
+        # a second identical request must not lose positions to the
+        # prefix cache
+        s2 = e.add_request(list(prompt), sp)
+        while e.has_work():
+            e.step()
+        return s1.prompt_logprob_values, s2.prompt_logprob_values
+
+    big1, big2 = run(4096)
+    small1, _ = run(16)  # forces 4 chunks
+    assert len(big1) == len(prompt) - 1
+    assert big1 == big2
+    assert len(small1) == len(big1)
+    assert all(math.isclose(a, b, rel_tol=1e-4, abs_tol=1e-5)
+               for a, b in zip(small1, big1))
+    assert all(v <= 0.0 for v in big1)

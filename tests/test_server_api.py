@@ -333,3 +333,41 @@ def test_non_stream_cancellation_aborts_engine(app):
         assert not engine.engine.has_work(), "sequence kept decoding after disconnect"
 
     run_with_client(app, fn)
+
+
+def test_completions_echo_with_logprobs(app):
+    """OpenAI echo contract: prompt text precedes the completion; with
+    logprobs the prompt tokens are scored (first one null)."""
+
+    async def fn(client):
+        r = await client.post("/v1/completions", json={
+            "model": "tiny", "prompt": "hello", "max_tokens": 3,
+            "temperature": 0, "ignore_eos": True,
+            "echo": True, "logprobs": 0,
+        })
+        assert r.status_code == 200, r.text
+        body = r.json()
+        choice = body["choices"][0]
+        assert choice["text"].startswith("hello")
+        lp = choice["logprobs"]
+        n_prompt = body["usage"]["prompt_tokens"]
+        assert len(lp["tokens"]) == n_prompt + 3
+        assert len(lp["token_logprobs"]) == n_prompt + 3
+        assert lp["token_logprobs"][0] is None
+        assert all(isinstance(v, float) and v <= 0.0
+                   for v in lp["token_logprobs"][1:])
+        # echo without logprobs: text only
+        r2 = await client.post("/v1/completions", json={
+            "model": "tiny", "prompt": "hello", "max_tokens": 3,
+            "temperature": 0, "ignore_eos": True, "echo": True,
+        })
+        assert r2.json()["choices"][0]["text"].startswith("hello")
+        assert r2.json()["choices"][0]["logprobs"] is None
+        # echo + stream rejected
+        r3 = await client.post("/v1/completions", json={
+            "model": "tiny", "prompt": "x", "max_tokens": 1,
+            "echo": True, "stream": True,
+        })
+        assert r3.status_code == 400
+
+    run_with_client(app, fn)
