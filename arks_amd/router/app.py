@@ -25,7 +25,7 @@ import json
 import httpx
 from fastapi import FastAPI, Request, Response
 from fastapi.responses import JSONResponse, StreamingResponse
-from prometheus_client import CollectorRegistry, Counter, generate_latest
+from prometheus_client import CollectorRegistry, Counter, Gauge, generate_latest
 
 HOP_HEADERS = {"host", "content-length", "connection", "accept-encoding"}
 
@@ -39,11 +39,24 @@ class RouterState:
         self.transport = transport
         self._rr_prefill = itertools.count()
         self._rr_decode = itertools.count()
+        self.inflight: dict[str, int] = {}  # decode url -> open requests
         self.registry = CollectorRegistry()
         self.requests_total = Counter(
             "arks_router_requests_total", "routed requests",
             labelnames=["decode", "prefill"], registry=self.registry,
         )
+        self.inflight_gauge = Gauge(
+            "arks_router_inflight", "open requests per decode worker",
+            labelnames=["decode"], registry=self.registry,
+        )
+
+    def acquire(self, decode: str) -> None:
+        self.inflight[decode] = self.inflight.get(decode, 0) + 1
+        self.inflight_gauge.labels(decode=decode).set(self.inflight[decode])
+
+    def release(self, decode: str) -> None:
+        self.inflight[decode] = max(self.inflight.get(decode, 0) - 1, 0)
+        self.inflight_gauge.labels(decode=decode).set(self.inflight[decode])
 
     def set_workers(self, prefill_urls: list[str] | None = None,
                     decode_urls: list[str] | None = None) -> None:
@@ -54,9 +67,16 @@ class RouterState:
 
     # --- policies ---
     def pick_decode(self) -> str:
+        """cache_aware: least-loaded decode worker (open-request count,
+        rotating tiebreak); round_robin: plain rotation."""
         if not self.decode_urls:
             raise LookupError("no decode workers")
-        return self.decode_urls[next(self._rr_decode) % len(self.decode_urls)]
+        n = len(self.decode_urls)
+        start = next(self._rr_decode) % n
+        if self.policy == "round_robin":
+            return self.decode_urls[start]
+        order = self.decode_urls[start:] + self.decode_urls[:start]
+        return min(order, key=lambda u: self.inflight.get(u, 0))
 
     def pick_prefill(self, prompt_key: str) -> str | None:
         if not self.prefill_urls:
@@ -106,18 +126,25 @@ def create_router_app(state: RouterState) -> FastAPI:
         if prefill:
             headers["x-arks-prefill-addr"] = _base(prefill)
         state.requests_total.labels(decode=decode, prefill=prefill or "").inc()
+        state.acquire(decode)
         client = httpx.AsyncClient(
             transport=state.transport, base_url=_base(decode), timeout=600.0
         )
         req = client.build_request("POST", path, content=body_bytes,
                                    headers=headers)
-        resp = await client.send(req, stream=True)
+        try:
+            resp = await client.send(req, stream=True)
+        except Exception:
+            state.release(decode)
+            await client.aclose()
+            raise
 
         async def relay():
             try:
                 async for chunk in resp.aiter_raw():
                     yield chunk
             finally:
+                state.release(decode)
                 await resp.aclose()
                 await client.aclose()
 

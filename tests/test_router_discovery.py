@@ -50,3 +50,27 @@ def test_discovery_updates_worker_lists():
     # only READY pods are registered
     assert state.prefill_urls == ["http://10.0.0.1:8080"]
     assert state.decode_urls == ["http://10.0.0.3:8080"]
+
+
+def test_least_loaded_decode_policy():
+    from arks_amd.router.app import RouterState
+
+    st = RouterState([], ["http://d1", "http://d2", "http://d3"],
+                     policy="cache_aware")
+    # no load: rotation spreads picks
+    picks = {st.pick_decode() for _ in range(3)}
+    assert len(picks) == 3
+    # d1 and d2 busy -> d3 wins regardless of rotation
+    st.acquire("http://d1")
+    st.acquire("http://d1")
+    st.acquire("http://d2")
+    assert all(st.pick_decode() == "http://d3" for _ in range(4))
+    # release evens things out again
+    st.release("http://d1")
+    st.release("http://d1")
+    st.release("http://d2")
+    assert len({st.pick_decode() for _ in range(3)}) == 3
+    # round_robin ignores load
+    rr = RouterState([], ["http://a", "http://b"], policy="round_robin")
+    rr.acquire("http://a")
+    assert {rr.pick_decode(), rr.pick_decode()} == {"http://a", "http://b"}
