@@ -112,7 +112,9 @@ def create_app(engine: AsyncEngine, served_model_name: str, tokenizer,
 
     @app.get("/health")
     async def health():
-        return Response(status_code=200)
+        # go unready after a fatal engine-loop error so the endpoint
+        # controller pulls this pod out of the HTTPRoute
+        return Response(status_code=503 if engine.failed else 200)
 
     @app.get("/v1/models")
     async def models():

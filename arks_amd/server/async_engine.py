@@ -96,6 +96,7 @@ class AsyncEngine:
         self.pending_aborts: list[str] = []
         self._wakeup: asyncio.Event | None = None
         self._task: asyncio.Task | None = None
+        self.failed: BaseException | None = None  # fatal engine-loop error
         # Collectives + forward run on ONE dedicated thread.
         self._executor = concurrent.futures.ThreadPoolExecutor(max_workers=1)
         self._stopped = False
@@ -225,9 +226,19 @@ class AsyncEngine:
                 continue
             msg = {"adds": self.pending_adds, "aborts": self.pending_aborts}
             self.pending_adds, self.pending_aborts = [], []
-            outputs, rejected = await loop.run_in_executor(
-                self._executor, self._sync_iteration, msg
-            )
+            try:
+                outputs, rejected = await loop.run_in_executor(
+                    self._executor, self._sync_iteration, msg
+                )
+            except Exception as exc:  # fatal: fail open streams, go unready
+                import traceback
+
+                self.failed = exc
+                traceback.print_exc()
+                for st in self.streams.values():
+                    st.queue.put_nowait(None)
+                self.streams.clear()
+                return
             for rid in rejected:
                 st = self.streams.pop(rid, None)
                 if st:

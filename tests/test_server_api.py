@@ -371,3 +371,28 @@ def test_completions_echo_with_logprobs(app):
         assert r3.status_code == 400
 
     run_with_client(app, fn)
+
+
+def test_engine_loop_failure_fails_streams_and_health(app):
+    """A fatal exception inside engine.step must end open streams (not hang
+    them) and flip /health to 503 for readiness-based ejection."""
+
+    async def fn(client):
+        eng = app.state.engine
+
+        def boom():
+            raise RuntimeError("injected engine fault")
+
+        eng.engine.step = boom
+        r = await client.post("/v1/completions", json={
+            "model": "tiny", "prompt": "x", "max_tokens": 5,
+            "temperature": 0, "ignore_eos": True,
+        })
+        # request completes (empty output) instead of hanging
+        assert r.status_code == 200
+        assert r.json()["usage"]["completion_tokens"] == 0
+        h = await client.get("/health")
+        assert h.status_code == 503
+        assert isinstance(eng.failed, RuntimeError)
+
+    run_with_client(app, fn)
