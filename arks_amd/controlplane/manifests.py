@@ -14,7 +14,6 @@ from typing import Any
 
 from ..crd.types import (
     LABEL_APPLICATION,
-    LABEL_DISAGG_ROLE,
     LABEL_MODEL,
     LABEL_WORKLOAD_ROLE,
     RESERVED_MOUNT_PATH,

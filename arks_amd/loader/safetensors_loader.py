@@ -93,8 +93,6 @@ def _chunk_complete(chunk: dict[str, torch.Tensor]) -> bool:
 def save_random_checkpoint(cfg, out_dir: str, seed: int = 0) -> None:
     """Write a random-init HF-layout checkpoint (config.json + safetensors)
     for tests of the loading path (no network for real checkpoints)."""
-    import dataclasses
-
     from safetensors.torch import save_file
 
     from ..models import create_model
