@@ -125,6 +125,9 @@ class AsyncEngine:
     def submit(self, request_id: str, token_ids: list[int],
                sampling: SamplingParams, hold_pages: bool = False) -> _Stream:
         st = _Stream()
+        if self.failed is not None:  # engine loop is gone — end immediately
+            st.queue.put_nowait(None)
+            return st
         self.streams[request_id] = st
         self.pending_adds.append(
             RequestAdd(request_id, token_ids, sampling.__dict__.copy(), hold_pages)

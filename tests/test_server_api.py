@@ -394,5 +394,12 @@ def test_engine_loop_failure_fails_streams_and_health(app):
         h = await client.get("/health")
         assert h.status_code == 503
         assert isinstance(eng.failed, RuntimeError)
+        # submissions after the failure end immediately instead of hanging
+        r2 = await client.post("/v1/completions", json={
+            "model": "tiny", "prompt": "y", "max_tokens": 3,
+            "temperature": 0, "ignore_eos": True,
+        })
+        assert r2.status_code == 200
+        assert r2.json()["usage"]["completion_tokens"] == 0
 
     run_with_client(app, fn)
