@@ -164,7 +164,7 @@ class LLMEngine:
         each sequence's history, extend its KV pages to cover them, verify
         in one forward, emit 1..k+1 tokens per seq (greedy-exact)."""
         from .kv_cache import BlockAllocator
-        from .spec import eligible, propose_ngram
+        from .spec import eligible, propose_ngram_cached
 
         alloc = self.scheduler.allocator
         bs = self.cfg.block_size
@@ -174,7 +174,7 @@ class LLMEngine:
             d: list[int] = []
             room = self.cfg.max_model_len - seq.num_tokens
             if room > 0 and eligible(seq):
-                d = propose_ngram(seq.all_token_ids, min(k, room))
+                d = propose_ngram_cached(seq, min(k, room))
             if d:
                 need = (
                     BlockAllocator.blocks_needed(seq.num_tokens + len(d), bs)

@@ -53,6 +53,35 @@ def propose_ngram(tokens: list[int], k: int) -> list[int]:
     return []
 
 
+def propose_ngram_cached(seq: Sequence, k: int) -> list[int]:
+    """Same result as propose_ngram, O(1) amortized per decode step: each
+    sequence carries {n-gram -> latest end position} maps, extended only
+    for tokens appended since the last call (a full backwards scan per step
+    costs O(T) Python time per sequence — milliseconds at serving batch
+    sizes, which would eat the speculative win)."""
+    st = getattr(seq, "_spec_ngrams", None)
+    if st is None:
+        st = seq._spec_ngrams = {"maps": {n: {} for n in NGRAM_SIZES}, "done": 0}
+    toks = seq.all_token_ids
+    T = len(toks)
+    for n in NGRAM_SIZES:
+        m = st["maps"][n]
+        # register grams ending at p (follower = toks[p]); the trailing
+        # gram (ending at T) has no follower yet and must not self-match
+        for p in range(max(st["done"], n), T):
+            m[tuple(toks[p - n:p])] = p
+    st["done"] = T
+    for n in NGRAM_SIZES:
+        if T <= n:
+            continue
+        p = st["maps"][n].get(tuple(toks[T - n:]))
+        if p is not None:
+            nxt = toks[p:p + k]
+            if nxt:
+                return list(nxt)
+    return []
+
+
 def eligible(seq: Sequence) -> bool:
     """Speculate only where greedy acceptance is exact."""
     sp = seq.sampling

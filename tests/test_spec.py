@@ -162,3 +162,23 @@ def test_spec_server_stream_e2e():
                 assert usage["completion_tokens"] == 12
 
     asyncio.new_event_loop().run_until_complete(go())
+
+
+def test_propose_ngram_cached_equals_pure():
+    """The incremental n-gram map must reproduce the pure backwards-scan
+    proposer as a sequence grows token by token."""
+    import random
+
+    from arks_amd.engine.sequence import Sequence
+    from arks_amd.engine.spec import propose_ngram, propose_ngram_cached
+
+    rng = random.Random(5)
+    for trial in range(20):
+        start = [rng.randrange(1, 5) for _ in range(rng.randrange(1, 10))]
+        seq = Sequence(list(start))
+        for _ in range(40):
+            toks = seq.all_token_ids
+            for k in (1, 4):
+                assert propose_ngram_cached(seq, k) == propose_ngram(toks, k), (
+                    trial, toks, k)
+            seq.append_token(rng.randrange(1, 5))
