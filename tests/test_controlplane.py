@@ -273,3 +273,25 @@ def test_disaggregated_app_flow():
     op.reconcile_until_stable()
     dapp = store.get("ArksDisaggregatedApplication", "default", "d1")
     assert dapp.status.phase is ApplicationPhase.RUNNING
+
+
+def test_app_replica_scale_propagates_to_workload():
+    """kubectl scale (spec.replicas change) must propagate to the generated
+    workload on the next reconcile — the HPA path (reference :400-457)."""
+    store = Store()
+    op = Operator(store)
+    store.create(mk_app())
+    make_model_ready(store, op)
+    op.reconcile_until_stable()
+    rbgs = store.get("RoleBasedGroupSet", "default", "a1")
+    assert rbgs["spec"]["replicas"] == 1
+
+    app = store.get("ArksApplication", "default", "a1")
+    app.spec.replicas = 4
+    store.update(app)
+    op.reconcile_until_stable()
+    rbgs = store.get("RoleBasedGroupSet", "default", "a1")
+    assert rbgs["spec"]["replicas"] == 4
+    # status no longer satisfies 4 replicas -> leaves Running
+    appx = store.get("ArksApplication", "default", "a1")
+    assert appx.status.phase is not None
