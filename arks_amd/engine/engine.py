@@ -164,7 +164,7 @@ class LLMEngine:
         each sequence's history, extend its KV pages to cover them, verify
         in one forward, emit 1..k+1 tokens per seq (greedy-exact)."""
         from .kv_cache import BlockAllocator
-        from .spec import eligible, propose_ngram_cached
+        from .spec import eligible, eligible_sampled, propose_ngram_cached
 
         alloc = self.scheduler.allocator
         bs = self.cfg.block_size
@@ -173,7 +173,7 @@ class LLMEngine:
         for seq in sb.seqs:
             d: list[int] = []
             room = self.cfg.max_model_len - seq.num_tokens
-            if room > 0 and eligible(seq):
+            if room > 0 and (eligible(seq) or eligible_sampled(seq)):
                 d = propose_ngram_cached(seq, min(k, room))
             if d:
                 need = (

@@ -463,7 +463,36 @@ class ModelRunner:
         logits = self.model(fb, self.kv_caches)  # [rows, vocab]
         greedy = ops.greedy_sample(logits.contiguous()).tolist()
         emitted: list[list[int] | None] = [None] * len(sb.seqs)
-        other = [i for i, s in enumerate(sb.seqs) if not eligible(s)]
+        # sampled-eligible seqs with drafts: exact rejection sampling
+        from .spec import eligible_sampled, reject_sample_token
+
+        rej = [i for i, (s, d) in enumerate(zip(sb.seqs, drafts))
+               if d and not eligible(s) and eligible_sampled(s)]
+        for i in rej:
+            seq, d = sb.seqs[i], drafts[i]
+            qn = cu[i + 1] - cu[i]
+            rows = logits[cu[i]:cu[i + 1]].float()
+            sp = seq.sampling
+            if sp.top_p < 1.0 or sp.top_k > 0 or sp.min_p > 0.0:
+                rows = self._apply_top_p_top_k(rows, [seq] * qn)
+            probs = torch.softmax(rows / sp.temperature, dim=-1)
+            u = torch.rand(2, qn, device=logits.device).tolist()
+            out: list[int] = []
+            for j, dt in enumerate(d):
+                ok, tok = reject_sample_token(probs[j], dt, u[0][j], u[1][j])
+                out.append(tok)
+                if not ok:
+                    break
+            else:
+                # full acceptance: bonus token by inverse CDF of the last row
+                cdfb = torch.cumsum(probs[qn - 1], 0)
+                out.append(int(torch.searchsorted(
+                    cdfb, torch.tensor(u[1][qn - 1], dtype=cdfb.dtype,
+                                       device=cdfb.device)
+                ).clamp(max=cdfb.numel() - 1)))
+            emitted[i] = out
+        other = [i for i, s in enumerate(sb.seqs)
+                 if not eligible(s) and emitted[i] is None]
         if other:
             rows = torch.tensor([cu[i + 1] - 1 for i in other],
                                 dtype=torch.long, device=logits.device)

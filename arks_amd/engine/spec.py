@@ -82,8 +82,33 @@ def propose_ngram_cached(seq: Sequence, k: int) -> list[int]:
     return []
 
 
+def reject_sample_token(probs, draft: int, u_accept: float,
+                        u_pick: float) -> tuple[bool, int]:
+    """One speculative-sampling step with a DETERMINISTIC proposal (the
+    n-gram draft): accept `draft` with probability probs[draft]; otherwise
+    sample from the residual (probs with the draft zeroed, renormalized)
+    by inverse CDF. Marginal over (u_accept, u_pick) is exactly `probs` —
+    the standard spec-sampling identity with q = delta(draft).
+    Returns (accepted, token)."""
+    import torch
+
+    pd = float(probs[draft])
+    if u_accept < pd:
+        return True, draft
+    residual = probs.clone()
+    residual[draft] = 0
+    z = float(residual.sum())
+    if z <= 0.0:  # degenerate: draft carries all mass
+        return True, draft
+    cdf = torch.cumsum(residual / z, 0)
+    tok = int(torch.searchsorted(
+        cdf, torch.tensor(u_pick, dtype=cdf.dtype)
+    ).clamp(max=cdf.numel() - 1))
+    return False, tok
+
+
 def eligible(seq: Sequence) -> bool:
-    """Speculate only where greedy acceptance is exact."""
+    """Greedy speculation: acceptance is plain argmax agreement."""
     sp = seq.sampling
     return (
         sp.temperature == 0.0
@@ -91,4 +116,23 @@ def eligible(seq: Sequence) -> bool:
         and sp.logprobs is None
         and not sp.logit_bias
         and sp.min_tokens == 0
+    )
+
+
+def eligible_sampled(seq: Sequence) -> bool:
+    """Sampled speculation via rejection sampling — distribution-exact for
+    temperature sampling with top-p/top-k/min-p filters (the filters are
+    applied to the verify logits before acceptance). Excluded: penalties
+    (their logits depend on the evolving token set inside the draft),
+    logprobs (reported values would mix accept/resample paths), logit_bias
+    / min_tokens (kept on the plain sampler), per-request seeds (their
+    generator's draw sequence is part of the reproducibility contract)."""
+    sp = seq.sampling
+    return (
+        sp.temperature > 0.0
+        and not sp.has_penalties
+        and sp.logprobs is None
+        and not sp.logit_bias
+        and sp.min_tokens == 0
+        and sp.seed is None
     )

@@ -182,3 +182,81 @@ def test_propose_ngram_cached_equals_pure():
                 assert propose_ngram_cached(seq, k) == propose_ngram(toks, k), (
                     trial, toks, k)
             seq.append_token(rng.randrange(1, 5))
+
+
+def test_reject_sample_token_math():
+    """Deterministic-proposal speculative sampling: acceptance boundary at
+    p(draft); residual removes the draft and renormalizes."""
+    from arks_amd.engine.spec import reject_sample_token
+
+    p = torch.tensor([0.5, 0.3, 0.2])
+    assert reject_sample_token(p, 0, 0.49, 0.0) == (True, 0)
+    # rejected: residual over [1, 2] is [0.6, 0.4]
+    assert reject_sample_token(p, 0, 0.51, 0.59) == (False, 1)
+    assert reject_sample_token(p, 0, 0.51, 0.61) == (False, 2)
+    # degenerate: draft carries all mass -> accept regardless
+    q = torch.tensor([1.0, 0.0, 0.0])
+    assert reject_sample_token(q, 0, 0.999, 0.5) == (True, 0)
+
+
+def test_reject_sample_marginal_matches_target():
+    """Empirical marginal of accept-or-resample equals the target
+    distribution (the spec-sampling identity), generous tolerance."""
+    from arks_amd.engine.spec import reject_sample_token
+
+    torch.manual_seed(0)
+    p = torch.tensor([0.6, 0.3, 0.1])
+    n = 20000
+    u = torch.rand(2, n).tolist()
+    counts = [0, 0, 0]
+    for i in range(n):
+        _, tok = reject_sample_token(p, 1, u[0][i], u[1][i])
+        counts[tok] += 1
+    freq = [c / n for c in counts]
+    assert all(abs(f - t) < 0.02 for f, t in zip(freq, p.tolist())), freq
+
+
+def test_spec_sampled_requests_deterministic_and_drafted():
+    """Unseeded temperature sampling under speculation: same global seed
+    -> same outputs, and drafts actually flow on repetitive text."""
+    prompts = [[1, 2, 3, 4] * 8, [2, 9] * 10]
+    sp = SamplingParams(max_tokens=16, ignore_eos=True, temperature=0.8,
+                        top_p=0.9)
+
+    def run():
+        torch.manual_seed(3)
+        e = LLMEngine(_cfg(spec="ngram"))
+        out = e.generate([list(p) for p in prompts], sp)
+        return out, e.spec_drafted_tokens, e.spec_accepted_tokens
+
+    out1, drafted1, accepted1 = run()
+    out2, drafted2, _ = run()
+    assert out1 == out2 and drafted1 == drafted2
+    assert 0 <= accepted1 <= drafted1
+    assert all(len(o) == 16 for o in out1)
+    # sampled outputs rarely repeat, so drive the rejection verify path
+    # directly: a decode batch with an injected draft emits 1..len(d)+1
+    # tokens deterministically under a fixed global seed
+    torch.manual_seed(4)
+    e = LLMEngine(_cfg(spec="ngram"))
+    seq = e.add_request([4, 9, 2, 7, 7, 1],
+                        SamplingParams(max_tokens=64, ignore_eos=True,
+                                       temperature=0.8, top_p=0.9))
+    while e.scheduler.num_waiting:
+        e.step()
+    sb = e.scheduler.schedule()
+    assert not sb.is_prefill and sb.seqs == [seq]
+    from arks_amd.engine.kv_cache import BlockAllocator
+
+    draft = [5, 9, 3]
+    need = (BlockAllocator.blocks_needed(seq.num_tokens + len(draft),
+                                         e.cfg.block_size)
+            - len(seq.block_table))
+    if need > 0:
+        seq.block_table.extend(e.scheduler.allocator.allocate(need))
+    torch.manual_seed(5)
+    got1 = e.runner.execute_spec(sb, [list(draft)])[0]
+    assert 1 <= len(got1) <= len(draft) + 1
+    torch.manual_seed(5)
+    got2 = e.runner.execute_spec(sb, [list(draft)])[0]
+    assert got1 == got2
