@@ -201,3 +201,31 @@ def test_spec_decode_gpu():
     # first token comes from the prefill forward — identical path/kernels
     # in both engines
     assert [o[0] for o in out1] == [o[0] for o in ref]
+
+
+def test_spec_rejection_sampling_gpu():
+    """Sampled speculation on GPU: inject a draft into a decode batch and
+    run the rejection-verify path on the native extend kernel."""
+    torch.manual_seed(0)
+    e = LLMEngine(EngineConfig(
+        preset="tiny-gpu", device="cuda", kv_cache_blocks=256,
+        max_model_len=512, max_num_seqs=16, speculative="ngram",
+    ))
+    seq = e.add_request([4, 9, 2, 7, 7, 1],
+                        SamplingParams(max_tokens=64, ignore_eos=True,
+                                       temperature=0.8, top_p=0.9))
+    while e.scheduler.num_waiting:
+        e.step()
+    sb = e.scheduler.schedule()
+    assert not sb.is_prefill
+    from arks_amd.engine.kv_cache import BlockAllocator
+
+    draft = [5, 9, 3]
+    need = (BlockAllocator.blocks_needed(seq.num_tokens + len(draft),
+                                         e.cfg.block_size)
+            - len(seq.block_table))
+    if need > 0:
+        seq.block_table.extend(e.scheduler.allocator.allocate(need))
+    got = e.runner.execute_spec(sb, [list(draft)])[0]
+    assert 1 <= len(got) <= len(draft) + 1
+    assert all(0 <= t < e.model_cfg.vocab_size for t in got)
