@@ -9,9 +9,11 @@ with exactly the greedy output the non-speculative engine produces.
 
 The reference's runtime images (vLLM/SGLang) ship the same speculator as
 `--speculative-model [ngram]` / prompt-lookup; here it is first-party.
-Opt-in via EngineConfig.speculative="ngram"; only greedy requests without
-penalties/logprobs are speculated (others decode normally in-batch), which
-keeps acceptance exact rather than distribution-approximate.
+Opt-in via EngineConfig.speculative="ngram". Greedy requests accept by
+argmax agreement; plain temperature requests accept by exact rejection
+sampling (reject_sample_token below); everything else (penalties, bias,
+logprobs, seeds) decodes normally in the same batch — acceptance is always
+exact, never distribution-approximate.
 
 Exactness caveat (same property as vLLM's greedy verify): "exact" means
 exact w.r.t. the verify forward's own logits. On GPU the verify runs the
@@ -102,7 +104,7 @@ def reject_sample_token(probs, draft: int, u_accept: float,
         return True, draft
     cdf = torch.cumsum(residual / z, 0)
     tok = int(torch.searchsorted(
-        cdf, torch.tensor(u_pick, dtype=cdf.dtype)
+        cdf, torch.tensor(u_pick, dtype=cdf.dtype, device=cdf.device)
     ).clamp(max=cdf.numel() - 1))
     return False, tok
 
