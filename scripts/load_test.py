@@ -34,6 +34,8 @@ def parse_args():
     p.add_argument("--input-len", type=int, default=512)
     p.add_argument("--output-len", type=int, default=64)
     p.add_argument("--server-args", default="", help="extra server flags")
+    p.add_argument("--prompt-style", choices=["random", "repetitive"],
+                   default="random")
     p.add_argument("--no-spawn", action="store_true",
                    help="assume a server is already on --port")
     return p.parse_args()
@@ -41,7 +43,14 @@ def parse_args():
 
 async def one_request(client, args, rid, results):
     # ByteTokenizer-friendly prompt of the requested token length
-    prompt = "".join(random.choice("abcdefgh ") for _ in range(args.input_len))
+    if args.prompt_style == "repetitive":
+        # periodic text: exercises prompt-lookup speculation
+        # (--server-args "--speculative ngram") and the prefix cache
+        unit = "".join(random.choice("abcdefgh ") for _ in range(16))
+        prompt = (unit * (args.input_len // 16 + 1))[: args.input_len]
+    else:
+        prompt = "".join(
+            random.choice("abcdefgh ") for _ in range(args.input_len))
     t0 = time.time()
     ttft = None
     n_tok = 0
