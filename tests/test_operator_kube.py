@@ -101,3 +101,34 @@ def test_leader_election_lease():
     lease = store.get_opt("Lease", "arks-system", a.name)
     assert lease["spec"]["holderIdentity"] == "op-a"
     assert lease["spec"]["leaseTransitions"] == 2
+
+
+def test_leader_election_renew_loop_and_loss_callback():
+    """run_renew keeps renewing while leading and fires on_lost exactly
+    once when another candidate has stolen the Lease."""
+    import threading
+
+    from arks_amd.controlplane.leaderelect import LeaderElector
+
+    fake, store = mk()
+    t = [0.0]
+    clock = lambda: t[0]  # noqa: E731
+    a = LeaderElector(store, "op-a", lease_s=15, renew_s=0.01, clock=clock)
+    b = LeaderElector(store, "op-b", lease_s=15, renew_s=5, clock=clock)
+    assert a.try_acquire()
+    lost = threading.Event()
+    th = threading.Thread(target=a.run_renew, args=(lost.set,), daemon=True)
+    th.start()
+    # renewals at a fixed clock keep op-a leading
+    import time as _time
+
+    _time.sleep(0.1)
+    assert a.is_leader and not lost.is_set()
+    # op-a's clock stalls 20s while op-b advances -> op-b steals; op-a's
+    # next renewal must fail and report loss
+    t[0] += 20.0
+    assert b.try_acquire()
+    lost.wait(timeout=5.0)
+    assert lost.is_set() and not a.is_leader
+    th.join(timeout=5.0)
+    assert not th.is_alive()
