@@ -403,3 +403,37 @@ def test_engine_loop_failure_fails_streams_and_health(app):
         assert r2.json()["usage"]["completion_tokens"] == 0
 
     run_with_client(app, fn)
+
+
+def test_chat_and_completion_token_logprobs(app):
+    """Per-generated-token logprobs on both endpoints (OpenAI shapes:
+    chat logprobs.content[], completions token_logprobs/top_logprobs)."""
+
+    async def fn(client):
+        r = await client.post("/v1/chat/completions", json={
+            "model": "tiny",
+            "messages": [{"role": "user", "content": "hi"}],
+            "max_tokens": 3, "temperature": 0, "ignore_eos": True,
+            "logprobs": True, "top_logprobs": 2,
+        })
+        assert r.status_code == 200, r.text
+        content = r.json()["choices"][0]["logprobs"]["content"]
+        assert len(content) == 3
+        for entry in content:
+            assert entry["logprob"] <= 0.0
+            assert len(entry["top_logprobs"]) == 2
+            # chosen-token logprob can't beat the best alternative
+            assert entry["logprob"] <= max(
+                t["logprob"] for t in entry["top_logprobs"]) + 1e-6
+
+        r2 = await client.post("/v1/completions", json={
+            "model": "tiny", "prompt": "abc", "max_tokens": 4,
+            "temperature": 0, "ignore_eos": True, "logprobs": 1,
+        })
+        lp = r2.json()["choices"][0]["logprobs"]
+        assert len(lp["token_logprobs"]) == 4
+        assert all(v <= 0.0 for v in lp["token_logprobs"])
+        assert len(lp["top_logprobs"]) == 4 and all(
+            len(d) == 1 for d in lp["top_logprobs"])
+
+    run_with_client(app, fn)
