@@ -280,6 +280,7 @@ def create_app(engine: AsyncEngine, served_model_name: str, tokenizer,
         finish = None
         tracker = StopStringTracker(_stop_list(req.stop))
         detok = IncrementalDetokenizer(tokenizer)
+        pending_lp: list[dict] = []  # per-token entries since the last emit
         try:
             async for out in engine.generate_stream(
             rid, token_ids, sp, prefill_addr=raw.headers.get("x-arks-prefill-addr")
@@ -288,6 +289,15 @@ def create_app(engine: AsyncEngine, served_model_name: str, tokenizer,
                     engine.abort(rid)
                     return
                 n_out += 1
+                if out.logprob is not None:
+                    pending_lp.append({
+                        "token": tokenizer.decode([out.new_token_id]),
+                        "logprob": out.logprob,
+                        "top_logprobs": [
+                            {"token": tokenizer.decode([t]), "logprob": v}
+                            for t, v in (out.top_logprobs or {}).items()
+                        ],
+                    })
                 text_in = detok.feed(out.new_token_id)
                 if out.finished:
                     text_in += detok.flush()
@@ -304,8 +314,12 @@ def create_app(engine: AsyncEngine, served_model_name: str, tokenizer,
                         id=rid, model=req.model, created=created,
                         choices=[ChatDeltaChoice(
                             delta={"content": piece},
+                            logprobs=(
+                                {"content": pending_lp} if pending_lp else None
+                            ),
                             finish_reason=finish if done else None)],
                     )
+                    pending_lp = []
                     yield f"data: {chunk.model_dump_json(exclude_none=True)}\n\n"
                 if stopped:
                     break

@@ -83,6 +83,7 @@ class ChatChoice(BaseModel):
 class ChatDeltaChoice(BaseModel):
     index: int = 0
     delta: dict[str, Any] = Field(default_factory=dict)
+    logprobs: dict[str, Any] | None = None  # {"content": [...]} when requested
     finish_reason: str | None = None
 
 

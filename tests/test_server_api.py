@@ -437,3 +437,31 @@ def test_chat_and_completion_token_logprobs(app):
             len(d) == 1 for d in lp["top_logprobs"])
 
     run_with_client(app, fn)
+
+
+def test_chat_stream_logprobs(app):
+    """Streamed chat with logprobs: chunks carry per-token logprob entries
+    (held-token merges allowed), totalling one entry per generated token."""
+
+    async def fn(client):
+        async with client.stream("POST", "/v1/chat/completions", json={
+            "model": "tiny",
+            "messages": [{"role": "user", "content": "hello"}],
+            "max_tokens": 4, "temperature": 0, "ignore_eos": True,
+            "stream": True, "logprobs": True, "top_logprobs": 1,
+        }) as r:
+            assert r.status_code == 200
+            entries = []
+            async for line in r.aiter_lines():
+                if not line.startswith("data: ") or line == "data: [DONE]":
+                    continue
+                c = json.loads(line[6:])
+                for ch in c.get("choices", []):
+                    lp = ch.get("logprobs")
+                    if lp:
+                        entries.extend(lp["content"])
+            assert len(entries) == 4
+            assert all(e["logprob"] <= 0.0 and len(e["top_logprobs"]) == 1
+                       for e in entries)
+
+    run_with_client(app, fn)
