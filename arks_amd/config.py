@@ -43,10 +43,14 @@ class ModelConfig:
     torch_dtype: str = "bfloat16"
 
     def __post_init__(self):
-        if self.sliding_window is not None:
+        # sliding_window is allowed to be SET (Mistral-v0.1-style configs
+        # declare 4096) — the engine enforces max_model_len <= window at
+        # startup so the window never actually binds (full attention below
+        # the window is identical to SWA); true windowed attention is a
+        # next-round kernel feature.
+        if self.sliding_window is not None and self.sliding_window <= 0:
             raise ValueError(
-                "sliding-window attention is not supported yet; this config "
-                f"sets sliding_window={self.sliding_window}"
+                f"invalid sliding_window={self.sliding_window}"
             )
 
     @classmethod

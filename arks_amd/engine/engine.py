@@ -18,6 +18,15 @@ class LLMEngine:
     def __init__(self, cfg: EngineConfig):
         self.cfg = cfg
         self.model_cfg = cfg.model_config()
+        sw = self.model_cfg.sliding_window
+        if sw is not None and cfg.max_model_len > sw:
+            # below the window, full attention == sliding-window attention;
+            # beyond it we would silently change the model's semantics
+            raise ValueError(
+                f"model declares sliding_window={sw} and windowed attention "
+                f"is not implemented: set max_model_len <= {sw} "
+                f"(requested {cfg.max_model_len})"
+            )
         self.runner = ModelRunner(cfg, self.model_cfg)
         t0 = time.time()
         self.runner.load_weights()

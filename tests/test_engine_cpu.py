@@ -277,3 +277,29 @@ def test_prompt_logprobs_chunked_matches_unchunked():
     assert all(math.isclose(a, b, rel_tol=1e-4, abs_tol=1e-5)
                for a, b in zip(small1, big1))
     assert all(v <= 0.0 for v in big1)
+
+
+def test_sliding_window_gate():
+    """Configs declaring sliding_window are servable while the window never
+    binds (max_model_len <= window); beyond it the engine refuses loudly
+    instead of silently changing attention semantics."""
+    import dataclasses
+
+    from arks_amd.config import PRESET_CONFIGS, EngineConfig
+    from arks_amd.engine import LLMEngine, SamplingParams
+
+    swcfg = dataclasses.replace(PRESET_CONFIGS["tiny"], sliding_window=128)
+    import arks_amd.config as C
+
+    C.PRESET_CONFIGS["tiny-swa"] = swcfg
+    try:
+        e = LLMEngine(EngineConfig(preset="tiny-swa", device="cpu",
+                                   kv_cache_blocks=64, max_model_len=128))
+        out = e.generate([[5, 2, 8]], SamplingParams(max_tokens=4,
+                                                     ignore_eos=True))
+        assert len(out[0]) == 4
+        with pytest.raises(ValueError, match="sliding_window"):
+            LLMEngine(EngineConfig(preset="tiny-swa", device="cpu",
+                                   kv_cache_blocks=64, max_model_len=256))
+    finally:
+        C.PRESET_CONFIGS.pop("tiny-swa", None)
