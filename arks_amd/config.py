@@ -77,7 +77,9 @@ class ModelConfig:
             qk_norm=arch.startswith("Qwen3"),
             num_local_experts=cfg.get("num_local_experts",
                                       cfg.get("num_experts", 0)),
-            sliding_window=cfg.get("sliding_window"),
+            # Qwen2-family configs declare a window but disable it
+            sliding_window=(cfg.get("sliding_window")
+                            if cfg.get("use_sliding_window", True) else None),
             num_experts_per_tok=cfg.get("num_experts_per_tok", 2),
             moe_intermediate_size=cfg.get("moe_intermediate_size"),
             shared_expert_intermediate_size=cfg.get(

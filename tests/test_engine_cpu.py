@@ -303,3 +303,19 @@ def test_sliding_window_gate():
                                    kv_cache_blocks=64, max_model_len=256))
     finally:
         C.PRESET_CONFIGS.pop("tiny-swa", None)
+
+
+def test_use_sliding_window_false_ignored():
+    """Qwen2.5-style configs: sliding_window declared but
+    use_sliding_window=false means full attention — no engine gate."""
+    from arks_amd.config import ModelConfig
+
+    mc = ModelConfig.from_hf_config({
+        "architectures": ["Qwen2ForCausalLM"], "sliding_window": 131072,
+        "use_sliding_window": False,
+    })
+    assert mc.sliding_window is None
+    mc2 = ModelConfig.from_hf_config({
+        "architectures": ["MistralForCausalLM"], "sliding_window": 4096,
+    })
+    assert mc2.sliding_window == 4096
