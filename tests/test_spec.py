@@ -260,3 +260,29 @@ def test_spec_sampled_requests_deterministic_and_drafted():
     torch.manual_seed(5)
     got2 = e.runner.execute_spec(sb, [list(draft)])[0]
     assert got1 == got2
+
+
+def test_spec_after_disagg_inject_matches_plain():
+    """Speculation on a decode instance over injected (remotely prefilled)
+    KV must emit the plain engine's greedy continuation."""
+    prompt = [1, 2, 3, 4] * 8
+    sp1 = SamplingParams(max_tokens=1, ignore_eos=True)
+    spn = SamplingParams(max_tokens=10, ignore_eos=True)
+
+    def run(spec):
+        torch.manual_seed(0)
+        a = LLMEngine(_cfg())
+        torch.manual_seed(0)
+        b = LLMEngine(_cfg(spec=spec))
+        seq = a.add_request(list(prompt), sp1, request_id="p", hold_pages=True)
+        while not seq.is_finished:
+            a.step()
+        _, kv = a.extract_prefilled("p")
+        b.add_prefilled(list(prompt), seq.output_token_ids[0], kv, spn, "d")
+        toks = [seq.output_token_ids[0]]
+        while b.has_work():
+            for o in b.step():
+                toks.append(o.new_token_id)
+        return toks
+
+    assert run("ngram") == run(None)
