@@ -184,6 +184,9 @@ def main():
                 "parallelism": f"{args.parallel}{world}",
                 "ttft_ms_p50": round(ttft_ms, 2) if ttft_ms is not None else None,
                 "weight_load_s": round(load_s, 2),
+                **({"spec_drafted": engine.spec_drafted_tokens,
+                    "spec_accepted": engine.spec_accepted_tokens}
+                   if args.speculative else {}),
             },
         }
         print(json.dumps(result))
