@@ -635,7 +635,15 @@ class ModelRunner:
         sub = logits[idx].float()
         sorted_logits, sorted_idx = sub.sort(dim=-1, descending=True)
         keep = torch.ones_like(sorted_logits, dtype=torch.bool)
-        probs = torch.softmax(sorted_logits, dim=-1)
+        # Temperature applies BEFORE the top-p/min-p nucleus (vLLM
+        # semantics): the kept set is computed from softmax(logits/T).
+        # Sort order is T-invariant, so only the probs need scaling; the
+        # masked logits stay unscaled (sampling divides by T later).
+        temps = torch.tensor(
+            [seqs[i].sampling.temperature for i in rows],
+            dtype=torch.float32, device=logits.device,
+        ).unsqueeze(1)
+        probs = torch.softmax(sorted_logits / temps, dim=-1)
         cum = probs.cumsum(dim=-1)
         for j, i in enumerate(rows):
             sp = seqs[i].sampling

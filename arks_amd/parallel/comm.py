@@ -85,8 +85,14 @@ def tp_all_gather(x: torch.Tensor, dim: int = -1) -> torch.Tensor:
     """All-gather shards along `dim` (no-op at TP=1)."""
     if _TP_WORLD <= 1:
         return x
-    parts = [torch.empty_like(x) for _ in range(_TP_WORLD)]
-    dist.all_gather(parts, x.contiguous(), group=_TP_GROUP)
+    # fp8 has no collective support on gloo (nor NCCL in common torch
+    # versions) — ship the bytes and view back.
+    fp8 = x.dtype == torch.float8_e4m3fn
+    wire = x.contiguous().view(torch.uint8) if fp8 else x.contiguous()
+    parts = [torch.empty_like(wire) for _ in range(_TP_WORLD)]
+    dist.all_gather(parts, wire, group=_TP_GROUP)
+    if fp8:
+        parts = [p.view(torch.float8_e4m3fn) for p in parts]
     return torch.cat(parts, dim=dim)
 
 

@@ -12,6 +12,7 @@ ServiceMonitor.
 from __future__ import annotations
 
 import asyncio
+import dataclasses
 import json
 import time
 import uuid
@@ -52,6 +53,15 @@ def _stop_list(stop) -> list[str]:
     if stop is None:
         return []
     return [stop] if isinstance(stop, str) else [x for x in stop if x]
+
+
+def _choice_sp(sp: SamplingParams, i: int) -> SamplingParams:
+    """Per-choice SamplingParams for n>1 fan-out: a fixed seed must still
+    produce n DISTINCT choices (identical seeds would sample n copies), so
+    each choice derives seed+i; choice 0 keeps the request's exact seed."""
+    if sp.seed is None or i == 0:
+        return sp
+    return dataclasses.replace(sp, seed=sp.seed + i)
 
 
 class StopStringTracker:
@@ -148,6 +158,10 @@ def create_app(engine: AsyncEngine, served_model_name: str, tokenizer,
     async def _collect_choice(rid, token_ids, sp, stops, prefill_addr):
         """One non-streamed choice: (token_ids, text, finish, n_generated)."""
         tracker = StopStringTracker(stops)
+        # incremental detok even in the non-stream path: per-token decode
+        # garbles multi-byte runes split across BPE tokens (and stop
+        # matching would then run over the garbled text)
+        detok = IncrementalDetokenizer(tokenizer)
         out_ids: list[int] = []
         text_acc = ""
         finish = None
@@ -156,8 +170,7 @@ def create_app(engine: AsyncEngine, served_model_name: str, tokenizer,
                                                     prefill_addr=prefill_addr):
                 out_ids.append(out.new_token_id)
                 if stops:
-                    emit, stopped = tracker.feed(
-                        tokenizer.decode([out.new_token_id]))
+                    emit, stopped = tracker.feed(detok.feed(out.new_token_id))
                     text_acc += emit
                     if stopped:
                         finish = "stop"
@@ -169,7 +182,8 @@ def create_app(engine: AsyncEngine, served_model_name: str, tokenizer,
             engine.abort(rid)  # client went away mid-request
             raise
         if stops and finish != "stop":
-            text_acc += tracker.flush()
+            emit, _ = tracker.feed(detok.flush())
+            text_acc += emit + tracker.flush()
         text = text_acc if stops else tokenizer.decode(out_ids)
         return out_ids, text, finish or "stop", len(out_ids)
 
@@ -195,7 +209,8 @@ def create_app(engine: AsyncEngine, served_model_name: str, tokenizer,
             paddr = raw.headers.get("x-arks-prefill-addr")
             stops_n = _stop_list(req.stop)
             results = await asyncio.gather(*[
-                _collect_choice(f"{rid}-{i}", token_ids, sp, stops_n, paddr)
+                _collect_choice(f"{rid}-{i}", token_ids, _choice_sp(sp, i),
+                                stops_n, paddr)
                 for i in range(req.n)
             ])
             total_out = sum(r[3] for r in results)
@@ -213,6 +228,7 @@ def create_app(engine: AsyncEngine, served_model_name: str, tokenizer,
             ).model_dump()
         stops = _stop_list(req.stop)
         tracker = StopStringTracker(stops)
+        detok = IncrementalDetokenizer(tokenizer)
         text_ids: list[int] = []
         text_acc = ""
         lp_content: list[dict] = []
@@ -233,8 +249,7 @@ def create_app(engine: AsyncEngine, served_model_name: str, tokenizer,
                         ],
                     })
                 if stops:
-                    emit, stopped = tracker.feed(
-                        tokenizer.decode([out.new_token_id]))
+                    emit, stopped = tracker.feed(detok.feed(out.new_token_id))
                     text_acc += emit
                     if stopped:
                         finish = "stop"
@@ -246,7 +261,8 @@ def create_app(engine: AsyncEngine, served_model_name: str, tokenizer,
             engine.abort(rid)  # client went away mid-request
             raise
         if stops and finish != "stop":
-            text_acc += tracker.flush()
+            emit, _ = tracker.feed(detok.flush())
+            text_acc += emit + tracker.flush()
         usage = Usage(
             prompt_tokens=len(token_ids),
             completion_tokens=len(text_ids),
@@ -377,7 +393,8 @@ def create_app(engine: AsyncEngine, served_model_name: str, tokenizer,
             paddr = raw.headers.get("x-arks-prefill-addr")
             stops_n = _stop_list(req.stop)
             results = await asyncio.gather(*[
-                _collect_choice(f"{rid}-{i}", token_ids, sp, stops_n, paddr)
+                _collect_choice(f"{rid}-{i}", token_ids, _choice_sp(sp, i),
+                                stops_n, paddr)
                 for i in range(req.n)
             ])
             total_out = sum(r[3] for r in results)
@@ -393,6 +410,7 @@ def create_app(engine: AsyncEngine, served_model_name: str, tokenizer,
             ).model_dump()
         stops = _stop_list(req.stop)
         tracker = StopStringTracker(stops)
+        detok = IncrementalDetokenizer(tokenizer)
         out_ids: list[int] = []
         text_acc = ""
         finish = None
@@ -414,8 +432,7 @@ def create_app(engine: AsyncEngine, served_model_name: str, tokenizer,
                         for t, v in (out.top_logprobs or {}).items()
                     })
                 if stops:
-                    emit, stopped = tracker.feed(
-                        tokenizer.decode([out.new_token_id]))
+                    emit, stopped = tracker.feed(detok.feed(out.new_token_id))
                     text_acc += emit
                     if stopped:
                         finish = "stop"
@@ -427,7 +444,8 @@ def create_app(engine: AsyncEngine, served_model_name: str, tokenizer,
             engine.abort(rid)  # client went away mid-request
             raise
         if stops and finish != "stop":
-            text_acc += tracker.flush()
+            emit, _ = tracker.feed(detok.flush())
+            text_acc += emit + tracker.flush()
         text = text_acc if stops else tokenizer.decode(out_ids)
         lp_obj = None
         if token_lps:
