@@ -242,6 +242,33 @@ def test_min_p_filter():
     assert out[1, 2] == float("-inf") and out[1, 3] == float("-inf")
 
 
+def test_temperature_scales_top_p_nucleus():
+    """The top-p/min-p kept set is computed from softmax(logits/T) (vLLM
+    semantics): a high temperature flattens the distribution so MORE tokens
+    enter the nucleus than at T=1."""
+    import torch
+
+    from arks_amd.engine.model_runner import ModelRunner
+    from arks_amd.engine.sequence import Sequence, SamplingParams
+
+    logits = torch.tensor([[4.0, 2.0, 0.0, -2.0]])
+    cold = [Sequence([1], SamplingParams(temperature=1.0, top_p=0.9))]
+    hot = [Sequence([1], SamplingParams(temperature=5.0, top_p=0.9))]
+    kept_cold = torch.isfinite(
+        ModelRunner._apply_top_p_top_k(logits.clone(), cold)[0]).sum()
+    kept_hot = torch.isfinite(
+        ModelRunner._apply_top_p_top_k(logits.clone(), hot)[0]).sum()
+    assert kept_hot > kept_cold
+    # and min_p: hot distribution keeps tokens the cold one drops
+    cold = [Sequence([1], SamplingParams(temperature=1.0, min_p=0.3))]
+    hot = [Sequence([1], SamplingParams(temperature=8.0, min_p=0.3))]
+    kept_cold = torch.isfinite(
+        ModelRunner._apply_top_p_top_k(logits.clone(), cold)[0]).sum()
+    kept_hot = torch.isfinite(
+        ModelRunner._apply_top_p_top_k(logits.clone(), hot)[0]).sum()
+    assert kept_hot > kept_cold
+
+
 def test_prompt_logprobs_chunked_matches_unchunked():
     """echo/prompt scoring: values accumulate across prefill chunks and
     must equal the single-chunk run; prefix-cache hits are bypassed so a
