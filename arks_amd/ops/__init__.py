@@ -114,27 +114,61 @@ def attention_prefill_varlen(q, k, v, cu_seqlens, seq_lens: list[int], scale: fl
     return ref.attention_prefill_varlen(q, k, v, cu_seqlens, scale)
 
 
+EXTEND2_QTILE = 256  # q rows per workgroup in the 8-wave ladder kernel
+# Sequences with at least this many new q tokens go to the 256-row ladder
+# kernel; shorter ones (decode-adjacent chunks, prefix-cache hits) stay on
+# the 64-row kernel where a mostly-empty 256-row tile would waste waves.
+EXTEND2_MIN_QLEN = 97
+
+
+def _split_extend_tiles(q_lens: list[int], kv_fp8: bool, device):
+    """Per-seq routing between the two extend kernels: (tiles64, tiles256)."""
+    t64, t256 = [], []
+    for i, n in enumerate(q_lens):
+        if kv_fp8 or n < EXTEND2_MIN_QLEN:
+            for q0 in range(0, n, PREFILL_QTILE):
+                t64.append((i, q0))
+        else:
+            for q0 in range(0, n, EXTEND2_QTILE):
+                t256.append((i, q0))
+    mk = lambda t: torch.tensor(t, dtype=torch.int32, device=device).reshape(-1, 2)
+    return mk(t64), mk(t256)
+
+
 def attention_extend_paged(
     q, k_cache, v_cache, block_tables, kv_lens, cu_seqlens_q,
-    q_lens: list[int], scale: float,
+    q_lens: list[int], scale: float, window: int = 0,
 ):
     """Causal attention of packed NEW tokens over each sequence's full paged
     KV history (cached prefix + new tokens already written by
     reshape_and_cache). The prefix-caching / chunked-prefill attention path.
     q_lens is the host-side per-seq new-token count (tile grid without a
-    D2H sync); kv_lens is the device int32 total kv length per seq."""
+    D2H sync); kv_lens is the device int32 total kv length per seq.
+    window > 0 = sliding-window attention.
+
+    Dispatch: big chunks run the 8-wave 32x32-MFMA ladder kernel
+    (attn_extend2.hip, 256-row q tiles); small chunks and fp8-KV runs use
+    the 4-wave 64-row kernel."""
     if q.is_cuda:
         out = torch.empty(
             (q.shape[0], q.shape[1], q.shape[2]), dtype=q.dtype, device=q.device
         )
-        tile_info = build_prefill_tiles(q_lens, q.device)
-        _native().attention_extend_paged(
-            out, q, k_cache, v_cache, block_tables, kv_lens, cu_seqlens_q,
-            tile_info, scale,
-        )
+        kv_fp8 = k_cache.dtype != torch.bfloat16
+        t64, t256 = _split_extend_tiles(q_lens, kv_fp8, q.device)
+        if t256.numel():
+            _native().attention_extend_paged2(
+                out, q, k_cache, v_cache, block_tables, kv_lens, cu_seqlens_q,
+                t256, scale, window,
+            )
+        if t64.numel():
+            _native().attention_extend_paged(
+                out, q, k_cache, v_cache, block_tables, kv_lens, cu_seqlens_q,
+                t64, scale, window,
+            )
         return out
     return ref.attention_extend_paged(
-        q, k_cache, v_cache, block_tables, kv_lens, cu_seqlens_q, scale
+        q, k_cache, v_cache, block_tables, kv_lens, cu_seqlens_q, scale,
+        window,
     )
 
 
@@ -153,7 +187,7 @@ def decode_num_partitions(num_seqs: int, num_kv_heads: int, max_blocks: int) -> 
 
 def attention_decode_paged(
     q, k_cache, v_cache, block_tables, seq_lens, scale: float,
-    num_partitions: int | None = None, part_out=None,
+    num_partitions: int | None = None, part_out=None, window: int = 0,
 ):
     if q.is_cuda:
         out = torch.empty(
@@ -174,10 +208,12 @@ def attention_decode_paged(
             part_out = q.new_empty(0, dtype=torch.float32)
         _native().attention_decode_paged(
             out, q, k_cache, v_cache, block_tables, seq_lens, scale,
-            part_out, num_partitions,
+            part_out, num_partitions, window,
         )
         return out
-    return ref.attention_decode_paged(q, k_cache, v_cache, block_tables, seq_lens, scale)
+    return ref.attention_decode_paged(
+        q, k_cache, v_cache, block_tables, seq_lens, scale, window
+    )
 
 
 _SKINNY_WS: dict = {}

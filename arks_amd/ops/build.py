@@ -25,6 +25,7 @@ SOURCES = [
     os.path.join(CSRC, "attn_decode.hip"),
     os.path.join(CSRC, "attn_prefill.hip"),
     os.path.join(CSRC, "attn_extend.hip"),
+    os.path.join(CSRC, "attn_extend2.hip"),
     os.path.join(CSRC, "quant_fp8.hip"),
     os.path.join(CSRC, "skinny_gemm.hip"),
     os.path.join(CSRC, "sampling.hip"),

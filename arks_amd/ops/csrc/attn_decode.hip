@@ -71,7 +71,7 @@ __global__ __launch_bounds__(256) void attn_decode_kernel(
     const int* __restrict__ block_tables,  // [S, max_blocks]
     const int* __restrict__ seq_lens,      // [S]
     const float scale, const int num_kv_heads, const int max_blocks,
-    const int64_t q_stride) {
+    const int64_t q_stride, const int window) {
   constexpr int NUM_WAVES = 4;
   constexpr int STEPS = HEAD_DIM / 32;   // QK^T contraction steps
   constexpr int CHUNKS = HEAD_DIM / 16;  // PV output dim chunks
@@ -141,11 +141,15 @@ __global__ __launch_bounds__(256) void attn_decode_kernel(
   // LDS producer->consumer wait is lgkmcnt-only so outstanding K/V global
   // prefetches keep flowing (a full s_waitcnt(0) here serialized every
   // chunk: v3 capped at ~3 TB/s of KV stream).
+  // Sliding window: tokens below L-window are masked; their pages may
+  // have been dropped by the allocator (block-table entry -1), so page ids
+  // are sanitized to 0 (any read from them is masked out).
+  auto page_of = [&](int pb) { int p = bt[pb]; return p < 0 ? 0 : p; };
   auto load_k = [&](int pb, ushort8 (*dst)[STEPS]) {
-    const int64_t pbase0 = (int64_t)bt[pb] * page_elems + head_off;
+    const int64_t pbase0 = (int64_t)page_of(pb) * page_elems + head_off;
     const bool have_p1 = pb + 1 < pb_hi;
     const int64_t pbase1 =
-        have_p1 ? (int64_t)bt[pb + 1] * page_elems + head_off : pbase0;
+        have_p1 ? (int64_t)page_of(pb + 1) * page_elems + head_off : pbase0;
 #pragma unroll
     for (int sub = 0; sub < 2; ++sub) {
       const int64_t pbase = sub == 0 ? pbase0 : pbase1;
@@ -170,14 +174,22 @@ __global__ __launch_bounds__(256) void attn_decode_kernel(
   };
 
   constexpr int NV = DEC_CHUNK_TOK * HEAD_DIM / 8 / WAVE_SIZE;
+  // Window lower bound: skip 2-page chunks wholly below it (chunk grid
+  // stays 2-page aligned to pb_lo so the wave stride is unchanged).
+  int pb_first = pb_lo;
+  if (window > 0 && L > window) {
+    const int win_lo_page = (L - window) / KV_BLOCK_SIZE;
+    if (win_lo_page > pb_lo)
+      pb_first = pb_lo + ((win_lo_page - pb_lo) / 2) * 2;
+  }
   ushort8 kreg[2][STEPS];
-  if (pb_lo + wave * 2 < pb_hi) load_k(pb_lo + wave * 2, kreg);
+  if (pb_first + wave * 2 < pb_hi) load_k(pb_first + wave * 2, kreg);
 
-  for (int pb = pb_lo + wave * 2; pb < pb_hi; pb += NUM_WAVES * 2) {
-    const int64_t pbase0 = (int64_t)bt[pb] * page_elems + head_off;
+  for (int pb = pb_first + wave * 2; pb < pb_hi; pb += NUM_WAVES * 2) {
+    const int64_t pbase0 = (int64_t)page_of(pb) * page_elems + head_off;
     const bool have_p1 = pb + 1 < pb_hi;
     const int64_t pbase1 =
-        have_p1 ? (int64_t)bt[pb + 1] * page_elems + head_off : pbase0;
+        have_p1 ? (int64_t)page_of(pb + 1) * page_elems + head_off : pbase0;
     const int tok0 = pb * KV_BLOCK_SIZE;  // first token of the chunk
     // valid tokens here: bounded by L AND by the partition's page range
     const int kmax =
@@ -236,7 +248,9 @@ __global__ __launch_bounds__(256) void attn_decode_kernel(
       for (int r = 0; r < 4; ++r) {
         const int t = sub * 16 + 4 * la + r;
         const int i = sub * 4 + r;
-        p[i] = (t < kmax) ? sc[sub][r] * scale : -FLT_MAX;
+        const bool vis =
+            t < kmax && (window <= 0 || tok0 + t >= L - window);
+        p[i] = vis ? sc[sub][r] * scale : -FLT_MAX;
         tile_max = fmaxf(tile_max, p[i]);
       }
     }
@@ -453,18 +467,19 @@ void launch_decode_gq(bf16* out, float* part_out, const bf16* q,
                       const bf16* kc, const bf16* vc, const int* bt,
                       const int* sl, float scale, int num_seqs,
                       int num_kv_heads, int gq, int max_blocks, int nparts,
-                      int64_t q_stride, bool kv_fp8, hipStream_t stream) {
+                      int64_t q_stride, bool kv_fp8, int window,
+                      hipStream_t stream) {
   dim3 grid(num_kv_heads, num_seqs, nparts), block(256);
 #define ARKS_CASE(G)                                                          \
   case G:                                                                     \
     if (kv_fp8) {                                                             \
       hipLaunchKernelGGL((attn_decode_kernel<HEAD_DIM, G, true>), grid,       \
                          block, 0, stream, out, part_out, q, kc, vc, bt, sl,  \
-                         scale, num_kv_heads, max_blocks, q_stride);          \
+                         scale, num_kv_heads, max_blocks, q_stride, window);  \
     } else {                                                                  \
       hipLaunchKernelGGL((attn_decode_kernel<HEAD_DIM, G, false>), grid,      \
                          block, 0, stream, out, part_out, q, kc, vc, bt, sl,  \
-                         scale, num_kv_heads, max_blocks, q_stride);          \
+                         scale, num_kv_heads, max_blocks, q_stride, window);  \
     }                                                                         \
     break;
   switch (gq) {
@@ -499,7 +514,7 @@ extern "C" void arks_attn_decode_paged(void* out, void* part_out, const void* q,
                                        int num_seqs, int num_q_heads,
                                        int num_kv_heads, int head_dim,
                                        int max_blocks, int nparts,
-                                       int64_t q_stride, int kv_fp8,
+                                       int64_t q_stride, int kv_fp8, int window,
                                        hipStream_t stream) {
   const int gq = num_q_heads / num_kv_heads;
   if (head_dim == 128) {
@@ -507,12 +522,12 @@ extern "C" void arks_attn_decode_paged(void* out, void* part_out, const void* q,
                           (const bf16*)k_cache, (const bf16*)v_cache,
                           (const int*)block_tables, (const int*)seq_lens,
                           scale, num_seqs, num_kv_heads, gq, max_blocks,
-                          nparts, q_stride, kv_fp8 != 0, stream);
+                          nparts, q_stride, kv_fp8 != 0, window, stream);
   } else if (head_dim == 64) {
     launch_decode_gq<64>((bf16*)out, (float*)part_out, (const bf16*)q,
                          (const bf16*)k_cache, (const bf16*)v_cache,
                          (const int*)block_tables, (const int*)seq_lens, scale,
                          num_seqs, num_kv_heads, gq, max_blocks, nparts,
-                         q_stride, kv_fp8 != 0, stream);
+                         q_stride, kv_fp8 != 0, window, stream);
   }
 }

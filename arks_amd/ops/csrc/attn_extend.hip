@@ -57,7 +57,7 @@ __global__ __launch_bounds__(256) void attn_extend_kernel(
     const int* __restrict__ cu_seqlens_q,  // [num_seqs + 1]
     const int* __restrict__ tile_info,     // [ntiles, 2] = (seq_idx, q0)
     const float scale, const int num_q_heads, const int num_kv_heads,
-    const int max_blocks, const int64_t q_stride) {
+    const int max_blocks, const int64_t q_stride, const int window) {
   constexpr int CHUNKS = HEAD_DIM / 16;
   constexpr int STEPS = HEAD_DIM / 32;
 
@@ -110,18 +110,23 @@ __global__ __launch_bounds__(256) void attn_extend_kernel(
   // Keys visible to this workgroup: strictly below kmax.
   const int kmax = min(kv_len, kv_off + q0 + EXT_QTILE);
   const int ntiles = (kmax + EXT_KTILE - 1) / EXT_KTILE;
+  // Sliding window: start at the first tile any row of this q-tile sees.
+  const int j0 =
+      (window > 0) ? max(0, (kv_off + q0 - window + 1) / EXT_KTILE) : 0;
 
-  for (int j = 0; j < ntiles; ++j) {
+  for (int j = j0; j < ntiles; ++j) {
     const int key_base = j * EXT_KTILE;
     // Page base offsets for this tile's two 16-key pages.
     const int64_t page_elems = (int64_t)num_kv_heads * PAGE * HEAD_DIM;
+    // window-dropped pages carry block-table entry -1; their tokens are
+    // masked, so the page id is sanitized to 0
+    const int pg0 = max(0, bt[key_base / PAGE]);
+    const int pg1 =
+        (key_base + PAGE < kmax) ? max(0, bt[key_base / PAGE + 1]) : pg0;
     const int64_t pbase0 =
-        (int64_t)bt[key_base / PAGE] * page_elems + (int64_t)kvh * PAGE * HEAD_DIM;
+        (int64_t)pg0 * page_elems + (int64_t)kvh * PAGE * HEAD_DIM;
     const int64_t pbase1 =
-        (key_base + PAGE < kmax)
-            ? (int64_t)bt[key_base / PAGE + 1] * page_elems +
-                  (int64_t)kvh * PAGE * HEAD_DIM
-            : pbase0;
+        (int64_t)pg1 * page_elems + (int64_t)kvh * PAGE * HEAD_DIM;
     {
       const int nvec = EXT_KTILE * HEAD_DIM / 8;
       for (int i = tid; i < nvec; i += 256) {
@@ -185,7 +190,8 @@ __global__ __launch_bounds__(256) void attn_extend_kernel(
       for (int r = 0; r < 4; ++r) {
         const int kg = key_base + sub * 16 + 4 * la + r;
         const int i = sub * 4 + r;
-        msk[i] = qrow_valid && (kg <= kv_off + my_qrow) && (kg < kmax);
+        msk[i] = qrow_valid && (kg <= kv_off + my_qrow) && (kg < kmax) &&
+                 (window <= 0 || kg > kv_off + my_qrow - window);
         p[i] = msk[i] ? sc[sub][r] * scale : -FLT_MAX;
         tile_max = fmaxf(tile_max, p[i]);
       }
@@ -279,7 +285,7 @@ extern "C" void arks_attn_extend_paged(
     const void* block_tables, const void* kv_lens, const void* cu_seqlens_q,
     const void* tile_info, int ntiles, float scale, int num_q_heads,
     int num_kv_heads, int head_dim, int max_blocks, int64_t q_stride,
-    int kv_fp8, hipStream_t stream) {
+    int kv_fp8, int window, hipStream_t stream) {
   dim3 grid(num_q_heads, ntiles), block(256);
   if (head_dim == 128 && kv_fp8) {
     hipLaunchKernelGGL((attn_extend_kernel<128, true>), grid, block, 0, stream,
@@ -287,27 +293,27 @@ extern "C" void arks_attn_extend_paged(
                        (const bf16*)v_cache, (const int*)block_tables,
                        (const int*)kv_lens, (const int*)cu_seqlens_q,
                        (const int*)tile_info, scale, num_q_heads, num_kv_heads,
-                       max_blocks, q_stride);
+                       max_blocks, q_stride, window);
   } else if (head_dim == 64 && kv_fp8) {
     hipLaunchKernelGGL((attn_extend_kernel<64, true>), grid, block, 0, stream,
                        (bf16*)out, (const bf16*)q, (const bf16*)k_cache,
                        (const bf16*)v_cache, (const int*)block_tables,
                        (const int*)kv_lens, (const int*)cu_seqlens_q,
                        (const int*)tile_info, scale, num_q_heads, num_kv_heads,
-                       max_blocks, q_stride);
+                       max_blocks, q_stride, window);
   } else if (head_dim == 128) {
     hipLaunchKernelGGL((attn_extend_kernel<128, false>), grid, block, 0, stream,
                        (bf16*)out, (const bf16*)q, (const bf16*)k_cache,
                        (const bf16*)v_cache, (const int*)block_tables,
                        (const int*)kv_lens, (const int*)cu_seqlens_q,
                        (const int*)tile_info, scale, num_q_heads, num_kv_heads,
-                       max_blocks, q_stride);
+                       max_blocks, q_stride, window);
   } else if (head_dim == 64) {
     hipLaunchKernelGGL((attn_extend_kernel<64, false>), grid, block, 0, stream,
                        (bf16*)out, (const bf16*)q, (const bf16*)k_cache,
                        (const bf16*)v_cache, (const int*)block_tables,
                        (const int*)kv_lens, (const int*)cu_seqlens_q,
                        (const int*)tile_info, scale, num_q_heads, num_kv_heads,
-                       max_blocks, q_stride);
+                       max_blocks, q_stride, window);
   }
 }

@@ -1,0 +1,373 @@
+// Paged "extend" attention, 8-wave 32x32-MFMA ladder (gfx950).
+//
+// Second-generation prefill/extend kernel: replaces the 4-wave 16x16
+// attn_extend for large q tiles. Structure follows the CDNA4 attention
+// ladder (guide "Fused attention prefill"): one workgroup = 8 waves = a
+// 256-row Q tile of one (seq, q-head); each wave owns 32 q rows. K/V are
+// iterated in 64-key tiles, double-buffered in dynamic LDS with register
+// staging (issue-early / write-late, T14), K rows XOR-swizzled for
+// ds_read_b128, V as a tr16 image read with ds_read_b64_tr_b16.
+//
+// QK^T uses swapped operands on v_mfma_f32_32x32x16_bf16 (A = K-tile,
+// B = Q) so each lane's C fragment holds 16 scores for ONE q-row
+// (col = lane&31); the online softmax is then fully in-register: row max =
+// in-lane fmax tree + one cross-half shuffle, P stays in VGPRs and is
+// packed into the PV A-fragment with v_cvt_pk_bf16_f32 + permlane32_swap
+// (16 cvt_pk + 8 swaps per 64-key tile) — no P round trip through LDS and
+// one barrier per KV tile. O-accumulator rescale is skipped while the
+// running max is within RESCALE_THR of the tile max (defer-max, T13).
+//
+// Sliding-window attention: `window > 0` restricts visibility to the last
+// `window` key positions per query (Mistral-style); the KV loop starts at
+// the first tile any of this workgroup's rows can see.
+//
+// This is the prefix-caching / chunked-prefill attention of the runtime
+// slot (the reference delegates it to vLLM/SGLang images
+// at arksapplication_controller.go:941-1002).
+#include "common.h"
+
+#include <cfloat>
+
+namespace arks {
+
+typedef __attribute__((ext_vector_type(8))) __bf16 e2_bf16x8;
+typedef __attribute__((ext_vector_type(4))) __bf16 e2_bf16x4;
+typedef __attribute__((ext_vector_type(16))) float f32x16;
+
+__device__ __forceinline__ f32x16 e2_mfma32(e2_bf16x8 a, e2_bf16x8 b, f32x16 c) {
+  return __builtin_amdgcn_mfma_f32_32x32x16_bf16(a, b, c, 0, 0, 0);
+}
+
+constexpr int E2_WAVES = 8;
+constexpr int E2_QTILE = 256;      // q rows per workgroup (32 per wave)
+constexpr int E2_KVBLK = 64;       // keys per LDS tile
+constexpr int E2_PAGE = 16;        // KV page size (tokens)
+constexpr float E2_LOG2E = 1.44269504088896340736f;
+constexpr float RESCALE_THR = 8.f; // defer-max threshold (exp2 domain)
+
+// v_cvt_pk_bf16_f32: pack two f32 into one dword of two bf16 (RNE).
+__device__ __forceinline__ uint32_t cvt_pk_bf16(float lo, float hi) {
+  uint32_t r;
+  asm volatile("v_cvt_pk_bf16_f32 %0, %1, %2" : "=v"(r) : "v"(lo), "v"(hi));
+  return r;
+}
+
+// V-image geometry: 4-key x 16-col blocks padded to 80 elements (160 B) so
+// the cooperative b128 staging writes hit all 32 write-banks (stride-160
+// spreads the 8-lane contiguous write groups; packed-128 is 4-way
+// conflicted) while the tr16 read pairs overlap by only 8 of 64 banks.
+constexpr int VBLK_ELEMS = 80;
+
+template <int HEAD_DIM>
+__global__ __launch_bounds__(512) void attn_extend2_kernel(
+    bf16* __restrict__ out,            // [Tq, Hq, D]
+    const bf16* __restrict__ q,        // [Tq, Hq, D] (packed new tokens)
+    const bf16* __restrict__ k_cache,  // [nblocks, Hkv, 16, D]
+    const bf16* __restrict__ v_cache,  // [nblocks, Hkv, 16, D]
+    const int* __restrict__ block_tables,  // [num_seqs, max_blocks]
+    const int* __restrict__ kv_lens,       // [num_seqs]
+    const int* __restrict__ cu_seqlens_q,  // [num_seqs + 1]
+    const int* __restrict__ tile_info,     // [ntiles, 2] = (seq_idx, q0)
+    const float scale, const int num_q_heads, const int num_kv_heads,
+    const int max_blocks, const int64_t q_stride, const int window) {
+  constexpr int STEPS = HEAD_DIM / 16;   // QK^T K-contraction steps
+  constexpr int DTILES = HEAD_DIM / 32;  // 32-dim output tiles
+  constexpr int CB = HEAD_DIM / 16;      // 16-col blocks per key row
+
+  const int h = blockIdx.x;
+  const int kvh = h / (num_q_heads / num_kv_heads);
+  const int seq_idx = tile_info[blockIdx.y * 2];
+  const int q0 = tile_info[blockIdx.y * 2 + 1];
+  const int q_start = cu_seqlens_q[seq_idx];
+  const int q_len = cu_seqlens_q[seq_idx + 1] - q_start;
+  const int kv_len = kv_lens[seq_idx];
+  const int kv_off = kv_len - q_len;  // cached-prefix length
+  const int* __restrict__ bt = block_tables + (int64_t)seq_idx * max_blocks;
+
+  const int tid = threadIdx.x;
+  const int wave = tid / WAVE_SIZE;
+  const int lane = tid % WAVE_SIZE;
+  const int col = lane & 31;   // q-row within wave (QK^T) / dim col (PV)
+  const int half = lane >> 5;  // lane half
+
+  // Dynamic LDS: K tiles (XOR-swizzled 2*HEAD_DIM-byte rows) then the V
+  // tr16 image, both double-buffered. 73.7 KiB at D=128 (> the 64 KiB
+  // static limit; the launcher raises the dynamic-LDS cap).
+  extern __shared__ __attribute__((aligned(16))) char lds_raw[];
+  bf16* k_lds = reinterpret_cast<bf16*>(lds_raw);  // [2][KVBLK][HEAD_DIM]
+  bf16* v_img = k_lds + 2 * E2_KVBLK * HEAD_DIM;   // [2][16][CB][80]
+  constexpr int KBUF = E2_KVBLK * HEAD_DIM;
+  constexpr int VBUF = 16 * CB * VBLK_ELEMS;
+
+  // --- Q fragments (registers, loaded once): lane's q-row = q0+32w+col,
+  // frag s holds dims 16s+8*half .. +7.
+  const int my_qrow = q0 + wave * 32 + col;
+  const bool qrow_valid = my_qrow < q_len;
+  const int my_pos = kv_off + my_qrow;  // global kv position of this q row
+  e2_bf16x8 qfrag[STEPS];
+  {
+    const int64_t qbase =
+        (int64_t)(q_start + (qrow_valid ? my_qrow : 0)) * q_stride +
+        (int64_t)h * HEAD_DIM + 8 * half;
+#pragma unroll
+    for (int s = 0; s < STEPS; ++s) {
+      ushort8 u = *reinterpret_cast<const ushort8*>(q + qbase + s * 16);
+      qfrag[s] = *reinterpret_cast<e2_bf16x8*>(&u);
+    }
+  }
+
+  // Online-softmax state (exp2 domain; per lane for q-row `my_qrow`,
+  // replicated across the two halves).
+  float m_run = -FLT_MAX;
+  float l_run = 0.f;
+  f32x16 oacc[DTILES];
+#pragma unroll
+  for (int n = 0; n < DTILES; ++n) oacc[n] = {};
+
+  const int kmax = min(kv_len, kv_off + q0 + E2_QTILE);
+  const int ntiles = (kmax + E2_KVBLK - 1) / E2_KVBLK;
+  // Sliding window: the first key visible to any row of this tile.
+  const int j0 =
+      (window > 0) ? max(0, (kv_off + q0 - window + 1) / E2_KVBLK) : 0;
+  const float scale2 = scale * E2_LOG2E;
+
+  // Cooperative staging: 512 threads x 2 vec8 cover one 64x128 tile (one
+  // vec8 per thread at D=64).
+  constexpr int NVEC = E2_KVBLK * HEAD_DIM / 8;
+  constexpr int VPT = NVEC / 512;
+  const int64_t page_elems = (int64_t)num_kv_heads * E2_PAGE * HEAD_DIM;
+  const int64_t head_off = (int64_t)kvh * E2_PAGE * HEAD_DIM;
+
+  auto load_tile = [&](int j, ushort8* kr, ushort8* vr) {
+    const int key_base = j * E2_KVBLK;
+#pragma unroll
+    for (int vv = 0; vv < VPT; ++vv) {
+      const int e = tid + 512 * vv;
+      const int key = e / (HEAD_DIM / 8);
+      const int col8 = (e % (HEAD_DIM / 8)) * 8;
+      const int kg = key_base + key;
+      ushort8 kv{}, vv8{};
+      if (kg < kmax) {
+        // window-dropped pages are -1 in the block table; sanitized to 0
+        // (their tokens are masked out)
+        const int64_t pg = max(0, bt[kg / E2_PAGE]);
+        const int64_t src = pg * page_elems + head_off +
+                            (int64_t)(kg % E2_PAGE) * HEAD_DIM + col8;
+        kv = *reinterpret_cast<const ushort8*>(k_cache + src);
+        vv8 = *reinterpret_cast<const ushort8*>(v_cache + src);
+      }
+      kr[vv] = kv;
+      vr[vv] = vv8;
+    }
+  };
+
+  auto store_tile = [&](int buf, const ushort8* kr, const ushort8* vr) {
+#pragma unroll
+    for (int vv = 0; vv < VPT; ++vv) {
+      const int e = tid + 512 * vv;
+      const int key = e / (HEAD_DIM / 8);
+      const int col8 = (e % (HEAD_DIM / 8)) * 8;
+      // K: XOR-swizzled byte offset within the 2*HEAD_DIM-byte row.
+      const int swz = (col8 * 2) ^ ((key & 7) << 4);
+      *reinterpret_cast<ushort8*>(
+          reinterpret_cast<char*>(k_lds + buf * KBUF + key * HEAD_DIM) + swz) =
+          kr[vv];
+      // V image: block (kb = key/4, cb = col8/16), row key%4, col col8%16.
+      bf16* blk = v_img + buf * VBUF + ((key >> 2) * CB + (col8 >> 4)) * VBLK_ELEMS;
+      *reinterpret_cast<ushort8*>(blk + (key & 3) * 16 + (col8 & 15)) = vr[vv];
+    }
+  };
+
+  {
+    ushort8 kr0[VPT], vr0[VPT];
+    load_tile(j0, kr0, vr0);
+    store_tile(0, kr0, vr0);
+  }
+  __syncthreads();
+
+  int buf = 0;
+  for (int j = j0; j < ntiles; ++j, buf ^= 1) {
+    const int key_base = j * E2_KVBLK;
+    // ---- prefetch tile j+1 into registers (overlaps with the MFMAs).
+    ushort8 krn[VPT], vrn[VPT];
+    if (j + 1 < ntiles) load_tile(j + 1, krn, vrn);
+
+    // ---- QK^T (swapped): A = K rows (32 keys), B = Q. C[key][qrow=col].
+    f32x16 sc[2];
+#pragma unroll
+    for (int sub = 0; sub < 2; ++sub) {
+      sc[sub] = {};
+      const int key = sub * 32 + col;
+      const char* krow = reinterpret_cast<const char*>(
+          k_lds + buf * KBUF + key * HEAD_DIM);
+      const int swz_mask = (key & 7) << 4;
+#pragma unroll
+      for (int s = 0; s < STEPS; ++s) {
+        const int off = ((s * 16 + 8 * half) * 2) ^ swz_mask;
+        ushort8 u = *reinterpret_cast<const ushort8*>(krow + off);
+        sc[sub] = e2_mfma32(*reinterpret_cast<e2_bf16x8*>(&u), qfrag[s], sc[sub]);
+      }
+    }
+
+    // ---- Masked softmax, fully in-register (exp2 domain). Lane's 32
+    // scores are all for q-row `my_qrow`; C reg 4g+r of subtile `sub` is
+    // key key_base + 32*sub + 8g + 4*half + r.
+    float p[32];
+    float tmax = -FLT_MAX;
+#pragma unroll
+    for (int sub = 0; sub < 2; ++sub) {
+      const int kb_s = key_base + 32 * sub + 4 * half;
+#pragma unroll
+      for (int r16 = 0; r16 < 16; ++r16) {
+        const int kg = kb_s + 8 * (r16 >> 2) + (r16 & 3);
+        bool ok = qrow_valid && kg <= my_pos && kg < kmax;
+        if (window > 0) ok = ok && (kg > my_pos - window);
+        const int i = sub * 16 + r16;
+        p[i] = ok ? sc[sub][r16] * scale2 : -FLT_MAX;
+        tmax = fmaxf(tmax, p[i]);
+      }
+    }
+    tmax = fmaxf(tmax, __shfl_xor(tmax, 32, WAVE_SIZE));
+
+    // Defer-max: only rescale O when the tile max meaningfully exceeds the
+    // running max (T13; exp2(THR)=256 stays comfortably finite).
+    if (!__all(tmax - m_run <= RESCALE_THR)) {
+      const float m_new = fmaxf(m_run, tmax);
+      const float alpha = __builtin_amdgcn_exp2f(m_run - m_new);
+      l_run *= alpha;
+      m_run = m_new;
+      // O rows live across regs (C layout): fetch each row's alpha from
+      // the lane that owns that q-row.
+#pragma unroll
+      for (int r16 = 0; r16 < 16; ++r16) {
+        const int row = 8 * (r16 >> 2) + 4 * half + (r16 & 3);
+        const float a_r = __shfl(alpha, 32 * (lane >> 5) + row, WAVE_SIZE);
+#pragma unroll
+        for (int n = 0; n < DTILES; ++n) oacc[n][r16] *= a_r;
+      }
+    }
+
+    float psum = 0.f;
+#pragma unroll
+    for (int i = 0; i < 32; ++i) {
+      p[i] = (p[i] == -FLT_MAX) ? 0.f : __builtin_amdgcn_exp2f(p[i] - m_run);
+      psum += p[i];
+    }
+    psum += __shfl_xor(psum, 32, WAVE_SIZE);
+    l_run += psum;
+
+    // ---- P -> bf16 PV A-fragments: per subtile, 8 cvt_pk words
+    // W[g][0] = pk(p[4g], p[4g+1]), W[g][1] = pk(p[4g+2], p[4g+3]); the
+    // permlane32_swap pairs (W[2t][*], W[2t+1][*]) then give each half its
+    // step-t fragment from both halves' key sets.
+    uint32_t afrag[4][4];  // [k-step t][4 dwords = 8 bf16]
+#pragma unroll
+    for (int sub = 0; sub < 2; ++sub) {
+      uint32_t w0[4], w1[4];
+#pragma unroll
+      for (int g = 0; g < 4; ++g) {
+        const int b = sub * 16 + 4 * g;
+        w0[g] = cvt_pk_bf16(p[b + 0], p[b + 1]);
+        w1[g] = cvt_pk_bf16(p[b + 2], p[b + 3]);
+      }
+#pragma unroll
+      for (int tau = 0; tau < 2; ++tau) {
+        auto r0 = __builtin_amdgcn_permlane32_swap(w0[2 * tau], w0[2 * tau + 1],
+                                                   false, false);
+        auto r1 = __builtin_amdgcn_permlane32_swap(w1[2 * tau], w1[2 * tau + 1],
+                                                   false, false);
+        const int t = sub * 2 + tau;
+        afrag[t][0] = r0[0];
+        afrag[t][1] = r1[0];
+        afrag[t][2] = r0[1];
+        afrag[t][3] = r1[1];
+      }
+    }
+
+    // ---- PV: O[qrow][dim] += P[qrow][key] V[key][dim]; B fragments via
+    // tr16 reads of the V image (per step t: keys 16t+8*half+0..7 at
+    // column 16*(colhalf)+l&15 -> blocks kb = 4t+2*half (+1), cb per group).
+    const int g4 = lane >> 4;  // 16-lane tr16 group
+    bf16* vbase = v_img + buf * VBUF;
+#pragma unroll
+    for (int n = 0; n < DTILES; ++n) {
+      const int cb = 2 * n + (g4 & 1);
+#pragma unroll
+      for (int t = 0; t < 4; ++t) {
+        const int kb = 4 * t + 2 * (g4 >> 1);
+        e2_bf16x4 v1 = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(
+            (__attribute__((address_space(3))) e2_bf16x4*)(
+                reinterpret_cast<char*>(vbase + (kb * CB + cb) * VBLK_ELEMS) +
+                (lane & 15) * 8));
+        e2_bf16x4 v2 = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(
+            (__attribute__((address_space(3))) e2_bf16x4*)(
+                reinterpret_cast<char*>(vbase + ((kb + 1) * CB + cb) * VBLK_ELEMS) +
+                (lane & 15) * 8));
+        e2_bf16x8 vb;
+#pragma unroll
+        for (int e = 0; e < 4; ++e) {
+          vb[e] = v1[e];
+          vb[e + 4] = v2[e];
+        }
+        oacc[n] = e2_mfma32(*reinterpret_cast<e2_bf16x8*>(&afrag[t][0]), vb,
+                            oacc[n]);
+      }
+    }
+
+    // ---- write-late: scatter the prefetched tile into the idle buffer.
+    if (j + 1 < ntiles) store_tile(buf ^ 1, krn, vrn);
+    __syncthreads();
+  }
+
+  // ---- Epilogue. O reg 4g+r of dim-tile n is q-row 8g+4*half+r, dim
+  // 32n+col; l_run for row `row` lives in lane row (either half).
+#pragma unroll
+  for (int r16 = 0; r16 < 16; ++r16) {
+    const int row = 8 * (r16 >> 2) + 4 * half + (r16 & 3);
+    const int qrow = q0 + wave * 32 + row;
+    const float lr = __shfl(l_run, 32 * (lane >> 5) + row, WAVE_SIZE);
+    if (qrow >= q_len) continue;
+    const float inv = lr > 0.f ? 1.f / lr : 0.f;
+    const int64_t obase =
+        ((int64_t)(q_start + qrow) * num_q_heads + h) * HEAD_DIM + col;
+#pragma unroll
+    for (int n = 0; n < DTILES; ++n) {
+      out[obase + 32 * n] = float_to_bf16_bits(oacc[n][r16] * inv);
+    }
+  }
+}
+
+}  // namespace arks
+
+using namespace arks;
+
+extern "C" void arks_attn_extend_paged2(
+    void* out, const void* q, const void* k_cache, const void* v_cache,
+    const void* block_tables, const void* kv_lens, const void* cu_seqlens_q,
+    const void* tile_info, int ntiles, float scale, int num_q_heads,
+    int num_kv_heads, int head_dim, int max_blocks, int64_t q_stride,
+    int window, hipStream_t stream) {
+  dim3 grid(num_q_heads, ntiles), block(512);
+  static bool attr_set[2] = {false, false};
+  auto launch = [&](auto kern, int lds_bytes, int idx) {
+    if (!attr_set[idx]) {
+      (void)hipFuncSetAttribute(reinterpret_cast<const void*>(kern),
+                          hipFuncAttributeMaxDynamicSharedMemorySize,
+                          lds_bytes);
+      attr_set[idx] = true;
+    }
+    hipLaunchKernelGGL(kern, grid, block, lds_bytes, stream, (bf16*)out,
+                       (const bf16*)q, (const bf16*)k_cache,
+                       (const bf16*)v_cache, (const int*)block_tables,
+                       (const int*)kv_lens, (const int*)cu_seqlens_q,
+                       (const int*)tile_info, scale, num_q_heads, num_kv_heads,
+                       max_blocks, q_stride, window);
+  };
+  if (head_dim == 128) {
+    constexpr int LDS = (2 * 64 * 128 + 2 * 16 * 8 * VBLK_ELEMS) * 2;
+    launch(attn_extend2_kernel<128>, LDS, 0);
+  } else if (head_dim == 64) {
+    constexpr int LDS = (2 * 64 * 64 + 2 * 16 * 4 * VBLK_ELEMS) * 2;
+    launch(attn_extend2_kernel<64>, LDS, 1);
+  }
+}

@@ -36,7 +36,7 @@ void arks_attn_decode_paged(void* out, void* part_out, const void* q,
                             float scale, int num_seqs, int num_q_heads,
                             int num_kv_heads, int head_dim, int max_blocks,
                             int nparts, int64_t q_stride, int kv_fp8,
-                            hipStream_t stream);
+                            int window, hipStream_t stream);
 void arks_attn_prefill_varlen(void* out, const void* q, const void* k,
                               const void* v, const void* cu_seqlens,
                               const void* tile_info, int ntiles, float scale,
@@ -49,7 +49,14 @@ void arks_attn_extend_paged(void* out, const void* q, const void* k_cache,
                             const void* tile_info, int ntiles, float scale,
                             int num_q_heads, int num_kv_heads, int head_dim,
                             int max_blocks, int64_t q_stride, int kv_fp8,
-                            hipStream_t stream);
+                            int window, hipStream_t stream);
+void arks_attn_extend_paged2(void* out, const void* q, const void* k_cache,
+                             const void* v_cache, const void* block_tables,
+                             const void* kv_lens, const void* cu_seqlens_q,
+                             const void* tile_info, int ntiles, float scale,
+                             int num_q_heads, int num_kv_heads, int head_dim,
+                             int max_blocks, int64_t q_stride, int window,
+                             hipStream_t stream);
 void arks_quant_fp8_rows(void* out, void* inv_scale, const void* x, int rows,
                          int cols, hipStream_t stream);
 void arks_rmsnorm_fp8(void* out, void* inv_scale, const void* input,
@@ -179,7 +186,7 @@ void attention_decode_paged(torch::Tensor out, torch::Tensor q,
                             torch::Tensor k_cache, torch::Tensor v_cache,
                             torch::Tensor block_tables, torch::Tensor seq_lens,
                             double scale, torch::Tensor part_out,
-                            int64_t nparts) {
+                            int64_t nparts, int64_t window) {
   check_bf16_contig(out, "out");
   check_bf16_rowstrided(q, "q");
   const bool kv_fp8 = check_kv_cache(k_cache, "k_cache");
@@ -209,7 +216,7 @@ void attention_decode_paged(torch::Tensor out, torch::Tensor q,
                          block_tables.data_ptr(), seq_lens.data_ptr(),
                          (float)scale, num_seqs, num_q_heads, num_kv_heads,
                          head_dim, max_blocks, (int)nparts, q.stride(0),
-                         kv_fp8 ? 1 : 0, current_stream());
+                         kv_fp8 ? 1 : 0, (int)window, current_stream());
 }
 
 void attention_prefill_varlen(torch::Tensor out, torch::Tensor q,
@@ -240,7 +247,7 @@ void attention_extend_paged(torch::Tensor out, torch::Tensor q,
                             torch::Tensor k_cache, torch::Tensor v_cache,
                             torch::Tensor block_tables, torch::Tensor kv_lens,
                             torch::Tensor cu_seqlens_q, torch::Tensor tile_info,
-                            double scale) {
+                            double scale, int64_t window) {
   check_bf16_contig(out, "out");
   check_bf16_rowstrided(q, "q");
   const bool kv_fp8 = check_kv_cache(k_cache, "k_cache");
@@ -262,7 +269,39 @@ void attention_extend_paged(torch::Tensor out, torch::Tensor q,
                          kv_lens.data_ptr(), cu_seqlens_q.data_ptr(),
                          tile_info.data_ptr(), ntiles, (float)scale,
                          num_q_heads, num_kv_heads, head_dim, max_blocks,
-                         q.stride(0), kv_fp8 ? 1 : 0, current_stream());
+                         q.stride(0), kv_fp8 ? 1 : 0, (int)window,
+                         current_stream());
+}
+
+// 8-wave 32x32-MFMA ladder (attn_extend2.hip): 256-row q tiles, bf16 KV.
+void attention_extend_paged2(torch::Tensor out, torch::Tensor q,
+                             torch::Tensor k_cache, torch::Tensor v_cache,
+                             torch::Tensor block_tables, torch::Tensor kv_lens,
+                             torch::Tensor cu_seqlens_q,
+                             torch::Tensor tile_info, double scale,
+                             int64_t window) {
+  check_bf16_contig(out, "out");
+  check_bf16_rowstrided(q, "q");
+  check_bf16_contig(k_cache, "k_cache");
+  check_bf16_contig(v_cache, "v_cache");
+  TORCH_CHECK(block_tables.scalar_type() == torch::kInt32);
+  TORCH_CHECK(kv_lens.scalar_type() == torch::kInt32);
+  TORCH_CHECK(cu_seqlens_q.scalar_type() == torch::kInt32);
+  TORCH_CHECK(tile_info.scalar_type() == torch::kInt32);
+  TORCH_CHECK(tile_info.dim() == 2 && tile_info.size(1) == 2);
+  const int num_q_heads = q.size(1);
+  const int head_dim = q.size(2);
+  const int num_kv_heads = k_cache.size(1);
+  TORCH_CHECK(k_cache.size(2) == 16, "KV block size must be 16");
+  TORCH_CHECK(head_dim == 64 || head_dim == 128, "head_dim must be 64 or 128");
+  const int ntiles = tile_info.size(0);
+  const int max_blocks = block_tables.size(1);
+  arks_attn_extend_paged2(out.data_ptr(), q.data_ptr(), k_cache.data_ptr(),
+                          v_cache.data_ptr(), block_tables.data_ptr(),
+                          kv_lens.data_ptr(), cu_seqlens_q.data_ptr(),
+                          tile_info.data_ptr(), ntiles, (float)scale,
+                          num_q_heads, num_kv_heads, head_dim, max_blocks,
+                          q.stride(0), (int)window, current_stream());
 }
 
 void skinny_gemm(torch::Tensor out, torch::Tensor part, torch::Tensor a,
@@ -391,6 +430,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("attention_decode_paged", &attention_decode_paged);
   m.def("attention_prefill_varlen", &attention_prefill_varlen);
   m.def("attention_extend_paged", &attention_extend_paged);
+  m.def("attention_extend_paged2", &attention_extend_paged2);
   m.def("quant_fp8_rows", &quant_fp8_rows);
   m.def("rmsnorm_fp8", &rmsnorm_fp8);
   m.def("silu_mul_fp8", &silu_mul_fp8);

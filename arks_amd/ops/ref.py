@@ -187,8 +187,10 @@ def attention_decode_paged(
     block_tables: torch.Tensor,  # [num_seqs, max_blocks] int32
     seq_lens: torch.Tensor,  # [num_seqs] int32 (length INCLUDING current token)
     scale: float,
+    window: int = 0,
 ) -> torch.Tensor:
-    """Single-token decode attention against the paged KV cache."""
+    """Single-token decode attention against the paged KV cache.
+    window > 0 = sliding-window (the query sees the last `window` keys)."""
     num_seqs, num_q_heads, head_dim = q.shape
     num_kv_heads = k_cache.shape[1]
     block_size = k_cache.shape[2]
@@ -205,6 +207,8 @@ def attention_decode_paged(
         vf = vs.float().repeat_interleave(rep, dim=1)
         qi = q[i].float()  # [H, D]
         scores = torch.einsum("hd,lhd->hl", qi, kf) * scale
+        if window > 0 and L > window:
+            scores[:, : L - window] = float("-inf")
         p = torch.softmax(scores, dim=-1)
         o = torch.einsum("hl,lhd->hd", p, vf)
         out[i] = o.to(q.dtype)
@@ -219,10 +223,12 @@ def attention_extend_paged(
     kv_lens: torch.Tensor,  # [num_seqs] int32 total kv length per seq
     cu_seqlens_q: torch.Tensor,  # [num_seqs + 1] int32 (packed new tokens)
     scale: float,
+    window: int = 0,
 ) -> torch.Tensor:
     """Causal attention of each sequence's new tokens over its full paged KV
     (prefix caching / chunked prefill). New token j of seq i sits at global
-    position kv_len - q_len + j."""
+    position kv_len - q_len + j. window > 0 = sliding-window attention
+    (each query sees only the last `window` key positions)."""
     num_q_heads, head_dim = q.shape[1], q.shape[2]
     num_kv_heads = k_cache.shape[1]
     block_size = k_cache.shape[2]
@@ -244,6 +250,8 @@ def attention_extend_paged(
         kpos = torch.arange(L, device=q.device)
         qpos = torch.arange(off, L, device=q.device)
         mask = kpos[None, :] > qpos[:, None]  # [qn, L] True = hidden
+        if window > 0:
+            mask |= kpos[None, :] <= qpos[:, None] - window
         scores = scores.masked_fill(mask, float("-inf"))
         p = torch.softmax(scores, dim=-1)
         o = torch.matmul(p, vf)  # [H, qn, D]

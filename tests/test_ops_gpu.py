@@ -431,3 +431,122 @@ def test_silu_mul_fp8_fused_matches_composed():
     y = ops.silu_mul(gu)
     dq = q.float() * s[:, None]
     torch.testing.assert_close(dq, y.float(), atol=0.08, rtol=0.12)
+
+
+def _mk_paged(spec, hq, hkv, hd, seed=13):
+    """Build scrambled paged KV fixtures: (q, kc, vc, bt, kvl, cu_q, q_lens)."""
+    torch.manual_seed(seed)
+    bs = 16
+    kv_lens = [s[0] for s in spec]
+    q_lens = [s[1] for s in spec]
+    S = len(spec)
+    Tq = sum(q_lens)
+    nb = [(n + bs - 1) // bs for n in kv_lens]
+    total_blocks = sum(nb) + 3
+    q = torch.randn(Tq, hq, hd, dtype=torch.bfloat16, device=DEV)
+    kc = torch.randn(total_blocks, hkv, bs, hd, dtype=torch.bfloat16, device=DEV)
+    vc = torch.randn_like(kc)
+    bt = torch.zeros(S, max(nb), dtype=torch.int32, device=DEV)
+    perm = torch.randperm(total_blocks - 1) + 1
+    idx = 0
+    for i in range(S):
+        for j in range(nb[i]):
+            bt[i, j] = perm[idx]
+            idx += 1
+    cu_q = torch.tensor([0] + list(torch.tensor(q_lens).cumsum(0)),
+                        dtype=torch.int32, device=DEV)
+    kvl = torch.tensor(kv_lens, dtype=torch.int32, device=DEV)
+    return q, kc, vc, bt, kvl, cu_q, q_lens
+
+
+@pytest.mark.parametrize(
+    "hq,hkv,hd,spec",
+    [
+        # big q tiles: the 8-wave 32x32 ladder kernel (attn_extend2)
+        (28, 4, 128, [(1024, 1024)]),
+        (8, 2, 128, [(700, 300), (257, 257), (1030, 97)]),
+        (4, 4, 64, [(600, 600), (300, 129)]),
+        # mixed routing: one seq per kernel in the same call
+        (8, 2, 128, [(512, 512), (40, 13)]),
+    ],
+)
+def test_attention_extend2_big_tiles(hq, hkv, hd, spec):
+    """The 256-row ladder extend kernel vs the torch fp32 reference,
+    including q lengths that straddle tile boundaries and calls that mix
+    both kernels."""
+    assert_native()
+    q, kc, vc, bt, kvl, cu_q, q_lens = _mk_paged(spec, hq, hkv, hd)
+    scale = 1.0 / math.sqrt(hd)
+    out = ops.attention_extend_paged(q, kc, vc, bt, kvl, cu_q, q_lens, scale)
+    expect = ref.attention_extend_paged(
+        q.float().cpu(), kc.float().cpu(), vc.float().cpu(), bt.cpu(),
+        kvl.cpu(), cu_q.cpu(), scale,
+    )
+    torch.testing.assert_close(out.float().cpu(), expect, atol=3e-2, rtol=3e-2)
+
+
+@pytest.mark.parametrize(
+    "hq,hkv,hd,spec,window",
+    [
+        (8, 2, 128, [(900, 800)], 128),     # ladder kernel, window binds
+        (8, 2, 128, [(900, 80), (64, 64)], 100),  # 64-row kernel
+        (4, 4, 64, [(512, 512)], 64),
+        (28, 4, 128, [(1024, 512)], 4096),  # window larger than kv: no-op
+    ],
+)
+def test_attention_extend_sliding_window(hq, hkv, hd, spec, window):
+    """Sliding-window masking in both extend kernels vs the reference."""
+    assert_native()
+    q, kc, vc, bt, kvl, cu_q, q_lens = _mk_paged(spec, hq, hkv, hd, seed=7)
+    scale = 1.0 / math.sqrt(hd)
+    out = ops.attention_extend_paged(
+        q, kc, vc, bt, kvl, cu_q, q_lens, scale, window=window
+    )
+    expect = ref.attention_extend_paged(
+        q.float().cpu(), kc.float().cpu(), vc.float().cpu(), bt.cpu(),
+        kvl.cpu(), cu_q.cpu(), scale, window=window,
+    )
+    torch.testing.assert_close(out.float().cpu(), expect, atol=3e-2, rtol=3e-2)
+
+
+@pytest.mark.parametrize("drop_pages", [False, True])
+def test_attention_decode_sliding_window(drop_pages):
+    """Decode kernel with a sliding window; out-of-window pages may be
+    dropped from the block table (-1) without changing the output."""
+    assert_native()
+    hq, hkv, hd = 8, 2, 128
+    window = 96
+    spec = [(400, 1), (97, 1), (64, 1)]
+    torch.manual_seed(11)
+    bs = 16
+    kv_lens = [s[0] for s in spec]
+    S = len(spec)
+    nb = [(n + bs - 1) // bs for n in kv_lens]
+    total_blocks = sum(nb) + 3
+    q = torch.randn(S, hq, hd, dtype=torch.bfloat16, device=DEV)
+    kc = torch.randn(total_blocks, hkv, bs, hd, dtype=torch.bfloat16, device=DEV)
+    vc = torch.randn_like(kc)
+    bt = torch.zeros(S, max(nb), dtype=torch.int32, device=DEV)
+    perm = torch.randperm(total_blocks - 1) + 1
+    idx = 0
+    for i in range(S):
+        for j in range(nb[i]):
+            bt[i, j] = perm[idx]
+            idx += 1
+    sl = torch.tensor(kv_lens, dtype=torch.int32, device=DEV)
+    scale = 1.0 / math.sqrt(hd)
+    expect = ref.attention_decode_paged(
+        q.float().cpu(), kc.float().cpu(), vc.float().cpu(), bt.cpu(),
+        sl.cpu(), scale, window=window,
+    )
+    if drop_pages:
+        # drop pages wholly below the window (allocator page-dropping)
+        for i, L in enumerate(kv_lens):
+            lo = max(0, L - window) // bs
+            bt[i, :lo] = -1
+    for nparts in (1, 4):
+        out = ops.attention_decode_paged(
+            q, kc, vc, bt, sl, scale, num_partitions=nparts, window=window
+        )
+        torch.testing.assert_close(out.float().cpu(), expect,
+                                   atol=3e-2, rtol=3e-2)
