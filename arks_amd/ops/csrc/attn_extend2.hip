@@ -138,8 +138,26 @@ __global__ __launch_bounds__(512) void attn_extend2_kernel(
   const int64_t page_elems = (int64_t)num_kv_heads * E2_PAGE * HEAD_DIM;
   const int64_t head_off = (int64_t)kvh * E2_PAGE * HEAD_DIM;
 
+  // Branchy per-element guards de-pipeline global loads (guide §5 trap (c)),
+  // so full in-range tiles take an unguarded path; only the final (ragged)
+  // tile pays the per-element branch.
   auto load_tile = [&](int j, ushort8* kr, ushort8* vr) {
     const int key_base = j * E2_KVBLK;
+    if (key_base + E2_KVBLK <= kmax) {
+#pragma unroll
+      for (int vv = 0; vv < VPT; ++vv) {
+        const int e = tid + 512 * vv;
+        const int key = e / (HEAD_DIM / 8);
+        const int col8 = (e % (HEAD_DIM / 8)) * 8;
+        const int kg = key_base + key;
+        const int64_t pg = max(0, bt[kg / E2_PAGE]);
+        const int64_t src = pg * page_elems + head_off +
+                            (int64_t)(kg % E2_PAGE) * HEAD_DIM + col8;
+        kr[vv] = *reinterpret_cast<const ushort8*>(k_cache + src);
+        vr[vv] = *reinterpret_cast<const ushort8*>(v_cache + src);
+      }
+      return;
+    }
 #pragma unroll
     for (int vv = 0; vv < VPT; ++vv) {
       const int e = tid + 512 * vv;
@@ -193,41 +211,63 @@ __global__ __launch_bounds__(512) void attn_extend2_kernel(
     if (j + 1 < ntiles) load_tile(j + 1, krn, vrn);
 
     // ---- QK^T (swapped): A = K rows (32 keys), B = Q. C[key][qrow=col].
+    // The two subtiles interleave so consecutive MFMAs hit independent
+    // accumulators (the 32x32 dependent-accumulator latency exceeds the
+    // 32-cycle issue interval).
     f32x16 sc[2];
+    sc[0] = {};
+    sc[1] = {};
+    const char* krow0 = reinterpret_cast<const char*>(
+        k_lds + buf * KBUF + col * HEAD_DIM);
+    const char* krow1 = krow0 + 32 * HEAD_DIM * 2;
+    const int swz_mask = (col & 7) << 4;
 #pragma unroll
-    for (int sub = 0; sub < 2; ++sub) {
-      sc[sub] = {};
-      const int key = sub * 32 + col;
-      const char* krow = reinterpret_cast<const char*>(
-          k_lds + buf * KBUF + key * HEAD_DIM);
-      const int swz_mask = (key & 7) << 4;
-#pragma unroll
-      for (int s = 0; s < STEPS; ++s) {
-        const int off = ((s * 16 + 8 * half) * 2) ^ swz_mask;
-        ushort8 u = *reinterpret_cast<const ushort8*>(krow + off);
-        sc[sub] = e2_mfma32(*reinterpret_cast<e2_bf16x8*>(&u), qfrag[s], sc[sub]);
-      }
+    for (int s = 0; s < STEPS; ++s) {
+      const int off = ((s * 16 + 8 * half) * 2) ^ swz_mask;
+      ushort8 u0 = *reinterpret_cast<const ushort8*>(krow0 + off);
+      ushort8 u1 = *reinterpret_cast<const ushort8*>(krow1 + off);
+      sc[0] = e2_mfma32(*reinterpret_cast<e2_bf16x8*>(&u0), qfrag[s], sc[0]);
+      sc[1] = e2_mfma32(*reinterpret_cast<e2_bf16x8*>(&u1), qfrag[s], sc[1]);
     }
 
     // ---- Masked softmax, fully in-register (exp2 domain). Lane's 32
     // scores are all for q-row `my_qrow`; C reg 4g+r of subtile `sub` is
-    // key key_base + 32*sub + 8g + 4*half + r.
+    // key key_base + 32*sub + 8g + 4*half + r. Tiles fully visible to
+    // every row of this wave (the bulk of the causal region) skip the 32
+    // per-key compares; the reductions are trees, not 31-deep chains.
     float p[32];
-    float tmax = -FLT_MAX;
+    const bool full_vis =
+        (q0 + wave * 32 + 31 < q_len) &&
+        (key_base + E2_KVBLK - 1 <= kv_off + q0 + wave * 32) &&
+        (key_base + E2_KVBLK <= kmax) &&
+        (window <= 0 || key_base > kv_off + q0 + wave * 32 + 31 - window);
+    if (full_vis) {
 #pragma unroll
-    for (int sub = 0; sub < 2; ++sub) {
-      const int kb_s = key_base + 32 * sub + 4 * half;
+      for (int sub = 0; sub < 2; ++sub)
 #pragma unroll
-      for (int r16 = 0; r16 < 16; ++r16) {
-        const int kg = kb_s + 8 * (r16 >> 2) + (r16 & 3);
-        bool ok = qrow_valid && kg <= my_pos && kg < kmax;
-        if (window > 0) ok = ok && (kg > my_pos - window);
-        const int i = sub * 16 + r16;
-        p[i] = ok ? sc[sub][r16] * scale2 : -FLT_MAX;
-        tmax = fmaxf(tmax, p[i]);
+        for (int r16 = 0; r16 < 16; ++r16)
+          p[sub * 16 + r16] = sc[sub][r16] * scale2;
+    } else {
+#pragma unroll
+      for (int sub = 0; sub < 2; ++sub) {
+        const int kb_s = key_base + 32 * sub + 4 * half;
+#pragma unroll
+        for (int r16 = 0; r16 < 16; ++r16) {
+          const int kg = kb_s + 8 * (r16 >> 2) + (r16 & 3);
+          bool ok = qrow_valid && kg <= my_pos && kg < kmax;
+          if (window > 0) ok = ok && (kg > my_pos - window);
+          p[sub * 16 + r16] = ok ? sc[sub][r16] * scale2 : -FLT_MAX;
+        }
       }
     }
-    tmax = fmaxf(tmax, __shfl_xor(tmax, 32, WAVE_SIZE));
+    float mt[16];
+#pragma unroll
+    for (int i = 0; i < 16; ++i) mt[i] = fmaxf(p[i], p[i + 16]);
+#pragma unroll
+    for (int w = 8; w > 0; w >>= 1)
+#pragma unroll
+      for (int i = 0; i < w; ++i) mt[i] = fmaxf(mt[i], mt[i + w]);
+    float tmax = fmaxf(mt[0], __shfl_xor(mt[0], 32, WAVE_SIZE));
 
     // Defer-max: only rescale O when the tile max meaningfully exceeds the
     // running max (T13; exp2(THR)=256 stays comfortably finite).
@@ -247,14 +287,17 @@ __global__ __launch_bounds__(512) void attn_extend2_kernel(
       }
     }
 
-    float psum = 0.f;
 #pragma unroll
-    for (int i = 0; i < 32; ++i) {
+    for (int i = 0; i < 32; ++i)
       p[i] = (p[i] == -FLT_MAX) ? 0.f : __builtin_amdgcn_exp2f(p[i] - m_run);
-      psum += p[i];
-    }
-    psum += __shfl_xor(psum, 32, WAVE_SIZE);
-    l_run += psum;
+    float st[16];
+#pragma unroll
+    for (int i = 0; i < 16; ++i) st[i] = p[i] + p[i + 16];
+#pragma unroll
+    for (int w = 8; w > 0; w >>= 1)
+#pragma unroll
+      for (int i = 0; i < w; ++i) st[i] += st[i + w];
+    l_run += st[0] + __shfl_xor(st[0], 32, WAVE_SIZE);
 
     // ---- P -> bf16 PV A-fragments: per subtile, 8 cvt_pk words
     // W[g][0] = pk(p[4g], p[4g+1]), W[g][1] = pk(p[4g+2], p[4g+3]); the
@@ -289,12 +332,14 @@ __global__ __launch_bounds__(512) void attn_extend2_kernel(
     // column 16*(colhalf)+l&15 -> blocks kb = 4t+2*half (+1), cb per group).
     const int g4 = lane >> 4;  // 16-lane tr16 group
     bf16* vbase = v_img + buf * VBUF;
+    // t outer / n inner: consecutive MFMAs accumulate into the DTILES
+    // independent oacc registers instead of chaining one.
 #pragma unroll
-    for (int n = 0; n < DTILES; ++n) {
-      const int cb = 2 * n + (g4 & 1);
+    for (int t = 0; t < 4; ++t) {
+      const int kb = 4 * t + 2 * (g4 >> 1);
 #pragma unroll
-      for (int t = 0; t < 4; ++t) {
-        const int kb = 4 * t + 2 * (g4 >> 1);
+      for (int n = 0; n < DTILES; ++n) {
+        const int cb = 2 * n + (g4 & 1);
         e2_bf16x4 v1 = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(
             (__attribute__((address_space(3))) e2_bf16x4*)(
                 reinterpret_cast<char*>(vbase + (kb * CB + cb) * VBLK_ELEMS) +
