@@ -19,6 +19,9 @@ class ForwardBatch:
     # decode only:
     block_tables: torch.Tensor | None = None  # [num_seqs, max_blocks] int32
     seq_lens: torch.Tensor | None = None  # [num_seqs] int32 (device)
+    # prefill-extend only: precomputed work-sorted (tiles64, tiles256) for
+    # the paged extend kernels, shared by every layer of the forward
+    ext_tiles: tuple | None = None
     # rows of the hidden states from which logits are needed (last token of
     # each sequence for prefill; everything for decode)
     logits_indices: torch.Tensor | None = None

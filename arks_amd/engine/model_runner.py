@@ -233,6 +233,7 @@ class ModelRunner:
             dev = self.device
             block_tables = None
             seq_lens = None
+            ext_tiles = None
             if use_extend:
                 max_blocks = max(len(s.block_table) for s in sb.seqs)
                 bt = torch.zeros((len(sb.seqs), max_blocks), dtype=torch.int32)
@@ -242,6 +243,10 @@ class ModelRunner:
                     )
                 block_tables = bt.to(dev)
                 seq_lens = torch.tensor(kv_lens, dtype=torch.int32, device=dev)
+                ext_tiles = ops.build_extend_tiles(
+                    q_lens, kv_lens,
+                    self.cfg.kv_cache_dtype == "fp8", dev,
+                )
             return ForwardBatch(
                 is_prefill=True,
                 input_ids=torch.tensor(input_ids, dtype=torch.int64, device=dev),
@@ -251,6 +256,7 @@ class ModelRunner:
                 seq_lens_list=q_lens,
                 block_tables=block_tables,
                 seq_lens=seq_lens,
+                ext_tiles=ext_tiles,
                 logits_indices=torch.tensor(logits_idx, dtype=torch.int64, device=dev),
             )
         # decode

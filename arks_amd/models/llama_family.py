@@ -99,6 +99,7 @@ class Attention(nn.Module):
                 out = ops.attention_extend_paged(
                     q, k_cache, v_cache, batch.block_tables, batch.seq_lens,
                     batch.cu_seqlens, batch.seq_lens_list, self.scale,
+                    tiles=batch.ext_tiles,
                 )
             else:
                 out = ops.attention_prefill_varlen(

@@ -121,23 +121,36 @@ EXTEND2_QTILE = 256  # q rows per workgroup in the 8-wave ladder kernel
 EXTEND2_MIN_QLEN = 97
 
 
-def _split_extend_tiles(q_lens: list[int], kv_fp8: bool, device):
-    """Per-seq routing between the two extend kernels: (tiles64, tiles256)."""
+def build_extend_tiles(q_lens: list[int], kv_lens: list[int] | None,
+                       kv_fp8: bool, device):
+    """Per-seq routing between the two extend kernels: (tiles64, tiles256).
+
+    Tiles are sorted by DESCENDING key depth: causal q-tiles differ up to
+    4x in KV work, and the dispatcher issues blocks in order, so launching
+    deep tiles first removes the straggler tail (PMC: unsorted launch left
+    the chip ~half idle at serving shapes). Build once per batch and reuse
+    across all layers."""
     t64, t256 = [], []
+    kvl = kv_lens if kv_lens is not None else q_lens
     for i, n in enumerate(q_lens):
+        off = kvl[i] - n
         if kv_fp8 or n < EXTEND2_MIN_QLEN:
             for q0 in range(0, n, PREFILL_QTILE):
-                t64.append((i, q0))
+                t64.append((i, q0, min(kvl[i], off + q0 + PREFILL_QTILE)))
         else:
             for q0 in range(0, n, EXTEND2_QTILE):
-                t256.append((i, q0))
-    mk = lambda t: torch.tensor(t, dtype=torch.int32, device=device).reshape(-1, 2)
+                t256.append((i, q0, min(kvl[i], off + q0 + EXTEND2_QTILE)))
+    t64.sort(key=lambda t: -t[2])
+    t256.sort(key=lambda t: -t[2])
+    mk = lambda t: torch.tensor(
+        [x[:2] for x in t], dtype=torch.int32, device=device
+    ).reshape(-1, 2)
     return mk(t64), mk(t256)
 
 
 def attention_extend_paged(
     q, k_cache, v_cache, block_tables, kv_lens, cu_seqlens_q,
-    q_lens: list[int], scale: float, window: int = 0,
+    q_lens: list[int], scale: float, window: int = 0, tiles=None,
 ):
     """Causal attention of packed NEW tokens over each sequence's full paged
     KV history (cached prefix + new tokens already written by
@@ -154,7 +167,9 @@ def attention_extend_paged(
             (q.shape[0], q.shape[1], q.shape[2]), dtype=q.dtype, device=q.device
         )
         kv_fp8 = k_cache.dtype != torch.bfloat16
-        t64, t256 = _split_extend_tiles(q_lens, kv_fp8, q.device)
+        t64, t256 = tiles if tiles is not None else build_extend_tiles(
+            q_lens, None, kv_fp8, q.device
+        )
         if t256.numel():
             _native().attention_extend_paged2(
                 out, q, k_cache, v_cache, block_tables, kv_lens, cu_seqlens_q,

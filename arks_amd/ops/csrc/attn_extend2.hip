@@ -116,9 +116,13 @@ __global__ __launch_bounds__(512) void attn_extend2_kernel(
     }
   }
 
-  // Online-softmax state (exp2 domain; per lane for q-row `my_qrow`,
-  // replicated across the two halves).
-  float m_run = -FLT_MAX;
+  // Online-softmax state (per lane for q-row `my_qrow`, replicated across
+  // the two halves). The running max is kept in RAW score units (scale is
+  // folded into the exp2 argument); masked scores use a sentinel so far
+  // below any real score that exp2(fma(sentinel, scale2, -m*scale2))
+  // underflows to 0 with no per-element select.
+  constexpr float MASKED = -3e38f;
+  float m_run = -1e30f;
   float l_run = 0.f;
   f32x16 oacc[DTILES];
 #pragma unroll
@@ -246,7 +250,7 @@ __global__ __launch_bounds__(512) void attn_extend2_kernel(
       for (int sub = 0; sub < 2; ++sub)
 #pragma unroll
         for (int r16 = 0; r16 < 16; ++r16)
-          p[sub * 16 + r16] = sc[sub][r16] * scale2;
+          p[sub * 16 + r16] = sc[sub][r16];
     } else {
 #pragma unroll
       for (int sub = 0; sub < 2; ++sub) {
@@ -256,7 +260,7 @@ __global__ __launch_bounds__(512) void attn_extend2_kernel(
           const int kg = kb_s + 8 * (r16 >> 2) + (r16 & 3);
           bool ok = qrow_valid && kg <= my_pos && kg < kmax;
           if (window > 0) ok = ok && (kg > my_pos - window);
-          p[sub * 16 + r16] = ok ? sc[sub][r16] * scale2 : -FLT_MAX;
+          p[sub * 16 + r16] = ok ? sc[sub][r16] : MASKED;
         }
       }
     }
@@ -267,13 +271,15 @@ __global__ __launch_bounds__(512) void attn_extend2_kernel(
     for (int w = 8; w > 0; w >>= 1)
 #pragma unroll
       for (int i = 0; i < w; ++i) mt[i] = fmaxf(mt[i], mt[i + w]);
-    float tmax = fmaxf(mt[0], __shfl_xor(mt[0], 32, WAVE_SIZE));
+    const float tmax = fmaxf(mt[0], __shfl_xor(mt[0], 32, WAVE_SIZE));
 
-    // Defer-max: only rescale O when the tile max meaningfully exceeds the
-    // running max (T13; exp2(THR)=256 stays comfortably finite).
-    if (!__all(tmax - m_run <= RESCALE_THR)) {
+    // Defer-max (T13): only rescale O when the tile max meaningfully
+    // exceeds the running max (threshold in raw units; exp2(THR)=256
+    // stays comfortably finite).
+    const float thr_raw = RESCALE_THR / scale2;
+    if (!__all(tmax - m_run <= thr_raw)) {
       const float m_new = fmaxf(m_run, tmax);
-      const float alpha = __builtin_amdgcn_exp2f(m_run - m_new);
+      const float alpha = __builtin_amdgcn_exp2f((m_run - m_new) * scale2);
       l_run *= alpha;
       m_run = m_new;
       // O rows live across regs (C layout): fetch each row's alpha from
@@ -287,9 +293,12 @@ __global__ __launch_bounds__(512) void attn_extend2_kernel(
       }
     }
 
+    // p = exp2((s - m) * scale2) as one fma + exp2; MASKED entries
+    // underflow to exactly 0 (|MASKED*scale2| >> |m*scale2|).
+    const float nms = -m_run * scale2;
 #pragma unroll
     for (int i = 0; i < 32; ++i)
-      p[i] = (p[i] == -FLT_MAX) ? 0.f : __builtin_amdgcn_exp2f(p[i] - m_run);
+      p[i] = __builtin_amdgcn_exp2f(__builtin_fmaf(p[i], scale2, nms));
     float st[16];
 #pragma unroll
     for (int i = 0; i < 16; ++i) st[i] = p[i] + p[i + 16];

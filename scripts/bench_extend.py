@@ -28,12 +28,16 @@ def bench(fn, iters=30):
     return (time.time() - t0) / iters * 1e6
 
 
-def tiles256(q_lens, device):
+def tiles256(q_lens, kv_lens, device):
+    """Work-sorted 256-row tiles (mirrors ops.build_extend_tiles)."""
     t = []
     for i, n in enumerate(q_lens):
+        off = kv_lens[i] - n
         for q0 in range(0, n, 256):
-            t.append((i, q0))
-    return torch.tensor(t, dtype=torch.int32, device=device).reshape(-1, 2)
+            t.append((i, q0, min(kv_lens[i], off + q0 + 256)))
+    t.sort(key=lambda x: -x[2])
+    return torch.tensor([x[:2] for x in t], dtype=torch.int32,
+                        device=device).reshape(-1, 2)
 
 
 def main():
@@ -60,7 +64,7 @@ def main():
         flops = 2 * 2 * hq * hd * pairs
         out = torch.empty_like(q)
         t64i = build_prefill_tiles([qlen] * S, "cuda")
-        t256 = tiles256([qlen] * S, "cuda")
+        t256 = tiles256([qlen] * S, [kvlen] * S, "cuda")
         nat = ops._native()
         t_old = bench(lambda: nat.attention_extend_paged(
             out, q, kc, vc, bt, kvl, cu, t64i, scale, 0))
