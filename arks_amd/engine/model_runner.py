@@ -183,26 +183,30 @@ class ModelRunner:
             # suffix runs as q rows; attention then reads the full paged KV
             # through the extend kernel.
             use_extend = any(s.num_cached_tokens > 0 for s in sb.seqs)
-            input_ids: list[int] = []
-            positions: list[int] = []
-            slot_mapping: list[int] = []
+            input_ids: list = []  # np.int64 arrays, one per seq
+            positions: list = []
+            slot_mapping: list = []
             cu = [0]
             q_lens: list[int] = []
             kv_lens: list[int] = []
             logits_idx: list[int] = []
             lp_meta: list[tuple[int, int, int, int]] = []
             self._lp_prefill = None
+            import numpy as np
+
             for i, seq in enumerate(sb.seqs):
                 toks = seq.all_token_ids
                 c = seq.num_cached_tokens
                 # chunked prefill: run exactly the scheduled token count
                 # (== everything remaining unless the budget split it)
                 n = c + sb.num_new_tokens[i]
-                input_ids.extend(toks[c:n])
-                positions.extend(range(c, n))
-                for pos in range(c, n):
-                    b = seq.block_table[pos // bs]
-                    slot_mapping.append(b * bs + pos % bs)
+                input_ids.append(np.asarray(toks[c:n], dtype=np.int64))
+                pos = np.arange(c, n, dtype=np.int64)
+                positions.append(pos)
+                # vectorized slot mapping (a per-token Python loop here cost
+                # ~2.8 ms of GPU idle per 8k-token chunk)
+                btn = np.asarray(seq.block_table, dtype=np.int64)
+                slot_mapping.append(btn[pos // bs] * bs + pos % bs)
                 cu.append(cu[-1] + (n - c))
                 q_lens.append(n - c)
                 kv_lens.append(n)
@@ -249,9 +253,9 @@ class ModelRunner:
                 )
             return ForwardBatch(
                 is_prefill=True,
-                input_ids=torch.tensor(input_ids, dtype=torch.int64, device=dev),
-                positions=torch.tensor(positions, dtype=torch.int64, device=dev),
-                slot_mapping=torch.tensor(slot_mapping, dtype=torch.int64, device=dev),
+                input_ids=torch.from_numpy(np.concatenate(input_ids)).to(dev),
+                positions=torch.from_numpy(np.concatenate(positions)).to(dev),
+                slot_mapping=torch.from_numpy(np.concatenate(slot_mapping)).to(dev),
                 cu_seqlens=torch.tensor(cu, dtype=torch.int32, device=dev),
                 seq_lens_list=q_lens,
                 block_tables=block_tables,
