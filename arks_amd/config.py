@@ -34,7 +34,7 @@ class ModelConfig:
     moe_intermediate_size: int | None = None  # per-expert FFN width (MoE)
     shared_expert_intermediate_size: int | None = None  # Qwen2-MoE
     norm_topk_prob: bool = True  # renormalize top-k routing weights
-    sliding_window: int | None = None  # SWA not implemented: must be None
+    sliding_window: int | None = None  # sliding-window attention size
     qk_norm: bool = False  # Qwen3: per-head RMSNorm on q/k before RoPE
     tie_word_embeddings: bool = False
     attention_bias: bool = True  # qwen2 has qkv bias; llama does not
@@ -43,11 +43,11 @@ class ModelConfig:
     torch_dtype: str = "bfloat16"
 
     def __post_init__(self):
-        # sliding_window is allowed to be SET (Mistral-v0.1-style configs
-        # declare 4096) — the engine enforces max_model_len <= window at
-        # startup so the window never actually binds (full attention below
-        # the window is identical to SWA); true windowed attention is a
-        # next-round kernel feature.
+        # Sliding-window attention (Mistral-v0.1-style configs declare
+        # 4096) is implemented end to end: the attention kernels mask keys
+        # outside each query's window and the engine drops out-of-window
+        # KV pages. Qwen2-style per-layer max_window_layers mixing is not
+        # modeled: a declared window applies to every layer.
         if self.sliding_window is not None and self.sliding_window <= 0:
             raise ValueError(
                 f"invalid sliding_window={self.sliding_window}"
@@ -169,6 +169,22 @@ PRESET_CONFIGS: dict[str, ModelConfig] = {
         attention_bias=False,
         eos_token_id=128001,
         bos_token_id=128000,
+    ),
+    "mistral-7b": ModelConfig(  # v0.1: global sliding window 4096
+        architecture="MistralForCausalLM",
+        vocab_size=32000,
+        hidden_size=4096,
+        intermediate_size=14336,
+        num_hidden_layers=32,
+        num_attention_heads=32,
+        num_key_value_heads=8,
+        head_dim=128,
+        rms_norm_eps=1e-5,
+        rope_theta=10000.0,
+        attention_bias=False,
+        sliding_window=4096,
+        eos_token_id=2,
+        bos_token_id=1,
     ),
     "mixtral-8x7b": ModelConfig(
         architecture="MixtralForCausalLM",

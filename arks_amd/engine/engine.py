@@ -18,15 +18,14 @@ class LLMEngine:
     def __init__(self, cfg: EngineConfig):
         self.cfg = cfg
         self.model_cfg = cfg.model_config()
-        sw = self.model_cfg.sliding_window
-        if sw is not None and cfg.max_model_len > sw:
-            # below the window, full attention == sliding-window attention;
-            # beyond it we would silently change the model's semantics
-            raise ValueError(
-                f"model declares sliding_window={sw} and windowed attention "
-                f"is not implemented: set max_model_len <= {sw} "
-                f"(requested {cfg.max_model_len})"
-            )
+        # Sliding-window attention (Mistral-style): the kernels mask keys
+        # outside each query's window and out-of-window KV pages are dropped
+        # back to the pool as sequences grow. Prefix caching is disabled for
+        # SWA models: a cached prefix page can be dropped+recycled while its
+        # digest is still matchable, so reuse would read stale pages.
+        self.window = self.model_cfg.sliding_window or 0
+        if self.window and cfg.enable_prefix_caching:
+            cfg.enable_prefix_caching = False
         self.runner = ModelRunner(cfg, self.model_cfg)
         t0 = time.time()
         self.runner.load_weights()
@@ -124,6 +123,8 @@ class LLMEngine:
 
     # ---- the step loop ----
     def step(self) -> list[StepOutput]:
+        if self.window:
+            self._drop_window_pages()
         sb = self.scheduler.schedule()
         if sb is None:
             return []
@@ -167,6 +168,26 @@ class LLMEngine:
             )
         self.scheduler.free_finished()
         return outputs
+
+    def _drop_window_pages(self) -> None:
+        """Free KV pages wholly below every running sequence's attention
+        window (block-table entries become -1; the kernels never read them).
+        Keeps long SWA generations at O(window) KV instead of O(length)."""
+        bs = self.cfg.block_size
+        for seq in self.scheduler.running:
+            if seq.num_tokens <= self.window:
+                continue
+            # pages [0, lim) hold only tokens < num_tokens - window
+            lim = (seq.num_tokens - self.window) // bs
+            start = getattr(seq, "window_dropped", 0)
+            if lim <= start:
+                continue
+            drop = [b for b in seq.block_table[start:lim] if b >= 0]
+            if drop:
+                self.scheduler.allocator.free(drop)
+            for p in range(start, lim):
+                seq.block_table[p] = -1
+            seq.window_dropped = lim
 
     def _spec_step(self, sb) -> list[StepOutput]:
         """Decode step with prompt-lookup speculation: propose drafts from

@@ -43,7 +43,8 @@ class BlockAllocator:
         return [self._free.pop() for _ in range(n)]
 
     def free(self, blocks: list[int]) -> None:
-        self._free.extend(reversed(blocks))
+        # -1 entries are sliding-window-dropped pages (already freed)
+        self._free.extend(b for b in reversed(blocks) if b >= 0)
 
     # --- prefix-caching hooks (no-ops here) ---
     def match_prefix(self, token_ids: list[int], max_tokens: int) -> tuple[list[int], int]:
@@ -112,6 +113,8 @@ class PrefixCachingAllocator(BlockAllocator):
 
     def free(self, blocks: list[int]) -> None:
         for b in blocks:
+            if b < 0:  # sliding-window-dropped page (already freed)
+                continue
             self._ref[b] -= 1
             assert self._ref[b] >= 0, f"double free of block {b}"
             if self._ref[b] == 0:

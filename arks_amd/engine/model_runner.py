@@ -148,7 +148,10 @@ class ModelRunner:
         transfer to a decode instance (the reference's PD
         split delegates this to SGLang's disaggregation-mode, reference
         arksdisaggregatedapplication_controller.go:1672-1724)."""
-        bt = torch.tensor(block_table, dtype=torch.long, device=self.device)
+        # sliding-window-dropped pages are -1: clamp to a valid page (the
+        # receiver never reads out-of-window content)
+        bt = torch.tensor(block_table, dtype=torch.long,
+                          device=self.device).clamp(min=0)
         layers = [torch.stack((kc[bt], vc[bt]), 0) for kc, vc in self.kv_caches]
         kv = torch.stack(layers, 0)
         # TP>1: each rank holds a contiguous nkv/tp head slice — all-gather

@@ -55,6 +55,9 @@ class Attention(nn.Module):
         self.nq_local = cfg.num_attention_heads // tp
         self.nkv_local = cfg.num_key_value_heads // tp
         self.scale = 1.0 / math.sqrt(cfg.head_dim)
+        # sliding-window attention (Mistral-style, applied globally when the
+        # config declares one; window masking happens inside the kernels)
+        self.window = cfg.sliding_window or 0
         self.qkv_proj = QKVParallelLinear(
             cfg.hidden_size, cfg.head_dim, cfg.num_attention_heads,
             cfg.num_key_value_heads, bias=cfg.attention_bias, dtype=dtype,
@@ -99,15 +102,17 @@ class Attention(nn.Module):
                 out = ops.attention_extend_paged(
                     q, k_cache, v_cache, batch.block_tables, batch.seq_lens,
                     batch.cu_seqlens, batch.seq_lens_list, self.scale,
-                    tiles=batch.ext_tiles,
+                    window=self.window, tiles=batch.ext_tiles,
                 )
             else:
                 out = ops.attention_prefill_varlen(
-                    q, k, v, batch.cu_seqlens, batch.seq_lens_list, self.scale
+                    q, k, v, batch.cu_seqlens, batch.seq_lens_list, self.scale,
+                    window=self.window,
                 )
         else:
             out = ops.attention_decode_paged(
-                q, k_cache, v_cache, batch.block_tables, batch.seq_lens, self.scale
+                q, k_cache, v_cache, batch.block_tables, batch.seq_lens,
+                self.scale, window=self.window,
             )
         return self.o_proj(out.view(T, -1))
 

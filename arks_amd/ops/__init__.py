@@ -103,15 +103,18 @@ def build_prefill_tiles(seq_lens: list[int], device) -> torch.Tensor:
     return torch.tensor(tiles, dtype=torch.int32, device=device).reshape(-1, 2)
 
 
-def attention_prefill_varlen(q, k, v, cu_seqlens, seq_lens: list[int], scale: float):
+def attention_prefill_varlen(q, k, v, cu_seqlens, seq_lens: list[int],
+                             scale: float, window: int = 0):
     """Causal varlen prefill attention. seq_lens is the host-side list of
-    per-sequence lengths (used to build the q-tile grid without a D2H sync)."""
+    per-sequence lengths (used to build the q-tile grid without a D2H sync).
+    window > 0 = sliding-window attention."""
     if q.is_cuda:
         out = torch.empty_like(q)
         tile_info = build_prefill_tiles(seq_lens, q.device)
-        _native().attention_prefill_varlen(out, q, k, v, cu_seqlens, tile_info, scale)
+        _native().attention_prefill_varlen(out, q, k, v, cu_seqlens, tile_info,
+                                           scale, window)
         return out
-    return ref.attention_prefill_varlen(q, k, v, cu_seqlens, scale)
+    return ref.attention_prefill_varlen(q, k, v, cu_seqlens, scale, window)
 
 
 EXTEND2_QTILE = 256  # q rows per workgroup in the 8-wave ladder kernel

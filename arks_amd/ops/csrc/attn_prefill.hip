@@ -50,7 +50,7 @@ __global__ __launch_bounds__(256) void attn_prefill_kernel(
     const int* __restrict__ cu_seqlens,  // [num_seqs + 1]
     const int* __restrict__ tile_info,   // [ntiles, 2] = (seq_idx, q0)
     const float scale, const int num_q_heads, const int num_kv_heads,
-    const int64_t q_stride, const int64_t kv_stride) {
+    const int64_t q_stride, const int64_t kv_stride, const int window) {
   constexpr int CHUNKS = HEAD_DIM / 16;  // dim chunks for PV output
   constexpr int STEPS = HEAD_DIM / 32;   // K-contraction steps for QK^T
 
@@ -112,6 +112,8 @@ __global__ __launch_bounds__(256) void attn_prefill_kernel(
   // Keys visible to this workgroup: strictly below kmax.
   const int kmax = min(seq_len, q0 + QTILE);
   const int ntiles = (kmax + KTILE - 1) / KTILE;
+  // Sliding window: first tile any row of this q-tile can see.
+  const int j0 = (window > 0) ? max(0, (q0 - window + 1) / KTILE) : 0;
 
   // Per-thread cooperative staging share: tile elements i = tid + 256*v.
   constexpr int NVEC = KTILE * HEAD_DIM / 8;      // vec8 per tile
@@ -157,13 +159,13 @@ __global__ __launch_bounds__(256) void attn_prefill_kernel(
 
   {
     ushort8 kr0[VPT], vr0[VPT];
-    load_tile(0, kr0, vr0);
+    load_tile(j0, kr0, vr0);
     store_tile(0, kr0, vr0);
   }
   __syncthreads();
 
   int buf = 0;
-  for (int j = 0; j < ntiles; ++j, buf ^= 1) {
+  for (int j = j0; j < ntiles; ++j, buf ^= 1) {
     const int key_base = j * KTILE;
     // ---- prefetch tile j+1 into registers (global loads overlap MFMAs).
     ushort8 krn[VPT], vrn[VPT];
@@ -197,7 +199,8 @@ __global__ __launch_bounds__(256) void attn_prefill_kernel(
       for (int r = 0; r < 4; ++r) {
         const int kg = key_base + sub * 16 + 4 * la + r;
         const int i = sub * 4 + r;
-        msk[i] = qrow_valid && (kg <= my_qrow) && (kg < kmax);
+        msk[i] = qrow_valid && (kg <= my_qrow) && (kg < kmax) &&
+                 (window <= 0 || kg > my_qrow - window);
         p[i] = msk[i] ? sc[sub][r] * scale : -FLT_MAX;
         tile_max = fmaxf(tile_max, p[i]);
       }
@@ -401,19 +404,19 @@ extern "C" void arks_attn_prefill_varlen(
     void* out, const void* q, const void* k, const void* v,
     const void* cu_seqlens, const void* tile_info, int ntiles, float scale,
     int num_q_heads, int num_kv_heads, int head_dim, int64_t q_stride,
-    int64_t kv_stride, hipStream_t stream) {
+    int64_t kv_stride, int window, hipStream_t stream) {
   dim3 grid(num_q_heads, ntiles), block(256);
   if (head_dim == 128) {
     hipLaunchKernelGGL((attn_prefill_kernel<128>), grid, block, 0, stream,
                        (bf16*)out, (const bf16*)q, (const bf16*)k,
                        (const bf16*)v, (const int*)cu_seqlens,
                        (const int*)tile_info, scale, num_q_heads, num_kv_heads,
-                       q_stride, kv_stride);
+                       q_stride, kv_stride, window);
   } else if (head_dim == 64) {
     hipLaunchKernelGGL((attn_prefill_kernel<64>), grid, block, 0, stream,
                        (bf16*)out, (const bf16*)q, (const bf16*)k,
                        (const bf16*)v, (const int*)cu_seqlens,
                        (const int*)tile_info, scale, num_q_heads, num_kv_heads,
-                       q_stride, kv_stride);
+                       q_stride, kv_stride, window);
   }
 }

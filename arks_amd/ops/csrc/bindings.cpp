@@ -41,7 +41,7 @@ void arks_attn_prefill_varlen(void* out, const void* q, const void* k,
                               const void* v, const void* cu_seqlens,
                               const void* tile_info, int ntiles, float scale,
                               int num_q_heads, int num_kv_heads, int head_dim,
-                              int64_t q_stride, int64_t kv_stride,
+                              int64_t q_stride, int64_t kv_stride, int window,
                               hipStream_t stream);
 void arks_attn_extend_paged(void* out, const void* q, const void* k_cache,
                             const void* v_cache, const void* block_tables,
@@ -222,7 +222,7 @@ void attention_decode_paged(torch::Tensor out, torch::Tensor q,
 void attention_prefill_varlen(torch::Tensor out, torch::Tensor q,
                               torch::Tensor k, torch::Tensor v,
                               torch::Tensor cu_seqlens, torch::Tensor tile_info,
-                              double scale) {
+                              double scale, int64_t window) {
   check_bf16_contig(out, "out");
   check_bf16_rowstrided(q, "q");
   check_bf16_rowstrided(k, "k");
@@ -240,7 +240,7 @@ void attention_prefill_varlen(torch::Tensor out, torch::Tensor q,
                            v.data_ptr(), cu_seqlens.data_ptr(),
                            tile_info.data_ptr(), ntiles, (float)scale,
                            num_q_heads, num_kv_heads, head_dim, q.stride(0),
-                           k.stride(0), current_stream());
+                           k.stride(0), (int)window, current_stream());
 }
 
 void attention_extend_paged(torch::Tensor out, torch::Tensor q,

@@ -156,9 +156,11 @@ def attention_prefill_varlen(
     v: torch.Tensor,
     cu_seqlens: torch.Tensor,  # [num_seqs + 1] int32
     scale: float,
+    window: int = 0,
 ) -> torch.Tensor:
     """Causal attention over packed variable-length sequences (full prefill:
-    keys == the packed k/v of the same forward). GQA by head repetition."""
+    keys == the packed k/v of the same forward). GQA by head repetition.
+    window > 0 = sliding-window attention."""
     num_q_heads = q.shape[1]
     num_kv_heads = k.shape[1]
     rep = num_q_heads // num_kv_heads
@@ -173,6 +175,11 @@ def attention_prefill_varlen(
         mask = torch.triu(
             torch.full((L, L), float("-inf"), device=q.device), diagonal=1
         )
+        if window > 0:
+            mask = mask + torch.tril(
+                torch.full((L, L), float("-inf"), device=q.device),
+                diagonal=-window,
+            )
         scores = scores + mask
         p = torch.softmax(scores, dim=-1)
         o = torch.matmul(p, vi)  # [H, L, D]
