@@ -229,3 +229,36 @@ def test_spec_rejection_sampling_gpu():
     got = e.runner.execute_spec(sb, [list(draft)])[0]
     assert 1 <= len(got) <= len(draft) + 1
     assert all(0 <= t < e.model_cfg.vocab_size for t in got)
+
+
+def test_sliding_window_engine_gpu():
+    """SWA served past its window on the native kernels: page dropping must
+    not change outputs (graphed decode included)."""
+    import dataclasses
+
+    import arks_amd.config as C
+
+    torch.manual_seed(0)
+    window = 64
+    swa = dataclasses.replace(C.PRESET_CONFIGS["tiny-gpu"],
+                              sliding_window=window)
+    C.PRESET_CONFIGS["tiny-gpu-swa"] = swa
+    try:
+        def run(drop: bool):
+            e = LLMEngine(EngineConfig(
+                preset="tiny-gpu-swa", device="cuda", kv_cache_blocks=512,
+                max_model_len=512, max_num_seqs=8,
+            ))
+            if not drop:
+                e._drop_window_pages = lambda: None
+            return e, e.generate(
+                [[(i * 11 + 2) % 99 for i in range(100)], [7, 3] * 40],
+                SamplingParams(max_tokens=80, ignore_eos=True),
+            )
+
+        e1, o1 = run(True)
+        _, o2 = run(False)
+        assert o1 == o2
+        assert all(len(o) == 80 for o in o1)
+    finally:
+        C.PRESET_CONFIGS.pop("tiny-gpu-swa", None)
