@@ -5,8 +5,9 @@ ext_proc side-call).
     python -m arks_amd.gateway --port 8080 [--kube-api https://...]
 
 Token/Quota/Endpoint CRs are read live from the cluster (KubeStore resync);
-rate-limit and quota state is in-process (fixed-window counters with the
-reference's redis key semantics — arks_amd/gateway/limiter.py).
+rate-limit and quota state is in-process by default (fixed-window counters
+with the reference's redis key semantics — arks_amd/gateway/limiter.py), or
+shared across gateway replicas via --redis-addr (RESP-backed stores).
 """
 
 from __future__ import annotations
@@ -23,11 +24,28 @@ def main(argv=None):
     ap.add_argument("--resync", type=float, default=5.0)
     ap.add_argument("--standalone", action="store_true",
                     help="in-memory store (no cluster; for local testing)")
+    ap.add_argument("--redis-addr", default=None, metavar="HOST:PORT",
+                    help="Redis for shared rate-limit/quota state (multi-"
+                         "replica gateways; reference cmd/gateway/main.go "
+                         "redis flags). Default: in-process counters.")
     args = ap.parse_args(argv)
 
     import uvicorn
 
     from .app import create_gateway_app
+
+    limiter = quota_service = None
+    if args.redis_addr:
+        from .limiter import RateLimiter, RedisCounterStore
+        from .quota import QuotaService, RedisQuotaStore
+        from .resp import RespClient
+
+        host, _, port = args.redis_addr.partition(":")
+        client = RespClient(host, int(port or 6379))
+        if not client.ping():
+            raise SystemExit(f"redis at {args.redis_addr} is unreachable")
+        limiter = RateLimiter(store=RedisCounterStore(client))
+        quota_service = QuotaService(store=RedisQuotaStore(client))
 
     if args.standalone:
         from ..controlplane.store import Store
@@ -39,7 +57,8 @@ def main(argv=None):
         store = KubeStore(api_base=args.kube_api)
         threading.Thread(target=store.run_resync, args=(args.resync,),
                          daemon=True).start()
-    app = create_gateway_app(store)
+    app = create_gateway_app(store, limiter=limiter,
+                             quota_service=quota_service)
     uvicorn.run(app, host=args.host, port=args.port, log_level="info")
 
 

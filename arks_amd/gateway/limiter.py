@@ -73,6 +73,41 @@ class InMemoryCounterStore:
                 del self._data[k]
 
 
+class RedisCounterStore:
+    """Redis(RESP)-backed counter store — the reference's production
+    backend (pkg/gateway/ratelimiter/redis_impl.go:115-165): pipelined
+    INCRBY + TTL, EXPIRE(window + jitter <= 1 s) set only when the key has
+    no TTL yet. Any number of gateway replicas share one budget."""
+
+    def __init__(self, client):
+        self.client = client  # arks_amd.gateway.resp.RespClient
+
+    def incr_by(self, key: str, amount: int, expire_s: float, now: float) -> int:
+        import random
+
+        val, ttl = self.client.pipeline(
+            [("INCRBY", key, amount), ("TTL", key)]
+        )
+        if isinstance(val, Exception):
+            raise val
+        if isinstance(ttl, int) and ttl < 0:
+            self.client.command(
+                "EXPIRE", key, int(expire_s + random.uniform(0.0, 1.0))
+            )
+        return int(val)
+
+    def get(self, key: str, now: float) -> int:
+        v = self.client.command("GET", key)
+        return int(v) if v is not None else 0
+
+    def set(self, key: str, value: int, now: float,
+            expire_s: float | None = None) -> None:
+        if expire_s:
+            self.client.command("SET", key, value, "EX", int(expire_s))
+        else:
+            self.client.command("SET", key, value)
+
+
 @dataclass(frozen=True)
 class LimitDescriptor:
     namespace: str

@@ -1,6 +1,8 @@
 """Cumulative quota service (reference pkg/gateway/quota/): lifetime
 counters (no window/TTL) keyed namespace/quota-name/type; over-limit when
-current > limit."""
+current > limit. The store is pluggable: in-memory for single-replica /
+test runs, Redis (RESP) for shared multi-replica state (reference
+quota/redis_impl.go:37-110)."""
 
 from __future__ import annotations
 
@@ -18,29 +20,60 @@ class QuotaDescriptor:
     limit: int
 
 
-class QuotaService:
-    KEY_PREFIX = "arks-quota"
-
+class InMemoryQuotaStore:
     def __init__(self):
         self._lock = threading.Lock()
         self._usage: dict[str, int] = {}
+
+    def incr_by(self, key: str, amount: int) -> int:
+        with self._lock:
+            self._usage[key] = self._usage.get(key, 0) + amount
+            return self._usage[key]
+
+    def set(self, key: str, value: int) -> None:
+        with self._lock:
+            self._usage[key] = value
+
+    def get(self, key: str) -> int:
+        with self._lock:
+            return self._usage.get(key, 0)
+
+
+class RedisQuotaStore:
+    """Cumulative lifetime counters in Redis — no window, no TTL
+    (reference quota/cache_key.go:41-57, redis_impl.go:37-110)."""
+
+    def __init__(self, client):
+        self.client = client  # arks_amd.gateway.resp.RespClient
+
+    def incr_by(self, key: str, amount: int) -> int:
+        return int(self.client.command("INCRBY", key, amount))
+
+    def set(self, key: str, value: int) -> None:
+        self.client.command("SET", key, value)
+
+    def get(self, key: str) -> int:
+        v = self.client.command("GET", key)
+        return int(v) if v is not None else 0
+
+
+class QuotaService:
+    KEY_PREFIX = "arks-quota"
+
+    def __init__(self, store=None):
+        self.store = store if store is not None else InMemoryQuotaStore()
 
     def _key(self, ns: str, name: str, type_: str) -> str:
         return f"{self.KEY_PREFIX}:ns={ns}:quota={name}:type={type_}"
 
     def incr_usage(self, ns: str, name: str, type_: str, amount: int) -> int:
-        with self._lock:
-            k = self._key(ns, name, type_)
-            self._usage[k] = self._usage.get(k, 0) + amount
-            return self._usage[k]
+        return self.store.incr_by(self._key(ns, name, type_), amount)
 
     def set_usage(self, ns: str, name: str, type_: str, value: int) -> None:
-        with self._lock:
-            self._usage[self._key(ns, name, type_)] = value
+        self.store.set(self._key(ns, name, type_), value)
 
     def get_usage(self, ns: str, name: str, type_: str) -> int:
-        with self._lock:
-            return self._usage.get(self._key(ns, name, type_), 0)
+        return self.store.get(self._key(ns, name, type_))
 
     def check(self, descriptors: list[QuotaDescriptor]) -> tuple[bool, str | None]:
         """Over-limit when current > limit (reference quota/redis_impl.go:101)."""

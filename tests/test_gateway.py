@@ -269,7 +269,7 @@ def test_quota_sync_loop_updates_cr_status(stack):
         used = {e.type: e.used for e in quota.status.quota_status}
         assert used["response"] == 3
         # crash recovery: wipe live counters, sync pushes CR values back
-        gw.state.quota_service._usage.clear()
+        gw.state.quota_service.store._usage.clear()
         gw.state.provider.sync_quota_usage()
         assert gw.state.quota_service.get_usage("default", "q1", "response") == 3
 
