@@ -24,6 +24,9 @@ def main(argv=None):
     ap.add_argument("--resync", type=float, default=5.0)
     ap.add_argument("--standalone", action="store_true",
                     help="in-memory store (no cluster; for local testing)")
+    ap.add_argument("--ext-proc-port", type=int, default=0,
+                    help="also serve the Envoy ext_proc gRPC protocol on "
+                         "this port (reference default 50052; 0 = off)")
     ap.add_argument("--redis-addr", default=None, metavar="HOST:PORT",
                     help="Redis for shared rate-limit/quota state (multi-"
                          "replica gateways; reference cmd/gateway/main.go "
@@ -59,6 +62,11 @@ def main(argv=None):
                          daemon=True).start()
     app = create_gateway_app(store, limiter=limiter,
                              quota_service=quota_service)
+    if args.ext_proc_port:
+        from .extproc import serve as serve_extproc
+
+        serve_extproc(app.state.provider, app.state.limiter,
+                      app.state.quota_service, port=args.ext_proc_port)
     uvicorn.run(app, host=args.host, port=args.port, log_level="info")
 
 
