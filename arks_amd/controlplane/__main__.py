@@ -67,10 +67,13 @@ def main(argv=None):
 
     health = ThreadingHTTPServer(("0.0.0.0", args.health_port), Probe)
     threading.Thread(target=health.serve_forever, daemon=True).start()
-    threading.Thread(target=store.run_resync, args=(args.resync,),
+    # watch streams (informer path); a slow resync loop remains as the
+    # safety net against missed events
+    store.run_watch()
+    threading.Thread(target=store.run_resync, args=(max(args.resync, 300.0),),
                      daemon=True).start()
-    print(f"arks operator running (resync {args.resync}s, "
-          f"health :{args.health_port})", flush=True)
+    print(f"arks operator running (watch + {max(args.resync, 300.0):.0f}s "
+          f"resync, health :{args.health_port})", flush=True)
     try:
         op.run()
     except KeyboardInterrupt:

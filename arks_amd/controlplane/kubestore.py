@@ -9,9 +9,9 @@ controller-runtime manager; here Operator + KubeStore fill that role):
   * workload kinds (Pod/Service/PVC/Deployment/LWS/RBGS/HTTPRoute) stay raw
     dicts;
   * status is written through the /status subresource when present;
-  * watch is a poll-based resync loop (list + diff) that feeds the
-    Operator's work queue — simple, restart-safe and sufficient at
-    control-plane rates.
+  * events come from real watch streams (run_watch: one list to seed,
+    then long-lived `?watch=1` requests per kind, relist on 410 Gone),
+    with the list+diff resync loop kept as a low-frequency safety net.
 """
 
 from __future__ import annotations
@@ -257,6 +257,112 @@ class KubeStore:
         while not self._stop.is_set():
             self.resync_once()
             self._stop.wait(interval_s)
+
+    # -- watch streams (the informer path; the resync loop above is the
+    #    fallback for servers without watch support) --
+    def _list_raw(self, kind: str) -> tuple[list[dict], str]:
+        """List a kind cluster-wide, returning (items, list resourceVersion)."""
+        prefix, plural = API_MAP[kind]
+        r = self._client.get(f"{prefix}/{plural}")
+        r.raise_for_status()
+        body = r.json()
+        return (body.get("items", []),
+                str(body.get("metadata", {}).get("resourceVersion", "")))
+
+    def _seed_kind(self, kind: str) -> str:
+        """Initial list: emit ADDED/MODIFIED/DELETED diffs vs the known
+        state, return the collection resourceVersion to watch from."""
+        items, rv = self._list_raw(kind)
+        seen = set()
+        for raw in items:
+            m = raw.get("metadata", {})
+            key = (kind, m.get("namespace", "default"), m.get("name", ""))
+            seen.add(key)
+            orv = str(m.get("resourceVersion", ""))
+            if self._known.get(key) != orv:
+                event = "MODIFIED" if key in self._known else "ADDED"
+                self._known[key] = orv
+                self._notify(event, _from_dict(kind, raw))
+        for key in [k for k in self._known
+                    if k[0] == kind and k not in seen]:
+            _, ns, name = key
+            del self._known[key]
+            self._notify("DELETED", {"kind": kind,
+                                     "metadata": {"namespace": ns,
+                                                  "name": name}})
+        return rv
+
+    def _watch_kind_once(self, kind: str, rv: str,
+                         timeout_s: float = 300.0) -> str | None:
+        """One watch request from resourceVersion `rv`; streams
+        {type, object} JSON lines and feeds the event bus. Returns the
+        last seen resourceVersion, or None on 410 Gone / error (caller
+        must relist)."""
+        import json as _json
+
+        prefix, plural = API_MAP[kind]
+        url = (f"{prefix}/{plural}?watch=1&allowWatchBookmarks=true"
+               f"&timeoutSeconds={int(timeout_s)}&resourceVersion={rv}")
+        try:
+            with self._client.stream("GET", url, timeout=timeout_s + 15) as r:
+                if r.status_code != 200:
+                    return None
+                for line in r.iter_lines():
+                    if self._stop.is_set():
+                        return rv
+                    if not line.strip():
+                        continue
+                    try:
+                        ev = _json.loads(line)
+                    except ValueError:
+                        return None
+                    etype = ev.get("type", "")
+                    obj = ev.get("object", {})
+                    orv = str(obj.get("metadata", {})
+                              .get("resourceVersion", rv))
+                    if etype == "BOOKMARK":
+                        rv = orv
+                        continue
+                    if etype == "ERROR":
+                        return None  # usually 410 Gone: relist
+                    m = obj.get("metadata", {})
+                    key = (kind, m.get("namespace", "default"),
+                           m.get("name", ""))
+                    rv = orv
+                    if etype == "DELETED":
+                        self._known.pop(key, None)
+                        self._notify("DELETED", obj)
+                    elif etype in ("ADDED", "MODIFIED"):
+                        self._known[key] = orv
+                        self._notify(etype, _from_dict(kind, obj))
+        except Exception:
+            return None
+        return rv
+
+    def _watch_loop(self, kind: str) -> None:
+        while not self._stop.is_set():
+            try:
+                rv = self._seed_kind(kind)
+            except Exception:
+                self._stop.wait(2.0)
+                continue
+            while not self._stop.is_set():
+                rv2 = self._watch_kind_once(kind, rv)
+                if rv2 is None:
+                    break  # relist
+                rv = rv2
+
+    def run_watch(self) -> list[threading.Thread]:
+        """Start one watch thread per watched kind (the controller-runtime
+        informer equivalent — no polling between events). Threads are
+        daemons; stop() ends them at their next event/timeout."""
+        threads = []
+        for kind in WATCHED_KINDS:
+            t = threading.Thread(target=self._watch_loop, args=(kind,),
+                                 daemon=True, name=f"watch-{kind}")
+            t.start()
+            threads.append(t)
+        return threads
 
     def stop(self) -> None:
         self._stop.set()

@@ -51,9 +51,25 @@ class Operator:
         if kind == "Pod" and name.startswith("arks-worker-"):
             self._queue.put(("ArksModel", ns, name[len("arks-worker-"):]))
         if kind in ("LeaderWorkerSet", "RoleBasedGroupSet", "Deployment"):
-            base = name.rsplit("-", 1)[0] if kind != "RoleBasedGroupSet" else name
-            self._queue.put(("ArksApplication", ns, name))
-            self._queue.put(("ArksDisaggregatedApplication", ns, base))
+            # route by ownerReferences (reference field-index wiring,
+            # arksapplication_controller.go:1091-1134) — name-splitting
+            # mis-routes hyphenated app names
+            owners = (
+                m.get("ownerReferences", []) if isinstance(m, dict) else []
+            )
+            routed = False
+            for ref in owners:
+                okind = ref.get("kind")
+                if okind in ("ArksApplication", "ArksDisaggregatedApplication"):
+                    self._queue.put((okind, ns, ref.get("name", "")))
+                    routed = True
+            if not routed:
+                # adopted/legacy objects without owner refs: fall back to
+                # name-prefix routing
+                base = (name.rsplit("-", 1)[0]
+                        if kind != "RoleBasedGroupSet" else name)
+                self._queue.put(("ArksApplication", ns, name))
+                self._queue.put(("ArksDisaggregatedApplication", ns, base))
         if kind in ("ArksApplication", "ArksDisaggregatedApplication"):
             # endpoint watches app readiness (reference filterApp :119-168)
             sname = (

@@ -16,10 +16,19 @@ class FakeKube:
     def __init__(self):
         self.objects: dict[str, dict] = {}
         self.rv = 0
+        # watch support: (rv, plural, event-dict) log
+        self.events: list[tuple[int, str, dict]] = []
+
+    def _record(self, etype: str, plural: str, obj: dict) -> None:
+        self.events.append((self.rv, plural, {"type": etype, "object": obj}))
 
     def handler(self, request: httpx.Request) -> httpx.Response:
         path = request.url.path
         method = request.method
+        query = dict(
+            kv.split("=", 1) for kv in str(request.url.query, "utf-8").split("&")
+            if "=" in kv
+        ) if request.url.query else {}
         status_sub = path.endswith("/status")
         if status_sub:
             path = path[: -len("/status")]
@@ -31,11 +40,21 @@ class FakeKube:
             name = parts[i + 3] if len(parts) > i + 3 else None
         else:
             ns, plural, name = None, parts[-1], None
+        if method == "GET" and name is None and query.get("watch") == "1":
+            # one-shot watch: emit events after resourceVersion, then end
+            # the stream (the client's watch loop re-issues)
+            since = int(query.get("resourceVersion") or 0)
+            evs = [e for (erv, epl, e) in self.events
+                   if epl == plural and erv > since]
+            body = "\n".join(json.dumps(e) for e in evs)
+            return httpx.Response(200, text=body)
         if method == "GET" and name is None:
             items = [o for k, o in self.objects.items()
                      if k.startswith(f"{plural}/") and
                      (ns is None or o["metadata"]["namespace"] == ns)]
-            return httpx.Response(200, json={"items": items})
+            return httpx.Response(
+                200, json={"items": items,
+                           "metadata": {"resourceVersion": str(self.rv)}})
         key = f"{plural}/{ns}/{name}"
         if method == "GET":
             if key not in self.objects:
@@ -51,6 +70,7 @@ class FakeKube:
             body["metadata"]["resourceVersion"] = str(self.rv)
             body["metadata"].setdefault("namespace", ns)
             self.objects[key] = body
+            self._record("ADDED", plural, body)
             return httpx.Response(201, json=body)
         if method == "PUT":
             if key not in self.objects:
@@ -63,13 +83,21 @@ class FakeKube:
                 cur["status"] = body.get("status")
                 cur["metadata"]["resourceVersion"] = str(self.rv)
                 self.objects[key] = cur
+                self._record("MODIFIED", plural, cur)
                 return httpx.Response(200, json=cur)
             body.setdefault("status", self.objects[key].get("status"))
             self.objects[key] = body
+            self._record("MODIFIED", plural, body)
             return httpx.Response(200, json=body)
         if method == "DELETE":
-            if self.objects.pop(key, None) is None:
+            gone = self.objects.pop(key, None)
+            if gone is None:
                 return httpx.Response(404, json={})
+            self.rv += 1
+            gone = dict(gone)
+            gone["metadata"] = {**gone["metadata"],
+                                "resourceVersion": str(self.rv)}
+            self._record("DELETED", plural, gone)
             return httpx.Response(200, json={})
         return httpx.Response(405)
 
@@ -139,3 +167,53 @@ def test_resync_events():
     store.delete("ArksModel", "d", "m1")
     store.resync_once()
     assert any(e == "DELETED" for e, _ in events)
+
+
+def test_watch_stream_delivers_events_without_polling():
+    """The watch path (informer equivalent): after the initial seed list,
+    mutations arrive as ADDED/MODIFIED/DELETED watch events — no list+diff
+    polling involved."""
+    fake, store = mk_store()
+    events = []
+    store.subscribe(lambda e, o: events.append((e, getattr(
+        o, "metadata", None) and o.metadata.name or o["metadata"]["name"])))
+
+    m = ArksModel(
+        metadata=ObjectMeta(name="m1", namespace="default"),
+        spec=ArksModelSpec(model="org/m"),
+    )
+    store.create(m)
+    # seed: list emits ADDED, returns the collection resourceVersion
+    rv = store._seed_kind("ArksModel")
+    assert ("ADDED", "m1") in events
+    events.clear()
+
+    # mutate: the next watch call must deliver MODIFIED without any list
+    got = store.get("ArksModel", "default", "m1")
+    got.status.phase = ModelPhase.READY
+    store.update(got)
+    rv2 = store._watch_kind_once("ArksModel", rv, timeout_s=1)
+    assert rv2 is not None and int(rv2) > int(rv)
+    assert ("MODIFIED", "m1") in events
+    events.clear()
+
+    store.delete("ArksModel", "default", "m1")
+    rv3 = store._watch_kind_once("ArksModel", rv2, timeout_s=1)
+    assert ("DELETED", "m1") in events
+    assert rv3 is not None
+
+
+def test_watch_seed_emits_deletes_for_vanished_objects():
+    fake, store = mk_store()
+    events = []
+    store.subscribe(lambda e, o: events.append(e))
+    m = ArksModel(
+        metadata=ObjectMeta(name="mgone", namespace="default"),
+        spec=ArksModelSpec(model="org/m"),
+    )
+    store.create(m)
+    store._seed_kind("ArksModel")
+    # delete behind the store's back (another client), reseed
+    fake.objects.clear()
+    store._seed_kind("ArksModel")
+    assert "DELETED" in events
