@@ -294,8 +294,8 @@ class ModelRunner:
     # ---------------- hipGraph decode ----------------
     def _init_graph_buffers(self) -> None:
         # round max_num_seqs up to the nearest capture size (capped at 256;
-        # MoE decode is only capture-safe on its dense path, <= 64 tokens —
-        # larger decode batches fall back to the eager sparse route)
+        # MoE decode uses the dense bmm path, capture-safe through the full
+        # 256-size graph ladder)
         cap = self.GRAPH_SIZES[-1]
         if self.model_cfg.num_local_experts > 0:
             from ..models.llama_family import MoEMLP

@@ -207,12 +207,14 @@ class MoEMLP(nn.Module):
                 requires_grad=False,
             )
 
-    # Below this many tokens the dense path wins: every expert's weights are
-    # streamed from HBM regardless (tokens scatter over all experts), so
-    # computing all tokens per expert costs nothing extra in bandwidth and
-    # removes the data-dependent nonzero/index_add — making decode steps
-    # hipGraph-capturable.
-    DENSE_TOKENS = 64
+    # Below this many tokens the dense path wins: every expert's weights
+    # are streamed from HBM regardless (tokens scatter over all experts),
+    # so decode is weight-bandwidth-bound and the extra MFMA work of
+    # computing all tokens per expert stays at or under the HBM floor up
+    # to ~256 tokens (E*T*3*I*H*2 flops vs E*3*I*H*2 bytes: the crossover
+    # at ~1.5 PF effective and ~6 TB/s is T ~ 250). Data-independent
+    # shapes make every decode batch size hipGraph-capturable.
+    DENSE_TOKENS = 256
 
     def forward(self, x):
         from ..parallel.comm import tp_all_reduce
