@@ -269,42 +269,46 @@ class MoEMLP(nn.Module):
         El = self.local_experts
         base = self.expert_base
         cap = max(counts_h[base:base + El] or [0])
-        # grouped only when the padded transients stay modest (the [El, cap,
-        # 2I] bmm buffer reaches GBs on wide-expert models at full prefill)
-        grouped_bytes = El * cap * 2 * self.inter * 2
-        if (cap > 0 and El * cap <= 4 * flat_sel.numel()
-                and grouped_bytes <= 512 * 1024 * 1024):
-            # Grouped: pad each local expert's segment to `cap` rows and run
-            # TWO strided-batch GEMMs for the whole block (384 segment GEMMs
-            # per Qwen3-MoE layer otherwise). Padding rows index row 0 with
-            # weight 0, so they are compute-only noise.
+        # Grouped dispatch with CLAMPED capacity: pad each local expert's
+        # segment to cap_eff rows and run TWO strided-batch GEMMs for the
+        # whole block (vs 2-4 launches PER EXPERT in a segment loop — 600+
+        # kernels per Qwen3-MoE layer, ~0.3 PF effective at the tiny
+        # per-segment M). The clamp bounds the padded transient to ~512 MB;
+        # experts overflowing cap_eff (heavy routing imbalance) finish in
+        # the per-expert loop below with large-M GEMMs, so the launch count
+        # stays O(#overflowing experts), not O(E).
+        cap_lim = max(1, (512 * 1024 * 1024) // (El * 2 * self.inter * 2))
+        cap_eff = min(cap, cap_lim)
+        if cap_eff > 0:
             local_counts = counts[base:base + El, None]        # [El, 1]
             local_starts = (offs[base:base + El] -
                             counts[base:base + El])[:, None]   # [El, 1]
-            ar = torch.arange(cap, device=x.device)[None, :]   # [1, cap]
-            valid = ar < local_counts
+            capped = torch.clamp(local_counts, max=cap_eff)
+            ar = torch.arange(cap_eff, device=x.device)[None, :]  # [1, cap]
+            valid = ar < capped
             idx = torch.where(valid, local_starts + ar,
                               torch.zeros_like(ar))
             flat = idx.reshape(-1)
-            wpad = torch.where(valid, w_sorted[flat].view(El, cap),
+            wpad = torch.where(valid, w_sorted[flat].view(El, cap_eff),
                                torch.zeros(1, dtype=w_sorted.dtype,
                                            device=x.device))
-            tpad = torch.where(valid, tok_sorted[flat].view(El, cap),
+            tpad = torch.where(valid, tok_sorted[flat].view(El, cap_eff),
                                torch.zeros(1, dtype=torch.long,
                                            device=x.device))
-            xp = x_g[flat].view(El, cap, self.hidden)
+            xp = x_g[flat].view(El, cap_eff, self.hidden)
             gu = torch.bmm(xp, self.w13)
-            h = ops.silu_mul(gu.reshape(El * cap, 2 * self.inter))
-            y = torch.bmm(h.view(El, cap, self.inter), self.w2)  # [El, cap, H]
+            h = ops.silu_mul(gu.reshape(El * cap_eff, 2 * self.inter))
+            y = torch.bmm(h.view(El, cap_eff, self.inter), self.w2)
             y = y * wpad[..., None].to(y.dtype)
-            out.index_add_(0, tpad.reshape(-1), y.reshape(El * cap, -1))
-        else:
+            out.index_add_(0, tpad.reshape(-1), y.reshape(El * cap_eff, -1))
+        if cap > cap_eff:
             for le in range(El):
                 ge = base + le
                 c = counts_h[ge]
-                if c == 0:
+                if c <= cap_eff:
                     continue
-                seg = slice(offs_h[ge] - c, offs_h[ge])
+                s0 = offs_h[ge] - c + cap_eff
+                seg = slice(s0, offs_h[ge])
                 h = ops.silu_mul(x_g[seg] @ self.w13[le])
                 y = h @ self.w2[le]
                 out.index_add_(
