@@ -221,28 +221,22 @@ class MoEMLP(nn.Module):
 
         T = x.shape[0]
         router_logits = torch.nn.functional.linear(x, self.gate).float()
-        probs = torch.softmax(router_logits, dim=-1)
-        weights, selected = probs.topk(self.top_k, dim=-1)  # [T, k]
-        if self.norm_topk:
-            weights = weights / weights.sum(dim=-1, keepdim=True)
+        # fused softmax+topk+renorm routing kernel (ops.moe_topk) replaces
+        # a 5-kernel torch chain per layer
+        weights, selected = ops.moe_topk(router_logits, self.top_k,
+                                         self.norm_topk)
         out = torch.zeros_like(x)
         if T <= self.DENSE_TOKENS:
             # Batched over experts: two strided-batch GEMMs (bmm) instead of
             # a per-expert loop — one launch pair regardless of E (Qwen3-MoE
-            # has 128 experts), and every expert's weights stream once.
-            wdense = torch.zeros(
-                T, self.num_experts, dtype=torch.float32, device=x.device
-            )
-            wdense.scatter_(1, selected, weights)
-            wdense = wdense.to(x.dtype)
+            # has 128 experts), and every expert's weights stream once. The
+            # weighted mix over experts is one gather kernel (ops.moe_mix).
             E = self.local_experts
             xb = x.unsqueeze(0).expand(E, T, self.hidden)
             gu = torch.bmm(xb, self.w13)  # [E, T, 2I]
             h = ops.silu_mul(gu.reshape(E * T, 2 * self.inter))
             y = torch.bmm(h.view(E, T, self.inter), self.w2)  # [E, T, H]
-            wl = wdense[:, self.expert_base:self.expert_base + E]  # [T, E]
-            out = torch.einsum("eth,te->th", y.float(),
-                               wl.float()).to(x.dtype)
+            out = ops.moe_mix(y, weights, selected, self.expert_base)
             out = tp_all_reduce(out)
             if self.shared is not None:
                 gate = torch.sigmoid(
@@ -254,7 +248,7 @@ class MoEMLP(nn.Module):
         # each expert sees a contiguous segment — one host sync for the
         # segment table instead of a .nonzero() sync per expert.
         k = self.top_k
-        flat_sel = selected.reshape(-1)
+        flat_sel = selected.reshape(-1).long()
         flat_tok = _arange_interleave(T, k, x.device)
         flat_w = weights.reshape(-1)
         order = torch.argsort(flat_sel, stable=True)

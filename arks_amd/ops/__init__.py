@@ -350,6 +350,33 @@ def quant_fp8_rows(x):
     return ref.quant_fp8_rows(x)
 
 
+def moe_topk(router_logits, k: int, renorm: bool):
+    """softmax + top-k + optional renorm in one kernel:
+    (weights [T,k] f32, ids [T,k] i32). torch.topk tie-breaking (lower
+    expert id wins)."""
+    if router_logits.is_cuda:
+        T = router_logits.shape[0]
+        weights = torch.empty(T, k, dtype=torch.float32,
+                              device=router_logits.device)
+        ids = torch.empty(T, k, dtype=torch.int32,
+                          device=router_logits.device)
+        _native().moe_topk(weights, ids, router_logits.contiguous(), k,
+                           1 if renorm else 0)
+        return weights, ids
+    return ref.moe_topk(router_logits, k, renorm)
+
+
+def moe_mix(y, weights, ids, expert_base: int = 0):
+    """out[t] = sum_i weights[t,i] * y[ids[t,i]-expert_base, t] with
+    out-of-slice experts contributing 0 (TP expert parallelism)."""
+    if y.is_cuda:
+        El, T, H = y.shape
+        out = torch.empty(T, H, dtype=y.dtype, device=y.device)
+        _native().moe_mix(out, y.contiguous(), weights, ids, expert_base, El)
+        return out
+    return ref.moe_mix(y, weights, ids, expert_base)
+
+
 def sample_tokens(logits, temperatures, uniform):
     """Gumbel-max categorical sampling; rows with temperature 0 are greedy."""
     if logits.is_cuda:

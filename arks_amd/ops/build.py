@@ -29,6 +29,7 @@ SOURCES = [
     os.path.join(CSRC, "quant_fp8.hip"),
     os.path.join(CSRC, "skinny_gemm.hip"),
     os.path.join(CSRC, "sampling.hip"),
+    os.path.join(CSRC, "moe.hip"),
 ]
 
 

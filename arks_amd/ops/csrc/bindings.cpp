@@ -73,6 +73,11 @@ void arks_greedy_sample(void* out, const void* logits, int rows, int vocab,
 void arks_gumbel_sample(void* out, const void* logits, const void* temperatures,
                         const void* uniform, int rows, int vocab,
                         hipStream_t stream);
+void arks_moe_topk(void* weights, void* ids, const void* logits, int T,
+                   int E, int k, int renorm, hipStream_t stream);
+void arks_moe_mix(void* out, const void* y, const void* weights,
+                  const void* ids, int T, int H, int k, int expert_base,
+                  int n_local, hipStream_t stream);
 void arks_mfma_probe(void* d, const void* a, const void* b, hipStream_t stream);
 void arks_mfma_probe32(void* d, const void* a, const void* b, hipStream_t stream);
 void arks_tr16_probe(void* out, int stride_bytes, hipStream_t stream);
@@ -410,6 +415,33 @@ void mfma_probe32(torch::Tensor d, torch::Tensor a, torch::Tensor b) {
   arks_mfma_probe32(d.data_ptr(), a.data_ptr(), b.data_ptr(), current_stream());
 }
 
+void moe_topk(torch::Tensor weights, torch::Tensor ids,
+              torch::Tensor logits, int64_t k, int64_t renorm) {
+  TORCH_CHECK(logits.is_cuda() && logits.is_contiguous() &&
+              logits.scalar_type() == torch::kFloat32, "logits f32 contig");
+  TORCH_CHECK(weights.scalar_type() == torch::kFloat32 &&
+              ids.scalar_type() == torch::kInt32);
+  const int T = logits.size(0), E = logits.size(1);
+  TORCH_CHECK(k >= 1 && k <= 16 && E <= 1024, "k<=16, E<=1024");
+  arks_moe_topk(weights.data_ptr(), ids.data_ptr(), logits.data_ptr(), T, E,
+                (int)k, (int)renorm, current_stream());
+}
+
+void moe_mix(torch::Tensor out, torch::Tensor y, torch::Tensor weights,
+             torch::Tensor ids, int64_t expert_base, int64_t n_local) {
+  check_bf16_contig(out, "out");
+  check_bf16_contig(y, "y");
+  TORCH_CHECK(weights.scalar_type() == torch::kFloat32 &&
+              ids.scalar_type() == torch::kInt32);
+  const int T = out.size(0), H = out.size(1);
+  const int k = weights.size(1);
+  TORCH_CHECK(k <= 16 && H % 2 == 0);
+  TORCH_CHECK(y.size(1) == T && y.size(2) == H && y.size(0) == n_local);
+  arks_moe_mix(out.data_ptr(), y.data_ptr(), weights.data_ptr(),
+               ids.data_ptr(), T, H, k, (int)expert_base, (int)n_local,
+               current_stream());
+}
+
 void mfma_probe(torch::Tensor d, torch::Tensor a, torch::Tensor b) {
   check_bf16_contig(a, "a");
   check_bf16_contig(b, "b");
@@ -437,6 +469,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("skinny_gemm", &skinny_gemm);
   m.def("greedy_sample", &greedy_sample);
   m.def("gumbel_sample", &gumbel_sample);
+  m.def("moe_topk", &moe_topk);
+  m.def("moe_mix", &moe_mix);
   m.def("mfma_probe", &mfma_probe);
   m.def("mfma_probe32", &mfma_probe32);
   m.def("tr16_probe", &tr16_probe);

@@ -266,6 +266,28 @@ def attention_extend_paged(
     return out
 
 
+def moe_topk(router_logits: torch.Tensor, k: int, renorm: bool):
+    """softmax + top-k (+ renorm): (weights [T,k] f32, ids [T,k] i32)."""
+    probs = torch.softmax(router_logits.float(), dim=-1)
+    weights, selected = probs.topk(k, dim=-1)
+    if renorm:
+        weights = weights / weights.sum(dim=-1, keepdim=True)
+    return weights, selected.to(torch.int32)
+
+
+def moe_mix(y: torch.Tensor, weights: torch.Tensor, ids: torch.Tensor,
+            expert_base: int = 0) -> torch.Tensor:
+    """Weighted mix of per-(local-)expert dense outputs y [El, T, H]."""
+    El, T, H = y.shape
+    le = ids.long() - expert_base
+    mine = (le >= 0) & (le < El)
+    le = le.clamp(0, El - 1)
+    t_idx = torch.arange(T, device=y.device)[:, None].expand_as(le)
+    gathered = y[le, t_idx].float()  # [T, k, H]
+    w = torch.where(mine, weights, torch.zeros_like(weights))
+    return torch.einsum("tkh,tk->th", gathered, w.float()).to(y.dtype)
+
+
 def quant_fp8_rows(x: torch.Tensor) -> tuple[torch.Tensor, torch.Tensor]:
     """Per-row symmetric quantization to OCP e4m3 (range +-448)."""
     amax = x.float().abs().amax(dim=1).clamp(min=1e-6)
