@@ -30,6 +30,7 @@ SOURCES = [
     os.path.join(CSRC, "skinny_gemm.hip"),
     os.path.join(CSRC, "sampling.hip"),
     os.path.join(CSRC, "moe.hip"),
+    os.path.join(CSRC, "allreduce.hip"),
 ]
 
 

@@ -5,7 +5,11 @@
 #include <c10/hip/HIPStream.h>
 #include <hip/hip_runtime.h>
 
+#include <cstring>
+#include <string>
 #include <vector>
+
+namespace py = pybind11;
 
 // --- kernel launchers (see the .hip files) ---
 extern "C" {
@@ -73,6 +77,10 @@ void arks_greedy_sample(void* out, const void* logits, int rows, int vocab,
 void arks_gumbel_sample(void* out, const void* logits, const void* temperatures,
                         const void* uniform, int rows, int vocab,
                         hipStream_t stream);
+void arks_ar_seq_inc(void* seq, hipStream_t stream);
+void arks_one_shot_allreduce(void* out, const void* src, void* mail_ptrs[8],
+                             void* flag_ptrs[8], void* seq, int rank,
+                             int world, int64_t n, hipStream_t stream);
 void arks_moe_topk(void* weights, void* ids, const void* logits, int T,
                    int E, int k, int renorm, hipStream_t stream);
 void arks_moe_mix(void* out, const void* y, const void* weights,
@@ -415,6 +423,58 @@ void mfma_probe32(torch::Tensor d, torch::Tensor a, torch::Tensor b) {
   arks_mfma_probe32(d.data_ptr(), a.data_ptr(), b.data_ptr(), current_stream());
 }
 
+// --- p2p one-shot all-reduce plumbing (allreduce.hip) ---
+py::bytes ipc_handle(torch::Tensor t) {
+  TORCH_CHECK(t.is_cuda(), "ipc_handle wants a CUDA tensor");
+  hipIpcMemHandle_t h;
+  auto err = hipIpcGetMemHandle(&h, t.data_ptr());
+  TORCH_CHECK(err == hipSuccess, "hipIpcGetMemHandle: ",
+              hipGetErrorString(err));
+  return py::bytes(reinterpret_cast<const char*>(&h), sizeof(h));
+}
+
+int64_t ipc_open(py::bytes handle) {
+  std::string s = handle;
+  TORCH_CHECK(s.size() == sizeof(hipIpcMemHandle_t), "bad handle size");
+  hipIpcMemHandle_t h;
+  memcpy(&h, s.data(), sizeof(h));
+  void* ptr = nullptr;
+  auto err = hipIpcOpenMemHandle(&ptr, h, hipIpcMemLazyEnablePeerAccess);
+  TORCH_CHECK(err == hipSuccess, "hipIpcOpenMemHandle: ",
+              hipGetErrorString(err));
+  return reinterpret_cast<int64_t>(ptr);
+}
+
+void ipc_close(int64_t ptr) {
+  (void)hipIpcCloseMemHandle(reinterpret_cast<void*>(ptr));
+}
+
+void one_shot_allreduce(torch::Tensor out, torch::Tensor src,
+                        std::vector<int64_t> mail,
+                        std::vector<int64_t> flags, torch::Tensor seq,
+                        int64_t rank) {
+  check_bf16_contig(out, "out");
+  check_bf16_contig(src, "src");
+  TORCH_CHECK(seq.is_cuda() && seq.scalar_type() == torch::kInt64);
+  const int world = (int)mail.size();
+  TORCH_CHECK(world == (int)flags.size());
+  TORCH_CHECK(world == 1 || world == 2 || world == 4 || world == 8,
+              "world must be 1/2/4/8");
+  const int64_t n = src.numel();
+  TORCH_CHECK(n % 8 == 0, "numel must be a multiple of 8");
+  TORCH_CHECK(out.numel() == n);
+  void* mp[8] = {};
+  void* fp[8] = {};
+  for (int p = 0; p < world; ++p) {
+    mp[p] = reinterpret_cast<void*>(mail[p]);
+    fp[p] = reinterpret_cast<void*>(flags[p]);
+  }
+  auto stream = current_stream();
+  arks_ar_seq_inc(seq.data_ptr(), stream);
+  arks_one_shot_allreduce(out.data_ptr(), src.data_ptr(), mp, fp,
+                          seq.data_ptr(), (int)rank, world, n, stream);
+}
+
 void moe_topk(torch::Tensor weights, torch::Tensor ids,
               torch::Tensor logits, int64_t k, int64_t renorm) {
   TORCH_CHECK(logits.is_cuda() && logits.is_contiguous() &&
@@ -469,6 +529,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("skinny_gemm", &skinny_gemm);
   m.def("greedy_sample", &greedy_sample);
   m.def("gumbel_sample", &gumbel_sample);
+  m.def("ipc_handle", &ipc_handle);
+  m.def("ipc_open", &ipc_open);
+  m.def("ipc_close", &ipc_close);
+  m.def("one_shot_allreduce", &one_shot_allreduce);
   m.def("moe_topk", &moe_topk);
   m.def("moe_mix", &moe_mix);
   m.def("mfma_probe", &mfma_probe);

@@ -74,9 +74,39 @@ def get_tp_world_size() -> int:
     return _TP_WORLD
 
 
+_P2P_AR = None  # lazily built OneShotAllReduce (or False after a failure)
+
+
+def _p2p_ar():
+    """Lazy one-shot p2p all-reduce setup (first CUDA all-reduce call,
+    which happens during eager warmup — before any hipGraph capture)."""
+    global _P2P_AR
+    if _P2P_AR is None:
+        try:
+            if dist.get_backend(_TP_GROUP) == "nccl":
+                from .p2p import OneShotAllReduce
+
+                _P2P_AR = OneShotAllReduce(
+                    _TP_GROUP, _TP_RANK, _TP_WORLD,
+                    torch.device("cuda", torch.cuda.current_device()),
+                )
+            else:
+                _P2P_AR = False
+        except Exception:
+            _P2P_AR = False
+    return _P2P_AR if _P2P_AR else None
+
+
 def tp_all_reduce(x: torch.Tensor) -> torch.Tensor:
-    """In-place sum all-reduce across the TP group (no-op at TP=1)."""
+    """Sum all-reduce across the TP group (no-op at TP=1). Small bf16 CUDA
+    tensors take the one-shot p2p xGMI kernel (parallel/p2p.py) when its
+    init-time self-check passed; everything else goes through RCCL."""
     if _TP_WORLD > 1:
+        if x.is_cuda:
+            ar = _p2p_ar()
+            if ar is not None and ar.usable(x):
+                # returns a NEW tensor (all callers consume the return)
+                return ar(x)
         dist.all_reduce(x, op=dist.ReduceOp.SUM, group=_TP_GROUP)
     return x
 

@@ -583,3 +583,45 @@ def test_moe_mix_matches_einsum():
     expect = ref.moe_mix(y.float().cpu(), w.cpu(), ids.cpu(), base)
     torch.testing.assert_close(out.float().cpu(), expect.float(),
                                atol=2e-2, rtol=2e-2)
+
+
+def test_one_shot_allreduce_single_device():
+    """p2p all-reduce data path on one GPU: world=1 is identity; a
+    simulated world=2 (two mailbox sets on one device, flags pre-armed so
+    the sequential launches never spin) reduces both contributions."""
+    assert_native()
+    nat = ops._native()
+    torch.manual_seed(3)
+    n = 8192
+    x0 = torch.randn(n, dtype=torch.bfloat16, device=DEV)
+    x1 = torch.randn(n, dtype=torch.bfloat16, device=DEV)
+
+    # world = 1: identity
+    mail = torch.zeros(n, dtype=torch.bfloat16, device=DEV)
+    flags = torch.zeros(64, dtype=torch.int64, device=DEV)
+    seq = torch.zeros(1, dtype=torch.int64, device=DEV)
+    out = torch.empty_like(x0)
+    nat.one_shot_allreduce(out, x0, [mail.data_ptr()], [flags.data_ptr()],
+                           seq, 0)
+    torch.cuda.synchronize()
+    torch.testing.assert_close(out, x0)
+
+    # world = 2 simulated on one device
+    m = [torch.zeros(2 * n, dtype=torch.bfloat16, device=DEV)
+         for _ in range(2)]
+    f = [torch.full((2 * 64,), 100, dtype=torch.int64, device=DEV)
+         for _ in range(2)]
+    seqs = [torch.zeros(1, dtype=torch.int64, device=DEV) for _ in range(2)]
+    mails = [t.data_ptr() for t in m]
+    flgs = [t.data_ptr() for t in f]
+    out0 = torch.empty_like(x0)
+    out1 = torch.empty_like(x0)
+    nat.one_shot_allreduce(out0, x0, mails, flgs, seqs[0], 0)
+    nat.one_shot_allreduce(out1, x1, mails, flgs, seqs[1], 1)
+    torch.cuda.synchronize()
+    # the SECOND launch sees both mailbox slots populated
+    torch.testing.assert_close(out1.float(), (x0.float() + x1.float()),
+                               atol=2e-2, rtol=2e-2)
+    # ipc handle round trip on this device's own buffer
+    h = nat.ipc_handle(m[0])
+    assert len(bytes(h)) == 64
