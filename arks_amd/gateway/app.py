@@ -69,14 +69,45 @@ class BackendResolver:
             lambda ns, svc: f"http://{svc}.{ns}.svc:8080"
         )
 
-    def resolve(self, namespace: str, model: str) -> str | None:
+    @staticmethod
+    def _match_entry(m: dict, path: str, headers: dict[str, str]) -> bool:
+        """One HTTPRouteMatch: path AND all header conditions must hold
+        (gateway-api semantics)."""
+        pm = m.get("path") or {}
+        if pm:
+            val = pm.get("value", "/")
+            typ = pm.get("type", "PathPrefix")
+            if typ == "Exact":
+                if path != val:
+                    return False
+            else:  # PathPrefix
+                if not path.startswith(val):
+                    return False
+        for hm in m.get("headers", []) or []:
+            if headers.get(hm.get("name", "").lower()) != hm.get("value"):
+                return False
+        return True
+
+    def resolve(self, namespace: str, model: str, path: str = "/",
+                headers: dict[str, str] | None = None) -> str | None:
+        """Pick a backend from the endpoint's HTTPRoute, evaluating EVERY
+        rule's match conditions (a rule matches when any of its match
+        entries does — user MatchConfigs from the ArksEndpoint spec are
+        honored, reference arksendpoint_controller.go:349-369); first
+        matching rule with backendRefs wins."""
         route = self.store.get_opt("HTTPRoute", namespace, model)
         if route is None:
             return None
-        rules = route.get("spec", {}).get("rules", [])
-        if not rules:
-            return None
-        refs = rules[0].get("backendRefs", [])
+        hdrs = {k.lower(): v for k, v in (headers or {}).items()}
+        hdrs.setdefault("namespace", namespace)
+        hdrs.setdefault("model", model)
+        refs = None
+        for rule in route.get("spec", {}).get("rules", []):
+            matches = rule.get("matches") or [{}]
+            if any(self._match_entry(m, path, hdrs) for m in matches):
+                refs = rule.get("backendRefs", [])
+                if refs:
+                    break
         if not refs:
             return None
         weights = [max(int(r.get("weight", 1)), 0) for r in refs]
@@ -173,7 +204,11 @@ def create_gateway_app(
             [d for d in descriptors if RULES[d.rule].type == TYPE_REQUEST], 1
         )
 
-        base = resolver.resolve(qos.namespace, model)
+        base = resolver.resolve(
+            qos.namespace, model, path=request.url.path,
+            headers={"model": model, "namespace": qos.namespace,
+                     "username": qos.user},
+        )
         if base is None:
             metrics.requests_total.labels(**labels, status="503").inc()
             return _err(503, "no ready backend for model")

@@ -274,3 +274,34 @@ def test_quota_sync_loop_updates_cr_status(stack):
         assert gw.state.quota_service.get_usage("default", "q1", "response") == 3
 
     run_stack(stack, fn)
+
+
+def test_backend_resolver_multi_rule_matching():
+    """Multi-rule HTTPRoutes: rules are evaluated with their path/header
+    match conditions (user MatchConfigs), not just rules[0]."""
+    from arks_amd.gateway import BackendResolver
+
+    store = Store()
+    store.apply({
+        "apiVersion": "gateway.networking.k8s.io/v1",
+        "kind": "HTTPRoute",
+        "metadata": {"name": "m", "namespace": "default"},
+        "spec": {"rules": [
+            # static rule for a specific path prefix
+            {"matches": [{"path": {"type": "PathPrefix",
+                                   "value": "/v1/special"}}],
+             "backendRefs": [{"name": "svc-special", "weight": 1}]},
+            # dynamic rule matched on the injected routing headers
+            {"matches": [{"path": {"type": "PathPrefix", "value": "/"},
+                          "headers": [{"name": "namespace",
+                                       "value": "default"},
+                                      {"name": "model", "value": "m"}]}],
+             "backendRefs": [{"name": "svc-app", "weight": 1}]},
+        ]},
+    })
+    r = BackendResolver(store, url_for_service=lambda ns, svc: svc)
+    assert r.resolve("default", "m", path="/v1/special/x") == "svc-special"
+    assert r.resolve("default", "m", path="/v1/chat/completions") == "svc-app"
+    # header mismatch on the dynamic rule -> no backend
+    assert r.resolve("default", "m", path="/v1/chat/completions",
+                     headers={"model": "other"}) is None
