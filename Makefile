@@ -1,7 +1,7 @@
 # Developer entry points (the reference's Makefile equivalents).
 PY ?= python3
 
-.PHONY: build test test-gpu bench crds installer docker
+.PHONY: build test test-gpu test-e2e bench crds installer docker
 
 build:  ## compile the gfx950 HIP extension in-tree
 	$(PY) -m arks_amd.ops.build
@@ -11,6 +11,10 @@ test:   ## CPU suite (engine, scheduler, server, gateway, control plane)
 
 test-gpu:  ## kernel numerics + engine e2e (needs an MI355X)
 	$(PY) -m pytest tests -q -m gpu
+
+test-e2e:  ## full-stack e2e: local sockets always; kind cluster when available
+	$(PY) -m pytest tests/test_e2e_local.py -q
+	@command -v kind >/dev/null 2>&1 && bash scripts/e2e_kind.sh || echo "kind not installed: skipped the real-cluster leg"
 
 bench:  ## the driver-contract benchmark (one GPU)
 	$(PY) bench.py --steps 16 --warmup 4
