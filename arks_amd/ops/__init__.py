@@ -126,18 +126,25 @@ EXTEND2_MIN_QLEN = 97
 
 def build_extend_tiles(q_lens: list[int], kv_lens: list[int] | None,
                        kv_fp8: bool, device):
-    """Per-seq routing between the two extend kernels: (tiles64, tiles256).
+    """Per-seq routing for the extend dispatcher:
+    (tiles64, tiles256, decode_rows).
 
-    Tiles are sorted by DESCENDING key depth: causal q-tiles differ up to
-    4x in KV work, and the dispatcher issues blocks in order, so launching
-    deep tiles first removes the straggler tail (PMC: unsorted launch left
-    the chip ~half idle at serving shapes). Build once per batch and reuse
-    across all layers."""
-    t64, t256 = [], []
+    q_len == 1 sequences (decode steps mixed into a prefill batch) go to
+    the flash-decode kernel — an extend q-tile for one query scans the
+    full KV with 255 masked rows (~6x slower than the decode kernel's
+    partitioned scan). Remaining sequences split between the 64-row and
+    256-row extend kernels; tiles are sorted by DESCENDING key depth:
+    causal q-tiles differ up to 4x in KV work and the dispatcher issues
+    blocks in order, so deep tiles first removes the straggler tail (PMC:
+    unsorted launch left the chip ~half idle at serving shapes). Build
+    once per batch and reuse across all layers."""
+    t64, t256, dec = [], [], []
     kvl = kv_lens if kv_lens is not None else q_lens
     for i, n in enumerate(q_lens):
         off = kvl[i] - n
-        if kv_fp8 or n < EXTEND2_MIN_QLEN:
+        if n == 1:
+            dec.append(i)
+        elif kv_fp8 or n < EXTEND2_MIN_QLEN:
             for q0 in range(0, n, PREFILL_QTILE):
                 t64.append((i, q0, min(kvl[i], off + q0 + PREFILL_QTILE)))
         else:
@@ -148,7 +155,18 @@ def build_extend_tiles(q_lens: list[int], kv_lens: list[int] | None,
     mk = lambda t: torch.tensor(
         [x[:2] for x in t], dtype=torch.int32, device=device
     ).reshape(-1, 2)
-    return mk(t64), mk(t256)
+    if dec:
+        # gather indices: the single q row of each decode seq sits at
+        # cu_seqlens_q[i]; seq-level metadata indexes with `dec` itself
+        cu = [0]
+        for n in q_lens:
+            cu.append(cu[-1] + n)
+        drows = torch.tensor([cu[i] for i in dec], dtype=torch.long,
+                             device=device)
+        dseqs = torch.tensor(dec, dtype=torch.long, device=device)
+    else:
+        drows = dseqs = None
+    return mk(t64), mk(t256), (drows, dseqs)
 
 
 def attention_extend_paged(
@@ -170,8 +188,9 @@ def attention_extend_paged(
             (q.shape[0], q.shape[1], q.shape[2]), dtype=q.dtype, device=q.device
         )
         kv_fp8 = k_cache.dtype != torch.bfloat16
-        t64, t256 = tiles if tiles is not None else build_extend_tiles(
-            q_lens, None, kv_fp8, q.device
+        t64, t256, (drows, dseqs) = (
+            tiles if tiles is not None
+            else build_extend_tiles(q_lens, None, kv_fp8, q.device)
         )
         if t256.numel():
             _native().attention_extend_paged2(
@@ -183,6 +202,14 @@ def attention_extend_paged(
                 out, q, k_cache, v_cache, block_tables, kv_lens, cu_seqlens_q,
                 t64, scale, window,
             )
+        if drows is not None:
+            dq = q[drows].contiguous()
+            dout = attention_decode_paged(
+                dq, k_cache, v_cache,
+                block_tables[dseqs].contiguous(),
+                kv_lens[dseqs].contiguous(), scale, window=window,
+            )
+            out[drows] = dout
         return out
     return ref.attention_extend_paged(
         q, k_cache, v_cache, block_tables, kv_lens, cu_seqlens_q, scale,
