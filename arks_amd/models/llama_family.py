@@ -245,15 +245,18 @@ class MoEMLP(nn.Module):
                 out = out + gate * self.shared(x)
             return out
         # Sparse path (prefill-sized T): sort token-expert pairs once so
-        # each expert sees a contiguous segment — one host sync for the
-        # segment table instead of a .nonzero() sync per expert.
+        # each expert sees a contiguous segment (one host sync per layer
+        # for the segment table), run the expert FFNs into ONE flat output
+        # buffer — a capacity-clamped grouped bmm plus large-M overflow
+        # GEMMs for heavily-routed experts — and combine with the
+        # moe_mix_rows gather kernel. No index_add_ scatters anywhere:
+        # torch's indexFuncLargeIndex was 36% of Qwen3-30B prefill GPU
+        # time (profiles/r02_final.md).
         k = self.top_k
         flat_sel = selected.reshape(-1).long()
         flat_tok = _arange_interleave(T, k, x.device)
-        flat_w = weights.reshape(-1)
         order = torch.argsort(flat_sel, stable=True)
         tok_sorted = flat_tok[order]
-        w_sorted = flat_w[order]
         counts = torch.bincount(flat_sel, minlength=self.num_experts)
         offs = torch.cumsum(counts, 0)
         seg = torch.stack((counts, offs)).cpu()  # ONE host sync per layer
@@ -262,52 +265,75 @@ class MoEMLP(nn.Module):
         x_g = x[tok_sorted]  # [T*k, H] gathered once
         El = self.local_experts
         base = self.expert_base
-        cap = max(counts_h[base:base + El] or [0])
-        # Grouped dispatch with CLAMPED capacity: pad each local expert's
-        # segment to cap_eff rows and run TWO strided-batch GEMMs for the
-        # whole block (vs 2-4 launches PER EXPERT in a segment loop — 600+
-        # kernels per Qwen3-MoE layer, ~0.3 PF effective at the tiny
-        # per-segment M). The clamp bounds the padded transient to ~512 MB;
-        # experts overflowing cap_eff (heavy routing imbalance) finish in
-        # the per-expert loop below with large-M GEMMs, so the launch count
-        # stays O(#overflowing experts), not O(E).
+        local_counts_h = counts_h[base:base + El]
+        cap = max(local_counts_h or [0])
         cap_lim = max(1, (512 * 1024 * 1024) // (El * 2 * self.inter * 2))
         cap_eff = min(cap, cap_lim)
-        if cap_eff > 0:
-            local_counts = counts[base:base + El, None]        # [El, 1]
-            local_starts = (offs[base:base + El] -
-                            counts[base:base + El])[:, None]   # [El, 1]
-            capped = torch.clamp(local_counts, max=cap_eff)
-            ar = torch.arange(cap_eff, device=x.device)[None, :]  # [1, cap]
-            valid = ar < capped
-            idx = torch.where(valid, local_starts + ar,
-                              torch.zeros_like(ar))
-            flat = idx.reshape(-1)
-            wpad = torch.where(valid, w_sorted[flat].view(El, cap_eff),
-                               torch.zeros(1, dtype=w_sorted.dtype,
-                                           device=x.device))
-            tpad = torch.where(valid, tok_sorted[flat].view(El, cap_eff),
-                               torch.zeros(1, dtype=torch.long,
-                                           device=x.device))
-            xp = x_g[flat].view(El, cap_eff, self.hidden)
-            gu = torch.bmm(xp, self.w13)
-            h = ops.silu_mul(gu.reshape(El * cap_eff, 2 * self.inter))
-            y = torch.bmm(h.view(El, cap_eff, self.inter), self.w2)
-            y = y * wpad[..., None].to(y.dtype)
-            out.index_add_(0, tpad.reshape(-1), y.reshape(El * cap_eff, -1))
-        if cap > cap_eff:
-            for le in range(El):
-                ge = base + le
-                c = counts_h[ge]
-                if c <= cap_eff:
-                    continue
-                s0 = offs_h[ge] - c + cap_eff
-                seg = slice(s0, offs_h[ge])
-                h = ops.silu_mul(x_g[seg] @ self.w13[le])
-                y = h @ self.w2[le]
-                out.index_add_(
-                    0, tok_sorted[seg], y * w_sorted[seg, None].to(y.dtype)
-                )
+        ovf = [(le, c - cap_eff) for le, c in enumerate(local_counts_h)
+               if c > cap_eff]
+        n_of = sum(o for _, o in ovf)
+        if cap_eff == 0:
+            out = tp_all_reduce(out)  # no pairs routed to this rank
+            if self.shared is not None:
+                gate = torch.sigmoid(
+                    torch.nn.functional.linear(x, self.shared_gate).float()
+                ).to(x.dtype)
+                out = out + gate * self.shared(x)
+            return out
+
+        y_all = x.new_empty(El * cap_eff + n_of, self.hidden)
+        # grouped part: pad each local expert's segment to cap_eff rows
+        # (two strided-batch GEMMs for the whole block; padding rows index
+        # row 0 and are never gathered)
+        local_counts = counts[base:base + El, None]        # [El, 1]
+        local_starts = (offs[base:base + El] -
+                        counts[base:base + El])[:, None]   # [El, 1]
+        capped = torch.clamp(local_counts, max=cap_eff)
+        ar = torch.arange(cap_eff, device=x.device)[None, :]  # [1, cap]
+        flat = torch.where(ar < capped, local_starts + ar,
+                           torch.zeros_like(ar)).reshape(-1)
+        xp = x_g[flat].view(El, cap_eff, self.hidden)
+        gu = torch.bmm(xp, self.w13)
+        h = ops.silu_mul(gu.reshape(El * cap_eff, 2 * self.inter))
+        torch.bmm(h.view(El, cap_eff, self.inter), self.w2,
+                  out=y_all[:El * cap_eff].view(El, cap_eff, self.hidden))
+        # overflow part: experts past cap_eff finish with plain large-M
+        # GEMMs appended to the same buffer
+        p0 = El * cap_eff
+        of_base_h = []
+        acc = 0
+        for le, o in ovf:
+            of_base_h.append((le, acc))
+            ge = base + le
+            sl = slice(offs_h[ge] - counts_h[ge] + cap_eff, offs_h[ge])
+            h2 = ops.silu_mul(x_g[sl] @ self.w13[le])
+            torch.mm(h2, self.w2[le], out=y_all[p0 + acc:p0 + acc + o])
+            acc += o
+
+        # row map [T, k]: each pair's position in y_all
+        inv = torch.empty_like(order)
+        inv[order] = torch.arange(T * k, device=x.device)
+        inv = inv.view(T, k)
+        sel = selected.long().view(T, k)
+        sel_local = sel - base
+        mine = (sel_local >= 0) & (sel_local < El)
+        seg_start = offs - counts  # [E] global segment starts
+        local_idx = inv - seg_start[sel]
+        in_cap = local_idx < cap_eff
+        sl_safe = sel_local.clamp(0, El - 1)
+        if ovf:
+            of_base_t = torch.zeros(El, dtype=torch.long, device=x.device)
+            of_base_t[torch.tensor([le for le, _ in of_base_h],
+                                   device=x.device)] = torch.tensor(
+                [a for _, a in of_base_h], device=x.device)
+            rows_of = p0 + of_base_t[sl_safe] + (local_idx - cap_eff)
+        else:
+            rows_of = torch.zeros_like(local_idx)
+        rows = torch.where(in_cap, sl_safe * cap_eff + local_idx, rows_of)
+        rows = torch.where(mine, rows, torch.zeros_like(rows))
+        w_masked = torch.where(mine, weights,
+                               torch.zeros_like(weights))
+        out = ops.moe_mix_rows(y_all, w_masked, rows.to(torch.int32))
         out = tp_all_reduce(out)
         if self.shared is not None:
             gate = torch.sigmoid(

@@ -404,6 +404,18 @@ def moe_mix(y, weights, ids, expert_base: int = 0):
     return ref.moe_mix(y, weights, ids, expert_base)
 
 
+def moe_mix_rows(y, weights, rows):
+    """out[t] = sum_j weights[t,j] * y[rows[t,j]] — the gather-side mix
+    over a flat per-pair expert-output buffer (zero-weight slots point at
+    row 0)."""
+    if y.is_cuda:
+        T, k = weights.shape
+        out = torch.empty(T, y.shape[1], dtype=y.dtype, device=y.device)
+        _native().moe_mix_rows(out, y.contiguous(), weights, rows)
+        return out
+    return ref.moe_mix_rows(y, weights, rows)
+
+
 def sample_tokens(logits, temperatures, uniform):
     """Gumbel-max categorical sampling; rows with temperature 0 are greedy."""
     if logits.is_cuda:

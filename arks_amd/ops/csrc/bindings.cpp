@@ -86,6 +86,9 @@ void arks_moe_topk(void* weights, void* ids, const void* logits, int T,
 void arks_moe_mix(void* out, const void* y, const void* weights,
                   const void* ids, int T, int H, int k, int expert_base,
                   int n_local, hipStream_t stream);
+void arks_moe_mix_rows(void* out, const void* y, const void* weights,
+                       const void* rows, int T, int H, int k,
+                       hipStream_t stream);
 void arks_mfma_probe(void* d, const void* a, const void* b, hipStream_t stream);
 void arks_mfma_probe32(void* d, const void* a, const void* b, hipStream_t stream);
 void arks_tr16_probe(void* out, int stride_bytes, hipStream_t stream);
@@ -502,6 +505,20 @@ void moe_mix(torch::Tensor out, torch::Tensor y, torch::Tensor weights,
                current_stream());
 }
 
+void moe_mix_rows(torch::Tensor out, torch::Tensor y,
+                  torch::Tensor weights, torch::Tensor rows) {
+  check_bf16_contig(out, "out");
+  check_bf16_contig(y, "y");
+  TORCH_CHECK(weights.scalar_type() == torch::kFloat32 &&
+              rows.scalar_type() == torch::kInt32);
+  const int T = out.size(0), H = out.size(1);
+  const int k = weights.size(1);
+  TORCH_CHECK(k <= 16 && H % 2 == 0);
+  TORCH_CHECK(rows.size(0) == T && rows.size(1) == k);
+  arks_moe_mix_rows(out.data_ptr(), y.data_ptr(), weights.data_ptr(),
+                    rows.data_ptr(), T, H, k, current_stream());
+}
+
 void mfma_probe(torch::Tensor d, torch::Tensor a, torch::Tensor b) {
   check_bf16_contig(a, "a");
   check_bf16_contig(b, "b");
@@ -535,6 +552,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("one_shot_allreduce", &one_shot_allreduce);
   m.def("moe_topk", &moe_topk);
   m.def("moe_mix", &moe_mix);
+  m.def("moe_mix_rows", &moe_mix_rows);
   m.def("mfma_probe", &mfma_probe);
   m.def("mfma_probe32", &mfma_probe32);
   m.def("tr16_probe", &tr16_probe);

@@ -121,6 +121,39 @@ __global__ __launch_bounds__(256) void moe_mix_kernel(
   }
 }
 
+// Row-indexed variant for the sparse/grouped prefill path: y is a flat
+// [R, H] buffer (padded grouped-bmm block + overflow segments); rows[t,k]
+// addresses each token's k expert outputs directly. Replaces torch
+// index_add_ scatters (36% of Qwen3-30B prefill GPU time) with one
+// bandwidth-bound gather; zero-weight slots point at row 0.
+__global__ __launch_bounds__(256) void moe_mix_rows_kernel(
+    bf16* __restrict__ out,          // [T, H]
+    const bf16* __restrict__ y,      // [R, H]
+    const float* __restrict__ weights,  // [T, k]
+    const int* __restrict__ rows,       // [T, k]
+    const int T, const int H, const int k) {
+  const int t = blockIdx.x;
+  __shared__ float w[16];
+  __shared__ int64_t base[16];
+  if (threadIdx.x < k) {
+    const int j = threadIdx.x;
+    w[j] = weights[(int64_t)t * k + j];
+    base[j] = (int64_t)rows[(int64_t)t * k + j] * H;
+  }
+  __syncthreads();
+  for (int h = threadIdx.x * 2; h < H; h += 256 * 2) {
+    float a0 = 0.f, a1 = 0.f;
+    for (int j = 0; j < k; ++j) {
+      const uint32_t u = *reinterpret_cast<const uint32_t*>(y + base[j] + h);
+      a0 += w[j] * bf16_bits_to_float((uint16_t)u);
+      a1 += w[j] * bf16_bits_to_float((uint16_t)(u >> 16));
+    }
+    uint32_t packed = (uint32_t)float_to_bf16_bits(a0) |
+                      ((uint32_t)float_to_bf16_bits(a1) << 16);
+    *reinterpret_cast<uint32_t*>(out + (int64_t)t * H + h) = packed;
+  }
+}
+
 }  // namespace arks
 
 using namespace arks;
@@ -132,6 +165,15 @@ extern "C" void arks_moe_topk(void* weights, void* ids, const void* logits,
   hipLaunchKernelGGL((moe_topk_kernel<16>), grid, block, 0, stream,
                      (float*)weights, (int*)ids, (const float*)logits, E, k,
                      renorm);
+}
+
+extern "C" void arks_moe_mix_rows(void* out, const void* y,
+                                  const void* weights, const void* rows,
+                                  int T, int H, int k, hipStream_t stream) {
+  dim3 grid(T), block(256);
+  hipLaunchKernelGGL(moe_mix_rows_kernel, grid, block, 0, stream, (bf16*)out,
+                     (const bf16*)y, (const float*)weights, (const int*)rows,
+                     T, H, k);
 }
 
 extern "C" void arks_moe_mix(void* out, const void* y, const void* weights,

@@ -288,6 +288,14 @@ def moe_mix(y: torch.Tensor, weights: torch.Tensor, ids: torch.Tensor,
     return torch.einsum("tkh,tk->th", gathered, w.float()).to(y.dtype)
 
 
+def moe_mix_rows(y: torch.Tensor, weights: torch.Tensor,
+                 rows: torch.Tensor) -> torch.Tensor:
+    """out[t] = sum_j weights[t,j] * y[rows[t,j]] over a flat [R, H]
+    expert-output buffer."""
+    gathered = y[rows.long()].float()  # [T, k, H]
+    return torch.einsum("tkh,tk->th", gathered, weights.float()).to(y.dtype)
+
+
 def quant_fp8_rows(x: torch.Tensor) -> tuple[torch.Tensor, torch.Tensor]:
     """Per-row symmetric quantization to OCP e4m3 (range +-448)."""
     amax = x.float().abs().amax(dim=1).clamp(min=1e-6)
