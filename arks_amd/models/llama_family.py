@@ -268,7 +268,12 @@ class MoEMLP(nn.Module):
         local_counts_h = counts_h[base:base + El]
         cap = max(local_counts_h or [0])
         cap_lim = max(1, (512 * 1024 * 1024) // (El * 2 * self.inter * 2))
-        cap_eff = min(cap, cap_lim)
+        # clamp the grouped capacity near 2x the mean load: under skewed
+        # routing, padding the whole block to the heaviest expert wastes
+        # multiples of the useful FLOPs, while spilled experts finish in
+        # efficient large-M overflow GEMMs below
+        mean2 = 2 * ((sum(local_counts_h) + El - 1) // El) if El else 0
+        cap_eff = min(cap, cap_lim, max(128, mean2))
         ovf = [(le, c - cap_eff) for le, c in enumerate(local_counts_h)
                if c > cap_eff]
         n_of = sum(o for _, o in ovf)
