@@ -21,3 +21,4 @@ def bundle(out, paths):
 bundle("dist/operator.yaml",
        sorted(glob.glob("deploy/crds/*.yaml")) + ["deploy/operator.yaml"])
 bundle("dist/gateway.yaml", ["deploy/gateway.yaml"])
+bundle("dist/gateway-envoy.yaml", ["deploy/gateway-envoy.yaml"])
