@@ -94,6 +94,10 @@ class ModelRunner:
             self.model.quantize_fp8()
         elif self.cfg.quantization is not None:
             raise ValueError(f"unknown quantization {self.cfg.quantization!r}")
+        # serving past max_position_embeddings: the rope table must cover
+        # every reachable position (OOB table reads fault on GPU)
+        if hasattr(self.model, "extend_rope_table"):
+            self.model.extend_rope_table(self.cfg.max_model_len)
         self._weights_loaded = True
 
     def profile_num_blocks(self) -> int:

@@ -411,13 +411,27 @@ class LlamaFamilyForCausalLM(nn.Module):
         quantize_module_fp8(self.lm_head)
         self.norm.fp8_out = True
 
-    def _build_cos_sin(self):
+    def _build_cos_sin(self, min_len: int | None = None):
         from ..ops import ref
 
         return ref.rope_cos_sin_cache(
-            self.cfg.head_dim, self.cfg.max_position_embeddings,
+            self.cfg.head_dim,
+            max(self.cfg.max_position_embeddings, min_len or 0),
             self.cfg.rope_theta, rope_scaling=self.cfg.rope_scaling,
         )
+
+    def extend_rope_table(self, min_len: int) -> None:
+        """Grow the RoPE cos/sin table to at least `min_len` positions
+        (serving past max_position_embeddings would otherwise read out of
+        bounds in the rope kernel; plain RoPE extrapolates)."""
+        if self.rope_cos_sin.shape[0] >= min_len:
+            return
+        cache = self._build_cos_sin(min_len).to(
+            device=self.rope_cos_sin.device, dtype=self.rope_cos_sin.dtype
+        )
+        self.register_buffer("rope_cos_sin", cache, persistent=False)
+        for layer in self.layers:
+            layer.self_attn._cos_sin = cache
 
     def _apply(self, fn, recurse=True):  # keep the shared cos_sin in sync
         out = super()._apply(fn, recurse)
