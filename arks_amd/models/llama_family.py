@@ -468,6 +468,17 @@ class LlamaFamilyForCausalLM(nn.Module):
             assert p.shape == value.shape, (name, p.shape, value.shape)
             p.data.copy_(value.to(p.dtype))
 
+        def put_cat(name, parts):
+            # fused params load piecewise: each (pinned) slice goes straight
+            # into its span of the parameter — no host-side torch.cat copy
+            p = own[name]
+            off = 0
+            for part in parts:
+                n = part.shape[0]
+                p.data[off:off + n].copy_(part.to(p.dtype))
+                off += n
+            assert off == p.shape[0], (name, off, p.shape)
+
         for name, w in tensors.items():
             name = name.removeprefix("model.")
             if name == "embed_tokens.weight":
@@ -496,17 +507,17 @@ class LlamaFamilyForCausalLM(nn.Module):
                     keys_w = {"q_proj.weight", "k_proj.weight", "v_proj.weight"}
                     keys_b = {"q_proj.bias", "k_proj.bias", "v_proj.bias"}
                     if keys_w <= d.keys():
-                        fused = layer.self_attn.qkv_proj.shard_qkv(
+                        parts = layer.self_attn.qkv_proj.shard_qkv_parts(
                             d["q_proj.weight"], d["k_proj.weight"], d["v_proj.weight"]
                         )
-                        put(f"layers.{li}.self_attn.qkv_proj.weight", fused)
+                        put_cat(f"layers.{li}.self_attn.qkv_proj.weight", parts)
                         for kk in keys_w:
                             del d[kk]
                     if layer.self_attn.qkv_proj.bias is not None and keys_b <= d.keys():
-                        fused = layer.self_attn.qkv_proj.shard_qkv(
+                        parts = layer.self_attn.qkv_proj.shard_qkv_parts(
                             d["q_proj.bias"], d["k_proj.bias"], d["v_proj.bias"]
                         )
-                        put(f"layers.{li}.self_attn.qkv_proj.bias", fused)
+                        put_cat(f"layers.{li}.self_attn.qkv_proj.bias", parts)
                         for kk in keys_b:
                             del d[kk]
                 elif sub == "self_attn.o_proj.weight":
@@ -518,10 +529,10 @@ class LlamaFamilyForCausalLM(nn.Module):
                     pending_mlp.setdefault(li, {})[sub] = w
                     d = pending_mlp[li]
                     if {"mlp.gate_proj.weight", "mlp.up_proj.weight"} <= d.keys():
-                        fused = layer.mlp.gate_up_proj.shard_merged(
+                        parts = layer.mlp.gate_up_proj.shard_merged_parts(
                             d["mlp.gate_proj.weight"], d["mlp.up_proj.weight"]
                         )
-                        put(f"layers.{li}.mlp.gate_up_proj.weight", fused)
+                        put_cat(f"layers.{li}.mlp.gate_up_proj.weight", parts)
                         d.clear()
                 elif sub == "mlp.down_proj.weight":
                     put(f"layers.{li}.mlp.down_proj.weight", layer.mlp.down_proj.shard(w))

@@ -120,14 +120,20 @@ class QKVParallelLinear(ColumnParallelLinear):
         out = (self.nq_local + 2 * self.nkv_local) * head_dim * tp  # per-rank x tp
         super().__init__(hidden, out, bias, dtype)
 
-    def shard_qkv(self, q_full, k_full, v_full) -> torch.Tensor:
-        """Build this rank's fused weight (or bias) from full q/k/v tensors."""
+    def shard_qkv_parts(self, q_full, k_full, v_full) -> list[torch.Tensor]:
+        """This rank's q/k/v slices in fused order (views — callers copy
+        each part straight into the fused parameter, no host-side cat)."""
         r = get_tp_rank()
         hd = self.head_dim
-        qs = q_full[r * self.nq_local * hd : (r + 1) * self.nq_local * hd]
-        ks = k_full[r * self.nkv_local * hd : (r + 1) * self.nkv_local * hd]
-        vs = v_full[r * self.nkv_local * hd : (r + 1) * self.nkv_local * hd]
-        return torch.cat([qs, ks, vs], dim=0)
+        return [
+            q_full[r * self.nq_local * hd: (r + 1) * self.nq_local * hd],
+            k_full[r * self.nkv_local * hd: (r + 1) * self.nkv_local * hd],
+            v_full[r * self.nkv_local * hd: (r + 1) * self.nkv_local * hd],
+        ]
+
+    def shard_qkv(self, q_full, k_full, v_full) -> torch.Tensor:
+        """Build this rank's fused weight (or bias) from full q/k/v tensors."""
+        return torch.cat(self.shard_qkv_parts(q_full, k_full, v_full), dim=0)
 
     def split_qkv(self, qkv: torch.Tensor):
         hd = self.head_dim
@@ -145,13 +151,15 @@ class MergedColumnParallelLinear(ColumnParallelLinear):
         self.each_out = each_out
         super().__init__(in_features, 2 * each_out, bias, dtype)
 
-    def shard_merged(self, gate_full: torch.Tensor, up_full: torch.Tensor) -> torch.Tensor:
+    def shard_merged_parts(self, gate_full, up_full) -> list[torch.Tensor]:
         r = get_tp_rank()
         tp = get_tp_world_size()
         per = self.each_out // tp
-        return torch.cat(
-            [gate_full[r * per : (r + 1) * per], up_full[r * per : (r + 1) * per]], dim=0
-        )
+        return [gate_full[r * per: (r + 1) * per],
+                up_full[r * per: (r + 1) * per]]
+
+    def shard_merged(self, gate_full: torch.Tensor, up_full: torch.Tensor) -> torch.Tensor:
+        return torch.cat(self.shard_merged_parts(gate_full, up_full), dim=0)
 
 
 class RowParallelLinear(nn.Module):
