@@ -21,7 +21,7 @@ import time
 from typing import Callable
 
 import httpx
-from prometheus_client import CollectorRegistry, Counter, Histogram, generate_latest
+from prometheus_client import CollectorRegistry, Counter, Histogram, generate_latest, Gauge
 from starlette.applications import Starlette
 from starlette.requests import Request
 from starlette.responses import JSONResponse, PlainTextResponse, Response, StreamingResponse
@@ -55,6 +55,17 @@ class GatewayMetrics:
         )
         self.errors_total = Counter(
             "gateway_errors_total", "errors", ["kind"], registry=self.registry
+        )
+        # quota gauges (the reference left these as TODO stubs,
+        # collector.go:58-75; populated here by the provider's 10 s
+        # quota sync loop)
+        self.quota_usage = Gauge(
+            "gateway_quota_usage", "cumulative quota usage",
+            ["namespace", "quota", "type"], registry=self.registry,
+        )
+        self.quota_limit = Gauge(
+            "gateway_quota_limit", "configured quota limit",
+            ["namespace", "quota", "type"], registry=self.registry,
         )
 
 
@@ -143,9 +154,9 @@ def create_gateway_app(
 ) -> Starlette:
     limiter = limiter or RateLimiter()
     quota_service = quota_service or QuotaService()
-    provider = ConfigProvider(store, quota_service)
-    resolver = resolver or BackendResolver(store)
     metrics = GatewayMetrics(registry)
+    provider = ConfigProvider(store, quota_service, metrics=metrics)
+    resolver = resolver or BackendResolver(store)
     client = httpx.AsyncClient(transport=transport, timeout=300.0)
 
     def _auth(request: Request) -> str | None:

@@ -35,9 +35,11 @@ class UserQos:
 
 
 class ConfigProvider:
-    def __init__(self, store: Store, quota_service: QuotaService):
+    def __init__(self, store: Store, quota_service: QuotaService,
+                 metrics=None):
         self.store = store
         self.quota_service = quota_service
+        self.metrics = metrics  # GatewayMetrics (quota gauges)
         self._stop = threading.Event()
 
     # --- lookups (token indexed by spec.token — arks_impl.go:59-73) ---
@@ -91,6 +93,11 @@ class ConfigProvider:
             changed = False
             for q in quota.spec.quotas:
                 live = self.quota_service.get_usage(ns, name, q.type)
+                if self.metrics is not None:
+                    self.metrics.quota_usage.labels(
+                        namespace=ns, quota=name, type=q.type).set(live)
+                    self.metrics.quota_limit.labels(
+                        namespace=ns, quota=name, type=q.type).set(q.value)
                 entry = next(
                     (e for e in quota.status.quota_status if e.type == q.type), None
                 )

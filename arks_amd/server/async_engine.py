@@ -198,7 +198,8 @@ class AsyncEngine:
             num_prompt_tokens=seq.num_prompt_tokens, num_output_tokens=1,
         ))
         if seq.is_finished:
-            self.metrics.request_success.inc()
+            self.metrics.record_finished(
+                seq.finish_reason, seq.num_prompt_tokens, 1)
             st.queue.put_nowait(None)
             self.streams.pop(request_id, None)
         if self._wakeup:
@@ -266,7 +267,9 @@ class AsyncEngine:
                 self.metrics.generation_tokens.inc()
                 st.queue.put_nowait(out)
                 if out.finished:
-                    self.metrics.request_success.inc()
+                    self.metrics.record_finished(
+                        out.finish_reason, out.num_prompt_tokens,
+                        out.num_output_tokens)
                     st.queue.put_nowait(None)  # sentinel
                     self.streams.pop(out.request_id, None)
             sched = self.engine.scheduler

@@ -22,6 +22,8 @@ _TPOT_BUCKETS = (0.01, 0.025, 0.05, 0.075, 0.1, 0.15, 0.2, 0.3, 0.4, 0.5,
                  0.75, 1.0, 2.5)
 _E2E_BUCKETS = (0.3, 0.5, 0.8, 1.0, 1.5, 2.0, 2.5, 5.0, 10.0, 15.0, 20.0,
                 30.0, 40.0, 50.0, 60.0)
+_LEN_BUCKETS = (1, 2, 5, 10, 20, 50, 100, 200, 500, 1000, 2000, 5000,
+                10000, 20000, 50000)
 
 
 class EngineMetrics:
@@ -44,7 +46,23 @@ class EngineMetrics:
         self.generation_tokens = mk(
             Counter, "vllm:generation_tokens", "generated tokens"
         )
-        self.request_success = mk(Counter, "vllm:request_success", "finished requests")
+        self._model_name = model_name
+        self._success = Counter(
+            "vllm:request_success", "finished requests",
+            labelnames=["model_name", "finished_reason"],
+            registry=self.registry,
+        )
+        # back-compat handle (unlabeled reason) for existing callers
+        self.request_success = self._success.labels(
+            model_name=model_name, finished_reason="stop")
+        self.request_prompt_tokens = mk(
+            Histogram, "vllm:request_prompt_tokens",
+            "per-request prompt length", buckets=_LEN_BUCKETS,
+        )
+        self.request_generation_tokens = mk(
+            Histogram, "vllm:request_generation_tokens",
+            "per-request generation length", buckets=_LEN_BUCKETS,
+        )
         self.ttft = mk(
             Histogram, "vllm:time_to_first_token_seconds", "TTFT",
             buckets=_TTFT_BUCKETS,
@@ -74,6 +92,16 @@ class EngineMetrics:
             Counter, "vllm:spec_decode_num_accepted_tokens",
             "speculative draft tokens accepted",
         )
+
+    def record_finished(self, reason: str | None, prompt_tokens: int,
+                        output_tokens: int) -> None:
+        """Per-request finish accounting: success counter by finish
+        reason + prompt/generation length histograms (the reference
+        dashboard's heatmap + finish-reason panels read these)."""
+        self._success.labels(model_name=self._model_name,
+                             finished_reason=reason or "stop").inc()
+        self.request_prompt_tokens.observe(prompt_tokens)
+        self.request_generation_tokens.observe(output_tokens)
 
     def render(self) -> bytes:
         return generate_latest(self.registry)

@@ -305,3 +305,25 @@ def test_backend_resolver_multi_rule_matching():
     # header mismatch on the dynamic rule -> no backend
     assert r.resolve("default", "m", path="/v1/chat/completions",
                      headers={"model": "other"}) is None
+
+
+def test_quota_gauges_exported():
+    """gateway_quota_usage/limit gauges (TODO stubs in the reference,
+    collector.go:58-75) are populated by the quota sync loop."""
+    from prometheus_client import generate_latest
+
+    from arks_amd.gateway.app import create_gateway_app
+    from arks_amd.crd.types import parse_manifest
+
+    store = Store()
+    store.apply(parse_manifest({
+        "apiVersion": "arks.ai/v1", "kind": "ArksQuota",
+        "metadata": {"name": "qg", "namespace": "default"},
+        "spec": {"quotas": [{"type": "total", "value": 500}]},
+    }))
+    gw = create_gateway_app(store)
+    gw.state.quota_service.incr_usage("default", "qg", "total", 42)
+    gw.state.provider.sync_quota_usage()
+    text = generate_latest(gw.state.provider.metrics.registry).decode()
+    assert 'gateway_quota_usage{namespace="default",quota="qg",type="total"} 42.0' in text
+    assert 'gateway_quota_limit{namespace="default",quota="qg",type="total"} 500.0' in text
