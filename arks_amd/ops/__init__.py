@@ -302,12 +302,9 @@ def _skinny_ws(nsplits: int, N: int, M: int, device):
 _USE_SKINNY = os.environ.get("ARKS_SKINNY_GEMM", "1") == "1"
 # Weight-elements threshold: the split-K streaming kernel beats hipBLASLt's
 # ~19 us latency floor on the small decode shapes (qkv/o: 1.0-1.6x,
-# bench_skinny v3). Wide-N streaming shapes (gate_up/lm_head) stay on the
-# tuned hipBLASLt algos (5.4-5.9 TB/s); narrow-N big-K shapes (down_proj:
-# N=3584, K=18944) take skinny too — few column tiles underutilize the
-# library there (2.3 TB/s lib vs 2.6 skinny at M=64, bench_skinny_big).
+# bench_skinny v3); the big streaming shapes (gate_up/down/lm_head) stay on
+# the tuned hipBLASLt algos which reach 5.4-6.4 TB/s there.
 _SKINNY_MAX_ELEMS = 34_000_000
-_SKINNY_MAX_N = 4096  # narrow-N shapes run skinny regardless of K
 
 
 def linear_bf16(x, weight, bias=None):
@@ -321,8 +318,7 @@ def linear_bf16(x, weight, bias=None):
         and x.dtype == torch.bfloat16
         and weight.shape[0] % 64 == 0
         and weight.shape[1] % 32 == 0
-        and (weight.numel() <= _SKINNY_MAX_ELEMS
-             or weight.shape[0] <= _SKINNY_MAX_N)
+        and weight.numel() <= _SKINNY_MAX_ELEMS
         and weight.is_contiguous()
         and native_available()
     ):
