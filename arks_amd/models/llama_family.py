@@ -87,13 +87,13 @@ class Attention(nn.Module):
             k = self.k_norm(
                 k.contiguous().view(-1, hd)).view(T, self.nkv_local * hd)
             v = v.contiguous()  # reshape_and_cache wants one shared kv stride
-        q, k = ops.rope_apply_inplace(
-            batch.positions, q, k, self._cos_sin, self.head_dim
+        k_cache, v_cache = kv_cache
+        q, k = ops.rope_and_cache(
+            batch.positions, q, k, v, k_cache, v_cache, batch.slot_mapping,
+            self._cos_sin, self.head_dim,
         )
         k = k.unflatten(-1, (self.nkv_local, self.head_dim))
         v = v.unflatten(-1, (self.nkv_local, self.head_dim))
-        k_cache, v_cache = kv_cache
-        ops.reshape_and_cache(k, v, k_cache, v_cache, batch.slot_mapping)
         q = q.unflatten(-1, (self.nq_local, self.head_dim))
         if batch.is_prefill:
             if batch.block_tables is not None:

@@ -83,6 +83,25 @@ def rope_apply_inplace(positions, q, k, cos_sin, head_dim: int):
     return ref.rope_apply(positions, q, k, cos_sin, head_dim)
 
 
+def rope_and_cache(positions, q, k, v, k_cache, v_cache, slot_mapping,
+                   cos_sin, head_dim: int):
+    """Fused RoPE + paged-cache write (GPU): rotates q/k in place and
+    scatters rotated k + v into the cache in one launch. CPU path runs the
+    two reference steps."""
+    if q.is_cuda:
+        _native().rope_and_cache(positions, q, k, v, k_cache, v_cache,
+                                 slot_mapping, cos_sin, head_dim)
+        return q, k
+    q2, k2 = ref.rope_apply(positions, q, k, cos_sin, head_dim)
+    T = q.shape[0]
+    nkv = k.numel() // max(T, 1) // head_dim
+    ref.reshape_and_cache(
+        k2.reshape(T, nkv, head_dim), v.reshape(T, nkv, head_dim),
+        k_cache, v_cache, slot_mapping,
+    )
+    return q2, k2
+
+
 def reshape_and_cache(k, v, k_cache, v_cache, slot_mapping):
     if k.is_cuda:
         _native().reshape_and_cache(k, v, k_cache, v_cache, slot_mapping)

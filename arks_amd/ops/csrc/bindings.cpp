@@ -34,6 +34,12 @@ void arks_reshape_and_cache_fp8(const void* k, const void* v, void* k_cache,
                                 int num_tokens, int num_kv_heads, int head_dim,
                                 int block_size, int64_t kv_stride,
                                 hipStream_t stream);
+void arks_rope_and_cache(const void* positions, void* q, void* k,
+                         const void* v, void* k_cache, void* v_cache,
+                         const void* slot_mapping, const void* cos_sin,
+                         int num_tokens, int head_dim, int num_q_heads,
+                         int num_kv_heads, int block_size, int64_t q_stride,
+                         int64_t k_stride, int kv_fp8, hipStream_t stream);
 void arks_attn_decode_paged(void* out, void* part_out, const void* q,
                             const void* k_cache, const void* v_cache,
                             const void* block_tables, const void* seq_lens,
@@ -196,6 +202,31 @@ void reshape_and_cache(torch::Tensor k, torch::Tensor v, torch::Tensor k_cache,
      v_cache.data_ptr(), slot_mapping.data_ptr(),
      num_tokens, num_kv_heads, head_dim, block_size,
      k.stride(0), current_stream());
+}
+
+void rope_and_cache(torch::Tensor positions, torch::Tensor q,
+                    torch::Tensor k, torch::Tensor v, torch::Tensor k_cache,
+                    torch::Tensor v_cache, torch::Tensor slot_mapping,
+                    torch::Tensor cos_sin, int64_t head_dim) {
+  TORCH_CHECK(positions.scalar_type() == torch::kInt64);
+  TORCH_CHECK(slot_mapping.scalar_type() == torch::kInt64);
+  TORCH_CHECK(cos_sin.scalar_type() == torch::kFloat32 &&
+              cos_sin.is_contiguous());
+  check_bf16_rowstrided(q, "q");
+  check_bf16_rowstrided(k, "k");
+  check_bf16_rowstrided(v, "v");
+  const bool kv_fp8 = check_kv_cache(k_cache, "k_cache");
+  check_kv_cache(v_cache, "v_cache");
+  TORCH_CHECK(k.stride(0) == v.stride(0), "k/v must share row stride");
+  const int T = q.size(0);
+  const int nq = (int)(q.numel() / std::max<int64_t>(T, 1) / head_dim);
+  const int nkv = (int)(k.numel() / std::max<int64_t>(T, 1) / head_dim);
+  const int block_size = k_cache.size(2);
+  arks_rope_and_cache(positions.data_ptr(), q.data_ptr(), k.data_ptr(),
+                      v.data_ptr(), k_cache.data_ptr(), v_cache.data_ptr(),
+                      slot_mapping.data_ptr(), cos_sin.data_ptr(), T,
+                      (int)head_dim, nq, nkv, block_size, q.stride(0),
+                      k.stride(0), kv_fp8 ? 1 : 0, current_stream());
 }
 
 void attention_decode_paged(torch::Tensor out, torch::Tensor q,
@@ -536,6 +567,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("silu_mul", &silu_mul);
   m.def("rope_inplace", &rope_inplace);
   m.def("reshape_and_cache", &reshape_and_cache);
+  m.def("rope_and_cache", &rope_and_cache);
   m.def("attention_decode_paged", &attention_decode_paged);
   m.def("attention_prefill_varlen", &attention_prefill_varlen);
   m.def("attention_extend_paged", &attention_extend_paged);

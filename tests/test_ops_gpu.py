@@ -625,3 +625,37 @@ def test_one_shot_allreduce_single_device():
     # ipc handle round trip on this device's own buffer
     h = nat.ipc_handle(m[0])
     assert len(bytes(h)) == 64
+
+
+@pytest.mark.parametrize("fp8", [False, True])
+def test_rope_and_cache_fused_matches_two_step(fp8):
+    """Fused RoPE+cache kernel vs the separate rope + reshape_and_cache
+    pair, including padded rows (slot -1: rotate but skip the cache)."""
+    assert_native()
+    torch.manual_seed(21)
+    T, hq, hkv, hd, bs = 9, 8, 2, 128, 16
+    nblocks = 8
+    dt = torch.float8_e4m3fn if fp8 else torch.bfloat16
+    qkv = torch.randn(T, (hq + 2 * hkv) * hd, dtype=torch.bfloat16, device=DEV)
+    q = qkv[:, : hq * hd]
+    k = qkv[:, hq * hd: (hq + hkv) * hd]
+    v = qkv[:, (hq + hkv) * hd:]
+    q2, k2, v2 = q.clone().contiguous(), k.clone().contiguous(), v.clone().contiguous()
+    cos_sin = ref.rope_cos_sin_cache(hd, 64, 10000.0).cuda()
+    pos = torch.randint(0, 64, (T,), dtype=torch.int64, device=DEV)
+    slots = torch.tensor([i * 3 if i != 4 else -1 for i in range(T)],
+                         dtype=torch.int64, device=DEV)
+    kc_a = torch.zeros(nblocks, hkv, bs, hd, dtype=dt, device=DEV)
+    vc_a = torch.zeros_like(kc_a)
+    kc_b = torch.zeros_like(kc_a)
+    vc_b = torch.zeros_like(kc_a)
+    # fused
+    ops.rope_and_cache(pos, q, k, v, kc_a, vc_a, slots, cos_sin, hd)
+    # two-step
+    ops.rope_apply_inplace(pos, q2, k2, cos_sin, hd)
+    ops.reshape_and_cache(k2.view(T, hkv, hd), v2.view(T, hkv, hd),
+                          kc_b, vc_b, slots)
+    torch.testing.assert_close(q, q2, atol=2e-2, rtol=2e-2)
+    torch.testing.assert_close(k, k2, atol=2e-2, rtol=2e-2)
+    torch.testing.assert_close(kc_a.float(), kc_b.float(), atol=3e-2, rtol=3e-2)
+    torch.testing.assert_close(vc_a.float(), vc_b.float(), atol=3e-2, rtol=3e-2)
