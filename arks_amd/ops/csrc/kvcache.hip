@@ -151,17 +151,14 @@ __global__ void rope_cache_kernel(
     ushort2v x1 = *reinterpret_cast<const ushort2v*>(base + p2);
     ushort2v x2 = *reinterpret_cast<const ushort2v*>(base + half + p2);
     ushort2v o1, o2;
-    float f1[2], f2[2];
 #pragma unroll
     for (int j = 0; j < 2; ++j) {
       const float c = cs[p2 + j];
       const float s = cs[half + p2 + j];
       const float a = bf16_bits_to_float(x1[j]);
       const float b = bf16_bits_to_float(x2[j]);
-      f1[j] = a * c - b * s;
-      f2[j] = b * c + a * s;
-      o1[j] = float_to_bf16_bits(f1[j]);
-      o2[j] = float_to_bf16_bits(f2[j]);
+      o1[j] = float_to_bf16_bits(a * c - b * s);
+      o2[j] = float_to_bf16_bits(b * c + a * s);
     }
     *reinterpret_cast<ushort2v*>(base + p2) = o1;
     *reinterpret_cast<ushort2v*>(base + half + p2) = o2;
@@ -171,7 +168,10 @@ __global__ void rope_cache_kernel(
         uint8_t* kc = reinterpret_cast<uint8_t*>(k_cache);
 #pragma unroll
         for (int j = 0; j < 2; ++j) {
-          __hip_fp8_e4m3 a(f1[j]), b(f2[j]);
+          // convert from the bf16-ROUNDED values so the cache is
+          // bit-identical to the separate rope + reshape_and_cache path
+          __hip_fp8_e4m3 a(bf16_bits_to_float(o1[j]));
+          __hip_fp8_e4m3 b(bf16_bits_to_float(o2[j]));
           kc[dst + p2 + j] = a.__x;
           kc[dst + half + p2 + j] = b.__x;
         }
