@@ -458,6 +458,33 @@ void mfma_probe32(torch::Tensor d, torch::Tensor a, torch::Tensor b) {
 }
 
 // --- p2p one-shot all-reduce plumbing (allreduce.hip) ---
+// IPC-shared buffers are raw hipMalloc allocations: a handle for a torch
+// caching-allocator tensor maps the UNDERLYING block, so the opened peer
+// pointer would miss the tensor's offset inside it.
+py::tuple ipc_alloc(int64_t nbytes) {
+  void* ptr = nullptr;
+  auto err = hipMalloc(&ptr, (size_t)nbytes);
+  TORCH_CHECK(err == hipSuccess, "hipMalloc: ", hipGetErrorString(err));
+  (void)hipMemset(ptr, 0, (size_t)nbytes);
+  hipIpcMemHandle_t h;
+  err = hipIpcGetMemHandle(&h, ptr);
+  TORCH_CHECK(err == hipSuccess, "hipIpcGetMemHandle: ",
+              hipGetErrorString(err));
+  return py::make_tuple(
+      reinterpret_cast<int64_t>(ptr),
+      py::bytes(reinterpret_cast<const char*>(&h), sizeof(h)));
+}
+
+void ipc_alloc_free(int64_t ptr) {
+  (void)hipFree(reinterpret_cast<void*>(ptr));
+}
+
+// test hook: arm a raw flag buffer with huge values (0x7f per byte) so the
+// single-GPU sequential harness never spins
+void arm_flags(int64_t ptr, int64_t n) {
+  (void)hipMemset(reinterpret_cast<void*>(ptr), 0x7f, (size_t)n * 8);
+}
+
 py::bytes ipc_handle(torch::Tensor t) {
   TORCH_CHECK(t.is_cuda(), "ipc_handle wants a CUDA tensor");
   hipIpcMemHandle_t h;
@@ -486,7 +513,7 @@ void ipc_close(int64_t ptr) {
 void one_shot_allreduce(torch::Tensor out, torch::Tensor src,
                         std::vector<int64_t> mail,
                         std::vector<int64_t> flags, torch::Tensor seq,
-                        int64_t rank) {
+                        int64_t rank) {  // mail/flags: ipc_alloc'd raw ptrs
   check_bf16_contig(out, "out");
   check_bf16_contig(src, "src");
   TORCH_CHECK(seq.is_cuda() && seq.scalar_type() == torch::kInt64);
@@ -578,6 +605,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("skinny_gemm", &skinny_gemm);
   m.def("greedy_sample", &greedy_sample);
   m.def("gumbel_sample", &gumbel_sample);
+  m.def("ipc_alloc", &ipc_alloc);
+  m.def("arm_flags", &arm_flags);
+  m.def("ipc_alloc_free", &ipc_alloc_free);
   m.def("ipc_handle", &ipc_handle);
   m.def("ipc_open", &ipc_open);
   m.def("ipc_close", &ipc_close);

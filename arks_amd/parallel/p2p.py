@@ -49,20 +49,21 @@ class OneShotAllReduce:
             return
         try:
             nat = ops._native()
-            self.mail = torch.zeros(world * MAX_ELEMS, dtype=torch.bfloat16,
-                                    device=device)
-            self.flags = torch.zeros(world * AR_MAX_BLOCKS,
-                                     dtype=torch.int64, device=device)
+            # raw hipMalloc allocations: an IPC handle for a torch
+            # caching-allocator tensor maps the underlying block, not the
+            # tensor's offset within it
+            self._mail_ptr, mail_h = nat.ipc_alloc(world * MAX_ELEMS * 2)
+            self._flag_ptr, flag_h = nat.ipc_alloc(world * AR_MAX_BLOCKS * 8)
+            self._owned = [self._mail_ptr, self._flag_ptr]
             self.seq = torch.zeros(1, dtype=torch.int64, device=device)
-            my = (bytes(nat.ipc_handle(self.mail)),
-                  bytes(nat.ipc_handle(self.flags)))
+            my = (bytes(mail_h), bytes(flag_h))
             handles: list = [None] * world
             dist.all_gather_object(handles, my, group=group)
             self.mail_ptrs, self.flag_ptrs = [], []
             for p, (mh, fh) in enumerate(handles):
                 if p == rank:
-                    self.mail_ptrs.append(self.mail.data_ptr())
-                    self.flag_ptrs.append(self.flags.data_ptr())
+                    self.mail_ptrs.append(self._mail_ptr)
+                    self.flag_ptrs.append(self._flag_ptr)
                 else:
                     mp = nat.ipc_open(mh)
                     fp = nat.ipc_open(fh)
@@ -108,3 +109,9 @@ class OneShotAllReduce:
             except Exception:
                 pass
         self._opened = []
+        for p in getattr(self, "_owned", []):
+            try:
+                self._nat.ipc_alloc_free(p)
+            except Exception:
+                pass
+        self._owned = []

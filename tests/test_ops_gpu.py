@@ -596,35 +596,39 @@ def test_one_shot_allreduce_single_device():
     x0 = torch.randn(n, dtype=torch.bfloat16, device=DEV)
     x1 = torch.randn(n, dtype=torch.bfloat16, device=DEV)
 
-    # world = 1: identity
-    mail = torch.zeros(n, dtype=torch.bfloat16, device=DEV)
-    flags = torch.zeros(64, dtype=torch.int64, device=DEV)
+    # world = 1: identity (ipc_alloc'd raw buffers, as production uses)
+    mail_ptr, mail_h = nat.ipc_alloc(n * 2)
+    flag_ptr, _ = nat.ipc_alloc(64 * 8)
+    assert len(bytes(mail_h)) == 64
     seq = torch.zeros(1, dtype=torch.int64, device=DEV)
     out = torch.empty_like(x0)
-    nat.one_shot_allreduce(out, x0, [mail.data_ptr()], [flags.data_ptr()],
-                           seq, 0)
+    nat.one_shot_allreduce(out, x0, [mail_ptr], [flag_ptr], seq, 0)
     torch.cuda.synchronize()
     torch.testing.assert_close(out, x0)
+    nat.ipc_alloc_free(mail_ptr)
+    nat.ipc_alloc_free(flag_ptr)
 
-    # world = 2 simulated on one device
-    m = [torch.zeros(2 * n, dtype=torch.bfloat16, device=DEV)
-         for _ in range(2)]
-    f = [torch.full((2 * 64,), 100, dtype=torch.int64, device=DEV)
-         for _ in range(2)]
+    # world = 2 simulated on one device; flags pre-armed (0x7f bytes =
+    # huge sequence values) so the two sequential launches never spin
+    mails, flgs = [], []
+    for _ in range(2):
+        mp, _h = nat.ipc_alloc(2 * n * 2)
+        fp, _h2 = nat.ipc_alloc(2 * 64 * 8)
+        mails.append(mp)
+        flgs.append(fp)
     seqs = [torch.zeros(1, dtype=torch.int64, device=DEV) for _ in range(2)]
-    mails = [t.data_ptr() for t in m]
-    flgs = [t.data_ptr() for t in f]
     out0 = torch.empty_like(x0)
     out1 = torch.empty_like(x0)
+    nat.arm_flags(flgs[0], 2 * 64)
+    nat.arm_flags(flgs[1], 2 * 64)
     nat.one_shot_allreduce(out0, x0, mails, flgs, seqs[0], 0)
     nat.one_shot_allreduce(out1, x1, mails, flgs, seqs[1], 1)
     torch.cuda.synchronize()
     # the SECOND launch sees both mailbox slots populated
     torch.testing.assert_close(out1.float(), (x0.float() + x1.float()),
                                atol=2e-2, rtol=2e-2)
-    # ipc handle round trip on this device's own buffer
-    h = nat.ipc_handle(m[0])
-    assert len(bytes(h)) == 64
+    for p in mails + flgs:
+        nat.ipc_alloc_free(p)
 
 
 @pytest.mark.parametrize("fp8", [False, True])
