@@ -257,6 +257,9 @@ class ModelRunner:
                 ext_tiles = ops.build_extend_tiles(
                     q_lens, kv_lens,
                     self.cfg.kv_cache_dtype == "fp8", dev,
+                    num_q_heads=max(
+                        self.model_cfg.num_attention_heads
+                        // get_tp_world_size(), 1),
                 )
             return ForwardBatch(
                 is_prefill=True,

@@ -144,19 +144,22 @@ EXTEND2_MIN_QLEN = 97
 
 
 def build_extend_tiles(q_lens: list[int], kv_lens: list[int] | None,
-                       kv_fp8: bool, device):
+                       kv_fp8: bool, device, num_q_heads: int = 32):
     """Per-seq routing for the extend dispatcher:
-    (tiles64, tiles256, decode_rows).
+    (tiles64, tiles256 [n,4], decode_rows, combine_table, ws_rows).
 
     q_len == 1 sequences (decode steps mixed into a prefill batch) go to
     the flash-decode kernel — an extend q-tile for one query scans the
     full KV with 255 masked rows (~6x slower than the decode kernel's
     partitioned scan). Remaining sequences split between the 64-row and
-    256-row extend kernels; tiles are sorted by DESCENDING key depth:
-    causal q-tiles differ up to 4x in KV work and the dispatcher issues
-    blocks in order, so deep tiles first removes the straggler tail (PMC:
-    unsorted launch left the chip ~half idle at serving shapes). Build
-    once per batch and reuse across all layers."""
+    256-row extend kernels; tiles are sorted by DESCENDING key depth
+    (causal q-tiles differ up to 4x in KV work and the dispatcher issues
+    blocks in order — deep tiles first removes the straggler tail). When
+    the batch is too small to fill the chip, deep 256-row tiles are
+    additionally SPLIT over their KV range flash-decode-style (part rows
+    + a combine pass over f32 partials); split parts sort first so the
+    partial workspace only covers them. Build once per batch and reuse
+    across all layers."""
     t64, t256, dec = [], [], []
     kvl = kv_lens if kv_lens is not None else q_lens
     for i, n in enumerate(q_lens):
@@ -171,9 +174,42 @@ def build_extend_tiles(q_lens: list[int], kv_lens: list[int] | None,
                 t256.append((i, q0, min(kvl[i], off + q0 + EXTEND2_QTILE)))
     t64.sort(key=lambda t: -t[2])
     t256.sort(key=lambda t: -t[2])
-    mk = lambda t: torch.tensor(
+    mk64 = lambda t: torch.tensor(
         [x[:2] for x in t], dtype=torch.int32, device=device
     ).reshape(-1, 2)
+
+    # KV-split policy for the 256-row kernel: when tiles x heads cannot
+    # fill ~2 workgroups per CU, split the deepest tiles' key ranges.
+    base_wgs = len(t256) * max(num_q_heads, 1)
+    factor = 1
+    if t256 and base_wgs < 512:
+        factor = min(8, -(-512 // base_wgs) + 1)
+    rows, combine = [], []
+    ws_rows = 0
+    if factor > 1:
+        split, plain = [], []
+        for (i, q0, kmax) in t256:
+            span = (kmax + E2_KVBLK - 1) // E2_KVBLK
+            p = min(factor, max(1, span // 4))
+            (split if p > 1 else plain).append((i, q0, p))
+        # cap the f32 partial workspace at ~512 MB
+        slab = num_q_heads * EXTEND2_QTILE * 130 * 4
+        if sum(p for _, _, p in split) * slab > 512 * 1024 * 1024:
+            split, plain = [], split + plain
+        for (i, q0, p) in split:
+            combine.append((len(rows), p, i, q0))
+            for part in range(p):
+                rows.append((i, q0, part, p))
+        ws_rows = len(rows)
+        for (i, q0, p) in plain:
+            rows.append((i, q0, 0, 1))
+    else:
+        rows = [(i, q0, 0, 1) for (i, q0, _) in t256]
+    t256_t = torch.tensor(rows, dtype=torch.int32,
+                          device=device).reshape(-1, 4)
+    ctab = torch.tensor(combine, dtype=torch.int32,
+                        device=device).reshape(-1, 4)
+
     if dec:
         # gather indices: the single q row of each decode seq sits at
         # cu_seqlens_q[i]; seq-level metadata indexes with `dec` itself
@@ -185,7 +221,7 @@ def build_extend_tiles(q_lens: list[int], kv_lens: list[int] | None,
         dseqs = torch.tensor(dec, dtype=torch.long, device=device)
     else:
         drows = dseqs = None
-    return mk(t64), mk(t256), (drows, dseqs)
+    return mk64(t64), t256_t, (drows, dseqs), ctab, ws_rows
 
 
 def attention_extend_paged(
@@ -207,15 +243,27 @@ def attention_extend_paged(
             (q.shape[0], q.shape[1], q.shape[2]), dtype=q.dtype, device=q.device
         )
         kv_fp8 = k_cache.dtype != torch.bfloat16
-        t64, t256, (drows, dseqs) = (
+        t64, t256, (drows, dseqs), ctab, ws_rows = (
             tiles if tiles is not None
-            else build_extend_tiles(q_lens, None, kv_fp8, q.device)
+            else build_extend_tiles(q_lens, None, kv_fp8, q.device,
+                                    num_q_heads=q.shape[1])
         )
         if t256.numel():
+            if ws_rows:
+                ws = _extend_ws(
+                    ws_rows * q.shape[1] * EXTEND2_QTILE * (q.shape[2] + 2),
+                    q.device,
+                )
+            else:
+                ws = q.new_empty(0, dtype=torch.float32)
             _native().attention_extend_paged2(
                 out, q, k_cache, v_cache, block_tables, kv_lens, cu_seqlens_q,
-                t256, scale, window,
+                t256, ws, scale, window,
             )
+            if ctab.numel():
+                _native().attention_extend2_combine(
+                    out, ws, ctab, cu_seqlens_q, scale,
+                )
         if t64.numel():
             _native().attention_extend_paged(
                 out, q, k_cache, v_cache, block_tables, kv_lens, cu_seqlens_q,
@@ -278,6 +326,18 @@ def attention_decode_paged(
     return ref.attention_decode_paged(
         q, k_cache, v_cache, block_tables, seq_lens, scale, window
     )
+
+
+_EXTEND_WS: dict = {}
+
+
+def _extend_ws(need: int, device):
+    key = (device.index,)
+    ws = _EXTEND_WS.get(key)
+    if ws is None or ws.numel() < need:
+        ws = torch.empty(need, dtype=torch.float32, device=device)
+        _EXTEND_WS[key] = ws
+    return ws
 
 
 _SKINNY_WS: dict = {}

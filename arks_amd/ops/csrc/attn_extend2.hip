@@ -67,7 +67,12 @@ __global__ __launch_bounds__(512) void attn_extend2_kernel(
     const int* __restrict__ block_tables,  // [num_seqs, max_blocks]
     const int* __restrict__ kv_lens,       // [num_seqs]
     const int* __restrict__ cu_seqlens_q,  // [num_seqs + 1]
-    const int* __restrict__ tile_info,     // [ntiles, 2] = (seq_idx, q0)
+    // [ntiles, 4] = (seq_idx, q0, part_idx, nparts): nparts > 1 splits the
+    // KV range flash-decode-style; partials go to `part_ws` and a combine
+    // kernel merges them (load balance: causal q-tiles differ 4x in KV
+    // depth and small batches leave the chip half idle otherwise)
+    const int* __restrict__ tile_info,
+    float* __restrict__ part_ws,  // [ntiles, Hq, QTILE, D+2] f32 slabs
     const float scale, const int num_q_heads, const int num_kv_heads,
     const int max_blocks, const int64_t q_stride, const int window) {
   constexpr int STEPS = HEAD_DIM / 16;   // QK^T K-contraction steps
@@ -76,8 +81,10 @@ __global__ __launch_bounds__(512) void attn_extend2_kernel(
 
   const int h = blockIdx.x;
   const int kvh = h / (num_q_heads / num_kv_heads);
-  const int seq_idx = tile_info[blockIdx.y * 2];
-  const int q0 = tile_info[blockIdx.y * 2 + 1];
+  const int seq_idx = tile_info[blockIdx.y * 4];
+  const int q0 = tile_info[blockIdx.y * 4 + 1];
+  const int part = tile_info[blockIdx.y * 4 + 2];
+  const int nparts = tile_info[blockIdx.y * 4 + 3];
   const int q_start = cu_seqlens_q[seq_idx];
   const int q_len = cu_seqlens_q[seq_idx + 1] - q_start;
   const int kv_len = kv_lens[seq_idx];
@@ -129,10 +136,15 @@ __global__ __launch_bounds__(512) void attn_extend2_kernel(
   for (int n = 0; n < DTILES; ++n) oacc[n] = {};
 
   const int kmax = min(kv_len, kv_off + q0 + E2_QTILE);
-  const int ntiles = (kmax + E2_KVBLK - 1) / E2_KVBLK;
+  const int all_tiles = (kmax + E2_KVBLK - 1) / E2_KVBLK;
   // Sliding window: the first key visible to any row of this tile.
-  const int j0 =
+  const int jw =
       (window > 0) ? max(0, (kv_off + q0 - window + 1) / E2_KVBLK) : 0;
+  // this part's contiguous slice of the KV tile range
+  const int span = all_tiles - jw;
+  const int per = (span + nparts - 1) / nparts;
+  const int j0 = jw + part * per;
+  const int ntiles = min(all_tiles, j0 + per);
   const float scale2 = scale * E2_LOG2E;
 
   // Cooperative staging: 512 threads x 2 vec8 cover one 64x128 tile (one
@@ -375,6 +387,26 @@ __global__ __launch_bounds__(512) void attn_extend2_kernel(
 
   // ---- Epilogue. O reg 4g+r of dim-tile n is q-row 8g+4*half+r, dim
   // 32n+col; l_run for row `row` lives in lane row (either half).
+  if (nparts > 1) {
+    // partial: UNNORMALIZED O + per-row (m, l) into this part's slab
+    // [QTILE rows][D+2] f32 (combined by attn_extend2_combine)
+    float* slab = part_ws +
+        ((int64_t)blockIdx.y * num_q_heads + h) * E2_QTILE * (HEAD_DIM + 2);
+#pragma unroll
+    for (int r16 = 0; r16 < 16; ++r16) {
+      const int row = 8 * (r16 >> 2) + 4 * half + (r16 & 3);
+      float* wr = slab + (int64_t)(wave * 32 + row) * (HEAD_DIM + 2) + col;
+#pragma unroll
+      for (int n = 0; n < DTILES; ++n) wr[32 * n] = oacc[n][r16];
+    }
+    if (half == 0) {
+      // lane `col` owns q-row `col` of this wave: write its m/l
+      float* wr = slab + (int64_t)(wave * 32 + col) * (HEAD_DIM + 2);
+      wr[HEAD_DIM] = m_run;
+      wr[HEAD_DIM + 1] = l_run;
+    }
+    return;
+  }
 #pragma unroll
   for (int r16 = 0; r16 < 16; ++r16) {
     const int row = 8 * (r16 >> 2) + 4 * half + (r16 & 3);
@@ -391,6 +423,53 @@ __global__ __launch_bounds__(512) void attn_extend2_kernel(
   }
 }
 
+// Merge the flash partials of split tiles: grid (Hq, n_split), one block
+// per (head, logical tile); combine_table rows = (first_y, nparts, seq,
+// q0). Standard flash merge in f32: m* = max m_p; out = sum_p
+// exp(m_p - m*) O_p / sum_p exp(m_p - m*) l_p (m is base-2: exp2).
+template <int HEAD_DIM>
+__global__ __launch_bounds__(256) void attn_extend2_combine_kernel(
+    bf16* __restrict__ out, const float* __restrict__ part_ws,
+    const int* __restrict__ combine_table,
+    const int* __restrict__ cu_seqlens_q, const float scale,
+    const int num_q_heads) {
+  const int h = blockIdx.x;
+  const int ci = blockIdx.y;
+  const int first_y = combine_table[ci * 4];
+  const int nparts = combine_table[ci * 4 + 1];
+  const int seq_idx = combine_table[ci * 4 + 2];
+  const int q0 = combine_table[ci * 4 + 3];
+  const int q_start = cu_seqlens_q[seq_idx];
+  const int q_len = cu_seqlens_q[seq_idx + 1] - q_start;
+  const float scale2 = scale * E2_LOG2E;
+  const int64_t slab_stride = (int64_t)num_q_heads * E2_QTILE * (HEAD_DIM + 2);
+  const float* base = part_ws + (int64_t)first_y * slab_stride +
+                      (int64_t)h * E2_QTILE * (HEAD_DIM + 2);
+  // two rows per iteration: 256 threads = 2 x 128 lanes over D
+  for (int row = threadIdx.x / HEAD_DIM; row < E2_QTILE;
+       row += 256 / HEAD_DIM) {
+    const int qrow = q0 + row;
+    if (qrow >= q_len) continue;
+    const int d = threadIdx.x % HEAD_DIM;
+    const float* r0 = base + (int64_t)row * (HEAD_DIM + 2);
+    float m_star = -1e30f;
+    for (int p = 0; p < nparts; ++p)
+      m_star = fmaxf(m_star, r0[p * slab_stride + HEAD_DIM]);
+    float l_tot = 0.f;
+    float acc = 0.f;
+    for (int p = 0; p < nparts; ++p) {
+      const float mp = r0[p * slab_stride + HEAD_DIM];
+      const float lp = r0[p * slab_stride + HEAD_DIM + 1];
+      const float a = __builtin_amdgcn_exp2f((mp - m_star) * scale2);
+      l_tot += a * lp;
+      acc += a * r0[p * slab_stride + d];
+    }
+    const float inv = l_tot > 0.f ? 1.f / l_tot : 0.f;
+    out[((int64_t)(q_start + qrow) * num_q_heads + h) * HEAD_DIM + d] =
+        float_to_bf16_bits(acc * inv);
+  }
+}
+
 }  // namespace arks
 
 using namespace arks;
@@ -398,9 +477,9 @@ using namespace arks;
 extern "C" void arks_attn_extend_paged2(
     void* out, const void* q, const void* k_cache, const void* v_cache,
     const void* block_tables, const void* kv_lens, const void* cu_seqlens_q,
-    const void* tile_info, int ntiles, float scale, int num_q_heads,
-    int num_kv_heads, int head_dim, int max_blocks, int64_t q_stride,
-    int window, hipStream_t stream) {
+    const void* tile_info, void* part_ws, int ntiles, float scale,
+    int num_q_heads, int num_kv_heads, int head_dim, int max_blocks,
+    int64_t q_stride, int window, hipStream_t stream) {
   dim3 grid(num_q_heads, ntiles), block(512);
   static bool attr_set[2] = {false, false};
   auto launch = [&](auto kern, int lds_bytes, int idx) {
@@ -414,8 +493,9 @@ extern "C" void arks_attn_extend_paged2(
                        (const bf16*)q, (const bf16*)k_cache,
                        (const bf16*)v_cache, (const int*)block_tables,
                        (const int*)kv_lens, (const int*)cu_seqlens_q,
-                       (const int*)tile_info, scale, num_q_heads, num_kv_heads,
-                       max_blocks, q_stride, window);
+                       (const int*)tile_info, (float*)part_ws, scale,
+                       num_q_heads, num_kv_heads, max_blocks, q_stride,
+                       window);
   };
   if (head_dim == 128) {
     constexpr int LDS = (2 * 64 * 128 + 2 * 16 * 8 * VBLK_ELEMS) * 2;
@@ -423,5 +503,23 @@ extern "C" void arks_attn_extend_paged2(
   } else if (head_dim == 64) {
     constexpr int LDS = (2 * 64 * 64 + 2 * 16 * 4 * VBLK_ELEMS) * 2;
     launch(attn_extend2_kernel<64>, LDS, 1);
+  }
+}
+
+extern "C" void arks_attn_extend2_combine(
+    void* out, const void* part_ws, const void* combine_table,
+    const void* cu_seqlens_q, int n_split, float scale, int num_q_heads,
+    int head_dim, hipStream_t stream) {
+  dim3 grid(num_q_heads, n_split), block(256);
+  if (head_dim == 128) {
+    hipLaunchKernelGGL((attn_extend2_combine_kernel<128>), grid, block, 0,
+                       stream, (bf16*)out, (const float*)part_ws,
+                       (const int*)combine_table, (const int*)cu_seqlens_q,
+                       scale, num_q_heads);
+  } else if (head_dim == 64) {
+    hipLaunchKernelGGL((attn_extend2_combine_kernel<64>), grid, block, 0,
+                       stream, (bf16*)out, (const float*)part_ws,
+                       (const int*)combine_table, (const int*)cu_seqlens_q,
+                       scale, num_q_heads);
   }
 }

@@ -63,10 +63,15 @@ void arks_attn_extend_paged(void* out, const void* q, const void* k_cache,
 void arks_attn_extend_paged2(void* out, const void* q, const void* k_cache,
                              const void* v_cache, const void* block_tables,
                              const void* kv_lens, const void* cu_seqlens_q,
-                             const void* tile_info, int ntiles, float scale,
-                             int num_q_heads, int num_kv_heads, int head_dim,
-                             int max_blocks, int64_t q_stride, int window,
-                             hipStream_t stream);
+                             const void* tile_info, void* part_ws, int ntiles,
+                             float scale, int num_q_heads, int num_kv_heads,
+                             int head_dim, int max_blocks, int64_t q_stride,
+                             int window, hipStream_t stream);
+void arks_attn_extend2_combine(void* out, const void* part_ws,
+                               const void* combine_table,
+                               const void* cu_seqlens_q, int n_split,
+                               float scale, int num_q_heads, int head_dim,
+                               hipStream_t stream);
 void arks_quant_fp8_rows(void* out, void* inv_scale, const void* x, int rows,
                          int cols, hipStream_t stream);
 void arks_rmsnorm_fp8(void* out, void* inv_scale, const void* input,
@@ -325,8 +330,8 @@ void attention_extend_paged2(torch::Tensor out, torch::Tensor q,
                              torch::Tensor k_cache, torch::Tensor v_cache,
                              torch::Tensor block_tables, torch::Tensor kv_lens,
                              torch::Tensor cu_seqlens_q,
-                             torch::Tensor tile_info, double scale,
-                             int64_t window) {
+                             torch::Tensor tile_info, torch::Tensor part_ws,
+                             double scale, int64_t window) {
   check_bf16_contig(out, "out");
   check_bf16_rowstrided(q, "q");
   check_bf16_contig(k_cache, "k_cache");
@@ -335,7 +340,8 @@ void attention_extend_paged2(torch::Tensor out, torch::Tensor q,
   TORCH_CHECK(kv_lens.scalar_type() == torch::kInt32);
   TORCH_CHECK(cu_seqlens_q.scalar_type() == torch::kInt32);
   TORCH_CHECK(tile_info.scalar_type() == torch::kInt32);
-  TORCH_CHECK(tile_info.dim() == 2 && tile_info.size(1) == 2);
+  TORCH_CHECK(tile_info.dim() == 2 && tile_info.size(1) == 4,
+              "tile_info rows are (seq, q0, part, nparts)");
   const int num_q_heads = q.size(1);
   const int head_dim = q.size(2);
   const int num_kv_heads = k_cache.size(1);
@@ -343,12 +349,33 @@ void attention_extend_paged2(torch::Tensor out, torch::Tensor q,
   TORCH_CHECK(head_dim == 64 || head_dim == 128, "head_dim must be 64 or 128");
   const int ntiles = tile_info.size(0);
   const int max_blocks = block_tables.size(1);
+  void* ws = nullptr;
+  if (part_ws.numel() > 0) {
+    // sized by the host: split rows sort first, so only their slabs exist
+    TORCH_CHECK(part_ws.scalar_type() == torch::kFloat32);
+    ws = part_ws.data_ptr();
+  }
   arks_attn_extend_paged2(out.data_ptr(), q.data_ptr(), k_cache.data_ptr(),
                           v_cache.data_ptr(), block_tables.data_ptr(),
                           kv_lens.data_ptr(), cu_seqlens_q.data_ptr(),
-                          tile_info.data_ptr(), ntiles, (float)scale,
+                          tile_info.data_ptr(), ws, ntiles, (float)scale,
                           num_q_heads, num_kv_heads, head_dim, max_blocks,
                           q.stride(0), (int)window, current_stream());
+}
+
+void attention_extend2_combine(torch::Tensor out, torch::Tensor part_ws,
+                               torch::Tensor combine_table,
+                               torch::Tensor cu_seqlens_q, double scale) {
+  check_bf16_rowstrided(out, "out");
+  TORCH_CHECK(part_ws.scalar_type() == torch::kFloat32);
+  TORCH_CHECK(combine_table.scalar_type() == torch::kInt32 &&
+              combine_table.dim() == 2 && combine_table.size(1) == 4);
+  const int num_q_heads = out.size(1);
+  const int head_dim = out.size(2);
+  arks_attn_extend2_combine(out.data_ptr(), part_ws.data_ptr(),
+                            combine_table.data_ptr(), cu_seqlens_q.data_ptr(),
+                            combine_table.size(0), (float)scale, num_q_heads,
+                            head_dim, current_stream());
 }
 
 void skinny_gemm(torch::Tensor out, torch::Tensor part, torch::Tensor a,
@@ -599,6 +626,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("attention_prefill_varlen", &attention_prefill_varlen);
   m.def("attention_extend_paged", &attention_extend_paged);
   m.def("attention_extend_paged2", &attention_extend_paged2);
+  m.def("attention_extend2_combine", &attention_extend2_combine);
   m.def("quant_fp8_rows", &quant_fp8_rows);
   m.def("rmsnorm_fp8", &rmsnorm_fp8);
   m.def("silu_mul_fp8", &silu_mul_fp8);

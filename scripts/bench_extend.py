@@ -29,15 +29,15 @@ def bench(fn, iters=30):
 
 
 def tiles256(q_lens, kv_lens, device):
-    """Work-sorted 256-row tiles (mirrors ops.build_extend_tiles)."""
+    """Work-sorted unsplit 256-row tiles [n,4] (part=0, nparts=1)."""
     t = []
     for i, n in enumerate(q_lens):
         off = kv_lens[i] - n
         for q0 in range(0, n, 256):
             t.append((i, q0, min(kv_lens[i], off + q0 + 256)))
     t.sort(key=lambda x: -x[2])
-    return torch.tensor([x[:2] for x in t], dtype=torch.int32,
-                        device=device).reshape(-1, 2)
+    return torch.tensor([(x[0], x[1], 0, 1) for x in t], dtype=torch.int32,
+                        device=device).reshape(-1, 4)
 
 
 def main():
@@ -68,12 +68,17 @@ def main():
         nat = ops._native()
         t_old = bench(lambda: nat.attention_extend_paged(
             out, q, kc, vc, bt, kvl, cu, t64i, scale, 0))
+        ws0 = torch.empty(0, dtype=torch.float32, device="cuda")
         t_new = bench(lambda: nat.attention_extend_paged2(
-            out, q, kc, vc, bt, kvl, cu, t256, scale, 0))
+            out, q, kc, vc, bt, kvl, cu, t256, ws0, scale, 0))
+        # split path through the full dispatcher (auto policy + combine)
+        t_split = bench(lambda: ops.attention_extend_paged(
+            q, kc, vc, bt, kvl, cu, [qlen] * S, scale))
         print(f"S={nseq:3d} q={qlen:5d} kv={kvlen:5d}  "
               f"old {t_old:9.1f}us {flops/t_old/1e6:7.1f}TF  "
               f"new {t_new:9.1f}us {flops/t_new/1e6:7.1f}TF  "
-              f"x{t_old/t_new:.2f}")
+              f"split {t_split:9.1f}us {flops/t_split/1e6:7.1f}TF  "
+              f"x{t_old/t_new:.2f}/x{t_old/t_split:.2f}")
 
 
 if __name__ == "__main__":

@@ -28,12 +28,13 @@ def main():
     tiles = []
     for i in range(S):
         for q0 in range(0, qlen, 256):
-            tiles.append((i, q0))
-    t256 = torch.tensor(tiles, dtype=torch.int32, device="cuda").reshape(-1, 2)
+            tiles.append((i, q0, 0, 1))
+    t256 = torch.tensor(tiles, dtype=torch.int32, device="cuda").reshape(-1, 4)
     out = torch.empty_like(q)
+    ws0 = torch.empty(0, dtype=torch.float32, device="cuda")
     nat = ops._native()
     for _ in range(20):
-        nat.attention_extend_paged2(out, q, kc, vc, bt, kvl, cu, t256,
+        nat.attention_extend_paged2(out, q, kc, vc, bt, kvl, cu, t256, ws0,
                                     1.0 / math.sqrt(hd), 0)
     torch.cuda.synchronize()
 

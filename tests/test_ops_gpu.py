@@ -468,6 +468,10 @@ def _mk_paged(spec, hq, hkv, hd, seed=13):
         (4, 4, 64, [(600, 600), (300, 129)]),
         # mixed routing: one seq per kernel in the same call
         (8, 2, 128, [(512, 512), (40, 13)]),
+        # deep KV behind a short chunk: the split+combine path (one
+        # 256-row tile over 8k keys forces kv-partitioning)
+        (28, 4, 128, [(8192, 256)]),
+        (8, 2, 64, [(4096, 200), (4096, 256)]),
     ],
 )
 def test_attention_extend2_big_tiles(hq, hkv, hd, spec):
