@@ -137,6 +137,7 @@ def attention_prefill_varlen(q, k, v, cu_seqlens, seq_lens: list[int],
 
 
 EXTEND2_QTILE = 256  # q rows per workgroup in the 8-wave ladder kernel
+E2_KVBLK = 64        # keys per LDS tile in the ladder kernel
 # Sequences with at least this many new q tokens go to the 256-row ladder
 # kernel; shorter ones (decode-adjacent chunks, prefix-cache hits) stay on
 # the 64-row kernel where a mostly-empty 256-row tile would waste waves.
