@@ -183,8 +183,11 @@ def build_extend_tiles(q_lens: list[int], kv_lens: list[int] | None,
     # fill ~2 workgroups per CU, split the deepest tiles' key ranges.
     base_wgs = len(t256) * max(num_q_heads, 1)
     factor = 1
-    if t256 and base_wgs < 512:
-        factor = min(8, -(-512 // base_wgs) + 1)
+    if t256 and base_wgs < 256:
+        # truly starved launches only (256 CUs, 1 block/CU kernel): the
+        # partial slabs + combine cost ~2x slab traffic, so marginal
+        # shortfalls are better left unsplit
+        factor = min(8, -(-512 // base_wgs))
     rows, combine = [], []
     ws_rows = 0
     if factor > 1:

@@ -46,7 +46,8 @@ def main():
     bs = 16
     # (nseq, q_len, kv_len): bench-shaped chunks + long-context chunks
     for nseq, qlen, kvlen in ((8, 1024, 1024), (8, 512, 1024), (4, 2048, 2048),
-                              (1, 8192, 8192), (16, 1024, 1024)):
+                              (1, 8192, 8192), (16, 1024, 1024),
+                              (1, 256, 8192), (2, 256, 16384)):
         S, Tq = nseq, nseq * qlen
         nb = (kvlen + bs - 1) // bs
         q = torch.randn(Tq, hq, hd, dtype=torch.bfloat16, device="cuda")
@@ -71,9 +72,11 @@ def main():
         ws0 = torch.empty(0, dtype=torch.float32, device="cuda")
         t_new = bench(lambda: nat.attention_extend_paged2(
             out, q, kc, vc, bt, kvl, cu, t256, ws0, scale, 0))
-        # split path through the full dispatcher (auto policy + combine)
+        # split path: prebuilt tiles (as the engine does once per batch)
+        tiles = ops.build_extend_tiles([qlen] * S, [kvlen] * S, False,
+                                       "cuda", num_q_heads=hq)
         t_split = bench(lambda: ops.attention_extend_paged(
-            q, kc, vc, bt, kvl, cu, [qlen] * S, scale))
+            q, kc, vc, bt, kvl, cu, [qlen] * S, scale, tiles=tiles))
         print(f"S={nseq:3d} q={qlen:5d} kv={kvlen:5d}  "
               f"old {t_old:9.1f}us {flops/t_old/1e6:7.1f}TF  "
               f"new {t_new:9.1f}us {flops/t_new/1e6:7.1f}TF  "
