@@ -423,10 +423,12 @@ __global__ __launch_bounds__(512) void attn_extend2_kernel(
   }
 }
 
-// Merge the flash partials of split tiles: grid (Hq, n_split), one block
-// per (head, logical tile); combine_table rows = (first_y, nparts, seq,
-// q0). Standard flash merge in f32: m* = max m_p; out = sum_p
-// exp(m_p - m*) O_p / sum_p exp(m_p - m*) l_p (m is base-2: exp2).
+// Merge the flash partials of split tiles: grid (Hq, n_split, row-groups
+// of 8) — the row-group axis exists purely for parallelism (a single
+// (head, tile) block is latency-bound on the strided part reads);
+// combine_table rows = (first_y, nparts, seq, q0). Standard flash merge
+// in f32: m* = max m_p; out = sum_p exp(m_p - m*) O_p / sum_p
+// exp(m_p - m*) l_p (m is base-2: exp2).
 template <int HEAD_DIM>
 __global__ __launch_bounds__(256) void attn_extend2_combine_kernel(
     bf16* __restrict__ out, const float* __restrict__ part_ws,
@@ -445,8 +447,9 @@ __global__ __launch_bounds__(256) void attn_extend2_combine_kernel(
   const int64_t slab_stride = (int64_t)num_q_heads * E2_QTILE * (HEAD_DIM + 2);
   const float* base = part_ws + (int64_t)first_y * slab_stride +
                       (int64_t)h * E2_QTILE * (HEAD_DIM + 2);
-  // two rows per iteration: 256 threads = 2 x 128 lanes over D
-  for (int row = threadIdx.x / HEAD_DIM; row < E2_QTILE;
+  // this block's 8-row group; 256 threads = 256/HEAD_DIM rows per pass
+  const int row0 = blockIdx.z * 8;
+  for (int row = row0 + threadIdx.x / HEAD_DIM; row < row0 + 8;
        row += 256 / HEAD_DIM) {
     const int qrow = q0 + row;
     if (qrow >= q_len) continue;
@@ -510,7 +513,7 @@ extern "C" void arks_attn_extend2_combine(
     void* out, const void* part_ws, const void* combine_table,
     const void* cu_seqlens_q, int n_split, float scale, int num_q_heads,
     int head_dim, hipStream_t stream) {
-  dim3 grid(num_q_heads, n_split), block(256);
+  dim3 grid(num_q_heads, n_split, E2_QTILE / 8), block(256);
   if (head_dim == 128) {
     hipLaunchKernelGGL((attn_extend2_combine_kernel<128>), grid, block, 0,
                        stream, (bf16*)out, (const float*)part_ws,
