@@ -110,19 +110,50 @@ class Operator:
                 seen_requeue.add((kind, ns, name))
         return n
 
-    def run(self, poll_interval: float = 1.0) -> None:
-        while not self._stop.is_set():
-            try:
-                kind, ns, name = self._queue.get(timeout=poll_interval)
-            except queue.Empty:
-                continue
-            delay = self._dispatch(kind, ns, name)
-            if delay == 0:
-                self._queue.put((kind, ns, name))
-            elif delay is not None:
-                t = threading.Timer(delay, lambda: self._queue.put((kind, ns, name)))
-                t.daemon = True
-                t.start()
+    def run(self, poll_interval: float = 1.0, workers: int = 8) -> None:
+        """Long-running form: `workers` reconcile threads drain the queue
+        (the reference runs MaxConcurrentReconciles=30 per controller;
+        reconcilers here are level-triggered and idempotent). The same
+        (kind, ns, name) never reconciles concurrently: an in-flight key
+        is re-queued by its finishing worker instead."""
+        inflight: set = set()
+        lock = threading.Lock()
+
+        def requeue_later(item, delay):
+            t = threading.Timer(delay, lambda: self._queue.put(item))
+            t.daemon = True
+            t.start()
+
+        def worker():
+            while not self._stop.is_set():
+                try:
+                    item = self._queue.get(timeout=poll_interval)
+                except queue.Empty:
+                    continue
+                with lock:
+                    if item in inflight:
+                        # coalesce: someone is reconciling this key now —
+                        # run it again shortly after they finish
+                        requeue_later(item, 0.05)
+                        continue
+                    inflight.add(item)
+                try:
+                    delay = self._dispatch(*item)
+                finally:
+                    with lock:
+                        inflight.discard(item)
+                if delay == 0:
+                    self._queue.put(item)
+                elif delay is not None:
+                    requeue_later(item, delay)
+
+        threads = [threading.Thread(target=worker, daemon=True,
+                                    name=f"reconcile-{i}")
+                   for i in range(max(workers, 1))]
+        for t in threads:
+            t.start()
+        for t in threads:
+            t.join()
 
     def stop(self):
         self._stop.set()

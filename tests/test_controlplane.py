@@ -295,3 +295,36 @@ def test_app_replica_scale_propagates_to_workload():
     # status no longer satisfies 4 replicas -> leaves Running
     appx = store.get("ArksApplication", "default", "a1")
     assert appx.status.phase is not None
+
+
+def test_operator_run_concurrent_workers():
+    """The threaded run() form reconciles events from multiple workers
+    without double-running one key concurrently."""
+    import threading
+    import time
+
+    from arks_amd.controlplane import Operator, Store
+    from arks_amd.crd.types import parse_manifest
+
+    store = Store()
+    op = Operator(store)
+    t = threading.Thread(target=op.run, kwargs={"poll_interval": 0.05,
+                                                "workers": 4}, daemon=True)
+    t.start()
+    for i in range(6):
+        store.apply(parse_manifest({
+            "apiVersion": "arks.ai/v1", "kind": "ArksModel",
+            "metadata": {"name": f"m{i}", "namespace": "default"},
+            "spec": {"model": f"org/m{i}",
+                     "storage": {"pvc": {"name": f"m{i}", "spec": {
+                         "accessModes": ["ReadWriteOnce"],
+                         "resources": {"requests": {"storage": "1Gi"}}}}}},
+        }))
+    deadline = time.time() + 10
+    while time.time() < deadline:
+        pvcs = store.list("PersistentVolumeClaim", "default")
+        if len(pvcs) >= 6:
+            break
+        time.sleep(0.1)
+    op.stop()
+    assert len(store.list("PersistentVolumeClaim", "default")) >= 6
