@@ -35,6 +35,10 @@ class ModelConfig:
     shared_expert_intermediate_size: int | None = None  # Qwen2-MoE
     norm_topk_prob: bool = True  # renormalize top-k routing weights
     sliding_window: int | None = None  # sliding-window attention size
+    # Qwen2-style per-layer mixing: layers >= max_window_layers slide,
+    # lower layers use full attention (HF layer_types overrides when set)
+    max_window_layers: int | None = None
+    layer_types: list | None = None
     qk_norm: bool = False  # Qwen3: per-head RMSNorm on q/k before RoPE
     tie_word_embeddings: bool = False
     attention_bias: bool = True  # qwen2 has qkv bias; llama does not
@@ -43,15 +47,37 @@ class ModelConfig:
     torch_dtype: str = "bfloat16"
 
     def __post_init__(self):
-        # Sliding-window attention (Mistral-v0.1-style configs declare
-        # 4096) is implemented end to end: the attention kernels mask keys
-        # outside each query's window and the engine drops out-of-window
-        # KV pages. Qwen2-style per-layer max_window_layers mixing is not
-        # modeled: a declared window applies to every layer.
+        # Sliding-window attention is implemented end to end: the
+        # attention kernels mask keys outside each query's per-layer
+        # window (layer_window: Mistral = all layers; Qwen2
+        # max_window_layers / layer_types mixing honored) and the engine
+        # drops out-of-window KV pages when EVERY layer slides
+        # (uniform_window).
         if self.sliding_window is not None and self.sliding_window <= 0:
             raise ValueError(
                 f"invalid sliding_window={self.sliding_window}"
             )
+
+    def layer_window(self, layer_idx: int) -> int:
+        """Effective attention window for one layer (0 = full attention).
+        HF Qwen2 semantics: layer_types wins when present; otherwise
+        layers >= max_window_layers slide and lower layers are full."""
+        if self.sliding_window is None:
+            return 0
+        if self.layer_types is not None:
+            return (self.sliding_window
+                    if self.layer_types[layer_idx] == "sliding_attention"
+                    else 0)
+        if self.max_window_layers is not None:
+            return self.sliding_window if layer_idx >= self.max_window_layers else 0
+        return self.sliding_window
+
+    def uniform_window(self) -> int:
+        """The window when EVERY layer slides (else 0): KV page dropping
+        and the prefix-cache gate need all layers windowed — a single
+        full-attention layer keeps the whole history live."""
+        ws = [self.layer_window(i) for i in range(self.num_hidden_layers)]
+        return ws[0] if ws and all(w == ws[0] and w > 0 for w in ws) else 0
 
     @classmethod
     def from_hf_config(cls, cfg: dict) -> "ModelConfig":
@@ -80,6 +106,8 @@ class ModelConfig:
             # Qwen2-family configs declare a window but disable it
             sliding_window=(cfg.get("sliding_window")
                             if cfg.get("use_sliding_window", True) else None),
+            max_window_layers=cfg.get("max_window_layers"),
+            layer_types=cfg.get("layer_types"),
             num_experts_per_tok=cfg.get("num_experts_per_tok", 2),
             moe_intermediate_size=cfg.get("moe_intermediate_size"),
             shared_expert_intermediate_size=cfg.get(

@@ -18,12 +18,14 @@ class LLMEngine:
     def __init__(self, cfg: EngineConfig):
         self.cfg = cfg
         self.model_cfg = cfg.model_config()
-        # Sliding-window attention (Mistral-style): the kernels mask keys
-        # outside each query's window and out-of-window KV pages are dropped
-        # back to the pool as sequences grow. Prefix caching is disabled for
-        # SWA models: a cached prefix page can be dropped+recycled while its
-        # digest is still matchable, so reuse would read stale pages.
-        self.window = self.model_cfg.sliding_window or 0
+        # Sliding-window attention: each layer masks keys outside its
+        # window (per-layer Qwen2 mixing honored via
+        # ModelConfig.layer_window). KV page dropping + the prefix-cache
+        # gate apply only when EVERY layer slides (uniform_window) — a
+        # single full-attention layer keeps the whole history live, and a
+        # cached prefix page must never be dropped+recycled while its
+        # digest is still matchable.
+        self.window = self.model_cfg.uniform_window()
         if self.window and cfg.enable_prefix_caching:
             cfg.enable_prefix_caching = False
         self.runner = ModelRunner(cfg, self.model_cfg)

@@ -48,16 +48,18 @@ class RMSNorm(nn.Module):
 
 
 class Attention(nn.Module):
-    def __init__(self, cfg: ModelConfig, dtype=torch.bfloat16):
+    def __init__(self, cfg: ModelConfig, dtype=torch.bfloat16,
+                 layer_idx: int = 0):
         super().__init__()
         tp = get_tp_world_size()
         self.head_dim = cfg.head_dim
         self.nq_local = cfg.num_attention_heads // tp
         self.nkv_local = cfg.num_key_value_heads // tp
         self.scale = 1.0 / math.sqrt(cfg.head_dim)
-        # sliding-window attention (Mistral-style, applied globally when the
-        # config declares one; window masking happens inside the kernels)
-        self.window = cfg.sliding_window or 0
+        # per-layer sliding window (HF Qwen2 max_window_layers/layer_types
+        # semantics; Mistral = every layer; 0 = full attention). Masking
+        # happens inside the kernels.
+        self.window = cfg.layer_window(layer_idx)
         self.qkv_proj = QKVParallelLinear(
             cfg.hidden_size, cfg.head_dim, cfg.num_attention_heads,
             cfg.num_key_value_heads, bias=cfg.attention_bias, dtype=dtype,
@@ -349,10 +351,11 @@ class MoEMLP(nn.Module):
 
 
 class DecoderLayer(nn.Module):
-    def __init__(self, cfg: ModelConfig, dtype=torch.bfloat16):
+    def __init__(self, cfg: ModelConfig, dtype=torch.bfloat16,
+                 layer_idx: int = 0):
         super().__init__()
         self.input_layernorm = RMSNorm(cfg.hidden_size, cfg.rms_norm_eps, dtype)
-        self.self_attn = Attention(cfg, dtype)
+        self.self_attn = Attention(cfg, dtype, layer_idx=layer_idx)
         self.post_attention_layernorm = RMSNorm(cfg.hidden_size, cfg.rms_norm_eps, dtype)
         self.mlp = (
             MoEMLP(cfg, dtype) if cfg.num_local_experts > 0 else MLP(cfg, dtype)
@@ -382,7 +385,8 @@ class LlamaFamilyForCausalLM(nn.Module):
         )
         self.embed_tokens.weight.requires_grad_(False)
         self.layers = nn.ModuleList(
-            DecoderLayer(cfg, dtype) for _ in range(cfg.num_hidden_layers)
+            DecoderLayer(cfg, dtype, layer_idx=i)
+            for i in range(cfg.num_hidden_layers)
         )
         self.norm = RMSNorm(cfg.hidden_size, cfg.rms_norm_eps, dtype)
         self.lm_head = ParallelLMHead(cfg.hidden_size, cfg.vocab_size, dtype)

@@ -426,3 +426,45 @@ def test_sliding_window_short_context_matches_full_attention():
         C.PRESET_CONFIGS["tiny"] = orig
     out_swa = swa.generate(prompts, sp)
     assert out_full == out_swa
+
+
+def test_per_layer_window_rule_and_mixed_generation():
+    """Qwen2 max_window_layers/layer_types semantics: layers below
+    max_window_layers are full attention; only all-sliding configs enable
+    KV page dropping."""
+    import dataclasses
+
+    import arks_amd.config as C
+    from arks_amd.config import ModelConfig
+
+    cfg = dataclasses.replace(C.PRESET_CONFIGS["tiny"], sliding_window=32,
+                              max_window_layers=1)
+    assert cfg.layer_window(0) == 0 and cfg.layer_window(1) == 32
+    assert cfg.uniform_window() == 0  # mixed -> no page dropping
+    cfg2 = dataclasses.replace(C.PRESET_CONFIGS["tiny"], sliding_window=32)
+    assert cfg2.uniform_window() == 32
+    cfg3 = dataclasses.replace(
+        C.PRESET_CONFIGS["tiny"], sliding_window=32,
+        layer_types=["full_attention", "sliding_attention"])
+    assert cfg3.layer_window(0) == 0 and cfg3.layer_window(1) == 32
+
+    # mixed-window model generates; no pages are dropped
+    C.PRESET_CONFIGS["tiny-mixed"] = cfg
+    try:
+        e = LLMEngine(EngineConfig(preset="tiny-mixed", device="cpu",
+                                   kv_cache_blocks=64, max_model_len=128))
+        assert e.window == 0
+        out = e.generate([[3, 1, 4] * 20],
+                         SamplingParams(max_tokens=40, ignore_eos=True))
+        assert len(out[0]) == 40
+        seq = None  # all pages retained
+    finally:
+        C.PRESET_CONFIGS.pop("tiny-mixed", None)
+
+    # HF parse: use_sliding_window + max_window_layers round-trip
+    mc = ModelConfig.from_hf_config({
+        "architectures": ["Qwen2ForCausalLM"], "sliding_window": 4096,
+        "use_sliding_window": True, "max_window_layers": 2,
+        "num_hidden_layers": 4,
+    })
+    assert mc.layer_window(0) == 0 and mc.layer_window(3) == 4096
