@@ -188,8 +188,13 @@ class ModelRunner:
         if sb.is_prefill:
             # Cached prefixes (prefix cache hits) are skipped: only the new
             # suffix runs as q rows; attention then reads the full paged KV
-            # through the extend kernel.
-            use_extend = any(s.num_cached_tokens > 0 for s in sb.seqs)
+            # through the extend kernel. On GPU, EVERY prefill takes the
+            # extend path (the 8-wave ladder kernel + kv-split live there;
+            # the contiguous-varlen kernel is slower and pages are written
+            # either way); the CPU reference keeps the contiguous path
+            # exercised for fresh prefills.
+            use_extend = (self.device.type == "cuda"
+                          or any(s.num_cached_tokens > 0 for s in sb.seqs))
             input_ids: list = []  # np.int64 arrays, one per seq
             positions: list = []
             slot_mapping: list = []
