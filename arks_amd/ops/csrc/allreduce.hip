@@ -24,6 +24,8 @@ namespace arks {
 
 constexpr int AR_MAX_WORLD = 8;
 constexpr int AR_MAX_BLOCKS = 64;
+// mailbox capacity per (parity, rank) slot — must match parallel/p2p.py
+constexpr int64_t AR_MAX_ELEMS = 1 << 19;
 
 struct ArPtrs {
   // peer p's mailbox base ([world][nelem] bf16) and flag base
@@ -46,6 +48,14 @@ __global__ __launch_bounds__(256) void one_shot_allreduce_kernel(
     const unsigned long long* __restrict__ seq, const int rank,
     const int64_t n) {
   const unsigned long long s = *seq;
+  // Parity double-buffering closes the consecutive-call race: rank B's
+  // call N+1 may start pushing while rank A still READS call N (B only
+  // waited for A's phase-1 flag, not A's reduce). Alternating mailbox
+  // halves by sequence parity makes the earliest same-slot reuse call
+  // N+2 — which transitively orders after A's call-N reduce (B's N+2
+  // waits A's N+1 flag; A posts it after its N reduce in stream order).
+  // Flags stay single-buffered: they are monotonic and waited with >=.
+  const int64_t pbase = (int64_t)(s & 1) * AR_MAX_WORLD * AR_MAX_ELEMS;
   const int nblk = gridDim.x;
   // per-block range stays 16-byte aligned (n is a multiple of 8)
   const int64_t per_blk = (((n + nblk - 1) / nblk) + 7) & ~(int64_t)7;
@@ -58,7 +68,8 @@ __global__ __launch_bounds__(256) void one_shot_allreduce_kernel(
     const uint4v v = *reinterpret_cast<const uint4v*>(src + i);
 #pragma unroll
     for (int p = 0; p < WORLD; ++p) {
-      bf16* m = reinterpret_cast<bf16*>(ptrs.mail[p]) + (int64_t)rank * n;
+      bf16* m =
+          reinterpret_cast<bf16*>(ptrs.mail[p]) + pbase + (int64_t)rank * n;
       *reinterpret_cast<uint4v*>(m + i) = v;
     }
   }
@@ -89,7 +100,7 @@ __global__ __launch_bounds__(256) void one_shot_allreduce_kernel(
   __builtin_amdgcn_fence(__ATOMIC_ACQUIRE, "");  // system-scope acquire
 
   // phase 3: reduce my (local) mailbox across the world slots
-  const bf16* mine = reinterpret_cast<const bf16*>(ptrs.mail[rank]);
+  const bf16* mine = reinterpret_cast<const bf16*>(ptrs.mail[rank]) + pbase;
   for (int64_t i = i0 + threadIdx.x * 8; i < i1; i += 256 * 8) {
     float acc[8] = {};
 #pragma unroll

@@ -52,7 +52,9 @@ class OneShotAllReduce:
             # raw hipMalloc allocations: an IPC handle for a torch
             # caching-allocator tensor maps the underlying block, not the
             # tensor's offset within it
-            self._mail_ptr, mail_h = nat.ipc_alloc(world * MAX_ELEMS * 2)
+            # x2: parity double-buffered mailbox halves (see
+            # allreduce.hip's consecutive-call race note)
+            self._mail_ptr, mail_h = nat.ipc_alloc(2 * 8 * MAX_ELEMS * 2)
             self._flag_ptr, flag_h = nat.ipc_alloc(world * AR_MAX_BLOCKS * 8)
             self._owned = [self._mail_ptr, self._flag_ptr]
             self.seq = torch.zeros(1, dtype=torch.int64, device=device)

@@ -600,8 +600,10 @@ def test_one_shot_allreduce_single_device():
     x0 = torch.randn(n, dtype=torch.bfloat16, device=DEV)
     x1 = torch.randn(n, dtype=torch.bfloat16, device=DEV)
 
-    # world = 1: identity (ipc_alloc'd raw buffers, as production uses)
-    mail_ptr, mail_h = nat.ipc_alloc(n * 2)
+    # world = 1: identity (ipc_alloc'd raw buffers, as production uses;
+    # mailboxes sized like production: 2 parities x 8 slots x MAX_ELEMS)
+    MAILBYTES = 2 * 8 * (1 << 19) * 2
+    mail_ptr, mail_h = nat.ipc_alloc(MAILBYTES)
     flag_ptr, _ = nat.ipc_alloc(64 * 8)
     assert len(bytes(mail_h)) == 64
     seq = torch.zeros(1, dtype=torch.int64, device=DEV)
@@ -616,7 +618,7 @@ def test_one_shot_allreduce_single_device():
     # huge sequence values) so the two sequential launches never spin
     mails, flgs = [], []
     for _ in range(2):
-        mp, _h = nat.ipc_alloc(2 * n * 2)
+        mp, _h = nat.ipc_alloc(MAILBYTES)
         fp, _h2 = nat.ipc_alloc(2 * 64 * 8)
         mails.append(mp)
         flgs.append(fp)
