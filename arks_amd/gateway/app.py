@@ -69,16 +69,53 @@ class GatewayMetrics:
         )
 
 
+class OutlierDetector:
+    """Passive outlier ejection for the no-Envoy mode: a backend with
+    `threshold` CONSECUTIVE 5xx/connect errors is ejected for
+    `ejection_s` (the reference delegates this to Envoy's
+    BackendTrafficPolicy passive health check: 3x5xx -> 30 s)."""
+
+    def __init__(self, threshold: int = 3, ejection_s: float = 30.0,
+                 clock=time.time):
+        self.threshold = threshold
+        self.ejection_s = ejection_s
+        self.clock = clock
+        self._consecutive: dict[str, int] = {}
+        self._ejected_until: dict[str, float] = {}
+
+    def record(self, backend: str, ok: bool) -> None:
+        if ok:
+            self._consecutive.pop(backend, None)
+            return
+        n = self._consecutive.get(backend, 0) + 1
+        self._consecutive[backend] = n
+        if n >= self.threshold:
+            self._ejected_until[backend] = self.clock() + self.ejection_s
+            self._consecutive[backend] = 0
+
+    def is_ejected(self, backend: str) -> bool:
+        until = self._ejected_until.get(backend)
+        if until is None:
+            return False
+        if until <= self.clock():
+            del self._ejected_until[backend]
+            return False
+        return True
+
+
 class BackendResolver:
     """Maps (namespace, model) -> base URL of a READY backend, by reading the
-    HTTPRoute objects the endpoint controller generates (weighted pick)."""
+    HTTPRoute objects the endpoint controller generates (weighted pick,
+    passive-outlier-aware)."""
 
     def __init__(self, store: Store,
-                 url_for_service: Callable[[str, str], str] | None = None):
+                 url_for_service: Callable[[str, str], str] | None = None,
+                 outliers: OutlierDetector | None = None):
         self.store = store
         self.url_for_service = url_for_service or (
             lambda ns, svc: f"http://{svc}.{ns}.svc:8080"
         )
+        self.outliers = outliers or OutlierDetector()
 
     @staticmethod
     def _match_entry(m: dict, path: str, headers: dict[str, str]) -> bool:
@@ -121,6 +158,13 @@ class BackendResolver:
                     break
         if not refs:
             return None
+        # skip ejected backends (fall back to the full set if ALL are
+        # ejected — Envoy's maxEjectionPercent analogue)
+        live = [r for r in refs
+                if not self.outliers.is_ejected(
+                    self.url_for_service(namespace, r["name"]))]
+        if live:
+            refs = live
         weights = [max(int(r.get("weight", 1)), 0) for r in refs]
         total = sum(weights)
         if total <= 0:
@@ -251,8 +295,10 @@ def create_gateway_app(
             try:
                 resp = await client.post(url, content=body, headers=fwd_headers)
             except httpx.HTTPError as e:
+                resolver.outliers.record(base, False)
                 metrics.requests_total.labels(**labels, status="502").inc()
                 return _err(502, f"backend error: {e}")
+            resolver.outliers.record(base, resp.status_code < 500)
             status = str(resp.status_code)
             if resp.status_code == 200:
                 try:
@@ -268,7 +314,13 @@ def create_gateway_app(
 
         # streaming: relay SSE, parse the final usage chunk
         req = client.build_request("POST", url, content=body, headers=fwd_headers)
-        upstream = await client.send(req, stream=True)
+        try:
+            upstream = await client.send(req, stream=True)
+        except httpx.HTTPError as e:
+            resolver.outliers.record(base, False)
+            metrics.requests_total.labels(**labels, status="502").inc()
+            return _err(502, f"backend error: {e}")
+        resolver.outliers.record(base, upstream.status_code < 500)
         if upstream.status_code != 200:
             content = await upstream.aread()
             await upstream.aclose()

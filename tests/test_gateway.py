@@ -327,3 +327,42 @@ def test_quota_gauges_exported():
     text = generate_latest(gw.state.provider.metrics.registry).decode()
     assert 'gateway_quota_usage{namespace="default",quota="qg",type="total"} 42.0' in text
     assert 'gateway_quota_limit{namespace="default",quota="qg",type="total"} 500.0' in text
+
+
+def test_passive_outlier_ejection():
+    """3 consecutive 5xx eject a backend for 30 s; traffic shifts to the
+    remaining backend; ejection expires (reference BackendTrafficPolicy
+    semantics for the no-Envoy mode)."""
+    from arks_amd.gateway.app import BackendResolver, OutlierDetector
+
+    now = [1000.0]
+    det = OutlierDetector(threshold=3, ejection_s=30.0, clock=lambda: now[0])
+    store = Store()
+    store.apply({
+        "apiVersion": "gateway.networking.k8s.io/v1",
+        "kind": "HTTPRoute",
+        "metadata": {"name": "m", "namespace": "default"},
+        "spec": {"rules": [{"backendRefs": [
+            {"name": "a", "weight": 1}, {"name": "b", "weight": 1},
+        ]}]},
+    })
+    r = BackendResolver(store, url_for_service=lambda ns, svc: svc,
+                        outliers=det)
+    det.record("a", False)
+    det.record("a", False)
+    assert not det.is_ejected("a")  # below threshold
+    det.record("a", True)           # success resets the streak
+    det.record("a", False)
+    det.record("a", False)
+    det.record("a", False)
+    assert det.is_ejected("a")
+    picks = {r.resolve("default", "m") for _ in range(20)}
+    assert picks == {"b"}
+    now[0] += 31  # ejection expires
+    assert not det.is_ejected("a")
+    picks = {r.resolve("default", "m") for _ in range(50)}
+    assert picks == {"a", "b"}
+    # all ejected -> fall back to the full set (never 0 backends)
+    det.record("a", False); det.record("a", False); det.record("a", False)
+    det.record("b", False); det.record("b", False); det.record("b", False)
+    assert r.resolve("default", "m") in {"a", "b"}
