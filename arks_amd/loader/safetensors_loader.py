@@ -108,8 +108,10 @@ def load_model_weights(model, model_path: str, device: torch.device,
 
     def reader():
         try:
-            # one flat (file, tensor-meta) list in offset order
-            pending: list = []
+            # fused-partner tensors straddling a file/buffer boundary are
+            # CLONED out of the recycled pinned buffer and re-joined with
+            # the next chunk (their group applies only when complete)
+            carry: dict[str, torch.Tensor] = {}
             for path in files:
                 data_start, ts = _parse_header(path)
                 with open(path, "rb", buffering=0) as f:
@@ -142,12 +144,16 @@ def load_model_weights(model, model_path: str, device: torch.device,
                         f.seek(data_start + span0)
                         got = f.readinto(mv)
                         assert got == span1 - span0, (got, span1 - span0)
-                        chunk: dict[str, torch.Tensor] = {}
+                        chunk: dict[str, torch.Tensor] = dict(carry)
+                        carry = {}
                         for name, (dt, shape, o0, o1) in names.items():
                             sl = buf[o0 - span0: o1 - span0]
                             chunk[name] = sl.view(dt).view(shape)
+                        for name in _dangling_names(chunk):
+                            carry[name] = chunk.pop(name).clone()
                         q.put((chunk, buf_id))
                         i = j
+            assert not carry, f"unpaired fused weights: {sorted(carry)}"
             q.put(None)
         except BaseException as exc:  # surface disk errors on the consumer
             q.put(exc)
@@ -168,6 +174,28 @@ def load_model_weights(model, model_path: str, device: torch.device,
         free_q.put((buf_id, ev))
     torch.cuda.current_stream(device).wait_stream(side)
     torch.cuda.synchronize(device)
+
+
+def _dangling_names(chunk) -> set:
+    """Names whose fused-weight partner group is incomplete in `chunk`
+    (q/k/v and gate/up must be applied together — a group split across a
+    file or staging-buffer boundary must carry over, or its weights would
+    silently never load)."""
+    names = set(chunk)
+    out = set()
+    for n in names:
+        if "q_proj" in n or "k_proj" in n or "v_proj" in n:
+            stem = n.rsplit(".", 2)[0]
+            kind = n.rsplit(".", 1)[1]
+            if any(f"{stem}.{p}.{kind}" not in names
+                   for p in ("q_proj", "k_proj", "v_proj")):
+                out.add(n)
+        if "gate_proj" in n or "up_proj" in n:
+            stem = n.rsplit(".", 2)[0]
+            if any(f"{stem}.{p}.weight" not in names
+                   for p in ("gate_proj", "up_proj")):
+                out.add(n)
+    return out
 
 
 def _chunk_complete(chunk: dict[str, torch.Tensor]) -> bool:

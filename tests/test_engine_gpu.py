@@ -311,3 +311,51 @@ def test_sliding_window_fp8_kv_gpu():
         assert o1 == o2 and len(o1[0]) == 60
     finally:
         C.PRESET_CONFIGS.pop("tiny-gpu-swa8", None)
+
+
+def test_sharded_checkpoint_split_fused_partners_gpu():
+    """Multi-file checkpoints can split a q/k/v (or gate/up) group across
+    file boundaries (HF shards by size); the streaming loader must carry
+    the dangling partners over — logits must equal the single-file load."""
+    import json
+    import os
+    import tempfile
+
+    from safetensors.torch import save_file
+
+    from arks_amd.config import PRESET_CONFIGS
+    from arks_amd.loader.safetensors_loader import save_random_checkpoint
+
+    cfg = PRESET_CONFIGS["tiny-gpu"]
+    with tempfile.TemporaryDirectory() as d1, \
+            tempfile.TemporaryDirectory() as d2:
+        save_random_checkpoint(cfg, d1, seed=5)
+        # re-shard: split so the cut lands INSIDE layer 0's q/k/v group
+        from safetensors import safe_open
+
+        f = [x for x in os.listdir(d1) if x.endswith(".safetensors")][0]
+        with safe_open(os.path.join(d1, f), framework="pt") as sf:
+            tensors = {k: sf.get_tensor(k) for k in sf.keys()}
+        names = sorted(tensors)
+        qn = next(n for n in names if "layers.0.self_attn.q_proj.weight" in n)
+        part1_names = [qn]  # file 1 = ONLY layer-0 q_proj
+        part2_names = [n for n in names if n != qn]
+        # file order matters: 0-prefix sorts first
+        save_file({n: tensors[n] for n in part1_names},
+                  os.path.join(d2, "0split.safetensors"))
+        save_file({n: tensors[n] for n in part2_names},
+                  os.path.join(d2, "1rest.safetensors"))
+        with open(os.path.join(d1, "config.json")) as cf:
+            cj = cf.read()
+        with open(os.path.join(d2, "config.json"), "w") as cf:
+            cf.write(cj)
+
+        def logits_of(path):
+            e = LLMEngine(EngineConfig(
+                model_path=path, device="cuda", kv_cache_blocks=64,
+                max_model_len=128, enforce_eager=True,
+            ))
+            return e.generate([[5, 9, 2, 7]],
+                              SamplingParams(max_tokens=4, ignore_eos=True))
+
+        assert logits_of(d2) == logits_of(d1)
