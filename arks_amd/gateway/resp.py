@@ -115,7 +115,11 @@ class RespClient:
                             out.append(e)
                     return out
                 except (OSError, ConnectionError):
-                    # broken connection: drop it and retry once
+                    # Broken connection: drop it and retry once. Note the
+                    # at-least-once caveat shared with go-redis retries: a
+                    # send that died after the server applied an INCRBY is
+                    # re-applied — for rate limiting that over-counts
+                    # (conservative direction), never under-counts.
                     try:
                         if self._sock is not None:
                             self._sock.close()

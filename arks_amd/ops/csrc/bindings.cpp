@@ -570,7 +570,8 @@ void moe_topk(torch::Tensor weights, torch::Tensor ids,
   TORCH_CHECK(weights.scalar_type() == torch::kFloat32 &&
               ids.scalar_type() == torch::kInt32);
   const int T = logits.size(0), E = logits.size(1);
-  TORCH_CHECK(k >= 1 && k <= 16 && E <= 1024, "k<=16, E<=1024");
+  TORCH_CHECK(k >= 1 && k <= 16 && E <= 1024 && k <= E,
+              "need 1 <= k <= min(16, E), E <= 1024");
   arks_moe_topk(weights.data_ptr(), ids.data_ptr(), logits.data_ptr(), T, E,
                 (int)k, (int)renorm, current_stream());
 }
