@@ -264,7 +264,6 @@ class MoEMLP(nn.Module):
         seg = torch.stack((counts, offs)).cpu()  # ONE host sync per layer
         counts_h = seg[0].tolist()
         offs_h = seg[1].tolist()
-        x_g = x[tok_sorted]  # [T*k, H] gathered once
         El = self.local_experts
         base = self.expert_base
         local_counts_h = counts_h[base:base + El]
@@ -299,7 +298,9 @@ class MoEMLP(nn.Module):
         ar = torch.arange(cap_eff, device=x.device)[None, :]  # [1, cap]
         flat = torch.where(ar < capped, local_starts + ar,
                            torch.zeros_like(ar)).reshape(-1)
-        xp = x_g[flat].view(El, cap_eff, self.hidden)
+        # single fused gather x[tok_sorted[flat]] — materializing an
+        # intermediate x_g[T*k, H] costs a full extra round trip per layer
+        xp = x[tok_sorted[flat]].view(El, cap_eff, self.hidden)
         gu = torch.bmm(xp, self.w13)
         h = ops.silu_mul(gu.reshape(El * cap_eff, 2 * self.inter))
         torch.bmm(h.view(El, cap_eff, self.inter), self.w2,
@@ -313,7 +314,7 @@ class MoEMLP(nn.Module):
             of_base_h.append((le, acc))
             ge = base + le
             sl = slice(offs_h[ge] - counts_h[ge] + cap_eff, offs_h[ge])
-            h2 = ops.silu_mul(x_g[sl] @ self.w13[le])
+            h2 = ops.silu_mul(x[tok_sorted[sl]] @ self.w13[le])
             torch.mm(h2, self.w2[le], out=y_all[p0 + acc:p0 + acc + o])
             acc += o
 
