@@ -78,3 +78,64 @@ def test_engine_invariants_under_random_ops(actions, speculative):
         assert guard < 2000, "engine failed to terminate"
     alloc = eng.scheduler.allocator
     assert alloc.num_free == alloc.num_blocks
+
+
+@settings(max_examples=12, deadline=None,
+          suppress_health_check=[HealthCheck.too_slow])
+@given(
+    actions=st.lists(
+        st.one_of(
+            st.tuples(st.just("add"), st.integers(8, 120), st.integers(1, 24)),
+            st.just(("step",)),
+            st.integers(0, 6).map(lambda i: ("abort", i)),
+        ),
+        min_size=4, max_size=24,
+    ),
+)
+def test_sliding_window_dropping_never_changes_outputs(actions):
+    """Fuzz: under random add/step/abort sequences on an SWA model, KV
+    page dropping must not change any emitted token (dropped pages are
+    outside every query's window by construction) and the allocator's
+    free count must never go negative."""
+    import dataclasses
+
+    import arks_amd.config as C
+
+    swa = dataclasses.replace(C.PRESET_CONFIGS["tiny"], sliding_window=32)
+    C.PRESET_CONFIGS["tiny-fuzz-swa"] = swa
+    try:
+        def run(drop: bool):
+            e = LLMEngine(EngineConfig(
+                preset="tiny-fuzz-swa", device="cpu", kv_cache_blocks=256,
+                max_model_len=256, seed=7,
+            ))
+            if not drop:
+                e._drop_window_pages = lambda: None
+            outs = {}
+            rid = 0
+            live = []
+            for act in actions:
+                if act[0] == "add":
+                    _, ln, maxt = act
+                    prompt = [(i * 17 + rid) % 90 for i in range(ln)]
+                    s = e.add_request(prompt, SamplingParams(
+                        max_tokens=maxt, ignore_eos=True),
+                        request_id=f"r{rid}")
+                    live.append(s)
+                    rid += 1
+                elif act[0] == "step":
+                    e.step()
+                else:
+                    i = act[1]
+                    if i < len(live):
+                        e.abort_request(live[i].request_id)
+            while e.has_work():
+                e.step()
+            assert e.scheduler.allocator.num_free >= 0
+            for s in live:
+                outs[s.request_id] = list(s.output_token_ids)
+            return outs
+
+        assert run(True) == run(False)
+    finally:
+        C.PRESET_CONFIGS.pop("tiny-fuzz-swa", None)
