@@ -350,13 +350,26 @@ _SKINNY_WS: dict = {}
 def skinny_gemm(x, weight, bias=None):
     """C = x @ weight^T for decode-sized M (<=64) via the split-K streaming
     MFMA kernel. Beats hipBLASLt's ~19 us latency floor on the small
-    qkv/o_proj shapes (bench_skinny logs); the dispatcher keeps the big
-    streaming shapes on the tuned library algos. Deterministic split-K
-    (fixed-order combine, no atomics)."""
+    qkv/o_proj shapes and its ~46 us in-graph down-proj algo (bench_skinny /
+    bench_down logs, r2); the dispatcher keeps the wide streaming shapes
+    (gate_up, lm_head: N >> K) on the tuned library algos. Deterministic
+    split-K (fixed-order combine, no atomics)."""
     M, K = x.shape
     N = weight.shape[0]
     out = torch.empty((M, N), dtype=x.dtype, device=x.device)
     ntiles = N // 64
+    if K > N and ntiles < 128:
+        # tall-K (down-proj shape): few N tiles, parallelism comes from
+        # K-splits; the deeper-staged AB=3 variant at nsplits ~= 448/ntiles
+        # measured fastest (34.6 us vs hipBLASLt's 46.4 in-graph, r2)
+        nsplits = min(-(-448 // ntiles), -(-K // 256))
+        k_per_split = -(-(-(-K // nsplits)) // 32) * 32
+        nsplits = -(-K // k_per_split)
+        part = (_skinny_ws(nsplits, N, 64, x.device) if nsplits > 1
+                else x.new_empty(0, dtype=torch.float32))
+        _native().skinny_gemm_v(out, part, x, weight, bias, k_per_split,
+                                nsplits, 7, False)
+        return out
     if ntiles >= 384:
         nsplits = 1
     else:
@@ -388,6 +401,10 @@ _USE_SKINNY = os.environ.get("ARKS_SKINNY_GEMM", "1") == "1"
 # bench_skinny v3); the big streaming shapes (gate_up/down/lm_head) stay on
 # the tuned hipBLASLt algos which reach 5.4-6.4 TB/s there.
 _SKINNY_MAX_ELEMS = 34_000_000
+# tall-K decode shapes (down-proj: K > N) stream W near-HBM-rate through the
+# split-K kernel and beat the library's in-graph algo (r2); wide-N shapes
+# (gate_up/lm_head) stay on hipBLASLt which reaches ~6 TB/s there.
+_SKINNY_TALLK_MAX_ELEMS = 80_000_000
 
 
 def linear_bf16(x, weight, bias=None):
@@ -401,7 +418,9 @@ def linear_bf16(x, weight, bias=None):
         and x.dtype == torch.bfloat16
         and weight.shape[0] % 64 == 0
         and weight.shape[1] % 32 == 0
-        and weight.numel() <= _SKINNY_MAX_ELEMS
+        and (weight.numel() <= _SKINNY_MAX_ELEMS
+             or (weight.shape[1] > weight.shape[0]
+                 and weight.numel() <= _SKINNY_TALLK_MAX_ELEMS))
         and weight.is_contiguous()
         and native_available()
     ):

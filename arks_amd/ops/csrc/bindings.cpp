@@ -83,6 +83,10 @@ void arks_skinny_gemm(void* part, void* out, const void* a, const void* w,
                       const void* bias, int m_rows, int n_total, int k_total,
                       int k_per_split, int nsplits, int64_t a_stride,
                       hipStream_t stream);
+void arks_skinny_gemm_v(void* part, void* out, const void* a, const void* w,
+                        const void* bias, int m_rows, int n_total, int k_total,
+                        int k_per_split, int nsplits, int64_t a_stride,
+                        int variant, bool fuse_silu, hipStream_t stream);
 void arks_greedy_sample(void* out, const void* logits, int rows, int vocab,
                         hipStream_t stream);
 void arks_gumbel_sample(void* out, const void* logits, const void* temperatures,
@@ -406,6 +410,38 @@ void skinny_gemm(torch::Tensor out, torch::Tensor part, torch::Tensor a,
                    current_stream());
 }
 
+// Variant/fused entry (MT fixed at 4, i.e. M <= 64). fuse_silu: `a` is the
+// gate_up output [M, >=2K] (gate cols [0,K), up cols [K,2K)).
+void skinny_gemm_v(torch::Tensor out, torch::Tensor part, torch::Tensor a,
+                   torch::Tensor w, c10::optional<torch::Tensor> bias,
+                   int64_t k_per_split, int64_t nsplits, int64_t variant,
+                   bool fuse_silu) {
+  check_bf16_contig(out, "out");
+  check_bf16_rowstrided(a, "a");
+  check_bf16_contig(w, "w");
+  const int m = a.size(0), k = w.size(1), n = w.size(0);
+  TORCH_CHECK(m >= 1 && m <= 64, "skinny_gemm_v needs 1 <= M <= 64");
+  TORCH_CHECK(n % 64 == 0 && k % 32 == 0, "N%64==0 and K%32==0 required");
+  TORCH_CHECK(a.size(1) >= (fuse_silu ? 2 * k : k), "A too narrow");
+  TORCH_CHECK(out.size(0) == m && out.size(1) == n);
+  TORCH_CHECK(0 <= variant && variant <= 7, "variant in 0..7");
+  const void* bias_ptr = nullptr;
+  if (bias.has_value()) {
+    check_bf16_contig(*bias, "bias");
+    TORCH_CHECK(bias->numel() == n);
+    bias_ptr = bias->data_ptr();
+  }
+  if (nsplits > 1) {
+    TORCH_CHECK(part.scalar_type() == torch::kFloat32 &&
+                part.numel() >= (int64_t)nsplits * n * 64,
+                "skinny_gemm_v workspace too small");
+  }
+  arks_skinny_gemm_v(nsplits > 1 ? part.data_ptr() : nullptr, out.data_ptr(),
+                     a.data_ptr(), w.data_ptr(), bias_ptr, m, n, k,
+                     (int)k_per_split, (int)nsplits, a.stride(0),
+                     (int)variant, fuse_silu, current_stream());
+}
+
 void rmsnorm_fp8(torch::Tensor out, torch::Tensor inv_scale,
                  torch::Tensor input, c10::optional<torch::Tensor> residual,
                  torch::Tensor weight, double eps) {
@@ -632,6 +668,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("rmsnorm_fp8", &rmsnorm_fp8);
   m.def("silu_mul_fp8", &silu_mul_fp8);
   m.def("skinny_gemm", &skinny_gemm);
+  m.def("skinny_gemm_v", &skinny_gemm_v);
   m.def("greedy_sample", &greedy_sample);
   m.def("gumbel_sample", &gumbel_sample);
   m.def("ipc_alloc", &ipc_alloc);
