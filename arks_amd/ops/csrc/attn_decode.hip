@@ -144,12 +144,19 @@ __global__ __launch_bounds__(256) void attn_decode_kernel(
   // Sliding window: tokens below L-window are masked; their pages may
   // have been dropped by the allocator (block-table entry -1), so page ids
   // are sanitized to 0 (any read from them is masked out).
-  auto page_of = [&](int pb) { int p = bt[pb]; return p < 0 ? 0 : p; };
-  auto load_k = [&](int pb, ushort8 (*dst)[STEPS]) {
-    const int64_t pbase0 = (int64_t)page_of(pb) * page_elems + head_off;
+  //
+  // Block-table ids are software-pipelined TWO iterations ahead: the K/V
+  // addresses depend on bt[pb], so loading it in the consuming iteration
+  // emits global_load_dword -> s_waitcnt vmcnt(0) before every tile — a
+  // dependent pointer-chase that drains all in-flight K/V prefetches per
+  // page (ISA audit r2, same class of stall as the skinny-GEMM staging).
+  auto page_base = [&](int id) {
+    return (int64_t)(id < 0 ? 0 : id) * page_elems + head_off;
+  };
+  auto raw_bt = [&](int pb_) { return pb_ < pb_hi ? bt[pb_] : 0; };
+  auto load_k = [&](int pb, ushort8 (*dst)[STEPS], int64_t pbase0,
+                    int64_t pbase1) {
     const bool have_p1 = pb + 1 < pb_hi;
-    const int64_t pbase1 =
-        have_p1 ? (int64_t)page_of(pb + 1) * page_elems + head_off : pbase0;
 #pragma unroll
     for (int sub = 0; sub < 2; ++sub) {
       const int64_t pbase = sub == 0 ? pbase0 : pbase1;
@@ -183,13 +190,20 @@ __global__ __launch_bounds__(256) void attn_decode_kernel(
       pb_first = pb_lo + ((win_lo_page - pb_lo) / 2) * 2;
   }
   ushort8 kreg[2][STEPS];
-  if (pb_first + wave * 2 < pb_hi) load_k(pb_first + wave * 2, kreg);
+  const int pb_w0 = pb_first + wave * 2;
+  // bt pipeline: cbt = this iteration's pages, nbt = next iteration's
+  // (already in flight); nnbt issues at each loop top for the one after.
+  int cbt0 = raw_bt(pb_w0), cbt1 = raw_bt(pb_w0 + 1);
+  int nbt0 = raw_bt(pb_w0 + NUM_WAVES * 2);
+  int nbt1 = raw_bt(pb_w0 + NUM_WAVES * 2 + 1);
+  if (pb_w0 < pb_hi)
+    load_k(pb_w0, kreg, page_base(cbt0), page_base(cbt1));
 
-  for (int pb = pb_first + wave * 2; pb < pb_hi; pb += NUM_WAVES * 2) {
-    const int64_t pbase0 = (int64_t)page_of(pb) * page_elems + head_off;
-    const bool have_p1 = pb + 1 < pb_hi;
-    const int64_t pbase1 =
-        have_p1 ? (int64_t)page_of(pb + 1) * page_elems + head_off : pbase0;
+  for (int pb = pb_w0; pb < pb_hi; pb += NUM_WAVES * 2) {
+    const int pnn = pb + 2 * NUM_WAVES * 2;
+    const int nnbt0 = raw_bt(pnn), nnbt1 = raw_bt(pnn + 1);
+    const int64_t pbase0 = page_base(cbt0);
+    const int64_t pbase1 = page_base(cbt1);
     const int tok0 = pb * KV_BLOCK_SIZE;  // first token of the chunk
     // valid tokens here: bounded by L AND by the partition's page range
     const int kmax =
@@ -224,7 +238,8 @@ __global__ __launch_bounds__(256) void attn_decode_kernel(
     // ---- prefetch next chunk's K while this chunk computes.
     ushort8 knext[2][STEPS];
     const int pnext = pb + NUM_WAVES * 2;
-    if (pnext < pb_hi) load_k(pnext, knext);
+    if (pnext < pb_hi)
+      load_k(pnext, knext, page_base(nbt0), page_base(nbt1));
 
     // ---- QK^T: A-frag = K registers, B-frag = Q registers.
     f32x4 sc[2];
@@ -335,6 +350,10 @@ __global__ __launch_bounds__(256) void attn_decode_kernel(
     for (int sub = 0; sub < 2; ++sub)
 #pragma unroll
       for (int st = 0; st < STEPS; ++st) kreg[sub][st] = knext[sub][st];
+    cbt0 = nbt0;
+    cbt1 = nbt1;
+    nbt0 = nnbt0;
+    nbt1 = nnbt1;
   }
 
   // ---- merge the 4 wave partials (flash-style) and emit.
