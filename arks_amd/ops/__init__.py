@@ -368,7 +368,7 @@ def skinny_gemm(x, weight, bias=None):
         part = (_skinny_ws(nsplits, N, 64, x.device) if nsplits > 1
                 else x.new_empty(0, dtype=torch.float32))
         _native().skinny_gemm_v(out, part, x, weight, bias, k_per_split,
-                                nsplits, 7, False)
+                                nsplits, _SKINNY_TALLK_VARIANT, False)
         return out
     if ntiles >= 384:
         nsplits = 1
@@ -405,6 +405,8 @@ _SKINNY_MAX_ELEMS = 34_000_000
 # split-K kernel and beat the library's in-graph algo (r2); wide-N shapes
 # (gate_up/lm_head) stay on hipBLASLt which reaches ~6 TB/s there.
 _SKINNY_TALLK_MAX_ELEMS = 80_000_000
+# 7 = AB=3 deep-staged; 8 = same + non-temporal W loads
+_SKINNY_TALLK_VARIANT = int(os.environ.get("ARKS_SKINNY_TALLK_V", "7"))
 
 
 def linear_bf16(x, weight, bias=None):

@@ -424,7 +424,7 @@ void skinny_gemm_v(torch::Tensor out, torch::Tensor part, torch::Tensor a,
   TORCH_CHECK(n % 64 == 0 && k % 32 == 0, "N%64==0 and K%32==0 required");
   TORCH_CHECK(a.size(1) >= (fuse_silu ? 2 * k : k), "A too narrow");
   TORCH_CHECK(out.size(0) == m && out.size(1) == n);
-  TORCH_CHECK(0 <= variant && variant <= 7, "variant in 0..7");
+  TORCH_CHECK(0 <= variant && variant <= 8, "variant in 0..8");
   const void* bias_ptr = nullptr;
   if (bias.has_value()) {
     check_bf16_contig(*bias, "bias");

@@ -49,7 +49,9 @@ constexpr int SK_AP = 8;   // a_lds row pad (bf16): rows stay 16B-aligned
 // KC = K elements per staged chunk (ping-pong pair in LDS).
 // PF = W register-prefetch distance in chunks (in-flight W = PF*KC*2 B/row).
 // FUSE_SILU: see header comment.
-template <int MT, int KC, int PF, bool FUSE_SILU, int AB = 2>
+// NT: non-temporal W loads (W is streamed exactly once; keeps it out of
+// the LLC so KV/activation lines survive across decode steps).
+template <int MT, int KC, int PF, bool FUSE_SILU, int AB = 2, bool NT = false>
 __global__ __launch_bounds__(256) void skinny_gemm_kernel(
     float* __restrict__ part,   // [nsplits, N, 16*MT] (nsplits > 1)
     bf16* __restrict__ out,     // [M, N] (nsplits == 1)
@@ -139,11 +141,16 @@ __global__ __launch_bounds__(256) void skinny_gemm_kernel(
     }
     const int kw = min(KC, ke - kc);
 #pragma unroll
-    for (int s = 0; s < KSTEPS; ++s)
-      dst[s] = (s * 32 < kw)
-                   ? *reinterpret_cast<const ushort8*>(wrow + kc + s * 32 +
-                                                       8 * la)
-                   : ushort8{};
+    for (int s = 0; s < KSTEPS; ++s) {
+      const ushort8* src =
+          reinterpret_cast<const ushort8*>(wrow + kc + s * 32 + 8 * la);
+      if (s * 32 >= kw)
+        dst[s] = ushort8{};
+      else if constexpr (NT)
+        dst[s] = __builtin_nontemporal_load(src);
+      else
+        dst[s] = *src;
+    }
   };
 
   // A staging issues BEFORE the W prefetch: vmcnt is an ISSUE-ORDERED
@@ -562,6 +569,13 @@ extern "C" void arks_skinny_gemm_v(void* part, void* out, const void* a,
                          stream, (float*)part, (bf16*)out, (const bf16*)a,    \
                          (const bf16*)w, (const bf16*)bias, m_rows, n_total,  \
                          k_total, k_per_split, nsplits, a_stride);            \
+      break;                                                                  \
+    case 8:                                                                   \
+      hipLaunchKernelGGL((skinny_gemm_kernel<4, 128, 2, FS, 3, true>), grid,  \
+                         block, 0, stream, (float*)part, (bf16*)out,          \
+                         (const bf16*)a, (const bf16*)w, (const bf16*)bias,   \
+                         m_rows, n_total, k_total, k_per_split, nsplits,      \
+                         a_stride);                                           \
       break;                                                                  \
     case 7:                                                                   \
       hipLaunchKernelGGL((skinny_gemm_kernel<4, 128, 2, FS, 3>), grid,        \
