@@ -358,10 +358,13 @@ def skinny_gemm(x, weight, bias=None):
     N = weight.shape[0]
     out = torch.empty((M, N), dtype=x.dtype, device=x.device)
     ntiles = N // 64
-    if K > N and ntiles < 128:
+    if K > N and ntiles < 128 and M > 48:
         # tall-K (down-proj shape): few N tiles, parallelism comes from
         # K-splits; the deeper-staged AB=3 variant at nsplits ~= 448/ntiles
-        # measured fastest (34.6 us vs hipBLASLt's 46.4 in-graph, r2)
+        # measured fastest (34.6 us vs hipBLASLt's 46.4 in-graph, r2).
+        # M-gated: the variant kernel is fixed at MT=4 (full 64-row cost)
+        # while hipBLASLt switches to a faster algo below M~48 — routing
+        # small-M down-proj here regressed b16x8k decode by 4% (r2)
         nsplits = min(-(-448 // ntiles), -(-K // 256))
         k_per_split = -(-(-(-K // nsplits)) // 32) * 32
         nsplits = -(-K // k_per_split)
@@ -422,6 +425,7 @@ def linear_bf16(x, weight, bias=None):
         and weight.shape[1] % 32 == 0
         and (weight.numel() <= _SKINNY_MAX_ELEMS
              or (weight.shape[1] > weight.shape[0]
+                 and x.shape[0] > 48
                  and weight.numel() <= _SKINNY_TALLK_MAX_ELEMS))
         and weight.is_contiguous()
         and native_available()
