@@ -7,13 +7,15 @@
 //
 //   grid = (N/64, nsplits): a workgroup owns a 64-row N-tile and a K-range.
 //   W is read exactly once, 16 B/lane (guide G13), staged PF chunks ahead
-//   in REGISTERS so (PF+1)*KC/32 16-byte loads per lane are in flight while
-//   the current chunk's MFMAs run — the decode shapes are HBM-latency bound
-//   at PF=1 (4 KB/wave in flight -> ~3 TB/s; PF>=2 is what reaches the
-//   streaming rate). A (the activations, tiny, L2/LLC-hot) is staged
-//   through ping-pong LDS buffers so the staging barrier is never on the
-//   critical path (PMC v1: 67% SQ_WAIT_ANY from the single-buffer
-//   round-trip — gpurun_out/pmc).
+//   in registers; A (the activations, tiny, L2/LLC-hot) is staged through
+//   a ring of AB LDS buffers. What actually governs throughput here (r2
+//   ISA audit, profiles/r02_final.md): the staging loops must be BATCHED
+//   and BRANCHLESS — with per-item guards the compiler emits
+//   global_load -> s_waitcnt vmcnt(0) -> ds_write per item, and because
+//   vmcnt is an issue-ordered counter each of those waits also drains
+//   every in-flight W prefetch, pinning the stream at ~3 TB/s no matter
+//   how deep PF is. With batched staging, PF=1/AB=3 reaches ~3.9 TB/s on
+//   the down-proj shape (hipBLASLt in-graph: 2.9).
 //   Swapped operands (A-frag = W rows, B-frag = activations) put the C
 //   fragment at [n, m], n = 4*la+r, col = lane%16 (same trick as the
 //   attention kernels, guide common-mistake #6).
@@ -233,12 +235,10 @@ __global__ __launch_bounds__(256) void skinny_gemm_kernel(
   }
 }
 
-// Wave-private variant: each wave stages its OWN copy of the A chunk into
-// its own LDS slice, so there are no cross-wave barriers at all — the only
-// ordering is the wave's own vmcnt/lgkmcnt, exactly like the raw streaming
-// probe (scripts/bench_membw.hip) that reaches 6.7 TB/s on this shape.
-// Costs 4x the A reads (all L2/LLC hits) and 4x the LDS footprint per
-// chunk, so the chunk is smaller (KC2 = 64).
+// Wave-private variant — MEASURED LOSER, kept for the record (bench_down
+// r2: 1.1 TB/s): each wave stages its OWN copy of the A chunk into its own
+// LDS slice so there are no cross-wave barriers, but the 4x A reads and 4x
+// staging instructions per W byte cost far more than the barriers did.
 template <int MT, bool FUSE_SILU>
 __global__ __launch_bounds__(256) void skinny_gemm_wave_kernel(
     float* __restrict__ part, bf16* __restrict__ out,
@@ -374,13 +374,12 @@ __global__ __launch_bounds__(256) void skinny_gemm_wave_kernel(
   }
 }
 
-// Direct (barrier-free) variant: both W and A stream straight from global —
-// A is tiny and L2/LLC-hot, and each lane's A fragment IS its MFMA operand
-// (b-frag lane index = the A row it reads), so no LDS staging or barriers
-// are needed. The inner loop is then a pure load+MFMA stream the compiler
-// pipelines like a memcpy: the raw-read probe (scripts/bench_membw.hip)
-// shows this pattern reaches 6.7 TB/s on the down-proj shape where the
-// LDS-staged kernel's barrier cadence caps at ~3 TB/s.
+// Direct (barrier-free) variant — MEASURED LOSER, kept for the record
+// (bench_down r2: 1.5 TB/s vs 3.9 staged): both W and A stream straight
+// from global. The dependent A loads share the vmcnt counter with W and
+// quadruple the per-wave load count, swamping the issue stream; the
+// raw-read probe (scripts/bench_membw.hip) only shows 6.7 TB/s for the
+// W pattern when it is the ONLY load stream.
 // Rows >= m_rows are clamped to the last valid row (duplicate finite reads;
 // their C columns are discarded by the epilogue/partial consumer).
 template <int MT, bool FUSE_SILU>
