@@ -20,6 +20,9 @@ def main(argv=None):
                     help="API server base URL (default: in-cluster)")
     ap.add_argument("--resync", type=float, default=5.0)
     ap.add_argument("--health-port", type=int, default=8081)
+    ap.add_argument("--metrics-port", type=int, default=8443,
+                    help="Prometheus metrics port (reference manager serves "
+                         "controller metrics on :8443; 0 disables)")
     ap.add_argument("--leader-elect", action="store_true",
                     help="gate reconciling on a coordination.k8s.io Lease "
                          "(reference cmd/main.go:198-216)")
@@ -67,6 +70,14 @@ def main(argv=None):
 
     health = ThreadingHTTPServer(("0.0.0.0", args.health_port), Probe)
     threading.Thread(target=health.serve_forever, daemon=True).start()
+
+    metrics_srv = None
+    if args.metrics_port:
+        from .metrics import make_metrics_server
+
+        metrics_srv = make_metrics_server(args.metrics_port)
+        threading.Thread(target=metrics_srv.serve_forever,
+                         daemon=True).start()
     # watch streams (informer path); a slow resync loop remains as the
     # safety net against missed events
     store.run_watch()
@@ -84,6 +95,8 @@ def main(argv=None):
         store.stop()
         op.stop()
         health.shutdown()
+        if metrics_srv is not None:
+            metrics_srv.shutdown()
 
 
 if __name__ == "__main__":

@@ -9,6 +9,7 @@ from __future__ import annotations
 
 import queue
 import threading
+import time
 
 from ..crd.types import served_model_name
 from .reconcilers import (
@@ -18,6 +19,7 @@ from .reconcilers import (
     ArksModelReconciler,
 )
 from .store import Store, obj_kind
+from . import metrics as opmetrics
 
 
 class Operator:
@@ -89,7 +91,19 @@ class Operator:
         }.get(kind)
         if rec is None:
             return None
-        return rec.reconcile(ns, name)
+        opmetrics.workqueue_depth.labels("operator").set(self._queue.qsize())
+        t0 = time.perf_counter()
+        try:
+            delay = rec.reconcile(ns, name)
+        except Exception:
+            opmetrics.reconcile_errors.labels(kind).inc()
+            opmetrics.reconcile_total.labels(kind, "error").inc()
+            raise
+        opmetrics.reconcile_time.labels(kind).observe(time.perf_counter() - t0)
+        result = ("success" if delay is None
+                  else "requeue" if delay == 0 else "requeue_after")
+        opmetrics.reconcile_total.labels(kind, result).inc()
+        return delay
 
     def reconcile_until_stable(self, max_iters: int = 200) -> int:
         """Drain the queue; follow zero-delay requeues. Returns iterations."""
@@ -139,6 +153,12 @@ class Operator:
                     inflight.add(item)
                 try:
                     delay = self._dispatch(*item)
+                except Exception as exc:  # noqa: BLE001
+                    # error backoff requeue (controller-runtime retries a
+                    # failed Reconcile with rate-limited backoff)
+                    print(f"reconcile {item} failed: {exc!r}", flush=True)
+                    delay = None
+                    requeue_later(item, 5.0)
                 finally:
                     with lock:
                         inflight.discard(item)

@@ -19,6 +19,7 @@ def bundle(out, paths):
 
 
 bundle("dist/operator.yaml",
-       sorted(glob.glob("deploy/crds/*.yaml")) + ["deploy/operator.yaml"])
+       sorted(glob.glob("deploy/crds/*.yaml")) + ["deploy/operator.yaml",
+                                                  "deploy/network-policy.yaml"])
 bundle("dist/gateway.yaml", ["deploy/gateway.yaml"])
 bundle("dist/gateway-envoy.yaml", ["deploy/gateway-envoy.yaml"])
