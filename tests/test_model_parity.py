@@ -371,3 +371,48 @@ def test_loader_chunked_matches_whole(tmp_path):
     sw = dict(whole.named_parameters())
     for name, p in chunked.named_parameters():
         assert torch.equal(p, sw[name]), name
+
+
+def test_parse_header_matches_safetensors_lib(tmp_path):
+    """The direct header parser (GPU fast path, loader/safetensors_loader.py
+    _parse_header) agrees with the official safetensors library on names,
+    dtypes, shapes, and byte ranges — including a __metadata__ entry and
+    0-d / empty tensors."""
+    import torch
+    from safetensors import safe_open
+    from safetensors.torch import save_file
+
+    from arks_amd.loader.safetensors_loader import _parse_header
+
+    tensors = {
+        "a.weight": torch.randn(3, 5, dtype=torch.bfloat16),
+        "b.bias": torch.randn(7, dtype=torch.float32),
+        "c.scalar": torch.tensor(2.5, dtype=torch.float16),
+        "d.empty": torch.empty(0, 4, dtype=torch.bfloat16),
+        "e.int": torch.arange(6, dtype=torch.int32).reshape(2, 3),
+    }
+    path = str(tmp_path / "model.safetensors")
+    save_file(tensors, path, metadata={"format": "pt"})
+
+    data_start, ts = _parse_header(path)
+    assert sorted(n for n, *_ in ts) == sorted(tensors)
+    by_name = {n: (dt, shape, o0, o1) for n, dt, shape, o0, o1 in ts}
+    with safe_open(path, framework="pt") as f:
+        for name, ref in tensors.items():
+            dt, shape, o0, o1 = by_name[name]
+            assert list(shape) == list(ref.shape)
+            assert dt == ref.dtype
+            assert o1 - o0 == ref.numel() * ref.element_size()
+            if ref.numel() == 0:
+                continue  # frombuffer rejects empty buffers
+            # byte range reproduces the tensor exactly
+            with open(path, "rb") as raw:
+                raw.seek(data_start + o0)
+                buf = raw.read(o1 - o0)
+            got = torch.frombuffer(bytearray(buf), dtype=ref.dtype)
+            assert torch.equal(got.reshape(ref.shape), f.get_tensor(name))
+    # offsets are sorted and non-overlapping (the chunked reader relies
+    # on this to stream spans sequentially)
+    offs = [(o0, o1) for *_, o0, o1 in ts]
+    assert offs == sorted(offs) and all(
+        offs[i][1] <= offs[i + 1][0] for i in range(len(offs) - 1))
