@@ -104,3 +104,47 @@ def test_redis_limiter_token_rules(redis):
     assert rl.check_limit(d)[0]  # token rules check at 0 increment
     rl.do_limit(d, 1)
     assert not rl.check_limit(d)[0]
+
+
+def test_resp_parser_fragmented_and_nested():
+    """Protocol-level: replies fragmented byte-by-byte across packets,
+    nested arrays, null bulks/arrays, and in-pipeline errors all parse
+    (the client buffers partial reads — resp.py _read_line/_read_exact)."""
+    import socket
+    import threading
+
+    # crafted reply stream for one pipeline of 6 commands
+    payload = (b"+OK\r\n"
+               b":42\r\n"
+               b"$-1\r\n"
+               b"*-1\r\n"
+               b"-ERR boom\r\n"
+               b"*3\r\n$3\r\nfoo\r\n:7\r\n*2\r\n+a\r\n$0\r\n\r\n")
+
+    srv = socket.socket()
+    srv.bind(("127.0.0.1", 0))
+    srv.listen(1)
+
+    def serve():
+        conn, _ = srv.accept()
+        conn.recv(65536)  # the pipelined commands (content irrelevant)
+        for i in range(len(payload)):  # dribble one byte per send
+            conn.sendall(payload[i:i + 1])
+        conn.close()
+
+    t = threading.Thread(target=serve, daemon=True)
+    t.start()
+    from arks_amd.gateway.resp import RespClient, RespError
+
+    c = RespClient("127.0.0.1", srv.getsockname()[1], timeout=5.0)
+    out = c.pipeline([("PING",), ("X",), ("GET", "m"), ("LRANGE", "k"),
+                      ("BAD",), ("NEST",)])
+    assert out[0] == "OK"
+    assert out[1] == 42
+    assert out[2] is None
+    assert out[3] is None
+    assert isinstance(out[4], RespError) and "boom" in str(out[4])
+    assert out[5] == [b"foo", 7, ["a", b""]]
+    c.close()
+    srv.close()
+    t.join(timeout=5)
