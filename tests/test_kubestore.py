@@ -217,3 +217,40 @@ def test_watch_seed_emits_deletes_for_vanished_objects():
     fake.objects.clear()
     store._seed_kind("ArksModel")
     assert "DELETED" in events
+
+
+def test_watch_error_event_triggers_relist_and_recovers():
+    """A watch ERROR event (the wire form of 410 Gone: compaction dropped
+    our resourceVersion) makes _watch_kind_once return None; the caller
+    relists via _seed_kind and the store converges on current state
+    (reference informer semantics)."""
+    fake, store = mk_store()
+    events = []
+    store.subscribe(lambda e, o: events.append((e, getattr(
+        o, "metadata", None) and o.metadata.name or o["metadata"]["name"])))
+
+    m = ArksModel(
+        metadata=ObjectMeta(name="m1", namespace="default"),
+        spec=ArksModelSpec(model="org/m"),
+    )
+    store.create(m)
+    rv = store._seed_kind("ArksModel")
+    events.clear()
+
+    # inject an ERROR watch event at the head of the stream (410 Gone)
+    fake.rv += 1
+    fake._record("ERROR", "arksmodels", {
+        "kind": "Status", "code": 410, "reason": "Gone",
+        "metadata": {}})
+    assert store._watch_kind_once("ArksModel", rv, timeout_s=1) is None
+
+    # while we were "disconnected", another client created m2
+    m2 = ArksModel(
+        metadata=ObjectMeta(name="m2", namespace="default"),
+        spec=ArksModelSpec(model="org/m2"),
+    )
+    store.create(m2)
+    events.clear()
+    rv2 = store._seed_kind("ArksModel")  # the relist
+    assert rv2 is not None
+    assert ("ADDED", "m2") in events
