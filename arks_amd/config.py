@@ -40,6 +40,10 @@ class ModelConfig:
     max_window_layers: int | None = None
     layer_types: list | None = None
     qk_norm: bool = False  # Qwen3: per-head RMSNorm on q/k before RoPE
+    # SmolLM3 NoPE: no_rope_layers[i] == 1 -> layer i USES RoPE, 0 -> the
+    # layer attends without positional encoding (HF configuration_smollm3
+    # semantics; default = one NoPE layer every no_rope_layer_interval)
+    no_rope_layers: list | None = None
     tie_word_embeddings: bool = False
     attention_bias: bool = True  # qwen2 has qkv bias; llama does not
     eos_token_id: int = 151645
@@ -57,6 +61,11 @@ class ModelConfig:
             raise ValueError(
                 f"invalid sliding_window={self.sliding_window}"
             )
+
+    def layer_uses_rope(self, layer_idx: int) -> bool:
+        if self.no_rope_layers is None:
+            return True
+        return bool(self.no_rope_layers[layer_idx])
 
     def layer_window(self, layer_idx: int) -> int:
         """Effective attention window for one layer (0 = full attention).
@@ -101,6 +110,11 @@ class ModelConfig:
             max_position_embeddings=cfg.get("max_position_embeddings", 32768),
             rope_scaling=cfg.get("rope_scaling"),
             qk_norm=arch.startswith("Qwen3"),
+            no_rope_layers=(
+                cfg["no_rope_layers"] if cfg.get("no_rope_layers") is not None
+                else [int((i + 1) % cfg.get("no_rope_layer_interval", 4) != 0)
+                      for i in range(cfg.get("num_hidden_layers", 28))]
+                if arch == "SmolLM3ForCausalLM" else None),
             num_local_experts=cfg.get("num_local_experts",
                                       cfg.get("num_experts", 0)),
             # Qwen2-family configs declare a window but disable it
@@ -348,6 +362,41 @@ PRESET_CONFIGS: dict[str, ModelConfig] = {
         attention_bias=False,
         num_local_experts=4,
         num_experts_per_tok=2,
+        eos_token_id=2,
+        bos_token_id=1,
+    ),
+    "smollm3-3b": ModelConfig(
+        architecture="SmolLM3ForCausalLM",
+        vocab_size=128256,
+        hidden_size=2048,
+        intermediate_size=11008,
+        num_hidden_layers=36,
+        num_attention_heads=16,
+        num_key_value_heads=4,
+        head_dim=128,
+        rms_norm_eps=1e-6,
+        rope_theta=5000000.0,
+        max_position_embeddings=65536,
+        tie_word_embeddings=True,
+        attention_bias=False,
+        no_rope_layers=[int((i + 1) % 4 != 0) for i in range(36)],
+        eos_token_id=128012,
+        bos_token_id=128000,
+    ),
+    "tiny-smollm3": ModelConfig(  # CPU-test-sized, NoPE on odd layers
+        architecture="SmolLM3ForCausalLM",
+        vocab_size=512,
+        hidden_size=128,
+        intermediate_size=256,
+        num_hidden_layers=2,
+        num_attention_heads=4,
+        num_key_value_heads=2,
+        head_dim=32,
+        rope_theta=10000.0,
+        max_position_embeddings=2048,
+        tie_word_embeddings=True,
+        attention_bias=False,
+        no_rope_layers=[1, 0],
         eos_token_id=2,
         bos_token_id=1,
     ),

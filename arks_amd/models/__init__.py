@@ -15,6 +15,8 @@ _REGISTRY = {
     "Qwen3ForCausalLM": LlamaFamilyForCausalLM,  # + per-head q/k RMSNorm
     "Qwen3MoeForCausalLM": LlamaFamilyForCausalLM,  # qk-norm + sparse MoE
     "Qwen2MoeForCausalLM": LlamaFamilyForCausalLM,  # MoE + shared expert
+    # llama arch + NoPE every no_rope_layer_interval-th layer + tied embeds
+    "SmolLM3ForCausalLM": LlamaFamilyForCausalLM,
 }
 
 

@@ -60,6 +60,9 @@ class Attention(nn.Module):
         # semantics; Mistral = every layer; 0 = full attention). Masking
         # happens inside the kernels.
         self.window = cfg.layer_window(layer_idx)
+        # SmolLM3 NoPE layers attend without positional encoding: k/v are
+        # cached unrotated (HF modeling_smollm3.py:200-226 semantics)
+        self.use_rope = cfg.layer_uses_rope(layer_idx)
         self.qkv_proj = QKVParallelLinear(
             cfg.hidden_size, cfg.head_dim, cfg.num_attention_heads,
             cfg.num_key_value_heads, bias=cfg.attention_bias, dtype=dtype,
@@ -90,10 +93,18 @@ class Attention(nn.Module):
                 k.contiguous().view(-1, hd)).view(T, self.nkv_local * hd)
             v = v.contiguous()  # reshape_and_cache wants one shared kv stride
         k_cache, v_cache = kv_cache
-        q, k = ops.rope_and_cache(
-            batch.positions, q, k, v, k_cache, v_cache, batch.slot_mapping,
-            self._cos_sin, self.head_dim,
-        )
+        if self.use_rope:
+            q, k = ops.rope_and_cache(
+                batch.positions, q, k, v, k_cache, v_cache,
+                batch.slot_mapping, self._cos_sin, self.head_dim,
+            )
+        else:
+            # NoPE layer: cache k/v as-is, q unrotated
+            ops.reshape_and_cache(
+                k.unflatten(-1, (self.nkv_local, self.head_dim)),
+                v.unflatten(-1, (self.nkv_local, self.head_dim)),
+                k_cache, v_cache, batch.slot_mapping,
+            )
         k = k.unflatten(-1, (self.nkv_local, self.head_dim))
         v = v.unflatten(-1, (self.nkv_local, self.head_dim))
         q = q.unflatten(-1, (self.nq_local, self.head_dim))

@@ -468,3 +468,27 @@ def test_per_layer_window_rule_and_mixed_generation():
         "num_hidden_layers": 4,
     })
     assert mc.layer_window(0) == 0 and mc.layer_window(3) == 4096
+
+
+def test_smollm3_nope_engine_decode_matches_prefill():
+    """SmolLM3 NoPE layers through the full engine: decode steps (paged
+    cache written by the no-rope branch) agree with a fresh prefill of the
+    same tokens — and generation is deterministic."""
+    from arks_amd.config import EngineConfig
+
+    torch.manual_seed(7)
+    e1 = LLMEngine(EngineConfig(preset="tiny-smollm3", device="cpu",
+                                kv_cache_blocks=128, max_model_len=512))
+    prompts = [[1, 5, 9, 20, 7], [3, 3, 7]]
+    sp = SamplingParams(max_tokens=8, ignore_eos=True)
+    out1 = e1.generate(prompts, sp)
+    torch.manual_seed(7)
+    e2 = LLMEngine(EngineConfig(preset="tiny-smollm3", device="cpu",
+                                kv_cache_blocks=128, max_model_len=512))
+    out2 = e2.generate(prompts, sp)
+    assert out1 == out2 and all(len(o) == 8 for o in out1)
+    # decode-vs-prefill consistency: feeding prompt+generated as one prefill
+    # must greedily re-derive the same final token
+    full = prompts[0] + out1[0][:-1]
+    out3 = e2.generate([full], SamplingParams(max_tokens=1, ignore_eos=True))
+    assert out3[0][0] == out1[0][-1]

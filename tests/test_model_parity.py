@@ -416,3 +416,57 @@ def test_parse_header_matches_safetensors_lib(tmp_path):
     offs = [(o0, o1) for *_, o0, o1 in ts]
     assert offs == sorted(offs) and all(
         offs[i][1] <= offs[i + 1][0] for i in range(len(offs) - 1))
+
+
+def test_smollm3_nope_logits_match_transformers(tmp_path):
+    """SmolLM3: llama arch with NoPE layers (no_rope_layers[i]==0 attends
+    without positional encoding — HF modeling_smollm3.py:200-226) and tied
+    embeddings, vs transformers."""
+    import dataclasses
+
+    from arks_amd.loader.safetensors_loader import (
+        load_model_weights,
+        save_random_checkpoint,
+    )
+
+    base = PRESET_CONFIGS["tiny"]
+    cfg = dataclasses.replace(
+        base,
+        architecture="SmolLM3ForCausalLM",
+        attention_bias=False,
+        tie_word_embeddings=True,
+        no_rope_layers=[1, 0],  # layer 1 is NoPE
+    )
+    save_random_checkpoint(cfg, str(tmp_path), seed=23)
+    ours = create_model(cfg, dtype=torch.float32)
+    load_model_weights(ours, str(tmp_path), torch.device("cpu"))
+
+    hf_cfg = transformers.SmolLM3Config(
+        vocab_size=cfg.vocab_size,
+        hidden_size=cfg.hidden_size,
+        intermediate_size=cfg.intermediate_size,
+        num_hidden_layers=cfg.num_hidden_layers,
+        num_attention_heads=cfg.num_attention_heads,
+        num_key_value_heads=cfg.num_key_value_heads,
+        rms_norm_eps=cfg.rms_norm_eps,
+        rope_theta=cfg.rope_theta,
+        max_position_embeddings=cfg.max_position_embeddings,
+        tie_word_embeddings=True,
+        use_sliding_window=False,
+        no_rope_layers=[1, 0],
+        attention_dropout=0.0,
+        pad_token_id=0,
+        bos_token_id=1,
+        eos_token_id=2,
+    )
+    hf = transformers.SmolLM3ForCausalLM.from_pretrained(
+        str(tmp_path), config=hf_cfg, torch_dtype=torch.float32
+    )
+    hf.eval()
+    ids = [3, 1, 4, 1, 5, 9, 2, 6, 5, 3]
+    logits = forward_ours(ours, cfg, ids)
+    with torch.no_grad():
+        hf_logits = hf(torch.tensor([ids])).logits[0]
+    diff = (logits - hf_logits).abs().max().item()
+    assert diff < 2e-3, f"max logits diff {diff}"
+    assert torch.equal(logits.argmax(-1), hf_logits.argmax(-1))
