@@ -39,6 +39,8 @@ def parse_args(argv=None):
     p.add_argument("--speculative", choices=["ngram"], default=None,
                    help="prompt-lookup speculative decoding (greedy requests)")
     p.add_argument("--num-speculative-tokens", type=int, default=4)
+    p.add_argument("--no-prefix-cache", action="store_true",
+                   help="disable the content-addressed prefix/radix cache")
     # multi-node group flags (LWS leader/worker topology): workers join the
     # leader's torch.distributed rendezvous
     p.add_argument("--leader-address", default=None)
@@ -69,6 +71,7 @@ def build_engine_config(args):
         kv_cache_dtype=args.kv_cache_dtype,
         speculative=args.speculative,
         num_speculative_tokens=args.num_speculative_tokens,
+        enable_prefix_caching=not args.no_prefix_cache,
     )
 
 
