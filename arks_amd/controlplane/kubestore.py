@@ -172,20 +172,24 @@ class KubeStore:
         ns = meta.get("namespace", "default")
         name = meta["name"]
         path = self._path(kind, ns, name)
-        # fetch live resourceVersion (reconcilers are level-triggered and
-        # idempotent; last-write-wins is the reference's RetryOnConflict
-        # pattern collapsed into one step)
-        live = self._client.get(path)
-        if live.status_code == 404:
-            raise NotFound(f"{kind}/{ns}/{name}")
-        live.raise_for_status()
-        live_obj = live.json()
-        meta["resourceVersion"] = live_obj["metadata"]["resourceVersion"]
         status = d.pop("status", None)
-        r = self._client.put(path, json=d)
-        if r.status_code == 409:
-            raise Conflict(r.text)
-        r.raise_for_status()
+        # refresh-and-retry on 409 (the reference's RetryOnConflict:
+        # reconcilers are level-triggered and idempotent, so re-reading the
+        # live resourceVersion and re-applying our mutation is safe)
+        for attempt in range(5):
+            live = self._client.get(path)
+            if live.status_code == 404:
+                raise NotFound(f"{kind}/{ns}/{name}")
+            live.raise_for_status()
+            live_obj = live.json()
+            meta["resourceVersion"] = live_obj["metadata"]["resourceVersion"]
+            r = self._client.put(path, json=d)
+            if r.status_code == 409:
+                if attempt == 4:
+                    raise Conflict(r.text)
+                continue
+            r.raise_for_status()
+            break
         out = r.json()
         if status is not None:
             sub = dict(out)

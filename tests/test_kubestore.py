@@ -16,6 +16,7 @@ class FakeKube:
     def __init__(self):
         self.objects: dict[str, dict] = {}
         self.rv = 0
+        self.conflicts_to_inject = 0  # chaos: next N PUTs 409 regardless
         # watch support: (rv, plural, event-dict) log
         self.events: list[tuple[int, str, dict]] = []
 
@@ -76,6 +77,15 @@ class FakeKube:
             if key not in self.objects:
                 return httpx.Response(404, json={})
             body = json.loads(request.content)
+            # real API-server optimistic concurrency: a stale
+            # resourceVersion (or an injected chaos conflict) is a 409
+            sent_rv = body.get("metadata", {}).get("resourceVersion")
+            cur_rv = self.objects[key]["metadata"].get("resourceVersion")
+            if self.conflicts_to_inject > 0:
+                self.conflicts_to_inject -= 1
+                return httpx.Response(409, json={"reason": "Conflict"})
+            if sent_rv is not None and sent_rv != cur_rv:
+                return httpx.Response(409, json={"reason": "Conflict"})
             self.rv += 1
             body["metadata"]["resourceVersion"] = str(self.rv)
             if status_sub:
@@ -254,3 +264,56 @@ def test_watch_error_event_triggers_relist_and_recovers():
     rv2 = store._seed_kind("ArksModel")  # the relist
     assert rv2 is not None
     assert ("ADDED", "m2") in events
+
+
+def test_update_retries_on_conflict():
+    """Optimistic-concurrency semantics (real API server / envtest): a 409
+    between the store's refresh and PUT is retried with a fresh
+    resourceVersion (reference RetryOnConflict); persistent conflicts
+    surface as Conflict."""
+    import pytest as _pytest
+
+    from arks_amd.controlplane.kubestore import Conflict
+
+    fake, store = mk_store()
+    m = ArksModel(
+        metadata=ObjectMeta(name="c1", namespace="default"),
+        spec=ArksModelSpec(model="org/m"),
+    )
+    store.create(m)
+    got = store.get("ArksModel", "default", "c1")
+    got.status.phase = ModelPhase.READY
+    fake.conflicts_to_inject = 2  # two racing writers, then success
+    out = store.update(got)
+    assert out.status.phase == ModelPhase.READY
+    # a conflict storm (more than the retry budget) surfaces
+    got2 = store.get("ArksModel", "default", "c1")
+    fake.conflicts_to_inject = 99
+    with _pytest.raises(Conflict):
+        store.update(got2)
+    fake.conflicts_to_inject = 0
+
+
+def test_stale_resource_version_put_rejected():
+    """FakeKube enforces resourceVersion like a real API server: a raw PUT
+    carrying a stale rv is rejected with 409 (this is what forces the
+    store's refresh-and-retry path to exist)."""
+    fake, store = mk_store()
+    m = ArksModel(
+        metadata=ObjectMeta(name="c2", namespace="default"),
+        spec=ArksModelSpec(model="org/m"),
+    )
+    store.create(m)
+    key = "arksmodels/default/c2"
+    stale = dict(fake.objects[key])
+    # another writer bumps the object
+    cur = store.get("ArksModel", "default", "c2")
+    store.update(cur)
+    # raw PUT with the old rv must 409
+    import httpx as _httpx
+    req = _httpx.Request(
+        "PUT",
+        "https://fake/apis/arks.ai/v1/namespaces/default/arksmodels/c2",
+        json=stale)
+    resp = fake.handler(req)
+    assert resp.status_code == 409
