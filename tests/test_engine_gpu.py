@@ -359,3 +359,31 @@ def test_sharded_checkpoint_split_fused_partners_gpu():
                               SamplingParams(max_tokens=4, ignore_eos=True))
 
         assert logits_of(d2) == logits_of(d1)
+
+
+def test_smollm3_nope_gpu_graph_matches_eager():
+    """SmolLM3 NoPE layers on the native kernels: the no-rope cache-write
+    branch (plain reshape_and_cache on the strided qkv views) produces the
+    same outputs graphed and eager."""
+    import dataclasses
+
+    import arks_amd.config as C
+
+    torch.manual_seed(0)
+    cfg = dataclasses.replace(
+        C.PRESET_CONFIGS["tiny-gpu"],
+        architecture="SmolLM3ForCausalLM",
+        attention_bias=False,
+        tie_word_embeddings=True,
+        no_rope_layers=[1, 0],  # layer 1 NoPE
+    )
+    C.PRESET_CONFIGS["tiny-gpu-smollm3"] = cfg
+    try:
+        prompts = [[1, 5, 9, 20, 31, 7], [3, 3, 7, 90], [17] * 40]
+        sp = SamplingParams(max_tokens=12, ignore_eos=True)
+        eager = mk(True, preset="tiny-gpu-smollm3").generate(prompts, sp)
+        graphed = mk(False, preset="tiny-gpu-smollm3").generate(prompts, sp)
+        assert eager == graphed
+        assert all(len(o) == 12 for o in eager)
+    finally:
+        C.PRESET_CONFIGS.pop("tiny-gpu-smollm3", None)
