@@ -206,22 +206,33 @@ class ExtProcServicer:
         return ProcessingResponse(response_body=CommonResponse())
 
     def _scan_sse(self, chunk: bytes, state: dict) -> None:
+        # Spec-correct SSE line parser (the reference uses openai-go's
+        # ssestream): CRLF endings tolerated, multi-line `data:` fields
+        # joined with \n, comments/other fields ignored. A naive
+        # split-on-"\n\n" misses events from runtimes that emit \r\n and
+        # silently drops their usage accounting.
         buf = state["buffer"] + chunk
-        while b"\n\n" in buf:
-            event, buf = buf.split(b"\n\n", 1)
-            if not event.startswith(b"data: "):
-                continue
-            data = event[len(b"data: "):]
-            if data.strip() == b"[DONE]":
-                continue
-            try:
-                obj = json.loads(data)
-            except Exception:
-                continue
-            # usage arrives in the final chunk with empty choices
-            # (handle_response.go:113-133)
-            if obj.get("usage") and not obj.get("choices"):
-                self._account(obj["usage"], state)
+        while b"\n" in buf:
+            line, buf = buf.split(b"\n", 1)
+            line = line.rstrip(b"\r")
+            if line.startswith(b"data:"):
+                state.setdefault("sse_data", []).append(
+                    line[len(b"data:"):].lstrip())
+            elif line == b"":
+                parts = state.pop("sse_data", None)
+                if not parts:
+                    continue
+                data = b"\n".join(parts)
+                if data.strip() == b"[DONE]":
+                    continue
+                try:
+                    obj = json.loads(data)
+                except Exception:
+                    continue
+                # usage arrives in the final chunk with empty choices
+                # (handle_response.go:113-133)
+                if obj.get("usage") and not obj.get("choices"):
+                    self._account(obj["usage"], state)
         state["buffer"] = buf
 
     def _account(self, usage: dict, state: dict) -> None:

@@ -260,3 +260,30 @@ def test_pb_roundtrip():
     assert back.request_headers.end_of_stream
     body = ProcessingRequest(request_body=HttpBody(body=b"xyz"))
     assert ProcessingRequest.decode(body.encode()).request_body.body == b"xyz"
+
+
+def test_streamed_usage_crlf_and_fragmented(stack):
+    """SSE with CRLF line endings, a multi-line data field, and the usage
+    chunk fragmented mid-line across two ext_proc body messages must still
+    account (spec-correct SSE parsing, not naive split-on-\\n\\n)."""
+    chan, limiter, quota, _ = stack
+    body = json.dumps({"model": "m1", "stream": True,
+                       "stream_options": {"include_usage": True}}).encode()
+    chunk1 = b'data: {"choices":[{"delta":{"content":"hi"}}]}\r\n\r\n'
+    final = (b'data: {"choices":[],"usage":{"prompt_tokens":5,\r\n'
+             b'data: "completion_tokens":6,"total_tokens":11}}\r\n\r\n')
+    # split the final event mid-line across two messages
+    cut = 37
+    msgs = [
+        _headers_msg([("authorization", "Bearer sk-test-1")]),
+        ProcessingRequest(request_body=HttpBody(body=body, end_of_stream=True)),
+        _resp_headers_msg("200"),
+        ProcessingRequest(response_body=HttpBody(body=chunk1)),
+        ProcessingRequest(response_body=HttpBody(body=final[:cut])),
+        ProcessingRequest(response_body=HttpBody(body=final[cut:])),
+        ProcessingRequest(response_body=HttpBody(
+            body=b"data: [DONE]\r\n\r\n", end_of_stream=True)),
+    ]
+    out = drive(chan, msgs)
+    assert all(r.immediate_response is None for r in out)
+    assert quota.get_usage("default", "q1", "total") == 11
