@@ -48,6 +48,7 @@ from .extproc_pb import (
 from .limiter import RULES, RateLimiter, TYPE_REQUEST, TYPE_TOKEN
 from .provider import ConfigProvider
 from .quota import QuotaService
+from .sse import SSEUsageScanner
 
 log = logging.getLogger("arks.gateway.extproc")
 
@@ -206,34 +207,14 @@ class ExtProcServicer:
         return ProcessingResponse(response_body=CommonResponse())
 
     def _scan_sse(self, chunk: bytes, state: dict) -> None:
-        # Spec-correct SSE line parser (the reference uses openai-go's
-        # ssestream): CRLF endings tolerated, multi-line `data:` fields
-        # joined with \n, comments/other fields ignored. A naive
-        # split-on-"\n\n" misses events from runtimes that emit \r\n and
-        # silently drops their usage accounting.
-        buf = state["buffer"] + chunk
-        while b"\n" in buf:
-            line, buf = buf.split(b"\n", 1)
-            line = line.rstrip(b"\r")
-            if line.startswith(b"data:"):
-                state.setdefault("sse_data", []).append(
-                    line[len(b"data:"):].lstrip())
-            elif line == b"":
-                parts = state.pop("sse_data", None)
-                if not parts:
-                    continue
-                data = b"\n".join(parts)
-                if data.strip() == b"[DONE]":
-                    continue
-                try:
-                    obj = json.loads(data)
-                except Exception:
-                    continue
-                # usage arrives in the final chunk with empty choices
-                # (handle_response.go:113-133)
-                if obj.get("usage") and not obj.get("choices"):
-                    self._account(obj["usage"], state)
-        state["buffer"] = buf
+        # shared spec-correct scanner (gateway/sse.py): CRLF endings and
+        # arbitrary fragmentation tolerated — a naive split on blank lines
+        # silently dropped usage from upstreams emitting CRLF
+        sc = state.get("sse_scanner")
+        if sc is None:
+            sc = state["sse_scanner"] = SSEUsageScanner(
+                lambda usage: self._account(usage, state))
+        sc.feed(chunk)
 
     def _account(self, usage: dict, state: dict) -> None:
         qos = state["qos"]
