@@ -125,19 +125,43 @@ def create_router_app(state: RouterState) -> FastAPI:
         }
         if prefill:
             headers["x-arks-prefill-addr"] = _base(prefill)
+        # connect-failure failover: service discovery lags a dead worker by
+        # up to one watch/poll interval — retry distinct decode workers
+        # before surfacing 502 (the reference's router relies on the same
+        # discovery loop and has the same window)
+        tried: set[str] = set()
+        resp = client = None
+        for _ in range(min(3, max(len(state.decode_urls), 1))):
+            if decode in tried:
+                try:
+                    decode = state.pick_decode()
+                except LookupError:
+                    break
+                if decode in tried:
+                    continue
+            tried.add(decode)
+            state.acquire(decode)
+            client = httpx.AsyncClient(
+                transport=state.transport, base_url=_base(decode),
+                timeout=600.0)
+            req = client.build_request("POST", path, content=body_bytes,
+                                       headers=headers)
+            try:
+                resp = await client.send(req, stream=True)
+                break
+            except (httpx.ConnectError, httpx.ConnectTimeout):
+                state.release(decode)
+                await client.aclose()
+                resp = client = None
+            except Exception:
+                state.release(decode)
+                await client.aclose()
+                raise
+        if resp is None:
+            return JSONResponse(
+                status_code=502,
+                content={"error": "no reachable decode worker"})
         state.requests_total.labels(decode=decode, prefill=prefill or "").inc()
-        state.acquire(decode)
-        client = httpx.AsyncClient(
-            transport=state.transport, base_url=_base(decode), timeout=600.0
-        )
-        req = client.build_request("POST", path, content=body_bytes,
-                                   headers=headers)
-        try:
-            resp = await client.send(req, stream=True)
-        except Exception:
-            state.release(decode)
-            await client.aclose()
-            raise
 
         async def relay():
             try:

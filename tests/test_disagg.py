@@ -195,3 +195,52 @@ def test_router_cache_aware_affinity():
 
     picks = {state.pick_prefill(_prompt_key(body)) for _ in range(10)}
     assert len(picks) == 1  # same prompt -> same prefill worker every time
+
+
+def test_router_fails_over_dead_decode_worker():
+    """A decode worker that refuses connections (service discovery hasn't
+    caught up) is skipped: the request lands on a live worker; with every
+    worker dead the router answers 502, not a raw 500."""
+    import asyncio
+
+    import httpx
+    from fastapi import FastAPI
+    from fastapi.responses import JSONResponse
+
+    from arks_amd.router.app import RouterState, create_router_app
+
+    live = FastAPI()
+
+    @live.post("/v1/chat/completions")
+    async def chat():
+        return JSONResponse({"ok": True, "served_by": "live"})
+
+    live_transport = httpx.ASGITransport(app=live)
+
+    class FlakyTransport(httpx.AsyncBaseTransport):
+        async def handle_async_request(self, request):
+            if request.url.host == "dead":
+                raise httpx.ConnectError("refused", request=request)
+            return await live_transport.handle_async_request(request)
+
+    state = RouterState([], ["http://dead", "http://live"],
+                        policy="round_robin")
+    state.transport = FlakyTransport()
+    app = create_router_app(state)
+    rt = httpx.ASGITransport(app=app)
+
+    async def go():
+        async with httpx.AsyncClient(transport=rt, base_url="http://r") as c:
+            # several requests: round_robin starts on either worker, all
+            # must succeed via failover
+            for _ in range(4):
+                r = await c.post("/v1/chat/completions", json={"messages": []})
+                assert r.status_code == 200 and r.json()["served_by"] == "live"
+        state.set_workers(decode_urls=["http://dead"])
+        async with httpx.AsyncClient(transport=rt, base_url="http://r") as c:
+            r = await c.post("/v1/chat/completions", json={"messages": []})
+            assert r.status_code == 502
+
+    asyncio.run(go())
+    # inflight counters fully released after failovers
+    assert all(v == 0 for v in state.inflight.values())
