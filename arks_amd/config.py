@@ -460,6 +460,9 @@ class EngineConfig:
     # penalties/logprobs — same surface as vLLM's ngram speculator)
     speculative: str | None = None
     num_speculative_tokens: int = 4  # max draft length per step
+    # speculative="draft": the draft model ("preset:<name>" or a checkpoint
+    # path; must share the target's tokenizer/vocab)
+    draft_model: str | None = None
     device: str = "cuda"
     dtype: str = "bfloat16"
     seed: int = 0

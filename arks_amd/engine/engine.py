@@ -43,10 +43,16 @@ class LLMEngine:
         # serving metrics
         self.total_prompt_tokens = 0
         self.total_output_tokens = 0
-        # speculative decoding (prompt-lookup) stats
-        self.spec_enabled = cfg.speculative == "ngram"
+        # speculative decoding (prompt-lookup or draft-model) stats
+        self.spec_enabled = cfg.speculative in ("ngram", "draft")
         self.spec_drafted_tokens = 0
         self.spec_accepted_tokens = 0
+        self.draft = None
+        if cfg.speculative == "draft":
+            from .draft import DraftRunner
+
+            self.draft = DraftRunner(cfg, self.model_cfg, self.runner.num_blocks,
+                                     self.runner.device, self.runner.dtype)
 
     @property
     def prefix_cache_stats(self) -> tuple[int, int]:
@@ -202,22 +208,45 @@ class LLMEngine:
         bs = self.cfg.block_size
         k = self.cfg.num_speculative_tokens
         drafts: list[list[int]] = []
-        for seq in sb.seqs:
-            d: list[int] = []
-            room = self.cfg.max_model_len - seq.num_tokens
-            if room > 0 and (eligible(seq) or eligible_sampled(seq)):
-                d = propose_ngram_cached(seq, min(k, room))
-            if d:
-                need = (
-                    BlockAllocator.blocks_needed(seq.num_tokens + len(d), bs)
-                    - len(seq.block_table)
-                )
-                if need > 0:
-                    if alloc.can_allocate(need):
-                        seq.block_table.extend(alloc.allocate(need))
-                    else:
-                        d = []  # no KV room — plain decode this step
-            drafts.append(d)
+        if self.draft is not None:
+            # draft-model proposal: pages must cover the drafted positions
+            # BEFORE propose() (the draft writes its mirrored KV there), so
+            # eligibility includes room for the full k and the allocation
+            # happens up front
+            cand: list = []
+            for seq in sb.seqs:
+                ok = (seq.num_tokens + k <= self.cfg.max_model_len
+                      and (eligible(seq) or eligible_sampled(seq)))
+                if ok:
+                    need = (BlockAllocator.blocks_needed(seq.num_tokens + k, bs)
+                            - len(seq.block_table))
+                    if need > 0:
+                        if alloc.can_allocate(need):
+                            seq.block_table.extend(alloc.allocate(need))
+                        else:
+                            ok = False
+                cand.append(ok)
+            picked = [s for s, ok in zip(sb.seqs, cand) if ok]
+            proposed = self.draft.propose(picked, k) if picked else []
+            it = iter(proposed)
+            drafts = [next(it) if ok else [] for ok in cand]
+        else:
+            for seq in sb.seqs:
+                d: list[int] = []
+                room = self.cfg.max_model_len - seq.num_tokens
+                if room > 0 and (eligible(seq) or eligible_sampled(seq)):
+                    d = propose_ngram_cached(seq, min(k, room))
+                if d:
+                    need = (
+                        BlockAllocator.blocks_needed(seq.num_tokens + len(d), bs)
+                        - len(seq.block_table)
+                    )
+                    if need > 0:
+                        if alloc.can_allocate(need):
+                            seq.block_table.extend(alloc.allocate(need))
+                        else:
+                            d = []  # no KV room — plain decode this step
+                drafts.append(d)
         if not any(drafts):
             # nothing to verify: take the normal (hipGraph) decode path
             emitted = [[t] for t in self.runner.execute(sb)]
@@ -232,6 +261,10 @@ class LLMEngine:
             self.spec_drafted_tokens += len(drafts[i])
             if drafts[i]:
                 self.spec_accepted_tokens += len(toks) - 1
+                if self.draft is not None:
+                    # accepted drafted positions already hold valid draft
+                    # KV; the correction/bonus token catches up next round
+                    seq._draft_len += len(toks) - 1
             for tok in toks:
                 seq.append_token(tok)
                 self.total_output_tokens += 1

@@ -240,6 +240,7 @@ class Scheduler:
         if seq.block_table:
             self.allocator.free(seq.block_table)
             seq.block_table = []
+        seq._draft_len = 0  # draft-model KV mirror follows the pages
 
     def free_finished(self) -> None:
         for seq in self.running:

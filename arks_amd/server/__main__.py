@@ -36,8 +36,12 @@ def parse_args(argv=None):
                    default=None)
     p.add_argument("--quantization", choices=["fp8"], default=None)
     p.add_argument("--kv-cache-dtype", choices=["auto", "fp8"], default="auto")
-    p.add_argument("--speculative", choices=["ngram"], default=None,
-                   help="prompt-lookup speculative decoding (greedy requests)")
+    p.add_argument("--speculative", choices=["ngram", "draft"], default=None,
+                   help="speculative decoding: ngram (prompt lookup) or "
+                        "draft (a smaller draft model, --draft-model)")
+    p.add_argument("--draft-model", default=None,
+                   help="draft checkpoint path (or preset:<name>); must "
+                        "share the target tokenizer")
     p.add_argument("--num-speculative-tokens", type=int, default=4)
     p.add_argument("--no-prefix-cache", action="store_true",
                    help="disable the content-addressed prefix/radix cache")
@@ -70,6 +74,7 @@ def build_engine_config(args):
         quantization=args.quantization,
         kv_cache_dtype=args.kv_cache_dtype,
         speculative=args.speculative,
+        draft_model=args.draft_model,
         num_speculative_tokens=args.num_speculative_tokens,
         enable_prefix_caching=not args.no_prefix_cache,
     )

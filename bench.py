@@ -43,7 +43,9 @@ def parse_args():
     # opt-in: prompt-lookup speculative decoding (random-token synthetic
     # prompts rarely repeat, so this mostly measures the no-draft overhead;
     # acceptance-driven wins need repetitive real text)
-    p.add_argument("--speculative", choices=["ngram"], default=None)
+    p.add_argument("--speculative", choices=["ngram", "draft"], default=None)
+    p.add_argument("--draft-model", default=None,
+                   help="draft checkpoint/preset for --speculative draft")
     # BASELINE's "model cold-start sec": write a random-init safetensors
     # checkpoint to disk once, drop the page cache, and time disk -> HBM
     # through the pinned double-buffer loader (not the synthetic
@@ -158,6 +160,7 @@ def main():
         quantization=args.quantization,
         kv_cache_dtype=args.kv_cache_dtype,
         speculative=args.speculative,
+        draft_model=args.draft_model,
     )
     t_load0 = time.time()
     engine = LLMEngine(cfg)
