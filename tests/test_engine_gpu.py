@@ -387,3 +387,29 @@ def test_smollm3_nope_gpu_graph_matches_eager():
         assert all(len(o) == 12 for o in eager)
     finally:
         C.PRESET_CONFIGS.pop("tiny-gpu-smollm3", None)
+
+
+def test_draft_model_speculation_gpu():
+    """Draft-model speculation on the native kernels: deterministic,
+    drafts flow, bookkeeping holds (the preset draft is random-init with a
+    different seed, so acceptance may be low — exactness is pinned on the
+    CPU fp32 path in tests/test_spec_draft.py)."""
+    prompts = [[1, 2, 3, 4] * 8, [9, 31, 7, 2, 55, 14, 3], [5, 6] * 12]
+    sp = SamplingParams(max_tokens=16, ignore_eos=True)
+
+    def run():
+        torch.manual_seed(0)
+        e = LLMEngine(EngineConfig(
+            preset="tiny-gpu", device="cuda", kv_cache_blocks=512,
+            max_model_len=1024, max_num_seqs=64, speculative="draft",
+            draft_model="preset:tiny-gpu",
+        ))
+        return e.generate(prompts, sp), e
+
+    out1, eng = run()
+    out2, _ = run()
+    assert out1 == out2, "draft speculation must be deterministic"
+    assert [len(o) for o in out1] == [16, 16, 16]
+    assert eng.spec_drafted_tokens > 0
+    assert 0 <= eng.spec_accepted_tokens <= eng.spec_drafted_tokens
+    assert eng.total_output_tokens == 48
