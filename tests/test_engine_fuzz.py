@@ -49,12 +49,13 @@ def audit(engine: LLMEngine) -> None:
         min_size=4, max_size=30,
     ),
 )
-@pytest.mark.parametrize("speculative", [None, "ngram"])
+@pytest.mark.parametrize("speculative", [None, "ngram", "draft"])
 def test_engine_invariants_under_random_ops(actions, speculative):
     eng = LLMEngine(EngineConfig(
         preset="tiny", device="cpu", kv_cache_blocks=24, max_model_len=96,
         max_num_batched_tokens=48, max_num_seqs=8, seed=5,
         speculative=speculative,
+        draft_model="preset:tiny" if speculative == "draft" else None,
     ))
     rid = 0
     for kind, a, b, c in actions:
