@@ -19,7 +19,7 @@ def _free_port() -> int:
     return port
 
 
-def test_bench_dp2_gloo_contract():
+def _run_bench(mode: str):
     repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
     env = dict(os.environ, MASTER_ADDR="127.0.0.1")
     cmd = [
@@ -29,9 +29,14 @@ def test_bench_dp2_gloo_contract():
         os.path.join(repo, "bench.py"),
         "--gpus", "2", "--steps", "3", "--warmup", "1",
         "--batch", "2", "--input-len", "32", "--model", "tiny",
+        "--parallel", mode,
     ]
-    out = subprocess.run(cmd, cwd=repo, env=env, capture_output=True,
-                         text=True, timeout=420)
+    return subprocess.run(cmd, cwd=repo, env=env, capture_output=True,
+                          text=True, timeout=420)
+
+
+def test_bench_dp2_gloo_contract():
+    out = _run_bench("dp")
     assert out.returncode == 0, f"bench failed:\n{out.stdout}\n{out.stderr}"
     lines = [ln for ln in out.stdout.splitlines()
              if ln.startswith("{") and '"metric"' in ln]
@@ -49,3 +54,19 @@ def test_bench_dp2_gloo_contract():
     expected_tokens = 2 * 2 * 3
     approx_tokens = rec["value"] * (rec["ms_per_step"] * 3 / 1000.0)
     assert abs(approx_tokens - expected_tokens) / expected_tokens < 0.05
+
+
+def test_bench_tp2_gloo_contract():
+    """The TP bench path (one engine sharded over both ranks): rank 0
+    emits the JSON line with strong scaling and un-multiplied tokens."""
+    out = _run_bench("tp")
+    assert out.returncode == 0, f"bench failed:\n{out.stdout}\n{out.stderr}"
+    lines = [ln for ln in out.stdout.splitlines()
+             if ln.startswith("{") and '"metric"' in ln]
+    assert len(lines) == 1, f"expected exactly one JSON line:\n{out.stdout}"
+    rec = json.loads(lines[0])
+    assert rec["n_gpus"] == 2
+    assert rec["scaling"] == "strong"
+    assert rec["config"]["parallelism"] == "tp2"
+    assert rec["config"]["global_batch"] == 2
+    assert rec["value"] > 0
