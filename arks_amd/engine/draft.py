@@ -54,12 +54,11 @@ class DraftRunner:
                 "draft-model speculation with sliding-window page dropping "
                 "is not supported")
         tp = get_tp_world_size()
-        if tp > 1:
-            # the draft forward runs outside the SPMD runner broadcast
-            # protocol; a sharded draft would deadlock worker ranks
-            raise ValueError(
-                "speculative='draft' is single-GPU for now (TP=1); "
-                "use speculative='ngram' with TP")
+        # TP: the engine runs SPMD (every rank executes _spec_step), so the
+        # sharded draft forward's collectives stay in lockstep and greedy
+        # proposals are rank-identical (tests/test_tp_gloo.py TP2 case)
+        if dcfg.num_key_value_heads % max(tp, 1) != 0:
+            raise ValueError("draft kv heads not divisible by TP degree")
         self.cfg = dcfg
         self.device = device
         self.block_size = engine_cfg.block_size

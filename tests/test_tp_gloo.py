@@ -31,7 +31,7 @@ def _tp1_reference(preset="tiny", quantization=None):
 
 
 def _tp_worker(rank: int, world: int, port: int, q, preset: str = "tiny",
-               quantization=None, speculative=None):
+               quantization=None, speculative=None, draft_model=None):
     os.environ["MASTER_ADDR"] = "127.0.0.1"
     os.environ["MASTER_PORT"] = str(port)
     os.environ["RANK"] = str(rank)
@@ -47,7 +47,7 @@ def _tp_worker(rank: int, world: int, port: int, q, preset: str = "tiny",
             EngineConfig(
                 preset=preset, device="cpu", kv_cache_blocks=128,
                 max_model_len=512, quantization=quantization,
-                speculative=speculative,
+                speculative=speculative, draft_model=draft_model,
             )
         )
         out = e.generate(PROMPTS, SamplingParams(max_tokens=MAX_TOKENS, ignore_eos=True))
@@ -308,3 +308,26 @@ def test_tp2_speculative_matches_tp1_plain():
         p.join(timeout=60)
     assert status == "ok", payload
     assert payload == ref, f"TP=2 spec output {payload} != TP=1 {ref}"
+
+
+@pytest.mark.timeout(180)
+def test_tp2_draft_speculation_matches_tp1():
+    """Draft-model speculation under TP: the sharded draft forward runs
+    SPMD on every rank (its all-reduces keep lockstep), proposals are
+    rank-identical, outputs match the plain TP=1 engine."""
+    ref = _tp1_reference()
+
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    port = _free_port()
+    procs = [ctx.Process(target=_tp_worker,
+                         args=(r, 2, port, q, "tiny", None, "draft",
+                               "preset:tiny"))
+             for r in range(2)]
+    for p in procs:
+        p.start()
+    status, payload = q.get(timeout=150)
+    for p in procs:
+        p.join(timeout=60)
+    assert status == "ok", payload
+    assert payload == ref, f"TP=2 draft-spec {payload} != TP=1 {ref}"
