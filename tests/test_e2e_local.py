@@ -179,3 +179,112 @@ spec:
       rateLimits:
         - {type: rpm, value: 1000}
 """
+
+
+MULTITENANT = """
+apiVersion: arks.ai/v1
+kind: ArksQuota
+metadata: {name: team-a-quota, namespace: default}
+spec:
+  quotas:
+    - {type: total, value: 4}
+---
+apiVersion: arks.ai/v1
+kind: ArksToken
+metadata: {name: team-a, namespace: default}
+spec:
+  token: sk-team-a
+  qos:
+    - arksEndpoint: {name: tiny-chat}
+      rateLimits:
+        - {type: rpm, value: 1000}
+      quota: {name: team-a-quota}
+---
+apiVersion: arks.ai/v1
+kind: ArksToken
+metadata: {name: team-b, namespace: default}
+spec:
+  token: sk-team-b
+  qos:
+    - arksEndpoint: {name: tiny-chat}
+      rateLimits:
+        - {type: rpm, value: 1000}
+"""
+
+
+@pytest.mark.timeout(120)
+def test_multitenant_quota_isolation_over_sockets():
+    """BASELINE config #5 shape: two tenants share one application through
+    the real gateway; team-a's tiny quota exhausts after one request (429
+    with x-error-* headers), team-b keeps working."""
+    store = Store()
+    op = Operator(store)
+    for d in yaml.safe_load_all(TINY_STACK + "---" + MULTITENANT):
+        if d:
+            store.apply(parse_manifest(d))
+    op.reconcile_until_stable()
+    from arks_amd.crd.types import ModelPhase
+
+    m = store.get("ArksModel", "default", "tiny-model")
+    m.status.phase = ModelPhase.READY
+    store.apply(m)
+    op.reconcile_until_stable()
+    for kind in ("RoleBasedGroupSet", "LeaderWorkerSet"):
+        for wl in store.list(kind, "default"):
+            wl["status"] = {"replicas": 1, "readyReplicas": 1,
+                            "updatedReplicas": 1}
+            store.update(wl)
+            op._queue.put(("ArksApplication", "default",
+                           wl["metadata"]["name"]))
+    op.reconcile_until_stable()
+
+    from arks_amd.server.api import create_app as create_server_app
+    from arks_amd.server.async_engine import AsyncEngine
+    from arks_amd.server.tokenizer import ByteTokenizer
+
+    ecfg = EngineConfig(preset="tiny", device="cpu", kv_cache_blocks=128,
+                        max_model_len=256)
+    mc = ecfg.model_config()
+    tok = ByteTokenizer(mc.vocab_size, mc.eos_token_id)
+    engine = AsyncEngine(ecfg, model_name="tiny-chat")
+    sapp = create_server_app(engine, "tiny-chat", tok)
+    eport = _free_port()
+    es = _serve(sapp, eport)
+
+    from arks_amd.gateway import BackendResolver
+    from arks_amd.gateway.app import create_gateway_app
+
+    resolver = BackendResolver(
+        store, url_for_service=lambda ns, svc: f"http://127.0.0.1:{eport}")
+    gw = create_gateway_app(store, resolver=resolver)
+    gport = _free_port()
+    gs = _serve(gw, gport)
+
+    def ask(token: str):
+        body = json.dumps({
+            "model": "tiny-chat",
+            "messages": [{"role": "user", "content": "hello there"}],
+            "max_tokens": 4,
+        }).encode()
+        req = urllib.request.Request(
+            f"http://127.0.0.1:{gport}/v1/chat/completions", data=body,
+            headers={"Authorization": f"Bearer {token}",
+                     "Content-Type": "application/json"})
+        try:
+            with urllib.request.urlopen(req, timeout=60) as r:
+                return r.status, dict(r.headers)
+        except urllib.error.HTTPError as e:
+            return e.code, dict(e.headers)
+
+    # team-a: first request passes (usage 0 <= 4), accounting pushes it over
+    code, _ = ask("sk-team-a")
+    assert code == 200
+    code, hdrs = ask("sk-team-a")
+    assert code == 429
+    assert hdrs.get("x-error-type") == "quota"
+    # team-b: unaffected by team-a's quota
+    for _ in range(2):
+        code, _ = ask("sk-team-b")
+        assert code == 200
+    es.should_exit = True
+    gs.should_exit = True
