@@ -366,3 +366,37 @@ def test_passive_outlier_ejection():
     det.record("a", False); det.record("a", False); det.record("a", False)
     det.record("b", False); det.record("b", False); det.record("b", False)
     assert r.resolve("default", "m") in {"a", "b"}
+
+
+def test_backend_resolver_weight_proportional():
+    """defaultWeight routing (reference HTTPRoute backendRef weights): a
+    3:1 weighting splits traffic ~3:1."""
+    import random
+
+    from arks_amd.controlplane import Store
+    from arks_amd.gateway import BackendResolver
+
+    random.seed(7)
+    store = Store()
+    store.apply({
+        "apiVersion": "gateway.networking.k8s.io/v1",
+        "kind": "HTTPRoute",
+        "metadata": {"name": "m", "namespace": "default"},
+        "spec": {"rules": [{"backendRefs": [
+            {"name": "heavy", "weight": 3}, {"name": "light", "weight": 1},
+        ]}]},
+    })
+    r = BackendResolver(store, url_for_service=lambda ns, svc: svc)
+    n = 4000
+    heavy = sum(r.resolve("default", "m") == "heavy" for _ in range(n))
+    assert 0.70 < heavy / n < 0.80, f"heavy fraction {heavy / n}"
+    # zero-weight backends never picked
+    store.apply({
+        "apiVersion": "gateway.networking.k8s.io/v1",
+        "kind": "HTTPRoute",
+        "metadata": {"name": "z", "namespace": "default"},
+        "spec": {"rules": [{"backendRefs": [
+            {"name": "on", "weight": 1}, {"name": "off", "weight": 0},
+        ]}]},
+    })
+    assert {r.resolve("default", "z") for _ in range(50)} == {"on"}
