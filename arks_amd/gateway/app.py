@@ -118,9 +118,13 @@ class BackendResolver:
         self.outliers = outliers or OutlierDetector()
 
     @staticmethod
-    def _match_entry(m: dict, path: str, headers: dict[str, str]) -> bool:
-        """One HTTPRouteMatch: path AND all header conditions must hold
-        (gateway-api semantics)."""
+    def _match_entry(m: dict, path: str, headers: dict[str, str],
+                     method: str | None = None) -> bool:
+        """One HTTPRouteMatch: method AND path AND all header conditions
+        must hold (gateway-api semantics; absent fields match anything)."""
+        want = m.get("method")
+        if want and method and want.upper() != method.upper():
+            return False
         pm = m.get("path") or {}
         if pm:
             val = pm.get("value", "/")
@@ -137,7 +141,8 @@ class BackendResolver:
         return True
 
     def resolve(self, namespace: str, model: str, path: str = "/",
-                headers: dict[str, str] | None = None) -> str | None:
+                headers: dict[str, str] | None = None,
+                method: str | None = None) -> str | None:
         """Pick a backend from the endpoint's HTTPRoute, evaluating EVERY
         rule's match conditions (a rule matches when any of its match
         entries does — user MatchConfigs from the ArksEndpoint spec are
@@ -152,7 +157,8 @@ class BackendResolver:
         refs = None
         for rule in route.get("spec", {}).get("rules", []):
             matches = rule.get("matches") or [{}]
-            if any(self._match_entry(m, path, hdrs) for m in matches):
+            if any(self._match_entry(m, path, hdrs, method)
+                   for m in matches):
                 refs = rule.get("backendRefs", [])
                 if refs:
                     break
@@ -263,6 +269,7 @@ def create_gateway_app(
             qos.namespace, model, path=request.url.path,
             headers={"model": model, "namespace": qos.namespace,
                      "username": qos.user},
+            method=request.method,
         )
         if base is None:
             metrics.requests_total.labels(**labels, status="503").inc()

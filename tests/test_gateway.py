@@ -400,3 +400,30 @@ def test_backend_resolver_weight_proportional():
         ]}]},
     })
     assert {r.resolve("default", "z") for _ in range(50)} == {"on"}
+
+
+def test_backend_resolver_method_matching():
+    """gateway-api method matching: a rule restricted to GET must not
+    capture POST traffic (reference MatchConfigs carry `method`)."""
+    from arks_amd.controlplane import Store
+    from arks_amd.gateway import BackendResolver
+
+    store = Store()
+    store.apply({
+        "apiVersion": "gateway.networking.k8s.io/v1",
+        "kind": "HTTPRoute",
+        "metadata": {"name": "m", "namespace": "default"},
+        "spec": {"rules": [
+            {"matches": [{"method": "GET",
+                          "path": {"type": "PathPrefix", "value": "/"}}],
+             "backendRefs": [{"name": "get-only", "weight": 1}]},
+            {"matches": [{"method": "POST",
+                          "path": {"type": "PathPrefix", "value": "/"}}],
+             "backendRefs": [{"name": "post-svc", "weight": 1}]},
+        ]},
+    })
+    r = BackendResolver(store, url_for_service=lambda ns, svc: svc)
+    assert r.resolve("default", "m", method="POST") == "post-svc"
+    assert r.resolve("default", "m", method="GET") == "get-only"
+    # method omitted by the caller: first rule wins (match-all fallback)
+    assert r.resolve("default", "m") == "get-only"
