@@ -465,3 +465,35 @@ def test_chat_stream_logprobs(app):
                        for e in entries)
 
     run_with_client(app, fn)
+
+
+def test_cli_flags_map_to_engine_config():
+    """The operator-composed command line (controlplane/commands.py contract)
+    parses into the right EngineConfig fields — including the speculative
+    and prefix-cache flags the docs reference."""
+    from arks_amd.server.__main__ import build_engine_config, parse_args
+
+    args = parse_args([
+        "--model", "/models/models/default/m", "--served-model-name", "m",
+        "--tensor-parallel-size", "2", "--port", "8080",
+        "--max-model-len", "4096", "--quantization", "fp8",
+        "--kv-cache-dtype", "fp8", "--speculative", "draft",
+        "--draft-model", "/models/models/default/m-draft",
+        "--num-speculative-tokens", "3", "--no-prefix-cache",
+    ])
+    cfg = build_engine_config(args)
+    assert cfg.model_path == "/models/models/default/m"
+    assert cfg.served_model_name == "m"
+    assert cfg.tensor_parallel_size == 2
+    assert cfg.max_model_len == 4096
+    assert cfg.quantization == "fp8"
+    assert cfg.kv_cache_dtype == "fp8"
+    assert cfg.speculative == "draft"
+    assert cfg.draft_model == "/models/models/default/m-draft"
+    assert cfg.num_speculative_tokens == 3
+    assert cfg.enable_prefix_caching is False
+    # preset form
+    args2 = parse_args(["--model", "preset:tiny"])
+    cfg2 = build_engine_config(args2)
+    assert cfg2.preset == "tiny" and cfg2.model_path is None
+    assert cfg2.enable_prefix_caching is True
