@@ -79,6 +79,7 @@ def worker_loop(cfg: EngineConfig) -> None:
 class _Stream:
     queue: asyncio.Queue = field(default_factory=asyncio.Queue)
     prev_token_time: float | None = None
+    arrival_time: float = field(default_factory=time.time)
 
 
 class AsyncEngine:
@@ -253,14 +254,10 @@ class AsyncEngine:
                 if st is None:
                     continue
                 if st.prev_token_time is None:
-                    # TTFT measured against request arrival
-                    seq = next(
-                        (s for s in self.engine.scheduler.running
-                         if s.request_id == out.request_id),
-                        None,
-                    )
-                    if seq is not None:
-                        self.metrics.ttft.observe(now - seq.arrival_time)
+                    # TTFT measured against request arrival (the stream's
+                    # own clock: a sequence that finishes on its first
+                    # token is already out of scheduler.running)
+                    self.metrics.ttft.observe(now - st.arrival_time)
                 else:
                     self.metrics.tpot.observe(now - st.prev_token_time)
                 st.prev_token_time = now
