@@ -39,3 +39,34 @@ def test_byte_tokenizer_roundtrip_ascii_and_eos():
     ids = TOK.encode("hello")
     assert TOK.decode(ids + [TOK.eos_token_id]) == "hello"  # eos stripped
     assert TOK.decode([]) == ""
+
+
+@settings(max_examples=300, deadline=None)
+@given(
+    text=st.text(alphabet="abcxyz", max_size=40),
+    stop=st.text(alphabet="abc", min_size=1, max_size=3),
+    cuts=st.lists(st.integers(1, 5), max_size=12),
+)
+def test_stop_tracker_streaming_equals_batch(text, stop, cuts):
+    """The streaming stop tracker (server/api.py StopStringTracker) must emit
+    exactly text.split(stop)[0] regardless of how the stream is chunked,
+    and never leak any part of the stop string."""
+    from arks_amd.server.api import StopStringTracker
+
+    tr = StopStringTracker([stop])
+    out, stopped = "", False
+    i = 0
+    for c in cuts + [len(text)]:
+        piece, hit = tr.feed(text[i:i + c])
+        out += piece
+        i += c
+        if hit:
+            stopped = True
+            break
+        if i >= len(text):
+            break
+    if not stopped:
+        out += tr.flush()
+    expect = text.split(stop)[0] if stop in text else text
+    assert out == expect
+    assert stopped == (stop in text)
